@@ -1,0 +1,59 @@
+"""Shared pytest configuration.
+
+Markers:
+- ``gpu``: requires an MI355X (run via gpurun / driver round-end on a GPU box)
+- ``slow``: long-running scale tests
+"""
+
+from __future__ import annotations
+
+import os
+
+import pytest
+
+
+def pytest_configure(config: pytest.Config) -> None:
+    config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
+    config.addinivalue_line("markers", "slow: long-running scale tests")
+
+
+def _has_gpu() -> bool:
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+HAS_GPU = _has_gpu()
+
+
+def pytest_collection_modifyitems(config: pytest.Config, items: list[pytest.Item]) -> None:
+    if HAS_GPU:
+        return
+    skip_gpu = pytest.mark.skip(reason="no GPU in this environment")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip_gpu)
+
+
+@pytest.fixture(autouse=True)
+def _reset_global_state():
+    """Snapshot/restore process-global mutable state between tests."""
+    from agentbom_amd.utils import version_utils
+
+    sink = version_utils._scan_warning_sink
+    yield
+    version_utils._scan_warning_sink = sink
+
+
+@pytest.fixture
+def env(monkeypatch):
+    """Helper to set AGENT_BOM_* env vars for one test."""
+
+    def _set(**kwargs):
+        for k, v in kwargs.items():
+            monkeypatch.setenv(k, str(v))
+
+    return _set
