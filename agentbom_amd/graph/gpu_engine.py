@@ -1,0 +1,353 @@
+"""GPU estate engine: CSR residency in HBM + the blast-radius pipeline.
+
+The estate graph (agents, servers, creds, tools, packages) is built as
+forward and reverse CSR on-device (torch sort/bincount/cumsum — GPU CSR
+construction) and stays resident in HBM3E.  On top of it:
+
+- ``match()``            bulk advisory matching (ops/csrc/match.hip)
+- ``dependency_reach()`` multi-source BFS from all agents (graph.hip)
+- ``blast_counts()``     segmented joins: per-finding distinct agent/cred/
+                         tool reach with CWE-impact filtering
+- ``score()``            GPU risk scoring (reference formula)
+- ``step()``             full findings pipeline: match -> reach -> join ->
+                         score -> rank (the bench.py timed region)
+- ``blast_radius_query()`` batched bounded BFS for p50 query latency /
+                         the /v1/graph read path
+
+Replaces the reference's scan_packages hot loop + graph dict walks
+(reference: src/agent_bom/scanners/package_scan.py:1307-2028,
+src/agent_bom/graph/{container,dependency_reach}.py) — see SURVEY.md §2.11.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+
+from agentbom_amd.models.cwe_impact import (
+    IMPACT_CODE,
+    IMPACT_CREDENTIAL_ACCESS,
+    IMPACT_CODE_EXECUTION,
+    IMPACT_DATA_LEAK,
+    IMPACT_FILE_ACCESS,
+    IMPACT_INJECTION,
+    IMPACT_SSRF,
+)
+from agentbom_amd.scan.synth import ET_CONTAINS, ET_HAS_CRED, ET_PROVIDES_TOOL, ET_USES, SyntheticEstate
+
+# Impact-code lookup tables (u8 code -> filter class), device-ready.
+_FULL_CRED = {IMPACT_CODE[c] for c in (IMPACT_CODE_EXECUTION, IMPACT_CREDENTIAL_ACCESS,
+                                       IMPACT_DATA_LEAK, IMPACT_FILE_ACCESS, IMPACT_SSRF)}
+_DB_CRED = {IMPACT_CODE[IMPACT_INJECTION]}
+_FULL_TOOL = _FULL_CRED
+_DB_TOOL = {IMPACT_CODE[IMPACT_INJECTION]}
+
+
+def _impact_lut(full: set, db: set):
+    """9-entry u8 LUT: 2 = full reach, 1 = db-only, 0 = none."""
+    lut = np.zeros(9, dtype=np.uint8)
+    for c in full:
+        lut[c] = 2
+    for c in db:
+        lut[c] = 1
+    return lut
+
+
+class EstateEngine:
+    """Device-resident estate + advisory arena + the findings pipeline."""
+
+    def __init__(self, estate: SyntheticEstate, device: str = "cuda"):
+        import torch
+
+        self.torch = torch
+        self.estate = estate
+        self.device = torch.device(device)
+        self.N = estate.num_nodes
+        dev = self.device
+
+        src = torch.from_numpy(estate.edge_src).to(dev)
+        dst = torch.from_numpy(estate.edge_dst).to(dev)
+        et = torch.from_numpy(estate.edge_type).to(dev)
+
+        self.fwd = self._build_csr(src, dst, et)
+        self.rev = self._build_csr(dst, src, et)
+
+        self.pkg_group_key = torch.from_numpy(estate.pkg_name_id.view(np.int64)).to(dev)
+        self.pkg_key_hi = torch.from_numpy(estate.pkg_key_hi.view(np.int64)).to(dev)
+        self.pkg_key_lo = torch.from_numpy(estate.pkg_key_lo.view(np.int64)).to(dev)
+        self.pkg_flags = torch.from_numpy(estate.pkg_flags).to(dev)
+
+        self.arena = estate.arena.to_torch(dev)
+        self.cred_is_db = torch.from_numpy(estate.cred_is_db).to(dev)
+        self.tool_is_db = torch.from_numpy(estate.tool_is_db).to(dev)
+        self.cred_lut = torch.from_numpy(_impact_lut(_FULL_CRED, _DB_CRED)).to(dev)
+        self.tool_lut = torch.from_numpy(_impact_lut(_FULL_TOOL, _DB_TOOL)).to(dev)
+
+        self.agent_ids = torch.arange(estate.n_agents, dtype=torch.int32, device=dev)
+        self._bfs_ws: dict = {}
+
+        # per-node db-credential / db-tool indicator (for impact filtering)
+        self.node_is_db_cred = torch.zeros(self.N, dtype=torch.uint8, device=dev)
+        self.node_is_db_cred[
+            estate.cred_base + torch.arange(estate.n_creds, device=dev)
+        ] = self.cred_is_db
+        self.node_is_db_tool = torch.zeros(self.N, dtype=torch.uint8, device=dev)
+        self.node_is_db_tool[
+            estate.tool_base + torch.arange(estate.n_tools, device=dev)
+        ] = self.tool_is_db
+
+    def _build_csr(self, src, dst, et):
+        """GPU CSR construction: stable sort by src, bincount offsets."""
+        torch = self.torch
+        order = torch.argsort(src, stable=True)
+        row_counts = torch.bincount(src, minlength=self.N)
+        row_off = torch.zeros(self.N + 1, dtype=torch.int64, device=src.device)
+        torch.cumsum(row_counts, 0, out=row_off[1:])
+        return {
+            "row_off": row_off,
+            "col": dst[order].to(torch.int32),
+            "etype": et[order].contiguous(),
+        }
+
+    # ── pipeline stages ────────────────────────────────────────────────────
+
+    @property
+    def use_gpu(self) -> bool:
+        return self.device.type == "cuda"
+
+    def match(self):
+        """(finding_pkg_idx, finding_win_idx) int64 device tensors."""
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            return native.match(
+                self.pkg_group_key, self.pkg_key_hi, self.pkg_key_lo, self.pkg_flags,
+                self.arena["group_keys"], self.arena["group_off"], self.arena["windows"],
+            )
+        # CPU oracle path (tests / dev boxes)
+        from agentbom_amd.ops import cpu_ref
+
+        torch = self.torch
+        win = {k: v.numpy().view(np.uint64) if v.dtype == torch.int64 else v.numpy()
+               for k, v in self.arena["windows"].items()}
+        pkg_i, win_i = cpu_ref.match(
+            self.pkg_group_key.numpy().view(np.uint64),
+            self.pkg_key_hi.numpy().view(np.uint64),
+            self.pkg_key_lo.numpy().view(np.uint64),
+            self.pkg_flags.numpy(),
+            self.arena["group_keys"].numpy().view(np.uint64),
+            self.arena["group_off"].numpy().view(np.uint32),
+            win,
+        )
+        return torch.from_numpy(pkg_i), torch.from_numpy(win_i)
+
+    def dependency_reach(self):
+        """u32 hop distance from the nearest agent, per node (multi-source BFS)."""
+        mask = (1 << ET_USES) | (1 << ET_CONTAINS) | (1 << ET_HAS_CRED) | (1 << ET_PROVIDES_TOOL)
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            return native.bfs(
+                self.fwd["row_off"], self.fwd["col"], self.agent_ids, self.N,
+                etype=self.fwd["etype"], allowed_mask=mask, workspace=self._bfs_ws,
+            )
+        from agentbom_amd.ops import cpu_ref
+
+        dist = cpu_ref.bfs(
+            self.fwd["row_off"].numpy(), self.fwd["col"].numpy(),
+            self.agent_ids.numpy(), self.N,
+            etype=self.fwd["etype"].numpy(), allowed_mask=mask,
+        )
+        return self.torch.from_numpy(dist.view(np.int32)).to(self.torch.int32)
+
+    def _expand(self, csr, rows, carry, want_type: int):
+        """Segmented gather: neighbors of ``rows`` with ``carry`` repeated.
+
+        Returns (neighbor_node, carried_value) filtered to edge type
+        ``want_type``.
+        """
+        torch = self.torch
+        row_off = csr["row_off"]
+        beg = row_off[rows]
+        cnt = row_off[rows + 1] - beg
+        total = int(cnt.sum().item())
+        if total == 0:
+            empty = torch.empty(0, dtype=torch.int64, device=self.device)
+            return empty, empty
+        rep = torch.repeat_interleave(torch.arange(rows.numel(), device=self.device), cnt)
+        cum = torch.zeros(rows.numel(), dtype=torch.int64, device=self.device)
+        torch.cumsum(cnt, 0, out=cum)
+        base = torch.cat([torch.zeros(1, dtype=torch.int64, device=self.device), cum[:-1]])
+        eidx = torch.arange(total, device=self.device) - base[rep] + beg[rep]
+        keep = csr["etype"][eidx] == want_type
+        nbr = csr["col"][eidx][keep].to(torch.int64)
+        return nbr, carry[rep[keep]]
+
+    def _distinct_counts(self, key_a, key_b, n_bins: int, weights=None):
+        """Count distinct (a, b) pairs per a; optional 0/1 weight per b kind."""
+        torch = self.torch
+        if key_a.numel() == 0:
+            z = torch.zeros(n_bins, dtype=torch.int32, device=self.device)
+            return z if weights is None else (z, z.clone())
+        pair = key_a * self.N + key_b
+        uniq = torch.unique(pair)
+        ua = uniq // self.N
+        ub = uniq % self.N
+        cnt = torch.bincount(ua, minlength=n_bins).to(torch.int32)
+        if weights is None:
+            return cnt
+        w = weights[ub].to(torch.int64)
+        cnt_w = torch.bincount(ua, weights=w.to(torch.float64), minlength=n_bins)
+        return cnt, cnt_w.to(torch.int32)
+
+    def blast_counts(self, finding_pkgs):
+        """Per unique finding-package reach counts via segmented joins.
+
+        Returns dict keyed by package node id -> n_servers/n_agents/
+        n_creds_all/n_creds_db/n_tools_all/n_tools_db as dense tensors
+        indexed by position in ``uniq_pkgs``.
+        """
+        torch = self.torch
+        uniq_pkgs = torch.unique(finding_pkgs)  # node ids (pkg space)
+        # position index for dense bincounts
+        n_up = uniq_pkgs.numel()
+        pos = torch.arange(n_up, device=self.device)
+
+        # package -> servers (reverse CONTAINS)
+        srv, carry = self._expand(self.rev, uniq_pkgs, pos, ET_CONTAINS)
+        n_servers = self._distinct_counts(carry, srv, n_up)
+
+        # servers -> agents (reverse USES), carry package position
+        ag, carry2 = self._expand(self.rev, srv, carry, ET_USES)
+        n_agents = self._distinct_counts(carry2, ag, n_up)
+
+        # servers -> creds / tools (forward)
+        cr, carry3 = self._expand(self.fwd, srv, carry, ET_HAS_CRED)
+        n_creds_all, n_creds_db = self._distinct_counts(
+            carry3, cr, n_up, weights=self.node_is_db_cred
+        )
+        tl, carry4 = self._expand(self.fwd, srv, carry, ET_PROVIDES_TOOL)
+        n_tools_all, n_tools_db = self._distinct_counts(
+            carry4, tl, n_up, weights=self.node_is_db_tool
+        )
+        return {
+            "uniq_pkgs": uniq_pkgs,
+            "n_servers": n_servers,
+            "n_agents": n_agents,
+            "n_creds_all": n_creds_all,
+            "n_creds_db": n_creds_db,
+            "n_tools_all": n_tools_all,
+            "n_tools_db": n_tools_db,
+        }
+
+    def step(self, blast_depth: int = 1, reach_dist=None):
+        """One full findings pass.  Returns summary dict.
+
+        ``reach_dist`` overrides the locally-computed dependency-reach
+        distances (multi-GPU mode passes the shard's slice of the
+        distributed BFS result)."""
+        torch = self.torch
+        pkg_idx, win_idx = self.match()
+        n_findings = pkg_idx.numel()
+
+        dist = reach_dist if reach_dist is not None else self.dependency_reach()
+
+        pkg_nodes = pkg_idx + self.estate.pkg_base
+        counts = self.blast_counts(pkg_nodes)
+
+        # map each finding to its unique-package position
+        pos = torch.searchsorted(counts["uniq_pkgs"], pkg_nodes)
+
+        # CWE-impact filtering via LUTs: 2=full, 1=db-only, 0=none
+        impact = self.arena["impact"].to(torch.int64)[win_idx]
+        cred_cls = self.cred_lut.to(torch.int64)[impact]
+        tool_cls = self.tool_lut.to(torch.int64)[impact]
+        n_creds = torch.where(
+            cred_cls == 2, counts["n_creds_all"][pos],
+            torch.where(cred_cls == 1, counts["n_creds_db"][pos],
+                        torch.zeros_like(pos, dtype=torch.int32)),
+        )
+        n_tools = torch.where(
+            tool_cls == 2, counts["n_tools_all"][pos],
+            torch.where(tool_cls == 1, counts["n_tools_db"][pos],
+                        torch.zeros_like(pos, dtype=torch.int32)),
+        )
+        n_agents = counts["n_agents"][pos]
+
+        sev = self.arena["severity"][win_idx]
+        kev = self.arena["kev"][win_idx]
+        epss = self.arena["epss"][win_idx]
+        flags = (kev.to(torch.uint8) << 1)  # bit1 kev; no ai_ctx/suppressed in synth
+        scorecard = torch.full((n_findings,), -1.0, dtype=torch.float32, device=self.device)
+        reach_known = dist[pkg_nodes] != 0xFFFFFFFF
+        reach = torch.where(
+            reach_known,
+            torch.ones(n_findings, dtype=torch.int8, device=self.device),
+            torch.zeros(n_findings, dtype=torch.int8, device=self.device),
+        )
+
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            scores = native.risk_score(
+                sev.contiguous(), n_agents.to(torch.int32).contiguous(),
+                n_creds.to(torch.int32).contiguous(), n_tools.to(torch.int32).contiguous(),
+                flags.contiguous(), epss.contiguous(), scorecard, reach.contiguous(),
+            )
+        else:
+            from agentbom_amd.ops import cpu_ref
+
+            scores = torch.from_numpy(
+                cpu_ref.risk_score(
+                    sev.numpy(), n_agents.numpy().astype(np.uint32),
+                    n_creds.numpy().astype(np.uint32), n_tools.numpy().astype(np.uint32),
+                    flags.numpy(), epss.numpy(), scorecard.numpy(), reach.numpy(),
+                )
+            )
+
+        # deterministic rank: score desc, finding index asc
+        order = torch.argsort(scores, descending=True, stable=True)
+        return {
+            "n_findings": int(n_findings),
+            "scores": scores,
+            "order": order,
+            "pkg_idx": pkg_idx,
+            "win_idx": win_idx,
+            "n_agents": n_agents,
+            "n_creds": n_creds,
+            "n_tools": n_tools,
+            "reach_dist": dist,
+        }
+
+    def blast_radius_query(self, node_ids, max_hops: int = 4, max_nodes: int = 4096):
+        """Bounded blast-radius neighborhoods (reverse direction: who reaches
+        this package / what this vuln exposes)."""
+        q = node_ids.to(self.torch.int32)
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            return native.impact_query(
+                self.rev["row_off"], self.rev["col"], q, etype=None,
+                allowed_mask=0xFFFFFFFF, max_hops=max_hops, max_nodes=max_nodes,
+            )
+        from agentbom_amd.ops import cpu_ref
+
+        torch = self.torch
+        results = cpu_ref.impact_query(
+            self.rev["row_off"].numpy(), self.rev["col"].numpy(), q.numpy(),
+            max_hops=max_hops, max_nodes=max_nodes,
+        )
+        Q = q.numel()
+        out_nodes = torch.zeros((Q, max_nodes), dtype=torch.int32)
+        out_hops = torch.zeros((Q, max_nodes), dtype=torch.uint8)
+        out_counts = torch.zeros(Q, dtype=torch.int32)
+        out_trunc = torch.zeros(Q, dtype=torch.uint8)
+        for qi, (hops, trunc) in enumerate(results):
+            items = list(hops.items())[:max_nodes]
+            for j, (node, hop) in enumerate(items):
+                out_nodes[qi, j] = node
+                out_hops[qi, j] = hop
+            out_counts[qi] = len(items)
+            out_trunc[qi] = 1 if trunc else 0
+        return out_nodes, out_hops, out_counts, out_trunc

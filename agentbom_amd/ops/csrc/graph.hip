@@ -1,0 +1,427 @@
+// Estate-graph kernels: level-synchronous multi-source BFS over CSR in HBM,
+// batched bounded blast-radius queries, and the risk-score formula.
+//
+// Replaces the reference's Python adjacency-dict walks
+// (reference: src/agent_bom/graph/dependency_reach.py:109-198 per-agent BFS,
+// src/agent_bom/graph/container.py:613-683 BFS/impact_of/traverse_subgraph)
+// with frontier kernels over a CSR whose 10M-100M nodes stay resident in
+// 288 GB of HBM3E.
+//
+// BFS design (CDNA4):
+// - one thread per frontier vertex, grid-stride; dist claims via atomicCAS
+//   from UNVISITED so each vertex is pushed exactly once (level-synchronous
+//   => first touch is the minimal hop).
+// - edge-type filtering via a 32-bit allowed-mask over per-edge type bytes
+//   (the reference traverses only USES/DEPENDS_ON/CONTAINS/PROVIDES_TOOL
+//   classes for dependency reach — types are a closed enum).
+// - high-degree frontier vertices (skewed estates: ~1% of agents fan out
+//   18-32x) are handled wave-cooperatively: degree >= 64 vertices are
+//   expanded by whole waves from a secondary queue filled in pass 1.
+//
+// The host-side level loop lives in abom_api.hip (C++, one sync per level).
+
+#include "abom_common.h"
+
+namespace abom {
+
+// Pass over the frontier: expand vertices with degree < WAVE_DEG inline;
+// defer heavy vertices to the heavy queue.
+__global__ void bfs_expand_kernel(
+    const uint64_t* __restrict__ row_off,   // [N+1]
+    const uint32_t* __restrict__ col,       // [E]
+    const uint8_t* __restrict__ etype,      // [E] or nullptr
+    uint32_t allowed_mask,
+    const uint32_t* __restrict__ frontier,
+    long long frontier_size,
+    uint32_t* __restrict__ dist,
+    uint32_t next_level,
+    uint32_t* __restrict__ next_frontier,
+    unsigned int* __restrict__ next_count,
+    uint32_t* __restrict__ heavy_queue,
+    unsigned int* __restrict__ heavy_count,
+    long long capacity) {
+    constexpr uint64_t WAVE_DEG = 64;
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < frontier_size;
+         i += stride) {
+        const uint32_t u = frontier[i];
+        const uint64_t beg = row_off[u];
+        const uint64_t end = row_off[u + 1];
+        if (end - beg >= WAVE_DEG) {
+            heavy_queue[atomicAdd(heavy_count, 1u)] = u;
+            continue;
+        }
+        for (uint64_t e = beg; e < end; ++e) {
+            if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
+            const uint32_t v = col[e];
+            if (dist[v] == ABOM_UNVISITED &&
+                atomicCAS(&dist[v], ABOM_UNVISITED, next_level) == ABOM_UNVISITED) {
+                const unsigned idx = atomicAdd(next_count, 1u);
+                if ((long long)idx < capacity) next_frontier[idx] = v;
+            }
+        }
+    }
+}
+
+// Wave-cooperative expansion of heavy vertices: one wave (64 lanes) walks one
+// vertex's adjacency with coalesced col[] reads.
+__global__ void bfs_expand_heavy_kernel(
+    const uint64_t* __restrict__ row_off,
+    const uint32_t* __restrict__ col,
+    const uint8_t* __restrict__ etype,
+    uint32_t allowed_mask,
+    const uint32_t* __restrict__ heavy_queue,
+    const unsigned int* __restrict__ heavy_size_ptr,
+    uint32_t* __restrict__ dist,
+    uint32_t next_level,
+    uint32_t* __restrict__ next_frontier,
+    unsigned int* __restrict__ next_count,
+    long long capacity) {
+    const unsigned heavy_size = *heavy_size_ptr;
+    const int lane = threadIdx.x & 63;
+    const long long wave = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const long long nwaves = ((long long)gridDim.x * blockDim.x) >> 6;
+    for (long long q = wave; q < heavy_size; q += nwaves) {
+        const uint32_t u = heavy_queue[q];
+        const uint64_t beg = row_off[u];
+        const uint64_t end = row_off[u + 1];
+        for (uint64_t e = beg + lane; e < end; e += 64) {
+            if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
+            const uint32_t v = col[e];
+            if (dist[v] == ABOM_UNVISITED &&
+                atomicCAS(&dist[v], ABOM_UNVISITED, next_level) == ABOM_UNVISITED) {
+                const unsigned idx = atomicAdd(next_count, 1u);
+                if ((long long)idx < capacity) next_frontier[idx] = v;
+            }
+        }
+    }
+}
+
+__global__ void init_dist_kernel(uint32_t* __restrict__ dist, long long n, uint32_t value) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        dist[i] = value;
+}
+
+__global__ void seed_sources_kernel(
+    const uint32_t* __restrict__ sources, long long n_sources,
+    uint32_t* __restrict__ dist, uint32_t* __restrict__ frontier) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n_sources;
+         i += stride) {
+        const uint32_t s = sources[i];
+        dist[s] = 0;
+        frontier[i] = s;
+    }
+}
+
+// ── Batched bounded blast-radius query (impact_of / traverse_subgraph) ─────
+//
+// One BLOCK per query: a small BFS bounded by max_hops (<=4 default, matching
+// container.impact_of) and max_nodes per query.  The visited set is an
+// open-addressed hash table in LDS (8K slots), the per-level frontier also
+// lives in LDS.  Collected (node, hop) pairs stream to the query's output
+// slab.  Used for p50 blast-radius query latency and the /v1/graph read path.
+
+constexpr int IQ_HASH = 8192;       // LDS hash slots (32 KB)
+constexpr int IQ_FRONTIER = 2048;   // per-level frontier cap in LDS
+
+__global__ void impact_query_kernel(
+    const uint64_t* __restrict__ row_off,
+    const uint32_t* __restrict__ col,
+    const uint8_t* __restrict__ etype,
+    uint32_t allowed_mask,
+    const uint32_t* __restrict__ query_sources,   // [Q]
+    int num_queries,
+    int max_hops,
+    int max_nodes_per_query,                      // slab stride
+    uint32_t* __restrict__ out_nodes,             // [Q * stride]
+    uint8_t* __restrict__ out_hops,               // [Q * stride]
+    uint32_t* __restrict__ out_counts,            // [Q] (clamped to stride)
+    uint8_t* __restrict__ out_truncated) {        // [Q]
+    __shared__ uint32_t h_table[IQ_HASH];
+    __shared__ uint32_t fr[2][IQ_FRONTIER];
+    __shared__ unsigned fr_size[2];
+    __shared__ unsigned out_cursor;
+    __shared__ unsigned truncated;
+
+    for (int q = blockIdx.x; q < num_queries; q += gridDim.x) {
+        // reset LDS state
+        for (int i = threadIdx.x; i < IQ_HASH; i += blockDim.x) h_table[i] = ABOM_UNVISITED;
+        if (threadIdx.x == 0) {
+            fr_size[0] = 1;
+            fr_size[1] = 0;
+            out_cursor = 0;
+            truncated = 0;
+            fr[0][0] = query_sources[q];
+            // seed visited
+            uint32_t s = query_sources[q];
+            uint32_t h = (s * 2654435761u) & (IQ_HASH - 1);
+            h_table[h] = s;
+        }
+        __syncthreads();
+
+        uint32_t* slab_nodes = out_nodes + (long long)q * max_nodes_per_query;
+        uint8_t* slab_hops = out_hops + (long long)q * max_nodes_per_query;
+        if (threadIdx.x == 0 && max_nodes_per_query > 0) {
+            slab_nodes[0] = fr[0][0];
+            slab_hops[0] = 0;
+            out_cursor = 1;
+        }
+        __syncthreads();
+
+        int cur = 0;
+        for (int hop = 1; hop <= max_hops; ++hop) {
+            const int nxt = cur ^ 1;
+            const unsigned fsz = fr_size[cur];
+            if (fsz == 0) break;
+            __syncthreads();
+            // Threads stride the (frontier x neighbor) work by frontier entry.
+            for (unsigned i = threadIdx.x; i < fsz; i += blockDim.x) {
+                const uint32_t u = fr[cur][i];
+                const uint64_t beg = row_off[u];
+                const uint64_t end = row_off[u + 1];
+                for (uint64_t e = beg; e < end; ++e) {
+                    if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
+                    const uint32_t v = col[e];
+                    // LDS open-addressing insert; full table => truncate.
+                    uint32_t h = (v * 2654435761u) & (IQ_HASH - 1);
+                    bool inserted = false, seen = false;
+                    for (int probe = 0; probe < 64; ++probe) {
+                        uint32_t prev = atomicCAS(&h_table[h], ABOM_UNVISITED, v);
+                        if (prev == ABOM_UNVISITED) { inserted = true; break; }
+                        if (prev == v) { seen = true; break; }
+                        h = (h + 1) & (IQ_HASH - 1);
+                    }
+                    if (seen) continue;
+                    if (!inserted) { truncated = 1; continue; }
+                    const unsigned oi = atomicAdd(&out_cursor, 1u);
+                    if ((int)oi < max_nodes_per_query) {
+                        slab_nodes[oi] = v;
+                        slab_hops[oi] = (uint8_t)hop;
+                    } else {
+                        truncated = 1;
+                    }
+                    const unsigned fi = atomicAdd(&fr_size[nxt], 1u);
+                    if ((int)fi < IQ_FRONTIER) fr[nxt][fi] = v;
+                    else truncated = 1;
+                }
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) {
+                if (fr_size[nxt] > IQ_FRONTIER) fr_size[nxt] = IQ_FRONTIER;
+                fr_size[cur] = 0;
+            }
+            cur = nxt;
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) {
+            out_counts[q] = out_cursor < (unsigned)max_nodes_per_query ? out_cursor
+                                                                       : (unsigned)max_nodes_per_query;
+            out_truncated[q] = (uint8_t)truncated;
+        }
+        __syncthreads();
+    }
+}
+
+// ── Risk-score kernel ──────────────────────────────────────────────────────
+// Implements models/blast.py risk_score_from_counts exactly (f32): the CPU
+// formula is the specification; parity is asserted in tests/test_ops_gpu.py.
+
+struct RiskWeights {
+    float base_critical, base_high, base_medium, base_low;
+    float agent_w, agent_cap, cred_w, cred_cap, tool_w, tool_cap;
+    float ai_boost, kev_boost, epss_boost, epss_threshold;
+    float sc_t1, sc_b1, sc_t2, sc_b2, sc_t3, sc_b3;
+    float reach_boost, unreach_penalty;
+};
+
+__global__ void risk_score_kernel(
+    const uint8_t* __restrict__ severity,     // SEVERITY_CODE (5=crit..2=low)
+    const uint32_t* __restrict__ n_agents,
+    const uint32_t* __restrict__ n_creds,
+    const uint32_t* __restrict__ n_tools,
+    const uint8_t* __restrict__ flags,        // bit0 ai_ctx, bit1 kev, bit2 suppressed
+    const float* __restrict__ epss,           // <0 => none
+    const float* __restrict__ scorecard,      // <0 => none
+    const int8_t* __restrict__ reach,         // -1 unknown / 0 unreachable / 1 reachable
+    float* __restrict__ out,
+    long long n, RiskWeights w) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        const uint8_t f = flags[i];
+        if (f & 4u) { out[i] = 0.0f; continue; }  // suppressed / VEX
+        float base = 0.0f;
+        switch (severity[i]) {
+            case 5: base = w.base_critical; break;
+            case 4: base = w.base_high; break;
+            case 3: base = w.base_medium; break;
+            case 2: base = w.base_low; break;
+            default: base = 0.0f;
+        }
+        const float af = fminf((float)n_agents[i] * w.agent_w, w.agent_cap);
+        const float cf = fminf((float)n_creds[i] * w.cred_w, w.cred_cap);
+        const float tf = fminf((float)n_tools[i] * w.tool_w, w.tool_cap);
+        const int ai_signals = (int)(f & 1u) + (n_creds[i] > 0) + (n_tools[i] > 0);
+        const float ai = ai_signals >= 2 ? w.ai_boost : 0.0f;
+        const float kev = (f & 2u) ? w.kev_boost : 0.0f;
+        const float ep = (epss[i] >= w.epss_threshold) ? w.epss_boost : 0.0f;
+        float sc = 0.0f;
+        const float s = scorecard[i];
+        if (s >= 0.0f) {
+            if (s < w.sc_t1) sc = w.sc_b1;
+            else if (s < w.sc_t2) sc = w.sc_b2;
+            else if (s < w.sc_t3) sc = w.sc_b3;
+        }
+        float ra = 0.0f;
+        if (reach[i] == 1) ra = w.reach_boost;
+        else if (reach[i] == 0) ra = -w.unreach_penalty;
+        out[i] = fmaxf(0.0f, fminf(base + af + cf + tf + ai + kev + ep + sc + ra, 10.0f));
+    }
+}
+
+// ── Segmented reductions for rollup ────────────────────────────────────────
+// Per-finding severity scatter into per-container histograms (6 severity
+// buckets) — container.rollup's descendant aggregates.
+__global__ void severity_histogram_kernel(
+    const uint32_t* __restrict__ owner,       // [F] container index per finding
+    const uint8_t* __restrict__ severity,     // [F] SEVERITY_CODE
+    unsigned int* __restrict__ hist,          // [C * 6]
+    long long n) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        atomicAdd(&hist[(long long)owner[i] * 6 + severity[i]], 1u);
+    }
+}
+
+}  // namespace abom
+
+// ── C ABI ──────────────────────────────────────────────────────────────────
+
+extern "C" int abom_bfs_init(void* dist, long long n, void* stream) {
+    const int block = 256;
+    hipLaunchKernelGGL(abom::init_dist_kernel, dim3(abom::grid_for(n, block)), dim3(block), 0,
+                       (hipStream_t)stream, (uint32_t*)dist, n, ABOM_UNVISITED);
+    return (int)hipGetLastError();
+}
+
+extern "C" int abom_bfs_seed(const void* sources, long long n_sources, void* dist,
+                             void* frontier, void* stream) {
+    const int block = 256;
+    hipLaunchKernelGGL(abom::seed_sources_kernel, dim3(abom::grid_for(n_sources, block)),
+                       dim3(block), 0, (hipStream_t)stream, (const uint32_t*)sources, n_sources,
+                       (uint32_t*)dist, (uint32_t*)frontier);
+    return (int)hipGetLastError();
+}
+
+extern "C" int abom_bfs_expand(
+    const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
+    const void* frontier, long long frontier_size, void* dist, unsigned int next_level,
+    void* next_frontier, void* next_count, void* heavy_queue, void* heavy_count,
+    long long capacity, void* stream) {
+    const int block = 256;
+    hipLaunchKernelGGL(abom::bfs_expand_kernel, dim3(abom::grid_for(frontier_size, block)),
+                       dim3(block), 0, (hipStream_t)stream, (const uint64_t*)row_off,
+                       (const uint32_t*)col, (const uint8_t*)etype, allowed_mask,
+                       (const uint32_t*)frontier, frontier_size, (uint32_t*)dist, next_level,
+                       (uint32_t*)next_frontier, (unsigned int*)next_count,
+                       (uint32_t*)heavy_queue, (unsigned int*)heavy_count, capacity);
+    return (int)hipGetLastError();
+}
+
+extern "C" int abom_bfs_expand_heavy(
+    const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
+    const void* heavy_queue, const void* heavy_size_ptr, void* dist, unsigned int next_level,
+    void* next_frontier, void* next_count, long long capacity, void* stream) {
+    const int block = 256;  // 4 waves per block; kernel grid-strides by wave
+    hipLaunchKernelGGL(abom::bfs_expand_heavy_kernel, dim3(2048), dim3(block), 0,
+                       (hipStream_t)stream, (const uint64_t*)row_off, (const uint32_t*)col,
+                       (const uint8_t*)etype, allowed_mask, (const uint32_t*)heavy_queue,
+                       (const unsigned int*)heavy_size_ptr, (uint32_t*)dist, next_level,
+                       (uint32_t*)next_frontier, (unsigned int*)next_count, capacity);
+    return (int)hipGetLastError();
+}
+
+// Full multi-source BFS: host-side level loop, one device->host count read
+// per level.  dist must be u32[N]; frontier_a/b u32[N]; heavy_queue u32[N];
+// counters = device u32[2] (next_count, heavy_count).  Returns negative hip
+// error or the number of levels run.
+extern "C" int abom_bfs_run(
+    const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
+    const void* sources, long long n_sources, void* dist, long long num_nodes,
+    void* frontier_a, void* frontier_b, void* heavy_queue, void* counters,
+    int max_levels, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    unsigned int* ctr = (unsigned int*)counters;
+    int rc = abom_bfs_init(dist, num_nodes, stream);
+    if (rc) return -rc;
+    rc = abom_bfs_seed(sources, n_sources, dist, frontier_a, stream);
+    if (rc) return -rc;
+
+    uint32_t* cur = (uint32_t*)frontier_a;
+    uint32_t* nxt = (uint32_t*)frontier_b;
+    long long frontier_size = n_sources;
+    int level = 0;
+    unsigned int host_count = 0;
+    while (frontier_size > 0 && level < max_levels) {
+        ++level;
+        ABOM_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(unsigned int), s));
+        rc = abom_bfs_expand(row_off, col, etype, allowed_mask, cur, frontier_size, dist,
+                             (unsigned int)level, nxt, ctr, heavy_queue, ctr + 1, num_nodes,
+                             stream);
+        if (rc) return -rc;
+        rc = abom_bfs_expand_heavy(row_off, col, etype, allowed_mask, heavy_queue, ctr + 1,
+                                   dist, (unsigned int)level, nxt, ctr, num_nodes, stream);
+        if (rc) return -rc;
+        ABOM_CHECK(hipMemcpyAsync(&host_count, ctr, sizeof(unsigned int),
+                                  hipMemcpyDeviceToHost, s));
+        ABOM_CHECK(hipStreamSynchronize(s));
+        frontier_size = host_count;
+        uint32_t* t = cur; cur = nxt; nxt = t;
+    }
+    return level;
+}
+
+extern "C" int abom_impact_query(
+    const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
+    const void* query_sources, int num_queries, int max_hops, int max_nodes_per_query,
+    void* out_nodes, void* out_hops, void* out_counts, void* out_truncated, void* stream) {
+    const int block = 256;
+    const int grid = num_queries < 2048 ? (num_queries > 0 ? num_queries : 1) : 2048;
+    hipLaunchKernelGGL(abom::impact_query_kernel, dim3(grid), dim3(block), 0,
+                       (hipStream_t)stream, (const uint64_t*)row_off, (const uint32_t*)col,
+                       (const uint8_t*)etype, allowed_mask, (const uint32_t*)query_sources,
+                       num_queries, max_hops, max_nodes_per_query, (uint32_t*)out_nodes,
+                       (uint8_t*)out_hops, (uint32_t*)out_counts, (uint8_t*)out_truncated);
+    return (int)hipGetLastError();
+}
+
+extern "C" int abom_risk_score(
+    const void* severity, const void* n_agents, const void* n_creds, const void* n_tools,
+    const void* flags, const void* epss, const void* scorecard, const void* reach,
+    void* out, long long n, const float* weights22, void* stream) {
+    abom::RiskWeights w;
+    const float* p = weights22;
+    w.base_critical = p[0]; w.base_high = p[1]; w.base_medium = p[2]; w.base_low = p[3];
+    w.agent_w = p[4]; w.agent_cap = p[5]; w.cred_w = p[6]; w.cred_cap = p[7];
+    w.tool_w = p[8]; w.tool_cap = p[9]; w.ai_boost = p[10]; w.kev_boost = p[11];
+    w.epss_boost = p[12]; w.epss_threshold = p[13]; w.sc_t1 = p[14]; w.sc_b1 = p[15];
+    w.sc_t2 = p[16]; w.sc_b2 = p[17]; w.sc_t3 = p[18]; w.sc_b3 = p[19];
+    w.reach_boost = p[20]; w.unreach_penalty = p[21];
+    const int block = 256;
+    hipLaunchKernelGGL(abom::risk_score_kernel, dim3(abom::grid_for(n, block)), dim3(block), 0,
+                       (hipStream_t)stream, (const uint8_t*)severity, (const uint32_t*)n_agents,
+                       (const uint32_t*)n_creds, (const uint32_t*)n_tools, (const uint8_t*)flags,
+                       (const float*)epss, (const float*)scorecard, (const int8_t*)reach,
+                       (float*)out, n, w);
+    return (int)hipGetLastError();
+}
+
+extern "C" int abom_severity_histogram(
+    const void* owner, const void* severity, void* hist, long long n, void* stream) {
+    const int block = 256;
+    hipLaunchKernelGGL(abom::severity_histogram_kernel, dim3(abom::grid_for(n, block)),
+                       dim3(block), 0, (hipStream_t)stream, (const uint32_t*)owner,
+                       (const uint8_t*)severity, (unsigned int*)hist, n);
+    return (int)hipGetLastError();
+}

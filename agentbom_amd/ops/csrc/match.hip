@@ -1,0 +1,107 @@
+// Bulk OSV/GHSA version-range matching on gfx950.
+//
+// Replaces the reference's per-(package, advisory-window) Python hot loop
+// (reference: src/agent_bom/scanners/package_scan.py:694-803 window walk,
+// src/agent_bom/db/lookup.py:549 batch lookup + Python version filtering)
+// with one kernel over the whole package batch against a GPU-resident
+// columnar advisory arena.
+//
+// Arena model: advisory windows for all ecosystems live in
+// HBM as sorted columnar arrays grouped by group_key = hash64(ecosystem,
+// normalized_name).  One row per window — multi-branch advisories keep one
+// row per (introduced, fixed, last_affected) triple, never collapsed.
+//
+// Matching semantics (fail-closed):
+// - introduced "0"/missing  -> window has no lower bound (WF_HAS_INTRO off)
+// - fixed is exclusive, last_affected inclusive
+// - commit-SHA / unencodable bounds  -> WF_CPU_FALLBACK, never matched here;
+//   the host resolves those windows with the exact CPU comparator
+// - packages with unencodable versions are skipped (PF_ENCODABLE off) and
+//   resolved on the host the same way.
+//
+// Each thread owns one package: binary-search the sorted group_key array
+// (top of the tree stays L2/L3-resident), then walk that group's windows
+// testing the package's u128 key against each window's bounds.  Matches are
+// appended through a device atomic cursor as (pkg_idx << 32 | window_idx);
+// the caller sorts the pairs for deterministic downstream ordering.
+
+#include "abom_common.h"
+
+namespace abom {
+
+__global__ void match_kernel(
+    const uint64_t* __restrict__ pkg_group_key,   // [P] hash64(eco, name)
+    const uint64_t* __restrict__ pkg_key_hi,      // [P]
+    const uint64_t* __restrict__ pkg_key_lo,      // [P]
+    const uint8_t* __restrict__ pkg_flags,        // [P]
+    long long num_packages,
+    const uint64_t* __restrict__ group_keys,      // [G] sorted ascending
+    const uint32_t* __restrict__ group_off,       // [G+1] -> window ranges
+    long long num_groups,
+    const uint64_t* __restrict__ w_intro_hi,      // [W]
+    const uint64_t* __restrict__ w_intro_lo,
+    const uint64_t* __restrict__ w_fixed_hi,
+    const uint64_t* __restrict__ w_fixed_lo,
+    const uint64_t* __restrict__ w_last_hi,
+    const uint64_t* __restrict__ w_last_lo,
+    const uint8_t* __restrict__ w_flags,
+    uint64_t* __restrict__ out_pairs,             // [capacity]
+    unsigned int* __restrict__ out_count,
+    long long capacity) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x; p < num_packages;
+         p += stride) {
+        if (!(pkg_flags[p] & PF_ENCODABLE)) continue;
+        const uint64_t gkey = pkg_group_key[p];
+
+        // Branchless-ish binary search over sorted group keys.
+        long long lo = 0, hi = num_groups;
+        while (lo < hi) {
+            long long mid = (lo + hi) >> 1;
+            if (group_keys[mid] < gkey) lo = mid + 1; else hi = mid;
+        }
+        if (lo >= num_groups || group_keys[lo] != gkey) continue;
+
+        const uint32_t wbeg = group_off[lo];
+        const uint32_t wend = group_off[lo + 1];
+        const uint64_t khi = pkg_key_hi[p];
+        const uint64_t klo = pkg_key_lo[p];
+
+        for (uint32_t w = wbeg; w < wend; ++w) {
+            const uint8_t f = w_flags[w];
+            if (f & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
+            if ((f & WF_HAS_INTRO) && key_lt(khi, klo, w_intro_hi[w], w_intro_lo[w])) continue;
+            if ((f & WF_HAS_FIXED) && key_ge(khi, klo, w_fixed_hi[w], w_fixed_lo[w])) continue;
+            if ((f & WF_HAS_LAST) && key_gt(khi, klo, w_last_hi[w], w_last_lo[w])) continue;
+            const unsigned idx = atomicAdd(out_count, 1u);
+            if ((long long)idx < capacity) {
+                out_pairs[idx] = ((uint64_t)p << 32) | (uint64_t)w;
+            }
+        }
+    }
+}
+
+}  // namespace abom
+
+extern "C" int abom_match(
+    const void* pkg_group_key, const void* pkg_key_hi, const void* pkg_key_lo,
+    const void* pkg_flags, long long num_packages,
+    const void* group_keys, const void* group_off, long long num_groups,
+    const void* w_intro_hi, const void* w_intro_lo,
+    const void* w_fixed_hi, const void* w_fixed_lo,
+    const void* w_last_hi, const void* w_last_lo,
+    const void* w_flags,
+    void* out_pairs, void* out_count, long long capacity, void* stream) {
+    const int block = 256;
+    const int grid = abom::grid_for(num_packages, block);
+    hipLaunchKernelGGL(abom::match_kernel, dim3(grid), dim3(block), 0, (hipStream_t)stream,
+                       (const uint64_t*)pkg_group_key, (const uint64_t*)pkg_key_hi,
+                       (const uint64_t*)pkg_key_lo, (const uint8_t*)pkg_flags, num_packages,
+                       (const uint64_t*)group_keys, (const uint32_t*)group_off, num_groups,
+                       (const uint64_t*)w_intro_hi, (const uint64_t*)w_intro_lo,
+                       (const uint64_t*)w_fixed_hi, (const uint64_t*)w_fixed_lo,
+                       (const uint64_t*)w_last_hi, (const uint64_t*)w_last_lo,
+                       (const uint8_t*)w_flags, (uint64_t*)out_pairs,
+                       (unsigned int*)out_count, capacity);
+    return (int)hipGetLastError();
+}

@@ -1,0 +1,305 @@
+"""ctypes bindings for the gfx950 HIP engine (_abom_gpu.so).
+
+Tensors are allocated and owned by PyTorch (the caching allocator manages
+HBM); raw device pointers + the current torch HIP stream are handed to the
+C ABI.  On a machine with a visible GPU the native library is REQUIRED —
+a missing .so raises instead of silently falling back to a slow path
+(AGENT_BOM_GPU_REQUIRED=0 opts out for CPU-only dev boxes).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from pathlib import Path
+from typing import Optional
+
+import numpy as np
+
+_SO_PATH = Path(__file__).resolve().parent / "_abom_gpu.so"
+_lib: Optional[ctypes.CDLL] = None
+
+_c = ctypes.c_void_p
+_i64 = ctypes.c_longlong
+_u32 = ctypes.c_uint
+_i32 = ctypes.c_int
+
+_SIGNATURES = {
+    "abom_abi_version": ([], _i32),
+    "abom_device_count": ([], _i32),
+    "abom_synchronize": ([_c], _i32),
+    "abom_error_string": ([_i32], ctypes.c_char_p),
+    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c, _c, _i64, _c], _i32),
+    "abom_bfs_init": ([_c, _i64, _c], _i32),
+    "abom_bfs_seed": ([_c, _i64, _c, _c, _c], _i32),
+    "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _i64, _c], _i32),
+    "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _i64, _c], _i32),
+    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c], _i32),
+    "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
+    "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
+    "abom_severity_histogram": ([_c, _c, _c, _i64, _c], _i32),
+}
+
+
+class NativeUnavailable(RuntimeError):
+    pass
+
+
+def load(required: Optional[bool] = None) -> ctypes.CDLL:
+    """Load the engine .so, binding signatures once."""
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not _SO_PATH.exists():
+        if required is None:
+            from agentbom_amd.utils import config as cfg
+
+            required = cfg.GPU_REQUIRED and _torch_has_gpu()
+        if required:
+            raise NativeUnavailable(
+                f"HIP engine missing: {_SO_PATH} not built. Run "
+                "`python -m agentbom_amd.ops.build` (requires hipcc). The GPU "
+                "path never silently falls back — set AGENT_BOM_GPU_REQUIRED=0 "
+                "only on CPU-only development machines."
+            )
+        raise NativeUnavailable(f"{_SO_PATH} not built")
+    lib = ctypes.CDLL(str(_SO_PATH))
+    for name, (argtypes, restype) in _SIGNATURES.items():
+        fn = getattr(lib, name)
+        fn.argtypes = argtypes
+        fn.restype = restype
+    _lib = lib
+    return lib
+
+
+def _torch_has_gpu() -> bool:
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def available() -> bool:
+    try:
+        load(required=False)
+        return True
+    except NativeUnavailable:
+        return False
+
+
+def _check(rc: int, what: str) -> None:
+    if rc != 0:
+        lib = load()
+        msg = lib.abom_error_string(abs(rc)).decode() if rc else ""
+        raise RuntimeError(f"{what} failed: hip error {rc} ({msg})")
+
+
+def _ptr(t) -> ctypes.c_void_p:
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _stream() -> ctypes.c_void_p:
+    import torch
+
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+# ── High-level ops (torch tensors on cuda device) ──────────────────────────
+
+
+def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_off,
+          windows: dict, capacity: Optional[int] = None):
+    """Bulk version-range match.  Returns sorted (pkg_idx u32, window_idx u32).
+
+    All tensors device-resident; ``windows`` carries intro/fixed/last hi+lo
+    (int64 bit-patterns) and flags (uint8).  Retries with a larger buffer if
+    capacity overflows.
+    """
+    import torch
+
+    lib = load()
+    P = pkg_group_key.numel()
+    G = group_keys.numel()
+    cap = capacity or max(1024, P // 4)
+    dev = pkg_group_key.device
+    while True:
+        out_pairs = torch.empty(cap, dtype=torch.int64, device=dev)
+        out_count = torch.zeros(1, dtype=torch.int32, device=dev)
+        rc = lib.abom_match(
+            _ptr(pkg_group_key), _ptr(pkg_key_hi), _ptr(pkg_key_lo), _ptr(pkg_flags), P,
+            _ptr(group_keys), _ptr(group_off), G,
+            _ptr(windows["intro_hi"]), _ptr(windows["intro_lo"]),
+            _ptr(windows["fixed_hi"]), _ptr(windows["fixed_lo"]),
+            _ptr(windows["last_hi"]), _ptr(windows["last_lo"]),
+            _ptr(windows["flags"]),
+            _ptr(out_pairs), _ptr(out_count), cap, _stream(),
+        )
+        _check(rc, "abom_match")
+        n = int(out_count.item())
+        if n <= cap:
+            pairs = out_pairs[:n]
+            pairs, _ = torch.sort(pairs)
+            pkg_idx = (pairs >> 32).to(torch.int64)
+            win_idx = (pairs & 0xFFFFFFFF).to(torch.int64)
+            return pkg_idx, win_idx
+        cap = int(n * 1.2) + 1024
+
+
+def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0xFFFFFFFF,
+        max_levels: int = 64, workspace: Optional[dict] = None):
+    """Multi-source BFS over CSR; returns u32 dist (UNVISITED = 0xFFFFFFFF).
+
+    ``workspace`` may carry preallocated buffers (dist/frontier_a/frontier_b/
+    heavy/counters) to avoid reallocation in steady-state serving.
+    """
+    import torch
+
+    lib = load()
+    dev = row_off.device
+    ws = workspace or {}
+    dist = ws.get("dist")
+    if dist is None or dist.numel() < num_nodes:
+        dist = torch.empty(num_nodes, dtype=torch.int32, device=dev)
+    fa = ws.get("frontier_a")
+    if fa is None or fa.numel() < num_nodes:
+        fa = torch.empty(num_nodes, dtype=torch.int32, device=dev)
+    fb = ws.get("frontier_b")
+    if fb is None or fb.numel() < num_nodes:
+        fb = torch.empty(num_nodes, dtype=torch.int32, device=dev)
+    hq = ws.get("heavy")
+    if hq is None or hq.numel() < num_nodes:
+        hq = torch.empty(num_nodes, dtype=torch.int32, device=dev)
+    ctr = ws.get("counters")
+    if ctr is None:
+        ctr = torch.zeros(2, dtype=torch.int32, device=dev)
+    if workspace is not None:
+        workspace.update(dist=dist, frontier_a=fa, frontier_b=fb, heavy=hq, counters=ctr)
+
+    et = _ptr(etype) if etype is not None else None
+    rc = lib.abom_bfs_run(
+        _ptr(row_off), _ptr(col), et, allowed_mask,
+        _ptr(sources), sources.numel(), _ptr(dist), num_nodes,
+        _ptr(fa), _ptr(fb), _ptr(hq), _ptr(ctr), max_levels, _stream(),
+    )
+    if rc < 0:
+        _check(-rc, "abom_bfs_run")
+    return dist[:num_nodes]
+
+
+def bfs_level(row_off, col, frontier, dist, level: int, etype=None,
+              allowed_mask: int = 0xFFFFFFFF, workspace: Optional[dict] = None):
+    """Expand ONE BFS level; returns the next frontier (claimed vertices).
+
+    ``dist`` may be larger than the local graph (distributed mode: the tail
+    indexes remote nodes and a claim there marks "queued for send").  Used by
+    parallel/dist_bfs.py between RCCL frontier exchanges.
+    """
+    import torch
+
+    lib = load()
+    dev = row_off.device
+    ws = workspace if workspace is not None else {}
+    cap = dist.numel()
+    nxt = ws.get("next")
+    if nxt is None or nxt.numel() < cap:
+        nxt = torch.empty(cap, dtype=torch.int32, device=dev)
+        ws["next"] = nxt
+    hq = ws.get("heavy")
+    if hq is None or hq.numel() < max(frontier.numel(), 1):
+        hq = torch.empty(max(frontier.numel(), 1024), dtype=torch.int32, device=dev)
+        ws["heavy"] = hq
+    ctr = ws.get("counters")
+    if ctr is None:
+        ctr = torch.zeros(2, dtype=torch.int32, device=dev)
+        ws["counters"] = ctr
+    ctr.zero_()
+    et = _ptr(etype) if etype is not None else None
+    rc = lib.abom_bfs_expand(
+        _ptr(row_off), _ptr(col), et, allowed_mask, _ptr(frontier), frontier.numel(),
+        _ptr(dist), level, _ptr(nxt), _ptr(ctr), _ptr(hq),
+        ctypes.c_void_p(ctr.data_ptr() + 4), cap, _stream(),
+    )
+    _check(rc, "abom_bfs_expand")
+    rc = lib.abom_bfs_expand_heavy(
+        _ptr(row_off), _ptr(col), et, allowed_mask, _ptr(hq),
+        ctypes.c_void_p(ctr.data_ptr() + 4), _ptr(dist), level, _ptr(nxt), _ptr(ctr),
+        cap, _stream(),
+    )
+    _check(rc, "abom_bfs_expand_heavy")
+    n = int(ctr[0].item())
+    return nxt[:n].clone()
+
+
+def impact_query(row_off, col, query_sources, etype=None, allowed_mask: int = 0xFFFFFFFF,
+                 max_hops: int = 4, max_nodes: int = 4096):
+    """Batched bounded blast-radius queries (one block per query).
+
+    Returns (nodes u32 [Q, max_nodes], hops u8 [Q, max_nodes], counts u32 [Q],
+    truncated u8 [Q]).
+    """
+    import torch
+
+    lib = load()
+    dev = row_off.device
+    Q = query_sources.numel()
+    out_nodes = torch.empty((Q, max_nodes), dtype=torch.int32, device=dev)
+    out_hops = torch.empty((Q, max_nodes), dtype=torch.uint8, device=dev)
+    out_counts = torch.zeros(Q, dtype=torch.int32, device=dev)
+    out_trunc = torch.zeros(Q, dtype=torch.uint8, device=dev)
+    et = _ptr(etype) if etype is not None else None
+    rc = lib.abom_impact_query(
+        _ptr(row_off), _ptr(col), et, allowed_mask, _ptr(query_sources), Q,
+        max_hops, max_nodes, _ptr(out_nodes), _ptr(out_hops), _ptr(out_counts),
+        _ptr(out_trunc), _stream(),
+    )
+    _check(rc, "abom_impact_query")
+    return out_nodes, out_hops, out_counts, out_trunc
+
+
+def risk_weights_array() -> np.ndarray:
+    """The 22-float weight vector, in kernel order, from live config."""
+    from agentbom_amd.utils import config as cfg
+
+    return np.array(
+        [
+            cfg.RISK_BASE_CRITICAL, cfg.RISK_BASE_HIGH, cfg.RISK_BASE_MEDIUM, cfg.RISK_BASE_LOW,
+            cfg.RISK_AGENT_WEIGHT, cfg.RISK_AGENT_CAP, cfg.RISK_CRED_WEIGHT, cfg.RISK_CRED_CAP,
+            cfg.RISK_TOOL_WEIGHT, cfg.RISK_TOOL_CAP, cfg.RISK_AI_BOOST, cfg.RISK_KEV_BOOST,
+            cfg.RISK_EPSS_BOOST, cfg.EPSS_CRITICAL_THRESHOLD,
+            cfg.RISK_SCORECARD_TIER1_THRESHOLD, cfg.RISK_SCORECARD_TIER1_BOOST,
+            cfg.RISK_SCORECARD_TIER2_THRESHOLD, cfg.RISK_SCORECARD_TIER2_BOOST,
+            cfg.RISK_SCORECARD_TIER3_THRESHOLD, cfg.RISK_SCORECARD_TIER3_BOOST,
+            cfg.RISK_REACHABLE_BOOST, cfg.RISK_UNREACHABLE_PENALTY,
+        ],
+        dtype=np.float32,
+    )
+
+
+def risk_score(severity, n_agents, n_creds, n_tools, flags, epss, scorecard, reach):
+    """GPU risk scoring — same formula as models/blast.risk_score_from_counts."""
+    import torch
+
+    lib = load()
+    n = severity.numel()
+    out = torch.empty(n, dtype=torch.float32, device=severity.device)
+    w = risk_weights_array()
+    rc = lib.abom_risk_score(
+        _ptr(severity), _ptr(n_agents), _ptr(n_creds), _ptr(n_tools), _ptr(flags),
+        _ptr(epss), _ptr(scorecard), _ptr(reach), _ptr(out), n,
+        w.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), _stream(),
+    )
+    _check(rc, "abom_risk_score")
+    return out
+
+
+def severity_histogram(owner, severity, num_containers: int):
+    import torch
+
+    lib = load()
+    hist = torch.zeros(num_containers * 6, dtype=torch.int32, device=owner.device)
+    rc = lib.abom_severity_histogram(_ptr(owner), _ptr(severity), _ptr(hist),
+                                     owner.numel(), _stream())
+    _check(rc, "abom_severity_histogram")
+    return hist.view(num_containers, 6)

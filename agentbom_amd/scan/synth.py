@@ -1,0 +1,202 @@
+"""Deterministic synthetic estate generator (10M-node scale, numpy).
+
+Rebuilds the reference's seeded benchmark-estate generator idea
+(reference: scripts/generate_graph_benchmark_estate.py — skewed fan-out,
+~1% of agents own 18-32 servers) at GPU scale: node-id ranges per entity
+class, typed edge lists, per-package encoded version keys, and a synthetic
+advisory arena with controllable match density.
+
+Node id space (contiguous ranges):
+  agents [0, A) | servers [A, A+S) | creds | tools | packages (last)
+
+Edge types (compact engine codes; the full RelationshipType enum of the
+graph layer maps onto these for traversal classes):
+  0 USES (agent->server)        1 CONTAINS (server->package)
+  2 HAS_CREDENTIAL (server->cred)  3 PROVIDES_TOOL (server->tool)
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import numpy as np
+
+from agentbom_amd.db.arena import AdvisoryArena, build_arena_from_columns
+from agentbom_amd.ops.cpu_ref import PF_ENCODABLE, WF_HAS_FIXED, WF_HAS_INTRO
+from agentbom_amd.utils.version_keys import RELEASE_RANK, pack_batch
+
+ET_USES = 0
+ET_CONTAINS = 1
+ET_HAS_CRED = 2
+ET_PROVIDES_TOOL = 3
+
+
+@dataclass
+class SyntheticEstate:
+    n_agents: int
+    n_servers: int
+    n_creds: int
+    n_tools: int
+    n_packages: int
+    # node-id range starts
+    agent_base: int
+    server_base: int
+    cred_base: int
+    tool_base: int
+    pkg_base: int
+    num_nodes: int
+    # typed edges (forward direction)
+    edge_src: np.ndarray  # int64
+    edge_dst: np.ndarray
+    edge_type: np.ndarray  # uint8
+    # package columns
+    pkg_name_id: np.ndarray  # uint64 group keys
+    pkg_key_hi: np.ndarray  # uint64
+    pkg_key_lo: np.ndarray
+    pkg_flags: np.ndarray  # uint8
+    # credential classification (for CWE impact filtering)
+    cred_is_db: np.ndarray  # uint8 [n_creds]
+    tool_is_db: np.ndarray  # uint8 [n_tools]
+    arena: AdvisoryArena
+    seed: int
+
+    @property
+    def num_edges(self) -> int:
+        return len(self.edge_src)
+
+
+def _triple_keys(rng, n, lo=(0, 0, 0), hi=(20, 30, 60)):
+    n1 = rng.integers(lo[0], hi[0], n, dtype=np.int64)
+    n2 = rng.integers(lo[1], hi[1], n, dtype=np.int64)
+    n3 = rng.integers(lo[2], hi[2], n, dtype=np.int64)
+    zeros = np.zeros(n, dtype=np.int64)
+    rank = np.full(n, RELEASE_RANK, dtype=np.int64)
+    hi_k, lo_k = pack_batch(zeros, n1, n2, n3, zeros, rank, zeros)
+    return (n1, n2, n3), (hi_k, lo_k)
+
+
+def generate_estate(
+    n_agents: int = 1000,
+    n_servers: int = 5000,
+    n_packages: int = 100_000,
+    name_catalog: int = 20_000,
+    vulnerable_name_fraction: float = 0.05,
+    windows_per_name: int = 2,
+    extra_pkg_share: float = 0.5,
+    creds_per_server: float = 2.0,
+    tools_per_server: float = 4.0,
+    seed: int = 1234,
+) -> SyntheticEstate:
+    """Generate a seeded estate.  Sizes are node counts; edges follow."""
+    rng = np.random.default_rng(seed)
+
+    n_creds = max(64, n_servers // 2)
+    n_tools = max(64, n_servers)
+    agent_base = 0
+    server_base = n_agents
+    cred_base = server_base + n_servers
+    tool_base = cred_base + n_creds
+    pkg_base = tool_base + n_tools
+    num_nodes = pkg_base + n_packages
+
+    # ── agent -> server (skewed: ~1% heavy agents own 18-32x the servers) ──
+    heavy = rng.random(n_agents) < 0.01
+    weights = np.where(heavy, 25.0, 1.0)
+    weights /= weights.sum()
+    owner = rng.choice(n_agents, size=n_servers, p=weights)
+    # ~10% of servers are shared with a second agent (lateral movement paths)
+    shared_mask = rng.random(n_servers) < 0.10
+    second = rng.integers(0, n_agents, n_servers)
+    shared_idx = np.nonzero(shared_mask & (second != owner))[0]
+    us_src = np.concatenate([owner, second[shared_idx]])
+    us_dst = np.concatenate([np.arange(n_servers), shared_idx]) + server_base
+
+    # ── server -> package ──────────────────────────────────────────────────
+    home_server = rng.integers(0, n_servers, n_packages)
+    sp_src = home_server + server_base
+    sp_dst = np.arange(n_packages, dtype=np.int64) + pkg_base
+    n_extra = int(n_packages * extra_pkg_share)
+    if n_extra:
+        # popularity-skewed extra containment (popular packages shared widely)
+        pop = (rng.pareto(1.5, n_extra) * n_packages / 50).astype(np.int64) % n_packages
+        ex_src = rng.integers(0, n_servers, n_extra) + server_base
+        ex_dst = pop + pkg_base
+        sp_src = np.concatenate([sp_src, ex_src])
+        sp_dst = np.concatenate([sp_dst, ex_dst])
+
+    # ── server -> cred / tool ─────────────────────────────────────────────
+    def attach(n_items, base, mean):
+        counts = rng.poisson(mean, n_servers)
+        total = int(counts.sum())
+        src = np.repeat(np.arange(n_servers), counts) + server_base
+        dst = rng.integers(0, n_items, total) + base
+        return src, dst
+
+    sc_src, sc_dst = attach(n_creds, cred_base, creds_per_server)
+    st_src, st_dst = attach(n_tools, tool_base, tools_per_server)
+
+    edge_src = np.concatenate([us_src, sp_src, sc_src, st_src]).astype(np.int64)
+    edge_dst = np.concatenate([us_dst, sp_dst, sc_dst, st_dst]).astype(np.int64)
+    edge_type = np.concatenate(
+        [
+            np.full(len(us_src), ET_USES, dtype=np.uint8),
+            np.full(len(sp_src), ET_CONTAINS, dtype=np.uint8),
+            np.full(len(sc_src), ET_HAS_CRED, dtype=np.uint8),
+            np.full(len(st_src), ET_PROVIDES_TOOL, dtype=np.uint8),
+        ]
+    )
+
+    # ── package identities + versions ─────────────────────────────────────
+    # zipf-ish name popularity over the catalog
+    name_id = (rng.pareto(1.2, n_packages) * name_catalog / 20).astype(np.uint64) % name_catalog
+    _, (key_hi, key_lo) = _triple_keys(rng, n_packages)
+    pkg_flags = np.full(n_packages, PF_ENCODABLE, dtype=np.uint8)
+
+    # ── synthetic advisory arena ──────────────────────────────────────────
+    n_vuln_names = max(1, int(name_catalog * vulnerable_name_fraction))
+    vuln_names = rng.choice(name_catalog, n_vuln_names, replace=False).astype(np.uint64)
+    W = n_vuln_names * windows_per_name
+    w_names = np.repeat(vuln_names, windows_per_name)
+    # introduced in the lower half of the version space, fixed above it: a
+    # package in a window iff introduced <= key < fixed
+    i1 = rng.integers(0, 10, W, dtype=np.int64)
+    i2 = rng.integers(0, 30, W, dtype=np.int64)
+    i3 = rng.integers(0, 60, W, dtype=np.int64)
+    span = rng.integers(1, 8, W, dtype=np.int64)
+    f1 = i1 + span
+    f2 = rng.integers(0, 30, W, dtype=np.int64)
+    f3 = rng.integers(0, 60, W, dtype=np.int64)
+    zeros = np.zeros(W, dtype=np.int64)
+    rank = np.full(W, RELEASE_RANK, dtype=np.int64)
+    ihi, ilo = pack_batch(zeros, i1, i2, i3, zeros, rank, zeros)
+    fhi, flo = pack_batch(zeros, f1, f2, f3, zeros, rank, zeros)
+    wflags = np.full(W, WF_HAS_INTRO | WF_HAS_FIXED, dtype=np.uint8)
+    sev = rng.choice([5, 4, 3, 2], size=W, p=[0.1, 0.3, 0.4, 0.2]).astype(np.uint8)
+    cvss = (rng.random(W) * 10).astype(np.float32)
+    epss = np.where(rng.random(W) < 0.7, rng.random(W) ** 2, -1.0).astype(np.float32)
+    kev = (rng.random(W) < 0.02).astype(np.uint8)
+    # CWE-impact mix (codes: 0 code-exec .. 7 client-side, 8 unknown)
+    impact = rng.choice(
+        np.arange(9, dtype=np.uint8), size=W,
+        p=[0.35, 0.08, 0.10, 0.04, 0.08, 0.12, 0.10, 0.05, 0.08],
+    ).astype(np.uint8)
+
+    arena = build_arena_from_columns(
+        w_names.astype(np.uint64), ihi, ilo, fhi, flo,
+        np.zeros(W, dtype=np.uint64), np.zeros(W, dtype=np.uint64),
+        wflags, sev, cvss, epss, kev, impact=impact,
+    )
+
+    return SyntheticEstate(
+        n_agents=n_agents, n_servers=n_servers, n_creds=n_creds, n_tools=n_tools,
+        n_packages=n_packages,
+        agent_base=agent_base, server_base=server_base, cred_base=cred_base,
+        tool_base=tool_base, pkg_base=pkg_base, num_nodes=num_nodes,
+        edge_src=edge_src, edge_dst=edge_dst, edge_type=edge_type,
+        pkg_name_id=name_id.astype(np.uint64), pkg_key_hi=key_hi, pkg_key_lo=key_lo,
+        pkg_flags=pkg_flags,
+        cred_is_db=(rng.random(n_creds) < 0.3).astype(np.uint8),
+        tool_is_db=(rng.random(n_tools) < 0.2).astype(np.uint8),
+        arena=arena,
+        seed=seed,
+    )

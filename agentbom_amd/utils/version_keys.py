@@ -1,0 +1,310 @@
+"""Order-preserving 128-bit version keys for the GPU bulk matcher.
+
+Encodes a version string into ``(hi, lo)`` u64 pair such that unsigned
+128-bit comparison of keys equals the ecosystem comparator's answer
+(:func:`agentbom_amd.utils.version_utils.compare_version_order`) for every
+pair of *encodable* versions of one ecosystem.  Anything the encoder cannot
+represent is flagged unencodable and the scan path falls back to the CPU
+comparator for exactly those (package, window) pairs — decisions are
+bit-identical either way (tested by tests/test_version_keys.py against the
+comparator oracle, including hypothesis-generated corpora).
+
+Key layout (128 bits, compared as (hi, lo) unsigned):
+
+  hi = epoch(8) | n1(28) | n2(28)
+  lo = n3(28) | n4(14) | rank(5) | num(17)
+
+- ``epoch``  : Debian/RPM/PEP440 epoch (0-255)
+- ``n1..n3`` : numeric release components (< 2^28 each; missing = 0)
+- ``n4``     : 4th numeric component / deb-apk revision (< 2^14)
+- ``rank``   : suffix class — pre-release tags rank 1..15 (< RELEASE),
+               RELEASE = 16, post-release tags 17..31
+- ``num``    : numeric argument of the suffix, stored as N+1 (0 = absent)
+               so ``alpha`` < ``alpha.0`` < ``alpha.1`` per SemVer
+
+Per-ecosystem rank tables encode each ecosystem's own suffix ordering
+(SemVer lexical identifiers, PEP 440 dev<a<b<rc<release<post, Maven
+qualifier order, apk _alpha.._rc < release < _cvs.._p, PHP version_compare,
+NuGet case-folded labels).  Windows/packages with commit SHAs, >4 numeric
+components, huge components, unknown tags, Go pseudo-versions, PEP 440
+local versions, or Debian/RPM alphanumeric interleaves are unencodable.
+
+The HIP matcher (ops/csrc/match.hip) compares these keys as two u64s; the
+reference semantics live in version_utils (reference:
+src/agent_bom/version_utils.py:927-1315).
+"""
+
+from __future__ import annotations
+
+import re
+from functools import lru_cache
+from typing import Optional
+
+from agentbom_amd.utils.version_utils import ECO_ALIASES, _looks_like_commit_sha
+
+RELEASE_RANK = 16
+
+# Bit widths
+_N1_BITS, _N2_BITS, _N3_BITS, _N4_BITS, _RANK_BITS, _NUM_BITS = 28, 28, 28, 14, 5, 17
+_N1_MAX = (1 << _N1_BITS) - 1
+_N2_MAX = (1 << _N2_BITS) - 1
+_N3_MAX = (1 << _N3_BITS) - 1
+_N4_MAX = (1 << _N4_BITS) - 1
+_NUM_MAX = (1 << _NUM_BITS) - 1
+
+# ── Suffix rank tables (per ecosystem family) ──────────────────────────────
+# SemVer / npm: prerelease identifiers compare ASCII-lexically; a bare
+# numeric identifier sorts below every alphanumeric one (rank 0).
+_SEMVER_RANKS = {
+    "alpha": 1, "beta": 2, "canary": 3, "dev": 4, "next": 5, "nightly": 6,
+    "pre": 7, "preview": 8, "rc": 9, "snapshot": 10,
+}
+_PYPI_RANKS = {"dev": 1, "a": 2, "alpha": 2, "b": 3, "beta": 3, "c": 4, "rc": 4,
+               "pre": 4, "preview": 4}
+_PYPI_POST = {"post": 17, "rev": 17, "r": 17}
+_MAVEN_RANKS = {"alpha": 1, "a": 1, "beta": 2, "b": 2, "milestone": 3, "m": 3,
+                "rc": 4, "cr": 4, "snapshot": 5}
+_MAVEN_RELEASE_ALIASES = {"", "ga", "final", "release"}
+_MAVEN_POST = {"sp": 17}
+_PHP_RANKS = {"dev": 1, "alpha": 2, "a": 2, "beta": 3, "b": 3, "rc": 4}
+_PHP_POST = {"pl": 17, "p": 17}
+_APK_RANKS = {"alpha": 1, "beta": 2, "pre": 3, "rc": 4}
+_APK_POST = {"cvs": 17, "svn": 18, "git": 19, "hg": 20, "p": 21}
+_GEM_RANKS = _SEMVER_RANKS  # letter segments compare lexically, same table
+
+_NUMERIC_DOTTED = re.compile(r"^\d+(?:\.\d+)*$")
+# X.Y[.Z[.W]] optionally followed by -tag[.N] / -tagN / .tagN
+_GENERIC_RE = re.compile(
+    r"^v?(\d+)(?:\.(\d+))?(?:\.(\d+))?(?:\.(\d+))?"
+    r"(?:[-._]([A-Za-z]+)[-._]?(\d+)?)?$"
+)
+
+
+def _pack(epoch: int, n1: int, n2: int, n3: int, n4: int, rank: int, num: int) -> tuple[int, int]:
+    hi = (epoch << 56) | (n1 << 28) | n2
+    lo = (n3 << 36) | (n4 << 22) | (rank << 17) | num
+    return hi, lo
+
+
+def _fits(epoch: int, n1: int, n2: int, n3: int, n4: int, num: int) -> bool:
+    return (
+        0 <= epoch <= 255 and 0 <= n1 <= _N1_MAX and 0 <= n2 <= _N2_MAX
+        and 0 <= n3 <= _N3_MAX and 0 <= n4 <= _N4_MAX and 0 <= num <= _NUM_MAX
+    )
+
+
+def _rank_tables(eco: str) -> Optional[tuple[dict, dict]]:
+    """(pre_ranks, post_ranks) for an ecosystem, or None when unsupported."""
+    if eco in ("npm", "npmjs", "yarn", "pnpm", "node", "javascript", "js", "cargo",
+               "crates.io", "hex", "pub", "swifturl", "go", "nuget"):
+        return _SEMVER_RANKS, {}
+    if eco == "pypi":
+        return _PYPI_RANKS, _PYPI_POST
+    if eco == "maven":
+        return _MAVEN_RANKS, _MAVEN_POST
+    if eco in ("packagist", "composer", "php"):
+        return _PHP_RANKS, _PHP_POST
+    if eco == "apk":
+        return _APK_RANKS, _APK_POST
+    if eco in ("rubygems", "gem", "gems"):
+        return _GEM_RANKS, {}
+    if eco in ("deb", "rpm"):
+        return {}, {}  # numeric-only forms
+    return None
+
+
+@lru_cache(maxsize=65536)
+def encode_version(version: str, ecosystem: str) -> tuple[int, int, bool]:
+    """Encode a version to ``(hi, lo, ok)``; ``ok=False`` → CPU fallback."""
+    eco = ECO_ALIASES.get((ecosystem or "").lower(), (ecosystem or "").lower())
+    v = (version or "").strip()
+    if not v or _looks_like_commit_sha(v):
+        return 0, 0, False
+    tables = _rank_tables(eco)
+    if tables is None:
+        return 0, 0, False
+    pre_ranks, post_ranks = tables
+
+    epoch = 0
+    revision = None
+
+    # Epoch (deb/rpm/pypi "N:" / "N!")
+    m = re.match(r"^(\d+)[:!](.*)$", v)
+    if m:
+        epoch = int(m.group(1))
+        v = m.group(2)
+        if epoch > 255:
+            return 0, 0, False
+
+    if eco in ("deb", "apk"):
+        # Trailing revision: deb "-N", apk "-rN"
+        if eco == "apk" and "-r" in v:
+            base, _, rev = v.rpartition("-r")
+            if rev.isdigit():
+                revision = int(rev)
+                v = base
+        elif eco == "deb" and "-" in v:
+            base, _, rev = v.rpartition("-")
+            if rev.isdigit():
+                revision = int(rev)
+                v = base
+            else:
+                return 0, 0, False
+        if "~" in v or "+" in v:
+            return 0, 0, False
+        if eco == "apk":
+            # apk significance: core < suffix(_alpha.._p) < revision(-rN), so
+            # the revision packs into the LOW 8 bits of num under the suffix
+            # ordinal (suffix N+1 in the high bits).  rev<256, suffixN<511.
+            rev = revision or 0
+            if rev > 255:
+                return 0, 0, False
+            core, _, suffix = v.partition("_")
+            rank, snum = RELEASE_RANK, 0
+            if suffix:
+                sm = re.match(r"^([a-z]+)(\d+)?$", suffix)
+                if not sm:
+                    return 0, 0, False
+                rank = _APK_RANKS.get(sm.group(1)) or _APK_POST.get(sm.group(1)) or 0
+                if rank == 0:
+                    return 0, 0, False
+                snum = int(sm.group(2)) + 1 if sm.group(2) else 0
+                if snum > 511:
+                    return 0, 0, False
+            return _encode_numeric(core, epoch, None, rank, (snum << 8) | rev)
+        return _encode_numeric(v, epoch, revision, RELEASE_RANK, 0)
+
+    if eco == "rpm":
+        if not _NUMERIC_DOTTED.match(v):
+            return 0, 0, False
+        return _encode_numeric(v, epoch, None, RELEASE_RANK, 0)
+
+    if eco == "go" and _looks_like_go_pseudo(v):
+        return 0, 0, False
+
+    if eco == "pypi":
+        return _encode_pypi(v, epoch)
+
+    # Generic X.Y.Z[-tag[.N]] families (semver, maven, php, gem, nuget)
+    m = _GENERIC_RE.match(v)
+    if m is None:
+        return 0, 0, False
+    n1 = int(m.group(1))
+    n2 = int(m.group(2)) if m.group(2) else 0
+    n3 = int(m.group(3)) if m.group(3) else 0
+    n4 = int(m.group(4)) if m.group(4) else 0
+    tag = m.group(5)
+    tagnum = m.group(6)
+    if tag is None:
+        rank, num = RELEASE_RANK, 0
+        if tagnum is not None:
+            return 0, 0, False
+    else:
+        t = tag.lower()
+        if eco == "maven" and t in _MAVEN_RELEASE_ALIASES:
+            rank, num = RELEASE_RANK, (int(tagnum) + 1 if tagnum else 0)
+        else:
+            r = pre_ranks.get(t) or post_ranks.get(t)
+            if r is None:
+                return 0, 0, False
+            rank = r
+            num = int(tagnum) + 1 if tagnum else 0
+    if not _fits(epoch, n1, n2, n3, n4, num):
+        return 0, 0, False
+    return (*_pack(epoch, n1, n2, n3, n4, rank, num), True)
+
+
+_GO_PSEUDO = re.compile(r"^v?\d+\.\d+\.\d+-(?:[0-9A-Za-z.\-]+\.)?\d{14}-[0-9a-f]{12}$")
+
+
+def _looks_like_go_pseudo(v: str) -> bool:
+    return _GO_PSEUDO.match(v) is not None
+
+
+def _encode_numeric(core: str, epoch: int, revision: Optional[int], rank: int, num: int) -> tuple[int, int, bool]:
+    if not _NUMERIC_DOTTED.match(core):
+        return 0, 0, False
+    parts = [int(p) for p in core.split(".")]
+    if len(parts) > 3:
+        return 0, 0, False
+    while len(parts) < 3:
+        parts.append(0)
+    n4 = revision or 0
+    if not _fits(epoch, parts[0], parts[1], parts[2], n4, num):
+        return 0, 0, False
+    return (*_pack(epoch, parts[0], parts[1], parts[2], n4, rank, num), True)
+
+
+_PYPI_RE = re.compile(
+    r"^v?(\d+)(?:\.(\d+))?(?:\.(\d+))?(?:\.(\d+))?"
+    r"(?:\.?(a|alpha|b|beta|c|rc|pre|preview)\.?(\d+)?)?"
+    r"(?:\.?(post|rev|r)(\d+)?)?"
+    r"(?:\.?dev(\d+)?)?$",
+    re.IGNORECASE,
+)
+
+
+def _encode_pypi(v: str, epoch: int) -> tuple[int, int, bool]:
+    if "+" in v:  # local versions: CPU fallback
+        return 0, 0, False
+    m = _PYPI_RE.match(v)
+    if m is None:
+        return 0, 0, False
+    n1 = int(m.group(1))
+    n2 = int(m.group(2)) if m.group(2) else 0
+    n3 = int(m.group(3)) if m.group(3) else 0
+    n4 = int(m.group(4)) if m.group(4) else 0
+    pre_tag, pre_n = m.group(5), m.group(6)
+    post_tag, post_n = m.group(7), m.group(8)
+    dev_n = m.group(9)
+    has_dev = v.lower().find("dev") != -1
+    # Only one suffix class encodable: dev-of-pre / pre-with-post etc. fall back.
+    n_classes = sum([pre_tag is not None, post_tag is not None, has_dev])
+    if n_classes > 1:
+        return 0, 0, False
+    if pre_tag is not None:
+        t = pre_tag.lower()
+        canonical = {"alpha": "a", "beta": "b", "c": "rc", "pre": "rc", "preview": "rc"}.get(t, t)
+        rank = _PYPI_RANKS.get(canonical)
+        if rank is None:
+            return 0, 0, False
+        num = int(pre_n) + 1 if pre_n else 0
+    elif post_tag is not None:
+        rank = 17
+        num = int(post_n) + 1 if post_n else 0
+    elif has_dev:
+        rank = _PYPI_RANKS["dev"]
+        num = int(dev_n) + 1 if dev_n else 0
+    else:
+        rank, num = RELEASE_RANK, 0
+    if not _fits(epoch, n1, n2, n3, n4, num):
+        return 0, 0, False
+    return (*_pack(epoch, n1, n2, n3, n4, rank, num), True)
+
+
+def compare_keys(a: tuple[int, int], b: tuple[int, int]) -> int:
+    """u128 comparison of two (hi, lo) keys."""
+    if a[0] != b[0]:
+        return (a[0] > b[0]) - (a[0] < b[0])
+    return (a[1] > b[1]) - (a[1] < b[1])
+
+
+# ── Batch (numpy) encoding for the scan/bench data path ─────────────────────
+
+
+def pack_batch(epoch, n1, n2, n3, n4, rank, num):
+    """Vectorized key packing from numpy integer arrays → (hi, lo) uint64."""
+    import numpy as np
+
+    hi = (
+        (epoch.astype(np.uint64) << np.uint64(56))
+        | (n1.astype(np.uint64) << np.uint64(28))
+        | n2.astype(np.uint64)
+    )
+    lo = (
+        (n3.astype(np.uint64) << np.uint64(36))
+        | (n4.astype(np.uint64) << np.uint64(22))
+        | (rank.astype(np.uint64) << np.uint64(17))
+        | num.astype(np.uint64)
+    )
+    return hi, lo
