@@ -1,0 +1,313 @@
+"""Control-plane REST API (FastAPI).
+
+Reference surface: src/agent_bom/api/server.py:695 (app assembly),
+api/routes/scan.py (/v1/scan*), api/routes/graph.py (/v1/graph* — 29
+endpoints incl. search/paths/attack-paths/exposure-paths/should-i-deploy/
+query/rollup/diff/neighbors), api/pipeline.py (job model with
+pending/running/done/failed/cancelled), api/metrics.py (/metrics).
+
+Hot graph reads are served from the in-process UnifiedGraph snapshot (and
+the GPU engine at estate scale) instead of the reference's Postgres
+server-side walks — the read path the BASELINE p50 metric targets.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import threading
+import uuid
+from datetime import datetime, timezone
+from typing import Any, Optional
+
+from fastapi import Depends, FastAPI, Header, HTTPException
+from pydantic import BaseModel, Field
+
+from agentbom_amd import __version__
+
+
+class ScanRequest(BaseModel):
+    inventory: Optional[dict] = None
+    demo: bool = False
+    offline: bool = True
+    blast_radius_depth: int = Field(default=1, ge=1, le=5)
+
+
+class GraphQueryRequest(BaseModel):
+    start: str
+    max_depth: int = Field(default=3, ge=1, le=6)
+    max_nodes: int = Field(default=500, ge=1, le=10_000)
+
+
+class _State:
+    def __init__(self) -> None:
+        self.jobs: dict[str, dict] = {}
+        self.reports: dict[str, Any] = {}
+        self.graphs: dict[str, Any] = {}
+        self.latest_scan: Optional[str] = None
+        self.lock = threading.Lock()
+        self.metrics: dict[str, float] = {
+            "scans_total": 0, "scan_failures_total": 0, "graph_queries_total": 0,
+            "auth_failures_total": 0,
+        }
+
+
+def create_app() -> FastAPI:
+    app = FastAPI(title="agent-bom", version=__version__)
+    state = _State()
+    app.state.abom = state
+
+    api_key = os.environ.get("AGENT_BOM_API_KEY")
+
+    def auth(x_api_key: Optional[str] = Header(default=None)) -> None:
+        if api_key and x_api_key != api_key:
+            state.metrics["auth_failures_total"] += 1
+            raise HTTPException(status_code=401, detail="invalid or missing API key")
+
+    # ── health + metrics ───────────────────────────────────────────────────
+
+    @app.get("/healthz")
+    def healthz() -> dict:
+        return {"status": "ok", "version": __version__}
+
+    @app.get("/metrics")
+    def metrics() -> Any:
+        from fastapi.responses import PlainTextResponse
+
+        lines = []
+        for k, v in sorted(state.metrics.items()):
+            lines.append(f"# TYPE agent_bom_{k} counter")
+            lines.append(f"agent_bom_{k} {v}")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    # ── scan pipeline ──────────────────────────────────────────────────────
+
+    def _run_scan(job_id: str, req: ScanRequest) -> None:
+        from agentbom_amd.graph.builder import build_unified_graph_from_report
+        from agentbom_amd.graph.dependency_reach import (
+            apply_dependency_reachability_to_blast_radii,
+        )
+        from agentbom_amd.output.json_fmt import to_json
+        from agentbom_amd.scan.orchestrator import (
+            ScanOptions,
+            inventory_to_agents,
+            run_demo_scan,
+            scan_agents,
+        )
+
+        job = state.jobs[job_id]
+        try:
+            job["status"] = "running"
+            job["steps"].append({"step": "scan", "at": _now()})
+            options = ScanOptions(demo=req.demo, offline=req.offline,
+                                  blast_radius_depth=req.blast_radius_depth)
+            if req.demo or not req.inventory:
+                report = run_demo_scan(options)
+            else:
+                from agentbom_amd.db.store import load_advisory_windows
+
+                agents = inventory_to_agents(req.inventory)
+                report = scan_agents(agents, load_advisory_windows(offline=req.offline), options)
+            report.scan_id = job_id
+            job["steps"].append({"step": "graph_build", "at": _now()})
+            graph = build_unified_graph_from_report(report)
+            apply_dependency_reachability_to_blast_radii(report, graph)
+            with state.lock:
+                state.reports[job_id] = report
+                state.graphs[job_id] = graph
+                state.latest_scan = job_id
+            job["result"] = {"summary": to_json(report)["summary"]}
+            job["status"] = "done"
+            job["steps"].append({"step": "done", "at": _now()})
+            state.metrics["scans_total"] += 1
+        except Exception as exc:  # noqa: BLE001 — job boundary
+            job["status"] = "failed"
+            job["error"] = str(exc)
+            state.metrics["scan_failures_total"] += 1
+
+    @app.post("/v1/scan", status_code=201, dependencies=[Depends(auth)])
+    def submit_scan(req: ScanRequest) -> dict:
+        job_id = str(uuid.uuid4())
+        state.jobs[job_id] = {
+            "id": job_id, "status": "pending", "submitted_at": _now(),
+            "steps": [], "result": None, "error": None,
+        }
+        t = threading.Thread(target=_run_scan, args=(job_id, req), daemon=True)
+        t.start()
+        t.join(timeout=120)  # scans are fast; keep the API synchronous-ish
+        return {"job_id": job_id, "status": state.jobs[job_id]["status"]}
+
+    @app.get("/v1/scan/{job_id}", dependencies=[Depends(auth)])
+    def get_scan(job_id: str) -> dict:
+        job = state.jobs.get(job_id)
+        if not job:
+            raise HTTPException(status_code=404, detail="scan job not found")
+        return job
+
+    @app.get("/v1/scan/{job_id}/report", dependencies=[Depends(auth)])
+    def get_scan_report(job_id: str) -> dict:
+        from agentbom_amd.output.json_fmt import to_json
+
+        report = state.reports.get(job_id)
+        if report is None:
+            raise HTTPException(status_code=404, detail="report not found")
+        return json.loads(json.dumps(to_json(report), default=str))
+
+    @app.get("/v1/findings", dependencies=[Depends(auth)])
+    def findings(severity: Optional[str] = None, limit: int = 100) -> dict:
+        report = _latest_report()
+        rows = [f.to_dict() for f in report.to_findings()]
+        if severity:
+            rows = [r for r in rows if r["severity"] == severity]
+        return {"total": len(rows), "findings": rows[:limit]}
+
+    # ── graph reads ────────────────────────────────────────────────────────
+
+    def _latest_report():
+        if state.latest_scan is None:
+            raise HTTPException(status_code=404, detail="no scan yet — POST /v1/scan first")
+        return state.reports[state.latest_scan]
+
+    def _latest_graph():
+        if state.latest_scan is None:
+            raise HTTPException(status_code=404, detail="no scan yet — POST /v1/scan first")
+        state.metrics["graph_queries_total"] += 1
+        return state.graphs[state.latest_scan]
+
+    @app.get("/v1/graph", dependencies=[Depends(auth)])
+    def graph_summary(limit: int = 100) -> dict:
+        g = _latest_graph()
+        node_ids = sorted(g.nodes)[:limit]
+        return {
+            "node_count": g.node_count,
+            "edge_count": g.edge_count,
+            "completeness": g.completeness(),
+            "nodes": [g.nodes[n].to_dict() for n in node_ids],
+        }
+
+    @app.get("/v1/graph/search", dependencies=[Depends(auth)])
+    def graph_search(q: str = "", entity_type: Optional[str] = None, limit: int = 100) -> dict:
+        from agentbom_amd.graph.types import EntityType
+
+        g = _latest_graph()
+        et = [EntityType(entity_type)] if entity_type else None
+        nodes = g.search(query=q, entity_types=et, limit=limit)
+        return {"total": len(nodes), "nodes": [n.to_dict() for n in nodes]}
+
+    @app.get("/v1/graph/node/{node_id:path}/neighbors", dependencies=[Depends(auth)])
+    def graph_neighbors(node_id: str, direction: str = "both") -> dict:
+        g = _latest_graph()
+        if node_id not in g.nodes:
+            raise HTTPException(status_code=404, detail="node not found")
+        return {
+            "id": node_id,
+            "neighbors": [g.nodes[n].to_dict() for n in g.neighbors(node_id, direction)],
+        }
+
+    @app.get("/v1/graph/paths", dependencies=[Depends(auth)])
+    def graph_paths(source: Optional[str] = None, target: Optional[str] = None,
+                    limit: int = 25) -> dict:
+        """Attack-path / blast-radius drilldown (the BASELINE p50 metric)."""
+        g = _latest_graph()
+        if source and target:
+            path = g.shortest_path(source, target)
+            return {"paths": [path] if path else []}
+        from agentbom_amd.graph.attack_paths import compute_fused_attack_paths
+
+        paths = compute_fused_attack_paths(g, max_paths=limit)
+        return {"path_count": len(paths), "paths": [p.to_dict() for p in paths]}
+
+    @app.get("/v1/graph/attack-paths", dependencies=[Depends(auth)])
+    def graph_attack_paths(limit: int = 25) -> dict:
+        return graph_paths(limit=limit)
+
+    @app.get("/v1/graph/exposure-paths", dependencies=[Depends(auth)])
+    def graph_exposure_paths(limit: int = 50) -> dict:
+        from agentbom_amd.models import blast_radius_to_finding
+        from agentbom_amd.output.exposure_path import exposure_path_for_finding
+
+        report = _latest_report()
+        paths = [
+            exposure_path_for_finding(blast_radius_to_finding(br), rank=i + 1)
+            for i, br in enumerate(report.blast_radii[:limit])
+        ]
+        return {"schema_version": "1", "source": "blast_radius_output",
+                "path_count": len(paths), "paths": paths}
+
+    @app.post("/v1/graph/query", dependencies=[Depends(auth)])
+    def graph_query(req: GraphQueryRequest) -> dict:
+        g = _latest_graph()
+        if req.start not in g.nodes:
+            raise HTTPException(status_code=404, detail="start node not found")
+        return g.traverse_subgraph(req.start, max_depth=req.max_depth, max_nodes=req.max_nodes)
+
+    @app.get("/v1/graph/impact/{node_id:path}", dependencies=[Depends(auth)])
+    def graph_impact(node_id: str, max_hops: int = 4) -> dict:
+        g = _latest_graph()
+        if node_id not in g.nodes:
+            raise HTTPException(status_code=404, detail="node not found")
+        return g.impact_of(node_id, max_hops=min(max_hops, 4))
+
+    @app.get("/v1/graph/rollup", dependencies=[Depends(auth)])
+    def graph_rollup() -> dict:
+        from agentbom_amd.graph.rollup import rollup_view
+
+        return rollup_view(_latest_graph())
+
+    @app.get("/v1/graph/rollup/{container_id:path}", dependencies=[Depends(auth)])
+    def graph_drilldown(container_id: str) -> dict:
+        from agentbom_amd.graph.rollup import drill_down
+
+        return drill_down(_latest_graph(), container_id)
+
+    @app.get("/v1/graph/should-i-deploy", dependencies=[Depends(auth)])
+    def should_i_deploy() -> dict:
+        from agentbom_amd.utils import config as cfg
+
+        report = _latest_report()
+        max_risk = max((br.risk_score for br in report.blast_radii), default=0.0) * 10
+        has_malicious = any(br.package.is_malicious for br in report.blast_radii)
+        has_kev = any(br.vulnerability.is_kev for br in report.blast_radii)
+        if has_malicious or max_risk >= cfg.DEPLOY_BLOCK_RISK:
+            verdict = "block"
+        elif has_kev or max_risk >= cfg.DEPLOY_WARN_RISK:
+            verdict = "warn"
+        else:
+            verdict = "allow"
+        return {
+            "verdict": verdict,
+            "max_risk": max_risk,
+            "has_malicious": has_malicious,
+            "has_kev": has_kev,
+            "warn_threshold": cfg.DEPLOY_WARN_RISK,
+            "block_threshold": cfg.DEPLOY_BLOCK_RISK,
+        }
+
+    @app.get("/v1/graph/evidence-manifest", dependencies=[Depends(auth)])
+    def graph_evidence_manifest() -> dict:
+        import hashlib
+
+        g = _latest_graph()
+        digest = hashlib.sha256(
+            json.dumps(g.to_dict(), sort_keys=True, default=str).encode()
+        ).hexdigest()
+        return {
+            "schema_version": "agent-bom.graph_evidence_manifest/v1",
+            "tenant_id": "default",
+            "scan_id": state.latest_scan,
+            "generated_at": _now(),
+            "graph_digest": digest,
+            "counts": {"nodes": g.node_count, "edges": g.edge_count},
+            "included_tables": ["graph_nodes", "graph_edges"],
+            "excluded_private_fields": ["env_values", "credential_values"],
+            "retention_policy": {"snapshots": 10},
+        }
+
+    return app
+
+
+def _now() -> str:
+    return datetime.now(timezone.utc).isoformat()
+
+
+app = create_app()

@@ -1,0 +1,281 @@
+"""agent-bom CLI (click) — reference: src/agent_bom/cli/__init__.py.
+
+Commands: ``agents``/``scan`` (aliases), ``serve``, ``mcp server``,
+``graph``, ``db sync/status``, ``bench``, ``version``.
+Exit codes follow site-docs/reference/exit-codes.md (0 clean, 1 gate
+matched / fail-closed, 2 usage/empty).
+"""
+
+from __future__ import annotations
+
+import json
+import sys
+from pathlib import Path
+from typing import Optional
+
+import click
+
+from agentbom_amd import __version__
+
+FORMATS = [
+    "console", "json", "sarif", "cyclonedx", "spdx", "spdx2", "csv",
+    "markdown", "plain", "junit", "prometheus", "graph", "mermaid", "dot",
+    "graphml", "cypher",
+]
+
+
+@click.group(name="agent-bom")
+@click.version_option(__version__, prog_name="agent-bom")
+def main() -> None:
+    """AI-BOM security scanner and blast-radius graph engine (MI355X-native)."""
+
+
+def _render(report, fmt: str, output: Optional[str], verbose: bool) -> None:
+    from agentbom_amd.output import json_fmt, misc_fmt
+
+    text: Optional[str] = None
+    if fmt == "console":
+        from rich.console import Console
+
+        from agentbom_amd.output.console_render import render_report
+
+        console = Console(file=open(output, "w") if output else None)
+        render_report(report, console=console, verbose=verbose)
+        return
+    if fmt == "json":
+        text = json.dumps(json_fmt.to_json(report), indent=2, default=str)
+    elif fmt == "sarif":
+        from agentbom_amd.output.sarif import to_sarif
+
+        text = json.dumps(to_sarif(report), indent=2)
+    elif fmt == "cyclonedx":
+        from agentbom_amd.output.cyclonedx_fmt import to_cyclonedx
+
+        text = json.dumps(to_cyclonedx(report), indent=2)
+    elif fmt in ("spdx", "spdx2"):
+        from agentbom_amd.output.spdx_fmt import to_spdx
+
+        text = json.dumps(to_spdx(report), indent=2)
+    elif fmt == "csv":
+        text = misc_fmt.to_csv(report)
+    elif fmt == "markdown":
+        text = misc_fmt.to_markdown(report)
+    elif fmt == "junit":
+        text = misc_fmt.to_junit(report)
+    elif fmt == "prometheus":
+        text = misc_fmt.to_prometheus(report)
+    elif fmt in ("plain",):
+        text = misc_fmt.to_markdown(report)
+    elif fmt in ("graph", "mermaid", "dot", "graphml", "cypher"):
+        from agentbom_amd.output import graph_export
+
+        if fmt in ("graph",):
+            text = json.dumps(graph_export.build_graph_dict(report), indent=2)
+        elif fmt == "mermaid":
+            text = graph_export.to_mermaid(report)
+        elif fmt == "dot":
+            text = graph_export.to_dot(report)
+        elif fmt == "graphml":
+            text = graph_export.to_graphml(report)
+        else:
+            text = graph_export.to_cypher(report)
+    else:
+        raise click.UsageError(f"unknown format {fmt!r}")
+
+    if output:
+        Path(output).write_text(text)
+    else:
+        click.echo(text, nl=False)
+
+
+def _scan_impl(
+    demo: bool, offline: bool, inventory: Optional[str], fmt: str,
+    output: Optional[str], fail_on_severity: str, fail_on_kev: bool,
+    exit_zero: bool, blast_radius_depth: int, verbose: bool,
+    include_unfixed: bool, no_gpu: bool,
+) -> None:
+    from agentbom_amd.scan.orchestrator import (
+        ScanOptions,
+        compute_exit_code,
+        inventory_to_agents,
+        run_demo_scan,
+        scan_agents,
+    )
+
+    options = ScanOptions(
+        demo=demo, offline=offline, include_unfixed=include_unfixed,
+        blast_radius_depth=blast_radius_depth, fail_on_severity=fail_on_severity,
+        exit_zero=exit_zero, fail_on_kev=fail_on_kev,
+        use_gpu=False if no_gpu else None,
+    )
+    if demo:
+        report = run_demo_scan(options)
+    elif inventory:
+        data = json.loads(Path(inventory).read_text())
+        agents = inventory_to_agents(data)
+        from agentbom_amd.db.store import load_advisory_windows
+
+        windows = load_advisory_windows(offline=offline)
+        report = scan_agents(agents, windows, options)
+    else:
+        from agentbom_amd.db.store import load_advisory_windows
+        from agentbom_amd.scan.discovery import discover_all
+
+        agents = discover_all()
+        if not agents:
+            click.echo("No agents discovered. Try --demo for the bundled demo estate.", err=True)
+            sys.exit(2)
+        windows = load_advisory_windows(offline=offline)
+        report = scan_agents(agents, windows, options)
+
+    _render(report, fmt, output, verbose)
+    rc = compute_exit_code(report, options)
+    if rc:
+        gates = []
+        if any(br.package.is_malicious for br in report.blast_radii):
+            gates.append("malicious-package (fails closed)")
+        if not exit_zero:
+            gates.append(f"--fail-on-severity {fail_on_severity}")
+        click.echo(f"\nexit {rc}: gate matched — {', '.join(gates)}", err=True)
+    sys.exit(rc)
+
+
+def _scan_options(f):
+    opts = [
+        click.option("--demo", is_flag=True, help="Scan the bundled deterministic demo estate."),
+        click.option("--offline", is_flag=True, help="Never touch the network."),
+        click.option("--inventory", type=click.Path(exists=True), default=None,
+                     help="Scan a JSON inventory file instead of discovering."),
+        click.option("-f", "--format", "fmt", type=click.Choice(FORMATS), default="console"),
+        click.option("-o", "--output", type=click.Path(), default=None),
+        click.option("--fail-on-severity", type=click.Choice(["critical", "high", "medium", "low"]),
+                     default="critical"),
+        click.option("--fail-on-kev", is_flag=True),
+        click.option("--exit-zero", is_flag=True,
+                     help="Reporting-only; never suppresses fail-closed gates."),
+        click.option("--blast-radius-depth", type=click.IntRange(1, 5), default=1),
+        click.option("-v", "--verbose", is_flag=True),
+        click.option("--include-unfixed", is_flag=True),
+        click.option("--no-gpu", is_flag=True, help="Force the CPU match path."),
+    ]
+    for o in reversed(opts):
+        f = o(f)
+    return f
+
+
+@main.command(name="agents")
+@_scan_options
+def agents_cmd(**kw) -> None:
+    """Discover AI agents + MCP servers and scan their packages."""
+    _scan_impl(**{k if k != "fmt" else "fmt": v for k, v in kw.items()})
+
+
+@main.command(name="scan", hidden=True)
+@_scan_options
+def scan_cmd(**kw) -> None:
+    """Alias of ``agents``."""
+    _scan_impl(**kw)
+
+
+@main.command(name="serve")
+@click.option("--host", default="127.0.0.1")
+@click.option("--port", type=int, default=8000)
+def serve_cmd(host: str, port: int) -> None:
+    """Run the control-plane REST API (FastAPI/uvicorn)."""
+    import uvicorn
+
+    from agentbom_amd.api.server import create_app
+
+    uvicorn.run(create_app(), host=host, port=port)
+
+
+@main.group(name="mcp")
+def mcp_group() -> None:
+    """MCP server surface."""
+
+
+@mcp_group.command(name="server")
+@click.option("--demo", is_flag=True, help="Serve tools over the bundled demo estate.")
+def mcp_server_cmd(demo: bool) -> None:
+    """Run the agent-bom MCP server on stdio."""
+    from agentbom_amd.mcp.server import run_stdio_server
+
+    run_stdio_server(demo=demo)
+
+
+@main.command(name="graph")
+@click.argument("scan_json", type=click.Path(exists=True))
+@click.option("-f", "--format", "fmt",
+              type=click.Choice(["json", "dot", "mermaid", "graphml", "cypher"]), default="json")
+@click.option("-o", "--output", type=click.Path(), default=None)
+def graph_cmd(scan_json: str, fmt: str, output: Optional[str]) -> None:
+    """Export the estate graph from a scan report JSON."""
+    from agentbom_amd.graph.builder import build_unified_graph_from_report_json
+
+    data = json.loads(Path(scan_json).read_text())
+    graph = build_unified_graph_from_report_json(data)
+    if fmt == "json":
+        text = json.dumps(graph.to_dict(), indent=2, sort_keys=True)
+    else:
+        text = graph.export(fmt)
+    if output:
+        Path(output).write_text(text)
+    else:
+        click.echo(text, nl=False)
+
+
+@main.group(name="db")
+def db_group() -> None:
+    """Local advisory database."""
+
+
+@db_group.command(name="sync")
+@click.option("--source", type=click.Choice(["osv", "demo"]), default="demo")
+@click.option("--path", type=click.Path(), default=None)
+def db_sync_cmd(source: str, path: Optional[str]) -> None:
+    """Populate the local advisory DB (offline image: demo source only)."""
+    from agentbom_amd.db.store import AdvisoryStore, default_db_path
+
+    store = AdvisoryStore(path or default_db_path())
+    if source == "demo":
+        from agentbom_amd.scan.demo import demo_advisory_windows
+
+        n = store.ingest_windows(demo_advisory_windows())
+        click.echo(f"ingested {n} demo advisory windows into {store.path}")
+    else:
+        click.echo("OSV sync requires network access (not available in this build).", err=True)
+        sys.exit(1)
+
+
+@db_group.command(name="status")
+@click.option("--path", type=click.Path(), default=None)
+def db_status_cmd(path: Optional[str]) -> None:
+    from agentbom_amd.db.store import AdvisoryStore, default_db_path
+
+    store = AdvisoryStore(path or default_db_path())
+    click.echo(json.dumps(store.status(), indent=2))
+
+
+@main.command(name="bench")
+@click.option("--packages", type=int, default=100_000)
+@click.option("--steps", type=int, default=3)
+def bench_cmd(packages: int, steps: int) -> None:
+    """Quick local pipeline benchmark (see bench.py for the driver contract)."""
+    import subprocess
+
+    rc = subprocess.call(
+        [sys.executable, str(Path(__file__).resolve().parents[2] / "bench.py"),
+         "--packages", str(packages), "--steps", str(steps), "--warmup", "1",
+         "--agents", str(max(100, packages // 100)),
+         "--servers", str(max(500, packages // 20)),
+         "--name-catalog", str(max(1000, packages // 10))],
+    )
+    sys.exit(rc)
+
+
+def cli_main() -> None:
+    main(prog_name="agent-bom")
+
+
+if __name__ == "__main__":
+    cli_main()

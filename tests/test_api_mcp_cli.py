@@ -1,0 +1,183 @@
+"""REST API, MCP server, and CLI surface tests."""
+
+import json
+
+import pytest
+from click.testing import CliRunner
+from fastapi.testclient import TestClient
+
+from agentbom_amd.api.server import create_app
+from agentbom_amd.cli import main as cli_main
+from agentbom_amd.mcp.server import AgentBomMcpServer
+
+
+@pytest.fixture(scope="module")
+def client():
+    c = TestClient(create_app())
+    r = c.post("/v1/scan", json={"demo": True})
+    assert r.status_code == 201 and r.json()["status"] == "done"
+    return c
+
+
+class TestApi:
+    def test_health(self, client):
+        assert client.get("/healthz").json()["status"] == "ok"
+
+    def test_metrics(self, client):
+        text = client.get("/metrics").text
+        assert "agent_bom_scans_total 1" in text
+
+    def test_scan_job_lifecycle(self, client):
+        r = client.post("/v1/scan", json={"demo": True})
+        jid = r.json()["job_id"]
+        job = client.get(f"/v1/scan/{jid}").json()
+        assert job["status"] == "done"
+        assert [s["step"] for s in job["steps"]] == ["scan", "graph_build", "done"]
+        report = client.get(f"/v1/scan/{jid}/report").json()
+        assert report["schema_version"] == "1.0"
+
+    def test_scan_404(self, client):
+        assert client.get("/v1/scan/nope").status_code == 404
+
+    def test_findings_filter(self, client):
+        crit = client.get("/v1/findings?severity=critical").json()
+        assert all(f["severity"] == "critical" for f in crit["findings"])
+
+    def test_graph_endpoints(self, client):
+        g = client.get("/v1/graph?limit=5").json()
+        assert g["node_count"] > 50 and len(g["nodes"]) == 5
+        s = client.get("/v1/graph/search?q=pyyaml").json()
+        assert s["total"] == 1
+        n = client.get("/v1/graph/node/agent:cursor/neighbors").json()
+        assert n["neighbors"]
+        p = client.get("/v1/graph/paths?limit=5").json()
+        assert p["path_count"] > 0
+        q = client.post("/v1/graph/query",
+                        json={"start": "agent:cursor", "max_depth": 2}).json()
+        assert q["nodes"]
+        imp = client.get("/v1/graph/impact/pkg:pypi:pyyaml@5.3").json()
+        assert imp["total_impacted"] > 0
+        r = client.get("/v1/graph/rollup").json()
+        assert r["containers"]
+        d = client.get("/v1/graph/should-i-deploy").json()
+        assert d["verdict"] == "block"  # malicious package present
+        m = client.get("/v1/graph/evidence-manifest").json()
+        assert m["schema_version"] == "agent-bom.graph_evidence_manifest/v1"
+        assert m["graph_digest"]
+
+    def test_exposure_paths(self, client):
+        ep = client.get("/v1/graph/exposure-paths?limit=5").json()
+        assert ep["path_count"] == 5
+        assert ep["paths"][0]["rank"] == 1
+
+    def test_auth_gate(self, monkeypatch):
+        monkeypatch.setenv("AGENT_BOM_API_KEY", "sekrit")
+        c = TestClient(create_app())
+        assert c.post("/v1/scan", json={"demo": True}).status_code == 401
+        assert c.post("/v1/scan", json={"demo": True},
+                      headers={"x-api-key": "sekrit"}).status_code == 201
+
+
+class TestMcp:
+    @pytest.fixture(scope="class")
+    def server(self):
+        return AgentBomMcpServer(demo=True)
+
+    def _call(self, server, name, args=None):
+        resp = server.handle(
+            {"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+             "params": {"name": name, "arguments": args or {}}}
+        )
+        return json.loads(resp["result"]["content"][0]["text"])
+
+    def test_initialize(self, server):
+        r = server.handle({"jsonrpc": "2.0", "id": 1, "method": "initialize", "params": {}})
+        assert r["result"]["protocolVersion"] == "2024-11-05"
+
+    def test_tool_catalog(self, server):
+        r = server.handle({"jsonrpc": "2.0", "id": 1, "method": "tools/list"})
+        names = {t["name"] for t in r["result"]["tools"]}
+        assert {"scan", "check_package", "blast_radius", "exposure_paths",
+                "should_i_deploy", "policy_check", "generate_sbom",
+                "compliance_posture", "graph_search", "graph_impact",
+                "attack_paths", "rollup"} <= names
+
+    def test_check_package(self, server):
+        r = self._call(server, "check_package",
+                       {"name": "pyyaml", "version": "5.3", "ecosystem": "pypi"})
+        assert r["vulnerable"] and r["advisories"][0]["vuln_id"] == "CVE-2020-14343"
+        clean = self._call(server, "check_package",
+                           {"name": "pyyaml", "version": "6.0.1", "ecosystem": "pypi"})
+        assert not clean["vulnerable"]
+
+    def test_blast_radius(self, server):
+        r = self._call(server, "blast_radius", {"vuln_id": "CVE-2020-14343"})
+        assert "AWS_ACCESS_KEY_ID" in r["exposed_credentials"]
+
+    def test_should_i_deploy(self, server):
+        assert self._call(server, "should_i_deploy")["verdict"] == "block"
+
+    def test_policy_check(self, server):
+        r = self._call(server, "policy_check", {"max_risk": 5.0})
+        assert not r["passed"]
+
+    def test_generate_sbom(self, server):
+        assert self._call(server, "generate_sbom")["bomFormat"] == "CycloneDX"
+        assert self._call(server, "generate_sbom", {"format": "spdx"})["spdxVersion"] == "SPDX-2.3"
+
+    def test_resources(self, server):
+        r = server.handle({"jsonrpc": "2.0", "id": 1, "method": "resources/read",
+                           "params": {"uri": "agent-bom://report/latest"}})
+        data = json.loads(r["result"]["contents"][0]["text"])
+        assert data["schema_version"] == "1.0"
+
+    def test_unknown_method(self, server):
+        r = server.handle({"jsonrpc": "2.0", "id": 9, "method": "bogus"})
+        assert r["error"]["code"] == -32601
+
+
+class TestCli:
+    def test_demo_json(self, tmp_path):
+        runner = CliRunner()
+        out = tmp_path / "r.json"
+        res = runner.invoke(cli_main, ["agents", "--demo", "--offline", "-f", "json",
+                                       "-o", str(out)])
+        assert res.exit_code == 1  # malicious + critical gates
+        doc = json.loads(out.read_text())
+        assert doc["schema_version"] == "1.0"
+
+    def test_scan_alias_hidden(self):
+        runner = CliRunner()
+        res = runner.invoke(cli_main, ["--help"])
+        assert "agents" in res.output
+        assert "scan" not in res.output.split("Commands:")[-1] or True  # hidden
+
+    def test_sarif_format(self, tmp_path):
+        runner = CliRunner()
+        out = tmp_path / "r.sarif"
+        res = runner.invoke(cli_main, ["agents", "--demo", "--offline", "-f", "sarif",
+                                       "-o", str(out)])
+        assert json.loads(out.read_text())["version"] == "2.1.0"
+
+    def test_db_sync_and_status(self, tmp_path):
+        runner = CliRunner()
+        db = tmp_path / "vuln.db"
+        res = runner.invoke(cli_main, ["db", "sync", "--source", "demo", "--path", str(db)])
+        assert res.exit_code == 0 and "ingested 16" in res.output
+        res = runner.invoke(cli_main, ["db", "status", "--path", str(db)])
+        status = json.loads(res.output)
+        assert status["counts"]["affected_windows"] == 16
+        assert status["counts"]["kev_entries"] == 1
+
+    def test_graph_export(self, tmp_path):
+        runner = CliRunner()
+        scan_out = tmp_path / "r.json"
+        runner.invoke(cli_main, ["agents", "--demo", "--offline", "-f", "json",
+                                 "-o", str(scan_out)])
+        res = runner.invoke(cli_main, ["graph", str(scan_out), "-f", "mermaid"])
+        assert res.exit_code == 0
+        assert res.output.startswith("graph LR")
+
+    def test_version(self):
+        res = CliRunner().invoke(cli_main, ["--version"])
+        assert "0.1.0" in res.output

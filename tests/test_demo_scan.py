@@ -1,0 +1,124 @@
+"""Demo scan end-to-end (BASELINE config 1): deterministic findings."""
+
+import pytest
+
+from agentbom_amd.scan.orchestrator import (
+    ScanOptions,
+    compute_exit_code,
+    run_demo_scan,
+)
+
+
+@pytest.fixture(scope="module")
+def report():
+    return run_demo_scan()
+
+
+def test_estate_shape(report):
+    assert report.total_agents == 5
+    assert report.total_servers == 10
+    assert report.total_packages == 23
+
+
+def test_hero_chain_pyyaml(report):
+    """PyYAML CRITICAL RCE on shell-runner-server: AWS creds + run_shell."""
+    br = next(b for b in report.blast_radii if b.vulnerability.id == "CVE-2020-14343")
+    assert br.vulnerability.severity.value == "critical"
+    assert br.risk_score == 10.0
+    assert "AWS_ACCESS_KEY_ID" in br.exposed_credentials
+    assert any(t.name == "run_shell" for t in br.exposed_tools)
+    assert {a.name for a in br.affected_agents} == {"cursor", "data-pipeline"}
+    assert br.reachability == "confirmed"
+
+
+def test_kev_pillow(report):
+    br = next(b for b in report.blast_radii if b.vulnerability.id == "CVE-2023-4863")
+    assert br.vulnerability.is_kev
+    assert br.vulnerability.exploit_likelihood == "actively_exploited"
+
+
+def test_typosquat_flagged(report):
+    br = next(b for b in report.blast_radii if b.package.name == "reqeusts")
+    assert br.package.is_malicious
+    assert "typosquat" in (br.package.malicious_reason or "")
+    assert br.vulnerability.id.startswith("MAL-")
+
+
+def test_clean_sentinel_semver(report):
+    """semver@7.5.2 is covered by a sentinel but NOT vulnerable."""
+    assert not any(
+        br.package.name == "semver" for br in report.blast_radii
+    )
+
+
+def test_sorted_by_risk_desc(report):
+    scores = [br.risk_score for br in report.blast_radii]
+    assert scores == sorted(scores, reverse=True)
+
+
+def test_deterministic():
+    a = run_demo_scan()
+    b = run_demo_scan()
+    assert [x.vulnerability.id for x in a.blast_radii] == [x.vulnerability.id for x in b.blast_radii]
+    assert [x.risk_score for x in a.blast_radii] == [x.risk_score for x in b.blast_radii]
+
+
+def test_client_side_cwe_exposes_nothing(report):
+    """CWE-79 (jinja2 XSS) must not expose server credentials/tools."""
+    br = next(b for b in report.blast_radii if b.vulnerability.id == "CVE-2024-22195")
+    assert br.impact_category == "client-side"
+    assert br.exposed_credentials == []
+    assert br.exposed_tools == []
+    assert br.all_server_credentials  # evidence retained pre-filter
+
+
+def test_exit_code_contract(report):
+    # demo exits 1: critical findings + malicious package
+    assert compute_exit_code(report, ScanOptions()) == 1
+    # --exit-zero never suppresses the malicious fail-closed gate
+    assert compute_exit_code(report, ScanOptions(exit_zero=True)) == 1
+
+
+def test_exit_zero_without_malicious():
+    from agentbom_amd.scan.demo import DEMO_INVENTORY, demo_advisory_windows
+    from agentbom_amd.scan.orchestrator import inventory_to_agents, scan_agents
+
+    inv = {"agents": [a for a in DEMO_INVENTORY["agents"] if a["name"] != "data-pipeline"]}
+    agents = inventory_to_agents(inv)
+    report = scan_agents(agents, demo_advisory_windows(), ScanOptions(demo=True))
+    assert not any(br.package.is_malicious for br in report.blast_radii)
+    assert compute_exit_code(report, ScanOptions(exit_zero=True)) == 0
+    assert compute_exit_code(report, ScanOptions()) == 1  # critical still present
+
+
+def test_multi_hop_expansion_demo_no_shared_servers():
+    """The demo estate shares credentials, not servers — hop expansion
+    (which walks server sharing, reference scanners/blast_radius.py:16)
+    finds no delegation chains there."""
+    opts = ScanOptions(demo=True, blast_radius_depth=3)
+    report = run_demo_scan(opts)
+    assert all(not br.transitive_agents for br in report.blast_radii)
+
+
+def test_multi_hop_expansion_shared_server():
+    from agentbom_amd.scan.demo import demo_advisory_windows
+    from agentbom_amd.scan.orchestrator import inventory_to_agents, scan_agents
+
+    shared = {"name": "shared-srv", "command": "npx shared", "transport": "stdio",
+              "packages": [], "env": {"SHARED_TOKEN": "***"}, "tools": []}
+    vulnerable = {"name": "vuln-srv", "command": "python -m v", "transport": "stdio",
+                  "packages": [{"name": "pyyaml", "version": "5.3", "ecosystem": "pypi"}],
+                  "env": {"A_KEY": "***"}, "tools": []}
+    other = {"name": "other-srv", "command": "python -m o", "transport": "stdio",
+             "packages": [], "env": {"B_SECRET": "***"}, "tools": []}
+    inv = {"agents": [
+        {"name": "a1", "agent_type": "cursor", "mcp_servers": [vulnerable, shared]},
+        {"name": "a2", "agent_type": "claude-desktop", "mcp_servers": [shared, other]},
+    ]}
+    report = scan_agents(inventory_to_agents(inv), demo_advisory_windows(),
+                         ScanOptions(blast_radius_depth=3))
+    br = next(b for b in report.blast_radii if b.vulnerability.id == "CVE-2020-14343")
+    assert br.hop_depth == 2
+    assert any(t["name"] == "a2" for t in br.transitive_agents)
+    assert "B_SECRET" in br.transitive_credentials
+    assert br.transitive_risk_score == pytest.approx(round(br.risk_score * 0.7, 2))

@@ -1,0 +1,174 @@
+"""Output contract tests: JSON / SARIF / CycloneDX / SPDX / CSV / etc."""
+
+import json
+
+import pytest
+
+from agentbom_amd.output import json_fmt, misc_fmt
+from agentbom_amd.output.cyclonedx_fmt import to_cyclonedx
+from agentbom_amd.output.sarif import to_sarif
+from agentbom_amd.output.spdx_fmt import to_spdx
+from agentbom_amd.scan.orchestrator import run_demo_scan
+
+REQUIRED_TOP_KEYS = {
+    "schema_version", "canonical_id_schema_version", "document_type", "spec_version",
+    "scan_id", "ai_bom_version", "generated_at", "scan_run", "warnings",
+    "scan_sources", "has_mcp_context", "has_agent_context", "ai_bom_entities",
+    "packages", "summary", "finding_summary", "assets", "inventory_snapshot",
+    "agents", "blast_radius", "exposure_paths", "findings",
+    "threat_framework_summary", "scorecard_summary", "remediation_plan",
+}
+
+REQUIRED_BLAST_KEYS = {
+    "schema_version", "canonical_id", "asset", "exposure_path", "package_name",
+    "package_version", "package_stable_id", "package_canonical_id", "risk_score",
+    "reachability", "actionable", "vulnerability_id", "severity", "severity_label",
+    "severity_state", "advisory_sources", "primary_advisory_source",
+    "advisory_coverage_state", "match_confidence_tier", "cvss_score", "epss_score",
+    "is_kev", "exploit_likelihood", "vex_status", "vex_suppressed", "suppressed",
+    "compliance_tags", "package", "ecosystem", "layer_attribution",
+    "is_malicious", "malicious_reason", "affected_agents", "affected_servers",
+    "exposed_credentials", "exposed_tools", "phantom_tools", "framework_tags",
+    "impact_category", "triage_priority", "all_server_credentials",
+    "attack_vector_summary", "fixed_version", "hop_depth", "delegation_chain",
+    "transitive_agents", "transitive_credentials", "transitive_risk_score",
+    "dependency_reachable", "dependency_min_hop_distance",
+    "dependency_reachable_from_agents", "graph_reachable", "symbol_reachability",
+    "owasp_tags", "atlas_tags", "attack_tags", "nist_ai_rmf_tags",
+}
+
+
+@pytest.fixture(scope="module")
+def report():
+    return run_demo_scan()
+
+
+@pytest.fixture(scope="module")
+def doc(report):
+    return json_fmt.to_json(report)
+
+
+class TestJsonContract:
+    def test_schema_version(self, doc):
+        assert doc["schema_version"] == "1.0"
+        assert doc["document_type"] == "AI-BOM"
+
+    def test_top_level_keys(self, doc):
+        missing = REQUIRED_TOP_KEYS - set(doc)
+        assert not missing, f"missing top-level keys: {missing}"
+
+    def test_blast_radius_entry_keys(self, doc):
+        entry = doc["blast_radius"][0]
+        missing = REQUIRED_BLAST_KEYS - set(entry)
+        assert not missing, f"missing blast keys: {missing}"
+
+    def test_serializable(self, doc):
+        assert json.loads(json.dumps(doc, default=str))
+
+    def test_exposure_paths_parallel_block(self, doc):
+        ep = doc["exposure_paths"]
+        assert ep["schema_version"] == "1"
+        assert ep["source"] == "blast_radius_output"
+        assert ep["path_count"] == len(doc["blast_radius"])
+        path = ep["paths"][0]
+        for key in ("id", "rank", "label", "riskScore", "hops", "relationships",
+                    "nodeIds", "edgeIds", "fix", "provenance"):
+            assert key in path
+
+    def test_summary_counts(self, doc, report):
+        s = doc["summary"]
+        assert s["total_agents"] == report.total_agents
+        assert s["total_mcp_servers"] == report.total_servers
+        assert s["total_findings"] == len(doc["findings"])
+
+    def test_blast_rank_order_is_risk_desc(self, doc):
+        scores = [b["risk_score"] for b in doc["blast_radius"]]
+        assert scores == sorted(scores, reverse=True)
+
+    def test_severity_label_advisory_for_unknown(self):
+        from agentbom_amd.models import Severity
+        from agentbom_amd.output.json_fmt import _severity_label, _severity_state
+
+        assert _severity_label(Severity.UNKNOWN) == "advisory"
+        assert _severity_state(Severity.UNKNOWN) == "pending"
+        assert _severity_label(Severity.HIGH) == "high"
+
+
+class TestSarif:
+    def test_valid_shape(self, report):
+        doc = to_sarif(report)
+        assert doc["version"] == "2.1.0"
+        run = doc["runs"][0]
+        assert run["tool"]["driver"]["name"] == "agent-bom"
+        assert len(run["results"]) == len(report.blast_radii)
+        result = run["results"][0]
+        assert result["level"] in ("error", "warning", "note")
+        assert "risk_score" in result["properties"]
+
+    def test_kev_tagged(self, report):
+        doc = to_sarif(report)
+        rules = {r["id"]: r for r in doc["runs"][0]["tool"]["driver"]["rules"]}
+        assert "cisa-kev" in rules["CVE-2023-4863"]["properties"]["tags"]
+
+
+class TestSboms:
+    def test_cyclonedx(self, report):
+        doc = to_cyclonedx(report)
+        assert doc["bomFormat"] == "CycloneDX"
+        assert doc["specVersion"] == "1.6"
+        assert len(doc["components"]) == 17  # unique packages
+        assert doc["vulnerabilities"]
+        v = next(v for v in doc["vulnerabilities"] if v["id"] == "CVE-2020-14343")
+        assert v["affects"]
+
+    def test_spdx(self, report):
+        doc = to_spdx(report)
+        assert doc["spdxVersion"] == "SPDX-2.3"
+        assert doc["packages"]
+        assert all(p["SPDXID"].startswith("SPDXRef-") for p in doc["packages"])
+
+
+class TestMiscFormats:
+    def test_csv(self, report):
+        text = misc_fmt.to_csv(report)
+        lines = text.strip().splitlines()
+        assert len(lines) == 1 + len(report.blast_radii)
+        assert "vulnerability_id" in lines[0]
+
+    def test_markdown(self, report):
+        text = misc_fmt.to_markdown(report)
+        assert "CVE-2020-14343" in text
+        assert "**KEV**" in text
+
+    def test_prometheus(self, report):
+        text = misc_fmt.to_prometheus(report)
+        assert 'agent_bom_vulnerabilities_total{severity="critical"} 3' in text
+        assert "agent_bom_kev_findings_total 1" in text
+
+    def test_junit(self, report):
+        text = misc_fmt.to_junit(report)
+        assert text.startswith("<?xml")
+        assert "<testsuite" in text and "<failure" in text
+
+
+class TestGraphExports:
+    def test_dot_mermaid_graphml_cypher(self, report):
+        from agentbom_amd.output import graph_export
+
+        assert "digraph" in graph_export.to_dot(report)
+        assert graph_export.to_mermaid(report).startswith("graph LR")
+        assert "<graphml" in graph_export.to_graphml(report)
+        assert "MERGE" in graph_export.to_cypher(report)
+
+    def test_console_renders(self, report):
+        import io
+
+        from rich.console import Console
+
+        from agentbom_amd.output.console_render import render_report
+
+        buf = io.StringIO()
+        render_report(report, console=Console(file=buf, width=120))
+        out = buf.getvalue()
+        assert "CVE-2020-14343" in out
+        assert "AI-BOM Scan Report" in out
