@@ -184,7 +184,7 @@ def parse_mcp_config(path: Path) -> list[MCPServer]:
     config_path = str(path)
     try:
         if suffix == ".toml":
-            import tomllib
+            from agentbom_amd.utils.compat import tomllib
 
             data = tomllib.loads(text)
             servers = data.get("mcp_servers") or data.get("mcpServers") or {}

@@ -299,6 +299,9 @@ def scan_agents(
                 vuln.cwe_ids, impact, creds, tools,
                 severity=vuln.severity.value, is_kev=vuln.is_kev,
             )
+            from agentbom_amd.scan.compliance import apply_framework_tags
+
+            apply_framework_tags(br)
             br.calculate_risk_score()
             blast_radii.append(br)
 
