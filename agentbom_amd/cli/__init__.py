@@ -256,6 +256,95 @@ def db_status_cmd(path: Optional[str]) -> None:
     click.echo(json.dumps(store.status(), indent=2))
 
 
+@main.command(name="proxy", context_settings={"ignore_unknown_options": True})
+@click.argument("command", nargs=-1, required=True, type=click.UNPROCESSED)
+@click.option("--policy", "policy_path", type=click.Path(exists=True), default=None)
+@click.option("--audit-log", type=click.Path(), default=None)
+@click.option("--block-on-warn", is_flag=True)
+def proxy_cmd(command: tuple[str, ...], policy_path: Optional[str],
+              audit_log: Optional[str], block_on_warn: bool) -> None:
+    """Wrap a target MCP server with inline runtime detectors."""
+    from agentbom_amd.runtime.proxy import AuditLog, McpProxy, ProxyPolicy
+
+    policy = ProxyPolicy.load(policy_path)
+    if block_on_warn:
+        policy.block_on_warn = True
+    proxy = McpProxy(list(command), policy=policy, audit=AuditLog(audit_log))
+    sys.exit(proxy.run())
+
+
+@main.group(name="gateway")
+def gateway_group() -> None:
+    """Central secure-by-default MCP relay."""
+
+
+@gateway_group.command(name="serve")
+@click.option("--host", default="127.0.0.1")
+@click.option("--port", type=int, default=8787)
+@click.option("--upstream", "upstreams", multiple=True,
+              help="name=http://host:port upstream registrations")
+def gateway_serve_cmd(host: str, port: int, upstreams: tuple[str, ...]) -> None:
+    """Serve the gateway relay over HTTP (/mcp/{upstream})."""
+    import uvicorn
+    from fastapi import FastAPI
+
+    from agentbom_amd.runtime.gateway import Gateway, Upstream
+
+    gw = Gateway()
+    for spec in upstreams:
+        name, _, url = spec.partition("=")
+
+        def make_handler(u):
+            def handler(frame):
+                import httpx
+
+                return httpx.post(u, json=frame, timeout=30.0).json()
+
+            return handler
+
+        gw.register(Upstream(name=name, handler=make_handler(url)))
+
+    app = FastAPI(title="agent-bom gateway")
+
+    @app.post("/mcp/{upstream}")
+    def relay(upstream: str, frame: dict) -> dict:
+        return gw.relay(upstream, frame)
+
+    @app.get("/metrics")
+    def metrics() -> dict:
+        return gw.metrics
+
+    uvicorn.run(app, host=host, port=port)
+
+
+@main.command(name="policy-check")
+@click.argument("policy_file", type=click.Path(exists=True))
+@click.option("--demo", is_flag=True)
+@click.option("--dry-run", is_flag=True)
+def policy_check_cmd(policy_file: str, demo: bool, dry_run: bool) -> None:
+    """Evaluate a policy file against the latest (demo) scan."""
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+    from agentbom_amd.scan.policy import evaluate_policy, load_policy
+
+    policy = load_policy(policy_file)
+    report = run_demo_scan()
+    result = evaluate_policy(policy, report.blast_radii, dry_run=dry_run)
+    click.echo(json.dumps(result, indent=2))
+    sys.exit(0 if result["passed"] else 1)
+
+
+@main.command(name="diff")
+@click.argument("old_report", type=click.Path(exists=True))
+@click.argument("new_report", type=click.Path(exists=True))
+def diff_cmd(old_report: str, new_report: str) -> None:
+    """Diff two scan report JSON files (new/resolved findings)."""
+    from agentbom_amd.scan.history import diff_reports
+
+    old = json.loads(Path(old_report).read_text())
+    new = json.loads(Path(new_report).read_text())
+    click.echo(json.dumps(diff_reports(old, new), indent=2, default=str))
+
+
 @main.command(name="bench")
 @click.option("--packages", type=int, default=100_000)
 @click.option("--steps", type=int, default=3)

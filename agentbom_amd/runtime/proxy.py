@@ -1,0 +1,174 @@
+"""MCP runtime proxy: wrap a target MCP server with inline detectors.
+
+Reference: src/agent_bom/proxy.py (stdio wrapping, per-message detector
+pipeline, audited allow/warn/block), proxy_audit.py (JSONL audit trail),
+proxy_policy.py (allow/deny tool lists).
+
+``agent-bom proxy -- <command>`` launches the target server as a child
+process, relays stdio JSON-RPC both ways, runs every frame through the
+detector pipeline, blocks on critical hits (the request is answered with a
+JSON-RPC error instead of being forwarded), and appends one audit record
+per frame to ``~/.agent-bom/proxy_audit.jsonl``.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import subprocess
+import sys
+import threading
+import time
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Any, Optional, TextIO
+
+from agentbom_amd.runtime.detectors import DetectorPipeline
+
+
+@dataclass
+class ProxyPolicy:
+    """Allow/deny tool lists + default action for detector warnings."""
+
+    allow_tools: Optional[list[str]] = None  # None = all
+    deny_tools: list[str] = field(default_factory=list)
+    block_on_warn: bool = False
+
+    def tool_allowed(self, tool: str) -> bool:
+        if tool in self.deny_tools:
+            return False
+        if self.allow_tools is not None and tool not in self.allow_tools:
+            return False
+        return True
+
+    @classmethod
+    def load(cls, path: Optional[str]) -> "ProxyPolicy":
+        if not path or not Path(path).exists():
+            return cls()
+        data = json.loads(Path(path).read_text())
+        return cls(
+            allow_tools=data.get("allow_tools"),
+            deny_tools=data.get("deny_tools", []),
+            block_on_warn=data.get("block_on_warn", False),
+        )
+
+
+class AuditLog:
+    """Append-only JSONL audit sink with chained integrity hashes."""
+
+    def __init__(self, path: Optional[str] = None):
+        import hashlib
+
+        self._hashlib = hashlib
+        self.path = Path(path or os.path.expanduser("~/.agent-bom/proxy_audit.jsonl"))
+        self.path.parent.mkdir(parents=True, exist_ok=True)
+        self._prev_hash = "0" * 64
+        self._lock = threading.Lock()
+
+    def record(self, entry: dict[str, Any]) -> None:
+        with self._lock:
+            entry = dict(entry)
+            entry["ts"] = time.time()
+            entry["prev_hash"] = self._prev_hash
+            body = json.dumps(entry, sort_keys=True, default=str)
+            entry_hash = self._hashlib.sha256(body.encode()).hexdigest()
+            entry["hash"] = entry_hash
+            self._prev_hash = entry_hash
+            with open(self.path, "a") as f:
+                f.write(json.dumps(entry, default=str) + "\n")
+
+    @staticmethod
+    def verify(path: str | Path) -> tuple[bool, int]:
+        """Replay the HMAC-style hash chain; returns (intact, entries)."""
+        import hashlib
+
+        prev = "0" * 64
+        n = 0
+        for line in Path(path).read_text().splitlines():
+            entry = json.loads(line)
+            claimed = entry.pop("hash")
+            if entry.get("prev_hash") != prev:
+                return False, n
+            body = json.dumps(entry, sort_keys=True, default=str)
+            if hashlib.sha256(body.encode()).hexdigest() != claimed:
+                return False, n
+            prev = claimed
+            n += 1
+        return True, n
+
+
+class McpProxy:
+    """Bidirectional stdio relay with inline detection."""
+
+    def __init__(self, command: list[str], policy: Optional[ProxyPolicy] = None,
+                 audit: Optional[AuditLog] = None,
+                 pipeline: Optional[DetectorPipeline] = None):
+        self.command = command
+        self.policy = policy or ProxyPolicy()
+        self.audit = audit or AuditLog()
+        self.pipeline = pipeline or DetectorPipeline()
+        self.blocked = 0
+        self.relayed = 0
+
+    # frame-level decision (unit-testable without processes)
+    def decide(self, frame: dict[str, Any], direction: str) -> tuple[str, list]:
+        method = frame.get("method", "")
+        if method == "tools/call":
+            tool = (frame.get("params") or {}).get("name", "")
+            if not self.policy.tool_allowed(tool):
+                return "block", [{"detector": "policy", "severity": "critical",
+                                  "message": f"tool {tool!r} denied by proxy policy",
+                                  "action": "block", "evidence": {}}]
+        action, alerts = self.pipeline.inspect(frame)
+        if action == "warn" and self.policy.block_on_warn:
+            action = "block"
+        return action, [a.to_dict() if hasattr(a, "to_dict") else a for a in alerts]
+
+    def _relay(self, src: TextIO, dst: TextIO, direction: str) -> None:
+        for line in src:
+            line = line.strip()
+            if not line:
+                continue
+            try:
+                frame = json.loads(line)
+            except json.JSONDecodeError:
+                dst.write(line + "\n")
+                dst.flush()
+                continue
+            action, alerts = self.decide(frame, direction)
+            self.audit.record({
+                "direction": direction,
+                "method": frame.get("method"),
+                "id": frame.get("id"),
+                "action": action,
+                "alerts": alerts,
+            })
+            if action == "block" and direction == "client->server":
+                self.blocked += 1
+                err = {"jsonrpc": "2.0", "id": frame.get("id"),
+                       "error": {"code": -32000,
+                                 "message": "blocked by agent-bom proxy: "
+                                            + "; ".join(a["message"] for a in alerts[:3])}}
+                sys.stdout.write(json.dumps(err) + "\n")
+                sys.stdout.flush()
+                continue
+            self.relayed += 1
+            dst.write(json.dumps(frame) + "\n")
+            dst.flush()
+
+    def run(self) -> int:
+        proc = subprocess.Popen(
+            self.command, stdin=subprocess.PIPE, stdout=subprocess.PIPE,
+            text=True, bufsize=1,
+        )
+        up = threading.Thread(target=self._relay,
+                              args=(sys.stdin, proc.stdin, "client->server"), daemon=True)
+        down = threading.Thread(target=self._relay,
+                                args=(proc.stdout, sys.stdout, "server->client"), daemon=True)
+        up.start()
+        down.start()
+        try:
+            return proc.wait()
+        except KeyboardInterrupt:
+            proc.terminate()
+            return 130
