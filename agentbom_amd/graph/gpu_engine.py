@@ -202,14 +202,49 @@ class EstateEngine:
         return cnt, cnt_w.to(torch.int32)
 
     def blast_counts(self, finding_pkgs):
-        """Per unique finding-package reach counts via segmented joins.
+        """Per unique finding-package reach counts.
 
-        Returns dict keyed by package node id -> n_servers/n_agents/
-        n_creds_all/n_creds_db/n_tools_all/n_tools_db as dense tensors
-        indexed by position in ``uniq_pkgs``.
+        GPU path: ONE fused wave-per-package HIP kernel (ops/csrc/blast.hip);
+        zipf-head packages that overflow its LDS caps fall back to the exact
+        sort-based join below.  CPU path: the sort-based join throughout.
+        Returns dense tensors indexed by position in ``uniq_pkgs``.
         """
         torch = self.torch
-        uniq_pkgs = torch.unique(finding_pkgs)  # node ids (pkg space)
+        uniq_pkgs = torch.unique(finding_pkgs)
+        if self.use_gpu:
+            return self._blast_counts_fused(uniq_pkgs)
+        return self._blast_counts_join(uniq_pkgs)
+
+    def _blast_counts_fused(self, uniq_pkgs):
+        from agentbom_amd.ops import native
+
+        torch = self.torch
+        counts, overflow = native.blast_counts(
+            uniq_pkgs.to(torch.int32).contiguous(), self.rev, self.fwd,
+            (ET_CONTAINS, ET_USES, ET_HAS_CRED, ET_PROVIDES_TOOL),
+            self.node_is_db_cred, self.node_is_db_tool,
+        )
+        result = {
+            "uniq_pkgs": uniq_pkgs,
+            "n_servers": counts[:, 0].to(torch.int32),
+            "n_agents": counts[:, 1].to(torch.int32),
+            "n_creds_all": counts[:, 2].to(torch.int32),
+            "n_creds_db": counts[:, 3].to(torch.int32),
+            "n_tools_all": counts[:, 4].to(torch.int32),
+            "n_tools_db": counts[:, 5].to(torch.int32),
+        }
+        ov_idx = torch.nonzero(overflow).flatten()
+        if ov_idx.numel():
+            heavy = self._blast_counts_join(uniq_pkgs[ov_idx])
+            for key in ("n_servers", "n_agents", "n_creds_all", "n_creds_db",
+                        "n_tools_all", "n_tools_db"):
+                result[key] = result[key].clone()
+                result[key][ov_idx] = heavy[key]
+        return result
+
+    def _blast_counts_join(self, uniq_pkgs):
+        """Sort-based exact join (torch ops) — CPU path + overflow fallback."""
+        torch = self.torch
         # position index for dense bincounts
         n_up = uniq_pkgs.numel()
         pos = torch.arange(n_up, device=self.device)

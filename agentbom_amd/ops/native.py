@@ -37,6 +37,7 @@ _SIGNATURES = {
     "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c], _i32),
     "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
+    "abom_blast_counts": ([_c, _i64] + [_c] * 6 + [_i32] * 4 + [_c] * 4 + [_c], _i32),
     "abom_severity_histogram": ([_c, _c, _c, _i64, _c], _i32),
 }
 
@@ -256,6 +257,30 @@ def impact_query(row_off, col, query_sources, etype=None, allowed_mask: int = 0x
     )
     _check(rc, "abom_impact_query")
     return out_nodes, out_hops, out_counts, out_trunc
+
+
+def blast_counts(pkgs, rev, fwd, et_codes: tuple[int, int, int, int],
+                 node_is_db_cred, node_is_db_tool):
+    """Fused per-package reach counts.  Returns (counts [n,6] int32,
+    overflow u8 [n]) — overflowed rows must be recomputed by the caller's
+    sort-based path."""
+    import torch
+
+    lib = load()
+    n = pkgs.numel()
+    dev = pkgs.device
+    out_counts = torch.zeros(n * 6, dtype=torch.int32, device=dev)
+    out_overflow = torch.zeros(n, dtype=torch.uint8, device=dev)
+    rc = lib.abom_blast_counts(
+        _ptr(pkgs), n,
+        _ptr(rev["row_off"]), _ptr(rev["col"]), _ptr(rev["etype"]),
+        _ptr(fwd["row_off"]), _ptr(fwd["col"]), _ptr(fwd["etype"]),
+        et_codes[0], et_codes[1], et_codes[2], et_codes[3],
+        _ptr(node_is_db_cred), _ptr(node_is_db_tool),
+        _ptr(out_counts), _ptr(out_overflow), _stream(),
+    )
+    _check(rc, "abom_blast_counts")
+    return out_counts.view(n, 6), out_overflow
 
 
 def risk_weights_array() -> np.ndarray:

@@ -189,6 +189,23 @@ def test_engine_step_gpu_vs_cpu(estate, dev):
     assert torch.equal(g["n_tools"].cpu(), c["n_tools"])
 
 
+def test_blast_counts_fused_vs_join(estate, dev):
+    """Fused wave-per-package kernel == sort-based join, incl. overflow rows."""
+    import torch as _torch
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+
+    eng = EstateEngine(estate, device=str(dev))
+    pkg_idx, _ = eng.match()
+    pkg_nodes = pkg_idx + estate.pkg_base
+    uniq = _torch.unique(pkg_nodes)
+    fused = eng._blast_counts_fused(uniq)
+    join = eng._blast_counts_join(uniq)
+    for key in ("n_servers", "n_agents", "n_creds_all", "n_creds_db",
+                "n_tools_all", "n_tools_db"):
+        assert _torch.equal(fused[key].cpu(), join[key].cpu()), key
+
+
 def test_severity_histogram_parity(dev):
     from agentbom_amd.ops import cpu_ref, native
 
