@@ -108,6 +108,8 @@ class EstateEngine:
             "row_off": row_off,
             "col": dst[order].to(torch.int32),
             "etype": et[order].contiguous(),
+            # col-aligned edge sources: enables edge-centric dense-frontier BFS
+            "src": src[order].to(torch.int32),
         }
 
     # ── pipeline stages ────────────────────────────────────────────────────
@@ -151,6 +153,7 @@ class EstateEngine:
             return native.bfs(
                 self.fwd["row_off"], self.fwd["col"], self.agent_ids, self.N,
                 etype=self.fwd["etype"], allowed_mask=mask, workspace=self._bfs_ws,
+                edge_src=self.fwd["src"],
             )
         from agentbom_amd.ops import cpu_ref
 

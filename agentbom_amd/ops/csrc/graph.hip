@@ -97,6 +97,36 @@ __global__ void bfs_expand_heavy_kernel(
     }
 }
 
+// Edge-centric expansion for DENSE frontiers: one thread per edge, fully
+// coalesced src/col streams.  Used when the frontier would touch a large
+// share of all edges (the agent->server->everything levels of estate
+// graphs), where per-vertex serial neighbor loops are latency-bound.
+// Requires the per-edge source array (edge_src, same order as col).
+__global__ void bfs_expand_edges_kernel(
+    const uint32_t* __restrict__ edge_src,   // [E]
+    const uint32_t* __restrict__ col,        // [E]
+    const uint8_t* __restrict__ etype,       // [E] or nullptr
+    uint32_t allowed_mask,
+    long long num_edges,
+    uint32_t* __restrict__ dist,
+    uint32_t cur_level,                      // frontier's level (claimed at cur_level)
+    uint32_t* __restrict__ next_frontier,
+    unsigned int* __restrict__ next_count,
+    long long capacity) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x; e < num_edges;
+         e += stride) {
+        if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
+        if (dist[edge_src[e]] != cur_level) continue;
+        const uint32_t v = col[e];
+        if (dist[v] == ABOM_UNVISITED &&
+            atomicCAS(&dist[v], ABOM_UNVISITED, cur_level + 1) == ABOM_UNVISITED) {
+            const unsigned idx = atomicAdd(next_count, 1u);
+            if ((long long)idx < capacity) next_frontier[idx] = v;
+        }
+    }
+}
+
 __global__ void init_dist_kernel(uint32_t* __restrict__ dist, long long n, uint32_t value) {
     const long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
@@ -342,15 +372,32 @@ extern "C" int abom_bfs_expand_heavy(
     return (int)hipGetLastError();
 }
 
+extern "C" int abom_bfs_expand_edges(
+    const void* edge_src, const void* col, const void* etype, unsigned int allowed_mask,
+    long long num_edges, void* dist, unsigned int cur_level, void* next_frontier,
+    void* next_count, long long capacity, void* stream) {
+    const int block = 256;
+    hipLaunchKernelGGL(abom::bfs_expand_edges_kernel, dim3(abom::grid_for(num_edges, block)),
+                       dim3(block), 0, (hipStream_t)stream, (const uint32_t*)edge_src,
+                       (const uint32_t*)col, (const uint8_t*)etype, allowed_mask, num_edges,
+                       (uint32_t*)dist, cur_level, (uint32_t*)next_frontier,
+                       (unsigned int*)next_count, capacity);
+    return (int)hipGetLastError();
+}
+
 // Full multi-source BFS: host-side level loop, one device->host count read
 // per level.  dist must be u32[N]; frontier_a/b u32[N]; heavy_queue u32[N];
-// counters = device u32[2] (next_count, heavy_count).  Returns negative hip
-// error or the number of levels run.
+// counters = device u32[2] (next_count, heavy_count).  ``edge_src`` (the
+// per-edge source array, col-aligned) may be null; when present, levels
+// whose frontier would touch > ~1/8 of all edges switch to the edge-centric
+// kernel (coalesced streams) instead of per-vertex expansion.
+// Returns negative hip error or the number of levels run.
 extern "C" int abom_bfs_run(
     const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
     const void* sources, long long n_sources, void* dist, long long num_nodes,
     void* frontier_a, void* frontier_b, void* heavy_queue, void* counters,
-    int max_levels, void* stream) {
+    int max_levels, const void* edge_src, long long num_edges, double avg_degree,
+    void* stream) {
     hipStream_t s = (hipStream_t)stream;
     unsigned int* ctr = (unsigned int*)counters;
     int rc = abom_bfs_init(dist, num_nodes, stream);
@@ -366,13 +413,21 @@ extern "C" int abom_bfs_run(
     while (frontier_size > 0 && level < max_levels) {
         ++level;
         ABOM_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(unsigned int), s));
-        rc = abom_bfs_expand(row_off, col, etype, allowed_mask, cur, frontier_size, dist,
-                             (unsigned int)level, nxt, ctr, heavy_queue, ctr + 1, num_nodes,
-                             stream);
-        if (rc) return -rc;
-        rc = abom_bfs_expand_heavy(row_off, col, etype, allowed_mask, heavy_queue, ctr + 1,
-                                   dist, (unsigned int)level, nxt, ctr, num_nodes, stream);
-        if (rc) return -rc;
+        const bool dense = edge_src != nullptr && num_edges > 0 &&
+                           (double)frontier_size * avg_degree > (double)num_edges / 8.0;
+        if (dense) {
+            rc = abom_bfs_expand_edges(edge_src, col, etype, allowed_mask, num_edges, dist,
+                                       (unsigned int)(level - 1), nxt, ctr, num_nodes, stream);
+            if (rc) return -rc;
+        } else {
+            rc = abom_bfs_expand(row_off, col, etype, allowed_mask, cur, frontier_size, dist,
+                                 (unsigned int)level, nxt, ctr, heavy_queue, ctr + 1,
+                                 num_nodes, stream);
+            if (rc) return -rc;
+            rc = abom_bfs_expand_heavy(row_off, col, etype, allowed_mask, heavy_queue, ctr + 1,
+                                       dist, (unsigned int)level, nxt, ctr, num_nodes, stream);
+            if (rc) return -rc;
+        }
         ABOM_CHECK(hipMemcpyAsync(&host_count, ctr, sizeof(unsigned int),
                                   hipMemcpyDeviceToHost, s));
         ABOM_CHECK(hipStreamSynchronize(s));

@@ -34,7 +34,8 @@ _SIGNATURES = {
     "abom_bfs_seed": ([_c, _i64, _c, _c, _c], _i32),
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _i64, _c], _i32),
     "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _i64, _c], _i32),
-    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c], _i32),
+    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c], _i32),
+    "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _u32, _c, _c, _i64, _c], _i32),
     "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
     "abom_blast_counts": ([_c, _i64] + [_c] * 6 + [_i32] * 4 + [_c] * 4 + [_c], _i32),
@@ -149,7 +150,7 @@ def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_of
 
 
 def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0xFFFFFFFF,
-        max_levels: int = 64, workspace: Optional[dict] = None):
+        max_levels: int = 64, workspace: Optional[dict] = None, edge_src=None):
     """Multi-source BFS over CSR; returns u32 dist (UNVISITED = 0xFFFFFFFF).
 
     ``workspace`` may carry preallocated buffers (dist/frontier_a/frontier_b/
@@ -179,10 +180,14 @@ def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0
         workspace.update(dist=dist, frontier_a=fa, frontier_b=fb, heavy=hq, counters=ctr)
 
     et = _ptr(etype) if etype is not None else None
+    es = _ptr(edge_src) if edge_src is not None else None
+    num_edges = col.numel()
+    avg_degree = num_edges / max(num_nodes, 1)
     rc = lib.abom_bfs_run(
         _ptr(row_off), _ptr(col), et, allowed_mask,
         _ptr(sources), sources.numel(), _ptr(dist), num_nodes,
-        _ptr(fa), _ptr(fb), _ptr(hq), _ptr(ctr), max_levels, _stream(),
+        _ptr(fa), _ptr(fb), _ptr(hq), _ptr(ctr), max_levels,
+        es, num_edges, float(avg_degree), _stream(),
     )
     if rc < 0:
         _check(-rc, "abom_bfs_run")
