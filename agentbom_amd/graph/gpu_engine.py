@@ -77,6 +77,15 @@ class EstateEngine:
         self.pkg_key_hi = torch.from_numpy(estate.pkg_key_hi.view(np.int64)).to(dev)
         self.pkg_key_lo = torch.from_numpy(estate.pkg_key_lo.view(np.int64)).to(dev)
         self.pkg_flags = torch.from_numpy(estate.pkg_flags).to(dev)
+        # Match-locality layout: packages sorted by group key so adjacent
+        # lanes binary-search identical tree paths (L1/L2-resident) and walk
+        # the same window runs.  ``pkg_perm`` maps sorted row -> original
+        # package index for output reporting.
+        self.pkg_perm = torch.argsort(self.pkg_group_key, stable=True)
+        self.pkg_group_key_sorted = self.pkg_group_key[self.pkg_perm].contiguous()
+        self.pkg_key_hi_sorted = self.pkg_key_hi[self.pkg_perm].contiguous()
+        self.pkg_key_lo_sorted = self.pkg_key_lo[self.pkg_perm].contiguous()
+        self.pkg_flags_sorted = self.pkg_flags[self.pkg_perm].contiguous()
 
         self.arena = estate.arena.to_torch(dev)
         self.cred_is_db = torch.from_numpy(estate.cred_is_db).to(dev)
@@ -123,10 +132,18 @@ class EstateEngine:
         if self.use_gpu:
             from agentbom_amd.ops import native
 
-            return native.match(
-                self.pkg_group_key, self.pkg_key_hi, self.pkg_key_lo, self.pkg_flags,
+            torch = self.torch
+            sp, sw = native.match(
+                self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
+                self.pkg_key_lo_sorted, self.pkg_flags_sorted,
                 self.arena["group_keys"], self.arena["group_off"], self.arena["windows"],
             )
+            # map sorted rows back to original package indices, re-sort for
+            # the deterministic (pkg, window) output order
+            orig = self.pkg_perm[sp]
+            packed = (orig << 32) | sw
+            packed, _ = torch.sort(packed)
+            return (packed >> 32), (packed & 0xFFFFFFFF)
         # CPU oracle path (tests / dev boxes)
         from agentbom_amd.ops import cpu_ref
 

@@ -39,6 +39,7 @@ __global__ void bfs_expand_kernel(
     unsigned int* __restrict__ next_count,
     uint32_t* __restrict__ heavy_queue,
     unsigned int* __restrict__ heavy_count,
+    unsigned int* __restrict__ next_degree_sum,
     long long capacity) {
     constexpr uint64_t WAVE_DEG = 64;
     const long long stride = (long long)gridDim.x * blockDim.x;
@@ -58,6 +59,7 @@ __global__ void bfs_expand_kernel(
                 atomicCAS(&dist[v], ABOM_UNVISITED, next_level) == ABOM_UNVISITED) {
                 const unsigned idx = atomicAdd(next_count, 1u);
                 if ((long long)idx < capacity) next_frontier[idx] = v;
+                atomicAdd(next_degree_sum, (unsigned)(row_off[v + 1] - row_off[v]));
             }
         }
     }
@@ -76,6 +78,7 @@ __global__ void bfs_expand_heavy_kernel(
     uint32_t next_level,
     uint32_t* __restrict__ next_frontier,
     unsigned int* __restrict__ next_count,
+    unsigned int* __restrict__ next_degree_sum,
     long long capacity) {
     const unsigned heavy_size = *heavy_size_ptr;
     const int lane = threadIdx.x & 63;
@@ -92,6 +95,7 @@ __global__ void bfs_expand_heavy_kernel(
                 atomicCAS(&dist[v], ABOM_UNVISITED, next_level) == ABOM_UNVISITED) {
                 const unsigned idx = atomicAdd(next_count, 1u);
                 if ((long long)idx < capacity) next_frontier[idx] = v;
+                atomicAdd(next_degree_sum, (unsigned)(row_off[v + 1] - row_off[v]));
             }
         }
     }
@@ -108,10 +112,12 @@ __global__ void bfs_expand_edges_kernel(
     const uint8_t* __restrict__ etype,       // [E] or nullptr
     uint32_t allowed_mask,
     long long num_edges,
+    const uint64_t* __restrict__ row_off,
     uint32_t* __restrict__ dist,
     uint32_t cur_level,                      // frontier's level (claimed at cur_level)
     uint32_t* __restrict__ next_frontier,
     unsigned int* __restrict__ next_count,
+    unsigned int* __restrict__ next_degree_sum,
     long long capacity) {
     const long long stride = (long long)gridDim.x * blockDim.x;
     for (long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x; e < num_edges;
@@ -123,6 +129,7 @@ __global__ void bfs_expand_edges_kernel(
             atomicCAS(&dist[v], ABOM_UNVISITED, cur_level + 1) == ABOM_UNVISITED) {
             const unsigned idx = atomicAdd(next_count, 1u);
             if ((long long)idx < capacity) next_frontier[idx] = v;
+            atomicAdd(next_degree_sum, (unsigned)(row_off[v + 1] - row_off[v]));
         }
     }
 }
@@ -135,13 +142,16 @@ __global__ void init_dist_kernel(uint32_t* __restrict__ dist, long long n, uint3
 
 __global__ void seed_sources_kernel(
     const uint32_t* __restrict__ sources, long long n_sources,
-    uint32_t* __restrict__ dist, uint32_t* __restrict__ frontier) {
+    const uint64_t* __restrict__ row_off,
+    uint32_t* __restrict__ dist, uint32_t* __restrict__ frontier,
+    unsigned int* __restrict__ degree_sum) {
     const long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n_sources;
          i += stride) {
         const uint32_t s = sources[i];
         dist[s] = 0;
         frontier[i] = s;
+        atomicAdd(degree_sum, (unsigned)(row_off[s + 1] - row_off[s]));
     }
 }
 
@@ -335,12 +345,13 @@ extern "C" int abom_bfs_init(void* dist, long long n, void* stream) {
     return (int)hipGetLastError();
 }
 
-extern "C" int abom_bfs_seed(const void* sources, long long n_sources, void* dist,
-                             void* frontier, void* stream) {
+extern "C" int abom_bfs_seed(const void* sources, long long n_sources, const void* row_off,
+                             void* dist, void* frontier, void* degree_sum, void* stream) {
     const int block = 256;
     hipLaunchKernelGGL(abom::seed_sources_kernel, dim3(abom::grid_for(n_sources, block)),
                        dim3(block), 0, (hipStream_t)stream, (const uint32_t*)sources, n_sources,
-                       (uint32_t*)dist, (uint32_t*)frontier);
+                       (const uint64_t*)row_off, (uint32_t*)dist, (uint32_t*)frontier,
+                       (unsigned int*)degree_sum);
     return (int)hipGetLastError();
 }
 
@@ -348,40 +359,45 @@ extern "C" int abom_bfs_expand(
     const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
     const void* frontier, long long frontier_size, void* dist, unsigned int next_level,
     void* next_frontier, void* next_count, void* heavy_queue, void* heavy_count,
-    long long capacity, void* stream) {
+    void* next_degree_sum, long long capacity, void* stream) {
     const int block = 256;
     hipLaunchKernelGGL(abom::bfs_expand_kernel, dim3(abom::grid_for(frontier_size, block)),
                        dim3(block), 0, (hipStream_t)stream, (const uint64_t*)row_off,
                        (const uint32_t*)col, (const uint8_t*)etype, allowed_mask,
                        (const uint32_t*)frontier, frontier_size, (uint32_t*)dist, next_level,
                        (uint32_t*)next_frontier, (unsigned int*)next_count,
-                       (uint32_t*)heavy_queue, (unsigned int*)heavy_count, capacity);
+                       (uint32_t*)heavy_queue, (unsigned int*)heavy_count,
+                       (unsigned int*)next_degree_sum, capacity);
     return (int)hipGetLastError();
 }
 
 extern "C" int abom_bfs_expand_heavy(
     const void* row_off, const void* col, const void* etype, unsigned int allowed_mask,
     const void* heavy_queue, const void* heavy_size_ptr, void* dist, unsigned int next_level,
-    void* next_frontier, void* next_count, long long capacity, void* stream) {
+    void* next_frontier, void* next_count, void* next_degree_sum, long long capacity,
+    void* stream) {
     const int block = 256;  // 4 waves per block; kernel grid-strides by wave
     hipLaunchKernelGGL(abom::bfs_expand_heavy_kernel, dim3(2048), dim3(block), 0,
                        (hipStream_t)stream, (const uint64_t*)row_off, (const uint32_t*)col,
                        (const uint8_t*)etype, allowed_mask, (const uint32_t*)heavy_queue,
                        (const unsigned int*)heavy_size_ptr, (uint32_t*)dist, next_level,
-                       (uint32_t*)next_frontier, (unsigned int*)next_count, capacity);
+                       (uint32_t*)next_frontier, (unsigned int*)next_count,
+                       (unsigned int*)next_degree_sum, capacity);
     return (int)hipGetLastError();
 }
 
 extern "C" int abom_bfs_expand_edges(
     const void* edge_src, const void* col, const void* etype, unsigned int allowed_mask,
-    long long num_edges, void* dist, unsigned int cur_level, void* next_frontier,
-    void* next_count, long long capacity, void* stream) {
+    long long num_edges, const void* row_off, void* dist, unsigned int cur_level,
+    void* next_frontier, void* next_count, void* next_degree_sum, long long capacity,
+    void* stream) {
     const int block = 256;
     hipLaunchKernelGGL(abom::bfs_expand_edges_kernel, dim3(abom::grid_for(num_edges, block)),
                        dim3(block), 0, (hipStream_t)stream, (const uint32_t*)edge_src,
                        (const uint32_t*)col, (const uint8_t*)etype, allowed_mask, num_edges,
-                       (uint32_t*)dist, cur_level, (uint32_t*)next_frontier,
-                       (unsigned int*)next_count, capacity);
+                       (const uint64_t*)row_off, (uint32_t*)dist, cur_level,
+                       (uint32_t*)next_frontier, (unsigned int*)next_count,
+                       (unsigned int*)next_degree_sum, capacity);
     return (int)hipGetLastError();
 }
 
@@ -398,40 +414,53 @@ extern "C" int abom_bfs_run(
     void* frontier_a, void* frontier_b, void* heavy_queue, void* counters,
     int max_levels, const void* edge_src, long long num_edges, double avg_degree,
     void* stream) {
+    // counters layout: [0] next_count, [1] heavy_count, [2] frontier degree sum
     hipStream_t s = (hipStream_t)stream;
     unsigned int* ctr = (unsigned int*)counters;
     int rc = abom_bfs_init(dist, num_nodes, stream);
     if (rc) return -rc;
-    rc = abom_bfs_seed(sources, n_sources, dist, frontier_a, stream);
+    ABOM_CHECK(hipMemsetAsync(ctr, 0, 3 * sizeof(unsigned int), s));
+    rc = abom_bfs_seed(sources, n_sources, row_off, dist, frontier_a, ctr + 2, stream);
     if (rc) return -rc;
+    unsigned int host_vals[2] = {0, 0};
+    ABOM_CHECK(hipMemcpyAsync(&host_vals[1], ctr + 2, sizeof(unsigned int),
+                              hipMemcpyDeviceToHost, s));
+    ABOM_CHECK(hipStreamSynchronize(s));
+    unsigned int frontier_degree = host_vals[1];
 
     uint32_t* cur = (uint32_t*)frontier_a;
     uint32_t* nxt = (uint32_t*)frontier_b;
     long long frontier_size = n_sources;
     int level = 0;
-    unsigned int host_count = 0;
     while (frontier_size > 0 && level < max_levels) {
         ++level;
-        ABOM_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(unsigned int), s));
+        ABOM_CHECK(hipMemsetAsync(ctr, 0, 3 * sizeof(unsigned int), s));
+        // Dense frontier (its edges are a big share of ALL edges): the
+        // one-thread-per-edge pass with coalesced src/col streams beats
+        // per-vertex serial neighbor loops.
         const bool dense = edge_src != nullptr && num_edges > 0 &&
-                           (double)frontier_size * avg_degree > (double)num_edges / 8.0;
+                           (double)frontier_degree > (double)num_edges / 8.0;
         if (dense) {
-            rc = abom_bfs_expand_edges(edge_src, col, etype, allowed_mask, num_edges, dist,
-                                       (unsigned int)(level - 1), nxt, ctr, num_nodes, stream);
+            rc = abom_bfs_expand_edges(edge_src, col, etype, allowed_mask, num_edges, row_off,
+                                       dist, (unsigned int)(level - 1), nxt, ctr, ctr + 2,
+                                       num_nodes, stream);
             if (rc) return -rc;
         } else {
             rc = abom_bfs_expand(row_off, col, etype, allowed_mask, cur, frontier_size, dist,
-                                 (unsigned int)level, nxt, ctr, heavy_queue, ctr + 1,
+                                 (unsigned int)level, nxt, ctr, heavy_queue, ctr + 1, ctr + 2,
                                  num_nodes, stream);
             if (rc) return -rc;
             rc = abom_bfs_expand_heavy(row_off, col, etype, allowed_mask, heavy_queue, ctr + 1,
-                                       dist, (unsigned int)level, nxt, ctr, num_nodes, stream);
+                                       dist, (unsigned int)level, nxt, ctr, ctr + 2,
+                                       num_nodes, stream);
             if (rc) return -rc;
         }
-        ABOM_CHECK(hipMemcpyAsync(&host_count, ctr, sizeof(unsigned int),
+        unsigned int host_pair[3] = {0, 0, 0};
+        ABOM_CHECK(hipMemcpyAsync(host_pair, ctr, 3 * sizeof(unsigned int),
                                   hipMemcpyDeviceToHost, s));
         ABOM_CHECK(hipStreamSynchronize(s));
-        frontier_size = host_count;
+        frontier_size = host_pair[0];
+        frontier_degree = host_pair[2];
         uint32_t* t = cur; cur = nxt; nxt = t;
     }
     return level;

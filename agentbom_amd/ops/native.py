@@ -31,11 +31,11 @@ _SIGNATURES = {
     "abom_error_string": ([_i32], ctypes.c_char_p),
     "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c, _c, _i64, _c], _i32),
     "abom_bfs_init": ([_c, _i64, _c], _i32),
-    "abom_bfs_seed": ([_c, _i64, _c, _c, _c], _i32),
-    "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _i64, _c], _i32),
-    "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _i64, _c], _i32),
+    "abom_bfs_seed": ([_c, _i64, _c, _c, _c, _c, _c], _i32),
+    "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
+    "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _c, _i64, _c], _i32),
     "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c], _i32),
-    "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _u32, _c, _c, _i64, _c], _i32),
+    "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _c, _u32, _c, _c, _c, _i64, _c], _i32),
     "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
     "abom_blast_counts": ([_c, _i64] + [_c] * 6 + [_i32] * 4 + [_c] * 4 + [_c], _i32),
@@ -174,8 +174,8 @@ def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0
     if hq is None or hq.numel() < num_nodes:
         hq = torch.empty(num_nodes, dtype=torch.int32, device=dev)
     ctr = ws.get("counters")
-    if ctr is None:
-        ctr = torch.zeros(2, dtype=torch.int32, device=dev)
+    if ctr is None or ctr.numel() < 4:
+        ctr = torch.zeros(4, dtype=torch.int32, device=dev)
     if workspace is not None:
         workspace.update(dist=dist, frontier_a=fa, frontier_b=fb, heavy=hq, counters=ctr)
 
@@ -217,21 +217,22 @@ def bfs_level(row_off, col, frontier, dist, level: int, etype=None,
         hq = torch.empty(max(frontier.numel(), 1024), dtype=torch.int32, device=dev)
         ws["heavy"] = hq
     ctr = ws.get("counters")
-    if ctr is None:
-        ctr = torch.zeros(2, dtype=torch.int32, device=dev)
+    if ctr is None or ctr.numel() < 4:
+        ctr = torch.zeros(4, dtype=torch.int32, device=dev)
         ws["counters"] = ctr
     ctr.zero_()
     et = _ptr(etype) if etype is not None else None
+    deg_ptr = ctypes.c_void_p(ctr.data_ptr() + 8)
     rc = lib.abom_bfs_expand(
         _ptr(row_off), _ptr(col), et, allowed_mask, _ptr(frontier), frontier.numel(),
         _ptr(dist), level, _ptr(nxt), _ptr(ctr), _ptr(hq),
-        ctypes.c_void_p(ctr.data_ptr() + 4), cap, _stream(),
+        ctypes.c_void_p(ctr.data_ptr() + 4), deg_ptr, cap, _stream(),
     )
     _check(rc, "abom_bfs_expand")
     rc = lib.abom_bfs_expand_heavy(
         _ptr(row_off), _ptr(col), et, allowed_mask, _ptr(hq),
         ctypes.c_void_p(ctr.data_ptr() + 4), _ptr(dist), level, _ptr(nxt), _ptr(ctr),
-        cap, _stream(),
+        deg_ptr, cap, _stream(),
     )
     _check(rc, "abom_bfs_expand_heavy")
     n = int(ctr[0].item())
