@@ -24,6 +24,12 @@
 
 namespace abom {
 
+// One atomic per wave: lanes accumulate locally, reduce, lane 0 adds.
+__device__ __forceinline__ void wave_add_degree(unsigned int* dst, unsigned v) {
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if ((threadIdx.x & 63) == 0 && v) atomicAdd(dst, v);
+}
+
 // Pass over the frontier: expand vertices with degree < WAVE_DEG inline;
 // defer heavy vertices to the heavy queue.
 __global__ void bfs_expand_kernel(
@@ -43,6 +49,7 @@ __global__ void bfs_expand_kernel(
     long long capacity) {
     constexpr uint64_t WAVE_DEG = 64;
     const long long stride = (long long)gridDim.x * blockDim.x;
+    unsigned my_deg = 0;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < frontier_size;
          i += stride) {
         const uint32_t u = frontier[i];
@@ -59,10 +66,11 @@ __global__ void bfs_expand_kernel(
                 atomicCAS(&dist[v], ABOM_UNVISITED, next_level) == ABOM_UNVISITED) {
                 const unsigned idx = atomicAdd(next_count, 1u);
                 if ((long long)idx < capacity) next_frontier[idx] = v;
-                atomicAdd(next_degree_sum, (unsigned)(row_off[v + 1] - row_off[v]));
+                my_deg += (unsigned)(row_off[v + 1] - row_off[v]);
             }
         }
     }
+    wave_add_degree(next_degree_sum, my_deg);
 }
 
 // Wave-cooperative expansion of heavy vertices: one wave (64 lanes) walks one
@@ -84,6 +92,7 @@ __global__ void bfs_expand_heavy_kernel(
     const int lane = threadIdx.x & 63;
     const long long wave = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const long long nwaves = ((long long)gridDim.x * blockDim.x) >> 6;
+    unsigned my_deg = 0;
     for (long long q = wave; q < heavy_size; q += nwaves) {
         const uint32_t u = heavy_queue[q];
         const uint64_t beg = row_off[u];
@@ -95,10 +104,11 @@ __global__ void bfs_expand_heavy_kernel(
                 atomicCAS(&dist[v], ABOM_UNVISITED, next_level) == ABOM_UNVISITED) {
                 const unsigned idx = atomicAdd(next_count, 1u);
                 if ((long long)idx < capacity) next_frontier[idx] = v;
-                atomicAdd(next_degree_sum, (unsigned)(row_off[v + 1] - row_off[v]));
+                my_deg += (unsigned)(row_off[v + 1] - row_off[v]);
             }
         }
     }
+    wave_add_degree(next_degree_sum, my_deg);
 }
 
 // Edge-centric expansion for DENSE frontiers: one thread per edge, fully
@@ -120,6 +130,7 @@ __global__ void bfs_expand_edges_kernel(
     unsigned int* __restrict__ next_degree_sum,
     long long capacity) {
     const long long stride = (long long)gridDim.x * blockDim.x;
+    unsigned my_deg = 0;
     for (long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x; e < num_edges;
          e += stride) {
         if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
@@ -129,9 +140,10 @@ __global__ void bfs_expand_edges_kernel(
             atomicCAS(&dist[v], ABOM_UNVISITED, cur_level + 1) == ABOM_UNVISITED) {
             const unsigned idx = atomicAdd(next_count, 1u);
             if ((long long)idx < capacity) next_frontier[idx] = v;
-            atomicAdd(next_degree_sum, (unsigned)(row_off[v + 1] - row_off[v]));
+            my_deg += (unsigned)(row_off[v + 1] - row_off[v]);
         }
     }
+    wave_add_degree(next_degree_sum, my_deg);
 }
 
 __global__ void init_dist_kernel(uint32_t* __restrict__ dist, long long n, uint32_t value) {
@@ -146,13 +158,15 @@ __global__ void seed_sources_kernel(
     uint32_t* __restrict__ dist, uint32_t* __restrict__ frontier,
     unsigned int* __restrict__ degree_sum) {
     const long long stride = (long long)gridDim.x * blockDim.x;
+    unsigned my_deg = 0;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n_sources;
          i += stride) {
         const uint32_t s = sources[i];
         dist[s] = 0;
         frontier[i] = s;
-        atomicAdd(degree_sum, (unsigned)(row_off[s + 1] - row_off[s]));
+        my_deg += (unsigned)(row_off[s + 1] - row_off[s]);
     }
+    wave_add_degree(degree_sum, my_deg);
 }
 
 // ── Batched bounded blast-radius query (impact_of / traverse_subgraph) ─────

@@ -180,6 +180,8 @@ def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0
         workspace.update(dist=dist, frontier_a=fa, frontier_b=fb, heavy=hq, counters=ctr)
 
     et = _ptr(etype) if etype is not None else None
+    if os.environ.get("AGENT_BOM_BFS_DENSE", "1") == "0":
+        edge_src = None  # A/B toggle: disable edge-centric dense-frontier mode
     es = _ptr(edge_src) if edge_src is not None else None
     num_edges = col.numel()
     avg_degree = num_edges / max(num_nodes, 1)
