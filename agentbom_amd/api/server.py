@@ -50,6 +50,12 @@ class _State:
             "scans_total": 0, "scan_failures_total": 0, "graph_queries_total": 0,
             "auth_failures_total": 0,
         }
+        self.graph_store = None
+        store_path = os.environ.get("AGENT_BOM_GRAPH_STORE")
+        if store_path:
+            from agentbom_amd.graph.store import SQLiteGraphStore
+
+            self.graph_store = SQLiteGraphStore(store_path)
 
 
 def create_app() -> FastAPI:
@@ -112,6 +118,21 @@ def create_app() -> FastAPI:
             job["steps"].append({"step": "graph_build", "at": _now()})
             graph = build_unified_graph_from_report(report)
             apply_dependency_reachability_to_blast_radii(report, graph)
+            from agentbom_amd.graph.toxic_combos import (
+                detect_toxic_combinations,
+                toxic_combination_to_finding,
+            )
+
+            combos = detect_toxic_combinations(graph)
+            report.findings = report.to_findings() + [
+                toxic_combination_to_finding(c) for c in combos
+            ]
+            report.toxic_combination_findings_data = [c.to_dict() for c in combos]
+            job["steps"].append({"step": "graph_persist", "at": _now()})
+            snapshot_id = None
+            if state.graph_store is not None:
+                snapshot_id = state.graph_store.save_snapshot(graph, scan_id=job_id)
+            job["snapshot_id"] = snapshot_id
             with state.lock:
                 state.reports[job_id] = report
                 state.graphs[job_id] = graph
@@ -283,10 +304,34 @@ def create_app() -> FastAPI:
             "block_threshold": cfg.DEPLOY_BLOCK_RISK,
         }
 
+    @app.get("/v1/graph/snapshots", dependencies=[Depends(auth)])
+    def graph_snapshots() -> dict:
+        if state.graph_store is None:
+            return {"snapshots": [], "note": "set AGENT_BOM_GRAPH_STORE to persist snapshots"}
+        return {"snapshots": state.graph_store.list_snapshots()}
+
+    @app.get("/v1/graph/diff", dependencies=[Depends(auth)])
+    def graph_diff(old: Optional[str] = None, new: Optional[str] = None) -> dict:
+        if state.graph_store is None:
+            raise HTTPException(status_code=404, detail="no graph store configured")
+        snaps = state.graph_store.list_snapshots()
+        if len(snaps) < 2 and not (old and new):
+            raise HTTPException(status_code=404, detail="need two snapshots to diff")
+        new = new or snaps[0]["snapshot_id"]
+        old = old or snaps[1]["snapshot_id"]
+        return state.graph_store.diff_snapshots(old, new)
+
+    @app.get("/v1/findings/toxic-combinations", dependencies=[Depends(auth)])
+    def toxic_combinations() -> dict:
+        report = _latest_report()
+        return {"combinations": report.toxic_combination_findings_data or []}
+
     @app.get("/v1/graph/evidence-manifest", dependencies=[Depends(auth)])
     def graph_evidence_manifest() -> dict:
         import hashlib
 
+        if state.graph_store is not None:
+            return state.graph_store.evidence_manifest(scan_id=state.latest_scan)
         g = _latest_graph()
         digest = hashlib.sha256(
             json.dumps(g.to_dict(), sort_keys=True, default=str).encode()

@@ -18,9 +18,9 @@ import click
 from agentbom_amd import __version__
 
 FORMATS = [
-    "console", "json", "sarif", "cyclonedx", "spdx", "spdx2", "csv",
-    "markdown", "plain", "junit", "prometheus", "graph", "mermaid", "dot",
-    "graphml", "cypher",
+    "console", "json", "html", "sarif", "cyclonedx", "spdx", "spdx2", "ocsf",
+    "csv", "markdown", "plain", "junit", "prometheus", "graph", "mermaid",
+    "dot", "graphml", "cypher",
 ]
 
 
@@ -56,6 +56,14 @@ def _render(report, fmt: str, output: Optional[str], verbose: bool) -> None:
         from agentbom_amd.output.spdx_fmt import to_spdx
 
         text = json.dumps(to_spdx(report), indent=2)
+    elif fmt == "html":
+        from agentbom_amd.output.html_fmt import to_html
+
+        text = to_html(report)
+    elif fmt == "ocsf":
+        from agentbom_amd.output.ocsf import to_ocsf
+
+        text = json.dumps(to_ocsf(report), indent=2, default=str)
     elif fmt == "csv":
         text = misc_fmt.to_csv(report)
     elif fmt == "markdown":
@@ -127,6 +135,26 @@ def _scan_impl(
             sys.exit(2)
         windows = load_advisory_windows(offline=offline)
         report = scan_agents(agents, windows, options)
+
+    # graph phase: reachability stamping + toxic-combination findings
+    from agentbom_amd.graph.builder import build_unified_graph_from_report
+    from agentbom_amd.graph.dependency_reach import (
+        apply_dependency_reachability_to_blast_radii,
+    )
+    from agentbom_amd.graph.toxic_combos import (
+        detect_toxic_combinations,
+        toxic_combination_to_finding,
+    )
+
+    graph = build_unified_graph_from_report(report)
+    apply_dependency_reachability_to_blast_radii(report, graph)
+    combos = detect_toxic_combinations(graph)
+    existing = {f.id for f in report.findings}
+    report.findings = report.to_findings() + [
+        toxic_combination_to_finding(c) for c in combos
+        if c.id not in existing
+    ]
+    report.toxic_combination_findings_data = [c.to_dict() for c in combos]
 
     _render(report, fmt, output, verbose)
     rc = compute_exit_code(report, options)

@@ -1,0 +1,99 @@
+"""Self-contained HTML report (no external assets).
+
+Reference: src/agent_bom/output/html*.py — a standalone single-file report
+with summary cards, findings table, and per-finding blast-radius details.
+"""
+
+from __future__ import annotations
+
+import html
+from typing import Any
+
+from agentbom_amd.models import AIBOMReport
+
+_CSS = """
+body { font-family: -apple-system, 'Segoe UI', Roboto, sans-serif; margin: 2rem;
+       background: #0d1117; color: #e6edf3; }
+h1, h2 { color: #f0f6fc; }
+.cards { display: flex; gap: 1rem; flex-wrap: wrap; margin: 1rem 0; }
+.card { background: #161b22; border: 1px solid #30363d; border-radius: 8px;
+        padding: 1rem 1.5rem; min-width: 8rem; }
+.card .num { font-size: 1.8rem; font-weight: 700; }
+.card.crit .num { color: #f85149; } .card.high .num { color: #ff8d67; }
+.card.med .num { color: #d29922; } .card.ok .num { color: #3fb950; }
+table { border-collapse: collapse; width: 100%; margin: 1rem 0; }
+th, td { border: 1px solid #30363d; padding: 0.4rem 0.7rem; text-align: left;
+         font-size: 0.9rem; }
+th { background: #161b22; }
+.sev-critical { color: #f85149; font-weight: 700; }
+.sev-high { color: #ff8d67; } .sev-medium { color: #d29922; } .sev-low { color: #79c0ff; }
+.badge { display: inline-block; padding: 0 0.4rem; border-radius: 6px;
+         font-size: 0.75rem; font-weight: 700; }
+.badge.kev { background: #8957e5; color: white; }
+.badge.mal { background: #f85149; color: white; }
+details { margin: 0.5rem 0; background: #161b22; border: 1px solid #30363d;
+          border-radius: 8px; padding: 0.5rem 1rem; }
+summary { cursor: pointer; }
+code { background: #21262d; padding: 0.1rem 0.3rem; border-radius: 4px; }
+"""
+
+
+def _e(x: Any) -> str:
+    return html.escape(str(x if x is not None else ""))
+
+
+def to_html(report: AIBOMReport) -> str:
+    counts = report.severity_counts()
+    rows = []
+    details = []
+    for br in report.blast_radii:
+        v = br.vulnerability
+        badges = ""
+        if v.is_kev:
+            badges += ' <span class="badge kev">KEV</span>'
+        if br.package.is_malicious:
+            badges += ' <span class="badge mal">MALICIOUS</span>'
+        rows.append(
+            f"<tr><td>{br.risk_score:.1f}</td>"
+            f'<td class="sev-{_e(v.severity.value)}">{_e(v.severity.value)}{badges}</td>'
+            f"<td><code>{_e(v.id)}</code></td>"
+            f"<td>{_e(br.package.name)}@{_e(br.package.version)}</td>"
+            f"<td>{_e(br.reachability)}</td>"
+            f"<td>{len(br.affected_agents)}</td><td>{len(br.exposed_credentials)}</td>"
+            f"<td>{len(br.exposed_tools)}</td>"
+            f"<td>{_e(v.fixed_version or '-')}</td></tr>"
+        )
+        details.append(
+            f"<details><summary><code>{_e(v.id)}</code> — {_e(br.package.name)} "
+            f"(risk {br.risk_score:.1f})</summary>"
+            f"<p>{_e(v.summary)}</p>"
+            f"<p><b>Attack vector:</b> {_e(br.attack_vector_summary or '')}</p>"
+            f"<p><b>Affected agents:</b> {_e(', '.join(a.name for a in br.affected_agents))}</p>"
+            f"<p><b>Affected servers:</b> {_e(', '.join(s.name for s in br.affected_servers))}</p>"
+            f"<p><b>Exposed credentials:</b> {_e(', '.join(br.exposed_credentials) or 'none')}</p>"
+            f"<p><b>Reachable tools:</b> {_e(', '.join(t.name for t in br.exposed_tools) or 'none')}</p>"
+            "</details>"
+        )
+
+    return f"""<!DOCTYPE html>
+<html><head><meta charset="utf-8"><title>agent-bom scan report</title>
+<style>{_CSS}</style></head><body>
+<h1>AI-BOM Scan Report</h1>
+<p>generated {_e(report.generated_at.isoformat())} · scan {_e(report.scan_id or 'local')}</p>
+<div class="cards">
+<div class="card"><div class="num">{report.total_agents}</div>agents</div>
+<div class="card"><div class="num">{report.total_servers}</div>MCP servers</div>
+<div class="card"><div class="num">{report.total_packages}</div>packages</div>
+<div class="card crit"><div class="num">{counts['critical']}</div>critical</div>
+<div class="card high"><div class="num">{counts['high']}</div>high</div>
+<div class="card med"><div class="num">{counts['medium']}</div>medium</div>
+<div class="card ok"><div class="num">{counts['low']}</div>low</div>
+</div>
+<h2>Findings</h2>
+<table><tr><th>risk</th><th>severity</th><th>vulnerability</th><th>package</th>
+<th>reach</th><th>agents</th><th>creds</th><th>tools</th><th>fix</th></tr>
+{''.join(rows)}
+</table>
+<h2>Blast radius</h2>
+{''.join(details)}
+</body></html>"""

@@ -32,7 +32,8 @@ class TestApi:
         jid = r.json()["job_id"]
         job = client.get(f"/v1/scan/{jid}").json()
         assert job["status"] == "done"
-        assert [s["step"] for s in job["steps"]] == ["scan", "graph_build", "done"]
+        assert [s["step"] for s in job["steps"]] == ["scan", "graph_build",
+                                                     "graph_persist", "done"]
         report = client.get(f"/v1/scan/{jid}/report").json()
         assert report["schema_version"] == "1.0"
 
