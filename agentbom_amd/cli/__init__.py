@@ -101,6 +101,8 @@ def _scan_impl(
     output: Optional[str], fail_on_severity: str, fail_on_kev: bool,
     exit_zero: bool, blast_radius_depth: int, verbose: bool,
     include_unfixed: bool, no_gpu: bool,
+    sbom: Optional[str] = None, filesystem: Optional[str] = None,
+    scan_secrets: bool = False, model_files: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -118,23 +120,53 @@ def _scan_impl(
     )
     if demo:
         report = run_demo_scan(options)
-    elif inventory:
-        data = json.loads(Path(inventory).read_text())
-        agents = inventory_to_agents(data)
-        from agentbom_amd.db.store import load_advisory_windows
-
-        windows = load_advisory_windows(offline=offline)
-        report = scan_agents(agents, windows, options)
     else:
-        from agentbom_amd.db.store import load_advisory_windows
-        from agentbom_amd.scan.discovery import discover_all
+        agents = []
+        if inventory:
+            agents += inventory_to_agents(json.loads(Path(inventory).read_text()))
+        if sbom:
+            from agentbom_amd.scan.sbom_ingest import sbom_to_agent
 
-        agents = discover_all()
+            agents.append(sbom_to_agent(sbom))
+        if filesystem:
+            from agentbom_amd.models import Agent, AgentType, MCPServer, ServerSurface
+            from agentbom_amd.scan.parsers import extract_packages
+
+            pkgs = extract_packages(filesystem)
+            agents.append(Agent(
+                name=f"filesystem:{Path(filesystem).name}", agent_type=AgentType.CUSTOM,
+                config_path=str(filesystem),
+                mcp_servers=[MCPServer(name=f"filesystem:{Path(filesystem).name}",
+                                       packages=pkgs, surface=ServerSurface.FILESYSTEM)],
+                source="filesystem_scan",
+            ))
+        if not agents:
+            from agentbom_amd.scan.discovery import discover_all
+
+            agents = discover_all()
         if not agents:
             click.echo("No agents discovered. Try --demo for the bundled demo estate.", err=True)
             sys.exit(2)
-        windows = load_advisory_windows(offline=offline)
-        report = scan_agents(agents, windows, options)
+        from agentbom_amd.db.store import load_advisory_windows
+
+        report = scan_agents(agents, load_advisory_windows(offline=offline), options)
+
+    # optional side scanners -> unified findings
+    if scan_secrets:
+        from agentbom_amd.scan.secrets import scan_paths, secret_hit_to_finding
+
+        root = filesystem or "."
+        hits = scan_paths(root)
+        report.findings.extend(secret_hit_to_finding(h) for h in hits)
+        report.ai_inventory_data = {"secrets": {"findings": [h.to_dict() for h in hits]}}
+    if model_files:
+        from agentbom_amd.scan.model_scan import model_result_to_finding, scan_model_tree
+
+        results = scan_model_tree(model_files)
+        report.extra_data["model_files"] = [r.to_dict() for r in results]
+        report.findings.extend(
+            f for f in (model_result_to_finding(r) for r in results) if f is not None
+        )
 
     # graph phase: reachability stamping + toxic-combination findings
     from agentbom_amd.graph.builder import build_unified_graph_from_report
@@ -185,6 +217,14 @@ def _scan_options(f):
         click.option("-v", "--verbose", is_flag=True),
         click.option("--include-unfixed", is_flag=True),
         click.option("--no-gpu", is_flag=True, help="Force the CPU match path."),
+        click.option("--sbom", type=click.Path(exists=True), default=None,
+                     help="Scan an existing CycloneDX/SPDX SBOM."),
+        click.option("--filesystem", type=click.Path(exists=True), default=None,
+                     help="Extract + scan packages from a directory tree."),
+        click.option("--scan-secrets", is_flag=True,
+                     help="Also scan for hardcoded secrets (redacted)."),
+        click.option("--model-files", type=click.Path(exists=True), default=None,
+                     help="Scan ML model artifacts for pickle payloads."),
     ]
     for o in reversed(opts):
         f = o(f)

@@ -109,16 +109,24 @@ class AdvisoryStore:
     def ingest_windows(self, windows: Sequence[AdvisoryWindow], source: str = "osv") -> int:
         n = 0
         for w in windows:
+            # NOTE: never INSERT OR REPLACE into vulns — REPLACE is
+            # DELETE+INSERT and the affected.vuln_id ON DELETE CASCADE would
+            # silently wipe sibling windows of a multi-branch advisory
+            # (exactly the collapse the schema-v5 PK exists to prevent).
             self.conn.execute(
-                "INSERT OR REPLACE INTO vulns(id, summary, severity, cvss_score, fixed_version,"
-                " cwe_ids, aliases, source) VALUES (?,?,?,?,?,?,?,?)",
+                "INSERT INTO vulns(id, summary, severity, cvss_score, fixed_version,"
+                " cwe_ids, aliases, source) VALUES (?,?,?,?,?,?,?,?)"
+                " ON CONFLICT(id) DO UPDATE SET summary=excluded.summary,"
+                " severity=excluded.severity, cvss_score=excluded.cvss_score,"
+                " fixed_version=excluded.fixed_version, cwe_ids=excluded.cwe_ids,"
+                " aliases=excluded.aliases, source=excluded.source",
                 (
                     w.vuln_id, w.summary, w.severity.value, w.cvss_score,
                     w.fixed_version or w.fixed, ",".join(w.cwe_ids), ",".join(w.aliases), source,
                 ),
             )
             self.conn.execute(
-                "INSERT OR REPLACE INTO affected(vuln_id, ecosystem, package_name, introduced,"
+                "INSERT OR IGNORE INTO affected(vuln_id, ecosystem, package_name, introduced,"
                 " fixed, last_affected) VALUES (?,?,?,?,?,?)",
                 (
                     w.vuln_id,
