@@ -103,6 +103,7 @@ def _scan_impl(
     include_unfixed: bool, no_gpu: bool,
     sbom: Optional[str] = None, filesystem: Optional[str] = None,
     scan_secrets: bool = False, model_files: Optional[str] = None,
+    code: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -167,6 +168,20 @@ def _scan_impl(
         report.findings.extend(
             f for f in (model_result_to_finding(r) for r in results) if f is not None
         )
+    if code:
+        from agentbom_amd.scan.ast_analysis import (
+            apply_symbol_reachability,
+            ast_finding_to_finding,
+            build_symbol_index,
+        )
+
+        idx = build_symbol_index(code)
+        report.findings.extend(ast_finding_to_finding(f) for f in idx.findings)
+        apply_symbol_reachability(report, idx)
+        report.ai_inventory_data = (report.ai_inventory_data or {}) | {
+            "ast_analysis": {"files_scanned": idx.files_scanned,
+                             "flow_findings": [f.to_dict() for f in idx.findings]}
+        }
 
     # graph phase: reachability stamping + toxic-combination findings
     from agentbom_amd.graph.builder import build_unified_graph_from_report
@@ -225,6 +240,8 @@ def _scan_options(f):
                      help="Also scan for hardcoded secrets (redacted)."),
         click.option("--model-files", type=click.Path(exists=True), default=None,
                      help="Scan ML model artifacts for pickle payloads."),
+        click.option("--code", type=click.Path(exists=True), default=None,
+                     help="AST security analysis + symbol-level CVE reachability."),
     ]
     for o in reversed(opts):
         f = o(f)
