@@ -56,6 +56,13 @@ class _State:
             from agentbom_amd.graph.store import SQLiteGraphStore
 
             self.graph_store = SQLiteGraphStore(store_path)
+        from agentbom_amd.api.fleet import BackpressureController, FleetRegistry
+
+        self.fleet = FleetRegistry()
+        self.backpressure = BackpressureController(
+            max_concurrent=int(os.environ.get("AGENT_BOM_MAX_CONCURRENT_SCANS", "8")))
+        self.scheduler = None
+        self.audit_entries: list[dict] = []
 
 
 def create_app() -> FastAPI:
@@ -148,6 +155,18 @@ def create_app() -> FastAPI:
 
     @app.post("/v1/scan", status_code=201, dependencies=[Depends(auth)])
     def submit_scan(req: ScanRequest) -> dict:
+        admitted, retry_after = state.backpressure.try_acquire()
+        if not admitted:
+            from fastapi.responses import JSONResponse
+
+            raise HTTPException(status_code=429, detail="scan concurrency limit reached",
+                                headers={"Retry-After": str(retry_after)})
+        try:
+            return _submit_scan_inner(req)
+        finally:
+            state.backpressure.release()
+
+    def _submit_scan_inner(req: ScanRequest) -> dict:
         job_id = str(uuid.uuid4())
         state.jobs[job_id] = {
             "id": job_id, "status": "pending", "submitted_at": _now(),
@@ -347,6 +366,78 @@ def create_app() -> FastAPI:
             "excluded_private_fields": ["env_values", "credential_values"],
             "retention_policy": {"snapshots": 10},
         }
+
+    # ── fleet / scheduler / audit ─────────────────────────────────────────
+
+    @app.post("/v1/fleet/heartbeat", dependencies=[Depends(auth)])
+    def fleet_heartbeat(payload: dict) -> dict:
+        return state.fleet.heartbeat(payload).to_dict()
+
+    @app.get("/v1/fleet", dependencies=[Depends(auth)])
+    def fleet_list() -> dict:
+        return {"members": state.fleet.list_members(),
+                "reconciliation": state.fleet.reconcile()}
+
+    @app.post("/v1/schedules", status_code=201, dependencies=[Depends(auth)])
+    def create_schedule(payload: dict) -> dict:
+        from agentbom_amd.api.fleet import ScanScheduler
+
+        if state.scheduler is None:
+            state.scheduler = ScanScheduler(
+                run_scan=lambda p: _run_scan(str(uuid.uuid4()), ScanRequest(demo=True)))
+            state.scheduler.start()
+        sched = state.scheduler.add(
+            payload.get("schedule_id") or str(uuid.uuid4()),
+            float(payload.get("interval_s", 3600)),
+            demo=bool(payload.get("demo", True)),
+        )
+        return sched.to_dict()
+
+    @app.get("/v1/schedules", dependencies=[Depends(auth)])
+    def list_schedules() -> dict:
+        if state.scheduler is None:
+            return {"schedules": []}
+        return {"schedules": [s.to_dict() for s in state.scheduler.schedules.values()]}
+
+    @app.delete("/v1/schedules/{schedule_id}", status_code=204, dependencies=[Depends(auth)])
+    def delete_schedule(schedule_id: str) -> None:
+        if state.scheduler is None or not state.scheduler.remove(schedule_id):
+            raise HTTPException(status_code=404, detail="schedule not found")
+
+    @app.post("/v1/proxy/audit", dependencies=[Depends(auth)])
+    def ingest_audit(payload: dict) -> dict:
+        entries = payload.get("entries", [])
+        if not isinstance(entries, list):
+            raise HTTPException(status_code=400, detail="entries must be a list")
+        state.audit_entries.extend(entries[:10_000])
+        return {"ingested": len(entries), "total": len(state.audit_entries)}
+
+    @app.get("/v1/proxy/audit", dependencies=[Depends(auth)])
+    def read_audit(limit: int = 100) -> dict:
+        return {"total": len(state.audit_entries),
+                "entries": state.audit_entries[-limit:]}
+
+    @app.get("/v1/compliance/{framework}/report", dependencies=[Depends(auth)])
+    def compliance_report(framework: str) -> dict:
+        from agentbom_amd.models import FRAMEWORK_TAG_FIELDS
+
+        report = _latest_report()
+        field_name = next((f for f, slug in FRAMEWORK_TAG_FIELDS if slug == framework), None)
+        if field_name is None:
+            raise HTTPException(status_code=404, detail=f"unknown framework {framework!r}")
+        from collections import Counter
+
+        counts: Counter = Counter()
+        rows = []
+        for br in report.blast_radii:
+            tags = getattr(br, field_name)
+            counts.update(tags)
+            if tags:
+                rows.append({"vulnerability_id": br.vulnerability.id,
+                             "package": f"{br.package.name}@{br.package.version}",
+                             "risk_score": br.risk_score, "controls": tags})
+        return {"framework": framework, "tagged_findings": len(rows),
+                "controls": dict(sorted(counts.items())), "findings": rows}
 
     return app
 
