@@ -1,0 +1,111 @@
+"""IaC scanning + Python SDK tests."""
+
+import sys
+import textwrap
+from pathlib import Path
+
+import pytest
+from fastapi.testclient import TestClient
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1] / "sdks" / "python"))
+
+from agentbom_amd.scan.iac import iac_finding_to_finding, scan_iac_text, scan_iac_tree
+
+
+class TestIac:
+    def test_terraform_rules(self):
+        tf = textwrap.dedent('''
+            resource "aws_security_group" "open" {
+              ingress { cidr_blocks = ["0.0.0.0/0"] }
+            }
+            resource "aws_db_instance" "db" {
+              publicly_accessible = true
+              password = "supersecret99"
+            }
+        ''')
+        hits = scan_iac_text(tf, "main.tf")
+        rules = {h.rule_id for h in hits}
+        assert {"TF001", "TF003", "TF004"} <= rules
+        assert any(h.severity == "critical" for h in hits)
+
+    def test_k8s_rules(self):
+        yaml_text = textwrap.dedent("""
+            apiVersion: v1
+            kind: Pod
+            spec:
+              hostNetwork: true
+              containers:
+              - name: x
+                securityContext:
+                  privileged: true
+                  runAsUser: 0
+        """)
+        hits = scan_iac_text(yaml_text, "pod.yaml")
+        assert {"K8S001", "K8S002", "K8S004"} <= {h.rule_id for h in hits}
+
+    def test_dockerfile_rules(self):
+        df = "FROM ubuntu\nUSER root\nRUN curl https://x.sh | bash\nENV API_KEY=abc123456\n"
+        hits = scan_iac_text(df, "Dockerfile")
+        assert {"DKR001", "DKR002", "DKR003", "DKR004"} <= {h.rule_id for h in hits}
+
+    def test_compose_rules(self):
+        text = "services:\n  app:\n    privileged: true\n    volumes:\n      - /var/run/docker.sock:/var/run/docker.sock\n"
+        hits = scan_iac_text(text, "docker-compose.yml")
+        assert {"CMP001", "CMP002"} <= {h.rule_id for h in hits}
+
+    def test_tree_walk_and_finding(self, tmp_path):
+        (tmp_path / "main.tf").write_text('acl = "public-read"\n')
+        (tmp_path / "node_modules").mkdir()
+        (tmp_path / "node_modules" / "x.tf").write_text('acl = "public-read"\n')
+        hits = scan_iac_tree(tmp_path)
+        assert len(hits) == 1
+        f = iac_finding_to_finding(hits[0])
+        assert f.finding_type.value == "CIS_FAIL"
+        assert f.attack_tags == ["T1530"]
+
+
+class TestSdk:
+    @pytest.fixture(scope="class")
+    def client(self):
+        import httpx
+        from agentbom_client import AgentBomClient
+
+        from agentbom_amd.api.server import create_app
+
+        tc = TestClient(create_app())
+
+        class _Bridge(httpx.BaseTransport):
+            """Route the SDK's sync httpx calls through the in-process app."""
+
+            def handle_request(self, request: httpx.Request) -> httpx.Response:
+                resp = tc.request(
+                    request.method,
+                    str(request.url.copy_with(scheme="http", host="testserver")),
+                    content=request.read(),
+                    headers=dict(request.headers),
+                )
+                return httpx.Response(resp.status_code, headers=resp.headers,
+                                      content=resp.content)
+
+        return AgentBomClient(base_url="http://testserver", transport=_Bridge())
+
+    def test_full_flow(self, client):
+        assert client.health()["status"] == "ok"
+        job = client.scan(demo=True)
+        assert job["status"] == "done"
+        report = client.scan_report(job["job_id"])
+        assert report["schema_version"] == "1.0"
+        assert client.findings(severity="critical")["findings"]
+        assert client.graph_search("pyyaml")["total"] == 1
+        assert client.should_i_deploy()["verdict"] == "block"
+        assert client.rollup()["containers"]
+        assert client.compliance_report("soc2")["tagged_findings"] > 0
+        hb = client.heartbeat("sdk-host", agents=1)
+        assert hb["status"] == "healthy"
+
+    def test_error_mapping(self, client):
+        from agentbom_client import AgentBomError
+
+        with pytest.raises(AgentBomError) as exc:
+            client.scan_job("missing")
+        assert exc.value.status_code == 404
