@@ -103,7 +103,7 @@ def _scan_impl(
     include_unfixed: bool, no_gpu: bool,
     sbom: Optional[str] = None, filesystem: Optional[str] = None,
     scan_secrets: bool = False, model_files: Optional[str] = None,
-    code: Optional[str] = None,
+    code: Optional[str] = None, iac: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -168,6 +168,12 @@ def _scan_impl(
         report.findings.extend(
             f for f in (model_result_to_finding(r) for r in results) if f is not None
         )
+    if iac:
+        from agentbom_amd.scan.iac import iac_finding_to_finding, scan_iac_tree
+
+        iac_hits = scan_iac_tree(iac)
+        report.iac_findings_data = {"findings": [h.to_dict() for h in iac_hits]}
+        report.findings.extend(iac_finding_to_finding(h) for h in iac_hits)
     if code:
         from agentbom_amd.scan.ast_analysis import (
             apply_symbol_reachability,
@@ -242,6 +248,8 @@ def _scan_options(f):
                      help="Scan ML model artifacts for pickle payloads."),
         click.option("--code", type=click.Path(exists=True), default=None,
                      help="AST security analysis + symbol-level CVE reachability."),
+        click.option("--iac", type=click.Path(exists=True), default=None,
+                     help="Scan Terraform/K8s/Dockerfile/compose for misconfigurations."),
     ]
     for o in reversed(opts):
         f = o(f)
