@@ -11,8 +11,12 @@ from __future__ import annotations
 
 import argparse
 import json
+import pathlib
 import statistics
+import sys
 import time
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parents[1]))
 
 
 def main() -> None:
