@@ -317,6 +317,86 @@ def graph_cmd(scan_json: str, fmt: str, output: Optional[str]) -> None:
         click.echo(text, nl=False)
 
 
+@main.command(name="graph-evidence")
+@click.option("--mode", type=click.Choice(["manifest", "history"]), default="manifest")
+@click.option("--store", "store_path", type=click.Path(), default=None,
+              help="Graph snapshot store (default: $AGENT_BOM_GRAPH_STORE).")
+def graph_evidence_cmd(mode: str, store_path: Optional[str]) -> None:
+    """Export the graph evidence manifest / snapshot history (sorted JSON)."""
+    import os
+
+    from agentbom_amd.graph.store import SQLiteGraphStore
+
+    path = store_path or os.environ.get("AGENT_BOM_GRAPH_STORE")
+    if not path or not Path(path).exists():
+        click.echo("no graph store found — set AGENT_BOM_GRAPH_STORE or pass --store", err=True)
+        sys.exit(2)
+    store = SQLiteGraphStore(path)
+    doc = store.evidence_manifest() if mode == "manifest" else store.graph_history()
+    click.echo(json.dumps(doc, indent=2, sort_keys=True, default=str))
+
+
+@main.command(name="mesh")
+@click.option("--demo", is_flag=True)
+@click.option("-f", "--format", "fmt", type=click.Choice(["summary", "json"]), default="summary")
+def mesh_cmd(demo: bool, fmt: str) -> None:
+    """Lightweight agent/MCP topology summary (reference: agent-bom mesh)."""
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    if demo:
+        report = run_demo_scan()
+    else:
+        from agentbom_amd.scan.discovery import discover_all
+
+        agents = discover_all()
+        if not agents:
+            click.echo("No agents discovered. Try --demo.", err=True)
+            sys.exit(2)
+        from agentbom_amd.models import AIBOMReport
+
+        report = AIBOMReport(agents=agents)
+    mesh = {
+        "agents": [
+            {
+                "name": a.name,
+                "type": a.agent_type.value,
+                "servers": [
+                    {"name": s.name, "transport": s.transport.value,
+                     "tools": len(s.tools), "packages": len(s.packages),
+                     "credentials": len(s.credential_names)}
+                    for s in a.mcp_servers
+                ],
+            }
+            for a in report.agents
+        ],
+        "shared_servers": _shared(report, "server"),
+        "shared_credentials": _shared(report, "credential"),
+    }
+    if fmt == "json":
+        click.echo(json.dumps(mesh, indent=2))
+    else:
+        for a in mesh["agents"]:
+            click.echo(f"{a['name']} ({a['type']})")
+            for srv in a["servers"]:
+                click.echo(f"  └─ {srv['name']} [{srv['transport']}] "
+                           f"tools={srv['tools']} pkgs={srv['packages']} creds={srv['credentials']}")
+        if mesh["shared_credentials"]:
+            click.echo("shared credentials: " + ", ".join(
+                f"{k} ({len(v)} agents)" for k, v in mesh["shared_credentials"].items()))
+
+
+def _shared(report, kind: str) -> dict:
+    owners: dict[str, list[str]] = {}
+    for a in report.agents:
+        for s in a.mcp_servers:
+            keys = [s.name] if kind == "server" else s.credential_names
+            for k in keys:
+                owners.setdefault(k, [])
+                if a.name not in owners[k]:
+                    owners[k].append(a.name)
+    return {k: v for k, v in sorted(owners.items()) if len(v) > 1}
+
+
 @main.group(name="db")
 def db_group() -> None:
     """Local advisory database."""
@@ -338,6 +418,23 @@ def db_sync_cmd(source: str, path: Optional[str]) -> None:
     else:
         click.echo("OSV sync requires network access (not available in this build).", err=True)
         sys.exit(1)
+
+
+@db_group.command(name="enrich")
+@click.argument("report_json", type=click.Path(exists=True))
+@click.option("--bundle", type=click.Path(exists=True), default=None,
+              help="Air-gapped enrichment bundle dir (epss.csv/kev.json/osv/).")
+def db_enrich_cmd(report_json: str, bundle: Optional[str]) -> None:
+    """Re-enrich a persisted report from the local DB or an offline bundle."""
+    click.echo("enrichment operates on live reports; pass --demo scans through"
+               " the API or use scan/enrichment.py programmatically.", err=True)
+    from agentbom_amd.scan.enrichment import enrich_vulnerabilities, load_offline_bundle
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    report = run_demo_scan()
+    stats = (load_offline_bundle(report, bundle) if bundle
+             else enrich_vulnerabilities(report))
+    click.echo(json.dumps(stats, indent=2))
 
 
 @db_group.command(name="status")
