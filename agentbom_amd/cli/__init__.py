@@ -104,6 +104,7 @@ def _scan_impl(
     sbom: Optional[str] = None, filesystem: Optional[str] = None,
     scan_secrets: bool = False, model_files: Optional[str] = None,
     code: Optional[str] = None, iac: Optional[str] = None,
+    aws_inventory: Optional[str] = None, endpoint: bool = False,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -168,6 +169,18 @@ def _scan_impl(
         report.findings.extend(
             f for f in (model_result_to_finding(r) for r in results) if f is not None
         )
+    if aws_inventory:
+        from agentbom_amd.scan.cloud import cis_result_to_finding, scan_cloud_inventory
+
+        cis = scan_cloud_inventory(aws_inventory)
+        report.extra_data["cis_benchmark_data"] = [r.to_dict() for r in cis]
+        report.findings.extend(
+            f for f in (cis_result_to_finding(r) for r in cis) if f is not None
+        )
+    if endpoint:
+        from agentbom_amd.scan.endpoint import collect_endpoint_inventory
+
+        report.extra_data["endpoint_inventory_data"] = collect_endpoint_inventory().to_dict()
     if iac:
         from agentbom_amd.scan.iac import iac_finding_to_finding, scan_iac_tree
 
@@ -250,6 +263,10 @@ def _scan_options(f):
                      help="AST security analysis + symbol-level CVE reachability."),
         click.option("--iac", type=click.Path(exists=True), default=None,
                      help="Scan Terraform/K8s/Dockerfile/compose for misconfigurations."),
+        click.option("--aws-inventory", type=click.Path(exists=True), default=None,
+                     help="Evaluate CIS checks over an exported AWS inventory JSON."),
+        click.option("--endpoint", is_flag=True,
+                     help="Collect bounded workstation endpoint inventory."),
     ]
     for o in reversed(opts):
         f = o(f)

@@ -105,3 +105,19 @@ class TestApiRoutes:
         assert r["tagged_findings"] > 0
         assert "RA-5" in r["controls"]
         assert client.get("/v1/compliance/bogus/report").status_code == 404
+
+
+class TestOpenApiContract:
+    def test_spec_matches_committed_snapshot(self):
+        """The committed docs/openapi/v1.json is the REST contract; route
+        changes must re-export it (reference: PROJECT_STRUCTURE.md:31)."""
+        import json
+        from pathlib import Path
+
+        from agentbom_amd.api.server import create_app
+
+        committed = json.loads(
+            (Path(__file__).resolve().parents[1] / "docs/openapi/v1.json").read_text())
+        live = create_app().openapi()
+        assert sorted(live["paths"]) == sorted(committed["paths"]), (
+            "REST surface changed — regenerate docs/openapi/v1.json")
