@@ -116,6 +116,10 @@ __global__ void bfs_expand_heavy_kernel(
 // share of all edges (the agent->server->everything levels of estate
 // graphs), where per-vertex serial neighbor loops are latency-bound.
 // Requires the per-edge source array (edge_src, same order as col).
+// build_frontier == 0: dist-driven mode for subsequent edge-centric levels —
+// claims are only counted (one wave-reduced atomic per wave), no frontier
+// append, no degree accumulation.  Removes the single-counter atomic storm
+// on claim-heavy levels (~10M returning atomicAdds on one word otherwise).
 __global__ void bfs_expand_edges_kernel(
     const uint32_t* __restrict__ edge_src,   // [E]
     const uint32_t* __restrict__ col,        // [E]
@@ -128,9 +132,11 @@ __global__ void bfs_expand_edges_kernel(
     uint32_t* __restrict__ next_frontier,
     unsigned int* __restrict__ next_count,
     unsigned int* __restrict__ next_degree_sum,
+    int build_frontier,
     long long capacity) {
     const long long stride = (long long)gridDim.x * blockDim.x;
     unsigned my_deg = 0;
+    unsigned my_claims = 0;
     for (long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x; e < num_edges;
          e += stride) {
         if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
@@ -138,12 +144,38 @@ __global__ void bfs_expand_edges_kernel(
         const uint32_t v = col[e];
         if (dist[v] == ABOM_UNVISITED &&
             atomicCAS(&dist[v], ABOM_UNVISITED, cur_level + 1) == ABOM_UNVISITED) {
-            const unsigned idx = atomicAdd(next_count, 1u);
-            if ((long long)idx < capacity) next_frontier[idx] = v;
-            my_deg += (unsigned)(row_off[v + 1] - row_off[v]);
+            if (build_frontier) {
+                const unsigned idx = atomicAdd(next_count, 1u);
+                if ((long long)idx < capacity) next_frontier[idx] = v;
+                my_deg += (unsigned)(row_off[v + 1] - row_off[v]);
+            } else {
+                ++my_claims;
+            }
         }
     }
     wave_add_degree(next_degree_sum, my_deg);
+    if (!build_frontier) wave_add_degree(next_count, my_claims);
+}
+
+// Rebuild a frontier from dist (nodes claimed at `level`): used when
+// dist-driven dense mode hands back to vertex-frontier mode after the
+// claim rate drops.  Claim counts are small here, so appends are cheap.
+__global__ void collect_frontier_kernel(
+    const uint32_t* __restrict__ dist, long long num_nodes, uint32_t level,
+    const uint64_t* __restrict__ row_off,
+    uint32_t* __restrict__ frontier, unsigned int* __restrict__ count,
+    unsigned int* __restrict__ degree_sum, long long capacity) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    unsigned my_deg = 0;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < num_nodes;
+         i += stride) {
+        if (dist[i] == level) {
+            const unsigned idx = atomicAdd(count, 1u);
+            if ((long long)idx < capacity) frontier[idx] = (uint32_t)i;
+            my_deg += (unsigned)(row_off[i + 1] - row_off[i]);
+        }
+    }
+    wave_add_degree(degree_sum, my_deg);
 }
 
 __global__ void init_dist_kernel(uint32_t* __restrict__ dist, long long n, uint32_t value) {
@@ -403,15 +435,15 @@ extern "C" int abom_bfs_expand_heavy(
 extern "C" int abom_bfs_expand_edges(
     const void* edge_src, const void* col, const void* etype, unsigned int allowed_mask,
     long long num_edges, const void* row_off, void* dist, unsigned int cur_level,
-    void* next_frontier, void* next_count, void* next_degree_sum, long long capacity,
-    void* stream) {
+    void* next_frontier, void* next_count, void* next_degree_sum, int build_frontier,
+    long long capacity, void* stream) {
     const int block = 256;
     hipLaunchKernelGGL(abom::bfs_expand_edges_kernel, dim3(abom::grid_for(num_edges, block)),
                        dim3(block), 0, (hipStream_t)stream, (const uint32_t*)edge_src,
                        (const uint32_t*)col, (const uint8_t*)etype, allowed_mask, num_edges,
                        (const uint64_t*)row_off, (uint32_t*)dist, cur_level,
                        (uint32_t*)next_frontier, (unsigned int*)next_count,
-                       (unsigned int*)next_degree_sum, capacity);
+                       (unsigned int*)next_degree_sum, build_frontier, capacity);
     return (int)hipGetLastError();
 }
 
@@ -446,17 +478,48 @@ extern "C" int abom_bfs_run(
     uint32_t* nxt = (uint32_t*)frontier_b;
     long long frontier_size = n_sources;
     int level = 0;
-    while (frontier_size > 0 && level < max_levels) {
+    bool stay_dense = false;
+    while (frontier_size != 0 && level < max_levels) {
         ++level;
         ABOM_CHECK(hipMemsetAsync(ctr, 0, 3 * sizeof(unsigned int), s));
         // Dense frontier (its edges are a big share of ALL edges): the
         // one-thread-per-edge pass with coalesced src/col streams beats
-        // per-vertex serial neighbor loops.
-        const bool dense = edge_src != nullptr && num_edges > 0 &&
-                           (double)frontier_degree > (double)num_edges / 8.0;
+        // per-vertex serial neighbor loops.  Once dense mode fires, stay
+        // dist-driven: subsequent edge-centric levels never materialize a
+        // frontier (build_frontier=0), claims are wave-counted only.
+        const bool dense = stay_dense ||
+                           (edge_src != nullptr && num_edges > 0 &&
+                            (double)frontier_degree > (double)num_edges / 8.0);
         if (dense) {
+            stay_dense = true;
             rc = abom_bfs_expand_edges(edge_src, col, etype, allowed_mask, num_edges, row_off,
                                        dist, (unsigned int)(level - 1), nxt, ctr, ctr + 2,
+                                       /*build_frontier=*/0, num_nodes, stream);
+            if (rc) return -rc;
+        } else if (frontier_size < 0) {
+            // dense mode handed back: rebuild the frontier from dist
+            hipLaunchKernelGGL(abom::collect_frontier_kernel,
+                               dim3(abom::grid_for(num_nodes, 256)), dim3(256), 0, s,
+                               (const uint32_t*)dist, num_nodes,
+                               (unsigned int)(level - 1), (const uint64_t*)row_off,
+                               cur, ctr + 1, ctr + 2, num_nodes);
+            rc = (int)hipGetLastError();
+            if (rc) return -rc;
+            unsigned int rebuilt[2] = {0, 0};
+            ABOM_CHECK(hipMemcpyAsync(&rebuilt[0], ctr + 1, sizeof(unsigned int),
+                                      hipMemcpyDeviceToHost, s));
+            ABOM_CHECK(hipStreamSynchronize(s));
+            frontier_size = rebuilt[0];
+            // reset all three counters: the rebuilt frontier's degree sum is
+            // the CURRENT level's, while the dense decision needs the NEXT
+            // frontier's (accumulated by the expand below)
+            ABOM_CHECK(hipMemsetAsync(ctr, 0, 3 * sizeof(unsigned int), s));
+            rc = abom_bfs_expand(row_off, col, etype, allowed_mask, cur, frontier_size, dist,
+                                 (unsigned int)level, nxt, ctr, heavy_queue, ctr + 1, ctr + 2,
+                                 num_nodes, stream);
+            if (rc) return -rc;
+            rc = abom_bfs_expand_heavy(row_off, col, etype, allowed_mask, heavy_queue, ctr + 1,
+                                       dist, (unsigned int)level, nxt, ctr, ctr + 2,
                                        num_nodes, stream);
             if (rc) return -rc;
         } else {
@@ -475,6 +538,13 @@ extern "C" int abom_bfs_run(
         ABOM_CHECK(hipStreamSynchronize(s));
         frontier_size = host_pair[0];
         frontier_degree = host_pair[2];
+        if (stay_dense && frontier_size > 0 &&
+            (double)frontier_size < (double)num_edges / 4096.0) {
+            // dense claims have trickled out: hand back to vertex mode next
+            // level (frontier rebuilt from dist — signalled by negative size)
+            stay_dense = false;
+            frontier_size = -1;
+        }
         uint32_t* t = cur; cur = nxt; nxt = t;
     }
     return level;

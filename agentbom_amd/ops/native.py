@@ -35,7 +35,7 @@ _SIGNATURES = {
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
     "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _c, _i64, _c], _i32),
     "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c], _i32),
-    "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _c, _u32, _c, _c, _c, _i64, _c], _i32),
+    "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _c, _u32, _c, _c, _c, _i32, _i64, _c], _i32),
     "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
     "abom_blast_counts": ([_c, _i64] + [_c] * 6 + [_i32] * 4 + [_c] * 4 + [_c], _i32),
