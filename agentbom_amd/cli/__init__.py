@@ -19,8 +19,8 @@ from agentbom_amd import __version__
 
 FORMATS = [
     "console", "json", "html", "sarif", "cyclonedx", "spdx", "spdx2", "ocsf",
-    "csv", "markdown", "plain", "junit", "prometheus", "graph", "mermaid",
-    "dot", "graphml", "cypher",
+    "csv", "markdown", "plain", "junit", "prometheus", "parquet", "svg",
+    "badge", "graph", "mermaid", "dot", "graphml", "cypher",
 ]
 
 
@@ -72,6 +72,13 @@ def _render(report, fmt: str, output: Optional[str], verbose: bool) -> None:
         text = misc_fmt.to_junit(report)
     elif fmt == "prometheus":
         text = misc_fmt.to_prometheus(report)
+    elif fmt == "parquet":
+        if not output:
+            raise click.UsageError("parquet requires -o/--output")
+        Path(output).write_bytes(misc_fmt.to_parquet_bytes(report))
+        return
+    elif fmt in ("svg", "badge"):
+        text = misc_fmt.to_badge_svg(report)
     elif fmt in ("plain",):
         text = misc_fmt.to_markdown(report)
     elif fmt in ("graph", "mermaid", "dot", "graphml", "cypher"):

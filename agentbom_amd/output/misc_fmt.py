@@ -133,3 +133,62 @@ def to_junit(report: AIBOMReport) -> str:
         out.append("  </testcase>")
     out.append("</testsuite>")
     return "\n".join(out) + "\n"
+
+
+def to_parquet_bytes(report: AIBOMReport) -> bytes:
+    """Findings table as Parquet (pyarrow) — the analytics export."""
+    import io as _io
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    rows = []
+    for f in report.to_findings():
+        ev = f.evidence if isinstance(f.evidence, dict) else {}
+        rows.append({
+            "finding_id": f.id, "finding_type": f.finding_type.value,
+            "severity": f.severity, "risk_score": float(f.risk_score),
+            "vulnerability_id": f.vulnerability_id or "",
+            "package": str(ev.get("package_name", "")),
+            "version": str(ev.get("package_version", "")),
+            "ecosystem": str(ev.get("ecosystem", "")),
+            "is_kev": bool(f.is_kev), "is_malicious": bool(f.is_malicious),
+            "epss_score": float(f.epss_score) if f.epss_score is not None else None,
+            "cvss_score": float(f.cvss_score) if f.cvss_score is not None else None,
+            "reachability": f.reachability or "",
+            "impact_category": f.impact_category or "",
+            "affected_agents": list(f.affected_agents),
+            "exposed_credentials": list(f.exposed_credentials),
+            "asset_name": f.asset.name, "asset_type": f.asset.asset_type,
+        })
+    table = pa.Table.from_pylist(rows)
+    buf = _io.BytesIO()
+    pq.write_table(table, buf)
+    return buf.getvalue()
+
+
+_BADGE_COLORS = {"critical": "#e05d44", "high": "#fe7d37", "medium": "#dfb317",
+                 "clean": "#4c1"}
+
+
+def to_badge_svg(report: AIBOMReport) -> str:
+    """Shields-style SVG badge summarizing the scan verdict."""
+    counts = report.severity_counts()
+    if counts["critical"]:
+        label, color = f"{counts['critical']} critical", _BADGE_COLORS["critical"]
+    elif counts["high"]:
+        label, color = f"{counts['high']} high", _BADGE_COLORS["high"]
+    elif counts["medium"]:
+        label, color = f"{counts['medium']} medium", _BADGE_COLORS["medium"]
+    else:
+        label, color = "clean", _BADGE_COLORS["clean"]
+    left, right = "agent-bom", label
+    lw, rw = 6 * len(left) + 10, 6 * len(right) + 10
+    return f"""<svg xmlns="http://www.w3.org/2000/svg" width="{lw + rw}" height="20">
+<rect width="{lw}" height="20" fill="#555"/>
+<rect x="{lw}" width="{rw}" height="20" fill="{color}"/>
+<g fill="#fff" text-anchor="middle" font-family="Verdana,sans-serif" font-size="11">
+<text x="{lw / 2}" y="14">{left}</text>
+<text x="{lw + rw / 2}" y="14">{right}</text>
+</g></svg>
+"""

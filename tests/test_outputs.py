@@ -172,3 +172,19 @@ class TestGraphExports:
         out = buf.getvalue()
         assert "CVE-2020-14343" in out
         assert "AI-BOM Scan Report" in out
+
+
+class TestParquetBadge:
+    def test_parquet(self, report, tmp_path):
+        import pyarrow.parquet as pq
+
+        data = misc_fmt.to_parquet_bytes(report)
+        p = tmp_path / "f.parquet"
+        p.write_bytes(data)
+        table = pq.read_table(p)
+        assert table.num_rows == len(report.to_findings())
+        assert "risk_score" in table.column_names
+
+    def test_badge(self, report):
+        svg = misc_fmt.to_badge_svg(report)
+        assert svg.startswith("<svg") and "critical" in svg
