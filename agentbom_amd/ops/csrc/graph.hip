@@ -142,15 +142,20 @@ __global__ void bfs_expand_edges_kernel(
         if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
         if (dist[edge_src[e]] != cur_level) continue;
         const uint32_t v = col[e];
-        if (dist[v] == ABOM_UNVISITED &&
-            atomicCAS(&dist[v], ABOM_UNVISITED, cur_level + 1) == ABOM_UNVISITED) {
-            if (build_frontier) {
+        if (build_frontier) {
+            if (dist[v] == ABOM_UNVISITED &&
+                atomicCAS(&dist[v], ABOM_UNVISITED, cur_level + 1) == ABOM_UNVISITED) {
                 const unsigned idx = atomicAdd(next_count, 1u);
                 if ((long long)idx < capacity) next_frontier[idx] = v;
                 my_deg += (unsigned)(row_off[v + 1] - row_off[v]);
-            } else {
-                ++my_claims;
             }
+        } else if (dist[v] == ABOM_UNVISITED) {
+            // dist-driven mode: every racer writes the SAME value, so a plain
+            // store is exact (level-synchronous); the claim count may double-
+            // count racers but is only used for termination (>0) and the
+            // hand-back threshold — both tolerant of overcounting.
+            dist[v] = cur_level + 1;
+            ++my_claims;
         }
     }
     wave_add_degree(next_degree_sum, my_deg);

@@ -49,21 +49,39 @@ __global__ void match_kernel(
     unsigned int* __restrict__ out_count,
     long long capacity) {
     const long long stride = (long long)gridDim.x * blockDim.x;
+    const int lane = threadIdx.x & 63;
     for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x; p < num_packages;
          p += stride) {
-        if (!(pkg_flags[p] & PF_ENCODABLE)) continue;
-        const uint64_t gkey = pkg_group_key[p];
+        const bool enc = (pkg_flags[p] & PF_ENCODABLE) != 0;
+        const uint64_t gkey = enc ? pkg_group_key[p] : 0;
 
-        // Branchless-ish binary search over sorted group keys.
-        long long lo = 0, hi = num_groups;
-        while (lo < hi) {
-            long long mid = (lo + hi) >> 1;
-            if (group_keys[mid] < gkey) lo = mid + 1; else hi = mid;
+        // Packages are laid out sorted by group key (engine build), so most
+        // lanes in a wave share their key with a predecessor: only segment
+        // LEADERS run the binary search; followers copy the result via a
+        // leader-index max-scan + shuffle (wave64 segmented broadcast).
+        const uint64_t prev_key = __shfl_up(gkey, 1, 64);
+        const bool leader = (lane == 0) || (gkey != prev_key) || !enc;
+        long long found = -1;
+        if (leader && enc) {
+            long long lo = 0, hi = num_groups;
+            while (lo < hi) {
+                long long mid = (lo + hi) >> 1;
+                if (group_keys[mid] < gkey) lo = mid + 1; else hi = mid;
+            }
+            found = (lo < num_groups && group_keys[lo] == gkey) ? lo : -1;
         }
-        if (lo >= num_groups || group_keys[lo] != gkey) continue;
+        // inclusive max-scan of leader lane indices -> my segment's leader
+        int leader_lane = leader ? lane : -1;
+        #pragma unroll
+        for (int off = 1; off < 64; off <<= 1) {
+            int up = __shfl_up(leader_lane, off, 64);
+            if (lane >= off && up > leader_lane) leader_lane = up;
+        }
+        found = __shfl(found, leader_lane, 64);
+        if (!enc || found < 0) continue;
 
-        const uint32_t wbeg = group_off[lo];
-        const uint32_t wend = group_off[lo + 1];
+        const uint32_t wbeg = group_off[found];
+        const uint32_t wend = group_off[found + 1];
         const uint64_t khi = pkg_key_hi[p];
         const uint64_t klo = pkg_key_lo[p];
 
