@@ -320,6 +320,75 @@ def mcp_server_cmd(demo: bool) -> None:
     run_stdio_server(demo=demo)
 
 
+@mcp_group.command(name="introspect")
+@click.option("--timeout", type=float, default=10.0, show_default=True)
+@click.option("--server", "server_spec", default=None,
+              help="Introspect one ad-hoc server: 'name:command arg1 arg2'.")
+@click.option("-o", "--output", type=click.Path(), default=None)
+def mcp_introspect_cmd(timeout: float, server_spec: Optional[str],
+                       output: Optional[str]) -> None:
+    """Connect to configured MCP servers (read-only) and diff runtime tools."""
+    import json as _json
+
+    from agentbom_amd.mcp.introspect import introspect_servers
+
+    servers = _resolve_introspection_targets(server_spec)
+    if not servers:
+        click.echo("no stdio MCP servers discovered", err=True)
+        raise SystemExit(2)
+    report = introspect_servers(servers, timeout=timeout)
+    text = _json.dumps(report.to_dict(), indent=2, default=str)
+    if output:
+        Path(output).write_text(text)
+        click.echo(f"wrote {output}")
+    else:
+        click.echo(text)
+    raise SystemExit(1 if report.drift_count else 0)
+
+
+@mcp_group.command(name="health")
+@click.option("--timeout", type=float, default=5.0, show_default=True)
+@click.option("--server", "server_spec", default=None,
+              help="Probe one ad-hoc server: 'name:command arg1 arg2'.")
+def mcp_health_cmd(timeout: float, server_spec: Optional[str]) -> None:
+    """Liveness-probe configured MCP servers (spawn + initialize only)."""
+    import json as _json
+
+    from agentbom_amd.mcp.introspect import health_check_servers
+
+    servers = _resolve_introspection_targets(server_spec)
+    if not servers:
+        click.echo("no stdio MCP servers discovered", err=True)
+        raise SystemExit(2)
+    statuses = health_check_servers(servers, timeout=timeout)
+    click.echo(_json.dumps([s.to_dict() for s in statuses], indent=2))
+    raise SystemExit(0 if all(s.healthy for s in statuses) else 1)
+
+
+def _resolve_introspection_targets(server_spec: Optional[str]):
+    from agentbom_amd.models import MCPServer
+
+    if server_spec:
+        name, _, cmdline = server_spec.partition(":")
+        parts = cmdline.split()
+        if not parts:
+            return []
+        return [MCPServer(name=name or parts[0], command=parts[0], args=parts[1:])]
+    from agentbom_amd.scan.discovery import discover_agents
+
+    servers = []
+    for agent in discover_agents():
+        servers.extend(s for s in agent.mcp_servers if s.command)
+    # de-dup by (command, args) — several clients often share one server
+    seen, out = set(), []
+    for s in servers:
+        key = (s.command, tuple(s.args))
+        if key not in seen:
+            seen.add(key)
+            out.append(s)
+    return out
+
+
 @main.command(name="graph")
 @click.argument("scan_json", type=click.Path(exists=True))
 @click.option("-f", "--format", "fmt",
