@@ -25,16 +25,100 @@ class McpTool:
 
 
 class AgentBomMcpServer:
-    """Tool registry + dispatch over a shared scan/graph session."""
+    """Tool registry + dispatch over a shared scan/graph session.
+
+    Guardrails on every call (reference: mcp_server_runtime.py):
+    response truncation to MAX_RESPONSE_BYTES, per-caller token-bucket rate
+    limiting, and per-tool latency/error metrics.
+    """
 
     PROTOCOL_VERSION = "2024-11-05"
+    MAX_RESPONSE_BYTES = 512 * 1024
+    RATE_LIMIT_CALLS = 120  # per caller per minute
 
     def __init__(self, demo: bool = False):
         self.demo = demo
         self._report = None
         self._graph = None
         self.tools: dict[str, McpTool] = {}
+        self.tool_metrics: dict[str, dict] = {}
+        self._rate_window: dict[str, list[float]] = {}
         self._register_tools()
+        from agentbom_amd.mcp.tools_operator import register_operator_tools
+        from agentbom_amd.mcp.tools_specialized import register_specialized_tools
+
+        register_operator_tools(self)
+        register_specialized_tools(self)
+
+    # shared operator state, built lazily so `tools/list` stays instant
+    _identity_store = None
+    _ticket_store = None
+    _campaign_store = None
+    _shield = None
+
+    @property
+    def identity_store(self):
+        if self._identity_store is None:
+            from agentbom_amd.identity import AgentIdentityStore
+
+            self._identity_store = AgentIdentityStore()
+        return self._identity_store
+
+    @property
+    def ticket_store(self):
+        if self._ticket_store is None:
+            from agentbom_amd.runtime.ticketing import TicketStore
+
+            self._ticket_store = TicketStore()
+        return self._ticket_store
+
+    @property
+    def campaign_store(self):
+        if self._campaign_store is None:
+            from agentbom_amd.scan.campaigns import CampaignStore
+
+            self._campaign_store = CampaignStore()
+        return self._campaign_store
+
+    @property
+    def shield(self):
+        if self._shield is None:
+            from agentbom_amd.runtime.shield import Shield
+
+            self._shield = Shield()
+        return self._shield
+
+    def tool(self, name: str, description: str, schema: Optional[dict] = None):
+        """Decorator used by the core registry and every tool group."""
+
+        def deco(fn):
+            self.tools[name] = McpTool(
+                name, description,
+                schema or {"type": "object", "properties": {},
+                           "additionalProperties": False},
+                fn,
+            )
+            return fn
+        return deco
+
+    def _check_rate_limit(self, caller: str) -> bool:
+        import time
+
+        now = time.monotonic()
+        window = self._rate_window.setdefault(caller, [])
+        window[:] = [t for t in window if now - t < 60.0]
+        if len(window) >= self.RATE_LIMIT_CALLS:
+            return False
+        window.append(now)
+        return True
+
+    def _record_metric(self, name: str, ms: float, error: bool) -> None:
+        m = self.tool_metrics.setdefault(
+            name, {"calls": 0, "errors": 0, "total_ms": 0.0})
+        m["calls"] += 1
+        m["total_ms"] += ms
+        if error:
+            m["errors"] += 1
 
     # ── session state ─────────────────────────────────────────────────────
 
@@ -54,15 +138,7 @@ class AgentBomMcpServer:
     # ── tools ─────────────────────────────────────────────────────────────
 
     def _register_tools(self) -> None:
-        def tool(name: str, description: str, schema: Optional[dict] = None):
-            def deco(fn):
-                self.tools[name] = McpTool(
-                    name, description,
-                    schema or {"type": "object", "properties": {}, "additionalProperties": False},
-                    fn,
-                )
-                return fn
-            return deco
+        tool = self.tool
 
         @tool("scan", "Security Scan: discover agents/MCP servers and scan their packages "
                       "for vulnerabilities, returning the findings summary.")
@@ -236,6 +312,171 @@ class AgentBomMcpServer:
                 ]
             }
 
+        @tool("intel_lookup", "Threat-intel view of one vulnerability: "
+                              "advisories, KEV, EPSS, CVSS, affected windows.",
+              {"type": "object", "properties": {"vuln_id": {"type": "string"}},
+               "required": ["vuln_id"]})
+        def intel_lookup(vuln_id: str) -> dict:
+            from agentbom_amd.db.store import load_advisory_windows
+
+            windows = [w for w in load_advisory_windows(offline=True)
+                       if w.vuln_id == vuln_id]
+            if not windows:
+                return {"vuln_id": vuln_id, "found": False}
+            w0 = windows[0]
+            return {"vuln_id": vuln_id, "found": True,
+                    "severity": w0.severity.value, "cvss_score": w0.cvss_score,
+                    "is_kev": w0.is_kev, "epss_score": w0.epss_score,
+                    "summary": w0.summary,
+                    "affected": [{"ecosystem": w.ecosystem,
+                                  "package": w.package_name,
+                                  "introduced": w.introduced, "fixed": w.fixed,
+                                  "last_affected": w.last_affected}
+                                 for w in windows]}
+
+        @tool("intel_match", "Match a package list against the advisory arena "
+                             "in one batch.",
+              {"type": "object", "properties": {
+                  "packages": {"type": "array", "items": {"type": "object"}}},
+               "required": ["packages"]})
+        def intel_match(packages: list) -> dict:
+            from agentbom_amd.db.store import load_advisory_windows
+            from agentbom_amd.utils.canonical_ids import normalize_package_name
+            from agentbom_amd.utils.version_utils import version_in_range
+
+            windows = load_advisory_windows(offline=True)
+            out = []
+            for p in packages[:10_000]:
+                name, version = str(p.get("name", "")), str(p.get("version", ""))
+                eco = str(p.get("ecosystem", ""))
+                norm = normalize_package_name(name, eco)
+                hits = [w.vuln_id for w in windows
+                        if w.ecosystem.lower() == eco.lower()
+                        and normalize_package_name(w.package_name, w.ecosystem) == norm
+                        and version_in_range(version, w.introduced, w.fixed,
+                                             w.last_affected, eco)]
+                if hits:
+                    out.append({"package": f"{name}@{version}",
+                                "ecosystem": eco, "vuln_ids": sorted(set(hits))})
+            return {"matched": len(out), "results": out}
+
+        @tool("intel_sources", "Advisory/intel sources bundled in this build "
+                               "and their record counts.")
+        def intel_sources() -> dict:
+            from agentbom_amd.db.store import load_advisory_windows
+
+            windows = load_advisory_windows(offline=True)
+            by_eco: dict[str, int] = {}
+            for w in windows:
+                by_eco[w.ecosystem] = by_eco.get(w.ecosystem, 0) + 1
+            return {"sources": ["osv-bundled", "cisa-kev-bundled",
+                                "epss-bundled", "malicious-registry"],
+                    "advisory_windows": len(windows),
+                    "kev_entries": sum(1 for w in windows if w.is_kev),
+                    "by_ecosystem": by_eco}
+
+        @tool("intel_daily_brief", "Operator brief: top risks, KEV exposure, "
+                                   "malicious hits, deploy verdict.")
+        def intel_daily_brief() -> dict:
+            report, _g = self._ensure_scan()
+            top = report.blast_radii[:5]
+            return {
+                "date_scope": "latest scan",
+                "headline_risks": [{
+                    "vuln_id": b.vulnerability.id, "risk_score": b.risk_score,
+                    "package": f"{b.package.name}@{b.package.version}",
+                    "is_kev": b.vulnerability.is_kev,
+                    "summary": b.attack_vector_summary} for b in top],
+                "kev_count": sum(1 for b in report.blast_radii
+                                 if b.vulnerability.is_kev),
+                "malicious_count": sum(1 for b in report.blast_radii
+                                       if b.package.is_malicious),
+                "deploy_verdict": self.tools["should_i_deploy"].fn(),
+            }
+
+        @tool("remediate", "Prioritized remediation plan: what to upgrade, in "
+                           "what order, and what it clears.")
+        def remediate() -> dict:
+            from agentbom_amd.output.json_fmt import _build_remediation_json
+
+            report, _g = self._ensure_scan()
+            return {"plan": _build_remediation_json(report)}
+
+        @tool("verify", "Re-scan and verify whether a finding is resolved.",
+              {"type": "object", "properties": {"vuln_id": {"type": "string"}},
+               "required": ["vuln_id"]})
+        def verify(vuln_id: str) -> dict:
+            self._report = None  # force a fresh scan
+            self._graph = None
+            report, _g = self._ensure_scan()
+            still = [b for b in report.blast_radii
+                     if b.vulnerability.id == vuln_id]
+            return {"vuln_id": vuln_id,
+                    "resolved": not still,
+                    "still_affected": [f"{b.package.name}@{b.package.version}"
+                                       for b in still]}
+
+        @tool("inventory_summary", "Asset inventory rollup for the latest scan.")
+        def inventory_summary() -> dict:
+            from agentbom_amd.output.json_fmt import to_json
+
+            report, _g = self._ensure_scan()
+            doc = to_json(report)
+            return doc["inventory_snapshot"]
+
+        @tool("inventory_list", "All inventoried assets with finding counts.",
+              {"type": "object", "properties": {
+                  "asset_type": {"type": "string", "default": ""}}})
+        def inventory_list(asset_type: str = "") -> dict:
+            from agentbom_amd.output.json_fmt import to_json
+
+            report, _g = self._ensure_scan()
+            assets = to_json(report)["assets"]
+            if asset_type:
+                assets = [a for a in assets if a.get("asset_type") == asset_type]
+            return {"total": len(assets), "assets": assets}
+
+        @tool("inventory_asset", "One asset's full detail incl. its findings.",
+              {"type": "object", "properties": {"name": {"type": "string"}},
+               "required": ["name"]})
+        def inventory_asset(name: str) -> dict:
+            from agentbom_amd.output.json_fmt import to_json
+
+            report, _g = self._ensure_scan()
+            doc = to_json(report)
+            asset = next((a for a in doc["assets"] if a.get("name") == name), None)
+            if asset is None:
+                return {"error": f"asset {name!r} not found"}
+            findings = [f for f in doc["findings"]
+                        if (f.get("asset") or {}).get("name") == name]
+            return {"asset": asset, "findings": findings}
+
+        @tool("registry_lookup", "Look up an MCP server in the known-server "
+                                 "registry (verification + risk notes).",
+              {"type": "object", "properties": {"name": {"type": "string"}},
+               "required": ["name"]})
+        def registry_lookup(name: str) -> dict:
+            report, _g = self._ensure_scan()
+            matches = [s for a in report.agents for s in a.mcp_servers
+                       if name.lower() in s.name.lower()]
+            if not matches:
+                return {"name": name, "found": False,
+                        "note": "not in latest scan; run marketplace_check for "
+                                "pre-install trust signals"}
+            from agentbom_amd.scan.risk import score_server_risk, server_risk_level
+
+            out = []
+            for s in matches:
+                score = score_server_risk(s)
+                out.append({"name": s.name, "command": s.command,
+                            "registry_verified": s.registry_verified,
+                            "tool_count": len(s.tools),
+                            "credentials": s.credential_names,
+                            "risk_score": score,
+                            "risk_level": server_risk_level(score),
+                            "security_warnings": s.security_warnings})
+            return {"name": name, "found": True, "servers": out}
+
         @tool("db_status", "Local advisory DB freshness and counts.")
         def db_status() -> dict:
             from agentbom_amd.db.store import AdvisoryStore, default_db_path
@@ -279,14 +520,31 @@ class AgentBomMcpServer:
                 ]
             })
         if method == "tools/call":
+            import time as _time
+
             name = params.get("name")
             tool = self.tools.get(name)
             if tool is None:
                 return err(-32602, f"unknown tool {name!r}")
+            caller = str((params.get("_meta") or {}).get("caller", "default"))
+            if not self._check_rate_limit(caller):
+                return ok({"content": [{"type": "text", "text": json.dumps(
+                    {"error": "rate limit exceeded", "retry_after_s": 60})}],
+                    "isError": True})
+            t0 = _time.perf_counter()
             try:
                 result = tool.fn(**(params.get("arguments") or {}))
-                return ok({"content": [{"type": "text", "text": json.dumps(result, default=str)}]})
+                text = json.dumps(result, default=str)
+                if len(text) > self.MAX_RESPONSE_BYTES:
+                    text = json.dumps({
+                        "truncated": True,
+                        "full_bytes": len(text),
+                        "preview": text[: self.MAX_RESPONSE_BYTES // 2],
+                    })
+                self._record_metric(name, (_time.perf_counter() - t0) * 1000, False)
+                return ok({"content": [{"type": "text", "text": text}]})
             except Exception as exc:  # noqa: BLE001 — tool boundary
+                self._record_metric(name, (_time.perf_counter() - t0) * 1000, True)
                 return ok({"content": [{"type": "text", "text": json.dumps({"error": str(exc)})}],
                            "isError": True})
         if method == "resources/list":
