@@ -63,6 +63,7 @@ class _State:
             max_concurrent=int(os.environ.get("AGENT_BOM_MAX_CONCURRENT_SCANS", "8")))
         self.scheduler = None
         self.audit_entries: list[dict] = []
+        self.identity_store = None
 
 
 def create_app() -> FastAPI:
@@ -438,6 +439,155 @@ def create_app() -> FastAPI:
                              "risk_score": br.risk_score, "controls": tags})
         return {"framework": framework, "tagged_findings": len(rows),
                 "controls": dict(sorted(counts.items())), "findings": rows}
+
+    # ── identity lifecycle (reference: api/routes/identities.py) ───────────
+
+    def _identity_store():
+        if state.identity_store is None:
+            from agentbom_amd.identity import AgentIdentityStore
+
+            state.identity_store = AgentIdentityStore()
+        return state.identity_store
+
+    def _gate_identity_write(action: str,
+                             x_operator_role: Optional[str],
+                             x_operator_scopes: Optional[str],
+                             x_audit_reason: Optional[str]) -> str:
+        from agentbom_amd.mcp.authz import authorize_write
+
+        ok, blocked = authorize_write(
+            action=action, operator_role=x_operator_role or "",
+            operator_scopes=x_operator_scopes or "",
+            reason=x_audit_reason or "", required_scope="identity:write")
+        if not ok:
+            raise HTTPException(status_code=403, detail=blocked)
+        return (x_audit_reason or "").strip()
+
+    @app.post("/v1/identities", status_code=201, dependencies=[Depends(auth)])
+    def identity_issue(payload: dict,
+                       x_operator_role: Optional[str] = Header(default=None),
+                       x_operator_scopes: Optional[str] = Header(default=None),
+                       x_audit_reason: Optional[str] = Header(default=None)) -> dict:
+        reason = _gate_identity_write("identity.issue", x_operator_role,
+                                      x_operator_scopes, x_audit_reason)
+        ident, raw = _identity_store().issue(
+            str(payload.get("agent_name", "")),
+            scopes=list(payload.get("scopes") or []),
+            allowed_tools=list(payload.get("allowed_tools") or []),
+            ttl_hours=float(payload.get("ttl_hours", 24)),
+            actor="api", reason=reason)
+        return {"identity": ident.to_public_dict(), "token": raw,
+                "note": "token is shown exactly once"}
+
+    @app.get("/v1/identities", dependencies=[Depends(auth)])
+    def identity_list(live_only: bool = False) -> dict:
+        idents = _identity_store().list(live_only=live_only)
+        return {"total": len(idents),
+                "identities": [i.to_public_dict() for i in idents]}
+
+    @app.post("/v1/identities/verify", dependencies=[Depends(auth)])
+    def identity_verify(payload: dict) -> dict:
+        return _identity_store().verify(
+            str(payload.get("token", "")),
+            tool=payload.get("tool"),
+            source_ip=payload.get("source_ip"))
+
+    @app.get("/v1/identities/reviews/access", dependencies=[Depends(auth)])
+    def identity_access_review() -> dict:
+        return _identity_store().access_review()
+
+    @app.get("/v1/identities/audit", dependencies=[Depends(auth)])
+    def identity_audit(limit: int = 200) -> dict:
+        store = _identity_store()
+        entries = store.audit_entries()
+        return {"total": len(entries), "chain_valid": store.audit_chain_valid(),
+                "entries": entries[-limit:]}
+
+    @app.post("/v1/identities/nhi/discover", dependencies=[Depends(auth)])
+    def nhi_discover(payload: dict) -> dict:
+        """Discover NHIs from inline export payloads (offline connectors)."""
+        import json as _json
+        import tempfile
+
+        from agentbom_amd.identity.nhi import discover_entra_nhis, discover_okta_nhis
+
+        out: dict[str, Any] = {}
+        for provider, fn in (("okta", discover_okta_nhis),
+                             ("entra", discover_entra_nhis)):
+            export = payload.get(provider)
+            if export is None:
+                continue
+            with tempfile.NamedTemporaryFile("w", suffix=".json",
+                                             delete=False) as f:
+                _json.dump(export, f)
+                path = f.name
+            try:
+                out[provider] = fn(export_path=path, env={}).to_dict()
+            finally:
+                os.unlink(path)
+        return out or {"note": "provide 'okta' and/or 'entra' export payloads"}
+
+    @app.get("/v1/identities/{identity_id}", dependencies=[Depends(auth)])
+    def identity_get(identity_id: str) -> dict:
+        ident = _identity_store().get(identity_id)
+        if ident is None:
+            raise HTTPException(status_code=404, detail="identity not found")
+        return {"identity": ident.to_public_dict(),
+                "active_scopes": _identity_store().active_scopes(identity_id),
+                "jit_grants": [g.to_public_dict() for g in
+                               _identity_store().list_jit_grants(identity_id)]}
+
+    @app.post("/v1/identities/{identity_id}/rotate", dependencies=[Depends(auth)])
+    def identity_rotate(identity_id: str,
+                        x_operator_role: Optional[str] = Header(default=None),
+                        x_operator_scopes: Optional[str] = Header(default=None),
+                        x_audit_reason: Optional[str] = Header(default=None)) -> dict:
+        reason = _gate_identity_write("identity.rotate", x_operator_role,
+                                      x_operator_scopes, x_audit_reason)
+        new, raw = _identity_store().rotate(identity_id, actor="api",
+                                            reason=reason)
+        if new is None:
+            raise HTTPException(status_code=404,
+                                detail="identity not found or not live")
+        return {"identity": new.to_public_dict(), "token": raw}
+
+    @app.delete("/v1/identities/{identity_id}", dependencies=[Depends(auth)])
+    def identity_revoke(identity_id: str,
+                        x_operator_role: Optional[str] = Header(default=None),
+                        x_operator_scopes: Optional[str] = Header(default=None),
+                        x_audit_reason: Optional[str] = Header(default=None)) -> dict:
+        reason = _gate_identity_write("identity.revoke", x_operator_role,
+                                      x_operator_scopes, x_audit_reason)
+        if not _identity_store().revoke(identity_id, actor="api", reason=reason):
+            raise HTTPException(status_code=404, detail="identity not found")
+        return {"revoked": True}
+
+    @app.post("/v1/identities/{identity_id}/jit", status_code=201,
+              dependencies=[Depends(auth)])
+    def identity_grant_jit(identity_id: str, payload: dict,
+                           x_operator_role: Optional[str] = Header(default=None),
+                           x_operator_scopes: Optional[str] = Header(default=None),
+                           x_audit_reason: Optional[str] = Header(default=None)) -> dict:
+        reason = _gate_identity_write("identity.grant_jit", x_operator_role,
+                                      x_operator_scopes, x_audit_reason)
+        g = _identity_store().grant_jit(
+            identity_id, list(payload.get("scopes") or []),
+            reason=reason, granted_by="api",
+            ttl_minutes=float(payload.get("ttl_minutes", 60)))
+        if g is None:
+            raise HTTPException(status_code=404, detail="identity not found")
+        return {"grant": g.to_public_dict()}
+
+    @app.delete("/v1/identities/jit/{grant_id}", dependencies=[Depends(auth)])
+    def identity_revoke_jit(grant_id: str,
+                            x_operator_role: Optional[str] = Header(default=None),
+                            x_operator_scopes: Optional[str] = Header(default=None),
+                            x_audit_reason: Optional[str] = Header(default=None)) -> dict:
+        reason = _gate_identity_write("identity.revoke_jit", x_operator_role,
+                                      x_operator_scopes, x_audit_reason)
+        if not _identity_store().revoke_jit(grant_id, actor="api", reason=reason):
+            raise HTTPException(status_code=404, detail="grant not found")
+        return {"revoked": True}
 
     return app
 

@@ -1,0 +1,98 @@
+"""REST identity-lifecycle route tests (/v1/identities)."""
+
+import pytest
+from fastapi.testclient import TestClient
+
+from agentbom_amd.api.server import create_app
+
+ADMIN_HEADERS = {
+    "X-Operator-Role": "admin",
+    "X-Operator-Scopes": "identity:write",
+    "X-Audit-Reason": "integration test issue",
+}
+
+
+@pytest.fixture()
+def client():
+    return TestClient(create_app())
+
+
+def _issue(client, name="bot", **kw):
+    resp = client.post("/v1/identities",
+                       json={"agent_name": name, **kw}, headers=ADMIN_HEADERS)
+    assert resp.status_code == 201, resp.text
+    return resp.json()
+
+
+class TestIdentityRoutes:
+    def test_issue_list_get(self, client):
+        out = _issue(client, scopes=["scan:read"])
+        assert out["token"].startswith("abi_")
+        iid = out["identity"]["identity_id"]
+        listed = client.get("/v1/identities").json()
+        assert listed["total"] == 1
+        one = client.get(f"/v1/identities/{iid}").json()
+        assert one["identity"]["agent_name"] == "bot"
+        assert one["active_scopes"] == ["scan:read"]
+
+    def test_writes_require_gate(self, client):
+        resp = client.post("/v1/identities", json={"agent_name": "x"})
+        assert resp.status_code == 403
+        assert resp.json()["detail"]["status"] == "blocked"
+        # wrong scope
+        resp = client.post("/v1/identities", json={"agent_name": "x"},
+                           headers={**ADMIN_HEADERS,
+                                    "X-Operator-Scopes": "scan:read"})
+        assert resp.status_code == 403
+
+    def test_verify_route(self, client):
+        out = _issue(client, allowed_tools=["read_file"])
+        ok = client.post("/v1/identities/verify",
+                         json={"token": out["token"], "tool": "read_file"}).json()
+        assert ok["valid"]
+        bad = client.post("/v1/identities/verify",
+                          json={"token": out["token"], "tool": "exec"}).json()
+        assert not bad["valid"]
+
+    def test_rotate_and_revoke(self, client):
+        out = _issue(client)
+        iid = out["identity"]["identity_id"]
+        rotated = client.post(f"/v1/identities/{iid}/rotate",
+                              headers=ADMIN_HEADERS)
+        assert rotated.status_code == 200
+        assert rotated.json()["identity"]["rotated_from"] == iid
+        resp = client.delete(f"/v1/identities/{iid}", headers=ADMIN_HEADERS)
+        assert resp.status_code == 200
+        # rotate a revoked identity -> 404
+        resp = client.post(f"/v1/identities/{iid}/rotate", headers=ADMIN_HEADERS)
+        assert resp.status_code == 404
+
+    def test_jit_routes(self, client):
+        iid = _issue(client)["identity"]["identity_id"]
+        g = client.post(f"/v1/identities/{iid}/jit",
+                        json={"scopes": ["shield:write"], "ttl_minutes": 5},
+                        headers=ADMIN_HEADERS)
+        assert g.status_code == 201
+        grant_id = g.json()["grant"]["grant_id"]
+        one = client.get(f"/v1/identities/{iid}").json()
+        assert "shield:write" in one["active_scopes"]
+        resp = client.delete(f"/v1/identities/jit/{grant_id}",
+                             headers=ADMIN_HEADERS)
+        assert resp.status_code == 200
+
+    def test_access_review_and_audit(self, client):
+        _issue(client, scopes=["*"])
+        review = client.get("/v1/identities/reviews/access").json()
+        assert review["live_identities"] == 1
+        assert review["wildcard_or_unscoped"]
+        audit = client.get("/v1/identities/audit").json()
+        assert audit["chain_valid"] and audit["total"] >= 1
+
+    def test_nhi_discover_inline_export(self, client):
+        payload = {"okta": {"apps": [
+            {"id": "a1", "label": "svc", "signOnMode": "OPENID_CONNECT",
+             "settings": {"oauthClient": {"application_type": "service"}}}],
+            "api_tokens": []}}
+        out = client.post("/v1/identities/nhi/discover", json=payload).json()
+        assert out["okta"]["status"] == "ok"
+        assert len(out["okta"]["identities"]) == 1
