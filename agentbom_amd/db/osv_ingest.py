@@ -34,17 +34,28 @@ _ECO_MAP = {
 }
 
 
+def _dict(value: Any) -> dict:
+    """Coerce an untrusted field to a dict (fail-soft ingestion contract)."""
+    return value if isinstance(value, dict) else {}
+
+
+def _list(value: Any) -> list:
+    return value if isinstance(value, list) else []
+
+
 def _severity_from_osv(record: dict[str, Any]) -> tuple[Severity, Optional[float]]:
     cvss = None
-    for sev in record.get("severity", []) or []:
-        if sev.get("type", "").startswith("CVSS"):
+    for sev in _list(record.get("severity")):
+        if not isinstance(sev, dict):
+            continue
+        if str(sev.get("type", "")).startswith("CVSS"):
             score = sev.get("score", "")
             # score may be a vector string or a number
             try:
                 cvss = float(score)
             except (TypeError, ValueError):
                 cvss = _cvss_base_from_vector(str(score))
-    db_sev = (record.get("database_specific", {}) or {}).get("severity", "")
+    db_sev = _dict(record.get("database_specific")).get("severity", "")
     label = str(db_sev).lower()
     if label in ("critical", "high", "medium", "low"):
         return Severity(label), cvss
@@ -91,27 +102,30 @@ def _cvss_base_from_vector(vector: str) -> Optional[float]:
 
 def parse_osv_record(record: dict[str, Any]) -> list[AdvisoryWindow]:
     """One OSV advisory -> windows (one per affected range segment)."""
-    vuln_id = record.get("id", "")
+    vuln_id = str(record.get("id") or "")
     if not vuln_id:
         return []
-    summary = record.get("summary") or record.get("details", "")[:200]
+    details = record.get("details")
+    summary = str(record.get("summary")
+                  or (details if isinstance(details, str) else ""))[:200]
     severity, cvss = _severity_from_osv(record)
-    aliases = tuple(record.get("aliases", []) or [])
-    cwes = tuple(
-        c for c in (record.get("database_specific", {}) or {}).get("cwe_ids", []) or []
-    )
+    aliases = tuple(str(a) for a in _list(record.get("aliases")))
+    cwes = tuple(str(c) for c in
+                 _list(_dict(record.get("database_specific")).get("cwe_ids")))
     out: list[AdvisoryWindow] = []
 
-    for affected in record.get("affected", []) or []:
-        pkg = affected.get("package", {}) or {}
+    for affected in _list(record.get("affected")):
+        if not isinstance(affected, dict):
+            continue
+        pkg = _dict(affected.get("package"))
         eco_raw = str(pkg.get("ecosystem", "")).split(":")[0].lower()
         eco = _ECO_MAP.get(eco_raw, eco_raw)
-        name = pkg.get("name", "")
+        name = str(pkg.get("name") or "")
         if not name:
             continue
-        aff_cwes = cwes or tuple(
-            c for c in (affected.get("database_specific", {}) or {}).get("cwes", []) or []
-        )
+        aff_cwes = cwes or tuple(str(c) for c in
+                                 _list(_dict(affected.get("database_specific"))
+                                       .get("cwes")))
 
         def mk(intro, fixed, last, unfixed=False):
             out.append(AdvisoryWindow(
@@ -123,25 +137,30 @@ def parse_osv_record(record: dict[str, Any]) -> list[AdvisoryWindow]:
             ))
 
         emitted = False
-        for rng in affected.get("ranges", []) or []:
+        for rng in _list(affected.get("ranges")):
+            if not isinstance(rng, dict):
+                continue
             if rng.get("type") == "GIT":
                 continue  # commit bounds: undecidable, fail closed
             intro: Optional[str] = None
             has_terminator = False
-            for event in rng.get("events", []) or []:
+            for event in _list(rng.get("events")):
+                if not isinstance(event, dict):
+                    continue
                 if "introduced" in event:
                     # a new introduced before a terminator closes the prior
                     # window as unfixed
                     if intro is not None and not has_terminator:
                         mk(intro, None, None, unfixed=True)
-                    intro = event["introduced"]
+                    intro = str(event["introduced"])
                     has_terminator = False
                 elif "fixed" in event:
-                    mk(intro if intro is not None else "0", event["fixed"], None)
+                    mk(intro if intro is not None else "0", str(event["fixed"]), None)
                     has_terminator = True
                     emitted = True
                 elif "last_affected" in event:
-                    mk(intro if intro is not None else "0", None, event["last_affected"])
+                    mk(intro if intro is not None else "0", None,
+                       str(event["last_affected"]))
                     has_terminator = True
                     emitted = True
             if intro is not None and not has_terminator:
@@ -149,8 +168,8 @@ def parse_osv_record(record: dict[str, Any]) -> list[AdvisoryWindow]:
                 emitted = True
 
         # explicit versions[] list: degenerate [v, v] windows
-        for v in affected.get("versions", []) or []:
-            mk(v, None, v)
+        for v in _list(affected.get("versions")):
+            mk(str(v), None, str(v))
             emitted = True
 
         if not emitted:

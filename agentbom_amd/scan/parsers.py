@@ -80,7 +80,12 @@ def parse_yarn_lock(text: str, path: str) -> list[Package]:
 def parse_pnpm_lock(text: str, path: str) -> list[Package]:
     import yaml
 
-    data = yaml.safe_load(text) or {}
+    try:
+        data = yaml.safe_load(text) or {}
+    except yaml.YAMLError:
+        return []
+    if not isinstance(data, dict):
+        return []
     out: dict[tuple, Package] = {}
     for key in (data.get("packages") or {}):
         # "/name@version(peer)" or "/@scope/name@version"
@@ -185,15 +190,21 @@ def parse_pyproject_toml(text: str, path: str) -> list[Package]:
 def parse_conda_env(text: str, path: str) -> list[Package]:
     import yaml
 
-    data = yaml.safe_load(text) or {}
+    try:
+        data = yaml.safe_load(text) or {}
+    except yaml.YAMLError:
+        return []
+    if not isinstance(data, dict):
+        return []
     out = []
-    for dep in data.get("dependencies", []):
+    deps = data.get("dependencies")
+    for dep in (deps if isinstance(deps, list) else []):
         if isinstance(dep, str):
             parts = dep.split("=")
             if len(parts) >= 2:
                 out.append(_pkg(parts[0], parts[1], "conda"))
         elif isinstance(dep, dict) and "pip" in dep:
-            for pip_dep in dep["pip"]:
+            for pip_dep in (dep["pip"] if isinstance(dep["pip"], list) else []):
                 m = _REQ_LINE.match(pip_dep)
                 if m:
                     out.append(_pkg(m.group(1), m.group(2), "pypi"))
@@ -359,7 +370,12 @@ def parse_mix_lock(text: str, path: str) -> list[Package]:
 def parse_pubspec_lock(text: str, path: str) -> list[Package]:
     import yaml
 
-    data = yaml.safe_load(text) or {}
+    try:
+        data = yaml.safe_load(text) or {}
+    except yaml.YAMLError:
+        return []
+    if not isinstance(data, dict):
+        return []
     out = []
     for name, meta in (data.get("packages") or {}).items():
         version = meta.get("version", "") if isinstance(meta, dict) else ""
