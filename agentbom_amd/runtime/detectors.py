@@ -279,10 +279,129 @@ class VectorDBInjectionDetector(BaseDetector):
         return []
 
 
+_HARM_PATTERNS = [
+    (re.compile(r"(?i)\b(kill yourself|kys)\b"), "self-harm-content", "critical"),
+    (re.compile(r"(?i)\b(all|those) (women|men|jews|muslims|christians|"
+                r"immigrants|blacks|whites|asians) (are|should)\b"),
+     "demographic-generalization", "warning"),
+    (re.compile(r"(?i)\byou (stupid|worthless|pathetic) (idiot|moron|fool)\b"),
+     "toxic-address", "warning"),
+]
+
+_FABRICATION_PATTERNS = [
+    re.compile(r"(?i)as (officially )?(confirmed|verified) by (the )?"
+               r"(CDC|WHO|NASA|FBI|government)"),
+    re.compile(r"(?i)100% (guaranteed|certain|proven) (cure|safe|effective)"),
+    re.compile(r"(?i)studies (show|prove) that .{0,60}(always|never)"),
+]
+
+
+class ContentSafetyDetector(BaseDetector):
+    """Toxicity / bias / fabrication patterns in tool RESPONSES.
+
+    Pattern-class parity with the reference's bias/toxicity/hallucination
+    detectors: responses carrying harmful address, sweeping demographic
+    claims, or authority-fabrication markers are flagged (warn, not block —
+    these are review signals, not exfiltration)."""
+
+    name = "content_safety"
+
+    def inspect(self, frame: dict[str, Any]) -> list[DetectorAlert]:
+        result = frame.get("result")
+        if result is None:
+            return []
+        text = str(result)
+        alerts = []
+        for pattern, kind, severity in _HARM_PATTERNS:
+            if pattern.search(text):
+                alerts.append(self._alert(severity, f"harmful content ({kind}) "
+                                          "in tool response", kind=kind))
+        for pattern in _FABRICATION_PATTERNS:
+            m = pattern.search(text)
+            if m:
+                alerts.append(self._alert(
+                    "warning", "authority-fabrication marker in tool response",
+                    match=m.group(0)[:60]))
+                break
+        return alerts
+
+
+class CrossAgentCorrelator(BaseDetector):
+    """The same credential fingerprint appearing from DIFFERENT sessions —
+    a leaked secret being replayed by another agent."""
+
+    name = "cross_agent_correlator"
+
+    def __init__(self) -> None:
+        super().__init__()
+        self._cred_sessions: dict[str, set[str]] = defaultdict(set)
+
+    def inspect(self, frame: dict[str, Any]) -> list[DetectorAlert]:
+        session = str(frame.get("_session") or frame.get("session") or "default")
+        text = _frame_text(frame)
+        alerts = []
+        for pattern, kind in _CRED_PATTERNS:
+            m = pattern.search(text)
+            if not m:
+                continue
+            fp = hashlib.sha256(m.group(0).encode()).hexdigest()[:16]
+            seen = self._cred_sessions[fp]
+            if seen and session not in seen:
+                alerts.append(self._alert(
+                    "critical",
+                    f"credential ({kind}) observed from multiple sessions "
+                    "(cross-agent reuse)",
+                    action="block", kind=kind, fingerprint=fp,
+                    sessions=len(seen) + 1))
+            seen.add(session)
+        return alerts
+
+
+_DATA_URI_RE = re.compile(r"data:image/[a-z+]+;base64,([A-Za-z0-9+/=]{64,})")
+_BASE64_BLOB_RE = re.compile(r"\b[A-Za-z0-9+/]{200,}={0,2}\b")
+
+
+class VisualLeakDetector(BaseDetector):
+    """Secret material smuggled inside image payloads / large base64 blobs.
+
+    Decodes embedded base64 (bounded) and re-runs the credential patterns
+    over the decoded bytes — an exfil channel plain-text scanners miss."""
+
+    name = "visual_leak"
+
+    MAX_DECODE = 256 * 1024
+
+    def inspect(self, frame: dict[str, Any]) -> list[DetectorAlert]:
+        import base64
+
+        text = _frame_text(frame)
+        blobs = _DATA_URI_RE.findall(text) or _BASE64_BLOB_RE.findall(text)
+        alerts = []
+        for blob in blobs[:8]:
+            try:
+                decoded = base64.b64decode(blob[: self.MAX_DECODE],
+                                           validate=False)
+            except (ValueError, TypeError):
+                continue
+            try:
+                decoded_text = decoded.decode("utf-8", errors="ignore")
+            except Exception:  # noqa: BLE001
+                continue
+            for pattern, kind in _CRED_PATTERNS:
+                if pattern.search(decoded_text):
+                    alerts.append(self._alert(
+                        "critical",
+                        f"credential ({kind}) hidden inside base64/image payload",
+                        action="block", kind=kind, blob_bytes=len(decoded)))
+                    break
+        return alerts
+
+
 DEFAULT_DETECTORS = (
     ToolDriftDetector, ArgumentAnalyzer, CredentialLeakDetector, ResponseInspector,
     CloakingDetector, RateLimitTracker, SequenceAnalyzer, ReplayDetector,
-    VectorDBInjectionDetector,
+    VectorDBInjectionDetector, ContentSafetyDetector, CrossAgentCorrelator,
+    VisualLeakDetector,
 )
 
 
