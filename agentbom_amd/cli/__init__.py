@@ -127,6 +127,13 @@ def _scan_impl(
         exit_zero=exit_zero, fail_on_kev=fail_on_kev,
         use_gpu=False if no_gpu else None,
     )
+    # project-level defaults + suppression file (.agent-bom.yaml / -ignore)
+    from agentbom_amd.utils.project_config import apply_to_scan_options, load_project_config
+
+    project_cfg = load_project_config(".")
+    apply_to_scan_options(project_cfg, options)
+    for warning in project_cfg.warnings:
+        click.echo(f"warning: {warning}", err=True)
     if demo:
         report = run_demo_scan(options)
     else:

@@ -1,0 +1,72 @@
+"""Project config (.agent-bom.yaml) + ignore file tests."""
+
+from agentbom_amd.scan.orchestrator import ScanOptions
+from agentbom_amd.utils.project_config import (
+    apply_to_scan_options,
+    load_project_config,
+    parse_ignore_line,
+)
+
+
+class TestIgnoreFile:
+    def test_parse_variants(self):
+        r = parse_ignore_line("CVE-2023-4863")
+        assert r.vuln_id == "CVE-2023-4863" and not r.expired
+        r = parse_ignore_line(
+            "CVE-2021-23337 until=2099-12-31 reason=accepted risk, sandboxed")
+        assert r.until == "2099-12-31"
+        assert r.reason == "accepted risk, sandboxed"
+        assert parse_ignore_line("# comment only") is None
+        assert parse_ignore_line("") is None
+
+    def test_expired_rules_fail_open(self, tmp_path):
+        (tmp_path / ".agent-bom-ignore").write_text(
+            "CVE-2020-14343 until=2020-01-01 reason=long gone\n"
+            "CVE-2023-4863 reason=current\n")
+        cfg = load_project_config(tmp_path)
+        assert cfg.active_ignore_ids == frozenset({"CVE-2023-4863"})
+        assert any("expired" in w for w in cfg.warnings)
+
+    def test_unparseable_expiry_is_expired(self):
+        assert parse_ignore_line("CVE-1 until=not-a-date").expired
+
+
+class TestYamlConfig:
+    def test_settings_overlay(self, tmp_path):
+        (tmp_path / ".agent-bom.yaml").write_text(
+            "offline: true\nfail_on_severity: high\nfail_on_kev: true\n"
+            "blast_radius_depth: 3\nignore:\n  - CVE-2024-0001\n"
+            "unknown_key: 1\n")
+        cfg = load_project_config(tmp_path)
+        assert any("unknown key" in w for w in cfg.warnings)
+        opts = ScanOptions()
+        apply_to_scan_options(cfg, opts)
+        assert opts.offline and opts.fail_on_kev
+        assert opts.fail_on_severity == "high"
+        assert opts.blast_radius_depth == 3
+        assert "CVE-2024-0001" in opts.ignore_ids
+
+    def test_malformed_yaml_degrades(self, tmp_path):
+        (tmp_path / ".agent-bom.yaml").write_text(": : :\n\t-")
+        cfg = load_project_config(tmp_path)
+        assert cfg.warnings and cfg.settings == {}
+
+    def test_missing_files(self, tmp_path):
+        cfg = load_project_config(tmp_path)
+        assert cfg.settings == {} and not cfg.ignore_rules
+
+
+class TestEndToEnd:
+    def test_ignore_suppresses_demo_finding(self, tmp_path, monkeypatch):
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        baseline = run_demo_scan()
+        assert any(br.vulnerability.id == "CVE-2020-14343"
+                   for br in baseline.blast_radii)
+        (tmp_path / ".agent-bom-ignore").write_text("CVE-2020-14343\n")
+        cfg = load_project_config(tmp_path)
+        opts = ScanOptions(demo=True)
+        apply_to_scan_options(cfg, opts)
+        report = run_demo_scan(opts)
+        assert not any(br.vulnerability.id == "CVE-2020-14343"
+                       for br in report.blast_radii)
