@@ -20,7 +20,7 @@ import uuid
 from datetime import datetime, timezone
 from typing import Any, Optional
 
-from fastapi import Depends, FastAPI, Header, HTTPException
+from fastapi import Depends, FastAPI, Header, HTTPException, Request
 from pydantic import BaseModel, Field
 
 from agentbom_amd import __version__
@@ -72,11 +72,45 @@ def create_app() -> FastAPI:
     app.state.abom = state
 
     api_key = os.environ.get("AGENT_BOM_API_KEY")
+    # RBAC: AGENT_BOM_API_KEYS="key1:admin,key2:operator,key3:viewer"
+    # (reference: api/middleware.py + rbac.py role model, collapsed to the
+    # three-tier contract: viewer=read, operator=read+scan, admin=all)
+    key_roles: dict[str, str] = {}
+    for part in (os.environ.get("AGENT_BOM_API_KEYS") or "").split(","):
+        if ":" in part:
+            k, _, role = part.strip().partition(":")
+            if k and role in ("admin", "operator", "viewer"):
+                key_roles[k] = role
+    if api_key and api_key not in key_roles:
+        key_roles[api_key] = "admin"  # legacy single-key = admin
 
-    def auth(x_api_key: Optional[str] = Header(default=None)) -> None:
-        if api_key and x_api_key != api_key:
+    _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet")
+
+    def _role_allows(role: str, method: str, path: str) -> bool:
+        if role == "admin":
+            return True
+        if method in ("GET", "HEAD", "OPTIONS"):
+            return True
+        if role == "operator":
+            # operators can launch scans and post audit evidence, but not
+            # touch identity/fleet/schedule write surfaces
+            return not any(path.startswith(p) for p in _WRITE_PREFIXES)
+        return False  # viewer: reads only
+
+    def auth(request: Request,
+             x_api_key: Optional[str] = Header(default=None)) -> None:
+        if not key_roles:
+            return  # auth disabled (no keys configured)
+        role = key_roles.get(x_api_key or "")
+        if role is None:
             state.metrics["auth_failures_total"] += 1
             raise HTTPException(status_code=401, detail="invalid or missing API key")
+        if not _role_allows(role, request.method, request.url.path):
+            state.metrics["auth_failures_total"] += 1
+            raise HTTPException(
+                status_code=403,
+                detail=f"role {role!r} may not {request.method} {request.url.path}")
+        request.state.role = role
 
     # ── health + metrics ───────────────────────────────────────────────────
 

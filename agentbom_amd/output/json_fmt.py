@@ -661,4 +661,5 @@ def to_json(report: AIBOMReport) -> dict[str, Any]:
             ),
         },
         "remediation_plan": _build_remediation_json(report),
+        "scan_performance": report.scan_performance_data,
     }

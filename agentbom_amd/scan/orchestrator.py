@@ -167,9 +167,19 @@ def scan_agents(
     options: Optional[ScanOptions] = None,
 ) -> AIBOMReport:
     """Match every package across the estate and build blast radii."""
+    import time as _time
+
     options = options or ScanOptions()
     warnings: list[str] = []
     set_scan_warning_sink(warnings.append)
+    perf: dict[str, float] = {}
+    _t0 = _time.perf_counter()
+
+    def _mark(stage: str) -> None:
+        nonlocal _t0
+        now = _time.perf_counter()
+        perf[f"{stage}_ms"] = round((now - _t0) * 1000, 3)
+        _t0 = now
 
     # ── collect + dedup packages across the estate ─────────────────────────
     pkg_refs: dict[tuple[str, str, str], list[Package]] = {}
@@ -185,9 +195,12 @@ def scan_agents(
 
     unique = list(pkg_refs.keys())
 
+    _mark("dedup")
+
     # malicious screening (fails closed)
     all_pkgs = [p for plist in pkg_refs.values() for p in plist]
     flag_malicious_packages(all_pkgs)
+    _mark("malicious_screen")
 
     # ── matching ───────────────────────────────────────────────────────────
     arena = build_arena(list(advisory_windows), include_unfixed=options.include_unfixed)
@@ -199,7 +212,9 @@ def scan_agents(
             use_gpu = torch.cuda.is_available() and len(unique) >= 10_000
         except Exception:
             use_gpu = False
+    _mark("arena_build")
     pairs = _match_packages(unique, arena, use_gpu)
+    _mark("match")
 
     # ── attach Vulnerability objects to packages ───────────────────────────
     # group matched windows by (package, vuln_id): several windows of one
@@ -305,14 +320,28 @@ def scan_agents(
             br.calculate_risk_score()
             blast_radii.append(br)
 
+    _mark("blast_radius")
     expand_blast_radius_hops(blast_radii, agents, max_depth=options.blast_radius_depth)
     blast_radii.sort(key=lambda b: -b.risk_score)
+    _mark("hop_expand_rank")
 
     report = AIBOMReport(agents=agents, blast_radii=blast_radii)
     report.warnings = warnings
     report.scan_sources = ["agent_discovery"] if not options.demo else ["demo_inventory"]
     report.scan_run = ScanRun()
     report.findings = report.to_findings()
+    _mark("finding_fusion")
+    # counter-in-report contract (reference: scan_performance counters,
+    # package_scan.py:1324 / models.py:1243): per-stage ms + volume counters
+    report.scan_performance_data = {
+        "stages": perf,
+        "total_ms": round(sum(perf.values()), 3),
+        "unique_packages": len(unique),
+        "advisory_windows": len(arena.windows),
+        "match_pairs": len(pairs),
+        "blast_radii": len(blast_radii),
+        "match_path": "gpu" if use_gpu else "cpu",
+    }
     set_scan_warning_sink(None)
     return report
 

@@ -96,3 +96,46 @@ class TestIdentityRoutes:
         out = client.post("/v1/identities/nhi/discover", json=payload).json()
         assert out["okta"]["status"] == "ok"
         assert len(out["okta"]["identities"]) == 1
+
+
+class TestRbac:
+    @pytest.fixture()
+    def rbac_client(self, monkeypatch):
+        monkeypatch.setenv("AGENT_BOM_API_KEYS",
+                           "adm-key:admin,op-key:operator,view-key:viewer")
+        monkeypatch.delenv("AGENT_BOM_API_KEY", raising=False)
+        return TestClient(create_app())
+
+    def test_viewer_reads_only(self, rbac_client):
+        assert rbac_client.get(
+            "/v1/fleet", headers={"X-API-Key": "view-key"}).status_code == 200
+        resp = rbac_client.post("/v1/scan", json={"demo": True},
+                                headers={"X-API-Key": "view-key"})
+        assert resp.status_code == 403
+
+    def test_operator_can_scan_not_identity(self, rbac_client):
+        resp = rbac_client.post("/v1/scan", json={"demo": True},
+                                headers={"X-API-Key": "op-key"})
+        assert resp.status_code == 201
+        resp = rbac_client.post(
+            "/v1/identities", json={"agent_name": "x"},
+            headers={"X-API-Key": "op-key", **ADMIN_HEADERS})
+        assert resp.status_code == 403
+
+    def test_admin_all_access(self, rbac_client):
+        resp = rbac_client.post(
+            "/v1/identities", json={"agent_name": "x"},
+            headers={"X-API-Key": "adm-key", **ADMIN_HEADERS})
+        assert resp.status_code == 201
+
+    def test_unknown_key_rejected(self, rbac_client):
+        assert rbac_client.get(
+            "/v1/fleet", headers={"X-API-Key": "nope"}).status_code == 401
+
+    def test_legacy_single_key_is_admin(self, monkeypatch):
+        monkeypatch.setenv("AGENT_BOM_API_KEY", "legacy")
+        monkeypatch.delenv("AGENT_BOM_API_KEYS", raising=False)
+        c = TestClient(create_app())
+        assert c.get("/v1/fleet",
+                     headers={"X-API-Key": "legacy"}).status_code == 200
+        assert c.get("/v1/fleet").status_code == 401

@@ -122,3 +122,20 @@ def test_multi_hop_expansion_shared_server():
     assert any(t["name"] == "a2" for t in br.transitive_agents)
     assert "B_SECRET" in br.transitive_credentials
     assert br.transitive_risk_score == pytest.approx(round(br.risk_score * 0.7, 2))
+
+
+class TestScanPerformanceCounters:
+    def test_counters_in_report_and_json(self):
+        from agentbom_amd.output.json_fmt import to_json
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        report = run_demo_scan()
+        perf = report.scan_performance_data
+        assert perf["match_path"] in ("cpu", "gpu")
+        assert perf["unique_packages"] == 17
+        assert perf["blast_radii"] == len(report.blast_radii)
+        for stage in ("dedup", "arena_build", "match", "blast_radius",
+                      "hop_expand_rank", "finding_fusion"):
+            assert f"{stage}_ms" in perf["stages"]
+        doc = to_json(report)
+        assert doc["scan_performance"]["match_pairs"] == perf["match_pairs"]
