@@ -460,9 +460,18 @@ class AgentBomMcpServer:
             matches = [s for a in report.agents for s in a.mcp_servers
                        if name.lower() in s.name.lower()]
             if not matches:
+                from agentbom_amd.mcp.registry import load_registry
+
+                reg = load_registry()
+                hits = [dict(e) for key, e in reg["servers"].items()
+                        if name.lower() in key.lower()
+                        or name.lower() in str(e.get("name", "")).lower()]
+                if hits:
+                    return {"name": name, "found": True, "source": "registry",
+                            "entries": hits[:10]}
                 return {"name": name, "found": False,
-                        "note": "not in latest scan; run marketplace_check for "
-                                "pre-install trust signals"}
+                        "note": "not in latest scan or registry; run "
+                                "marketplace_check for trust signals"}
             from agentbom_amd.scan.risk import score_server_risk, server_risk_level
 
             out = []

@@ -298,6 +298,7 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
            "required": ["name"]})
     def marketplace_check(name: str, ecosystem: str = "npm") -> dict:
         from agentbom_amd.db.store import load_advisory_windows
+        from agentbom_amd.mcp.registry import check_blocklist, lookup_package
         from agentbom_amd.models import Package
         from agentbom_amd.scan.malicious import check_typosquat, flag_malicious_packages
         from agentbom_amd.utils.canonical_ids import normalize_package_name
@@ -309,11 +310,20 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
         advisories = [w.vuln_id for w in load_advisory_windows(offline=True)
                       if w.ecosystem.lower() == ecosystem.lower()
                       and normalize_package_name(w.package_name, w.ecosystem) == norm]
-        verdict = ("block" if pkg.is_malicious
-                   else "warn" if typo or advisories else "allow")
+        registry_entry = lookup_package(name)
+        blocked = check_blocklist(name)
+        high_risk = bool(registry_entry
+                         and registry_entry.get("risk_level") in ("high", "critical"))
+        verdict = ("block" if pkg.is_malicious or blocked
+                   else "warn" if typo or advisories or high_risk else "allow")
         return {"package": name, "ecosystem": ecosystem, "verdict": verdict,
                 "is_malicious": pkg.is_malicious,
                 "malicious_reason": pkg.malicious_reason,
+                "blocklist_hit": blocked,
+                "registry": ({k: registry_entry.get(k) for k in
+                              ("name", "risk_level", "verified", "category",
+                               "credential_env_vars", "risk_justification")}
+                             if registry_entry else None),
                 "typosquat_of": typo, "known_advisories": sorted(set(advisories))}
 
     # ── graph ─────────────────────────────────────────────────────────────

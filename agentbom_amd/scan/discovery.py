@@ -171,6 +171,13 @@ def _parse_servers_dict(servers: dict[str, Any], config_path: str) -> list[MCPSe
                 discovery_sources=["global_config"],
             )
         )
+    # pre-spawn enrichment from the known-server registry (verification,
+    # inferred tool/credential surface, blocklist fail-closed)
+    from agentbom_amd.mcp.registry import stamp_server_from_registry
+
+    for server in out:
+        if server.command:
+            stamp_server_from_registry(server)
     return out
 
 

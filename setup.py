@@ -18,6 +18,7 @@ setup(
     python_requires=">=3.10",
     packages=find_packages(include=["agentbom_amd*"]),
     include_package_data=True,
-    package_data={"agentbom_amd.ops": ["csrc/*", "_abom_gpu.so"]},
+    package_data={"agentbom_amd.ops": ["csrc/*", "_abom_gpu.so"],
+                  "agentbom_amd": ["data/*.json"]},
     entry_points={"console_scripts": ["agent-bom = agentbom_amd.cli:cli_main"]},
 )
