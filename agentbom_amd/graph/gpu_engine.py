@@ -95,6 +95,7 @@ class EstateEngine:
 
         self.agent_ids = torch.arange(estate.n_agents, dtype=torch.int32, device=dev)
         self._bfs_ws: dict = {}
+        self._match_stream = None  # side stream: match ∥ reach BFS overlap
 
         # per-node db-credential / db-tool indicator (for impact filtering)
         self.node_is_db_cred = torch.zeros(self.N, dtype=torch.uint8, device=dev)
@@ -303,10 +304,33 @@ class EstateEngine:
         distances (multi-GPU mode passes the shard's slice of the
         distributed BFS result)."""
         torch = self.torch
-        pkg_idx, win_idx = self.match()
-        n_findings = pkg_idx.numel()
+        if self.use_gpu:
+            # Overlap: the match kernel has no dependency on the reach BFS,
+            # so it runs on a side HIP stream while the BFS host loop (which
+            # syncs its own stream every level) drives the default stream.
+            from agentbom_amd.ops import native
 
-        dist = reach_dist if reach_dist is not None else self.dependency_reach()
+            if self._match_stream is None:
+                self._match_stream = torch.cuda.Stream(device=self.device)
+            side = self._match_stream
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                pending = native.match_launch(
+                    self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
+                    self.pkg_key_lo_sorted, self.pkg_flags_sorted,
+                    self.arena["group_keys"], self.arena["group_off"],
+                    self.arena["windows"])
+            dist = reach_dist if reach_dist is not None else self.dependency_reach()
+            torch.cuda.current_stream().wait_stream(side)
+            sp, sw = native.match_finalize(pending)
+            orig = self.pkg_perm[sp]
+            packed = (orig << 32) | sw
+            packed, _ = torch.sort(packed)
+            pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
+        else:
+            pkg_idx, win_idx = self.match()
+            dist = reach_dist if reach_dist is not None else self.dependency_reach()
+        n_findings = pkg_idx.numel()
 
         pkg_nodes = pkg_idx + self.estate.pkg_base
         counts = self.blast_counts(pkg_nodes)

@@ -111,13 +111,12 @@ def _stream() -> ctypes.c_void_p:
 # ── High-level ops (torch tensors on cuda device) ──────────────────────────
 
 
-def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_off,
-          windows: dict, capacity: Optional[int] = None):
-    """Bulk version-range match.  Returns sorted (pkg_idx u32, window_idx u32).
+def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
+                 group_off, windows: dict, capacity: Optional[int] = None) -> dict:
+    """Launch the match kernel on the CURRENT stream without syncing.
 
-    All tensors device-resident; ``windows`` carries intro/fixed/last hi+lo
-    (int64 bit-patterns) and flags (uint8).  Retries with a larger buffer if
-    capacity overflows.
+    Returns a pending handle for :func:`match_finalize`.  Used by the engine
+    to overlap the match with the reach BFS on a second HIP stream.
     """
     import torch
 
@@ -126,27 +125,50 @@ def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_of
     G = group_keys.numel()
     cap = capacity or max(1024, P // 4)
     dev = pkg_group_key.device
-    while True:
-        out_pairs = torch.empty(cap, dtype=torch.int64, device=dev)
-        out_count = torch.zeros(1, dtype=torch.int32, device=dev)
-        rc = lib.abom_match(
-            _ptr(pkg_group_key), _ptr(pkg_key_hi), _ptr(pkg_key_lo), _ptr(pkg_flags), P,
-            _ptr(group_keys), _ptr(group_off), G,
-            _ptr(windows["intro_hi"]), _ptr(windows["intro_lo"]),
-            _ptr(windows["fixed_hi"]), _ptr(windows["fixed_lo"]),
-            _ptr(windows["last_hi"]), _ptr(windows["last_lo"]),
-            _ptr(windows["flags"]),
-            _ptr(out_pairs), _ptr(out_count), cap, _stream(),
-        )
-        _check(rc, "abom_match")
-        n = int(out_count.item())
-        if n <= cap:
-            pairs = out_pairs[:n]
-            pairs, _ = torch.sort(pairs)
-            pkg_idx = (pairs >> 32).to(torch.int64)
-            win_idx = (pairs & 0xFFFFFFFF).to(torch.int64)
-            return pkg_idx, win_idx
-        cap = int(n * 1.2) + 1024
+    out_pairs = torch.empty(cap, dtype=torch.int64, device=dev)
+    out_count = torch.zeros(1, dtype=torch.int32, device=dev)
+    rc = lib.abom_match(
+        _ptr(pkg_group_key), _ptr(pkg_key_hi), _ptr(pkg_key_lo), _ptr(pkg_flags), P,
+        _ptr(group_keys), _ptr(group_off), G,
+        _ptr(windows["intro_hi"]), _ptr(windows["intro_lo"]),
+        _ptr(windows["fixed_hi"]), _ptr(windows["fixed_lo"]),
+        _ptr(windows["last_hi"]), _ptr(windows["last_lo"]),
+        _ptr(windows["flags"]),
+        _ptr(out_pairs), _ptr(out_count), cap, _stream(),
+    )
+    _check(rc, "abom_match")
+    return {"out_pairs": out_pairs, "out_count": out_count, "cap": cap,
+            "args": (pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags,
+                     group_keys, group_off, windows)}
+
+
+def match_finalize(pending: dict):
+    """Read the pending match count and produce sorted (pkg_idx, win_idx).
+
+    On capacity overflow (count > cap) the match is relaunched synchronously
+    with a grown buffer — rare, the launch heuristic is P//4.
+    """
+    import torch
+
+    n = int(pending["out_count"].item())
+    if n > pending["cap"]:
+        return match(*pending["args"], capacity=int(n * 1.2) + 1024)
+    pairs = pending["out_pairs"][:n]
+    pairs, _ = torch.sort(pairs)
+    return (pairs >> 32).to(torch.int64), (pairs & 0xFFFFFFFF).to(torch.int64)
+
+
+def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_off,
+          windows: dict, capacity: Optional[int] = None):
+    """Bulk version-range match.  Returns sorted (pkg_idx u32, window_idx u32).
+
+    All tensors device-resident; ``windows`` carries intro/fixed/last hi+lo
+    (int64 bit-patterns) and flags (uint8).  Retries with a larger buffer if
+    capacity overflows.
+    """
+    pending = match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags,
+                           group_keys, group_off, windows, capacity=capacity)
+    return match_finalize(pending)
 
 
 def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0xFFFFFFFF,
