@@ -474,6 +474,36 @@ def create_app() -> FastAPI:
         return {"framework": framework, "tagged_findings": len(rows),
                 "controls": dict(sorted(counts.items())), "findings": rows}
 
+    @app.get("/v1/graph/attack-flow", dependencies=[Depends(auth)])
+    def attack_flow(cve: Optional[str] = None, min_severity: Optional[str] = None,
+                    agent: Optional[str] = None) -> dict:
+        from agentbom_amd.output.flow_fmt import build_attack_flow
+
+        state.metrics["graph_queries_total"] += 1
+        return build_attack_flow(_latest_report(), cve=cve,
+                                 min_severity=min_severity, agent=agent)
+
+    @app.get("/v1/mesh", dependencies=[Depends(auth)])
+    def agent_mesh() -> dict:
+        from agentbom_amd.output.flow_fmt import build_agent_mesh
+
+        return build_agent_mesh(_latest_report())
+
+    @app.get("/v1/findings/delta", dependencies=[Depends(auth)])
+    def findings_delta(format: str = "ndjson") -> dict:
+        """Delta events (new/resolved/changed) since the previous call."""
+        from agentbom_amd.output.delta_stream import DeltaStreamer
+
+        if getattr(state, "delta_streamer", None) is None:
+            state.delta_streamer = DeltaStreamer(fmt="ndjson")
+        events = state.delta_streamer.emit(_latest_report())
+        if format == "ocsf":
+            from agentbom_amd.output.delta_stream import _to_ocsf_event
+
+            return {"watermark": state.delta_streamer.watermark,
+                    "events": [_to_ocsf_event(e) for e in events]}
+        return {"watermark": state.delta_streamer.watermark, "events": events}
+
     # ── identity lifecycle (reference: api/routes/identities.py) ───────────
 
     def _identity_store():
