@@ -16,6 +16,35 @@ import sys
 from typing import Any, Callable, Optional
 
 
+def resolve_mcp_tenant_id() -> str:
+    """Sanctioned tenant resolution for the MCP surface (reference:
+    mcp_tenant.py).  MCP clients pass no authenticated tenant identity, so
+    the launching operator decides via AGENT_BOM_MCP_TENANT_ID (preferred)
+    with AGENT_BOM_TENANT_ID as fallback; "default" otherwise."""
+    import os
+
+    return (os.environ.get("AGENT_BOM_MCP_TENANT_ID")
+            or os.environ.get("AGENT_BOM_TENANT_ID")
+            or "default").strip() or "default"
+
+
+def _check_strict_args(schema: dict, arguments: dict) -> Optional[dict]:
+    """Reject unknown and missing-required tool arguments (fail loud).
+
+    Returns a structured error dict, or None when the call is well-formed.
+    """
+    props = schema.get("properties") or {}
+    unknown = sorted(set(arguments) - set(props))
+    if unknown:
+        return {"error": "unknown arguments rejected (strict-args contract)",
+                "unknown": unknown, "accepted": sorted(props)}
+    missing = sorted(set(schema.get("required") or []) - set(arguments))
+    if missing:
+        return {"error": "missing required arguments",
+                "missing": missing, "accepted": sorted(props)}
+    return None
+
+
 class McpTool:
     def __init__(self, name: str, description: str, schema: dict, fn: Callable[..., Any]):
         self.name = name
@@ -517,7 +546,8 @@ class AgentBomMcpServer:
             return ok({
                 "protocolVersion": self.PROTOCOL_VERSION,
                 "capabilities": {"tools": {}, "resources": {}, "prompts": {}},
-                "serverInfo": {"name": "agent-bom", "version": "0.1.0"},
+                "serverInfo": {"name": "agent-bom", "version": "0.1.0",
+                               "tenant": resolve_mcp_tenant_id()},
             })
         if method == "notifications/initialized":
             return None
@@ -540,9 +570,18 @@ class AgentBomMcpServer:
                 return ok({"content": [{"type": "text", "text": json.dumps(
                     {"error": "rate limit exceeded", "retry_after_s": 60})}],
                     "isError": True})
+            # strict-args contract (reference: mcp_strict_args.py — silent
+            # arg drops turn typo'd gates into false-clean verdicts): unknown
+            # or missing-required arguments are rejected, never dropped.
+            arguments = params.get("arguments") or {}
+            strict_error = _check_strict_args(tool.schema, arguments)
+            if strict_error is not None:
+                return ok({"content": [{"type": "text",
+                                        "text": json.dumps(strict_error)}],
+                           "isError": True})
             t0 = _time.perf_counter()
             try:
-                result = tool.fn(**(params.get("arguments") or {}))
+                result = tool.fn(**arguments)
                 text = json.dumps(result, default=str)
                 if len(text) > self.MAX_RESPONSE_BYTES:
                     text = json.dumps({

@@ -827,6 +827,64 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
         return {"members": registry.list_members(),
                 "schedules": [s.to_dict() for s in scheduler.schedules.values()]}
 
+    @tool("mcp_auth_posture", "MCP server auth posture: remote no-auth, "
+                              "plaintext transport, static tokens vs OAuth.")
+    def mcp_auth_posture() -> dict:
+        from agentbom_amd.scan.auth_posture import assess_estate
+
+        report, _g = server._ensure_scan()
+        return assess_estate(report.agents)
+
+    @tool("self_posture", "agent-bom audits its OWN deployment hardening "
+                          "(auth, RBAC, tenancy, persistence, signing).")
+    def self_posture() -> dict:
+        from agentbom_amd.scan.self_posture import evaluate_self_posture
+
+        return evaluate_self_posture()
+
+    @tool("attest_server", "Sign a DSSE scan attestation for one scanned "
+                           "MCP server (key from AGENT_BOM_ATTESTATION_KEY).",
+          {"type": "object", "properties": {
+              "server_name": {"type": "string"},
+              "verdict": {"type": "string", "enum": ["pass", "warn", "block"],
+                          "default": "pass"}},
+           "required": ["server_name"]})
+    def attest_server(server_name: str, verdict: str = "pass") -> dict:
+        import os
+
+        from agentbom_amd.mcp.server import resolve_mcp_tenant_id
+        from agentbom_amd.utils.attestation import attest_scanned_server
+
+        key_hex = os.environ.get("AGENT_BOM_ATTESTATION_KEY", "")
+        if not key_hex:
+            return {"error": "AGENT_BOM_ATTESTATION_KEY not configured"}
+        report, _g = server._ensure_scan()
+        target = next((s for a in report.agents for s in a.mcp_servers
+                       if s.name == server_name), None)
+        if target is None:
+            return {"error": f"server {server_name!r} not in latest scan"}
+        return attest_scanned_server(
+            target, verdict, bytes.fromhex(key_hex),
+            key_id=os.environ.get("AGENT_BOM_ATTESTATION_KEY_ID", "operator"),
+            tenant_id=resolve_mcp_tenant_id())
+
+    @tool("verify_attestation", "Verify a DSSE scan attestation against the "
+                                "operator trust policy (pinned keys only).",
+          {"type": "object", "properties": {
+              "envelope": {"type": "object"},
+              "expected_tenant": {"type": "string", "default": ""}},
+           "required": ["envelope"]})
+    def verify_attestation_tool(envelope: dict, expected_tenant: str = "") -> dict:
+        import os
+
+        from agentbom_amd.utils.attestation import verify_attestation
+
+        key_hex = os.environ.get("AGENT_BOM_ATTESTATION_KEY", "")
+        key_id = os.environ.get("AGENT_BOM_ATTESTATION_KEY_ID", "operator")
+        policy = {"keys": ({key_id: key_hex} if key_hex else {}),
+                  "expected_tenant": expected_tenant or None}
+        return verify_attestation(envelope, policy)
+
     @tool("tool_metrics", "Per-tool call/latency/error counters for this "
                           "MCP session.")
     def tool_metrics() -> dict:
