@@ -108,7 +108,8 @@ def _scan_impl(
     output: Optional[str], fail_on_severity: str, fail_on_kev: bool,
     exit_zero: bool, blast_radius_depth: int, verbose: bool,
     include_unfixed: bool, no_gpu: bool,
-    sbom: Optional[str] = None, filesystem: Optional[str] = None,
+    sbom: Optional[str] = None, image: Optional[str] = None,
+    filesystem: Optional[str] = None,
     scan_secrets: bool = False, model_files: Optional[str] = None,
     code: Optional[str] = None, iac: Optional[str] = None,
     aws_inventory: Optional[str] = None, endpoint: bool = False,
@@ -144,6 +145,13 @@ def _scan_impl(
             from agentbom_amd.scan.sbom_ingest import sbom_to_agent
 
             agents.append(sbom_to_agent(sbom))
+        if image:
+            from agentbom_amd.scan.oci import oci_result_to_agent, scan_image
+
+            result = scan_image(image)
+            for w in result.warnings:
+                click.echo(f"warning: {w}", err=True)
+            agents.append(oci_result_to_agent(result))
         if filesystem:
             from agentbom_amd.models import Agent, AgentType, MCPServer, ServerSurface
             from agentbom_amd.scan.parsers import extract_packages
@@ -267,6 +275,8 @@ def _scan_options(f):
         click.option("--no-gpu", is_flag=True, help="Force the CPU match path."),
         click.option("--sbom", type=click.Path(exists=True), default=None,
                      help="Scan an existing CycloneDX/SPDX SBOM."),
+        click.option("--image", type=click.Path(exists=True), default=None,
+                     help="Scan a container image (docker-save tar or OCI layout)."),
         click.option("--filesystem", type=click.Path(exists=True), default=None,
                      help="Extract + scan packages from a directory tree."),
         click.option("--scan-secrets", is_flag=True,
