@@ -175,40 +175,20 @@ def _scan_impl(
 
         report = scan_agents(agents, load_advisory_windows(offline=offline), options)
 
-    # optional side scanners -> unified findings
+    # optional side scanners -> unified findings, run through the executor
+    # so each driver's DECLARED failure mode applies (scan/registry.py)
+    from agentbom_amd.scan.registry import run_scanner_driver
+
     if scan_secrets:
-        from agentbom_amd.scan.secrets import scan_paths, secret_hit_to_finding
-
-        root = filesystem or "."
-        hits = scan_paths(root)
-        report.findings.extend(secret_hit_to_finding(h) for h in hits)
-        report.ai_inventory_data = {"secrets": {"findings": [h.to_dict() for h in hits]}}
+        run_scanner_driver("secrets", report, filesystem or ".")
     if model_files:
-        from agentbom_amd.scan.model_scan import model_result_to_finding, scan_model_tree
-
-        results = scan_model_tree(model_files)
-        report.extra_data["model_files"] = [r.to_dict() for r in results]
-        report.findings.extend(
-            f for f in (model_result_to_finding(r) for r in results) if f is not None
-        )
+        run_scanner_driver("model_files", report, model_files)
     if aws_inventory:
-        from agentbom_amd.scan.cloud import cis_result_to_finding, scan_cloud_inventory
-
-        cis = scan_cloud_inventory(aws_inventory)
-        report.extra_data["cis_benchmark_data"] = [r.to_dict() for r in cis]
-        report.findings.extend(
-            f for f in (cis_result_to_finding(r) for r in cis) if f is not None
-        )
+        run_scanner_driver("cloud_cis", report, aws_inventory)
     if endpoint:
-        from agentbom_amd.scan.endpoint import collect_endpoint_inventory
-
-        report.extra_data["endpoint_inventory_data"] = collect_endpoint_inventory().to_dict()
+        run_scanner_driver("endpoint", report)
     if iac:
-        from agentbom_amd.scan.iac import iac_finding_to_finding, scan_iac_tree
-
-        iac_hits = scan_iac_tree(iac)
-        report.iac_findings_data = {"findings": [h.to_dict() for h in iac_hits]}
-        report.findings.extend(iac_finding_to_finding(h) for h in iac_hits)
+        run_scanner_driver("iac", report, iac)
     if code:
         from agentbom_amd.scan.ast_analysis import (
             apply_symbol_reachability,
