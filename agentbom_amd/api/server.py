@@ -144,9 +144,23 @@ def create_app() -> FastAPI:
         )
 
         job = state.jobs[job_id]
+
+        def _cancelled() -> bool:
+            """Cooperative cancel checked at phase boundaries."""
+            if job.get("cancel_requested"):
+                job["status"] = "cancelled"
+                job["steps"].append({"step": "cancelled", "at": _now()})
+                return True
+            return False
+
         try:
             job["status"] = "running"
+            import time as _time
+
+            job["started_monotonic"] = _time.monotonic()
             job["steps"].append({"step": "scan", "at": _now()})
+            if _cancelled():
+                return
             options = ScanOptions(demo=req.demo, offline=req.offline,
                                   blast_radius_depth=req.blast_radius_depth)
             if req.demo or not req.inventory:
@@ -157,6 +171,8 @@ def create_app() -> FastAPI:
                 agents = inventory_to_agents(req.inventory)
                 report = scan_agents(agents, load_advisory_windows(offline=req.offline), options)
             report.scan_id = job_id
+            if _cancelled():
+                return
             job["steps"].append({"step": "graph_build", "at": _now()})
             graph = build_unified_graph_from_report(report)
             apply_dependency_reachability_to_blast_radii(report, graph)
@@ -170,6 +186,8 @@ def create_app() -> FastAPI:
                 toxic_combination_to_finding(c) for c in combos
             ]
             report.toxic_combination_findings_data = [c.to_dict() for c in combos]
+            if _cancelled():
+                return
             job["steps"].append({"step": "graph_persist", "at": _now()})
             snapshot_id = None
             if state.graph_store is not None:
@@ -212,8 +230,47 @@ def create_app() -> FastAPI:
         t.join(timeout=120)  # scans are fast; keep the API synchronous-ish
         return {"job_id": job_id, "status": state.jobs[job_id]["status"]}
 
+    def reap_stuck_jobs(max_age_s: float = 600.0) -> list[str]:
+        """Mark running/pending jobs past the deadline as failed (reaper).
+
+        Reference: api/server.py stuck-job reaper — a crashed worker thread
+        must not leave a job 'running' forever.  Called lazily on job reads
+        and exposed for schedulers.
+        """
+        import time as _time
+
+        now = _time.monotonic()
+        reaped = []
+        with state.lock:
+            for job_id, job in state.jobs.items():
+                if job["status"] not in ("running", "pending"):
+                    continue
+                started = job.get("started_monotonic")
+                if started is not None and now - started > max_age_s:
+                    job["status"] = "failed"
+                    job["error"] = f"reaped: exceeded max runtime {max_age_s:.0f}s"
+                    job["steps"].append({"step": "reaped", "at": _now()})
+                    reaped.append(job_id)
+        return reaped
+
+    app.state.reap_stuck_jobs = reap_stuck_jobs
+
+    @app.post("/v1/scan/{job_id}/cancel", dependencies=[Depends(auth)])
+    def cancel_scan(job_id: str) -> dict:
+        job = state.jobs.get(job_id)
+        if not job:
+            raise HTTPException(status_code=404, detail="scan job not found")
+        if job["status"] in ("done", "failed", "cancelled"):
+            return {"job_id": job_id, "status": job["status"],
+                    "note": "job already finished"}
+        job["cancel_requested"] = True
+        return {"job_id": job_id, "status": job["status"],
+                "cancel_requested": True}
+
     @app.get("/v1/scan/{job_id}", dependencies=[Depends(auth)])
     def get_scan(job_id: str) -> dict:
+        reap_stuck_jobs(float(os.environ.get("AGENT_BOM_JOB_MAX_RUNTIME_S",
+                                             "600")))
         job = state.jobs.get(job_id)
         if not job:
             raise HTTPException(status_code=404, detail="scan job not found")

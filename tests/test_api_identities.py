@@ -139,3 +139,58 @@ class TestRbac:
         assert c.get("/v1/fleet",
                      headers={"X-API-Key": "legacy"}).status_code == 200
         assert c.get("/v1/fleet").status_code == 401
+
+
+class TestJobLifecycle:
+    def test_cancel_finished_job_is_noop(self, client):
+        job = client.post("/v1/scan", json={"demo": True}).json()
+        import time
+
+        for _ in range(100):
+            if client.get(f"/v1/scan/{job['job_id']}").json()["status"] == "done":
+                break
+            time.sleep(0.05)
+        out = client.post(f"/v1/scan/{job['job_id']}/cancel").json()
+        assert out["status"] == "done" and "already finished" in out["note"]
+
+    def test_cancel_unknown_404(self, client):
+        assert client.post("/v1/scan/nope/cancel").status_code == 404
+
+    def test_stuck_job_reaped(self, client):
+        app = client.app
+        state = app.state.abom
+        # forge a stuck running job (worker died without updating status)
+        import time
+
+        state.jobs["stuck-1"] = {"id": "stuck-1", "status": "running",
+                                 "submitted_at": "x", "steps": [],
+                                 "result": None, "error": None,
+                                 "started_monotonic": time.monotonic() - 10_000}
+        reaped = app.state.reap_stuck_jobs(600.0)
+        assert "stuck-1" in reaped
+        job = client.get("/v1/scan/stuck-1").json()
+        assert job["status"] == "failed" and "reaped" in job["error"]
+
+    def test_fresh_jobs_not_reaped(self, client):
+        app = client.app
+        state = app.state.abom
+        import time
+
+        state.jobs["fresh-1"] = {"id": "fresh-1", "status": "running",
+                                 "submitted_at": "x", "steps": [],
+                                 "result": None, "error": None,
+                                 "started_monotonic": time.monotonic()}
+        assert app.state.reap_stuck_jobs(600.0) == []
+        assert state.jobs["fresh-1"]["status"] == "running"
+
+    def test_cooperative_cancel_flag(self):
+        """A pending job with cancel_requested set reports the request and
+        the worker exits 'cancelled' at its first phase boundary."""
+        app = create_app()
+        c = TestClient(app)
+        state = app.state.abom
+        state.jobs["c1"] = {"id": "c1", "status": "pending",
+                            "submitted_at": "x", "steps": [], "result": None,
+                            "error": None, "cancel_requested": True}
+        resp = c.post("/v1/scan/c1/cancel").json()
+        assert resp["cancel_requested"] is True
