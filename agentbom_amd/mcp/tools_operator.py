@@ -827,6 +827,22 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
         return {"members": registry.list_members(),
                 "schedules": [s.to_dict() for s in scheduler.schedules.values()]}
 
+    @tool("effective_permissions", "Per-agent effective tool permissions "
+                                   "with grant provenance + least-privilege "
+                                   "recommendations.")
+    def effective_permissions() -> dict:
+        from agentbom_amd.graph.effective_permissions import (
+            compute_effective_permissions,
+            least_privilege_recommendations,
+        )
+        from agentbom_amd.graph.nhi_overlay import apply_issued_identity_overlay
+
+        _r, g = server._ensure_scan()
+        apply_issued_identity_overlay(g, server.identity_store)
+        perms = compute_effective_permissions(g)
+        perms["recommendations"] = least_privilege_recommendations(g)
+        return perms
+
     @tool("mcp_auth_posture", "MCP server auth posture: remote no-auth, "
                               "plaintext transport, static tokens vs OAuth.")
     def mcp_auth_posture() -> dict:

@@ -86,3 +86,55 @@ class TestEndpointOverlay:
     def test_noop(self, graph):
         assert apply_endpoint_overlay(graph, {}) == {
             "nodes_added": 0, "edges_added": 0}
+
+
+class TestEffectivePermissions:
+    def test_direct_and_identity_paths(self, graph, report):
+        from agentbom_amd.graph.effective_permissions import (
+            compute_effective_permissions,
+        )
+        from agentbom_amd.graph.nhi_overlay import apply_issued_identity_overlay
+        from agentbom_amd.identity import AgentIdentityStore
+
+        store = AgentIdentityStore()
+        store.issue(report.agents[0].name, scopes=["*"])
+        apply_issued_identity_overlay(graph, store)
+
+        perms = compute_effective_permissions(graph)
+        assert perms["summary"]["total_agents"] == len(report.agents)
+        first = next(a for a in perms["agents"]
+                     if a["agent"] == report.agents[0].name)
+        assert first["effective_tools"] > 0
+        kinds = {p["kind"] for g in first["grants"] for p in g["paths"]}
+        assert "direct" in kinds
+        assert perms["summary"]["wildcard_identities"]
+
+    def test_evidence_informed_unused_grants(self, graph, report):
+        from agentbom_amd.graph.effective_permissions import (
+            compute_effective_permissions,
+            least_privilege_recommendations,
+        )
+
+        called_tool = next(t.name for b in report.blast_radii
+                           for t in b.exposed_tools)
+        apply_evidence_overlay(graph, [
+            {"method": "tools/call", "tool": called_tool, "ts": 1.0}])
+        perms = compute_effective_permissions(graph)
+        assert perms["evidence_informed"]
+        # every granted tool except the called one is an unused grant
+        assert perms["summary"]["total_unused_grants"] > 0
+        for agent in perms["agents"]:
+            assert called_tool not in agent["unused_grants"] or not any(
+                g["tool"] == called_tool and g["observed_calls"] > 0
+                for g in agent["grants"])
+        recs = least_privilege_recommendations(graph)
+        assert any(r["kind"] == "revoke_unused_grant" for r in recs)
+
+    def test_no_evidence_means_no_unused_claims(self, graph):
+        from agentbom_amd.graph.effective_permissions import (
+            compute_effective_permissions,
+        )
+
+        perms = compute_effective_permissions(graph)
+        assert not perms["evidence_informed"]
+        assert perms["summary"]["total_unused_grants"] is None
