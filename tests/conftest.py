@@ -1,4 +1,5 @@
 """Shared pytest configuration.
+import pytest
 
 Markers:
 - ``gpu``: requires an MI355X (run via gpurun / driver round-end on a GPU box)
@@ -57,3 +58,15 @@ def env(monkeypatch):
             monkeypatch.setenv(k, str(v))
 
     return _set
+
+
+@pytest.fixture(autouse=True)
+def _reset_scanner_registry():
+    """Global-state reset: tests registering scanners never leak them
+    (reference analog: tests/conftest.py reset_global_test_state)."""
+    from agentbom_amd.scan import registry as _reg
+
+    snapshot = dict(_reg._REGISTRY)
+    yield
+    _reg._REGISTRY.clear()
+    _reg._REGISTRY.update(snapshot)
