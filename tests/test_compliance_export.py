@@ -1,0 +1,88 @@
+"""Signed compliance bundle tests + 10k-control signing perf gate."""
+
+import time
+
+import pytest
+
+from agentbom_amd.output.compliance_export import (
+    export_compliance_bundle,
+    export_compliance_bundle_timed,
+    verify_compliance_bundle,
+)
+from agentbom_amd.scan.orchestrator import run_demo_scan
+
+KEY = b"audit-key"
+
+
+@pytest.fixture(scope="module")
+def report():
+    return run_demo_scan()
+
+
+class TestBundle:
+    def test_evidence_backed_bundle(self, report):
+        bundle = export_compliance_bundle(report, "owasp_llm", hmac_key=KEY)
+        m = bundle["manifest"]
+        assert m["framework"] == "owasp_llm"
+        assert m["control_count"] > 0
+        assert m["completeness"] == "evidence_backed"
+        assert bundle["signature"]["status"] == "signed"
+        assert verify_compliance_bundle(bundle, KEY)["valid"]
+
+    def test_unsigned_is_honest(self, report, monkeypatch):
+        monkeypatch.delenv("AGENT_BOM_AUDIT_HMAC_KEY", raising=False)
+        bundle = export_compliance_bundle(report, "owasp_llm")
+        assert bundle["signature"]["status"] == "unsigned_local_bundle"
+        assert not verify_compliance_bundle(bundle, KEY)["valid"]
+
+    def test_tamper_detection(self, report):
+        bundle = export_compliance_bundle(report, "mitre_attack", hmac_key=KEY)
+        bundle["controls"][0]["evidence"][0]["risk_score"] = 0.0
+        out = verify_compliance_bundle(bundle, KEY)
+        assert not out["valid"] and "digest" in out["reason"]
+        # wrong key
+        bundle2 = export_compliance_bundle(report, "mitre_attack", hmac_key=KEY)
+        assert not verify_compliance_bundle(bundle2, b"other")["valid"]
+
+    def test_unknown_framework(self, report):
+        with pytest.raises(ValueError):
+            export_compliance_bundle(report, "iso-nope")
+
+    def test_all_tag_frameworks_exportable(self, report):
+        """15 per-finding tag frameworks (AISVS, the 16th registered
+        framework, flows through the benchmark path instead — exactly the
+        reference's split)."""
+        from agentbom_amd.models import FRAMEWORK_TAG_FIELDS
+        from agentbom_amd.scan.compliance import FRAMEWORK_REGISTRY
+
+        assert len(FRAMEWORK_TAG_FIELDS) == 15
+        assert len(FRAMEWORK_REGISTRY) == 16 and "aisvs" in FRAMEWORK_REGISTRY
+        for _field, slug in FRAMEWORK_TAG_FIELDS:
+            bundle = export_compliance_bundle(report, slug, hmac_key=KEY)
+            assert verify_compliance_bundle(bundle, KEY)["valid"]
+
+
+class TestSigningPerf:
+    def test_10k_control_bundle_under_baseline(self, report):
+        """Reference baseline: 10k-control HMAC-signed bundle in 19.4 ms.
+
+        Build a synthetic report slice producing >=10k control rows and
+        require signing end-to-end faster than the published number."""
+        import copy
+
+        big = copy.copy(report)
+        big.blast_radii = list(report.blast_radii)
+        # synthesize 10k distinct controls on cloned findings
+        base = report.blast_radii[0]
+        clones = []
+        for i in range(10_000):
+            br = copy.copy(base)
+            br.owasp_tags = [f"LLM{i:05d}"]
+            clones.append(br)
+        big.blast_radii = clones
+        big.findings = []
+        bundle = export_compliance_bundle_timed(big, "owasp_llm", hmac_key=KEY)
+        assert bundle["manifest"]["control_count"] == 10_000
+        assert bundle["generated_in_ms"] < 19.4 * 10  # CPU-box slack 10x
+        # the number we actually publish is measured on the GPU box; here
+        # we only pin the order of magnitude so regressions surface
