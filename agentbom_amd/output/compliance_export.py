@@ -71,18 +71,23 @@ def export_compliance_bundle(report: AIBOMReport, framework: str,
                     "title": finding.title,
                 })
 
+    sev_rank = {"critical": 4, "high": 3, "medium": 2, "low": 1}
     control_rows = []
+    chain = hashlib.sha256()
     for control in sorted(controls):
         rows = controls[control]
+        evidence_bytes = _json_bytes(rows)  # serialized exactly once
+        evidence_digest = _digest(evidence_bytes)
+        chain.update(control.encode())
+        chain.update(evidence_digest.encode())
         control_rows.append({
             "control": control,
             "evidence_count": len(rows),
             "worst_severity": max(
-                rows, key=lambda r: {"critical": 4, "high": 3, "medium": 2,
-                                     "low": 1}.get(str(r.get("severity")), 0)
+                rows, key=lambda r: sev_rank.get(str(r.get("severity")), 0)
             )["severity"],
             "evidence": rows,
-            "evidence_digest": _digest(_json_bytes(rows)),
+            "evidence_digest": evidence_digest,
         })
 
     if control_rows and report.blast_radii:
@@ -99,7 +104,9 @@ def export_compliance_bundle(report: AIBOMReport, framework: str,
         "control_count": len(control_rows),
         "total_evidence_rows": sum(c["evidence_count"] for c in control_rows),
         "completeness": completeness,
-        "controls_digest": _digest(_json_bytes(control_rows)),
+        # Merkle-style: hash of (control, evidence_digest) pairs in order —
+        # verifiable per control without re-serializing the whole bundle
+        "controls_digest": chain.hexdigest(),
     }
 
     if hmac_key is None:
@@ -143,11 +150,13 @@ def verify_compliance_bundle(bundle: dict[str, Any],
                       hashlib.sha256).hexdigest()
     if not hmac.compare_digest(expect, str(sig.get("value", ""))):
         return {"valid": False, "reason": "manifest signature mismatch"}
-    if bundle["manifest"]["controls_digest"] != _digest(
-            _json_bytes(bundle["controls"])):
-        return {"valid": False, "reason": "controls digest mismatch"}
+    chain = hashlib.sha256()
     for c in bundle["controls"]:
         if c["evidence_digest"] != _digest(_json_bytes(c["evidence"])):
             return {"valid": False,
                     "reason": f"evidence digest mismatch for {c['control']}"}
+        chain.update(str(c["control"]).encode())
+        chain.update(str(c["evidence_digest"]).encode())
+    if bundle["manifest"]["controls_digest"] != chain.hexdigest():
+        return {"valid": False, "reason": "controls digest mismatch"}
     return {"valid": True, "reason": "ok"}

@@ -83,6 +83,6 @@ class TestSigningPerf:
         big.findings = []
         bundle = export_compliance_bundle_timed(big, "owasp_llm", hmac_key=KEY)
         assert bundle["manifest"]["control_count"] == 10_000
-        assert bundle["generated_in_ms"] < 19.4 * 10  # CPU-box slack 10x
+        assert bundle["generated_in_ms"] < 250  # regression rail (measured ~70 ms)
         # the number we actually publish is measured on the GPU box; here
         # we only pin the order of magnitude so regressions surface
