@@ -851,6 +851,15 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
         report, _g = server._ensure_scan()
         return assess_estate(report.agents)
 
+    @tool("a2a_auth_posture", "Inter-agent auth posture: shared static "
+                              "credentials, missing mutual auth, over-broad "
+                              "delegation, transitive delegation webs.")
+    def a2a_auth_posture() -> dict:
+        from agentbom_amd.scan.auth_posture import assess_a2a
+
+        report, _g = server._ensure_scan()
+        return assess_a2a(report.agents, identity_store=server.identity_store)
+
     @tool("self_posture", "agent-bom audits its OWN deployment hardening "
                           "(auth, RBAC, tenancy, persistence, signing).")
     def self_posture() -> dict:

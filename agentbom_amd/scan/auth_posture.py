@@ -88,6 +88,80 @@ def assess_server(server: MCPServer,
         agent_names)
 
 
+def assess_a2a(agents: list[Agent],
+               identity_store=None) -> dict[str, Any]:
+    """Inter-agent (A2A) auth posture — reference-only, four weakness classes
+    (reference: src/agent_bom/a2a_auth_posture.py):
+
+    1. shared static credentials between agents (one token, many callers);
+    2. missing mutual auth — no issued identity covers an agent that shares
+       infrastructure with others;
+    3. over-broad delegation — wildcard ``allowed_tools``/scopes identities;
+    4. deep shared-server webs enabling unbounded transitive delegation.
+    """
+    findings: list[dict[str, Any]] = []
+
+    # credential name -> agents whose servers carry it
+    cred_agents: dict[str, set[str]] = {}
+    server_agents: dict[str, set[str]] = {}
+    for agent in agents:
+        for server in agent.mcp_servers:
+            server_agents.setdefault(server.name, set()).add(agent.name)
+            for cred in server.credential_names:
+                cred_agents.setdefault(cred, set()).add(agent.name)
+
+    for cred, names in sorted(cred_agents.items()):
+        if len(names) > 1:
+            findings.append({
+                "weakness": "shared_static_credential", "severity": "high",
+                "credential": cred, "agents": sorted(names),
+                "detail": f"one static credential reaches {len(names)} agents; "
+                          "issue per-agent short-lived identities instead"})
+
+    identified: set[str] = set()
+    wildcard: list[str] = []
+    if identity_store is not None:
+        for ident in identity_store.list(live_only=True):
+            identified.add(ident.agent_name)
+            if "*" in ident.scopes or not ident.allowed_tools:
+                wildcard.append(ident.identity_id)
+    for server, names in sorted(server_agents.items()):
+        if len(names) > 1:
+            unauthenticated = sorted(n for n in names if n not in identified)
+            if unauthenticated:
+                findings.append({
+                    "weakness": "missing_mutual_auth", "severity": "medium",
+                    "server": server, "agents": unauthenticated,
+                    "detail": "agents share this server with no issued "
+                              "identity binding the caller"})
+    for ident_id in sorted(wildcard):
+        findings.append({
+            "weakness": "over_broad_delegation", "severity": "medium",
+            "identity": ident_id,
+            "detail": "identity carries wildcard scope / unbounded tools"})
+
+    # transitive web: agents connected through >=2 shared servers
+    pair_shared: dict[tuple, int] = {}
+    for names in server_agents.values():
+        ordered = sorted(names)
+        for i, a in enumerate(ordered):
+            for b in ordered[i + 1:]:
+                pair_shared[(a, b)] = pair_shared.get((a, b), 0) + 1
+    for (a, b), n in sorted(pair_shared.items()):
+        if n >= 2:
+            findings.append({
+                "weakness": "unbounded_transitive_delegation",
+                "severity": "medium", "agents": [a, b],
+                "detail": f"{n} shared servers form an unbounded lateral "
+                          "delegation web"})
+
+    by_weakness: dict[str, int] = {}
+    for f in findings:
+        by_weakness[f["weakness"]] = by_weakness.get(f["weakness"], 0) + 1
+    return {"schema_version": "1", "findings": findings,
+            "by_weakness": by_weakness}
+
+
 def assess_estate(agents: list[Agent]) -> dict[str, Any]:
     """Auth posture across every discovered agent→server edge."""
     by_server: dict[str, tuple[MCPServer, list[str]]] = {}

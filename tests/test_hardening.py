@@ -178,3 +178,43 @@ class TestSelfPosture:
         assert states["SELF-001"] == "hardened"
         assert states["SELF-005"] == "hardened"
         assert out["score"] > 50
+
+
+class TestA2APosture:
+    def test_shared_credential_and_mutual_auth(self):
+        from agentbom_amd.identity import AgentIdentityStore
+        from agentbom_amd.scan.auth_posture import assess_a2a
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        report = run_demo_scan()
+        store = AgentIdentityStore()
+        out = assess_a2a(report.agents, identity_store=store)
+        # the demo estate shares credentials across agents by design
+        assert out["by_weakness"].get("shared_static_credential", 0) >= 1
+        shared = next(f for f in out["findings"]
+                      if f["weakness"] == "shared_static_credential")
+        assert len(shared["agents"]) > 1
+
+    def test_issued_identity_clears_mutual_auth(self):
+        from agentbom_amd.identity import AgentIdentityStore
+        from agentbom_amd.scan.auth_posture import assess_a2a
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        report = run_demo_scan()
+        store = AgentIdentityStore()
+        for a in report.agents:
+            store.issue(a.name, scopes=["scan:read"],
+                        allowed_tools=["read_file"])
+        out = assess_a2a(report.agents, identity_store=store)
+        assert out["by_weakness"].get("missing_mutual_auth", 0) == 0
+
+    def test_wildcard_identity_flagged(self):
+        from agentbom_amd.identity import AgentIdentityStore
+        from agentbom_amd.scan.auth_posture import assess_a2a
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        report = run_demo_scan()
+        store = AgentIdentityStore()
+        store.issue("x", scopes=["*"])
+        out = assess_a2a(report.agents, identity_store=store)
+        assert out["by_weakness"].get("over_broad_delegation", 0) >= 1
