@@ -386,6 +386,65 @@ def _resolve_introspection_targets(server_spec: Optional[str]):
     return out
 
 
+@main.command(name="posture")
+@click.option("--demo", is_flag=True, help="Assess the bundled demo estate.")
+@click.option("--a2a", is_flag=True, help="Include inter-agent (A2A) posture.")
+def posture_cmd(demo: bool, a2a: bool) -> None:
+    """Auth + self posture: MCP server auth surface and own-deployment audit."""
+    from agentbom_amd.scan.auth_posture import assess_a2a, assess_estate
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+    from agentbom_amd.scan.self_posture import evaluate_self_posture
+
+    if demo:
+        agents = run_demo_scan().agents
+    else:
+        from agentbom_amd.scan.discovery import discover_all
+
+        agents = discover_all()
+    out = {
+        "mcp_auth_posture": assess_estate(agents),
+        "self_posture": evaluate_self_posture(),
+    }
+    if a2a:
+        out["a2a_auth_posture"] = assess_a2a(agents)
+    click.echo(json.dumps(out, indent=2, default=str))
+    critical = out["mcp_auth_posture"]["critical_exposures"]
+    raise SystemExit(1 if critical else 0)
+
+
+@main.command(name="compliance-bundle")
+@click.argument("framework")
+@click.option("--demo", is_flag=True, help="Bundle the demo estate scan.")
+@click.option("-o", "--output", type=click.Path(), default=None)
+def compliance_bundle_cmd(framework: str, demo: bool,
+                          output: Optional[str]) -> None:
+    """Export a signed per-framework compliance evidence bundle."""
+    from agentbom_amd.output.compliance_export import export_compliance_bundle_timed
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    if not demo:
+        click.echo("note: scanning live estate", err=True)
+    from agentbom_amd.db.store import load_advisory_windows
+    from agentbom_amd.scan.discovery import discover_all
+    from agentbom_amd.scan.orchestrator import scan_agents
+
+    report = (run_demo_scan() if demo else
+              scan_agents(discover_all(), load_advisory_windows(offline=True)))
+    try:
+        bundle = export_compliance_bundle_timed(report, framework)
+    except ValueError as exc:
+        click.echo(str(exc), err=True)
+        raise SystemExit(2)
+    text = json.dumps(bundle, indent=2, default=str)
+    if output:
+        Path(output).write_text(text)
+        click.echo(f"wrote {output} "
+                   f"({bundle['signature']['status']}, "
+                   f"{bundle['manifest']['control_count']} controls)")
+    else:
+        click.echo(text)
+
+
 @main.command(name="graph")
 @click.argument("scan_json", type=click.Path(exists=True))
 @click.option("-f", "--format", "fmt",

@@ -218,3 +218,37 @@ class TestA2APosture:
         store.issue("x", scopes=["*"])
         out = assess_a2a(report.agents, identity_store=store)
         assert out["by_weakness"].get("over_broad_delegation", 0) >= 1
+
+
+class TestPostureCli:
+    def test_posture_command(self):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        out = CliRunner().invoke(main, ["posture", "--demo", "--a2a"])
+        assert out.exit_code in (0, 1), out.output
+        doc = json.loads(out.output[out.output.index("{"):])
+        assert "mcp_auth_posture" in doc and "self_posture" in doc
+        assert "a2a_auth_posture" in doc
+
+    def test_compliance_bundle_command(self, tmp_path, monkeypatch):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        monkeypatch.setenv("AGENT_BOM_AUDIT_HMAC_KEY", "k")
+        out = CliRunner().invoke(main, [
+            "compliance-bundle", "owasp_llm", "--demo",
+            "-o", str(tmp_path / "b.json")])
+        assert out.exit_code == 0, out.output
+        doc = json.loads((tmp_path / "b.json").read_text())
+        assert doc["signature"]["status"] == "signed"
+
+    def test_compliance_bundle_unknown_framework(self):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        out = CliRunner().invoke(main, ["compliance-bundle", "nope", "--demo"])
+        assert out.exit_code == 2

@@ -218,3 +218,19 @@ def test_severity_histogram_parity(dev):
     ).cpu().numpy()
     ref = cpu_ref.severity_histogram(owner, sev, C)
     assert np.array_equal(gpu.view(np.uint32) if gpu.dtype != np.uint32 else gpu, ref.astype(gpu.dtype))
+
+
+def test_engine_step_deterministic(estate, dev):
+    """Atomics reorder match pairs, but the sorted pipeline output must be
+    bit-identical across repeated steps (production determinism gate)."""
+    import torch as _torch
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+
+    eng = EstateEngine(estate, device=str(dev))
+    a = eng.step()
+    b = eng.step()
+    assert a["n_findings"] == b["n_findings"]
+    for key in ("pkg_idx", "win_idx", "scores", "order", "n_agents",
+                "n_creds", "n_tools"):
+        assert _torch.equal(a[key].cpu(), b[key].cpu()), key
