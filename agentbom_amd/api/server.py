@@ -64,6 +64,9 @@ class _State:
         self.scheduler = None
         self.audit_entries: list[dict] = []
         self.identity_store = None
+        from agentbom_amd.api.webhooks import WebhookRegistry
+
+        self.webhooks = WebhookRegistry()
 
 
 def create_app() -> FastAPI:
@@ -201,9 +204,14 @@ def create_app() -> FastAPI:
             job["status"] = "done"
             job["steps"].append({"step": "done", "at": _now()})
             state.metrics["scans_total"] += 1
+            state.webhooks.emit("scan.completed", {
+                "job_id": job_id, "summary": job["result"]["summary"],
+                "snapshot_id": snapshot_id})
         except Exception as exc:  # noqa: BLE001 — job boundary
             job["status"] = "failed"
             job["error"] = str(exc)
+            state.webhooks.emit("scan.failed",
+                                {"job_id": job_id, "error": str(exc)})
             state.metrics["scan_failures_total"] += 1
 
     @app.post("/v1/scan", status_code=201, dependencies=[Depends(auth)])
@@ -560,6 +568,33 @@ def create_app() -> FastAPI:
             return {"watermark": state.delta_streamer.watermark,
                     "events": [_to_ocsf_event(e) for e in events]}
         return {"watermark": state.delta_streamer.watermark, "events": events}
+
+    # ── webhooks ───────────────────────────────────────────────────────────
+
+    @app.post("/v1/webhooks", status_code=201, dependencies=[Depends(auth)])
+    def webhook_subscribe(payload: dict) -> dict:
+        try:
+            return state.webhooks.subscribe(
+                url=str(payload.get("url", "")),
+                events=list(payload.get("events") or []),
+                secret=payload.get("secret"))
+        except ValueError as exc:
+            raise HTTPException(status_code=400, detail=str(exc))
+
+    @app.get("/v1/webhooks", dependencies=[Depends(auth)])
+    def webhook_list() -> dict:
+        return {"webhooks": state.webhooks.list(),
+                "dead_letters": len(state.webhooks.dead_letters)}
+
+    @app.delete("/v1/webhooks/{webhook_id}", dependencies=[Depends(auth)])
+    def webhook_unsubscribe(webhook_id: str) -> dict:
+        if not state.webhooks.unsubscribe(webhook_id):
+            raise HTTPException(status_code=404, detail="webhook not found")
+        return {"removed": True}
+
+    @app.get("/v1/webhooks/dead-letters", dependencies=[Depends(auth)])
+    def webhook_dead_letters(limit: int = 50) -> dict:
+        return {"dead_letters": state.webhooks.dead_letters[-limit:]}
 
     # ── identity lifecycle (reference: api/routes/identities.py) ───────────
 

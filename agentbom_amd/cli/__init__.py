@@ -19,8 +19,8 @@ from agentbom_amd import __version__
 
 FORMATS = [
     "console", "json", "html", "sarif", "cyclonedx", "spdx", "spdx2", "ocsf",
-    "csv", "markdown", "plain", "junit", "prometheus", "parquet", "svg",
-    "badge", "graph", "mermaid", "dot", "graphml", "cypher",
+    "csv", "markdown", "plain", "junit", "prometheus", "parquet", "pdf",
+    "svg", "badge", "graph", "mermaid", "dot", "graphml", "cypher",
 ]
 
 
@@ -76,6 +76,13 @@ def _render(report, fmt: str, output: Optional[str], verbose: bool) -> None:
         if not output:
             raise click.UsageError("parquet requires -o/--output")
         Path(output).write_bytes(misc_fmt.to_parquet_bytes(report))
+        return
+    elif fmt == "pdf":
+        from agentbom_amd.output.pdf_fmt import to_pdf_bytes
+
+        if not output:
+            raise click.UsageError("pdf requires -o/--output")
+        Path(output).write_bytes(to_pdf_bytes(report))
         return
     elif fmt in ("svg", "badge"):
         text = misc_fmt.to_badge_svg(report)

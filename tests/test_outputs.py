@@ -188,3 +188,52 @@ class TestParquetBadge:
     def test_badge(self, report):
         svg = misc_fmt.to_badge_svg(report)
         assert svg.startswith("<svg") and "critical" in svg
+
+
+class TestPdf:
+    def test_structurally_valid(self, report):
+        import re
+        import zlib
+
+        from agentbom_amd.output.pdf_fmt import to_pdf_bytes
+
+        data = to_pdf_bytes(report)
+        assert data.startswith(b"%PDF-1.4")
+        assert data.rstrip().endswith(b"%%EOF")
+        text = data.decode("latin-1")
+        # every xref offset points at the matching "N 0 obj" header
+        xref_at = int(text.rsplit("startxref\n", 1)[1].split("\n")[0])
+        assert text[xref_at:xref_at + 4] == "xref"
+        entries = re.findall(r"^(\d{10}) 00000 n",
+                             text[xref_at:], flags=re.M)
+        for i, off in enumerate(entries, start=1):
+            assert text[int(off):].startswith(f"{i} 0 obj")
+        # content streams decompress and carry the headline text
+        streams = re.findall(rb"stream\n(.*?)\nendstream", data, re.S)
+        assert streams
+        first = zlib.decompress(streams[0]).decode("latin-1")
+        assert "AI-BOM Scan Report" in first
+        assert "CVE-2020-14343" in first
+
+    def test_many_findings_paginate(self, report):
+        import copy
+        import re
+
+        from agentbom_amd.output.pdf_fmt import to_pdf_bytes
+
+        big = copy.copy(report)
+        big.blast_radii = report.blast_radii * 20  # ~320 rows
+        data = to_pdf_bytes(big)
+        n_pages = len(re.findall(rb"/Type /Page\b(?! s)", data))
+        assert n_pages >= 4
+
+    def test_cli_pdf_output(self, report, tmp_path):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        out = CliRunner().invoke(main, ["scan", "--demo", "--offline",
+                                        "-f", "pdf",
+                                        "-o", str(tmp_path / "r.pdf")])
+        assert out.exit_code == 1  # demo estate gates by design
+        assert (tmp_path / "r.pdf").read_bytes().startswith(b"%PDF")
