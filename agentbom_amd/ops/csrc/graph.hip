@@ -162,6 +162,67 @@ __global__ void bfs_expand_edges_kernel(
     if (!build_frontier) wave_add_degree(next_count, my_claims);
 }
 
+// Bitmap-mode dense expansion.  The random dist[] probes in the edge pass
+// are 64B-granule bound (two probes x E edges x a cache line each); the
+// same membership tests against 1-bit-per-node bitmaps keep the whole
+// working set (3 x N/8 bytes ~ 4 MB at 11M nodes) resident in L2/LLC:
+//   cur_bits      nodes claimed at cur_level (the implicit frontier)
+//   visited_bits  any claimed node
+//   next_bits     nodes claimed this level (becomes cur next level)
+// atomicOr claims are exact (first setter wins), so the claim count is not
+// overcounted; dist[] is written once per claim for the output contract.
+__global__ void build_bits_kernel(
+    const uint32_t* __restrict__ dist, long long num_nodes, uint32_t cur_level,
+    uint32_t* __restrict__ cur_bits, uint32_t* __restrict__ visited_bits) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    const long long words = (num_nodes + 31) >> 5;
+    for (long long w = (long long)blockIdx.x * blockDim.x + threadIdx.x; w < words;
+         w += stride) {
+        uint32_t cur = 0, vis = 0;
+        const long long base = w << 5;
+        const int lanes = (int)((num_nodes - base) < 32 ? (num_nodes - base) : 32);
+        for (int b = 0; b < lanes; ++b) {
+            const uint32_t d = dist[base + b];
+            if (d != ABOM_UNVISITED) vis |= (1u << b);
+            if (d == cur_level) cur |= (1u << b);
+        }
+        cur_bits[w] = cur;
+        visited_bits[w] = vis;
+    }
+}
+
+__global__ void bfs_expand_edges_bits_kernel(
+    const uint32_t* __restrict__ edge_src,   // [E]
+    const uint32_t* __restrict__ col,        // [E]
+    const uint8_t* __restrict__ etype,       // [E] or nullptr
+    uint32_t allowed_mask,
+    long long num_edges,
+    uint32_t* __restrict__ dist,
+    const uint32_t* __restrict__ cur_bits,
+    uint32_t* __restrict__ visited_bits,
+    uint32_t* __restrict__ next_bits,
+    uint32_t cur_level,
+    unsigned int* __restrict__ next_count) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    unsigned my_claims = 0;
+    for (long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x; e < num_edges;
+         e += stride) {
+        if (etype && !((allowed_mask >> etype[e]) & 1u)) continue;
+        const uint32_t u = edge_src[e];
+        if (!((cur_bits[u >> 5] >> (u & 31)) & 1u)) continue;
+        const uint32_t v = col[e];
+        const uint32_t bit = 1u << (v & 31);
+        if ((visited_bits[v >> 5] & bit)) continue;
+        const uint32_t old = atomicOr(&visited_bits[v >> 5], bit);
+        if (!(old & bit)) {  // exact first claim
+            dist[v] = cur_level + 1;
+            atomicOr(&next_bits[v >> 5], bit);
+            ++my_claims;
+        }
+    }
+    wave_add_degree(next_count, my_claims);
+}
+
 // Rebuild a frontier from dist (nodes claimed at `level`): used when
 // dist-driven dense mode hands back to vertex-frontier mode after the
 // claim rate drops.  Claim counts are small here, so appends are cheap.
@@ -464,10 +525,17 @@ extern "C" int abom_bfs_run(
     const void* sources, long long n_sources, void* dist, long long num_nodes,
     void* frontier_a, void* frontier_b, void* heavy_queue, void* counters,
     int max_levels, const void* edge_src, long long num_edges, double avg_degree,
-    void* stream) {
+    void* bits, void* stream) {
     // counters layout: [0] next_count, [1] heavy_count, [2] frontier degree sum
+    // bits (optional): u32[3 * ceil(N/32)] — cur/next/visited bitmaps for the
+    // L2-resident dense mode; nullptr falls back to dist-probe dense mode.
     hipStream_t s = (hipStream_t)stream;
     unsigned int* ctr = (unsigned int*)counters;
+    const long long bit_words = (num_nodes + 31) >> 5;
+    uint32_t* cur_bits = (uint32_t*)bits;
+    uint32_t* next_bits = cur_bits ? cur_bits + bit_words : nullptr;
+    uint32_t* visited_bits = cur_bits ? cur_bits + 2 * bit_words : nullptr;
+    bool bits_ready = false;
     int rc = abom_bfs_init(dist, num_nodes, stream);
     if (rc) return -rc;
     ABOM_CHECK(hipMemsetAsync(ctr, 0, 3 * sizeof(unsigned int), s));
@@ -497,10 +565,35 @@ extern "C" int abom_bfs_run(
                             (double)frontier_degree > (double)num_edges / 8.0);
         if (dense) {
             stay_dense = true;
-            rc = abom_bfs_expand_edges(edge_src, col, etype, allowed_mask, num_edges, row_off,
-                                       dist, (unsigned int)(level - 1), nxt, ctr, ctr + 2,
-                                       /*build_frontier=*/0, num_nodes, stream);
-            if (rc) return -rc;
+            if (cur_bits) {
+                if (!bits_ready) {
+                    hipLaunchKernelGGL(abom::build_bits_kernel,
+                                       dim3(abom::grid_for((num_nodes + 31) >> 5, 256)),
+                                       dim3(256), 0, s,
+                                       (const uint32_t*)dist, num_nodes,
+                                       (unsigned int)(level - 1), cur_bits, visited_bits);
+                    rc = (int)hipGetLastError();
+                    if (rc) return -rc;
+                    bits_ready = true;
+                }
+                ABOM_CHECK(hipMemsetAsync(next_bits, 0,
+                                          bit_words * sizeof(uint32_t), s));
+                hipLaunchKernelGGL(abom::bfs_expand_edges_bits_kernel,
+                                   dim3(abom::grid_for(num_edges, 256)), dim3(256), 0, s,
+                                   (const uint32_t*)edge_src, (const uint32_t*)col,
+                                   (const uint8_t*)etype, allowed_mask, num_edges,
+                                   (uint32_t*)dist, cur_bits, visited_bits, next_bits,
+                                   (unsigned int)(level - 1), ctr);
+                rc = (int)hipGetLastError();
+                if (rc) return -rc;
+                uint32_t* tb = cur_bits; cur_bits = next_bits; next_bits = tb;
+            } else {
+                rc = abom_bfs_expand_edges(edge_src, col, etype, allowed_mask, num_edges,
+                                           row_off, dist, (unsigned int)(level - 1), nxt,
+                                           ctr, ctr + 2, /*build_frontier=*/0,
+                                           num_nodes, stream);
+                if (rc) return -rc;
+            }
         } else if (frontier_size < 0) {
             // dense mode handed back: rebuild the frontier from dist
             hipLaunchKernelGGL(abom::collect_frontier_kernel,
@@ -549,6 +642,7 @@ extern "C" int abom_bfs_run(
             // level (frontier rebuilt from dist — signalled by negative size)
             stay_dense = false;
             frontier_size = -1;
+            bits_ready = false;  // re-entry rebuilds bitmaps from dist
         }
         uint32_t* t = cur; cur = nxt; nxt = t;
     }

@@ -34,7 +34,7 @@ _SIGNATURES = {
     "abom_bfs_seed": ([_c, _i64, _c, _c, _c, _c, _c], _i32),
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
     "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _c, _i64, _c], _i32),
-    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c], _i32),
+    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c, _c], _i32),
     "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _c, _u32, _c, _c, _c, _i32, _i64, _c], _i32),
     "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
@@ -198,12 +198,26 @@ def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0
     ctr = ws.get("counters")
     if ctr is None or ctr.numel() < 4:
         ctr = torch.zeros(4, dtype=torch.int32, device=dev)
+    # dense-mode bitmaps: 3 x ceil(N/32) u32 words, L2/LLC-resident
+    bit_words = (num_nodes + 31) // 32
+    bits = ws.get("bits")
+    if bits is None or bits.numel() < 3 * bit_words:
+        bits = torch.empty(3 * bit_words, dtype=torch.int32, device=dev)
     if workspace is not None:
-        workspace.update(dist=dist, frontier_a=fa, frontier_b=fb, heavy=hq, counters=ctr)
+        workspace.update(dist=dist, frontier_a=fa, frontier_b=fb, heavy=hq,
+                         counters=ctr, bits=bits)
 
     et = _ptr(etype) if etype is not None else None
     if os.environ.get("AGENT_BOM_BFS_DENSE", "1") == "0":
         edge_src = None  # A/B toggle: disable edge-centric dense-frontier mode
+    # Bitmap dense mode measured 2x SLOWER than dist-probe mode on MI355X:
+    # atomicOr packs 32 nodes per cache line and the 8 XCDs' private L2s
+    # ping-pong those shared lines through the coherence point, while the
+    # scattered dist[] writes touch each line once.  Kept behind an opt-in
+    # flag for future single-XCD / small-graph experiments.
+    bits_ptr = None
+    if os.environ.get("AGENT_BOM_BFS_BITS", "0") == "1":
+        bits_ptr = _ptr(bits)
     es = _ptr(edge_src) if edge_src is not None else None
     num_edges = col.numel()
     avg_degree = num_edges / max(num_nodes, 1)
@@ -211,7 +225,7 @@ def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0
         _ptr(row_off), _ptr(col), et, allowed_mask,
         _ptr(sources), sources.numel(), _ptr(dist), num_nodes,
         _ptr(fa), _ptr(fb), _ptr(hq), _ptr(ctr), max_levels,
-        es, num_edges, float(avg_degree), _stream(),
+        es, num_edges, float(avg_degree), bits_ptr, _stream(),
     )
     if rc < 0:
         _check(-rc, "abom_bfs_run")
