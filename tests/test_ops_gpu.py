@@ -234,3 +234,52 @@ def test_engine_step_deterministic(estate, dev):
     for key in ("pkg_idx", "win_idx", "scores", "order", "n_agents",
                 "n_creds", "n_tools"):
         assert _torch.equal(a[key].cpu(), b[key].cpu()), key
+
+
+def test_orchestrator_scan_gpu_vs_cpu(dev):
+    """The REAL scan path (AdvisoryArena from OSV-style windows + fallback
+    comparator rows) must produce identical findings on GPU and CPU."""
+    import random
+
+    from agentbom_amd.db.arena import AdvisoryWindow
+    from agentbom_amd.models import Agent, AgentType, MCPServer, Package, Severity
+    from agentbom_amd.scan.orchestrator import ScanOptions, scan_agents
+
+    rng = random.Random(7)
+    ecos = ["pypi", "npm", "cargo", "deb", "maven"]
+    windows = []
+    pkgs = []
+    for i in range(2_000):
+        eco = ecos[i % len(ecos)]
+        name = f"pkg{i % 600}"
+        lo = f"{rng.randint(0, 3)}.{rng.randint(0, 9)}.0"
+        hi = f"{rng.randint(4, 9)}.{rng.randint(0, 9)}.0"
+        windows.append(AdvisoryWindow(
+            ecosystem=eco, package_name=name, vuln_id=f"CVE-T-{i}",
+            introduced=lo, fixed=hi, last_affected=None,
+            severity=Severity.HIGH, summary="t"))
+    for i in range(12_000):  # above the >=10k GPU threshold
+        eco = ecos[i % len(ecos)]
+        version = f"{rng.randint(0, 9)}.{rng.randint(0, 9)}.{rng.randint(0, 9)}"
+        if i % 97 == 0:
+            version = f"{version}+weird.build.meta"  # exercise fallback rows
+        pkgs.append(Package(name=f"pkg{i % 700}", version=version,
+                            ecosystem=eco))
+    server = MCPServer(name="srv", command="x", packages=pkgs)
+    agent = Agent(name="a", agent_type=AgentType.CUSTOM, config_path="/x",
+                  mcp_servers=[server])
+
+    def findings(use_gpu):
+        # fresh model objects each run: scan_agents mutates packages
+        import copy
+
+        report = scan_agents(copy.deepcopy([agent]), windows,
+                             ScanOptions(use_gpu=use_gpu))
+        return sorted((br.vulnerability.id, br.package.name,
+                       br.package.version, round(br.risk_score, 6))
+                      for br in report.blast_radii)
+
+    gpu = findings(True)
+    cpu = findings(False)
+    assert gpu == cpu
+    assert len(gpu) > 100  # the corpus must actually match things
