@@ -294,8 +294,21 @@ def create_app() -> FastAPI:
         return json.loads(json.dumps(to_json(report), default=str))
 
     @app.get("/v1/findings", dependencies=[Depends(auth)])
-    def findings(severity: Optional[str] = None, limit: int = 100) -> dict:
+    def findings(severity: Optional[str] = None, limit: int = 100,
+                 view: Optional[str] = None) -> dict:
         report = _latest_report()
+        if view:
+            from agentbom_amd.output import finding_views as fv
+
+            views = {"by_severity": fv.by_severity, "by_package": fv.by_package,
+                     "by_agent": fv.by_agent, "by_framework": fv.by_framework,
+                     "compact": fv.to_compact}
+            fn = views.get(view)
+            if fn is None:
+                raise HTTPException(status_code=400,
+                                    detail=f"unknown view {view!r}; "
+                                           f"one of {sorted(views)}")
+            return fn(report)
         rows = [f.to_dict() for f in report.to_findings()]
         if severity:
             rows = [r for r in rows if r["severity"] == severity]

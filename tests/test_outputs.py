@@ -237,3 +237,31 @@ class TestPdf:
                                         "-o", str(tmp_path / "r.pdf")])
         assert out.exit_code == 1  # demo estate gates by design
         assert (tmp_path / "r.pdf").read_bytes().startswith(b"%PDF")
+
+
+class TestFindingViews:
+    def test_views_and_compact(self, report):
+        from agentbom_amd.output import finding_views as fv
+
+        sev = fv.by_severity(report)
+        assert list(sev["groups"]) == [s for s in
+                                       ("critical", "high", "medium", "low",
+                                        "unknown", "none") if s in sev["groups"]]
+        total = sum(len(v) for v in sev["groups"].values())
+        assert total == len(report.blast_radii)
+
+        pkg = fv.by_package(report)
+        firsts = [max(r["risk_score"] for r in v)
+                  for v in pkg["groups"].values()]
+        assert firsts == sorted(firsts, reverse=True)
+
+        ag = fv.by_agent(report)
+        assert ag["groups"]
+
+        fw = fv.by_framework(report)
+        assert all(rows for rows in fw["groups"].values())
+
+        compact = fv.to_compact(report, top=5)
+        assert len(compact["top_findings"]) == 5
+        assert compact["summary"]["total_findings"] == len(report.blast_radii)
+        assert compact["summary"]["kev_count"] >= 1
