@@ -158,6 +158,10 @@ def main():
         rng_idx = torch.arange(args.queries, dtype=torch.int64) * 997 % est.n_packages
         nodes = (rng_idx + est.pkg_base).to(device)
         lat = []
+        # serving mode: the query is captured once into a hipGraph and each
+        # request replays it (no per-query launch construction)
+        if use_gpu:
+            engine.enable_query_graph(batch=1, max_hops=4)
         # warm the query path
         engine.blast_radius_query(nodes[:1], max_hops=4)
         if use_gpu:

@@ -283,3 +283,27 @@ def test_orchestrator_scan_gpu_vs_cpu(dev):
     cpu = findings(False)
     assert gpu == cpu
     assert len(gpu) > 100  # the corpus must actually match things
+
+
+def test_query_graph_replay_matches_eager(estate, dev):
+    """hipGraph-captured blast query == eager launches, across inputs."""
+    import torch as _torch
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+
+    eng = EstateEngine(estate, device=str(dev))
+    nodes = _torch.tensor([estate.pkg_base + 7, estate.pkg_base + 1234,
+                           estate.pkg_base + 50_000], dtype=_torch.int32,
+                          device=dev)
+    eager = [eng.blast_radius_query(nodes[i:i + 1], max_hops=4)
+             for i in range(3)]
+    eng.enable_query_graph(batch=1, max_hops=4)
+    for i in range(3):
+        gn, gh, gc, gt = eng.blast_radius_query(nodes[i:i + 1], max_hops=4)
+        en, eh, ec, et_ = eager[i]
+        assert _torch.equal(gc.cpu(), ec.cpu())
+        k = int(ec[0])
+        # node sets match (order within a hop may differ across launches)
+        assert (set(gn[0, :k].cpu().tolist())
+                == set(en[0, :k].cpu().tolist()))
+        assert _torch.equal(gt.cpu(), et_.cpu())
