@@ -120,6 +120,7 @@ def _scan_impl(
     scan_secrets: bool = False, model_files: Optional[str] = None,
     code: Optional[str] = None, iac: Optional[str] = None,
     aws_inventory: Optional[str] = None, endpoint: bool = False,
+    notebooks: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -196,6 +197,8 @@ def _scan_impl(
         run_scanner_driver("endpoint", report)
     if iac:
         run_scanner_driver("iac", report, iac)
+    if notebooks:
+        run_scanner_driver("notebooks", report, notebooks)
     if code:
         from agentbom_amd.scan.ast_analysis import (
             apply_symbol_reachability,
@@ -278,6 +281,8 @@ def _scan_options(f):
                      help="Evaluate CIS checks over an exported AWS inventory JSON."),
         click.option("--endpoint", is_flag=True,
                      help="Collect bounded workstation endpoint inventory."),
+        click.option("--notebooks", type=click.Path(exists=True), default=None,
+                     help="Scan Jupyter notebooks (pip installs, secrets in outputs, sinks)."),
     ]
     for o in reversed(opts):
         f = o(f)

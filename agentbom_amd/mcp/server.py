@@ -423,13 +423,19 @@ class AgentBomMcpServer:
                 "deploy_verdict": self.tools["should_i_deploy"].fn(),
             }
 
-        @tool("remediate", "Prioritized remediation plan: what to upgrade, in "
-                           "what order, and what it clears.")
+        @tool("remediate", "Prioritized remediation plan with executable "
+                           "per-ecosystem upgrade commands.")
         def remediate() -> dict:
             from agentbom_amd.output.json_fmt import _build_remediation_json
+            from agentbom_amd.scan.remediation import (
+                remediation_commands,
+                remediation_script,
+            )
 
             report, _g = self._ensure_scan()
-            return {"plan": _build_remediation_json(report)}
+            return {"plan": _build_remediation_json(report),
+                    "commands": remediation_commands(report),
+                    "script": remediation_script(report)}
 
         @tool("verify", "Re-scan and verify whether a finding is resolved.",
               {"type": "object", "properties": {"vuln_id": {"type": "string"}},

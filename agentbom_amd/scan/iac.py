@@ -106,8 +106,37 @@ _COMPOSE_RULES = [
      "compose service on host network", "high", "T1610"),
 ]
 
+# GitHub Actions workflow supply-chain rules (reference: github_actions.py)
+_GHA_RULES = [
+    ("GHA001", re.compile(r"uses:\s*\S+@(?:main|master|v?\d+(?:\.\d+)?)\s*$",
+                          re.M),
+     "action pinned to a mutable ref (tag/branch) instead of a commit SHA",
+     "medium", "T1195.002"),
+    ("GHA002", re.compile(r"(?i)on:\s*\n?\s*-?\s*pull_request_target"),
+     "pull_request_target grants secrets to fork PRs (pwn-request surface)",
+     "high", "T1195"),
+    ("GHA003", re.compile(r"\$\{\{\s*github\.event\.(?:issue|comment|"
+                          r"pull_request)\.(?:title|body)[^}]*\}\}"),
+     "untrusted event text interpolated into a run step (script injection)",
+     "critical", "T1059"),
+    ("GHA004", re.compile(r"(?i)permissions:\s*write-all"),
+     "workflow requests write-all token permissions", "medium", "T1528"),
+    ("GHA005", re.compile(r"(?i)(?:curl|wget)[^|\n]*\|\s*(?:bash|sh)\b"),
+     "remote script piped to shell inside a workflow", "critical", "T1059"),
+    ("GHA006", re.compile(r"(?i)ACTIONS_ALLOW_UNSECURE_COMMANDS\s*:\s*true"),
+     "deprecated unsecure workflow commands enabled", "high", "T1059"),
+]
+
+
+def _is_workflow(p: Path) -> bool:
+    return (p.suffix in (".yaml", ".yml")
+            and "workflows" in p.parts
+            and ".github" in p.parts)
+
+
 _FILE_RULES: list[tuple[Callable[[Path], bool], list]] = [
     (lambda p: p.suffix == ".tf", _TF_RULES),
+    (_is_workflow, _GHA_RULES),
     (lambda p: p.suffix in (".yaml", ".yml") and p.name.startswith(("docker-compose", "compose")),
      _COMPOSE_RULES),
     (lambda p: p.suffix in (".yaml", ".yml"), _K8S_RULES),

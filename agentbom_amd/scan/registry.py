@@ -181,6 +181,14 @@ def _run_endpoint(report: AIBOMReport, target: Optional[str]) -> int:
     return len(inv.processes)
 
 
+def _run_notebooks(report: AIBOMReport, target: str) -> int:
+    from agentbom_amd.scan.notebooks import scan_notebook_tree
+
+    results = scan_notebook_tree(target)
+    report.extra_data["notebooks"] = [r.to_dict() for r in results]
+    return sum(r.cells_scanned for r in results)
+
+
 def _ensure_builtins() -> None:
     if "secrets" in _REGISTRY:
         return
@@ -195,3 +203,5 @@ def _ensure_builtins() -> None:
     register_scanner(ScannerRegistration(
         "endpoint", "endpoint", _run_endpoint, failure_mode=SKIP,
         requires_target=False))
+    register_scanner(ScannerRegistration(
+        "notebooks", "notebooks", _run_notebooks, failure_mode=WARN_CONTINUE))
