@@ -216,6 +216,11 @@ def build_symbol_index(root: str | Path, max_files: int = 2000) -> SymbolIndex:
         except SyntaxError:
             continue
         index.files_scanned += 1
+    # polyglot sources (JS/TS, Go, Java, Ruby, Rust, PHP, C#) feed the same
+    # index so symbol reachability spans the whole repository
+    from agentbom_amd.scan.ast_polyglot import extend_symbol_index
+
+    extend_symbol_index(index, root, max_files=max_files)
     return index
 
 

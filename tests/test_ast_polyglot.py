@@ -1,0 +1,91 @@
+"""Polyglot sink detection + cross-language symbol reachability."""
+
+import pytest
+
+from agentbom_amd.scan.ast_polyglot import (
+    analyze_source,
+    extract_imports,
+    language_for,
+)
+
+CASES = [
+    ("js", "app.js", 'eval(userInput)', "eval", "critical"),
+    ("js", "srv.ts", 'const { execSync } = require("child_process");\n'
+                     'execSync(cmd)', "child_process.exec", "critical"),
+    ("js", "dom.jsx", 'el.innerHTML = data', "innerHTML", "high"),
+    ("go", "main.go", 'out, _ := exec.Command(name, args...).Output()',
+     "exec.Command", "high"),
+    ("go", "tpl.go", 'return template.HTML(userInput)', "template.HTML", "high"),
+    ("java", "App.java",
+     'Process p = Runtime.getRuntime().exec(cmd);', "Runtime.exec", "critical"),
+    ("java", "Ser.java", 'ObjectInputStream in = new ObjectInputStream(s);'
+                         ' in.readObject();',
+     "ObjectInputStream.readObject", "critical"),
+    ("ruby", "job.rb", 'Marshal.load(payload)', "Marshal.load", "critical"),
+    ("ruby", "sh.rb", 'system("ls #{dir}")', "system/backtick", "high"),
+    ("rust", "lib.rs", 'unsafe { *ptr }', "unsafe block", "medium"),
+    ("php", "index.php", 'eval($_GET["x"]);', "eval", "critical"),
+    ("php", "up.php", 'unserialize($data);', "unserialize", "critical"),
+    ("csharp", "P.cs", 'var f = new BinaryFormatter();', "BinaryFormatter",
+     "critical"),
+]
+
+
+class TestSinks:
+    @pytest.mark.parametrize("lang,fname,code,call,severity", CASES)
+    def test_sink_detected(self, lang, fname, code, call, severity):
+        findings, _ = analyze_source(code, fname, lang)
+        hit = next((f for f in findings if f.call == call), None)
+        assert hit is not None, [f.call for f in findings]
+        assert hit.severity == severity
+
+    def test_comments_skipped(self):
+        findings, _ = analyze_source("// eval(x)\n# nothing", "a.js", "js")
+        assert findings == []
+
+    def test_language_detection(self):
+        assert language_for("x.tsx") == "js"
+        assert language_for("x.go") == "go"
+        assert language_for("x.py") is None
+
+    def test_entrypoint_attribution(self):
+        code = "func handler(w http.ResponseWriter) {\n" \
+               "  exec.Command(name)\n}\n"
+        findings, calls = analyze_source(code, "h.go", "go")
+        assert findings[0].entrypoint == "handler"
+        assert "handler" in calls
+
+
+class TestImports:
+    def test_per_language(self):
+        assert "lodash" in extract_imports(
+            'import _ from "lodash"', "js")
+        assert "express" in extract_imports(
+            'const e = require("express")', "js")
+        assert "gin" in extract_imports(
+            'import "github.com/gin-gonic/gin"', "go")
+        assert "com" in extract_imports(
+            "import com.fasterxml.jackson.databind.ObjectMapper;", "java")
+        assert "rails" in extract_imports("require 'rails'", "ruby")
+        assert "serde" in extract_imports("use serde::Deserialize;", "rust")
+
+
+class TestCrossLanguageReachability:
+    def test_symbol_index_spans_languages(self, tmp_path):
+        from agentbom_amd.scan.ast_analysis import build_symbol_index
+
+        (tmp_path / "a.py").write_text("import yaml\nyaml.load(x)\n")
+        (tmp_path / "b.js").write_text(
+            'const lodash = require("lodash");\nfunction merge(a){'
+            ' lodash.merge(a); }\n')
+        (tmp_path / "c.go").write_text(
+            'package main\nimport "github.com/lib/pq"\n'
+            'func run() { exec.Command("ls") }\n')
+        idx = build_symbol_index(tmp_path)
+        assert idx.files_scanned >= 3
+        assert idx.calls_symbol("merge")
+        assert idx.calls_symbol("run")
+        assert idx.imports_module("lodash")
+        # sinks from all languages land in one finding stream
+        calls = {f.call for f in idx.findings}
+        assert "exec.Command" in calls
