@@ -398,6 +398,67 @@ def _resolve_introspection_targets(server_spec: Optional[str]):
     return out
 
 
+@main.command(name="watch")
+@click.option("--filesystem", type=click.Path(exists=True), default=".",
+              show_default=True, help="Tree to watch for manifest changes.")
+@click.option("--interval", type=float, default=10.0, show_default=True)
+@click.option("--max-iterations", type=int, default=0,
+              help="Stop after N checks (0 = forever); used by tests/CI.")
+@click.option("--offline", is_flag=True)
+def watch_cmd(filesystem: str, interval: float, max_iterations: int,
+              offline: bool) -> None:
+    """Watch a tree; rescan on manifest change and stream finding deltas."""
+    import time
+
+    from agentbom_amd.models import Agent, AgentType, MCPServer
+    from agentbom_amd.output.delta_stream import DeltaStreamer
+    from agentbom_amd.db.store import load_advisory_windows
+    from agentbom_amd.scan.orchestrator import ScanOptions, scan_agents
+    from agentbom_amd.scan.parsers import BUILTIN_INVENTORY_PARSERS, extract_packages
+
+    patterns = [p for p, _fn in BUILTIN_INVENTORY_PARSERS]
+    windows = load_advisory_windows(offline=offline)
+    streamer = DeltaStreamer()
+
+    def fingerprint() -> tuple:
+        root = Path(filesystem)
+        sig = []
+        for pat in patterns:
+            for f in sorted(root.rglob(pat)):
+                try:
+                    st = f.stat()
+                    sig.append((str(f), st.st_mtime_ns, st.st_size))
+                except OSError:
+                    continue
+        return tuple(sig)
+
+    def rescan() -> int:
+        pkgs = extract_packages(filesystem)
+        agent = Agent(name=f"watch:{Path(filesystem).name}",
+                      agent_type=AgentType.CUSTOM, config_path=filesystem,
+                      mcp_servers=[MCPServer(name="watched", command="",
+                                             packages=pkgs)])
+        report = scan_agents([agent], windows, ScanOptions(offline=offline))
+        events = streamer.emit(report)
+        for ev in events:
+            click.echo(json.dumps(ev, default=str))
+        return len(events)
+
+    last = fingerprint()
+    n_events = rescan()
+    click.echo(f"# watching {filesystem} ({len(last)} manifests, "
+               f"{n_events} initial findings)", err=True)
+    iterations = 0
+    while max_iterations == 0 or iterations < max_iterations:
+        iterations += 1
+        time.sleep(interval)
+        cur = fingerprint()
+        if cur != last:
+            last = cur
+            changed = rescan()
+            click.echo(f"# change detected -> {changed} delta events", err=True)
+
+
 @main.command(name="posture")
 @click.option("--demo", is_flag=True, help="Assess the bundled demo estate.")
 @click.option("--a2a", is_flag=True, help="Include inter-agent (A2A) posture.")

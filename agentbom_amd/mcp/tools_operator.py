@@ -910,6 +910,24 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
                   "expected_tenant": expected_tenant or None}
         return verify_attestation(envelope, policy)
 
+    @tool("trust_score", "Supply-chain trust score (0-100 + reasons) for a "
+                         "package: malicious/typosquat/registry/advisory/"
+                         "name signals.",
+          {"type": "object", "properties": {
+              "name": {"type": "string"},
+              "version": {"type": "string", "default": ""},
+              "ecosystem": {"type": "string", "default": "npm"}},
+           "required": ["name"]})
+    def trust_score_tool(name: str, version: str = "",
+                         ecosystem: str = "npm") -> dict:
+        from agentbom_amd.db.store import load_advisory_windows
+        from agentbom_amd.models import Package
+        from agentbom_amd.scan.trust import trust_score
+
+        return trust_score(
+            Package(name=name, version=version or "0.0.0", ecosystem=ecosystem),
+            advisory_windows=load_advisory_windows(offline=True))
+
     @tool("tool_metrics", "Per-tool call/latency/error counters for this "
                           "MCP session.")
     def tool_metrics() -> dict:

@@ -122,3 +122,86 @@ class TestGithubActionsRules:
         findings = [f for f in scan_iac_tree(tmp_path)
                     if f.rule_id.startswith("GHA")]
         assert findings == []
+
+
+class TestWatch:
+    def test_watch_detects_change(self, tmp_path):
+        import threading
+        import time
+
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        (tmp_path / "requirements.txt").write_text("pyyaml==5.3\n")
+        results = {}
+
+        def run():
+            results["out"] = CliRunner().invoke(
+                main, ["watch", "--filesystem", str(tmp_path),
+                       "--interval", "0.2", "--max-iterations", "8",
+                       "--offline"])
+
+        t = threading.Thread(target=run)
+        t.start()
+        time.sleep(0.5)
+        # introduce a new vulnerable package mid-watch
+        (tmp_path / "requirements.txt").write_text(
+            "pyyaml==5.3\nrequests==2.25.0\n")
+        t.join(timeout=30)
+        out = results["out"]
+        assert out.exit_code == 0, out.output
+        import json as _json
+
+        events = [_json.loads(l) for l in out.output.splitlines()
+                  if l.startswith("{")]
+        kinds = {e["kind"] for e in events}
+        assert "new" in kinds
+        # the mid-watch change produced at least one later event
+        assert "change detected" in out.output
+
+
+class TestTrustScore:
+    def test_clean_verified_package_grades_high(self):
+        from agentbom_amd.models import Package
+        from agentbom_amd.scan.trust import trust_score
+
+        out = trust_score(Package(
+            name="@modelcontextprotocol/server-memory", version="1.0",
+            ecosystem="npm"))
+        assert out["grade"] in ("A", "B")
+        assert any("verified" in r["reason"] for r in out["reasons"])
+
+    def test_malicious_grades_f(self):
+        from agentbom_amd.models import Package
+        from agentbom_amd.scan.trust import trust_score
+
+        out = trust_score(Package(name="reqeusts", version="1.0",
+                                  ecosystem="pypi"))
+        assert out["grade"] == "F"
+        assert out["score"] < 20
+
+    def test_advisory_density_and_kev(self):
+        from agentbom_amd.db.store import load_advisory_windows
+        from agentbom_amd.models import Package
+        from agentbom_amd.scan.trust import trust_score
+
+        windows = load_advisory_windows(offline=True)
+        out = trust_score(Package(name="pillow", version="9.0",
+                                  ecosystem="pypi"),
+                          advisory_windows=windows)
+        assert any("KEV" in r["reason"] for r in out["reasons"])
+        clean = trust_score(Package(name="totally-clean-lib", version="1.0",
+                                    ecosystem="pypi"),
+                            advisory_windows=windows)
+        assert clean["score"] > out["score"]
+
+    def test_scorecard_signal(self):
+        from agentbom_amd.models import Package
+        from agentbom_amd.scan.trust import trust_score
+
+        good = Package(name="well-run", version="1", ecosystem="pypi")
+        good.scorecard_score = 9.0
+        bad = Package(name="well-run", version="1", ecosystem="pypi")
+        bad.scorecard_score = 2.0
+        assert trust_score(good)["score"] > trust_score(bad)["score"]
