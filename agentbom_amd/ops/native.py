@@ -277,6 +277,43 @@ def bfs_level(row_off, col, frontier, dist, level: int, etype=None,
     return nxt[:n].clone()
 
 
+def bfs_level_edges(row_off, col, edge_src, frontier_dist_level: int, dist,
+                    etype=None, allowed_mask: int = 0xFFFFFFFF,
+                    workspace: Optional[dict] = None):
+    """Edge-centric ONE-level expansion (dense frontiers, distributed mode).
+
+    Claims every unvisited neighbor of vertices whose dist equals
+    ``frontier_dist_level`` and RETURNS the claimed list (local + remote ids
+    alike) — the materialized frontier the RCCL exchange needs.  Used by
+    parallel/dist_bfs.py when the frontier's degree sum is a large share of
+    the shard's edges (same threshold as the single-GPU dense mode).
+    """
+    import torch
+
+    lib = load()
+    dev = row_off.device
+    ws = workspace if workspace is not None else {}
+    cap = dist.numel()
+    nxt = ws.get("next")
+    if nxt is None or nxt.numel() < cap:
+        nxt = torch.empty(cap, dtype=torch.int32, device=dev)
+        ws["next"] = nxt
+    ctr = ws.get("counters")
+    if ctr is None or ctr.numel() < 4:
+        ctr = torch.zeros(4, dtype=torch.int32, device=dev)
+        ws["counters"] = ctr
+    ctr.zero_()
+    et = _ptr(etype) if etype is not None else None
+    rc = lib.abom_bfs_expand_edges(
+        _ptr(edge_src), _ptr(col), et, allowed_mask, col.numel(),
+        _ptr(row_off), _ptr(dist), frontier_dist_level, _ptr(nxt), _ptr(ctr),
+        ctypes.c_void_p(ctr.data_ptr() + 8), 1, cap, _stream(),
+    )
+    _check(rc, "abom_bfs_expand_edges")
+    n = int(ctr[0].item())
+    return nxt[:n].clone()
+
+
 def impact_query(row_off, col, query_sources, etype=None, allowed_mask: int = 0xFFFFFFFF,
                  max_hops: int = 4, max_nodes: int = 4096):
     """Batched bounded blast-radius queries (one block per query).

@@ -39,9 +39,26 @@ def _expand_level(csr, frontier, dist, level, etype, allowed_mask, use_gpu, ws):
     if use_gpu:
         from agentbom_amd.ops import native
 
+        frontier = frontier.to(dist.device, dtype=frontier.dtype)
+        # dense-frontier handoff, same threshold as the single-GPU engine:
+        # when the frontier's outgoing degree is a big share of the shard's
+        # edges, the one-thread-per-edge pass with coalesced src/col streams
+        # beats per-vertex neighbor loops.  Claims stay materialized (the
+        # RCCL exchange needs the list), so this is the build_frontier=1
+        # edge kernel, not the dist-driven one.
+        if csr.get("src") is not None and frontier.numel():
+            import torch
+
+            f64 = frontier.to(torch.int64)
+            deg = (csr["row_off"][f64 + 1] - csr["row_off"][f64]).sum()
+            if int(deg) > csr["col"].numel() // 8:
+                return native.bfs_level_edges(
+                    csr["row_off"], csr["col"], csr["src"], level - 1, dist,
+                    etype=etype, allowed_mask=allowed_mask, workspace=ws,
+                )
         return native.bfs_level(
-            csr["row_off"], csr["col"], frontier.to(dist.device, dtype=frontier.dtype),
-            dist, level, etype=etype, allowed_mask=allowed_mask, workspace=ws,
+            csr["row_off"], csr["col"], frontier, dist, level,
+            etype=etype, allowed_mask=allowed_mask, workspace=ws,
         )
     import torch
 

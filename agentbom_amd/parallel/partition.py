@@ -58,4 +58,7 @@ def build_global_csr(edges: dict, device) -> dict:
         "row_off": row_off,
         "col": dst[order].to(torch.int32),
         "etype": et[order].contiguous(),
+        # col-aligned edge sources: enables the edge-centric dense-frontier
+        # level expansion in dist_bfs (same layout as the single-GPU engine)
+        "src": src[order].to(torch.int32),
     }

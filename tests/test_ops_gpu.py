@@ -307,3 +307,31 @@ def test_query_graph_replay_matches_eager(estate, dev):
         assert (set(gn[0, :k].cpu().tolist())
                 == set(en[0, :k].cpu().tolist()))
         assert _torch.equal(gt.cpu(), et_.cpu())
+
+
+def test_bfs_level_edges_matches_vertex_level(estate, dev):
+    """Edge-centric one-level expansion (dist dense mode) claims exactly the
+    same vertex set as the vertex-frontier kernel."""
+    import torch as _torch
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+    from agentbom_amd.ops import native
+
+    eng = EstateEngine(estate, device=str(dev))
+    mask = 0xFFFFFFFF
+    # seed: all agents at dist 0
+    N = estate.num_nodes
+    base_dist = _torch.full((N,), -1, dtype=_torch.int32, device=dev)
+    base_dist[: estate.n_agents] = 0
+    frontier = _torch.arange(estate.n_agents, dtype=_torch.int32, device=dev)
+
+    dist_a = base_dist.clone()
+    nxt_a = native.bfs_level(eng.fwd["row_off"], eng.fwd["col"], frontier,
+                             dist_a, 1, etype=eng.fwd["etype"],
+                             allowed_mask=mask)
+    dist_b = base_dist.clone()
+    nxt_b = native.bfs_level_edges(eng.fwd["row_off"], eng.fwd["col"],
+                                   eng.fwd["src"], 0, dist_b,
+                                   etype=eng.fwd["etype"], allowed_mask=mask)
+    assert set(nxt_a.cpu().tolist()) == set(nxt_b.cpu().tolist())
+    assert _torch.equal(dist_a.cpu(), dist_b.cpu())
