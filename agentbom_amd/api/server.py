@@ -582,6 +582,44 @@ def create_app() -> FastAPI:
                     "events": [_to_ocsf_event(e) for e in events]}
         return {"watermark": state.delta_streamer.watermark, "events": events}
 
+    @app.get("/v1/remediation", dependencies=[Depends(auth)])
+    def remediation(script: bool = False) -> Any:
+        from agentbom_amd.scan.remediation import (
+            remediation_commands,
+            remediation_script,
+        )
+
+        report = _latest_report()
+        if script:
+            from fastapi.responses import PlainTextResponse
+
+            return PlainTextResponse(remediation_script(report),
+                                     media_type="text/x-shellscript")
+        return {"commands": remediation_commands(report)}
+
+    @app.get("/v1/posture", dependencies=[Depends(auth)])
+    def posture() -> dict:
+        from agentbom_amd.scan.auth_posture import assess_a2a, assess_estate
+        from agentbom_amd.scan.self_posture import evaluate_self_posture
+
+        report = _latest_report()
+        return {
+            "mcp_auth_posture": assess_estate(report.agents),
+            "a2a_auth_posture": assess_a2a(report.agents,
+                                           identity_store=state.identity_store),
+            "self_posture": evaluate_self_posture(),
+        }
+
+    @app.get("/v1/trust/{ecosystem}/{name:path}", dependencies=[Depends(auth)])
+    def trust(ecosystem: str, name: str, version: str = "0.0.0") -> dict:
+        from agentbom_amd.db.store import load_advisory_windows
+        from agentbom_amd.models import Package
+        from agentbom_amd.scan.trust import trust_score
+
+        return trust_score(
+            Package(name=name, version=version, ecosystem=ecosystem),
+            advisory_windows=load_advisory_windows(offline=True))
+
     # ── webhooks ───────────────────────────────────────────────────────────
 
     @app.post("/v1/webhooks", status_code=201, dependencies=[Depends(auth)])

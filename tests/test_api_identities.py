@@ -194,3 +194,22 @@ class TestJobLifecycle:
                             "error": None, "cancel_requested": True}
         resp = c.post("/v1/scan/c1/cancel").json()
         assert resp["cancel_requested"] is True
+
+
+class TestPostureRemediationTrust:
+    def test_routes(self, client):
+        import time
+
+        job = client.post("/v1/scan", json={"demo": True}).json()
+        for _ in range(100):
+            if client.get(f"/v1/scan/{job['job_id']}").json()["status"] == "done":
+                break
+            time.sleep(0.05)
+        rem = client.get("/v1/remediation").json()
+        assert rem["commands"]
+        script = client.get("/v1/remediation?script=true")
+        assert script.text.startswith("#!/bin/sh")
+        posture = client.get("/v1/posture").json()
+        assert "mcp_auth_posture" in posture and "self_posture" in posture
+        trust = client.get("/v1/trust/pypi/reqeusts").json()
+        assert trust["grade"] == "F"
