@@ -139,7 +139,7 @@ class TestWatch:
         def run():
             results["out"] = CliRunner().invoke(
                 main, ["watch", "--filesystem", str(tmp_path),
-                       "--interval", "0.2", "--max-iterations", "8",
+                       "--interval", "0.2", "--max-iterations", "25",
                        "--offline"])
 
         t = threading.Thread(target=run)
