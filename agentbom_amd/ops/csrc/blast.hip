@@ -53,6 +53,41 @@ __device__ __forceinline__ int collect(
     return count;
 }
 
+// One pass over `node`'s FORWARD edge row appending creds and tools into
+// their LDS arrays simultaneously — the fwd row of a server is dominated by
+// CONTAINS->package edges, so scanning it once instead of once per type
+// saves the bulk of this kernel's edge traffic.
+__device__ __forceinline__ void collect_fwd2(
+    const uint64_t* __restrict__ row_off, const uint32_t* __restrict__ col,
+    const uint8_t* __restrict__ etype, uint32_t node,
+    uint8_t want_a, uint32_t* lds_a, int* count_a, int cap_a,
+    uint8_t want_b, uint32_t* lds_b, int* count_b, int cap_b,
+    int lane, bool* overflow) {
+    const uint64_t beg = row_off[node];
+    const uint64_t end = row_off[node + 1];
+    for (uint64_t base = beg; base < end; base += 64) {
+        const uint64_t e = base + lane;
+        const bool in = e < end;
+        const uint8_t et = in ? etype[e] : (uint8_t)0xFF;
+        const uint32_t v = in ? col[e] : 0u;
+        const bool va = in && et == want_a;
+        const bool vb = in && et == want_b;
+        const unsigned long long ma = __ballot(va);
+        const unsigned long long mb = __ballot(vb);
+        if (va) {
+            const int idx = *count_a + __popcll(ma & ((1ull << lane) - 1));
+            if (idx < cap_a) lds_a[idx] = v;
+        }
+        if (vb) {
+            const int idx = *count_b + __popcll(mb & ((1ull << lane) - 1));
+            if (idx < cap_b) lds_b[idx] = v;
+        }
+        *count_a += __popcll(ma);
+        *count_b += __popcll(mb);
+        if (*count_a > cap_a || *count_b > cap_b) { *overflow = true; return; }
+    }
+}
+
 // Distinct count of lds[0..m): lane-strided first-occurrence scan.
 // Also counts distinct elements whose db_flag[node] is set.
 __device__ __forceinline__ void distinct_count(
@@ -109,11 +144,10 @@ __global__ void __launch_bounds__(256) blast_count_kernel(
                 n_ag = collect(rev_off, rev_col, rev_et, srv, et_uses,
                                lds_ag[wid], n_ag, AG_CAP, lane, &overflow);
                 if (overflow) break;
-                n_cr = collect(fwd_off, fwd_col, fwd_et, srv, et_cred,
-                               lds_cr[wid], n_cr, CR_CAP, lane, &overflow);
-                if (overflow) break;
-                n_tl = collect(fwd_off, fwd_col, fwd_et, srv, et_tool,
-                               lds_tl[wid], n_tl, TL_CAP, lane, &overflow);
+                collect_fwd2(fwd_off, fwd_col, fwd_et, srv,
+                             et_cred, lds_cr[wid], &n_cr, CR_CAP,
+                             et_tool, lds_tl[wid], &n_tl, TL_CAP,
+                             lane, &overflow);
             }
         }
 
