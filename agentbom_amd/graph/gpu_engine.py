@@ -257,10 +257,13 @@ class EstateEngine:
         ov_idx = torch.nonzero(overflow).flatten()
         if ov_idx.numel():
             heavy = self._blast_counts_join(uniq_pkgs[ov_idx])
-            for key in ("n_servers", "n_agents", "n_creds_all", "n_creds_db",
-                        "n_tools_all", "n_tools_db"):
+            counts = counts.clone()
+            for k, key in enumerate(("n_servers", "n_agents", "n_creds_all",
+                                     "n_creds_db", "n_tools_all", "n_tools_db")):
                 result[key] = result[key].clone()
                 result[key][ov_idx] = heavy[key]
+                counts[ov_idx, k] = heavy[key]
+        result["counts2d"] = counts  # [U,6] — feeds the fused score_gather
         return result
 
     def _blast_counts_join(self, uniq_pkgs):
@@ -338,6 +341,28 @@ class EstateEngine:
         # map each finding to its unique-package position
         pos = torch.searchsorted(counts["uniq_pkgs"], pkg_nodes)
 
+        if self.use_gpu:
+            # fused gather + score: arena gathers, CWE-impact LUT count
+            # selection, reachability and the risk formula in ONE kernel
+            from agentbom_amd.ops import native
+
+            scores, n_agents, n_creds, n_tools = native.score_gather(
+                win_idx.contiguous(), pkg_nodes.contiguous(), pos.contiguous(),
+                self.arena["severity"], self.arena["kev"], self.arena["epss"],
+                self.arena["impact"], self.cred_lut, self.tool_lut,
+                counts["counts2d"].contiguous(), dist)
+            return {
+                "n_findings": int(n_findings),
+                "scores": scores,
+                "order": torch.argsort(scores, descending=True, stable=True),
+                "pkg_idx": pkg_idx,
+                "win_idx": win_idx,
+                "n_agents": n_agents,
+                "n_creds": n_creds,
+                "n_tools": n_tools,
+                "reach_dist": dist,
+            }
+
         # CWE-impact filtering via LUTs: 2=full, 1=db-only, 0=none
         impact = self.arena["impact"].to(torch.int64)[win_idx]
         cred_cls = self.cred_lut.to(torch.int64)[impact]
@@ -366,14 +391,8 @@ class EstateEngine:
             torch.zeros(n_findings, dtype=torch.int8, device=self.device),
         )
 
-        if self.use_gpu:
-            from agentbom_amd.ops import native
-
-            scores = native.risk_score(
-                sev.contiguous(), n_agents.to(torch.int32).contiguous(),
-                n_creds.to(torch.int32).contiguous(), n_tools.to(torch.int32).contiguous(),
-                flags.contiguous(), epss.contiguous(), scorecard, reach.contiguous(),
-            )
+        if False:
+            pass
         else:
             from agentbom_amd.ops import cpu_ref
 

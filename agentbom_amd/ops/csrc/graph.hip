@@ -388,6 +388,80 @@ struct RiskWeights {
     float reach_boost, unreach_penalty;
 };
 
+__device__ __forceinline__ float score_one(
+    uint8_t sev, uint32_t n_agents, uint32_t n_creds, uint32_t n_tools,
+    uint8_t f, float epss, float scorecard, int8_t reach, const RiskWeights& w) {
+    if (f & 4u) return 0.0f;  // suppressed / VEX
+    float base = 0.0f;
+    switch (sev) {
+        case 5: base = w.base_critical; break;
+        case 4: base = w.base_high; break;
+        case 3: base = w.base_medium; break;
+        case 2: base = w.base_low; break;
+        default: base = 0.0f;
+    }
+    const float af = fminf((float)n_agents * w.agent_w, w.agent_cap);
+    const float cf = fminf((float)n_creds * w.cred_w, w.cred_cap);
+    const float tf = fminf((float)n_tools * w.tool_w, w.tool_cap);
+    const int ai_signals = (int)(f & 1u) + (n_creds > 0) + (n_tools > 0);
+    const float ai = ai_signals >= 2 ? w.ai_boost : 0.0f;
+    const float kev = (f & 2u) ? w.kev_boost : 0.0f;
+    const float ep = (epss >= w.epss_threshold) ? w.epss_boost : 0.0f;
+    float sc = 0.0f;
+    if (scorecard >= 0.0f) {
+        if (scorecard < w.sc_t1) sc = w.sc_b1;
+        else if (scorecard < w.sc_t2) sc = w.sc_b2;
+        else if (scorecard < w.sc_t3) sc = w.sc_b3;
+    }
+    float ra = 0.0f;
+    if (reach == 1) ra = w.reach_boost;
+    else if (reach == 0) ra = -w.unreach_penalty;
+    return fmaxf(0.0f, fminf(base + af + cf + tf + ai + kev + ep + sc + ra, 10.0f));
+}
+
+// Fused per-finding gather + score: replaces the ~14-op torch chain
+// (arena gathers, CWE-impact LUT classification, count selection,
+// reachability compare, score) with one pass.
+__global__ void score_gather_kernel(
+    const long long* __restrict__ win_idx,    // [n] -> arena window rows
+    const long long* __restrict__ pkg_nodes,  // [n] -> global node ids
+    const long long* __restrict__ pos,        // [n] -> rows into counts2d
+    const uint8_t* __restrict__ a_sev,        // [W]
+    const uint8_t* __restrict__ a_kev,        // [W]
+    const float* __restrict__ a_epss,         // [W]
+    const uint8_t* __restrict__ a_impact,     // [W]
+    const uint8_t* __restrict__ cred_lut,     // [9]
+    const uint8_t* __restrict__ tool_lut,     // [9]
+    const int* __restrict__ counts2d,         // [U*6]
+    const uint32_t* __restrict__ dist,        // [N]
+    float* __restrict__ out_scores,
+    int* __restrict__ out_agents,
+    int* __restrict__ out_creds,
+    int* __restrict__ out_tools,
+    long long n, RiskWeights w) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        const long long wrow = win_idx[i];
+        const long long crow = pos[i] * 6;
+        const uint8_t impact = a_impact[wrow];
+        const uint8_t ccls = cred_lut[impact];
+        const uint8_t tcls = tool_lut[impact];
+        const int n_creds = ccls == 2 ? counts2d[crow + 2]
+                          : ccls == 1 ? counts2d[crow + 3] : 0;
+        const int n_tools = tcls == 2 ? counts2d[crow + 4]
+                          : tcls == 1 ? counts2d[crow + 5] : 0;
+        const int n_agents = counts2d[crow + 1];
+        const uint8_t flags = (uint8_t)(a_kev[wrow] ? 2u : 0u);
+        const int8_t reach = dist[pkg_nodes[i]] != ABOM_UNVISITED ? 1 : 0;
+        out_scores[i] = score_one(a_sev[wrow], (uint32_t)n_agents,
+                                  (uint32_t)n_creds, (uint32_t)n_tools, flags,
+                                  a_epss[wrow], -1.0f, reach, w);
+        out_agents[i] = n_agents;
+        out_creds[i] = n_creds;
+        out_tools[i] = n_tools;
+    }
+}
+
 __global__ void risk_score_kernel(
     const uint8_t* __restrict__ severity,     // SEVERITY_CODE (5=crit..2=low)
     const uint32_t* __restrict__ n_agents,
@@ -401,34 +475,8 @@ __global__ void risk_score_kernel(
     long long n, RiskWeights w) {
     const long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        const uint8_t f = flags[i];
-        if (f & 4u) { out[i] = 0.0f; continue; }  // suppressed / VEX
-        float base = 0.0f;
-        switch (severity[i]) {
-            case 5: base = w.base_critical; break;
-            case 4: base = w.base_high; break;
-            case 3: base = w.base_medium; break;
-            case 2: base = w.base_low; break;
-            default: base = 0.0f;
-        }
-        const float af = fminf((float)n_agents[i] * w.agent_w, w.agent_cap);
-        const float cf = fminf((float)n_creds[i] * w.cred_w, w.cred_cap);
-        const float tf = fminf((float)n_tools[i] * w.tool_w, w.tool_cap);
-        const int ai_signals = (int)(f & 1u) + (n_creds[i] > 0) + (n_tools[i] > 0);
-        const float ai = ai_signals >= 2 ? w.ai_boost : 0.0f;
-        const float kev = (f & 2u) ? w.kev_boost : 0.0f;
-        const float ep = (epss[i] >= w.epss_threshold) ? w.epss_boost : 0.0f;
-        float sc = 0.0f;
-        const float s = scorecard[i];
-        if (s >= 0.0f) {
-            if (s < w.sc_t1) sc = w.sc_b1;
-            else if (s < w.sc_t2) sc = w.sc_b2;
-            else if (s < w.sc_t3) sc = w.sc_b3;
-        }
-        float ra = 0.0f;
-        if (reach[i] == 1) ra = w.reach_boost;
-        else if (reach[i] == 0) ra = -w.unreach_penalty;
-        out[i] = fmaxf(0.0f, fminf(base + af + cf + tf + ai + kev + ep + sc + ra, 10.0f));
+        out[i] = score_one(severity[i], n_agents[i], n_creds[i], n_tools[i],
+                           flags[i], epss[i], scorecard[i], reach[i], w);
     }
 }
 
@@ -681,6 +729,33 @@ extern "C" int abom_risk_score(
                        (const uint32_t*)n_creds, (const uint32_t*)n_tools, (const uint8_t*)flags,
                        (const float*)epss, (const float*)scorecard, (const int8_t*)reach,
                        (float*)out, n, w);
+    return (int)hipGetLastError();
+}
+
+extern "C" int abom_score_gather(
+    const void* win_idx, const void* pkg_nodes, const void* pos,
+    const void* a_sev, const void* a_kev, const void* a_epss, const void* a_impact,
+    const void* cred_lut, const void* tool_lut, const void* counts2d,
+    const void* dist, void* out_scores, void* out_agents, void* out_creds,
+    void* out_tools, long long n, const float* weights22, void* stream) {
+    abom::RiskWeights w;
+    const float* p = weights22;
+    w.base_critical = p[0]; w.base_high = p[1]; w.base_medium = p[2]; w.base_low = p[3];
+    w.agent_w = p[4]; w.agent_cap = p[5]; w.cred_w = p[6]; w.cred_cap = p[7];
+    w.tool_w = p[8]; w.tool_cap = p[9]; w.ai_boost = p[10]; w.kev_boost = p[11];
+    w.epss_boost = p[12]; w.epss_threshold = p[13]; w.sc_t1 = p[14]; w.sc_b1 = p[15];
+    w.sc_t2 = p[16]; w.sc_b2 = p[17]; w.sc_t3 = p[18]; w.sc_b3 = p[19];
+    w.reach_boost = p[20]; w.unreach_penalty = p[21];
+    const int block = 256;
+    hipLaunchKernelGGL(abom::score_gather_kernel, dim3(abom::grid_for(n, block)),
+                       dim3(block), 0, (hipStream_t)stream,
+                       (const long long*)win_idx, (const long long*)pkg_nodes,
+                       (const long long*)pos, (const uint8_t*)a_sev,
+                       (const uint8_t*)a_kev, (const float*)a_epss,
+                       (const uint8_t*)a_impact, (const uint8_t*)cred_lut,
+                       (const uint8_t*)tool_lut, (const int*)counts2d,
+                       (const uint32_t*)dist, (float*)out_scores,
+                       (int*)out_agents, (int*)out_creds, (int*)out_tools, n, w);
     return (int)hipGetLastError();
 }
 

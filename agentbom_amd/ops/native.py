@@ -40,6 +40,7 @@ _SIGNATURES = {
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
     "abom_blast_counts": ([_c, _i64] + [_c] * 6 + [_i32] * 4 + [_c] * 4 + [_c], _i32),
     "abom_severity_histogram": ([_c, _c, _c, _i64, _c], _i32),
+    "abom_score_gather": ([_c] * 11 + [_c] * 4 + [_i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
 }
 
 
@@ -398,6 +399,32 @@ def risk_score(severity, n_agents, n_creds, n_tools, flags, epss, scorecard, rea
     )
     _check(rc, "abom_risk_score")
     return out
+
+
+def score_gather(win_idx, pkg_nodes, pos, arena_sev, arena_kev, arena_epss,
+                 arena_impact, cred_lut, tool_lut, counts2d, dist):
+    """Fused per-finding gather + risk score (one kernel for the whole
+    post-counts chain).  Returns (scores f32, n_agents i32, n_creds i32,
+    n_tools i32)."""
+    import torch
+
+    lib = load()
+    n = win_idx.numel()
+    dev = win_idx.device
+    out_scores = torch.empty(n, dtype=torch.float32, device=dev)
+    out_agents = torch.empty(n, dtype=torch.int32, device=dev)
+    out_creds = torch.empty(n, dtype=torch.int32, device=dev)
+    out_tools = torch.empty(n, dtype=torch.int32, device=dev)
+    w = risk_weights_array()
+    rc = lib.abom_score_gather(
+        _ptr(win_idx), _ptr(pkg_nodes), _ptr(pos), _ptr(arena_sev),
+        _ptr(arena_kev), _ptr(arena_epss), _ptr(arena_impact), _ptr(cred_lut),
+        _ptr(tool_lut), _ptr(counts2d), _ptr(dist), _ptr(out_scores),
+        _ptr(out_agents), _ptr(out_creds), _ptr(out_tools), n,
+        w.ctypes.data_as(ctypes.POINTER(ctypes.c_float)), _stream(),
+    )
+    _check(rc, "abom_score_gather")
+    return out_scores, out_agents, out_creds, out_tools
 
 
 def severity_histogram(owner, severity, num_containers: int):
