@@ -139,7 +139,7 @@ class TestWatch:
         def run():
             results["out"] = CliRunner().invoke(
                 main, ["watch", "--filesystem", str(tmp_path),
-                       "--interval", "0.2", "--max-iterations", "25",
+                       "--interval", "0.1", "--max-iterations", "100",
                        "--offline"])
 
         t = threading.Thread(target=run)
@@ -148,7 +148,7 @@ class TestWatch:
         # introduce a new vulnerable package mid-watch
         (tmp_path / "requirements.txt").write_text(
             "pyyaml==5.3\nrequests==2.25.0\n")
-        t.join(timeout=30)
+        t.join(timeout=60)
         out = results["out"]
         assert out.exit_code == 0, out.output
         import json as _json
