@@ -499,10 +499,11 @@ __global__ void severity_histogram_kernel(
 // ── C ABI ──────────────────────────────────────────────────────────────────
 
 extern "C" int abom_bfs_init(void* dist, long long n, void* stream) {
-    const int block = 256;
-    hipLaunchKernelGGL(abom::init_dist_kernel, dim3(abom::grid_for(n, block)), dim3(block), 0,
-                       (hipStream_t)stream, (uint32_t*)dist, n, ABOM_UNVISITED);
-    return (int)hipGetLastError();
+    // ABOM_UNVISITED is 0xFFFFFFFF — a byte-uniform pattern, so the DMA
+    // memset path beats a grid-stride store kernel (and frees the CUs for
+    // the match kernel running concurrently on the side stream).
+    return (int)hipMemsetAsync(dist, 0xFF, (size_t)n * sizeof(uint32_t),
+                               (hipStream_t)stream);
 }
 
 extern "C" int abom_bfs_seed(const void* sources, long long n_sources, const void* row_off,
