@@ -171,7 +171,7 @@ class EstateEngine:
             return native.bfs(
                 self.fwd["row_off"], self.fwd["col"], self.agent_ids, self.N,
                 etype=self.fwd["etype"], allowed_mask=mask, workspace=self._bfs_ws,
-                edge_src=self.fwd["src"],
+                edge_src=self.fwd["src"], rev=self.rev,
             )
         from agentbom_amd.ops import cpu_ref
 

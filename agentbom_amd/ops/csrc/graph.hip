@@ -223,6 +223,39 @@ __global__ void bfs_expand_edges_bits_kernel(
     wave_add_degree(next_count, my_claims);
 }
 
+// Direction-optimized bottom-up step: every UNVISITED vertex probes its
+// REVERSE edges for a parent in the current frontier and claims itself.
+// dist[v] reads/writes are coalesced (v is the loop index), the per-vertex
+// probe loop EARLY-EXITS on the first frontier parent, and no frontier is
+// materialized — the dense-level complement of the edge-centric push pass.
+__global__ void bfs_bottom_up_kernel(
+    const uint64_t* __restrict__ rev_off,
+    const uint32_t* __restrict__ rev_col,
+    const uint8_t* __restrict__ rev_et,   // type of edge (parent -> v)
+    uint32_t allowed_mask,
+    long long num_nodes,
+    uint32_t* __restrict__ dist,
+    uint32_t cur_level,
+    unsigned int* __restrict__ next_count) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    unsigned my_claims = 0;
+    for (long long v = (long long)blockIdx.x * blockDim.x + threadIdx.x; v < num_nodes;
+         v += stride) {
+        if (dist[v] != ABOM_UNVISITED) continue;
+        const uint64_t beg = rev_off[v];
+        const uint64_t end = rev_off[v + 1];
+        for (uint64_t e = beg; e < end; ++e) {
+            if (rev_et && !((allowed_mask >> rev_et[e]) & 1u)) continue;
+            if (dist[rev_col[e]] == cur_level) {
+                dist[v] = cur_level + 1;
+                ++my_claims;
+                break;
+            }
+        }
+    }
+    wave_add_degree(next_count, my_claims);
+}
+
 // Rebuild a frontier from dist (nodes claimed at `level`): used when
 // dist-driven dense mode hands back to vertex-frontier mode after the
 // claim rate drops.  Claim counts are small here, so appends are cheap.
@@ -574,7 +607,8 @@ extern "C" int abom_bfs_run(
     const void* sources, long long n_sources, void* dist, long long num_nodes,
     void* frontier_a, void* frontier_b, void* heavy_queue, void* counters,
     int max_levels, const void* edge_src, long long num_edges, double avg_degree,
-    void* bits, void* stream) {
+    void* bits, const void* rev_off, const void* rev_col, const void* rev_et,
+    void* stream) {
     // counters layout: [0] next_count, [1] heavy_count, [2] frontier degree sum
     // bits (optional): u32[3 * ceil(N/32)] — cur/next/visited bitmaps for the
     // L2-resident dense mode; nullptr falls back to dist-probe dense mode.
@@ -614,7 +648,15 @@ extern "C" int abom_bfs_run(
                             (double)frontier_degree > (double)num_edges / 8.0);
         if (dense) {
             stay_dense = true;
-            if (cur_bits) {
+            if (rev_off && rev_col) {
+                hipLaunchKernelGGL(abom::bfs_bottom_up_kernel,
+                                   dim3(abom::grid_for(num_nodes, 256)), dim3(256), 0, s,
+                                   (const uint64_t*)rev_off, (const uint32_t*)rev_col,
+                                   (const uint8_t*)rev_et, allowed_mask, num_nodes,
+                                   (uint32_t*)dist, (unsigned int)(level - 1), ctr);
+                rc = (int)hipGetLastError();
+                if (rc) return -rc;
+            } else if (cur_bits) {
                 if (!bits_ready) {
                     hipLaunchKernelGGL(abom::build_bits_kernel,
                                        dim3(abom::grid_for((num_nodes + 31) >> 5, 256)),

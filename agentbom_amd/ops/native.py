@@ -34,7 +34,7 @@ _SIGNATURES = {
     "abom_bfs_seed": ([_c, _i64, _c, _c, _c, _c, _c], _i32),
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
     "abom_bfs_expand_heavy": ([_c, _c, _c, _u32, _c, _c, _c, _u32, _c, _c, _c, _i64, _c], _i32),
-    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c, _c], _i32),
+    "abom_bfs_run": ([_c, _c, _c, _u32, _c, _i64, _c, _i64, _c, _c, _c, _c, _i32, _c, _i64, ctypes.c_double, _c, _c, _c, _c, _c], _i32),
     "abom_bfs_expand_edges": ([_c, _c, _c, _u32, _i64, _c, _c, _u32, _c, _c, _c, _i32, _i64, _c], _i32),
     "abom_impact_query": ([_c, _c, _c, _u32, _c, _i32, _i32, _i32, _c, _c, _c, _c, _c], _i32),
     "abom_risk_score": ([_c] * 8 + [_c, _i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
@@ -173,7 +173,8 @@ def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_of
 
 
 def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0xFFFFFFFF,
-        max_levels: int = 64, workspace: Optional[dict] = None, edge_src=None):
+        max_levels: int = 64, workspace: Optional[dict] = None, edge_src=None,
+        rev=None):
     """Multi-source BFS over CSR; returns u32 dist (UNVISITED = 0xFFFFFFFF).
 
     ``workspace`` may carry preallocated buffers (dist/frontier_a/frontier_b/
@@ -222,11 +223,18 @@ def bfs(row_off, col, sources, num_nodes: int, etype=None, allowed_mask: int = 0
     es = _ptr(edge_src) if edge_src is not None else None
     num_edges = col.numel()
     avg_degree = num_edges / max(num_nodes, 1)
+    # direction-optimized dense levels: pass the reverse CSR for bottom-up
+    rev_off_p = rev_col_p = rev_et_p = None
+    if rev is not None and os.environ.get("AGENT_BOM_BFS_BOTTOMUP", "1") != "0":
+        rev_off_p = _ptr(rev["row_off"])
+        rev_col_p = _ptr(rev["col"])
+        rev_et_p = _ptr(rev["etype"]) if rev.get("etype") is not None else None
     rc = lib.abom_bfs_run(
         _ptr(row_off), _ptr(col), et, allowed_mask,
         _ptr(sources), sources.numel(), _ptr(dist), num_nodes,
         _ptr(fa), _ptr(fb), _ptr(hq), _ptr(ctr), max_levels,
-        es, num_edges, float(avg_degree), bits_ptr, _stream(),
+        es, num_edges, float(avg_degree), bits_ptr,
+        rev_off_p, rev_col_p, rev_et_p, _stream(),
     )
     if rc < 0:
         _check(-rc, "abom_bfs_run")
