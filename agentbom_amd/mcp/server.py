@@ -45,6 +45,56 @@ def _check_strict_args(schema: dict, arguments: dict) -> Optional[dict]:
     return None
 
 
+# Prompt catalog: (description, required args, message template).
+# Read-first guidance — each prompt steers the model toward the matching
+# read tools before any write-gated action.
+_PROMPTS = {
+    "triage": (
+        "Triage the highest-risk findings.", (),
+        "Call findings_triage and risk_summary, then rank the top 10 "
+        "findings by fused priority. For each: why it ranks there (KEV, "
+        "EPSS, reachability, blast radius) and the single next action."),
+    "remediate": (
+        "Draft a remediation plan.", (),
+        "Call remediate and group the commands into waves by risk. "
+        "Produce a reviewable plan: wave, command, what it clears, "
+        "rollback note."),
+    "investigate-cve": (
+        "Deep-dive one CVE across the estate.", ("cve",),
+        "Investigate {cve}: call intel_lookup, blast_radius and "
+        "exposure_paths for it. Summarize which agents/credentials/tools "
+        "it reaches, whether runtime evidence confirms exposure, and the "
+        "exact fix."),
+    "blast-analysis": (
+        "Explain the blast radius of a package.", ("package",),
+        "For package {package}: call check_package, then graph_search and "
+        "graph_impact on its node. Explain every hop from the package to "
+        "credentials and agents in plain language."),
+    "compliance-report": (
+        "Summarize posture for one framework.", ("framework",),
+        "Call compliance_posture and analytics_query, then write an "
+        "auditor-facing summary for {framework}: controls with evidence, "
+        "gaps, and the top 3 remediations that close the most controls."),
+    "harden-mcp": (
+        "Harden the MCP server estate.", (),
+        "Call mcp_auth_posture, a2a_auth_posture and self_posture. "
+        "Produce a hardening checklist ordered by severity with the "
+        "specific config change for each item."),
+    "incident-response": (
+        "Respond to a suspected compromise.", ("indicator",),
+        "Indicator: {indicator}. Call runtime_correlate and proxy_alerts "
+        "to scope it, effective_permissions for the implicated agents, "
+        "then propose shield actions (quarantine/block/revoke) with the "
+        "exact admin-gated tool calls an operator should approve."),
+    "executive-summary": (
+        "Executive one-pager for the latest scan.", (),
+        "Call scan, intel_daily_brief and should_i_deploy. Write one page: "
+        "estate size, risk trend, deploy verdict, top 3 risks in business "
+        "language, and what was fixed since last scan (diff if a baseline "
+        "is available)."),
+}
+
+
 class McpTool:
     def __init__(self, name: str, description: str, schema: dict, fn: Callable[..., Any]):
         self.name = name
@@ -607,6 +657,16 @@ class AgentBomMcpServer:
                  "mimeType": "application/json"},
                 {"uri": "agent-bom://graph/latest", "name": "latest estate graph",
                  "mimeType": "application/json"},
+                {"uri": "agent-bom://findings/latest", "name": "unified findings",
+                 "mimeType": "application/json"},
+                {"uri": "agent-bom://remediation/latest",
+                 "name": "remediation plan + commands",
+                 "mimeType": "application/json"},
+                {"uri": "agent-bom://posture/latest",
+                 "name": "auth + self posture", "mimeType": "application/json"},
+                {"uri": "agent-bom://registry/servers",
+                 "name": "known-MCP-server security registry",
+                 "mimeType": "application/json"},
             ]})
         if method == "resources/read":
             uri = params.get("uri", "")
@@ -617,14 +677,49 @@ class AgentBomMcpServer:
                 text = json.dumps(to_json(report), default=str)
             elif uri == "agent-bom://graph/latest":
                 text = json.dumps(graph.to_dict(), default=str)
+            elif uri == "agent-bom://findings/latest":
+                text = json.dumps([f.to_dict() for f in report.to_findings()],
+                                  default=str)
+            elif uri == "agent-bom://remediation/latest":
+                from agentbom_amd.scan.remediation import remediation_commands
+
+                text = json.dumps({"commands": remediation_commands(report)},
+                                  default=str)
+            elif uri == "agent-bom://posture/latest":
+                from agentbom_amd.scan.auth_posture import assess_estate
+                from agentbom_amd.scan.self_posture import evaluate_self_posture
+
+                text = json.dumps({"mcp_auth_posture": assess_estate(report.agents),
+                                   "self_posture": evaluate_self_posture()},
+                                  default=str)
+            elif uri == "agent-bom://registry/servers":
+                from agentbom_amd.mcp.registry import load_registry
+
+                text = json.dumps(load_registry(), default=str)
             else:
                 return err(-32602, f"unknown resource {uri!r}")
             return ok({"contents": [{"uri": uri, "mimeType": "application/json", "text": text}]})
         if method == "prompts/list":
             return ok({"prompts": [
-                {"name": "triage", "description": "Triage the highest-risk findings."},
-                {"name": "remediate", "description": "Draft a remediation plan."},
+                {"name": p_name, "description": desc,
+                 "arguments": [{"name": a, "required": True} for a in args]}
+                for p_name, (desc, args, _tpl) in _PROMPTS.items()
             ]})
+        if method == "prompts/get":
+            p_name = params.get("name", "")
+            entry = _PROMPTS.get(p_name)
+            if entry is None:
+                return err(-32602, f"unknown prompt {p_name!r}")
+            desc, arg_names, template = entry
+            values = params.get("arguments") or {}
+            missing = [a for a in arg_names if a not in values]
+            if missing:
+                return err(-32602, f"missing prompt arguments {missing}")
+            return ok({"description": desc, "messages": [{
+                "role": "user",
+                "content": {"type": "text",
+                            "text": template.format(**values)},
+            }]})
         if method == "ping":
             return ok({})
         return err(-32601, f"method {method!r} not found")

@@ -182,3 +182,38 @@ class TestCli:
     def test_version(self):
         res = CliRunner().invoke(cli_main, ["--version"])
         assert "0.1.0" in res.output
+
+
+class TestMcpResourcesPrompts:
+    def test_six_resources_readable(self):
+        from agentbom_amd.mcp.server import AgentBomMcpServer
+
+        s = AgentBomMcpServer()
+        listed = s.handle({"jsonrpc": "2.0", "id": 1,
+                           "method": "resources/list"})["result"]["resources"]
+        assert len(listed) == 6
+        import json as _json
+
+        for r in listed:
+            out = s.handle({"jsonrpc": "2.0", "id": 1,
+                            "method": "resources/read",
+                            "params": {"uri": r["uri"]}})
+            assert "error" not in out, r["uri"]
+            _json.loads(out["result"]["contents"][0]["text"])
+
+    def test_eight_prompts_and_get(self):
+        from agentbom_amd.mcp.server import AgentBomMcpServer
+
+        s = AgentBomMcpServer()
+        listed = s.handle({"jsonrpc": "2.0", "id": 1,
+                           "method": "prompts/list"})["result"]["prompts"]
+        assert len(listed) == 8
+        out = s.handle({"jsonrpc": "2.0", "id": 1, "method": "prompts/get",
+                        "params": {"name": "investigate-cve",
+                                   "arguments": {"cve": "CVE-2020-14343"}}})
+        text = out["result"]["messages"][0]["content"]["text"]
+        assert "CVE-2020-14343" in text and "intel_lookup" in text
+        # missing required arg fails loudly
+        out = s.handle({"jsonrpc": "2.0", "id": 1, "method": "prompts/get",
+                        "params": {"name": "investigate-cve"}})
+        assert "error" in out
