@@ -231,7 +231,9 @@ class EstateEngine:
         Returns dense tensors indexed by position in ``uniq_pkgs``.
         """
         torch = self.torch
-        uniq_pkgs = torch.unique(finding_pkgs)
+        # finding_pkgs arrives sorted (the match-finalize pair sort), so
+        # unique_consecutive avoids torch.unique's internal re-sort
+        uniq_pkgs = torch.unique_consecutive(finding_pkgs)
         if self.use_gpu:
             return self._blast_counts_fused(uniq_pkgs)
         return self._blast_counts_join(uniq_pkgs)
