@@ -265,3 +265,40 @@ class TestFindingViews:
         assert len(compact["top_findings"]) == 5
         assert compact["summary"]["total_findings"] == len(report.blast_radii)
         assert compact["summary"]["kev_count"] >= 1
+
+
+class TestNarratives:
+    def test_advisory_text(self, report):
+        from agentbom_amd.output.narratives import advisory_text
+
+        text = advisory_text(report)
+        assert "CVE-2020-14343" in text
+        assert "KNOWN MALICIOUS" in text  # reqeusts typosquat
+        assert "Known Exploited Vulnerabilities" in text  # KEV Pillow
+        assert "Fix: upgrade to" in text
+
+    def test_compliance_narrative(self, report):
+        from agentbom_amd.output.narratives import compliance_narrative
+
+        text = compliance_narrative(report)
+        assert "OWASP LLM Top 10" in text
+        assert "worst residual risk" in text
+        empty = compliance_narrative(report, framework_field="no_such_field",
+                                     framework_label="X")
+        assert "No findings mapped" in empty
+
+    def test_cis_posture_text(self):
+        from agentbom_amd.output.narratives import cis_posture_text
+
+        rows = [
+            {"check_id": "CIS-1.1", "title": "root MFA", "severity": "critical",
+             "status": "fail", "resource": "iam:root", "detail": "no MFA"},
+            {"check_id": "CIS-2.1", "title": "bucket public", "severity": "high",
+             "status": "pass", "resource": "s3:a"},
+            {"check_id": "CIS-3.1", "title": "trail", "severity": "high",
+             "status": "error", "resource": "ct"},
+        ]
+        text = cis_posture_text(rows)
+        assert "1/3 checks pass" in text
+        assert "CIS-1.1" in text and "no MFA" in text
+        assert "UNVERIFIED" in text
