@@ -11,17 +11,21 @@
 // Zipf-head packages that overflow the LDS caps are flagged; the host
 // resolves just those with the sort-based path (hybrid, exact either way).
 // CDNA4 notes: one workgroup = 4 waves = 4 packages; LDS slice per wave =
-// (SRV_CAP + AG_CAP + CR_CAP + TL_CAP) * 4 B = 8.5 KiB -> 34 KiB/block,
-// fine within 160 KiB/CU at several blocks per CU.
+// (SRV_CAP + AG_CAP + CR_CAP + TL_CAP) * 4 B = 4.5 KiB -> 18 KiB/block,
+// sized so LDS is not the occupancy limiter (see cap comment below).
 
 #include "abom_common.h"
 
 namespace abom {
 
+// Caps sized for occupancy: 4.1 KiB LDS per wave -> 16.5 KiB per 4-wave
+// block -> 9 blocks/CU (full 32-wave occupancy; the previous 8.5 KiB/wave
+// layout measured 11.1/32).  Zipf-head packages beyond a cap overflow to
+// the exact sort-based join, so caps trade fallback rate for occupancy.
 constexpr int SRV_CAP = 64;
-constexpr int AG_CAP = 512;
-constexpr int CR_CAP = 512;
-constexpr int TL_CAP = 1024;
+constexpr int AG_CAP = 256;
+constexpr int CR_CAP = 256;
+constexpr int TL_CAP = 512;
 constexpr int WAVES_PER_BLOCK = 4;
 
 __device__ __forceinline__ int wave_reduce_add(int v) {
