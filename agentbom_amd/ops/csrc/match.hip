@@ -101,6 +101,19 @@ __global__ void match_kernel(
 
 }  // namespace abom
 
+namespace abom {
+// Match-specific grid: zipf-head advisory groups make per-package work
+// highly variable, so give the scheduler FINE work units (many more
+// workgroups than resident waves) instead of the default 2048-block
+// grid-stride — the hardware queue load-balances the long walks.
+inline int match_grid_for(long long work, int block) {
+    long long blocks = (work + block - 1) / block;
+    if (blocks > 16384) blocks = 16384;
+    if (blocks < 1) blocks = 1;
+    return (int)blocks;
+}
+}  // namespace abom
+
 extern "C" int abom_match(
     const void* pkg_group_key, const void* pkg_key_hi, const void* pkg_key_lo,
     const void* pkg_flags, long long num_packages,
@@ -111,7 +124,7 @@ extern "C" int abom_match(
     const void* w_flags,
     void* out_pairs, void* out_count, long long capacity, void* stream) {
     const int block = 256;
-    const int grid = abom::grid_for(num_packages, block);
+    const int grid = abom::match_grid_for(num_packages, block);
     hipLaunchKernelGGL(abom::match_kernel, dim3(grid), dim3(block), 0, (hipStream_t)stream,
                        (const uint64_t*)pkg_group_key, (const uint64_t*)pkg_key_hi,
                        (const uint64_t*)pkg_key_lo, (const uint8_t*)pkg_flags, num_packages,
