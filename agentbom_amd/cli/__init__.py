@@ -120,7 +120,7 @@ def _scan_impl(
     scan_secrets: bool = False, model_files: Optional[str] = None,
     code: Optional[str] = None, iac: Optional[str] = None,
     aws_inventory: Optional[str] = None, endpoint: bool = False,
-    notebooks: Optional[str] = None,
+    notebooks: Optional[str] = None, skills: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -199,6 +199,8 @@ def _scan_impl(
         run_scanner_driver("iac", report, iac)
     if notebooks:
         run_scanner_driver("notebooks", report, notebooks)
+    if skills:
+        run_scanner_driver("skills", report, skills)
     if code:
         from agentbom_amd.scan.ast_analysis import (
             apply_symbol_reachability,
@@ -283,6 +285,8 @@ def _scan_options(f):
                      help="Collect bounded workstation endpoint inventory."),
         click.option("--notebooks", type=click.Path(exists=True), default=None,
                      help="Scan Jupyter notebooks (pip installs, secrets in outputs, sinks)."),
+        click.option("--skills", type=click.Path(exists=True), default=None,
+                     help="Scan agent skill bundles (SKILL.md injection/grants/scripts)."),
     ]
     for o in reversed(opts):
         f = o(f)

@@ -74,6 +74,22 @@ def register_specialized_tools(server) -> None:  # noqa: C901 — one registrar
     _PATH_SCHEMA = {"type": "object", "properties": {"path": {"type": "string"}},
                     "required": ["path"]}
 
+    @tool("skill_bundle_scan", "Scan agent skill bundles (SKILL.md): "
+                               "injection surfaces, wildcard tool grants, "
+                               "risky bundled scripts; optional policy.",
+          {"type": "object", "properties": {
+              "path": {"type": "string"},
+              "policy": {"type": "object"}},
+           "required": ["path"]})
+    def skill_bundle_scan(path: str, policy: dict = None) -> dict:
+        from agentbom_amd.scan.skills import evaluate_skills_policy, scan_skills_tree
+
+        bundles = scan_skills_tree(path)
+        out = {"bundles": [b.to_dict() for b in bundles]}
+        if policy:
+            out["policy"] = evaluate_skills_policy(bundles, policy)
+        return out
+
     @tool("model_file_scan", "Scan model artifacts (pickle/pytorch/safetensors/"
                              "gguf/onnx) for embedded code and unsafe formats.",
           _PATH_SCHEMA)

@@ -189,6 +189,14 @@ def _run_notebooks(report: AIBOMReport, target: str) -> int:
     return sum(r.cells_scanned for r in results)
 
 
+def _run_skills(report: AIBOMReport, target: str) -> int:
+    from agentbom_amd.scan.skills import scan_skills_tree
+
+    bundles = scan_skills_tree(target)
+    report.extra_data["skills"] = [b.to_dict() for b in bundles]
+    return len(bundles)
+
+
 def _ensure_builtins() -> None:
     if "secrets" in _REGISTRY:
         return
@@ -205,3 +213,5 @@ def _ensure_builtins() -> None:
         requires_target=False))
     register_scanner(ScannerRegistration(
         "notebooks", "notebooks", _run_notebooks, failure_mode=WARN_CONTINUE))
+    register_scanner(ScannerRegistration(
+        "skills", "skills", _run_skills, failure_mode=WARN_CONTINUE))
