@@ -201,6 +201,8 @@ def _scan_impl(
         run_scanner_driver("notebooks", report, notebooks)
     if skills:
         run_scanner_driver("skills", report, skills)
+    if filesystem:
+        run_scanner_driver("floating_refs", report, filesystem)
     if code:
         from agentbom_amd.scan.ast_analysis import (
             apply_symbol_reachability,

@@ -28,11 +28,18 @@ from agentbom_amd.runtime.detectors import DetectorPipeline
 
 @dataclass
 class ProxyPolicy:
-    """Allow/deny tool lists + default action for detector warnings."""
+    """Allow/deny tool lists + default action for detector warnings.
+
+    ``mode``: "enforce" (default — blocks are enforced) or "observe" —
+    every decision is still computed and AUDITED (with the action the
+    policy WOULD have taken), but nothing is blocked.  Observe-first is
+    the standard rollout path for a new detector pipeline (reference:
+    observe_enforce.py)."""
 
     allow_tools: Optional[list[str]] = None  # None = all
     deny_tools: list[str] = field(default_factory=list)
     block_on_warn: bool = False
+    mode: str = "enforce"  # enforce | observe
 
     def tool_allowed(self, tool: str) -> bool:
         if tool in self.deny_tools:
@@ -50,6 +57,7 @@ class ProxyPolicy:
             allow_tools=data.get("allow_tools"),
             deny_tools=data.get("deny_tools", []),
             block_on_warn=data.get("block_on_warn", False),
+            mode=data.get("mode", "enforce"),
         )
 
 
@@ -116,13 +124,23 @@ class McpProxy:
         if method == "tools/call":
             tool = (frame.get("params") or {}).get("name", "")
             if not self.policy.tool_allowed(tool):
-                return "block", [{"detector": "policy", "severity": "critical",
-                                  "message": f"tool {tool!r} denied by proxy policy",
-                                  "action": "block", "evidence": {}}]
+                denial = [{"detector": "policy", "severity": "critical",
+                           "message": f"tool {tool!r} denied by proxy policy",
+                           "action": "block", "evidence": {}}]
+                if self.policy.mode == "observe":
+                    denial[0]["would_block"] = True
+                    return "warn", denial
+                return "block", denial
         action, alerts = self.pipeline.inspect(frame)
         if action == "warn" and self.policy.block_on_warn:
             action = "block"
-        return action, [a.to_dict() if hasattr(a, "to_dict") else a for a in alerts]
+        out = [a.to_dict() if hasattr(a, "to_dict") else a for a in alerts]
+        if self.policy.mode == "observe" and action == "block":
+            # observe mode: record what WOULD have happened, enforce nothing
+            for a in out:
+                a["would_block"] = True
+            return "warn", out
+        return action, out
 
     def _relay(self, src: TextIO, dst: TextIO, direction: str) -> None:
         for line in src:

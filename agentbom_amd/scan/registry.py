@@ -197,6 +197,14 @@ def _run_skills(report: AIBOMReport, target: str) -> int:
     return len(bundles)
 
 
+def _run_floating_refs(report: AIBOMReport, target: str) -> int:
+    from agentbom_amd.scan.floating_refs import scan_floating_refs
+
+    refs = scan_floating_refs(target)
+    report.extra_data["floating_refs"] = [r.to_dict() for r in refs]
+    return len(refs)
+
+
 def _ensure_builtins() -> None:
     if "secrets" in _REGISTRY:
         return
@@ -215,3 +223,6 @@ def _ensure_builtins() -> None:
         "notebooks", "notebooks", _run_notebooks, failure_mode=WARN_CONTINUE))
     register_scanner(ScannerRegistration(
         "skills", "skills", _run_skills, failure_mode=WARN_CONTINUE))
+    register_scanner(ScannerRegistration(
+        "floating_refs", "pinning", _run_floating_refs,
+        failure_mode=WARN_CONTINUE))

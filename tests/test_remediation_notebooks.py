@@ -205,3 +205,46 @@ class TestTrustScore:
         bad = Package(name="well-run", version="1", ecosystem="pypi")
         bad.scorecard_score = 2.0
         assert trust_score(good)["score"] > trust_score(bad)["score"]
+
+
+class TestFloatingRefs:
+    def test_npm_specs_classified(self, tmp_path):
+        from agentbom_amd.scan.floating_refs import scan_floating_refs
+
+        (tmp_path / "package.json").write_text(json.dumps({"dependencies": {
+            "left-pad": "*", "lodash": "latest", "express": "^4.18.0",
+            "react": "~18.2.0", "weird": "github:someone/weird",
+            "pinned": "1.2.3", "ranged": ">=2.0"}}))
+        refs = {r.name: r for r in scan_floating_refs(tmp_path)}
+        assert refs["left-pad"].spec_class == "wildcard"
+        assert refs["left-pad"].severity == "critical"
+        assert refs["lodash"].spec_class == "wildcard"
+        assert refs["express"].spec_class == "caret-range"
+        assert refs["react"].spec_class == "tilde-range"
+        assert refs["weird"].spec_class == "git-branch"
+        assert refs["ranged"].spec_class == "open-lower-bound"
+        assert "pinned" not in refs
+
+    def test_lockfile_suppresses(self, tmp_path):
+        from agentbom_amd.scan.floating_refs import scan_floating_refs
+
+        (tmp_path / "package.json").write_text(json.dumps({"dependencies": {
+            "left-pad": "*"}}))
+        (tmp_path / "package-lock.json").write_text("{}")
+        assert scan_floating_refs(tmp_path) == []
+
+    def test_requirements_and_pyproject(self, tmp_path):
+        from agentbom_amd.scan.floating_refs import scan_floating_refs
+
+        (tmp_path / "requirements.txt").write_text(
+            "flask\nrequests>=2.0\ndjango==4.2\n")
+        (tmp_path / "pyproject.toml").write_text(
+            '[project]\ndependencies = ["numpy", "scipy==1.11"]\n'
+            "[tool.poetry.dependencies]\npython = \"^3.10\"\ntorch = \"^2.0\"\n")
+        refs = scan_floating_refs(tmp_path)
+        names = {(r.name, r.spec_class) for r in refs}
+        assert ("flask", "wildcard") in names
+        assert ("requests", "open-lower-bound") in names
+        assert ("numpy", "wildcard") in names
+        assert ("torch", "caret-range") in names
+        assert not any(r.name in ("django", "scipy", "python") for r in refs)
