@@ -217,7 +217,38 @@ class AdvisoryStore:
             for row in self.conn.execute("SELECT source, last_synced, record_count FROM sync_meta")
         }
         return {"path": str(self.path), "schema_version": SCHEMA_VERSION,
-                "counts": counts, "sync": sync}
+                "counts": counts, "sync": sync,
+                "freshness": grade_freshness(sync)}
+
+
+def grade_freshness(sync: dict, now=None) -> dict:
+    """Advisory staleness grading (reference: vuln_freshness.py).
+
+    fresh < 3 days <= stale < 30 days <= expired; never-synced sources are
+    "unknown".  Scanning against an expired DB is surfaced as a scan
+    warning — decisions made on month-old advisories mislead.
+    """
+    from datetime import datetime, timezone
+
+    now = now or datetime.now(timezone.utc)
+    per_source = {}
+    worst = "fresh"
+    order = {"fresh": 0, "stale": 1, "expired": 2, "unknown": 3}
+    for source, meta in sync.items():
+        last = meta.get("last_synced")
+        try:
+            age_days = (now - datetime.fromisoformat(str(last))).total_seconds() / 86400
+        except (TypeError, ValueError):
+            per_source[source] = {"grade": "unknown", "age_days": None}
+            worst = max(worst, "unknown", key=lambda g: order[g])
+            continue
+        grade = ("fresh" if age_days < 3 else
+                 "stale" if age_days < 30 else "expired")
+        per_source[source] = {"grade": grade, "age_days": round(age_days, 2)}
+        worst = max(worst, grade, key=lambda g: order[g])
+    if not per_source:
+        worst = "unknown"
+    return {"overall": worst, "sources": per_source}
 
 
 def _now() -> str:

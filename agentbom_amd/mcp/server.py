@@ -459,8 +459,11 @@ class AgentBomMcpServer:
         def intel_daily_brief() -> dict:
             report, _g = self._ensure_scan()
             top = report.blast_radii[:5]
+            from agentbom_amd.scan.risk import estate_exec_score
+
             return {
                 "date_scope": "latest scan",
+                "estate_score": estate_exec_score(report),
                 "headline_risks": [{
                     "vuln_id": b.vulnerability.id, "risk_score": b.risk_score,
                     "package": f"{b.package.name}@{b.package.version}",

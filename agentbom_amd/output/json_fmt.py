@@ -492,6 +492,12 @@ def _build_framework_summary(blast_radii: list[BlastRadius]) -> dict[str, Any]:
     }
 
 
+def _estate_score(report: AIBOMReport) -> dict[str, Any]:
+    from agentbom_amd.scan.risk import estate_exec_score
+
+    return estate_exec_score(report)
+
+
 def _build_remediation_json(report: AIBOMReport) -> list[dict[str, Any]]:
     """Group findings by (package, fix) into prioritized remediation items."""
     items: dict[tuple[str, str], dict[str, Any]] = {}
@@ -662,4 +668,5 @@ def to_json(report: AIBOMReport) -> dict[str, Any]:
         },
         "remediation_plan": _build_remediation_json(report),
         "scan_performance": report.scan_performance_data,
+        "estate_score": _estate_score(report),
     }

@@ -85,3 +85,59 @@ def server_risk_level(score: float) -> str:
     if score >= cfg.SERVER_RISK_MEDIUM_THRESHOLD:
         return "medium"
     return "low"
+
+
+def estate_exec_score(report) -> dict:
+    """Executive estate score 0-100 (higher = healthier) with drivers.
+
+    Reference analog: exec_score.py — one number for leadership, derived
+    entirely from scan evidence: finding pressure (severity-weighted per
+    asset), confirmed-exploit exposure (KEV/malicious), reachability
+    discount, and coverage honesty (PARTIAL scans cap the score).
+    """
+    from agentbom_amd.models import ScanOutcome, active_blast_radii
+
+    brs = active_blast_radii(report.blast_radii)
+    assets = max(report.total_agents + report.total_servers, 1)
+    sev_w = {"critical": 10.0, "high": 5.0, "medium": 2.0, "low": 0.5}
+    pressure = sum(sev_w.get(b.vulnerability.severity.value, 0.0) for b in brs)
+    pressure_per_asset = pressure / assets
+
+    score = 100.0
+    drivers = []
+
+    deduction = min(pressure_per_asset * 8.0, 50.0)
+    if deduction:
+        drivers.append({"driver": "finding_pressure", "delta": -round(deduction, 1),
+                        "detail": f"{len(brs)} active findings across {assets} assets"})
+    score -= deduction
+
+    kev = sum(1 for b in brs if b.vulnerability.is_kev)
+    if kev:
+        d = min(10.0 + 5.0 * (kev - 1), 25.0)
+        drivers.append({"driver": "known_exploited", "delta": -round(d, 1),
+                        "detail": f"{kev} KEV finding(s)"})
+        score -= d
+    mal = sum(1 for b in brs if b.package.is_malicious)
+    if mal:
+        d = min(15.0 + 5.0 * (mal - 1), 30.0)
+        drivers.append({"driver": "malicious_packages", "delta": -round(d, 1),
+                        "detail": f"{mal} known-malicious package(s)"})
+        score -= d
+
+    unreachable = sum(1 for b in brs if b.reachability == "unreachable")
+    if brs and unreachable:
+        credit = min(10.0 * unreachable / len(brs), 10.0)
+        drivers.append({"driver": "reachability_pruning", "delta": round(credit, 1),
+                        "detail": f"{unreachable}/{len(brs)} findings proven unreachable"})
+        score += credit
+
+    if report.scan_run and report.scan_run.outcome is not ScanOutcome.COMPLETE:
+        drivers.append({"driver": "incomplete_evidence", "delta": "cap 60",
+                        "detail": f"scan outcome {report.scan_run.outcome.value}"})
+        score = min(score, 60.0)
+
+    score = max(0.0, min(100.0, score))
+    grade = ("A" if score >= 90 else "B" if score >= 75 else
+             "C" if score >= 60 else "D" if score >= 40 else "F")
+    return {"score": round(score, 1), "grade": grade, "drivers": drivers}

@@ -252,3 +252,54 @@ class TestPostureCli:
 
         out = CliRunner().invoke(main, ["compliance-bundle", "nope", "--demo"])
         assert out.exit_code == 2
+
+
+class TestExecScoreFreshness:
+    def test_demo_estate_scores_poorly(self):
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+        from agentbom_amd.scan.risk import estate_exec_score
+
+        out = estate_exec_score(run_demo_scan())
+        assert out["grade"] in ("D", "F")  # KEV + malicious by design
+        drivers = {d["driver"] for d in out["drivers"]}
+        assert {"known_exploited", "malicious_packages"} <= drivers
+
+    def test_clean_estate_scores_high(self):
+        from agentbom_amd.models import AIBOMReport
+        from agentbom_amd.scan.risk import estate_exec_score
+
+        out = estate_exec_score(AIBOMReport(agents=[], blast_radii=[]))
+        assert out["score"] == 100.0 and out["grade"] == "A"
+
+    def test_partial_scan_caps_score(self):
+        from agentbom_amd.models import AIBOMReport
+        from agentbom_amd.models.report import ScanIssue, ScanRun
+
+        from agentbom_amd.scan.risk import estate_exec_score
+
+        report = AIBOMReport(agents=[], blast_radii=[])
+        report.scan_run = ScanRun(issues=[ScanIssue(
+            code="x", stage="s", source="t", message="m",
+            severity="error", affects_coverage=True)])
+        out = estate_exec_score(report)
+        assert out["score"] <= 60.0
+
+    def test_freshness_grading(self):
+        from datetime import datetime, timedelta, timezone
+
+        from agentbom_amd.db.store import grade_freshness
+
+        now = datetime.now(timezone.utc)
+        sync = {
+            "osv": {"last_synced": (now - timedelta(days=1)).isoformat()},
+            "epss": {"last_synced": (now - timedelta(days=10)).isoformat()},
+            "kev": {"last_synced": (now - timedelta(days=90)).isoformat()},
+            "ghsa": {"last_synced": None},
+        }
+        out = grade_freshness(sync, now=now)
+        assert out["sources"]["osv"]["grade"] == "fresh"
+        assert out["sources"]["epss"]["grade"] == "stale"
+        assert out["sources"]["kev"]["grade"] == "expired"
+        assert out["sources"]["ghsa"]["grade"] == "unknown"
+        assert out["overall"] == "unknown"
+        assert grade_freshness({}, now=now)["overall"] == "unknown"
