@@ -90,6 +90,19 @@ def register_specialized_tools(server) -> None:  # noqa: C901 — one registrar
             out["policy"] = evaluate_skills_policy(bundles, policy)
         return out
 
+    @tool("create_mcp_server_scaffold",
+          "Generate a minimal, hardened MCP server project (stdio, strict "
+          "schemas, env-only secrets) — returned as files, nothing written.",
+          {"type": "object", "properties": {
+              "name": {"type": "string"},
+              "tools": {"type": "array", "items": {"type": "string"},
+                        "default": []}},
+           "required": ["name"]})
+    def create_mcp_server_scaffold(name: str, tools: list = None) -> dict:
+        from agentbom_amd.mcp.scaffold import generate_mcp_server_scaffold
+
+        return generate_mcp_server_scaffold(name, tools)
+
     @tool("model_file_scan", "Scan model artifacts (pickle/pytorch/safetensors/"
                              "gguf/onnx) for embedded code and unsafe formats.",
           _PATH_SCHEMA)

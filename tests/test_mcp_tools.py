@@ -392,3 +392,24 @@ class TestDispatchGuardrails:
         m = call(server, "tool_metrics")
         assert m["runtime_blueprints"]["calls"] >= 1
         assert "avg_ms" in m["runtime_blueprints"]
+
+
+class TestScaffold:
+    def test_generated_server_actually_speaks_mcp(self, server, tmp_path):
+        """The scaffold must be a WORKING stdio MCP server: write it out
+        and introspect it with our own runtime introspection client."""
+        import sys
+
+        from agentbom_amd.mcp.introspect import introspect_server
+        from agentbom_amd.models import MCPServer
+
+        out = call(server, "create_mcp_server_scaffold",
+                   name="Weather Tools", tools=["get_forecast", "get_alerts"])
+        assert set(out["files"]) == {"server.py", "smithery.yaml", "README.md"}
+        (tmp_path / "server.py").write_text(out["files"]["server.py"])
+        target = MCPServer(name="scaffold", command=sys.executable,
+                           args=[str(tmp_path / "server.py")])
+        r = introspect_server(target, timeout=20)
+        assert r.success, r.error
+        assert {t.name for t in r.runtime_tools} == {"get_forecast",
+                                                     "get_alerts"}
