@@ -335,3 +335,38 @@ def test_bfs_level_edges_matches_vertex_level(estate, dev):
                                    etype=eng.fwd["etype"], allowed_mask=mask)
     assert set(nxt_a.cpu().tolist()) == set(nxt_b.cpu().tolist())
     assert _torch.equal(dist_a.cpu(), dist_b.cpu())
+
+
+def test_distributed_reach_gpu_world1_matches_local(estate, dev):
+    """The FULL distributed code path (incl. the edge-centric dense-level
+    branch) on GPU with world_size=1 must equal the local engine BFS."""
+    import torch as _torch
+    import torch.distributed as dist_mod
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+    from agentbom_amd.parallel.dist_bfs import distributed_reach
+    from agentbom_amd.scan.synth import (
+        ET_CONTAINS,
+        ET_HAS_CRED,
+        ET_PROVIDES_TOOL,
+        ET_USES,
+    )
+
+    if not dist_mod.is_initialized():
+        store = dist_mod.TCPStore("127.0.0.1", 29751, 1, True)
+        dist_mod.init_process_group("gloo", store=store, rank=0, world_size=1)
+    try:
+        eng = EstateEngine(estate, device=str(dev))
+        mask = ((1 << ET_USES) | (1 << ET_CONTAINS) | (1 << ET_HAS_CRED)
+                | (1 << ET_PROVIDES_TOOL))
+        local = eng.dependency_reach()
+
+        csr = {"row_off": eng.fwd["row_off"], "col": eng.fwd["col"],
+               "etype": eng.fwd["etype"], "src": eng.fwd["src"]}
+        dist_out = distributed_reach(
+            csr, eng.agent_ids, estate.num_nodes, estate.num_nodes,
+            etype=eng.fwd["etype"], allowed_mask=mask)
+        assert _torch.equal(dist_out.cpu().view(_torch.int32),
+                            local.cpu().view(_torch.int32))
+    finally:
+        dist_mod.destroy_process_group()
