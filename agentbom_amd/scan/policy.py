@@ -190,7 +190,13 @@ class _Parser:
             return value == "true"
         if ttype == "IDENT":
             if value in FIELD_GETTERS:
-                return FIELD_GETTERS[value](self.br)
+                try:
+                    return FIELD_GETTERS[value](self.br)
+                except AttributeError as exc:
+                    # partial finding objects (dict imports, external scans)
+                    # must degrade to a policy error, never an internal crash
+                    raise ValueError(
+                        f"field {value!r} unavailable on this finding: {exc}")
             if value.upper() in SEVERITY_POLICY_ORDER:
                 return SEVERITY_POLICY_ORDER[value.upper()]
             raise ValueError(f"unknown field {value!r}")
