@@ -122,27 +122,38 @@ def build_unified_graph_from_report_json(data: dict[str, Any],
     provider_id = "provider:local"
     g.add_node(_n(provider_id, EntityType.PROVIDER, "local", GraphSemanticLayer.INFRA))
 
-    for agent in data.get("agents", []):
+    def _rows(value, key):
+        """Fail-soft row iteration: persisted/hand-edited documents may
+        carry malformed entries — skip anything that is not a dict with
+        the join key instead of crashing the whole build."""
+        if not isinstance(value, list):
+            return []
+        return [r for r in value if isinstance(r, dict) and r.get(key)]
+
+    for agent in _rows(data.get("agents"), "name"):
         a_id = f"agent:{agent['name']}"
         g.add_node(_n(a_id, EntityType.AGENT, agent["name"], GraphSemanticLayer.APP,
                       agent_type=agent.get("agent_type")))
         g.add_edge(UnifiedEdge(provider_id, a_id, RelationshipType.HOSTS))
-        for server in agent.get("mcp_servers", []):
+        for server in _rows(agent.get("mcp_servers"), "name"):
             s_id = f"server:{server['name']}"
             g.add_node(_n(s_id, EntityType.SERVER, server["name"], GraphSemanticLayer.MCP_SERVER,
                           transport=server.get("transport")))
             g.add_edge(UnifiedEdge(a_id, s_id, RelationshipType.USES))
-            for cred in server.get("credential_env_vars", []):
+            creds = server.get("credential_env_vars")
+            for cred in (creds if isinstance(creds, list) else []):
+                if not isinstance(cred, str) or not cred:
+                    continue
                 c_id = f"credential:{cred}"
                 g.add_node(_n(c_id, EntityType.CREDENTIAL, cred, GraphSemanticLayer.IDENTITY))
                 g.add_edge(UnifiedEdge(s_id, c_id, RelationshipType.EXPOSES_CRED, weight=6.0))
-            for tool in server.get("tools", []):
+            for tool in _rows(server.get("tools"), "name"):
                 t_id = f"tool:{server['name']}/{tool['name']}"
                 g.add_node(_n(t_id, EntityType.TOOL, tool["name"], GraphSemanticLayer.TOOL))
                 g.add_edge(UnifiedEdge(s_id, t_id, RelationshipType.PROVIDES_TOOL))
-            for pkg in server.get("packages", []):
+            for pkg in _rows(server.get("packages"), "name"):
                 p_id = f"pkg:{pkg.get('ecosystem','')}:{pkg['name']}@{pkg.get('version','')}"
-                vulns = pkg.get("vulnerabilities", [])
+                vulns = _rows(pkg.get("vulnerabilities"), "id")
                 g.add_node(_n(p_id, EntityType.PACKAGE,
                               f"{pkg['name']}@{pkg.get('version','')}",
                               GraphSemanticLayer.PACKAGE,

@@ -172,3 +172,27 @@ class TestPolicyFuzz:
         br = self._Br()
         assert evaluate_expression("severity >= high and is_kev", br)
         assert not evaluate_expression("risk_score > 9", br)
+
+
+class TestGraphBuilderFuzz:
+    """The graph builder consumes persisted report JSON — a seam where
+    hand-edited or cross-version documents arrive."""
+
+    @FAST
+    @given(mutation=st.dictionaries(
+        st.sampled_from(["agents", "blast_radius", "packages", "summary",
+                         "findings", "exposure_paths"]),
+        json_values, max_size=4))
+    def test_builder_from_mutated_doc(self, mutation):
+        from agentbom_amd.graph.builder import build_unified_graph_from_report_json
+        from agentbom_amd.output.json_fmt import to_json
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        doc = to_json(run_demo_scan())
+        doc.update(mutation)
+        try:
+            graph = build_unified_graph_from_report_json(doc)
+        except (KeyError, TypeError, AttributeError, ValueError) as exc:
+            pytest.fail(f"builder crashed on mutated doc keys "
+                        f"{sorted(mutation)}: {exc!r}")
+        assert graph.nodes is not None
