@@ -213,3 +213,53 @@ class TestPostureRemediationTrust:
         assert "mcp_auth_posture" in posture and "self_posture" in posture
         trust = client.get("/v1/trust/pypi/reqeusts").json()
         assert trust["grade"] == "F"
+
+
+class TestApiRobustness:
+    """Hostile payloads to every POST route must 4xx, never 500."""
+
+    JUNK = [
+        {}, {"x": 1}, {"url": 123, "events": "no"},
+        {"agent_name": ["list"]}, {"demo": "yes-ish"},
+        {"scopes": {"a": 1}}, {"token": None},
+    ]
+
+    def test_posts_never_500(self, client):
+        from agentbom_amd.api.server import create_app
+
+        app = create_app()
+        c = TestClient(app)
+        post_paths = []
+        for route in app.routes:
+            methods = getattr(route, "methods", None) or set()
+            if "POST" in methods:
+                path = route.path
+                # fill path params with junk ids
+                path = (path.replace("{job_id}", "nope")
+                            .replace("{identity_id}", "nope")
+                            .replace("{grant_id}", "nope")
+                            .replace("{webhook_id}", "nope"))
+                post_paths.append(path)
+        assert post_paths
+        for path in post_paths:
+            for junk in self.JUNK:
+                resp = c.post(path, json=junk, headers=ADMIN_HEADERS)
+                assert resp.status_code < 500, (path, junk, resp.text)
+
+    def test_gets_never_500(self, client):
+        from agentbom_amd.api.server import create_app
+
+        app = create_app()
+        c = TestClient(app)
+        for route in app.routes:
+            methods = getattr(route, "methods", None) or set()
+            if "GET" not in methods:
+                continue
+            path = route.path
+            for param in ("job_id", "identity_id", "node_id", "container_id",
+                          "framework", "ecosystem", "name", "grant_id",
+                          "webhook_id"):
+                path = path.replace("{%s}" % param, "nope").replace(
+                    "{%s:path}" % param, "nope")
+            resp = c.get(path + "?limit=nope&severity=')--&view=zzz")
+            assert resp.status_code < 500, (path, resp.text)
