@@ -413,3 +413,40 @@ class TestScaffold:
         assert r.success, r.error
         assert {t.name for t in r.runtime_tools} == {"get_forecast",
                                                      "get_alerts"}
+
+
+class TestToolCatalogRobustness:
+    """Every tool invoked with junk-but-schema-valid arguments must return
+    a structured result or error — the dispatch boundary never crashes."""
+
+    def _junk_for(self, schema_prop):
+        t = schema_prop.get("type")
+        if t == "integer":
+            return 1
+        if t == "number":
+            return 1.0
+        if t == "boolean":
+            return True
+        if t == "array":
+            return []
+        if t == "object":
+            return {}
+        enum = schema_prop.get("enum")
+        if enum:
+            return enum[0]
+        return "/nonexistent/zz"  # hostile path-ish string
+
+    def test_every_tool_survives_junk_args(self, server):
+        skipped = {"verify"}  # forces a full re-scan; covered elsewhere
+        for name, tool in sorted(server.tools.items()):
+            if name in skipped:
+                continue
+            args = {req: self._junk_for(tool.schema["properties"][req])
+                    for req in tool.schema.get("required", [])}
+            resp = server.handle({
+                "jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                "params": {"name": name, "arguments": args,
+                           "_meta": {"caller": f"robust-{name}"}}})
+            assert "result" in resp, name
+            payload = resp["result"]["content"][0]["text"]
+            json.loads(payload)  # always structured JSON
