@@ -404,6 +404,196 @@ def _resolve_introspection_targets(server_spec: Optional[str]):
     return out
 
 
+@main.command(name="check")
+@click.argument("spec")
+@click.option("--ecosystem", "-e", default="pypi", show_default=True)
+@click.option("--offline", is_flag=True)
+def check_cmd(spec: str, ecosystem: str, offline: bool) -> None:
+    """Check one package@version against the advisory data (exit 1 if hit)."""
+    from agentbom_amd.db.store import load_advisory_windows
+    from agentbom_amd.utils.canonical_ids import normalize_package_name
+    from agentbom_amd.utils.version_utils import version_in_range
+
+    name, _, version = spec.partition("@")
+    if not version:
+        click.echo("usage: agent-bom check NAME@VERSION", err=True)
+        raise SystemExit(2)
+    norm = normalize_package_name(name, ecosystem)
+    hits = []
+    for w in load_advisory_windows(offline=offline):
+        if (w.ecosystem.lower() == ecosystem.lower()
+                and normalize_package_name(w.package_name, w.ecosystem) == norm
+                and version_in_range(version, w.introduced, w.fixed,
+                                     w.last_affected, ecosystem)):
+            hits.append({"vuln_id": w.vuln_id, "severity": w.severity.value,
+                         "fixed": w.fixed, "is_kev": w.is_kev,
+                         "summary": w.summary})
+    click.echo(json.dumps({"package": spec, "ecosystem": ecosystem,
+                           "vulnerable": bool(hits), "advisories": hits},
+                          indent=2))
+    raise SystemExit(1 if hits else 0)
+
+
+@main.command(name="doctor")
+def doctor_cmd() -> None:
+    """Environment self-check: GPU, HIP engine, advisory data, versions."""
+    import platform
+
+    checks = []
+
+    def check(name: str, ok, detail: str) -> None:
+        checks.append({"check": name, "ok": bool(ok), "detail": detail})
+
+    check("python", True, platform.python_version())
+    try:
+        import torch
+
+        check("torch", True, torch.__version__)
+        gpu = torch.cuda.is_available()
+        check("gpu", gpu,
+              torch.cuda.get_device_name(0) if gpu else "no GPU visible "
+              "(CPU oracle path active)")
+    except Exception as exc:  # noqa: BLE001 — doctor reports, never crashes
+        check("torch", False, str(exc))
+    from agentbom_amd.ops import native
+
+    check("hip_engine", native.available(),
+          "in-tree _abom_gpu.so built" if native.available()
+          else "run `python -m agentbom_amd.ops.build` (needs hipcc)")
+    from agentbom_amd.db.store import load_advisory_windows
+
+    windows = load_advisory_windows(offline=True)
+    check("advisories", len(windows) > 0,
+          f"{len(windows)} bundled windows (run `agent-bom db sync` for live data)")
+    from agentbom_amd.mcp.registry import load_registry
+
+    reg = load_registry()
+    check("mcp_registry", len(reg["servers"]) > 0,
+          f"{len(reg['servers'])} known servers, "
+          f"{len(reg['blocklist'])} blocklist rules")
+    from agentbom_amd.scan.self_posture import evaluate_self_posture
+
+    posture = evaluate_self_posture()
+    check("self_posture", posture["score"] >= 50,
+          f"hardening score {posture['score']} (agent-bom posture for detail)")
+    ok = all(c["ok"] for c in checks if c["check"] != "gpu")
+    click.echo(json.dumps({"healthy": ok, "checks": checks}, indent=2))
+    raise SystemExit(0 if ok else 1)
+
+
+@main.command(name="trust")
+@click.argument("spec")
+@click.option("--ecosystem", "-e", default="npm", show_default=True)
+def trust_cmd(spec: str, ecosystem: str) -> None:
+    """Supply-chain trust score for a package (exit 1 on grade D/F)."""
+    from agentbom_amd.db.store import load_advisory_windows
+    from agentbom_amd.models import Package
+    from agentbom_amd.scan.trust import trust_score
+
+    name, _, version = spec.partition("@")
+    out = trust_score(
+        Package(name=name, version=version or "0.0.0", ecosystem=ecosystem),
+        advisory_windows=load_advisory_windows(offline=True))
+    click.echo(json.dumps(out, indent=2))
+    raise SystemExit(1 if out["grade"] in ("D", "F") else 0)
+
+
+@main.command(name="remediate")
+@click.option("--demo", is_flag=True)
+@click.option("--script", "as_script", is_flag=True,
+              help="Emit a reviewable shell script instead of JSON.")
+def remediate_cmd(demo: bool, as_script: bool) -> None:
+    """Prioritized remediation commands for the latest scan."""
+    from agentbom_amd.db.store import load_advisory_windows
+    from agentbom_amd.scan.orchestrator import run_demo_scan, scan_agents
+    from agentbom_amd.scan.remediation import remediation_commands, remediation_script
+
+    if demo:
+        report = run_demo_scan()
+    else:
+        from agentbom_amd.scan.discovery import discover_all
+
+        report = scan_agents(discover_all(),
+                             load_advisory_windows(offline=True))
+    if as_script:
+        click.echo(remediation_script(report))
+    else:
+        click.echo(json.dumps({"commands": remediation_commands(report)},
+                              indent=2))
+
+
+@main.command(name="attest")
+@click.argument("server_name")
+@click.option("--demo", is_flag=True)
+@click.option("--verdict", type=click.Choice(["pass", "warn", "block"]),
+              default="pass", show_default=True)
+@click.option("--verify", "verify_path", type=click.Path(exists=True),
+              default=None, help="Verify an envelope file instead of signing.")
+@click.option("-o", "--output", type=click.Path(), default=None)
+def attest_cmd(server_name: str, demo: bool, verdict: str,
+               verify_path: Optional[str], output: Optional[str]) -> None:
+    """Sign (or verify) a DSSE scan attestation for one MCP server."""
+    import os
+
+    from agentbom_amd.mcp.server import resolve_mcp_tenant_id
+    from agentbom_amd.utils.attestation import attest_scanned_server, verify_attestation
+
+    key_hex = os.environ.get("AGENT_BOM_ATTESTATION_KEY", "")
+    key_id = os.environ.get("AGENT_BOM_ATTESTATION_KEY_ID", "operator")
+    if verify_path:
+        envelope = json.loads(Path(verify_path).read_text())
+        out = verify_attestation(envelope, {
+            "keys": ({key_id: key_hex} if key_hex else {})})
+        click.echo(json.dumps(out, indent=2, default=str))
+        raise SystemExit(0 if out["valid"] else 1)
+    if not key_hex:
+        click.echo("AGENT_BOM_ATTESTATION_KEY not set", err=True)
+        raise SystemExit(2)
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    report = run_demo_scan() if demo else None
+    if report is None:
+        from agentbom_amd.db.store import load_advisory_windows
+        from agentbom_amd.scan.discovery import discover_all
+        from agentbom_amd.scan.orchestrator import scan_agents
+
+        report = scan_agents(discover_all(),
+                             load_advisory_windows(offline=True))
+    target = next((s for a in report.agents for s in a.mcp_servers
+                   if s.name == server_name), None)
+    if target is None:
+        click.echo(f"server {server_name!r} not in scan", err=True)
+        raise SystemExit(2)
+    env = attest_scanned_server(target, verdict, bytes.fromhex(key_hex),
+                                key_id=key_id,
+                                tenant_id=resolve_mcp_tenant_id())
+    text = json.dumps(env, indent=2)
+    if output:
+        Path(output).write_text(text)
+        click.echo(f"wrote {output}")
+    else:
+        click.echo(text)
+
+
+@main.command(name="quickstart")
+def quickstart_cmd() -> None:
+    """Guided first steps."""
+    click.echo("""agent-bom quickstart
+====================
+1. Demo scan (exits 1 by design — the demo estate is vulnerable):
+     agent-bom agents --demo --offline
+2. Scan YOUR machine's agents + MCP servers:
+     agent-bom agents --offline -f json -o scan.json
+3. One package:            agent-bom check pyyaml@5.3 -e pypi
+4. Trust signals:          agent-bom trust left-pad -e npm
+5. Fix commands:           agent-bom remediate --demo --script
+6. Posture (MCP + self):   agent-bom posture --demo --a2a
+7. Serve REST + MCP:       agent-bom serve   |   agent-bom mcp server
+8. Watch a repo:           agent-bom watch --filesystem . --interval 30
+9. Health check:           agent-bom doctor
+Docs: docs/OPERATIONS.md · docs/ARCHITECTURE.md · docs/PERFORMANCE.md""")
+
+
 @main.command(name="watch")
 @click.option("--filesystem", type=click.Path(exists=True), default=".",
               show_default=True, help="Tree to watch for manifest changes.")

@@ -217,3 +217,67 @@ class TestMcpResourcesPrompts:
         out = s.handle({"jsonrpc": "2.0", "id": 1, "method": "prompts/get",
                         "params": {"name": "investigate-cve"}})
         assert "error" in out
+
+
+class TestFocusedCliCommands:
+    def _run(self, args, env=None):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        return CliRunner().invoke(main, args, env=env)
+
+    def test_check_hit_and_clean(self):
+        out = self._run(["check", "pyyaml@5.3", "-e", "pypi", "--offline"])
+        assert out.exit_code == 1
+        import json as _json
+
+        doc = _json.loads(out.output)
+        assert doc["vulnerable"] and any(
+            a["vuln_id"] == "CVE-2020-14343" for a in doc["advisories"])
+        clean = self._run(["check", "totally-fine@9.9", "-e", "pypi",
+                           "--offline"])
+        assert clean.exit_code == 0
+
+    def test_doctor(self):
+        out = self._run(["doctor"])
+        import json as _json
+
+        doc = _json.loads(out.output)
+        names = {c["check"] for c in doc["checks"]}
+        assert {"python", "torch", "gpu", "hip_engine", "advisories",
+                "mcp_registry", "self_posture"} <= names
+
+    def test_trust(self):
+        out = self._run(["trust", "reqeusts", "-e", "pypi"])
+        assert out.exit_code == 1  # malicious -> F
+        ok = self._run(["trust", "@modelcontextprotocol/server-memory",
+                        "-e", "npm"])
+        assert ok.exit_code == 0
+
+    def test_remediate_script(self):
+        out = self._run(["remediate", "--demo", "--script"])
+        assert out.exit_code == 0
+        assert out.output.startswith("#!/bin/sh")
+
+    def test_attest_sign_and_verify(self, tmp_path):
+        key = "ab" * 32
+        env = {"AGENT_BOM_ATTESTATION_KEY": key}
+        # find a demo server name
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        name = run_demo_scan().agents[0].mcp_servers[0].name
+        out = self._run(["attest", name, "--demo", "-o",
+                         str(tmp_path / "env.json")], env=env)
+        assert out.exit_code == 0, out.output
+        ver = self._run(["attest", "x", "--verify",
+                         str(tmp_path / "env.json")], env=env)
+        assert ver.exit_code == 0, ver.output
+        bad = self._run(["attest", "x", "--verify",
+                         str(tmp_path / "env.json")],
+                        env={"AGENT_BOM_ATTESTATION_KEY": "cd" * 32})
+        assert bad.exit_code == 1
+
+    def test_quickstart(self):
+        out = self._run(["quickstart"])
+        assert out.exit_code == 0 and "Demo scan" in out.output
