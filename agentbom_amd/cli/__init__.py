@@ -575,6 +575,115 @@ def attest_cmd(server_name: str, demo: bool, verdict: str,
         click.echo(text)
 
 
+@main.command(name="skills")
+@click.argument("path", type=click.Path(exists=True))
+@click.option("--policy", type=click.Path(exists=True), default=None,
+              help="JSON policy: blocklist / max_risk_level / require_frontmatter.")
+def skills_cmd(path: str, policy: Optional[str]) -> None:
+    """Scan agent skill bundles (SKILL.md) for injection/grants/scripts."""
+    from agentbom_amd.scan.skills import evaluate_skills_policy, scan_skills_tree
+
+    bundles = scan_skills_tree(path)
+    out = {"bundles": [b.to_dict() for b in bundles]}
+    denied = 0
+    if policy:
+        decision = evaluate_skills_policy(
+            bundles, json.loads(Path(policy).read_text()))
+        out["policy"] = decision
+        denied = decision["denied"]
+    click.echo(json.dumps(out, indent=2))
+    critical = any(b.risk_level == "critical" for b in bundles)
+    raise SystemExit(1 if critical or denied else 0)
+
+
+@main.group(name="identity")
+def identity_group() -> None:
+    """Agent identity lifecycle (local store)."""
+
+
+def _identity_store_from_opt(store_path: str):
+    from agentbom_amd.identity import AgentIdentityStore
+
+    return AgentIdentityStore(store_path)
+
+
+_STORE_OPT = click.option(
+    "--store", default="~/.agent-bom/identities.db", show_default=True,
+    help="SQLite identity store path.")
+
+
+@identity_group.command(name="issue")
+@click.argument("agent_name")
+@click.option("--scopes", default="", help="Comma-separated scopes.")
+@click.option("--allowed-tools", default="")
+@click.option("--ttl-hours", type=float, default=24.0, show_default=True)
+@_STORE_OPT
+def identity_issue_cmd(agent_name: str, scopes: str, allowed_tools: str,
+                       ttl_hours: float, store: str) -> None:
+    """Issue a scoped identity; the raw token prints EXACTLY ONCE."""
+    s = _identity_store_from_opt(str(Path(store).expanduser()))
+    ident, raw = s.issue(
+        agent_name,
+        scopes=[x for x in scopes.split(",") if x.strip()],
+        allowed_tools=[x for x in allowed_tools.split(",") if x.strip()],
+        ttl_hours=ttl_hours, actor="cli", reason="cli issue")
+    click.echo(json.dumps({"identity": ident.to_public_dict(),
+                           "token": raw,
+                           "note": "token shown once; only its hash is stored"},
+                          indent=2))
+
+
+@identity_group.command(name="list")
+@_STORE_OPT
+def identity_list_cmd(store: str) -> None:
+    s = _identity_store_from_opt(str(Path(store).expanduser()))
+    click.echo(json.dumps(
+        [i.to_public_dict() for i in s.list()], indent=2))
+
+
+@identity_group.command(name="revoke")
+@click.argument("identity_id")
+@click.option("--reason", default="cli revocation", show_default=True)
+@_STORE_OPT
+def identity_revoke_cmd(identity_id: str, reason: str, store: str) -> None:
+    s = _identity_store_from_opt(str(Path(store).expanduser()))
+    ok = s.revoke(identity_id, actor="cli", reason=reason)
+    click.echo(json.dumps({"revoked": ok}))
+    raise SystemExit(0 if ok else 1)
+
+
+@identity_group.command(name="verify")
+@click.argument("token")
+@click.option("--tool", default=None)
+@_STORE_OPT
+def identity_verify_cmd(token: str, tool: Optional[str], store: str) -> None:
+    s = _identity_store_from_opt(str(Path(store).expanduser()))
+    out = s.verify(token, tool=tool)
+    click.echo(json.dumps(out, indent=2))
+    raise SystemExit(0 if out["valid"] else 1)
+
+
+@main.command(name="history")
+@click.option("--save", "save_report", type=click.Path(exists=True),
+              default=None, help="Save a report JSON as a snapshot.")
+@click.option("--diff", "diff_pair", nargs=2, type=click.Path(exists=True),
+              default=None, help="Diff two report JSON files (old new).")
+def history_cmd(save_report: Optional[str],
+                diff_pair: Optional[tuple]) -> None:
+    """Report snapshot history: save, list, diff."""
+    from agentbom_amd.scan.history import diff_reports, list_snapshots, save_report_snapshot
+
+    if save_report:
+        path = save_report_snapshot(json.loads(Path(save_report).read_text()))
+        click.echo(f"saved {path}")
+        return
+    if diff_pair:
+        old, new = (json.loads(Path(p).read_text()) for p in diff_pair)
+        click.echo(json.dumps(diff_reports(old, new), indent=2, default=str))
+        return
+    click.echo(json.dumps([str(p) for p in list_snapshots()], indent=2))
+
+
 @main.command(name="quickstart")
 def quickstart_cmd() -> None:
     """Guided first steps."""
