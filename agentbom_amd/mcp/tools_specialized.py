@@ -467,3 +467,64 @@ def register_specialized_tools(server) -> None:  # noqa: C901 — one registrar
                 "by_level": dict(defaultdict(int, {
                     lv: sum(1 for i in ingested if i["level"] == lv)
                     for lv in {i["level"] for i in ingested}}))}
+
+    @tool("threat_intel_search",
+          "Search the local threat-intel IOC store (packages, CVEs, domains, "
+          "hashes, actors; reference: intel_lookup.py). Air-gapped: bundled "
+          "feed + AGENT_BOM_INTEL_DB overlay only, no egress.",
+          {"type": "object", "properties": {
+              "query": {"type": "string"},
+              "kind": {"type": "string",
+                       "enum": ["auto", "package", "cve", "domain", "hash",
+                                "actor"], "default": "auto"}},
+           "required": ["query"]})
+    def threat_intel_search(query: str, kind: str = "auto") -> dict:
+        from agentbom_amd.scan.intel import IntelStore
+
+        store = IntelStore.load()
+        hits = []
+        if kind == "auto":
+            q = query.lower()
+            exact = (store.lookup_cve(q) if q.startswith("cve-")
+                     else store.lookup("package", q))
+            if exact is not None:
+                hits.append(exact)
+            hits.extend(i for i in store.search(query) if i is not exact)
+        else:
+            exact = store.lookup(kind, query)
+            hits = [exact] if exact is not None else []
+        return {"query": query, "kind": kind,
+                "indicators": [h.to_dict() for h in hits[:20]],
+                "store_size": len(store)}
+
+    @tool("vendor_advisory_check",
+          "Match packages against vendor security bulletins (AMD PSIRT / "
+          "NVIDIA CSAF / Intel) from the local feed — sources OSV never "
+          "carries. Tiers: vendor_range (bounded) / vendor_prefix.",
+          {"type": "object", "properties": {
+              "packages": {"type": "array", "items": {
+                  "type": "object", "properties": {
+                      "ecosystem": {"type": "string"},
+                      "name": {"type": "string"},
+                      "version": {"type": "string"}},
+                  "required": ["name", "version"]}}},
+           "required": ["packages"]})
+    def vendor_advisory_check(packages: list) -> dict:
+        from agentbom_amd.scan.vendor_advisories import (
+            check_vendor_advisories, vendor_for_package)
+
+        tuples = [(str(p.get("ecosystem") or "PyPI"), str(p.get("name") or ""),
+                   str(p.get("version") or "")) for p in packages[:500]
+                  if isinstance(p, dict)]
+        hits = check_vendor_advisories(tuples)
+        return {"checked": len(tuples),
+                "vendor_stack": {f"{e}:{n}": vendor_for_package(n)
+                                 for (e, n, _v) in tuples
+                                 if vendor_for_package(n)},
+                "matches": {f"{e}:{n}@{v}": [
+                    {"id": x.id, "severity": x.severity.value,
+                     "tier": x.match_confidence_tier,
+                     "fixed_version": x.fixed_version,
+                     "sources": x.advisory_sources}
+                    for x in vulns]
+                    for (e, n, v), vulns in hits.items()}}

@@ -106,6 +106,7 @@ class ScanOptions:
     fail_on_kev: bool = False
     use_gpu: Optional[bool] = None  # None = auto
     ignore_ids: frozenset = frozenset()
+    vendor_advisories: bool = True  # AMD PSIRT / NVIDIA CSAF / Intel feeds
 
 
 def _match_packages(
@@ -242,6 +243,21 @@ def scan_agents(
         for pkg in pkg_refs[key]:
             if all(v.id != vuln.id for v in pkg.vulnerabilities):
                 pkg.vulnerabilities.append(vuln)
+
+    # vendor-bulletin supplemental pass (AMD PSIRT / NVIDIA CSAF / Intel):
+    # bulletins that never reach OSV, matched fail-closed and tiered
+    # (reference: scanners/{amd,nvidia,ghsa}_advisory.py supplemental joins)
+    if options.vendor_advisories:
+        from agentbom_amd.scan.vendor_advisories import check_vendor_advisories
+
+        for key, extra in check_vendor_advisories(unique).items():
+            for vuln in extra:
+                if vuln.id in options.ignore_ids:
+                    continue
+                for pkg in pkg_refs[key]:
+                    if all(v.id != vuln.id for v in pkg.vulnerabilities):
+                        pkg.vulnerabilities.append(vuln)
+    _mark("vendor_advisories")
 
     # malicious packages surface as synthetic advisory rows (fail closed)
     for pkg in all_pkgs:
