@@ -121,6 +121,7 @@ def _scan_impl(
     code: Optional[str] = None, iac: Optional[str] = None,
     aws_inventory: Optional[str] = None, endpoint: bool = False,
     notebooks: Optional[str] = None, skills: Optional[str] = None,
+    semgrep: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -203,14 +204,25 @@ def _scan_impl(
         run_scanner_driver("skills", report, skills)
     if filesystem:
         run_scanner_driver("floating_refs", report, filesystem)
-    if code:
+    if code or semgrep:
         from agentbom_amd.scan.ast_analysis import (
+            SymbolIndex,
             apply_symbol_reachability,
             ast_finding_to_finding,
             build_symbol_index,
         )
 
-        idx = build_symbol_index(code)
+        idx = build_symbol_index(code) if code else SymbolIndex()
+        if semgrep:
+            from agentbom_amd.scan.sast_ingest import (
+                extend_symbol_index_from_semgrep,
+                load_semgrep_file,
+                semgrep_to_findings,
+            )
+
+            sg = load_semgrep_file(semgrep)
+            report.findings.extend(semgrep_to_findings(sg))
+            extend_symbol_index_from_semgrep(idx, sg)
         report.findings.extend(ast_finding_to_finding(f) for f in idx.findings)
         apply_symbol_reachability(report, idx)
         report.ai_inventory_data = (report.ai_inventory_data or {}) | {
@@ -289,6 +301,9 @@ def _scan_options(f):
                      help="Scan Jupyter notebooks (pip installs, secrets in outputs, sinks)."),
         click.option("--skills", type=click.Path(exists=True), default=None,
                      help="Scan agent skill bundles (SKILL.md injection/grants/scripts)."),
+        click.option("--semgrep", type=click.Path(exists=True), default=None,
+                     help="Ingest a semgrep --json result file (SAST findings "
+                          "+ symbol-reachability joins)."),
     ]
     for o in reversed(opts):
         f = o(f)

@@ -435,13 +435,26 @@ def register_specialized_tools(server) -> None:  # noqa: C901 — one registrar
         return {"ingested": len(rows), "kind": kind,
                 "session_totals": {k: len(v) for k, v in store.items()}}
 
-    @tool("ingest_external_scan", "Ingest a third-party SARIF file and merge "
-                                  "its results into the finding view.",
+    @tool("ingest_external_scan", "Ingest a third-party scan result file "
+                                  "(SARIF, or semgrep --json, auto-detected) "
+                                  "and merge its results into the finding view.",
           {"type": "object", "properties": {"sarif_path": {"type": "string"}},
            "required": ["sarif_path"]})
     def ingest_external_scan(sarif_path: str) -> dict:
         doc = json.loads(Path(sarif_path).read_text())
         ingested = []
+        from agentbom_amd.scan.sast_ingest import (
+            looks_like_semgrep, parse_semgrep_json)
+
+        if looks_like_semgrep(doc):
+            for r in parse_semgrep_json(doc):
+                ingested.append({
+                    "source_tool": "semgrep",
+                    "rule_id": r.check_id,
+                    "level": {"high": "error", "medium": "warning",
+                              "low": "note"}[r.severity],
+                    "message": r.message[:300],
+                    "location": f"{r.path}:{r.line}"})
         for run in doc.get("runs", []):
             tool_name = (run.get("tool", {}).get("driver", {})
                          .get("name", "external"))
