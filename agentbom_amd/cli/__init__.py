@@ -121,7 +121,7 @@ def _scan_impl(
     code: Optional[str] = None, iac: Optional[str] = None,
     aws_inventory: Optional[str] = None, endpoint: bool = False,
     notebooks: Optional[str] = None, skills: Optional[str] = None,
-    semgrep: Optional[str] = None,
+    semgrep: Optional[str] = None, cloud_inventory: tuple = (),
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -194,6 +194,8 @@ def _scan_impl(
         run_scanner_driver("model_files", report, model_files)
     if aws_inventory:
         run_scanner_driver("cloud_cis", report, aws_inventory)
+    for spec in cloud_inventory:
+        run_scanner_driver("cloud_estate", report, spec)
     if endpoint:
         run_scanner_driver("endpoint", report)
     if iac:
@@ -304,6 +306,10 @@ def _scan_options(f):
         click.option("--semgrep", type=click.Path(exists=True), default=None,
                      help="Ingest a semgrep --json result file (SAST findings "
                           "+ symbol-reachability joins)."),
+        click.option("--cloud-inventory", "cloud_inventory", multiple=True,
+                     help="provider:path — evaluate an exported cloud "
+                          "inventory (azure/gcp/snowflake/databricks/aws; "
+                          "includes IAM, audit-trail and DSPM sections)."),
     ]
     for o in reversed(opts):
         f = o(f)

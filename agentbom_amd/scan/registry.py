@@ -173,6 +173,23 @@ def _run_cloud_cis(report: AIBOMReport, target: str) -> int:
     return len(cis)
 
 
+def _run_cloud_estate(report: AIBOMReport, target: str) -> int:
+    """target = 'provider:path' (azure/gcp/snowflake/databricks/aws) — the
+    multi-cloud inventory evaluator incl. IAM / audit-trail / DSPM sections."""
+    from agentbom_amd.scan.cloud import cis_result_to_finding
+    from agentbom_amd.scan.cloud_estate import scan_cloud_estate
+
+    provider, _, path = target.partition(":")
+    if not path:
+        provider, path = "aws", provider
+    cis = scan_cloud_estate(path, provider)
+    rows = report.extra_data.setdefault("cis_benchmark_data", [])
+    rows.extend(r.to_dict() for r in cis)
+    report.findings.extend(
+        f for f in (cis_result_to_finding(r, provider) for r in cis) if f)
+    return len(cis)
+
+
 def _run_endpoint(report: AIBOMReport, target: Optional[str]) -> int:
     from agentbom_amd.scan.endpoint import collect_endpoint_inventory
 
@@ -216,6 +233,8 @@ def _ensure_builtins() -> None:
         "iac", "iac", _run_iac, failure_mode=WARN_CONTINUE))
     register_scanner(ScannerRegistration(
         "cloud_cis", "cloud_cis", _run_cloud_cis, failure_mode=FAIL_CLOSED))
+    register_scanner(ScannerRegistration(
+        "cloud_estate", "cloud_cis", _run_cloud_estate, failure_mode=FAIL_CLOSED))
     register_scanner(ScannerRegistration(
         "endpoint", "endpoint", _run_endpoint, failure_mode=SKIP,
         requires_target=False))
