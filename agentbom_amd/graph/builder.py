@@ -24,7 +24,15 @@ def _n(nid: str, et: EntityType, label: str, layer: GraphSemanticLayer,
 
 def build_unified_graph_from_report(report: AIBOMReport,
                                     node_budget: int = 250_000) -> UnifiedGraph:
-    g = UnifiedGraph(node_budget=node_budget)
+    # estate-scale builds spill node payloads to a SQLite workspace
+    # (reference auto-wire: store_backed.py for >= 5,000 entities)
+    from agentbom_amd.graph.store_backed import maybe_store_backed
+
+    expected = sum(
+        1 + len(a.mcp_servers) + sum(len(s.packages) + len(s.tools)
+                                     for s in a.mcp_servers)
+        for a in report.agents) + len(report.blast_radii)
+    g = maybe_store_backed(expected, node_budget=node_budget)
     provider_id = "provider:local"
     g.add_node(_n(provider_id, EntityType.PROVIDER, "local", GraphSemanticLayer.INFRA))
 

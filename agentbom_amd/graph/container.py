@@ -322,6 +322,27 @@ class UnifiedGraph:
         }
         return sorted(degrees.items(), key=lambda kv: (-kv[1], kv[0]))[:top_n]
 
+    def pagerank(self, damping: float = 0.85, top_n: Optional[int] = None,
+                 backend: Optional[str] = None) -> dict[str, float]:
+        """Influence ranking via the pluggable analytics backend
+        (graph/backend.py: native numpy by default, networkx when asked
+        for and installed — reference graph_backend.py:25)."""
+        from agentbom_amd.graph.backend import get_backend
+
+        ranks = get_backend(backend).pagerank(self, damping=damping)
+        if top_n is not None:
+            top = sorted(ranks.items(), key=lambda kv: (-kv[1], kv[0]))[:top_n]
+            return dict(top)
+        return ranks
+
+    def betweenness(self, sample: int = 64,
+                    backend: Optional[str] = None) -> dict[str, float]:
+        """Brandes betweenness through the analytics backend (exact when
+        sample >= |V|); `bottlenecks()` remains the cheap approximation."""
+        from agentbom_amd.graph.backend import get_backend
+
+        return get_backend(backend).betweenness(self, sample=sample)
+
     def bottlenecks(self, top_n: int = 10, sample: int = 50) -> list[tuple[str, float]]:
         """Approximate betweenness via BFS from a deterministic node sample."""
         import itertools
