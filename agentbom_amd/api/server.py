@@ -87,7 +87,8 @@ def create_app() -> FastAPI:
     if api_key and api_key not in key_roles:
         key_roles[api_key] = "admin"  # legacy single-key = admin
 
-    _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet")
+    _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
+                       "/scim", "/v1/delegation-tokens")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -100,14 +101,58 @@ def create_app() -> FastAPI:
             return not any(path.startswith(p) for p in _WRITE_PREFIXES)
         return False  # viewer: reads only
 
+    from agentbom_amd.api.auth import (
+        AuthError,
+        DelegationTokens,
+        QuotaTracker,
+        ScimUserStore,
+        role_from_claims,
+        verify_oidc_bearer,
+    )
+
+    state.delegation = DelegationTokens()
+    state.quotas = QuotaTracker()
+    state.scim_users = ScimUserStore()
+    oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET"))
+
+    def _resolve_role(request: Request, x_api_key: Optional[str]) -> Optional[str]:
+        """API key (static table or SCIM-bound) > OIDC bearer > delegation
+        token.  Returns None when no credential resolves."""
+        if x_api_key:
+            role = key_roles.get(x_api_key) or state.scim_users.role_for_key(x_api_key)
+            if role:
+                request.state.principal = f"key:{x_api_key[:8]}"
+                return role
+        authz = request.headers.get("Authorization") or ""
+        if oidc_enabled and authz.startswith("Bearer "):
+            try:
+                claims = verify_oidc_bearer(authz[7:])
+            except AuthError:
+                return None
+            request.state.principal = f"oidc:{claims.get('sub', '?')}"
+            return role_from_claims(claims)
+        dtok = request.headers.get("X-Delegation-Token")
+        if dtok and state.delegation.enabled:
+            try:
+                payload = state.delegation.verify(dtok)
+            except AuthError:
+                return None
+            request.state.principal = f"delegated:{payload.get('jti', '?')[:8]}"
+            request.state.tenant_id = payload.get("tenant_id", "default")
+            return payload.get("role", "viewer")
+        return None
+
     def auth(request: Request,
              x_api_key: Optional[str] = Header(default=None)) -> None:
-        if not key_roles:
-            return  # auth disabled (no keys configured)
-        role = key_roles.get(x_api_key or "")
+        if not key_roles and not oidc_enabled and not state.delegation.enabled:
+            request.state.role = "admin"
+            request.state.principal = "anonymous"
+            return  # auth disabled (nothing configured)
+        role = _resolve_role(request, x_api_key)
         if role is None:
             state.metrics["auth_failures_total"] += 1
-            raise HTTPException(status_code=401, detail="invalid or missing API key")
+            raise HTTPException(status_code=401,
+                                detail="invalid or missing credential")
         if not _role_allows(role, request.method, request.url.path):
             state.metrics["auth_failures_total"] += 1
             raise HTTPException(
@@ -215,7 +260,13 @@ def create_app() -> FastAPI:
             state.metrics["scan_failures_total"] += 1
 
     @app.post("/v1/scan", status_code=201, dependencies=[Depends(auth)])
-    def submit_scan(req: ScanRequest) -> dict:
+    def submit_scan(req: ScanRequest, request: Request) -> dict:
+        principal = getattr(request.state, "principal", "anonymous")
+        ok, retry = state.quotas.check_and_record(principal)
+        if not ok:
+            raise HTTPException(status_code=429,
+                                detail="scan quota exceeded for this principal",
+                                headers={"Retry-After": str(int(retry))})
         admitted, retry_after = state.backpressure.try_acquire()
         if not admitted:
             from fastapi.responses import JSONResponse
@@ -795,6 +846,97 @@ def create_app() -> FastAPI:
         if not _identity_store().revoke_jit(grant_id, actor="api", reason=reason):
             raise HTTPException(status_code=404, detail="grant not found")
         return {"revoked": True}
+
+    # ── delegation tokens + SCIM provisioning (reference: api/{scim,oidc}.py,
+    # delegation tokens in middleware; SURVEY §2.6 auth/tenancy row) ───────
+
+    @app.post("/v1/delegation-tokens", status_code=201, dependencies=[Depends(auth)])
+    def mint_delegation_token(payload: dict) -> dict:
+        if not state.delegation.enabled:
+            raise HTTPException(status_code=400,
+                                detail="AGENT_BOM_DELEGATION_SECRET not configured")
+        try:
+            token = state.delegation.mint(
+                role=str(payload.get("role") or "viewer"),
+                scopes=[str(s) for s in (payload.get("scopes") or [])],
+                ttl_s=float(payload.get("ttl_s", 3600)),
+                tenant_id=str(payload.get("tenant_id") or "default"))
+        except AuthError as exc:
+            raise HTTPException(status_code=400, detail=str(exc))
+        except (TypeError, ValueError) as exc:
+            raise HTTPException(status_code=422, detail=str(exc))
+        return {"token": token, "note": "store securely; shown once"}
+
+    @app.delete("/v1/delegation-tokens/{jti}", dependencies=[Depends(auth)])
+    def revoke_delegation_token(jti: str) -> dict:
+        state.delegation.revoke(jti)
+        return {"revoked": jti}
+
+    @app.get("/scim/v2/ServiceProviderConfig", dependencies=[Depends(auth)])
+    def scim_config() -> dict:
+        return {
+            "schemas": ["urn:ietf:params:scim:schemas:core:2.0:ServiceProviderConfig"],
+            "patch": {"supported": True}, "bulk": {"supported": False},
+            "filter": {"supported": False}, "sort": {"supported": False},
+            "authenticationSchemes": [{"type": "httpheader",
+                                       "name": "X-API-Key"}],
+        }
+
+    @app.get("/scim/v2/Users", dependencies=[Depends(auth)])
+    def scim_list_users() -> dict:
+        users = state.scim_users.list()
+        return {"schemas": ["urn:ietf:params:scim:api:messages:2.0:ListResponse"],
+                "totalResults": len(users),
+                "Resources": [u.to_scim() for u in users]}
+
+    @app.post("/scim/v2/Users", status_code=201, dependencies=[Depends(auth)])
+    def scim_create_user(payload: dict) -> dict:
+        try:
+            return state.scim_users.create(payload).to_scim()
+        except AuthError as exc:
+            raise HTTPException(status_code=409, detail=str(exc))
+
+    @app.get("/scim/v2/Users/{uid}", dependencies=[Depends(auth)])
+    def scim_get_user(uid: str) -> dict:
+        u = state.scim_users.get(uid)
+        if u is None:
+            raise HTTPException(status_code=404, detail="user not found")
+        return u.to_scim()
+
+    @app.patch("/scim/v2/Users/{uid}", dependencies=[Depends(auth)])
+    def scim_patch_user(uid: str, payload: dict) -> dict:
+        active: Optional[bool] = None
+        for op in payload.get("Operations") or []:
+            if not isinstance(op, dict):
+                continue
+            if str(op.get("op", "")).lower() == "replace":
+                val = op.get("value")
+                if isinstance(val, dict) and "active" in val:
+                    active = bool(val["active"])
+                elif str(op.get("path", "")).lower() == "active":
+                    active = bool(val) if not isinstance(val, str) \
+                        else val.lower() == "true"
+        if active is None:
+            raise HTTPException(status_code=422,
+                                detail="only replace-active is supported")
+        u = state.scim_users.set_active(uid, active)
+        if u is None:
+            raise HTTPException(status_code=404, detail="user not found")
+        return u.to_scim()
+
+    @app.delete("/scim/v2/Users/{uid}", status_code=204, dependencies=[Depends(auth)])
+    def scim_delete_user(uid: str) -> None:
+        if not state.scim_users.delete(uid):
+            raise HTTPException(status_code=404, detail="user not found")
+
+    @app.post("/scim/v2/Users/{uid}/api-key", status_code=201,
+              dependencies=[Depends(auth)])
+    def scim_bind_key(uid: str) -> dict:
+        """Mint + bind an API key to a SCIM user (hash-only stored)."""
+        key = f"abk_{uuid.uuid4().hex}"
+        if not state.scim_users.bind_key(uid, key):
+            raise HTTPException(status_code=404, detail="user not found")
+        return {"api_key": key, "note": "store securely; shown once"}
 
     return app
 
