@@ -110,6 +110,34 @@ def _render(report, fmt: str, output: Optional[str], verbose: bool) -> None:
         click.echo(text, nl=False)
 
 
+def _apply_profile(kw: dict) -> dict:
+    """Merge a named .agent-bom.yaml profile into the scan kwargs.
+
+    Only parameters the user left at their DEFAULT are overridden — an
+    explicit CLI flag always wins over the profile (reference:
+    cli/_profiles.py precedence)."""
+    name = kw.pop("profile", None)
+    if not name:
+        return kw
+    from agentbom_amd.utils.project_config import get_profile, load_project_config
+
+    try:
+        body = get_profile(load_project_config("."), name)
+    except ValueError as exc:
+        raise click.UsageError(str(exc))
+    ctx = click.get_current_context(silent=True)
+    for key, value in body.items():
+        param = "fmt" if key == "format" else key.replace("-", "_")
+        if param not in kw:
+            raise click.UsageError(
+                f"profile {name!r}: unknown flag {key!r}")
+        src = ctx.get_parameter_source(param) if ctx else None
+        if src is None or src.name == "DEFAULT":
+            kw[param] = tuple(value) if isinstance(kw[param], tuple) and \
+                isinstance(value, list) else value
+    return kw
+
+
 def _scan_impl(
     demo: bool, offline: bool, inventory: Optional[str], fmt: str,
     output: Optional[str], fail_on_severity: str, fail_on_kev: bool,
@@ -310,6 +338,9 @@ def _scan_options(f):
                      help="provider:path — evaluate an exported cloud "
                           "inventory (azure/gcp/snowflake/databricks/aws; "
                           "includes IAM, audit-trail and DSPM sections)."),
+        click.option("--profile", "profile", default=None,
+                     help="Apply a named flag-set from .agent-bom.yaml "
+                          "profiles: (explicit flags still win)."),
     ]
     for o in reversed(opts):
         f = o(f)
@@ -320,14 +351,14 @@ def _scan_options(f):
 @_scan_options
 def agents_cmd(**kw) -> None:
     """Discover AI agents + MCP servers and scan their packages."""
-    _scan_impl(**{k if k != "fmt" else "fmt": v for k, v in kw.items()})
+    _scan_impl(**_apply_profile(kw))
 
 
 @main.command(name="scan", hidden=True)
 @_scan_options
 def scan_cmd(**kw) -> None:
     """Alias of ``agents``."""
-    _scan_impl(**kw)
+    _scan_impl(**_apply_profile(kw))
 
 
 @main.command(name="serve")

@@ -39,6 +39,7 @@ IGNORE_FILENAME = ".agent-bom-ignore"
 _KNOWN_KEYS = {
     "offline", "include_unfixed", "fail_on_severity", "fail_on_kev",
     "blast_radius_depth", "exit_zero", "policy", "ignore", "use_gpu",
+    "profiles",
 }
 
 _RULE_RE = re.compile(r"^(?P<id>\S+)(?P<rest>.*)$")
@@ -163,3 +164,26 @@ def apply_to_scan_options(cfg: ProjectConfig, options) -> None:
     if "use_gpu" in s:
         options.use_gpu = bool(s["use_gpu"])
     options.ignore_ids = frozenset(options.ignore_ids) | cfg.active_ignore_ids
+
+
+def get_profile(cfg: ProjectConfig, name: str) -> dict[str, Any]:
+    """Named CLI flag-set from the project file (reference: cli/_profiles.py).
+
+    .agent-bom.yaml::
+
+        profiles:
+          ci:       {offline: true, fail_on_severity: high, format: sarif}
+          deep:     {blast_radius_depth: 3, include_unfixed: true}
+
+    Unknown profile -> ValueError listing what exists; non-mapping entries
+    are rejected the same way (never silently empty)."""
+    profiles = cfg.settings.get("profiles") or {}
+    if not isinstance(profiles, dict):
+        raise ValueError(".agent-bom.yaml: profiles must be a mapping")
+    if name not in profiles:
+        raise ValueError(
+            f"unknown profile {name!r}; defined: {sorted(profiles) or '(none)'}")
+    body = profiles[name]
+    if not isinstance(body, dict):
+        raise ValueError(f"profile {name!r} must be a mapping of flags")
+    return dict(body)

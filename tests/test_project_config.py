@@ -1,5 +1,7 @@
 """Project config (.agent-bom.yaml) + ignore file tests."""
 
+import pytest
+
 from agentbom_amd.scan.orchestrator import ScanOptions
 from agentbom_amd.utils.project_config import (
     apply_to_scan_options,
@@ -70,3 +72,68 @@ class TestEndToEnd:
         report = run_demo_scan(opts)
         assert not any(br.vulnerability.id == "CVE-2020-14343"
                        for br in report.blast_radii)
+
+
+class TestCliProfiles:
+    """Named flag-sets in .agent-bom.yaml (reference: cli/_profiles.py)."""
+
+    def _write_cfg(self, tmp_path):
+        (tmp_path / ".agent-bom.yaml").write_text(
+            "profiles:\n"
+            "  ci:\n"
+            "    offline: true\n"
+            "    format: json\n"
+            "    exit_zero: true\n"
+            "  broken: [not, a, mapping]\n")
+
+    def test_get_profile(self, tmp_path, monkeypatch):
+        from agentbom_amd.utils.project_config import get_profile, load_project_config
+
+        self._write_cfg(tmp_path)
+        cfg = load_project_config(tmp_path)
+        assert get_profile(cfg, "ci")["format"] == "json"
+        with pytest.raises(ValueError, match="unknown profile"):
+            get_profile(cfg, "nope")
+        with pytest.raises(ValueError, match="must be a mapping"):
+            get_profile(cfg, "broken")
+
+    def test_cli_profile_applies_defaults(self, tmp_path, monkeypatch):
+        import json as _json
+
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        self._write_cfg(tmp_path)
+        monkeypatch.chdir(tmp_path)
+        out = tmp_path / "r.json"
+        res = CliRunner().invoke(main, [
+            "agents", "--demo", "--profile", "ci", "-o", str(out)])
+        assert res.exit_code in (0, 1), res.output
+        # profile switched the default console format to json
+        _json.loads(out.read_text())
+
+    def test_cli_explicit_flag_beats_profile(self, tmp_path, monkeypatch):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        self._write_cfg(tmp_path)
+        monkeypatch.chdir(tmp_path)
+        out = tmp_path / "r.txt"
+        res = CliRunner().invoke(main, [
+            "agents", "--demo", "--profile", "ci", "-f", "markdown",
+            "-o", str(out)])
+        assert res.exit_code in (0, 1), res.output
+        assert out.read_text().lstrip().startswith("#")  # markdown, not json
+
+    def test_unknown_profile_is_usage_error(self, tmp_path, monkeypatch):
+        from click.testing import CliRunner
+
+        from agentbom_amd.cli import main
+
+        self._write_cfg(tmp_path)
+        monkeypatch.chdir(tmp_path)
+        res = CliRunner().invoke(main, ["agents", "--demo", "--profile", "zz"])
+        assert res.exit_code == 2
+        assert "unknown profile" in res.output
