@@ -427,10 +427,20 @@ def register_specialized_tools(server) -> None:  # noqa: C901 — one registrar
     def runtime_evidence_ingest(path: str, kind: str = "audit") -> dict:
         from agentbom_amd.mcp.tools_operator import _load_jsonl
 
-        rows = _load_jsonl(path)
         store = getattr(server, "_evidence", None)
         if store is None:
             store = server._evidence = {"audit": [], "spans": []}
+        if kind == "spans":
+            # OTLP JSON / Langfuse exports / span JSONL all land here
+            from agentbom_amd.utils.otel_ingest import ingest_trace_file
+
+            result = ingest_trace_file(path)
+            store["spans"].append({"source": path, **result})
+            return {"ingested": result["spans"], "kind": kind,
+                    "activity": result["activity"],
+                    "anomalies": result["anomalies"],
+                    "session_totals": {k: len(v) for k, v in store.items()}}
+        rows = _load_jsonl(path)
         store[kind].extend(rows)
         return {"ingested": len(rows), "kind": kind,
                 "session_totals": {k: len(v) for k, v in store.items()}}
