@@ -169,6 +169,7 @@ class AIBOMReport:
     coverage_warnings: list[dict[str, Any]] = field(default_factory=list)
     warnings: list[str] = field(default_factory=list)
     scan_performance_data: Optional[dict[str, Any]] = None
+    intel_matches: Optional[list[dict[str, Any]]] = None
     vuln_data_freshness: Optional[dict[str, Any]] = None
     # Optional side blocks (subset of the reference's; extend as surfaces land)
     iac_findings_data: Optional[dict[str, Any]] = None

@@ -668,5 +668,6 @@ def to_json(report: AIBOMReport) -> dict[str, Any]:
         },
         "remediation_plan": _build_remediation_json(report),
         "scan_performance": report.scan_performance_data,
+        "intel_matches": report.intel_matches or [],
         "estate_score": _estate_score(report),
     }

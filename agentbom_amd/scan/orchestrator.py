@@ -107,6 +107,7 @@ class ScanOptions:
     use_gpu: Optional[bool] = None  # None = auto
     ignore_ids: frozenset = frozenset()
     vendor_advisories: bool = True  # AMD PSIRT / NVIDIA CSAF / Intel feeds
+    threat_intel: bool = True       # local IOC store enrichment
 
 
 def _match_packages(
@@ -345,6 +346,14 @@ def scan_agents(
     report.warnings = warnings
     report.scan_sources = ["agent_discovery"] if not options.demo else ["demo_inventory"]
     report.scan_run = ScanRun()
+    # local threat-intel pass: active-exploitation rescore + malicious
+    # indicators (reference: intel_lookup during enrichment) — runs BEFORE
+    # finding fusion so findings carry the boosted scores
+    if options.threat_intel:
+        from agentbom_amd.scan.intel import enrich_report_with_intel
+
+        enrich_report_with_intel(report)
+        _mark("threat_intel")
     report.findings = report.to_findings()
     _mark("finding_fusion")
     # counter-in-report contract (reference: scan_performance counters,

@@ -233,3 +233,20 @@ class TestIntelEnrichment:
         report = run_demo_scan()
         matches = enrich_report_with_intel(report, IntelStore())
         assert all(m["entity"] != "live" for m in matches)
+
+
+class TestOrchestratorIntelWiring:
+    def test_scan_runs_intel_stage_and_report_carries_matches(self):
+        report = run_demo_scan()
+        assert "threat_intel_ms" in report.scan_performance_data["stages"]
+        assert report.intel_matches is not None  # list, possibly empty
+
+    def test_json_contract_has_intel_matches(self):
+        from agentbom_amd.output.json_fmt import to_json
+
+        doc = to_json(run_demo_scan())
+        assert "intel_matches" in doc
+
+    def test_opt_out(self):
+        report = run_demo_scan(ScanOptions(demo=True, threat_intel=False))
+        assert "threat_intel_ms" not in report.scan_performance_data["stages"]
