@@ -24,6 +24,20 @@ from typing import Any, Optional
 
 from agentbom_amd.scan.cloud import CisCheckResult, _check
 
+
+def _rows(inv: Any, key: str) -> list[dict]:
+    """Tolerant list-of-dicts extraction: hostile/hand-edited inventory
+    documents may put anything under these keys (fuzz-found)."""
+    v = inv.get(key) if isinstance(inv, dict) else None
+    if not isinstance(v, list):
+        return []
+    return [r for r in v if isinstance(r, dict)]
+
+
+def _section(inv: Any, key: str) -> Optional[dict]:
+    v = inv.get(key) if isinstance(inv, dict) else None
+    return v if isinstance(v, dict) else None
+
 # ── Azure CIS subset ────────────────────────────────────────────────────────
 
 
@@ -34,7 +48,7 @@ def evaluate_azure_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
      "aad_users": [...], "ml_workspaces": [...], "activity_log": {...}}
     """
     results: list[CisCheckResult] = []
-    for sa in inv.get("storage_accounts", []) or []:
+    for sa in _rows(inv, "storage_accounts"):
         name = sa.get("name", "?")
         _check(results, "AZ-3.1", "Storage account denies public blob access", "high",
                f"storage:{name}", not sa.get("allowBlobPublicAccess", False))
@@ -42,34 +56,34 @@ def evaluate_azure_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
                f"storage:{name}", bool(sa.get("supportsHttpsTrafficOnly", True)))
         _check(results, "AZ-3.8", "Storage soft delete enabled", "low",
                f"storage:{name}", bool(sa.get("softDelete", True)))
-    for nsg in inv.get("nsgs", []) or []:
+    for nsg in _rows(inv, "nsgs"):
         name = nsg.get("name", "?")
         open_world = any(
             r.get("sourceAddressPrefix") in ("*", "0.0.0.0/0", "Internet")
             and str(r.get("destinationPortRange", "")) in ("22", "3389", "*")
             and r.get("access", "Allow") == "Allow"
-            for r in nsg.get("rules", []) or [])
+            for r in nsg.get("rules", []) or [] if isinstance(r, dict))
         _check(results, "AZ-6.1", "No admin ports open to the internet", "critical",
                f"nsg:{name}", not open_world,
                "SSH/RDP open to Internet" if open_world else "")
-    for sql in inv.get("sql_servers", []) or []:
+    for sql in _rows(inv, "sql_servers"):
         name = sql.get("name", "?")
         _check(results, "AZ-4.1", "SQL server auditing enabled", "medium",
                f"sql:{name}", bool(sql.get("auditingEnabled", True)))
         _check(results, "AZ-4.2", "SQL public network access disabled", "critical",
                f"sql:{name}", not sql.get("publicNetworkAccess", False))
-    for user in inv.get("aad_users", []) or []:
+    for user in _rows(inv, "aad_users"):
         name = user.get("userPrincipalName", "?")
         _check(results, "AZ-1.1", "MFA enabled for privileged users", "high",
                f"aad:{name}",
                bool(user.get("mfaEnabled", True)) or not user.get("privileged", False))
-    for ws in inv.get("ml_workspaces", []) or []:
+    for ws in _rows(inv, "ml_workspaces"):
         name = ws.get("name", "?")
         _check(results, "AIINF-1", "ML workspace endpoint requires auth", "critical",
                f"mlworkspace:{name}", bool(ws.get("authRequired", True)))
         _check(results, "AIINF-2", "ML workspace not internet-exposed", "high",
                f"mlworkspace:{name}", not ws.get("publicEndpoint", False))
-    log = inv.get("activity_log")
+    log = _section(inv, "activity_log")
     if log is not None:
         _check(results, "AZ-5.1", "Activity log export configured", "high",
                "activity-log", bool(log.get("exportEnabled", False)))
@@ -86,7 +100,7 @@ def evaluate_gcp_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
      "cloudsql": [...], "vertex_endpoints": [...], "audit_config": {...}}
     """
     results: list[CisCheckResult] = []
-    for b in inv.get("buckets", []) or []:
+    for b in _rows(inv, "buckets"):
         name = b.get("name", "?")
         public = any(m in ("allUsers", "allAuthenticatedUsers")
                      for m in b.get("iamMembers", []) or [])
@@ -95,14 +109,15 @@ def evaluate_gcp_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
                "allUsers/allAuthenticatedUsers granted" if public else "")
         _check(results, "GCP-5.2", "Bucket uniform access enabled", "medium",
                f"gcs:{name}", bool(b.get("uniformBucketLevelAccess", True)))
-    for fw in inv.get("firewalls", []) or []:
+    for fw in _rows(inv, "firewalls"):
         name = fw.get("name", "?")
         open_world = ("0.0.0.0/0" in (fw.get("sourceRanges") or []) and any(
-            str(p) in ("22", "3389", "all") for a in (fw.get("allowed") or [])
+            str(p) in ("22", "3389", "all")
+            for a in (fw.get("allowed") or []) if isinstance(a, dict)
             for p in (a.get("ports") or ["all"])))
         _check(results, "GCP-3.6", "No admin ports open to 0.0.0.0/0", "critical",
                f"firewall:{name}", not open_world)
-    for sa in inv.get("service_accounts", []) or []:
+    for sa in _rows(inv, "service_accounts"):
         email = sa.get("email", "?")
         _check(results, "GCP-1.5", "Service account has no Owner/Editor role", "high",
                f"sa:{email}",
@@ -110,16 +125,16 @@ def evaluate_gcp_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
                        for r in sa.get("roles", []) or []))
         _check(results, "GCP-1.4", "Service account keys rotated <= 90 days", "medium",
                f"sa:{email}", not sa.get("keyStale", False))
-    for db in inv.get("cloudsql", []) or []:
+    for db in _rows(inv, "cloudsql"):
         name = db.get("name", "?")
         _check(results, "GCP-6.5", "Cloud SQL not open to the world", "critical",
                f"cloudsql:{name}",
                "0.0.0.0/0" not in (db.get("authorizedNetworks") or []))
-    for ep in inv.get("vertex_endpoints", []) or []:
+    for ep in _rows(inv, "vertex_endpoints"):
         name = ep.get("name", "?")
         _check(results, "AIINF-1", "Vertex endpoint requires auth", "critical",
                f"vertex:{name}", bool(ep.get("authRequired", True)))
-    audit = inv.get("audit_config")
+    audit = _section(inv, "audit_config")
     if audit is not None:
         _check(results, "GCP-2.1", "Cloud audit logging configured", "high",
                "audit-config", bool(audit.get("enabled", False)))
@@ -131,18 +146,18 @@ def evaluate_gcp_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
 
 def evaluate_snowflake_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
     results: list[CisCheckResult] = []
-    for u in inv.get("users", []) or []:
+    for u in _rows(inv, "users"):
         name = u.get("name", "?")
         _check(results, "SF-1.1", "Snowflake user has MFA or key-pair auth", "high",
                f"sf-user:{name}",
                bool(u.get("mfaEnabled") or u.get("hasRsaKey")))
         _check(results, "SF-1.3", "ACCOUNTADMIN not used as default role", "high",
                f"sf-user:{name}", u.get("defaultRole") != "ACCOUNTADMIN")
-    for sh in inv.get("shares", []) or []:
+    for sh in _rows(inv, "shares"):
         name = sh.get("name", "?")
         _check(results, "SF-2.1", "Share restricted to named accounts", "medium",
                f"sf-share:{name}", not sh.get("public", False))
-    params = inv.get("account_parameters") or {}
+    params = _section(inv, "account_parameters") or {}
     if params:
         _check(results, "SF-3.1", "Network policy attached to account", "medium",
                "sf-account", bool(params.get("networkPolicy")))
@@ -151,18 +166,18 @@ def evaluate_snowflake_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
 
 def evaluate_databricks_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
     results: list[CisCheckResult] = []
-    for c in inv.get("clusters", []) or []:
+    for c in _rows(inv, "clusters"):
         name = c.get("name", "?")
         _check(results, "DBX-1.1", "Cluster not publicly reachable", "critical",
                f"dbx-cluster:{name}", not c.get("publicIp", False))
         _check(results, "DBX-1.2", "Credential passthrough / UC governance on", "medium",
                f"dbx-cluster:{name}",
                bool(c.get("unityCatalog") or c.get("credentialPassthrough")))
-    for t in inv.get("tokens", []) or []:
+    for t in _rows(inv, "tokens"):
         _check(results, "DBX-2.1", "PAT lifetime bounded", "medium",
                f"dbx-token:{t.get('comment', '?')}",
                t.get("lifetimeDays") is not None and t.get("lifetimeDays", 999) <= 90)
-    for sv in inv.get("model_serving", []) or []:
+    for sv in _rows(inv, "model_serving"):
         name = sv.get("name", "?")
         _check(results, "AIINF-1", "Model serving endpoint requires auth", "critical",
                f"dbx-serving:{name}", bool(sv.get("authRequired", True)))

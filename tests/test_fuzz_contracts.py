@@ -196,3 +196,83 @@ class TestGraphBuilderFuzz:
             pytest.fail(f"builder crashed on mutated doc keys "
                         f"{sorted(mutation)}: {exc!r}")
         assert graph.nodes is not None
+
+
+class TestNewSeamFuzz:
+    """Round-1 late additions parse operator/attacker-supplied files too:
+    cloud inventories, OTLP traces, vendor feeds, deps bundles, semgrep."""
+
+    @FAST
+    @given(inv=json_values)
+    def test_cloud_estate_evaluators_never_raise(self, inv):
+        from agentbom_amd.scan import cloud_estate as ce
+
+        doc = inv if isinstance(inv, dict) else {"buckets": inv}
+        for fn in (ce.evaluate_azure_inventory, ce.evaluate_gcp_inventory,
+                   ce.evaluate_snowflake_inventory,
+                   ce.evaluate_databricks_inventory):
+            try:
+                out = fn(doc)
+            except (AttributeError, TypeError, KeyError) as exc:
+                pytest.fail(f"{fn.__name__} crashed on {doc!r}: {exc!r}")
+            assert isinstance(out, list)
+
+    @FAST
+    @given(rows=st.lists(json_values, max_size=6))
+    def test_iam_and_audit_never_raise(self, rows):
+        from agentbom_amd.scan.cloud_estate import (
+            evaluate_iam_policies,
+            ingest_audit_trail,
+        )
+
+        assert isinstance(evaluate_iam_policies(rows), list)
+        assert isinstance(ingest_audit_trail(rows), list)
+
+    @FAST
+    @given(doc=json_values)
+    def test_otlp_and_langfuse_never_raise(self, doc):
+        from agentbom_amd.utils.otel_ingest import (
+            parse_langfuse_export,
+            parse_otlp_json,
+            summarize_agent_activity,
+        )
+
+        spans = parse_otlp_json(doc)
+        spans += parse_langfuse_export(doc)
+        assert isinstance(summarize_agent_activity(spans), dict)
+
+    @FAST
+    @given(doc=json_values)
+    def test_vendor_feed_and_csaf_never_raise(self, doc, tmp_path):
+        from agentbom_amd.scan.vendor_advisories import (
+            load_vendor_feed,
+            parse_csaf_document,
+        )
+
+        assert isinstance(parse_csaf_document(doc), list)
+        f = tmp_path / "feed.json"
+        f.write_text(json.dumps(doc))
+        assert isinstance(load_vendor_feed(str(f)), list)
+
+    @FAST
+    @given(doc=json_values)
+    def test_semgrep_parse_never_raises(self, doc):
+        from agentbom_amd.scan.sast_ingest import parse_semgrep_json
+
+        rows = parse_semgrep_json(doc)
+        assert isinstance(rows, list)
+
+    @FAST
+    @given(doc=json_values)
+    def test_deps_bundle_never_raises(self, doc, tmp_path):
+        from agentbom_amd.models.core import Package
+        from agentbom_amd.scan.deps_meta import (
+            enrich_packages_with_deps_meta,
+            load_deps_bundle,
+        )
+
+        f = tmp_path / "b.json"
+        f.write_text(json.dumps(doc))
+        bundle = load_deps_bundle(str(f))
+        p = Package(name="x", version="1", ecosystem="PyPI")
+        enrich_packages_with_deps_meta([p], bundle=bundle)
