@@ -551,3 +551,32 @@ def register_specialized_tools(server) -> None:  # noqa: C901 — one registrar
                      "sources": x.advisory_sources}
                     for x in vulns]
                     for (e, n, v), vulns in hits.items()}}
+
+    @tool("youcom_search",
+          "Live threat-intel web search (reference: the only third-party "
+          "egress tool). Requires AGENT_BOM_YOUCOM_KEY + an injectable "
+          "client; air-gapped deployments get an explicit refusal, never "
+          "a silent empty result.",
+          {"type": "object", "properties": {
+              "query": {"type": "string"},
+              "count": {"type": "integer", "default": 5}},
+           "required": ["query"]})
+    def youcom_search(query: str, count: int = 5) -> dict:
+        import os
+
+        key = os.environ.get("AGENT_BOM_YOUCOM_KEY")
+        client = getattr(server, "youcom_client", None)
+        if not key or client is None:
+            return {"error": "live search unavailable: set AGENT_BOM_YOUCOM_KEY "
+                             "and attach a client (air-gapped deployments use "
+                             "threat_intel_search against the local store)",
+                    "offline_alternative": "threat_intel_search",
+                    "results": []}
+        try:
+            rows = client(query, count=min(max(1, count), 10)) or []
+        except Exception as exc:  # network boundary — never crash the server
+            return {"error": f"live search failed: {exc}", "results": []}
+        return {"query": query,
+                "results": [{k: str(r.get(k, ""))[:300]
+                             for k in ("title", "url", "snippet")}
+                            for r in rows if isinstance(r, dict)][:10]}

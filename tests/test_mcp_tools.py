@@ -450,3 +450,42 @@ class TestToolCatalogRobustness:
             assert "result" in resp, name
             payload = resp["result"]["content"][0]["text"]
             json.loads(payload)  # always structured JSON
+
+
+class TestYoucomSearch:
+    """Live-search egress tool: gated off in air-gapped deployments with an
+    explicit refusal (reference: youcom_search, the only egress tool)."""
+
+    def _call(self, server, **args):
+        import json as _json
+
+        resp = server.handle({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                              "params": {"name": "youcom_search",
+                                         "arguments": args}})
+        return _json.loads(resp["result"]["content"][0]["text"])
+
+    def test_offline_refusal_names_alternative(self, monkeypatch):
+        from agentbom_amd.mcp.server import AgentBomMcpServer
+
+        monkeypatch.delenv("AGENT_BOM_YOUCOM_KEY", raising=False)
+        out = self._call(AgentBomMcpServer(), query="CVE-2021-44228")
+        assert out["results"] == []
+        assert out["offline_alternative"] == "threat_intel_search"
+
+    def test_injected_client_and_failure(self, monkeypatch):
+        from agentbom_amd.mcp.server import AgentBomMcpServer
+
+        monkeypatch.setenv("AGENT_BOM_YOUCOM_KEY", "k")
+        s = AgentBomMcpServer()
+        s.youcom_client = lambda q, count=5: [
+            {"title": "writeup", "url": "https://x", "snippet": "…"}, "junk"]
+        out = self._call(s, query="q", count=3)
+        assert out["results"] == [{"title": "writeup", "url": "https://x",
+                                   "snippet": "…"}]
+
+        def boom(q, count=5):
+            raise ConnectionError("egress blocked")
+
+        s.youcom_client = boom
+        out = self._call(s, query="q")
+        assert "failed" in out["error"] and out["results"] == []
