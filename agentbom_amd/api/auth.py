@@ -78,7 +78,9 @@ def verify_oidc_bearer(token: str,
         raise AuthError("signature mismatch")
     t = now if now is not None else time.time()
     exp = claims.get("exp")
-    if isinstance(exp, (int, float)) and t > exp + 60:
+    if not isinstance(exp, (int, float)) or isinstance(exp, bool):
+        raise AuthError("token missing numeric exp claim")
+    if t > exp + 60:
         raise AuthError("token expired")
     nbf = claims.get("nbf")
     if isinstance(nbf, (int, float)) and t < nbf - 60:

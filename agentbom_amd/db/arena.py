@@ -39,6 +39,15 @@ from agentbom_amd.utils.version_utils import _looks_like_commit_sha, normalize_i
 _FNV_OFFSET = np.uint64(0xCBF29CE484222325)
 _FNV_PRIME = np.uint64(0x100000001B3)
 
+# Unfixed-advisory suppression applies ONLY to OS distro ecosystems: app
+# ecosystems (npm/PyPI/...) keep unfixed windows matched by default (the
+# reference gates on package_scan._OS_DISTRO_ECOSYSTEMS, package_scan.py:203,244).
+OS_DISTRO_ECOSYSTEMS = frozenset({"deb", "apk", "rpm"})
+
+
+def _suppress_unfixed(ecosystem: str) -> bool:
+    return normalize_package_ecosystem(ecosystem) in OS_DISTRO_ECOSYSTEMS
+
 
 def hash_name(ecosystem: str, name: str) -> int:
     """Stable FNV-1a 64 over ``eco:name`` (normalized) — the group key."""
@@ -177,7 +186,7 @@ def build_arena(windows: Sequence[AdvisoryWindow], include_unfixed: bool = False
             f |= WF_CPU_FALLBACK  # host drops these windows entirely
         elif not encodable:
             f |= WF_CPU_FALLBACK  # host resolves with the exact comparator
-        if w.unfixed and not include_unfixed:
+        if w.unfixed and not include_unfixed and _suppress_unfixed(w.ecosystem):
             f |= WF_UNFIXED_SUPPRESSED
         flags[i] = f
         sev[i] = SEVERITY_CODE.get(w.severity, 1)
@@ -288,7 +297,7 @@ def match_cpu_fallback(
             )
             if sha_bound:
                 continue  # dropped window, never matches
-            if w.unfixed and not include_unfixed:
+            if w.unfixed and not include_unfixed and _suppress_unfixed(w.ecosystem):
                 continue
             if version_in_range(version, w.introduced, w.fixed, w.last_affected, eco):
                 out.add((idx, w_sorted))

@@ -123,6 +123,8 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
 
     lib = load()
     P = pkg_group_key.numel()
+    if P >= 1 << 32:
+        raise ValueError(f"match: {P} packages exceeds the u32 pair encoding; shard the batch")
     G = group_keys.numel()
     cap = capacity or max(1024, P // 4)
     dev = pkg_group_key.device
@@ -152,6 +154,10 @@ def match_finalize(pending: dict):
     import torch
 
     n = int(pending["out_count"].item())
+    if n < 0:
+        # int32 device counter wrapped: >2^31 candidate matches. Fail loudly
+        # rather than slicing with a wrapped count.
+        raise RuntimeError("match: device match count overflowed int32; shard the package batch")
     if n > pending["cap"]:
         return match(*pending["args"], capacity=int(n * 1.2) + 1024)
     pairs = pending["out_pairs"][:n]

@@ -114,6 +114,7 @@ def _match_packages(
     unique_pkgs: list[tuple[str, str, str]],
     arena: AdvisoryArena,
     use_gpu: bool,
+    include_unfixed: bool = False,
 ) -> list[tuple[int, int]]:
     """Match unique (eco, name, version) tuples; returns (pkg_i, window_i)."""
     P = len(unique_pkgs)
@@ -159,6 +160,7 @@ def _match_packages(
         arena,
         [(i, e, n, v) for i, (e, n, v) in enumerate(unique_pkgs)],
         unencodable,
+        include_unfixed=include_unfixed,
     )
     return sorted(set(pairs) | set(fallback))
 
@@ -215,7 +217,8 @@ def scan_agents(
         except Exception:
             use_gpu = False
     _mark("arena_build")
-    pairs = _match_packages(unique, arena, use_gpu)
+    pairs = _match_packages(unique, arena, use_gpu,
+                            include_unfixed=options.include_unfixed)
     _mark("match")
 
     # ── attach Vulnerability objects to packages ───────────────────────────

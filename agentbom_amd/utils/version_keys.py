@@ -73,10 +73,21 @@ _APK_POST = {"cvs": 17, "svn": 18, "git": 19, "hg": 20, "p": 21}
 _GEM_RANKS = _SEMVER_RANKS  # letter segments compare lexically, same table
 
 _NUMERIC_DOTTED = re.compile(r"^\d+(?:\.\d+)*$")
-# X.Y[.Z[.W]] optionally followed by -tag[.N] / -tagN / .tagN
+# X.Y[.Z[.W]] optionally followed by <sep1>tag[<sep2>N]; separators captured
+# because the exact comparators are separator-sensitive in some families
+# (strict SemVer: "rc.2" < "rc2" lexically; Maven: "alpha.1" > "alpha1";
+# Gem: "-rc2" != ".rc2") — keys must only be issued where order is preserved.
 _GENERIC_RE = re.compile(
     r"^v?(\d+)(?:\.(\d+))?(?:\.(\d+))?(?:\.(\d+))?"
-    r"(?:[-._]([A-Za-z]+)[-._]?(\d+)?)?$"
+    r"(?:([-._])([A-Za-z]+)([-._]?)(\d+)?)?$"
+)
+
+# Families where compare_version_order is STRICT SemVer (identifier-wise,
+# lexical for alphanumeric identifiers): only "-tag" / "-tag.N" with a
+# lowercase tag maps to an order-preserving (rank, num) key; "tagN"/"tag-N"
+# compare lexically and take the exact-comparator fallback instead.
+_STRICT_SEMVER_ECOS = frozenset(
+    {"npm", "npmjs", "yarn", "pnpm", "node", "javascript", "js", "nuget"}
 )
 
 
@@ -193,13 +204,32 @@ def encode_version(version: str, ecosystem: str) -> tuple[int, int, bool]:
     n2 = int(m.group(2)) if m.group(2) else 0
     n3 = int(m.group(3)) if m.group(3) else 0
     n4 = int(m.group(4)) if m.group(4) else 0
-    tag = m.group(5)
-    tagnum = m.group(6)
+    sep1, tag, sep2, tagnum = m.group(5), m.group(6), m.group(7), m.group(8)
     if tag is None:
         rank, num = RELEASE_RANK, 0
         if tagnum is not None:
             return 0, 0, False
     else:
+        if sep2 and tagnum is None:
+            return 0, 0, False  # trailing separator ("1.0.0-rc.")
+        # Separator-sensitivity per family (oracle: compare_version_order):
+        if eco in _STRICT_SEMVER_ECOS:
+            # strict SemVer path: "-tag" / "-tag.N" only; "tagN" and "tag-N"
+            # are single alphanumeric identifiers (lexical compare) and
+            # ".tag"-prefixed forms take the normalizing fallback path.
+            if sep1 != "-" or not tag.islower():
+                return 0, 0, False
+            if tagnum is not None and sep2 != ".":
+                return 0, 0, False
+        elif eco == "maven":
+            # Maven: "alpha1" == "alpha-1" but "alpha.1" orders differently.
+            if tagnum is not None and sep2 == ".":
+                return 0, 0, False
+        elif eco in ("rubygems", "gem", "gems"):
+            # Gem: "-tag" canonicalizes through an extra "pre" segment and
+            # orders differently from ".tag"; only dot/underscore forms keyed.
+            if sep1 == "-":
+                return 0, 0, False
         t = tag.lower()
         if eco == "maven" and t in _MAVEN_RELEASE_ALIASES:
             rank, num = RELEASE_RANK, (int(tagnum) + 1 if tagnum else 0)

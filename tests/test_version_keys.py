@@ -84,6 +84,45 @@ def test_sentinel_and_release_ordering():
     assert compare_keys((hi0, lo0), (hi1, lo1)) == -1
 
 
+def test_separator_sensitive_forms():
+    """Separator variants that the exact comparator distinguishes must either
+    encode order-preservingly or be unencodable ('1.0.0-rc.2' and
+    '1.0.0-rc2' previously shared one key while the oracle says rc.2 < rc2)."""
+    # strict-SemVer family: tagN / tag-N / tag_N are lexical -> unencodable
+    for v in ("1.0.0-rc2", "1.0.0-rc-2", "1.0.0-rc_2", "1.0.0-RC.2",
+              "1.0.0.rc.2", "1.0.0-rc."):
+        for eco in ("npm", "nuget"):
+            assert not encode_version(v, eco)[2], (v, eco)
+    assert encode_version("1.0.0-rc.2", "npm")[2]
+    assert encode_version("1.0.0-rc", "npm")[2]
+    # maven: dot-joined number orders differently -> unencodable; tagN/tag-N fine
+    assert not encode_version("1.0-alpha.1", "maven")[2]
+    assert encode_version("1.0-alpha1", "maven")[2]
+    assert encode_version("1.0-alpha-1", "maven")[2]
+    # gem: hyphen-prefixed tag canonicalizes via an extra "pre" segment
+    assert not encode_version("1.0.0-rc2", "rubygems")[2]
+    assert encode_version("1.0.0.rc2", "rubygems")[2]
+    # cross-check every encodable separator variant against the oracle
+    variants = {
+        "npm": ["1.0.0-rc.2", "1.0.0-rc.10", "1.0.0-rc", "1.0.0-alpha.1", "1.0.0"],
+        "maven": ["1.0-alpha1", "1.0-alpha-2", "1.0-rc1", "1.0-rc10", "1.0"],
+        "rubygems": ["1.0.0.rc2", "1.0.0.rc10", "1.0.0.beta1", "1.0.0"],
+        "cargo": ["1.0.0-rc.2", "1.0.0-rc2", "1.0.0-rc10", "1.0.0"],
+        "composer": ["1.0.0-rc.2", "1.0.0-rc2", "1.0.0-rc10", "1.0.0"],
+    }
+    for eco, versions in variants.items():
+        for left, right in itertools.combinations(versions, 2):
+            lhi, llo, lok = encode_version(left, eco)
+            rhi, rlo, rok = encode_version(right, eco)
+            if not (lok and rok):
+                continue
+            oracle = compare_version_order(left, right, eco)
+            if oracle is None:
+                continue
+            got = compare_keys((lhi, llo), (rhi, rlo))
+            assert got == oracle, f"{eco}: {left} vs {right}: key {got} != oracle {oracle}"
+
+
 _num = st.integers(min_value=0, max_value=400)
 _tag = st.sampled_from(["alpha", "beta", "rc", "pre", ""])
 
