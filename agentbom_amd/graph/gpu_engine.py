@@ -386,25 +386,20 @@ class EstateEngine:
         epss = self.arena["epss"][win_idx]
         flags = (kev.to(torch.uint8) << 1)  # bit1 kev; no ai_ctx/suppressed in synth
         scorecard = torch.full((n_findings,), -1.0, dtype=torch.float32, device=self.device)
-        reach_known = dist[pkg_nodes] != 0xFFFFFFFF
-        reach = torch.where(
-            reach_known,
-            torch.ones(n_findings, dtype=torch.int8, device=self.device),
-            torch.zeros(n_findings, dtype=torch.int8, device=self.device),
-        )
+        # dist is int32 with UNVISITED as the -1 bit pattern (comparing the
+        # u32 literal would promote to int64 and never match)
+        reach_known = dist[pkg_nodes] != -1
+        reach = reach_known.to(torch.int8)
 
-        if False:
-            pass
-        else:
-            from agentbom_amd.ops import cpu_ref
+        from agentbom_amd.ops import cpu_ref
 
-            scores = torch.from_numpy(
-                cpu_ref.risk_score(
-                    sev.numpy(), n_agents.numpy().astype(np.uint32),
-                    n_creds.numpy().astype(np.uint32), n_tools.numpy().astype(np.uint32),
-                    flags.numpy(), epss.numpy(), scorecard.numpy(), reach.numpy(),
-                )
+        scores = torch.from_numpy(
+            cpu_ref.risk_score(
+                sev.numpy(), n_agents.numpy().astype(np.uint32),
+                n_creds.numpy().astype(np.uint32), n_tools.numpy().astype(np.uint32),
+                flags.numpy(), epss.numpy(), scorecard.numpy(), reach.numpy(),
             )
+        )
 
         # deterministic rank: score desc, finding index asc
         order = torch.argsort(scores, descending=True, stable=True)

@@ -1,0 +1,249 @@
+"""Distributed estate engine: one hash-partitioned estate, full pipeline.
+
+``DistEstateEngine`` is the multi-GPU counterpart of
+graph/gpu_engine.EstateEngine: the SAME global estate (same seed, same
+graph) is hash-partitioned across ranks (parallel/partition.py), and every
+pipeline stage runs distributed:
+
+  match        owned packages vs the replicated advisory arena (HIP kernel;
+               packages are dealt round-robin by the modular hash, so match
+               load is balanced by construction)
+  reach BFS    distributed multi-source BFS from ALL agents (each rank
+               seeds its owned agents; parallel/dist_bfs.py)
+  blast joins  owner-computes two-shuffle distinct joins
+               (parallel/dist_blast.py) — cross-shard reach lands in the
+               counts, unlike round 1's shard-local joins
+  score/rank   fused score_gather HIP kernel per rank + local rank order
+
+The union of per-rank findings provably equals a single-engine run on the
+unpartitioned estate (tests/test_dist_engine.py, world 2 and 4 over gloo;
+the identical code path runs over RCCL on the 8-GPU node).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from agentbom_amd.graph.gpu_engine import _DB_CRED, _DB_TOOL, _FULL_CRED, _FULL_TOOL, _impact_lut
+from agentbom_amd.parallel.dist_blast import distributed_blast_counts
+from agentbom_amd.parallel.dist_bfs import distributed_reach
+from agentbom_amd.parallel.partition import build_csr, partition_estate
+from agentbom_amd.scan.synth import (
+    ET_CONTAINS,
+    ET_HAS_CRED,
+    ET_PROVIDES_TOOL,
+    ET_USES,
+    SyntheticEstate,
+)
+
+
+class DistEstateEngine:
+    """One rank's engine over its hash partition of the global estate."""
+
+    def __init__(self, estate: SyntheticEstate, rank: int, world: int,
+                 device: str = "cuda", group=None):
+        import torch
+
+        self.torch = torch
+        self.estate = estate
+        self.rank = rank
+        self.world = world
+        self.group = group
+        self.device = torch.device(device)
+        self.N = estate.num_nodes
+        dev = self.device
+
+        part = partition_estate(estate, rank, world)
+        self.part = part
+        self.fwd = build_csr(part.fwd_src, part.fwd_dst, part.fwd_type, self.N, dev)
+        self.rev = build_csr(part.rev_src, part.rev_dst, part.rev_type, self.N, dev)
+
+        def t(a, dtype):
+            return torch.from_numpy(np.ascontiguousarray(a)).to(device=dev, dtype=dtype)
+
+        self.own_pkg_idx = t(part.own_pkg_idx, torch.int64)
+        gk = t(part.pkg_name_id.view(np.int64), torch.int64)
+        khi = t(part.pkg_key_hi.view(np.int64), torch.int64)
+        klo = t(part.pkg_key_lo.view(np.int64), torch.int64)
+        fl = t(part.pkg_flags, torch.uint8)
+        # match-locality layout (same as the single engine): sort owned rows
+        # by group key so adjacent lanes walk shared tree paths
+        self.pkg_perm = torch.argsort(gk, stable=True)
+        self.pkg_group_key_sorted = gk[self.pkg_perm].contiguous()
+        self.pkg_key_hi_sorted = khi[self.pkg_perm].contiguous()
+        self.pkg_key_lo_sorted = klo[self.pkg_perm].contiguous()
+        self.pkg_flags_sorted = fl[self.pkg_perm].contiguous()
+
+        self.arena = estate.arena.to_torch(dev)
+        self.cred_lut = torch.from_numpy(_impact_lut(_FULL_CRED, _DB_CRED)).to(dev)
+        self.tool_lut = torch.from_numpy(_impact_lut(_FULL_TOOL, _DB_TOOL)).to(dev)
+
+        # replicated per-node classification flags (small next to the CSR)
+        self.node_is_db_cred = torch.zeros(self.N, dtype=torch.uint8, device=dev)
+        self.node_is_db_cred[
+            estate.cred_base + torch.arange(estate.n_creds, device=dev)
+        ] = torch.from_numpy(estate.cred_is_db).to(dev)
+        self.node_is_db_tool = torch.zeros(self.N, dtype=torch.uint8, device=dev)
+        self.node_is_db_tool[
+            estate.tool_base + torch.arange(estate.n_tools, device=dev)
+        ] = torch.from_numpy(estate.tool_is_db).to(dev)
+
+        self.own_agents = t(part.own_agents, torch.int32)
+        self._ws: dict = {}
+        self._match_stream = None
+
+    @property
+    def use_gpu(self) -> bool:
+        return self.device.type == "cuda"
+
+    # ── stages ─────────────────────────────────────────────────────────────
+
+    def _match_local(self):
+        """(local_row, window) for owned packages (sorted pair order)."""
+        torch = self.torch
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            sp, sw = native.match(
+                self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
+                self.pkg_key_lo_sorted, self.pkg_flags_sorted,
+                self.arena["group_keys"], self.arena["group_off"],
+                self.arena["windows"],
+            )
+            return sp, sw
+        from agentbom_amd.ops import cpu_ref
+
+        win = {k: v.numpy().view(np.uint64) if v.dtype == torch.int64 else v.numpy()
+               for k, v in self.arena["windows"].items()}
+        sp, sw = cpu_ref.match(
+            self.pkg_group_key_sorted.numpy().view(np.uint64),
+            self.pkg_key_hi_sorted.numpy().view(np.uint64),
+            self.pkg_key_lo_sorted.numpy().view(np.uint64),
+            self.pkg_flags_sorted.numpy(),
+            self.arena["group_keys"].numpy().view(np.uint64),
+            self.arena["group_off"].numpy().view(np.uint32),
+            win,
+        )
+        return torch.from_numpy(np.ascontiguousarray(sp)).to(torch.int64), \
+            torch.from_numpy(np.ascontiguousarray(sw)).to(torch.int64)
+
+    def dependency_reach(self):
+        """Distributed multi-source BFS from all agents (u32 global dist)."""
+        mask = (1 << ET_USES) | (1 << ET_CONTAINS) | (1 << ET_HAS_CRED) | (1 << ET_PROVIDES_TOOL)
+        return distributed_reach(
+            self.fwd, self.own_agents, self.N, self.world, self.rank,
+            etype=self.fwd["etype"], allowed_mask=mask, group=self.group,
+            workspace=self._ws,
+        )
+
+    def step(self, reach_dist=None):
+        """One full distributed findings pass; returns the local summary."""
+        torch = self.torch
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            # overlap: match on a side stream while the BFS level loop (host
+            # driven, syncs its own stream) runs collectives on the default
+            if self._match_stream is None:
+                self._match_stream = torch.cuda.Stream(device=self.device)
+            side = self._match_stream
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                pending = native.match_launch(
+                    self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
+                    self.pkg_key_lo_sorted, self.pkg_flags_sorted,
+                    self.arena["group_keys"], self.arena["group_off"],
+                    self.arena["windows"])
+            dist = reach_dist if reach_dist is not None else self.dependency_reach()
+            torch.cuda.current_stream().wait_stream(side)
+            sp, sw = native.match_finalize(pending)
+        else:
+            sp, sw = self._match_local()
+            dist = reach_dist if reach_dist is not None else self.dependency_reach()
+
+        # local sorted row -> GLOBAL package index, deterministic pair order
+        orig = self.own_pkg_idx[self.pkg_perm[sp]]
+        packed = (orig << 32) | sw
+        packed, _ = torch.sort(packed)
+        pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
+        n_findings = pkg_idx.numel()
+
+        pkg_nodes = pkg_idx + self.estate.pkg_base
+        uniq_pkgs = torch.unique_consecutive(pkg_nodes)
+        counts = distributed_blast_counts(
+            uniq_pkgs, self.rev, self.fwd,
+            (ET_CONTAINS, ET_USES, ET_HAS_CRED, ET_PROVIDES_TOOL),
+            self.node_is_db_cred, self.node_is_db_tool,
+            self.rank, self.world, group=self.group, ws=self._ws,
+        )
+        pos = torch.searchsorted(counts["uniq_pkgs"], pkg_nodes)
+
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            scores, n_agents, n_creds, n_tools = native.score_gather(
+                win_idx.contiguous(), pkg_nodes.contiguous(), pos.contiguous(),
+                self.arena["severity"], self.arena["kev"], self.arena["epss"],
+                self.arena["impact"], self.cred_lut, self.tool_lut,
+                counts["counts2d"].contiguous(), dist)
+        else:
+            impact = self.arena["impact"].to(torch.int64)[win_idx]
+            cred_cls = self.cred_lut.to(torch.int64)[impact]
+            tool_cls = self.tool_lut.to(torch.int64)[impact]
+            n_creds = torch.where(
+                cred_cls == 2, counts["n_creds_all"][pos],
+                torch.where(cred_cls == 1, counts["n_creds_db"][pos],
+                            torch.zeros_like(pos, dtype=torch.int32)))
+            n_tools = torch.where(
+                tool_cls == 2, counts["n_tools_all"][pos],
+                torch.where(tool_cls == 1, counts["n_tools_db"][pos],
+                            torch.zeros_like(pos, dtype=torch.int32)))
+            n_agents = counts["n_agents"][pos]
+            sev = self.arena["severity"][win_idx]
+            kev = self.arena["kev"][win_idx]
+            epss = self.arena["epss"][win_idx]
+            flags = (kev.to(torch.uint8) << 1)
+            scorecard = torch.full((n_findings,), -1.0, dtype=torch.float32,
+                                   device=self.device)
+            reach_known = dist[pkg_nodes] != -1
+            reach = reach_known.to(torch.int8)
+            from agentbom_amd.ops import cpu_ref
+
+            scores = torch.from_numpy(
+                cpu_ref.risk_score(
+                    sev.numpy(), n_agents.numpy().astype(np.uint32),
+                    n_creds.numpy().astype(np.uint32),
+                    n_tools.numpy().astype(np.uint32),
+                    flags.numpy(), epss.numpy(), scorecard.numpy(),
+                    reach.numpy(),
+                ))
+
+        order = torch.argsort(scores, descending=True, stable=True)
+        return {
+            "n_findings": int(n_findings),
+            "scores": scores,
+            "order": order,
+            "pkg_idx": pkg_idx,
+            "win_idx": win_idx,
+            "n_agents": n_agents,
+            "n_creds": n_creds,
+            "n_tools": n_tools,
+            "reach_dist": dist,
+        }
+
+    def blast_radius_query(self, node_id: int, max_hops: int = 4):
+        """Distributed bounded blast-radius query (reverse direction).
+
+        All ranks participate; the rank owning ``node_id`` seeds it.
+        Returns the global dist tensor (hops; -1 bit pattern = unreached).
+        """
+        torch = self.torch
+        if node_id % self.world == self.rank:
+            src = torch.tensor([node_id], dtype=torch.int32, device=self.device)
+        else:
+            src = torch.empty(0, dtype=torch.int32, device=self.device)
+        return distributed_reach(
+            self.rev, src, self.N, self.world, self.rank,
+            etype=None, allowed_mask=0xFFFFFFFF, max_levels=max_hops,
+            group=self.group, workspace=self._ws,
+        )

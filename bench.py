@@ -1,19 +1,24 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: findings/sec on the 10M-pkg synthetic estate.
+"""Flagship benchmark: findings/sec on the 10M-pkg-per-GPU synthetic estate.
 
-One step = one full findings pass on each GPU's estate shard:
-  bulk version-range match (HIP kernel, 10M packages vs advisory arena)
-  -> dependency-reach BFS (HIP frontier kernels; multi-GPU: hash-partitioned
-     CSR with RCCL all-to-all frontier exchange over xGMI)
-  -> blast-radius joins (GPU segmented gathers: distinct agents/creds/tools
-     per finding, CWE-impact filtered)
+One step = one full findings pass over the estate:
+  bulk version-range match (HIP kernel, packages vs advisory arena)
+  -> dependency-reach BFS (HIP frontier kernels; multi-GPU: ONE global
+     estate hash-partitioned by node id with padded RCCL all-to-all
+     frontier exchange over xGMI)
+  -> blast-radius joins (GPU joins: distinct agents/creds/tools per
+     finding, CWE-impact filtered; multi-GPU: owner-computes two-shuffle
+     distributed joins — cross-shard reach included)
   -> risk scoring (HIP kernel, reference formula)
   -> ranking (sort by score desc).
 
-value = aggregate findings/sec over all N GPUs (weak scaling: one 10M-pkg
-shard per GPU).  After the timed region, rank 0 measures p50 blast-radius
-query latency (single bounded-BFS queries, one launch + sync each) — the
-second BASELINE metric — reported in config.
+Multi-GPU is weak scaling of ONE estate: the global estate is N x the
+per-GPU size, every rank owns the node ids congruent to its rank
+(provably equal to a single-engine run on the same estate —
+tests/test_dist_engine.py).  value = aggregate findings/sec over all N
+GPUs.  After the timed region, p50 blast-radius query latency (the second
+BASELINE metric) is measured: single-GPU via hipGraph replay; multi-GPU
+via collective distributed bounded BFS queries.
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--packages P]
 For N>1 the driver launches via torch.distributed.run (one rank per GPU).
@@ -71,46 +76,32 @@ def main():
         dist_mod.init_process_group(backend)
     device = f"cuda:{local_rank}" if use_gpu else "cpu"
 
-    from agentbom_amd.graph.gpu_engine import EstateEngine
-    from agentbom_amd.parallel.partition import build_global_csr, generate_shard
+    from agentbom_amd.scan.synth import generate_estate
 
+    # ONE global estate, weak scaling: global sizes = per-GPU sizes x world
     estate_kw = dict(
-        n_agents=args.agents, n_servers=args.servers, n_packages=args.packages,
-        name_catalog=args.name_catalog,
+        n_agents=args.agents * n_gpus, n_servers=args.servers * n_gpus,
+        n_packages=args.packages * n_gpus, name_catalog=args.name_catalog * n_gpus,
     )
 
     t_gen = time.perf_counter()
-    if distributed:
-        est, edges = generate_shard(rank, world, cross_fraction=0.05, seed=args.seed,
-                                    **estate_kw)
-    else:
-        from agentbom_amd.scan.synth import generate_estate
-
-        est = generate_estate(seed=args.seed, **estate_kw)
-        edges = None
+    est = generate_estate(seed=args.seed, **estate_kw)
     gen_s = time.perf_counter() - t_gen
 
     t_build = time.perf_counter()
-    engine = EstateEngine(est, device=device)
-    global_csr = None
     if distributed:
-        global_csr = build_global_csr(edges, torch.device(device))
+        from agentbom_amd.parallel.dist_engine import DistEstateEngine
+
+        engine = DistEstateEngine(est, rank, world, device=device)
+    else:
+        from agentbom_amd.graph.gpu_engine import EstateEngine
+
+        engine = EstateEngine(est, device=device)
     if use_gpu:
         torch.cuda.synchronize()
     build_s = time.perf_counter() - t_build
 
     def one_step():
-        if distributed:
-            from agentbom_amd.parallel.dist_bfs import distributed_reach
-
-            sources = (torch.arange(est.n_agents, dtype=torch.int32, device=device)
-                       + rank * edges["stride"])
-            dglobal = distributed_reach(
-                global_csr, sources, edges["num_global"], edges["stride"],
-                etype=global_csr["etype"],
-            )
-            own = dglobal[rank * edges["stride"]: (rank + 1) * edges["stride"]]
-            return engine.step(reach_dist=own)
         return engine.step()
 
     # warmup
@@ -152,9 +143,26 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     findings_per_sec = total_findings_per_step * args.steps / elapsed
 
-    # p50 blast-radius query latency (rank 0, individual launches)
+    # p50 blast-radius query latency
     p50_ms = None
-    if rank == 0 and args.queries > 0:
+    if distributed and args.queries > 0:
+        # collective distributed bounded-BFS queries: every rank participates,
+        # rank 0 reports the median wall time per query
+        qidx = torch.arange(args.queries, dtype=torch.int64) * 997 % est.n_packages
+        qnodes = (qidx + est.pkg_base).tolist()
+        engine.blast_radius_query(qnodes[0], max_hops=4)  # warm
+        if use_gpu:
+            torch.cuda.synchronize()
+        dist_mod.barrier()
+        lat = []
+        for qn in qnodes:
+            tq = time.perf_counter()
+            engine.blast_radius_query(qn, max_hops=4)
+            if use_gpu:
+                torch.cuda.synchronize()
+            lat.append((time.perf_counter() - tq) * 1000.0)
+        p50_ms = statistics.median(lat)
+    elif rank == 0 and args.queries > 0:
         rng_idx = torch.arange(args.queries, dtype=torch.int64) * 997 % est.n_packages
         nodes = (rng_idx + est.pkg_base).to(device)
         lat = []
@@ -196,9 +204,11 @@ def main():
                 "global_batch": total_findings_per_step,
                 "seq_len": args.packages,
                 "parallelism": f"hashpart{n_gpus}" if n_gpus > 1 else "single",
-                "estate": f"synthetic {args.packages/1e6:.0f}M-pkg/{est.num_nodes/1e6:.1f}M-node per GPU",
-                "nodes_per_gpu": est.num_nodes,
-                "edges_per_gpu": est.num_edges,
+                "estate": (f"synthetic ONE global {args.packages*n_gpus/1e6:.0f}M-pkg/"
+                           f"{est.num_nodes/1e6:.1f}M-node estate, "
+                           f"hash-partitioned {n_gpus}-way ({args.packages/1e6:.0f}M pkg/GPU)"),
+                "nodes_global": est.num_nodes,
+                "edges_global": est.num_edges,
                 "arena_windows": est.arena.num_windows,
                 "findings_per_step": total_findings_per_step,
                 "p50_blast_query_ms": p50_ms,

@@ -1,9 +1,10 @@
-"""Distributed frontier-exchange BFS: 2-process gloo test (CPU).
+"""Distributed frontier-exchange BFS over ONE hash-partitioned estate.
 
-Validates the partitioned traversal (parallel/dist_bfs.py) against a
-single-process BFS over the union graph — hop distances must be identical
-for every owned node on every rank.  The same code path runs over RCCL on
-the 8-GPU node (backend "nccl"); only the exchange primitive differs.
+Validates parallel/dist_bfs.distributed_reach (padded equal-split
+all-to-all, counts piggybacked — the identical code path RCCL runs on the
+8-GPU node) against a single-process BFS over the unpartitioned estate:
+hop distances must be identical for every owned node on every rank.  Also
+unit-tests the exchange overflow/retry protocol.
 """
 
 from __future__ import annotations
@@ -16,35 +17,14 @@ import torch
 import torch.multiprocessing as mp
 
 from agentbom_amd.ops import cpu_ref
-from agentbom_amd.parallel.partition import build_global_csr, generate_shard
+from agentbom_amd.scan.synth import generate_estate
 
 WORLD = 2
-ESTATE_KW = dict(n_agents=40, n_servers=150, n_packages=1500, name_catalog=400)
+ESTATE_KW = dict(n_agents=40, n_servers=150, n_packages=1500, name_catalog=400,
+                 seed=99)
 
 
-def _union_graph(world: int, seed: int = 99):
-    srcs, dsts, ets = [], [], []
-    stride = num_global = None
-    agents = []
-    for r in range(world):
-        est, edges = generate_shard(r, world, cross_fraction=0.1, seed=seed, **ESTATE_KW)
-        srcs.append(edges["src"])
-        dsts.append(edges["dst"])
-        ets.append(edges["etype"])
-        stride = edges["stride"]
-        num_global = edges["num_global"]
-        agents.append(np.arange(est.n_agents) + r * stride)
-    src = np.concatenate(srcs)
-    dst = np.concatenate(dsts)
-    et = np.concatenate(ets)
-    order = np.argsort(src, kind="stable")
-    counts = np.bincount(src, minlength=num_global)
-    row_off = np.zeros(num_global + 1, dtype=np.int64)
-    np.cumsum(counts, out=row_off[1:])
-    return row_off, dst[order], et[order], np.concatenate(agents), stride, num_global
-
-
-def _worker(rank: int, world: int, port: int, result_dir: str):
+def _worker(rank: int, world: int, port: int, result_dir: str, cap: int):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     import torch.distributed as dist_mod
@@ -52,21 +32,29 @@ def _worker(rank: int, world: int, port: int, result_dir: str):
     dist_mod.init_process_group("gloo", rank=rank, world_size=world)
     try:
         from agentbom_amd.parallel.dist_bfs import distributed_reach
+        from agentbom_amd.parallel.partition import build_csr, partition_estate
 
-        est, edges = generate_shard(rank, world, cross_fraction=0.1, seed=99, **ESTATE_KW)
-        csr = build_global_csr(edges, torch.device("cpu"))
-        sources = torch.arange(est.n_agents, dtype=torch.int32) + rank * edges["stride"]
+        est = generate_estate(**ESTATE_KW)
+        part = partition_estate(est, rank, world)
+        csr = build_csr(part.fwd_src, part.fwd_dst, part.fwd_type,
+                        est.num_nodes, torch.device("cpu"))
+        sources = torch.from_numpy(part.own_agents).to(torch.int32)
         dist = distributed_reach(
-            csr, sources, edges["num_global"], edges["stride"], etype=csr["etype"],
+            csr, sources, est.num_nodes, world, rank, etype=csr["etype"],
+            cap=cap,
         )
-        own = dist[rank * edges["stride"]: (rank + 1) * edges["stride"]]
-        np.save(os.path.join(result_dir, f"dist_{rank}.npy"), own.numpy())
+        own = np.arange(est.num_nodes) % world == rank
+        np.save(os.path.join(result_dir, f"dist_{rank}.npy"),
+                dist.numpy()[own])
     finally:
         dist_mod.destroy_process_group()
 
 
 @pytest.mark.timeout(180)
-def test_distributed_reach_matches_union(tmp_path):
+@pytest.mark.parametrize("cap", [0, 8], ids=["auto_cap", "tiny_cap_overflow_retry"])
+def test_distributed_reach_matches_single(tmp_path, cap):
+    """cap=8 forces the padded exchange to overflow and retry collectively
+    (grown cap), exercising the recovery path end-to-end."""
     import socket
 
     with socket.socket() as s:
@@ -74,17 +62,33 @@ def test_distributed_reach_matches_union(tmp_path):
         port = s.getsockname()[1]
 
     ctx = mp.get_context("spawn")
-    procs = [ctx.Process(target=_worker, args=(r, WORLD, port, str(tmp_path))) for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker, args=(r, WORLD, port, str(tmp_path), cap))
+             for r in range(WORLD)]
     for p in procs:
         p.start()
     for p in procs:
         p.join(timeout=150)
         assert p.exitcode == 0
 
-    row_off, col, et, agents, stride, num_global = _union_graph(WORLD)
-    expected = cpu_ref.bfs(row_off, col, agents, num_global, etype=et)
+    est = generate_estate(**ESTATE_KW)
+    row_counts = np.bincount(est.edge_src, minlength=est.num_nodes)
+    row_off = np.zeros(est.num_nodes + 1, dtype=np.int64)
+    np.cumsum(row_counts, out=row_off[1:])
+    order = np.argsort(est.edge_src, kind="stable")
+    expected = cpu_ref.bfs(
+        row_off, est.edge_dst[order], np.arange(est.n_agents), est.num_nodes,
+        etype=est.edge_type[order],
+    )
 
     for r in range(WORLD):
         got = np.load(tmp_path / f"dist_{r}.npy").view(np.uint32)
-        exp = expected[r * stride: (r + 1) * stride]
-        assert np.array_equal(got, exp), f"rank {r} dist mismatch"
+        own = np.arange(est.num_nodes) % WORLD == r
+        assert np.array_equal(got, expected[own]), f"rank {r} dist mismatch"
+
+
+def test_padded_exchange_world1_passthrough():
+    from agentbom_amd.parallel.exchange import exchange_sized
+
+    v = torch.arange(10, dtype=torch.int64)
+    local, recv = exchange_sized(v, torch.zeros(10, dtype=torch.int64), 0, 1)
+    assert torch.equal(local, v) and recv.numel() == 0

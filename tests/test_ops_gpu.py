@@ -341,7 +341,6 @@ def test_distributed_reach_gpu_world1_matches_local(estate, dev):
     """The FULL distributed code path (incl. the edge-centric dense-level
     branch) on GPU with world_size=1 must equal the local engine BFS."""
     import torch as _torch
-    import torch.distributed as dist_mod
 
     from agentbom_amd.graph.gpu_engine import EstateEngine
     from agentbom_amd.parallel.dist_bfs import distributed_reach
@@ -352,21 +351,29 @@ def test_distributed_reach_gpu_world1_matches_local(estate, dev):
         ET_USES,
     )
 
-    if not dist_mod.is_initialized():
-        store = dist_mod.TCPStore("127.0.0.1", 29751, 1, True)
-        dist_mod.init_process_group("gloo", store=store, rank=0, world_size=1)
-    try:
-        eng = EstateEngine(estate, device=str(dev))
-        mask = ((1 << ET_USES) | (1 << ET_CONTAINS) | (1 << ET_HAS_CRED)
-                | (1 << ET_PROVIDES_TOOL))
-        local = eng.dependency_reach()
+    eng = EstateEngine(estate, device=str(dev))
+    mask = ((1 << ET_USES) | (1 << ET_CONTAINS) | (1 << ET_HAS_CRED)
+            | (1 << ET_PROVIDES_TOOL))
+    local = eng.dependency_reach()
 
-        csr = {"row_off": eng.fwd["row_off"], "col": eng.fwd["col"],
-               "etype": eng.fwd["etype"], "src": eng.fwd["src"]}
-        dist_out = distributed_reach(
-            csr, eng.agent_ids, estate.num_nodes, estate.num_nodes,
-            etype=eng.fwd["etype"], allowed_mask=mask)
-        assert _torch.equal(dist_out.cpu().view(_torch.int32),
-                            local.cpu().view(_torch.int32))
-    finally:
-        dist_mod.destroy_process_group()
+    csr = {"row_off": eng.fwd["row_off"], "col": eng.fwd["col"],
+           "etype": eng.fwd["etype"], "src": eng.fwd["src"]}
+    dist_out = distributed_reach(
+        csr, eng.agent_ids, estate.num_nodes, world=1, rank=0,
+        etype=eng.fwd["etype"], allowed_mask=mask)
+    assert _torch.equal(dist_out.cpu().view(_torch.int32),
+                        local.cpu().view(_torch.int32))
+
+
+def test_dist_engine_gpu_world1_matches_single(estate, dev):
+    """DistEstateEngine full step on GPU at world=1 == EstateEngine step
+    (covers the dist match/join/score path with the native kernels)."""
+    import torch as _torch
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+    from agentbom_amd.parallel.dist_engine import DistEstateEngine
+
+    ref = EstateEngine(estate, device=str(dev)).step()
+    got = DistEstateEngine(estate, 0, 1, device=str(dev)).step()
+    for key in ("pkg_idx", "win_idx", "n_agents", "n_creds", "n_tools", "scores"):
+        assert _torch.equal(got[key].cpu(), ref[key].cpu()), key
