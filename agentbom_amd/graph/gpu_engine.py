@@ -415,6 +415,99 @@ class EstateEngine:
             "reach_dist": dist,
         }
 
+    # ── attack / exposure paths (GPU path DP) ──────────────────────────────
+
+    ET_LATERAL = 4  # reversed USES edge (server -> agent): lateral movement
+
+    def _path_edges(self):
+        """Edge arrays for the path DP: forward edges + reversed USES edges
+        (lateral movement via shared servers).  Cached device tensors."""
+        cached = getattr(self, "_path_edges_cache", None)
+        if cached is not None:
+            return cached
+        torch = self.torch
+        est = self.estate
+        uses = est.edge_type == ET_USES
+        src = np.concatenate([est.edge_src, est.edge_dst[uses]])
+        dst = np.concatenate([est.edge_dst, est.edge_src[uses]])
+        et = np.concatenate([est.edge_type,
+                             np.full(int(uses.sum()), self.ET_LATERAL, dtype=np.uint8)])
+        dev = self.device
+        cached = (
+            torch.from_numpy(src).to(device=dev, dtype=torch.int32),
+            torch.from_numpy(dst).to(device=dev, dtype=torch.int32),
+            torch.from_numpy(et).to(device=dev, dtype=torch.uint8),
+        )
+        self._path_edges_cache = cached
+        return cached
+
+    def attack_paths(self, step_res=None, k: int = 100, max_depth: int = 6):
+        """Top-k attack/exposure paths over the estate (HIP path DP).
+
+        Entry class: agents.  Crown jewels: credentials and tools.  Gate:
+        the path must traverse a server that contains a matched-vulnerable
+        package (the estate analog of the vulnerable_to/exploitable_via
+        edge gate in graph/attack_paths.py).  Node boosts mirror
+        _node_boost: vulnerable +1.5, severity map, KEV +2.0.
+        Replaces the reference's DFS at estate scale
+        (src/agent_bom/graph/attack_path_fusion.py:194).
+        """
+        from agentbom_amd.graph.path_engine import run_path_dp
+
+        torch = self.torch
+        est = self.estate
+        if step_res is None:
+            step_res = self.step()
+
+        pkg_nodes = (step_res["pkg_idx"] + est.pkg_base).to(torch.int64)
+        win_idx = step_res["win_idx"].to(torch.int64)
+        sev = self.arena["severity"].to(torch.int64)[win_idx]
+        kev = self.arena["kev"].to(torch.int64)[win_idx]
+        sev_boost = torch.zeros_like(sev, dtype=torch.float32)
+        sev_boost[sev == 5] = 2.0   # critical
+        sev_boost[sev == 4] = 1.2   # high
+        sev_boost[sev == 3] = 0.6   # medium
+        boost = 1.5 + sev_boost + kev.to(torch.float32) * 2.0
+
+        node_boost = torch.zeros(self.N, dtype=torch.float32, device=self.device)
+        node_boost.scatter_reduce_(0, pkg_nodes, boost, reduce="amax")
+
+        # gated servers: contain >= 1 matched package; server boost = the
+        # best boost among its matched packages (compromise value)
+        uniq = torch.unique(pkg_nodes)
+        pos = torch.arange(uniq.numel(), device=self.device)
+        srv, carry = self._expand(self.rev, uniq, pos, ET_CONTAINS)
+        node_gate = torch.zeros(self.N, dtype=torch.uint8, device=self.device)
+        if srv.numel():
+            node_gate[srv] = 1
+            pkg_best = torch.zeros(uniq.numel(), dtype=torch.float32, device=self.device)
+            pkg_best.scatter_reduce_(0, torch.searchsorted(uniq, pkg_nodes),
+                                     boost, reduce="amax")
+            node_boost.scatter_reduce_(0, srv, pkg_best[carry], reduce="amax")
+
+        etype_boost = np.zeros(256, dtype=np.float32)
+        etype_boost[ET_USES] = 0.3
+        etype_boost[ET_CONTAINS] = 0.3
+        etype_boost[ET_HAS_CRED] = 1.5       # exposes_cred class
+        etype_boost[ET_PROVIDES_TOOL] = 0.3
+        etype_boost[self.ET_LATERAL] = 1.0   # shares_server class
+        etype_trav = np.zeros(256, dtype=np.uint8)
+        etype_trav[[ET_USES, ET_CONTAINS, ET_HAS_CRED, ET_PROVIDES_TOOL,
+                    self.ET_LATERAL]] = 1
+        etype_gate = np.zeros(256, dtype=np.uint8)
+
+        target_mask = np.zeros(self.N, dtype=np.uint8)
+        target_mask[est.cred_base:est.cred_base + est.n_creds] = 1
+        target_mask[est.tool_base:est.tool_base + est.n_tools] = 1
+
+        src, dst, et = self._path_edges()
+        return run_path_dp(
+            src, dst, et, None, self.N, self.agent_ids.to(torch.int64),
+            node_boost, etype_boost, etype_trav, etype_gate, node_gate,
+            target_mask, max_depth=max_depth, k=k,
+            device=self.device if self.use_gpu else None,
+        )
+
     def enable_query_graph(self, batch: int = 1, max_hops: int = 4,
                            max_nodes: int = 4096) -> None:
         """Capture the blast-radius query into a hipGraph for serving.

@@ -171,3 +171,63 @@ def severity_histogram(owner, severity, num_containers: int):
     hist = np.zeros((num_containers, 6), dtype=np.uint32)
     np.add.at(hist, (np.asarray(owner, dtype=np.int64), np.asarray(severity, dtype=np.int64)), 1)
     return hist
+
+
+# ── attack/exposure path DP (oracle for csrc/paths.hip) ─────────────────────
+
+PATH_NO_EDGE = np.uint64(0xFFFFFFFF)
+
+
+def _ordered_f32(f):
+    """Order-preserving u32 image of non-negative float32 scores."""
+    u = np.asarray(f, dtype=np.float32).view(np.uint32)
+    return (u | np.uint32(0x80000000)).astype(np.uint64)
+
+
+def path_unordered_f32(u):
+    """Inverse of _ordered_f32 for valid (non-negative-score) labels."""
+    raw = (np.asarray(u, dtype=np.uint64) & np.uint64(0x7FFFFFFF)).astype(np.uint32)
+    return raw.view(np.float32)
+
+
+def path_pack(score, edge):
+    return (_ordered_f32(score) << np.uint64(32)) | np.asarray(edge, dtype=np.uint64)
+
+
+def path_relax(edge_src, col, etype, edge_weight, cur, node_boost,
+               etype_boost, etype_trav, etype_gate, node_gate):
+    """One hop of the max-score path DP; exact mirror of abom_path_relax.
+
+    ``cur``/returned ``nxt`` are u64 [N*2] packed labels
+    ((ordered_f32(score) << 32) | winner_edge); slot 0 = ungated path,
+    slot 1 = path that used a gate edge or touched a gated node.  The
+    kernel's atomicMax == np.maximum.at (both order-independent).
+    """
+    nxt = np.zeros_like(cur)
+    et = np.asarray(etype)
+    keep = etype_trav[et].astype(bool)
+    e_idx = np.nonzero(keep)[0]
+    if not len(e_idx):
+        return nxt
+    u = np.asarray(edge_src, dtype=np.int64)[e_idx]
+    v = np.asarray(col, dtype=np.int64)[e_idx]
+    step = etype_boost[et[e_idx]].astype(np.float32) + node_boost[v].astype(np.float32)
+    if edge_weight is not None:
+        step = step + edge_weight[e_idx].astype(np.float32) * np.float32(0.3)
+    gate = etype_gate[et[e_idx]].astype(bool)
+    if node_gate is not None:
+        gate = gate | node_gate[v].astype(bool)
+
+    lu0 = cur[u * 2]
+    m = lu0 != 0
+    if m.any():
+        s = path_unordered_f32(lu0[m] >> np.uint64(32)) + step[m]
+        cand = path_pack(s, e_idx[m])
+        np.maximum.at(nxt, v[m] * 2 + gate[m], cand)
+    lu1 = cur[u * 2 + 1]
+    m = lu1 != 0
+    if m.any():
+        s = path_unordered_f32(lu1[m] >> np.uint64(32)) + step[m]
+        cand = path_pack(s, e_idx[m])
+        np.maximum.at(nxt, v[m] * 2 + 1, cand)
+    return nxt

@@ -41,6 +41,7 @@ _SIGNATURES = {
     "abom_blast_counts": ([_c, _i64] + [_c] * 6 + [_i32] * 4 + [_c] * 4 + [_c], _i32),
     "abom_severity_histogram": ([_c, _c, _c, _i64, _c], _i32),
     "abom_score_gather": ([_c] * 11 + [_c] * 4 + [_i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
+    "abom_path_relax": ([_c] * 11 + [_i64, _c], _i32),
 }
 
 
@@ -450,3 +451,23 @@ def severity_histogram(owner, severity, num_containers: int):
                                      owner.numel(), _stream())
     _check(rc, "abom_severity_histogram")
     return hist.view(num_containers, 6)
+
+
+def path_relax(edge_src, col, etype, edge_weight, cur, nxt, node_boost,
+               etype_boost, etype_trav, etype_gate, node_gate):
+    """One hop of the max-score path DP (ops/csrc/paths.hip).
+
+    ``cur``/``nxt``: int64 [N*2] tensors holding the packed u64 labels
+    ((ordered_f32(score) << 32) | winner_edge); ``nxt`` must be pre-zeroed.
+    ``edge_weight`` and ``node_gate`` may be None.
+    """
+    lib = load()
+    rc = lib.abom_path_relax(
+        _ptr(edge_src), _ptr(col), _ptr(etype),
+        _ptr(edge_weight) if edge_weight is not None else None,
+        _ptr(cur), _ptr(nxt), _ptr(node_boost), _ptr(etype_boost),
+        _ptr(etype_trav), _ptr(etype_gate),
+        _ptr(node_gate) if node_gate is not None else None,
+        etype.numel(), _stream(),
+    )
+    _check(rc, "abom_path_relax")

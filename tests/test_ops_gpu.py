@@ -377,3 +377,51 @@ def test_dist_engine_gpu_world1_matches_single(estate, dev):
     got = DistEstateEngine(estate, 0, 1, device=str(dev)).step()
     for key in ("pkg_idx", "win_idx", "n_agents", "n_creds", "n_tools", "scores"):
         assert _torch.equal(got[key].cpu(), ref[key].cpu()), key
+
+
+def test_path_relax_parity(estate, dev):
+    """paths.hip relax levels vs cpu_ref.path_relax — bit-for-bit equal
+    packed labels per hop (atomicMax == np.maximum.at, both order-free)."""
+    from agentbom_amd.graph.path_engine import _run_gpu
+    from agentbom_amd.ops import cpu_ref
+
+    src = estate.edge_src.astype(np.int32)
+    dst = estate.edge_dst.astype(np.int32)
+    et = estate.edge_type
+    eb = np.zeros(256, dtype=np.float32)
+    eb[[0, 1, 2, 3]] = [0.3, 0.3, 1.5, 0.3]
+    tv = np.zeros(256, dtype=np.uint8)
+    tv[[0, 1, 2, 3]] = 1
+    eg = np.zeros(256, dtype=np.uint8)
+    eg[2] = 1
+    rng = np.random.default_rng(9)
+    nb = (rng.random(estate.num_nodes) * 2).astype(np.float32)
+    ng = (rng.random(estate.num_nodes) < 0.1).astype(np.uint8)
+
+    seed = np.zeros(estate.num_nodes * 2, dtype=np.uint64)
+    entries = np.arange(estate.n_agents, dtype=np.int64)
+    from agentbom_amd.graph.path_engine import _SEED_PACK
+
+    seed[entries * 2] = np.uint64(_SEED_PACK)
+
+    gpu_levels = _run_gpu(src, dst, et, None, seed, nb, eb, tv, eg, ng,
+                          4, dev)
+    cur = seed
+    for d in range(1, 5):
+        cur = cpu_ref.path_relax(src, dst, et, None, cur, nb, eb, tv, eg, ng)
+        assert np.array_equal(gpu_levels[d], cur), f"hop {d} labels diverge"
+
+
+def test_engine_attack_paths_gpu_matches_cpu(estate, dev):
+    """engine.attack_paths end-to-end: GPU DP == CPU oracle paths."""
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+
+    eng_g = EstateEngine(estate, device=str(dev))
+    hits_g = eng_g.attack_paths(k=50)
+    eng_c = EstateEngine(estate, device="cpu")
+    hits_c = eng_c.attack_paths(k=50)
+    assert len(hits_g) == len(hits_c)
+    for a, b in zip(hits_g, hits_c):
+        assert a.nodes == b.nodes
+        assert a.score == b.score
+        assert a.etypes == b.etypes
