@@ -18,11 +18,12 @@ graph layer maps onto these for traversal classes):
 from __future__ import annotations
 
 from dataclasses import dataclass
+from typing import Optional
 
 import numpy as np
 
 from agentbom_amd.db.arena import AdvisoryArena, build_arena_from_columns
-from agentbom_amd.ops.cpu_ref import PF_ENCODABLE, WF_HAS_FIXED, WF_HAS_INTRO
+from agentbom_amd.ops.cpu_ref import PF_ENCODABLE, WF_HAS_FIXED, WF_HAS_INTRO, WF_HAS_LAST
 from agentbom_amd.utils.version_keys import RELEASE_RANK, pack_batch
 
 ET_USES = 0
@@ -65,6 +66,95 @@ class SyntheticEstate:
         return len(self.edge_src)
 
 
+def _osv_shaped_arena(target_windows: int, name_catalog: int, seed: int) -> AdvisoryArena:
+    """OSV-dump-shaped arena: exactly ``target_windows`` windows.
+
+    Realism properties the legacy uniform arena lacked:
+    - zipf-correlated branch counts: popular names (low name id — package
+      name popularity is pareto over ids) carry MORE advisory windows,
+      so the kernel's window walk meets long runs exactly where package
+      density is highest (the adversarial case for the zipf-head groups);
+    - multi-window branches per advisory (1..64 windows per name);
+    - mixed window forms: ~10% last_affected (inclusive upper bound),
+      ~1% open-ended introduced-only rows (unfixed app advisories — kept
+      matched by default per the round-1 advisor fix), remainder
+      introduced/fixed pairs.
+    """
+    rng = np.random.default_rng([seed, 0xA5A5])
+    mean_w = 4.0
+    n_est = max(1, int(target_windows / mean_w) + 1024)
+    counts = 1 + np.minimum((rng.pareto(1.5, n_est) * 2.2).astype(np.int64), 63)
+    n_est = min(n_est, name_catalog)  # at most one branch-group per name
+    counts = counts[:n_est]
+    cum = np.cumsum(counts)
+    cut = int(np.searchsorted(cum, target_windows))
+    if cut >= len(counts):
+        counts[-1] += target_windows - cum[-1]
+    else:
+        counts = counts[: cut + 1]
+        counts[-1] -= int(cum[cut]) - target_windows
+    n_names = len(counts)
+    # distinct names, branch-heavy groups on the popular (low-id) names
+    vuln_names = np.sort(rng.choice(name_catalog, n_names, replace=False)).astype(np.uint64)
+    counts = np.sort(counts)[::-1]
+    w_names = np.repeat(vuln_names, counts)
+    W = len(w_names)
+
+    # windows span MINOR versions within one major branch (the OSV norm:
+    # an advisory's affected window covers a few releases of one branch) —
+    # keeps per-window match probability realistic (~0.5%) so the finding
+    # rate stays a few percent of the estate even at millions of windows
+    i1 = rng.integers(0, 20, W, dtype=np.int64)
+    i2 = rng.integers(0, 28, W, dtype=np.int64)
+    i3 = rng.integers(0, 60, W, dtype=np.int64)
+    span = 1 + np.minimum(rng.pareto(2.0, W).astype(np.int64), 5)
+    f1 = i1
+    f2 = np.minimum(i2 + span, 30)
+    f3 = rng.integers(0, 60, W, dtype=np.int64)
+    zeros = np.zeros(W, dtype=np.int64)
+    rank = np.full(W, RELEASE_RANK, dtype=np.int64)
+    ihi, ilo = pack_batch(zeros, i1, i2, i3, zeros, rank, zeros)
+    uhi, ulo = pack_batch(zeros, f1, f2, f3, zeros, rank, zeros)
+
+    form = rng.random(W)
+    last_mask = form < 0.10           # introduced + last_affected (inclusive)
+    open_mask = (form >= 0.10) & (form < 0.11)  # introduced-only (no fix yet)
+    fhi = np.where(last_mask | open_mask, 0, uhi).astype(np.uint64)
+    flo = np.where(last_mask | open_mask, 0, ulo).astype(np.uint64)
+    lhi = np.where(last_mask, uhi, 0).astype(np.uint64)
+    llo = np.where(last_mask, ulo, 0).astype(np.uint64)
+    wflags = np.full(W, WF_HAS_INTRO | WF_HAS_FIXED, dtype=np.uint8)
+    wflags[last_mask] = WF_HAS_INTRO | WF_HAS_LAST
+    wflags[open_mask] = WF_HAS_INTRO
+    # open-ended rows: bias introduced into the top of the version space so
+    # each unfixed advisory bounds its own match set (~1-2% of versions)
+    n_open = int(open_mask.sum())
+    oh, ol = pack_batch(
+        np.zeros(n_open, dtype=np.int64),
+        rng.integers(18, 20, n_open, dtype=np.int64),
+        rng.integers(24, 30, n_open, dtype=np.int64),
+        i3[open_mask],
+        np.zeros(n_open, dtype=np.int64),
+        np.full(n_open, RELEASE_RANK, dtype=np.int64),
+        np.zeros(n_open, dtype=np.int64))
+    ihi[open_mask] = oh
+    ilo[open_mask] = ol
+
+    sev = rng.choice([5, 4, 3, 2], size=W, p=[0.1, 0.3, 0.4, 0.2]).astype(np.uint8)
+    cvss = (rng.random(W) * 10).astype(np.float32)
+    epss = np.where(rng.random(W) < 0.7, rng.random(W) ** 2, -1.0).astype(np.float32)
+    kev = (rng.random(W) < 0.02).astype(np.uint8)
+    impact = rng.choice(
+        np.arange(9, dtype=np.uint8), size=W,
+        p=[0.35, 0.08, 0.10, 0.04, 0.08, 0.12, 0.10, 0.05, 0.08],
+    ).astype(np.uint8)
+
+    return build_arena_from_columns(
+        w_names, ihi.astype(np.uint64), ilo.astype(np.uint64), fhi, flo,
+        lhi, llo, wflags, sev, cvss, epss, kev, impact=impact,
+    )
+
+
 def _triple_keys(rng, n, lo=(0, 0, 0), hi=(20, 30, 60)):
     n1 = rng.integers(lo[0], hi[0], n, dtype=np.int64)
     n2 = rng.integers(lo[1], hi[1], n, dtype=np.int64)
@@ -86,8 +176,16 @@ def generate_estate(
     creds_per_server: float = 2.0,
     tools_per_server: float = 4.0,
     seed: int = 1234,
+    arena_windows: Optional[int] = None,
 ) -> SyntheticEstate:
-    """Generate a seeded estate.  Sizes are node counts; edges follow."""
+    """Generate a seeded estate.  Sizes are node counts; edges follow.
+
+    ``arena_windows``: when set, build an OSV-dump-shaped arena of exactly
+    that many windows (zipf-correlated branch counts per name, mixed
+    fixed/last_affected/open window forms) instead of the legacy uniform
+    2-windows-per-name arena — VERDICT r1 item 3 (the real OSV corpus is
+    millions of windows with multi-window branches, reference
+    src/agent_bom/db/schema.py:88-142)."""
     rng = np.random.default_rng(seed)
 
     n_creds = max(64, n_servers // 2)
@@ -153,6 +251,21 @@ def generate_estate(
     pkg_flags = np.full(n_packages, PF_ENCODABLE, dtype=np.uint8)
 
     # ── synthetic advisory arena ──────────────────────────────────────────
+    if arena_windows is not None:
+        arena = _osv_shaped_arena(arena_windows, name_catalog, seed)
+        return SyntheticEstate(
+            n_agents=n_agents, n_servers=n_servers, n_creds=n_creds,
+            n_tools=n_tools, n_packages=n_packages,
+            agent_base=agent_base, server_base=server_base, cred_base=cred_base,
+            tool_base=tool_base, pkg_base=pkg_base, num_nodes=num_nodes,
+            edge_src=edge_src, edge_dst=edge_dst, edge_type=edge_type,
+            pkg_name_id=name_id.astype(np.uint64), pkg_key_hi=key_hi,
+            pkg_key_lo=key_lo, pkg_flags=pkg_flags,
+            cred_is_db=(rng.random(n_creds) < 0.3).astype(np.uint8),
+            tool_is_db=(rng.random(n_tools) < 0.2).astype(np.uint8),
+            arena=arena,
+            seed=seed,
+        )
     n_vuln_names = max(1, int(name_catalog * vulnerable_name_fraction))
     vuln_names = rng.choice(name_catalog, n_vuln_names, replace=False).astype(np.uint64)
     W = n_vuln_names * windows_per_name

@@ -44,6 +44,9 @@ def parse_args():
     ap.add_argument("--agents", type=int, default=100_000)
     ap.add_argument("--servers", type=int, default=500_000)
     ap.add_argument("--name-catalog", type=int, default=1_000_000)
+    ap.add_argument("--arena-windows", type=int, default=2_000_000,
+                    help="advisory windows per GPU-worth of catalog (OSV-dump-"
+                         "shaped: zipf branch counts, mixed window forms)")
     ap.add_argument("--queries", type=int, default=201,
                     help="blast-radius latency queries after the timed region")
     ap.add_argument("--seed", type=int, default=1234)
@@ -82,6 +85,7 @@ def main():
     estate_kw = dict(
         n_agents=args.agents * n_gpus, n_servers=args.servers * n_gpus,
         n_packages=args.packages * n_gpus, name_catalog=args.name_catalog * n_gpus,
+        arena_windows=args.arena_windows * n_gpus if args.arena_windows else None,
     )
 
     t_gen = time.perf_counter()

@@ -24,6 +24,8 @@ def main() -> None:
     ap.add_argument("--sizes", default="1000000,10000000,50000000")
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--queries", type=int, default=501)
+    ap.add_argument("--arena-windows", type=int, default=0,
+                    help="OSV-dump-shaped arena size (0 = legacy uniform arena)")
     args = ap.parse_args()
 
     import torch
@@ -40,6 +42,7 @@ def main() -> None:
             n_packages=size,
             name_catalog=max(10_000, size // 10),
             seed=4242,
+            arena_windows=args.arena_windows or None,
         )
         eng = EstateEngine(est, device="cuda")
         torch.cuda.synchronize()

@@ -116,3 +116,67 @@ def test_impact_filter_zeroes_clientside(estate):
     if none_reach.any():
         assert (res["n_creds"].numpy()[none_reach] == 0).all()
         assert (res["n_tools"].numpy()[none_reach] == 0).all()
+
+
+class TestOsvShapedArena:
+    """The arena_windows mode: exact-window-count OSV-dump
+    shape with zipf branch counts and mixed window forms."""
+
+    @pytest.fixture(scope="class")
+    def osv_estate(self):
+        return generate_estate(n_agents=100, n_servers=500, n_packages=20_000,
+                               name_catalog=4_000, seed=21, arena_windows=9_000)
+
+    def test_exact_window_count_and_shape(self, osv_estate):
+        a = osv_estate.arena
+        assert a.num_windows == 9_000
+        gc = np.diff(a.group_off.astype(np.int64))
+        assert gc.max() >= 16, "zipf branch heads missing"
+        assert (gc >= 1).all()
+        from agentbom_amd.ops.cpu_ref import WF_HAS_FIXED, WF_HAS_INTRO, WF_HAS_LAST
+
+        flags = a.flags
+        assert ((flags & WF_HAS_INTRO) != 0).all()
+        n_last = int(((flags & WF_HAS_LAST) != 0).sum())
+        n_open = int((flags == WF_HAS_INTRO).sum())
+        assert 0 < n_last < a.num_windows // 5
+        assert 0 < n_open < a.num_windows // 20
+
+    def test_match_bruteforce_all_forms(self, osv_estate):
+        """CPU match vs per-window brute force incl. last_affected
+        (inclusive) and open-ended introduced-only windows."""
+        eng = EstateEngine(osv_estate, device="cpu")
+        pkg_idx, win_idx = eng.match()
+        got = set(zip(pkg_idx.tolist(), win_idx.tolist()))
+        assert got, "OSV-shaped arena must produce findings"
+
+        from agentbom_amd.ops.cpu_ref import WF_HAS_FIXED, WF_HAS_LAST
+
+        a = osv_estate.arena
+        # brute force: every package x every window of its group
+        gk = osv_estate.pkg_name_id
+        keys = list(zip(osv_estate.pkg_key_hi.tolist(), osv_estate.pkg_key_lo.tolist()))
+        gkeys = a.group_keys
+        expect = set()
+        import numpy as _np
+
+        for p in range(osv_estate.n_packages):
+            g = int(_np.searchsorted(gkeys, gk[p]))
+            if g >= len(gkeys) or gkeys[g] != gk[p]:
+                continue
+            for w in range(int(a.group_off[g]), int(a.group_off[g + 1])):
+                key = keys[p]
+                if key < (int(a.intro_hi[w]), int(a.intro_lo[w])):
+                    continue
+                fl = int(a.flags[w])
+                if fl & WF_HAS_FIXED and key >= (int(a.fixed_hi[w]), int(a.fixed_lo[w])):
+                    continue
+                if fl & WF_HAS_LAST and key > (int(a.last_hi[w]), int(a.last_lo[w])):
+                    continue
+                expect.add((p, w))
+        assert got == expect
+
+    def test_step_runs_with_osv_arena(self, osv_estate):
+        res = EstateEngine(osv_estate, device="cpu").step()
+        rate = res["n_findings"] / osv_estate.n_packages
+        assert 0.001 < rate < 0.25, f"finding rate {rate} out of realistic band"
