@@ -142,7 +142,7 @@ def _run_gpu(edge_src, col, etype, edge_weight, seed, node_boost, etype_boost,
     def dev_t(a, dtype):
         if a is None:
             return None
-        if hasattr(a, "device"):
+        if torch.is_tensor(a):  # numpy 2.x arrays also expose .device
             return a.to(device=device, dtype=dtype)
         return torch.from_numpy(np.ascontiguousarray(a)).to(device=device, dtype=dtype)
 
