@@ -9,18 +9,20 @@ reference's DFS walk (src/agent_bom/graph/attack_path_fusion.py:194-377):
   with gate state g (g=1 once the path used a vuln-class edge or touched a
   gated node), packed ((ordered_f32(score) << 32) | winner_edge) so one
   atomicMax ranks by score with deterministic edge-index tie-breaks;
-- per-hop label arrays are kept (N*2 u64 per hop, HBM-resident at estate
-  scale) so every reported path is reconstructed EXACTLY, score re-derived
-  along the walk;
-- targets = gated labels on target-class nodes, best over all hop depths,
-  ranked by (score desc, node id asc) — deterministic.
+- the relaxation never walks straight back to a label's predecessor
+  (2-cycle guard — on lateral agent<->server edges the unguarded max-score
+  walk always ping-pongs the best-boosted server);
+- candidates are ALL (target, depth) gated labels ranked by score; each is
+  reconstructed exactly (score re-derived along the walk) and non-simple
+  walks are rejected — depth-2 labels (agent -> gated server -> jewel) are
+  simple by construction, so every reachable target eventually yields a
+  valid path; the first (best) accepted candidate per target wins;
+- on GPU the label levels stay in HBM: selection is a device topk and
+  reconstruction pulls only the touched labels (small gathers), not the
+  level arrays.
 
-Semantics vs the DFS: the DFS enumerates per (entry, target) and keeps the
-best; the DP merges entries and keeps the best per target (with the entry
-recovered from the reconstruction).  On acyclic estates the best-per-
-target results coincide (tests/test_attack_paths.py asserts it); on cyclic
-graphs the DP may report a non-simple walk, which reconstruction rejects
-(dropped from the top-k, the next candidate takes its place).
+Ranking is (score desc, packed-label tie-break, node asc) — deterministic
+(graph contract: determinism given the same inventory).
 """
 
 from __future__ import annotations
@@ -32,6 +34,7 @@ import numpy as np
 
 NO_EDGE = 0xFFFFFFFF
 _SEED_PACK = (0x80000000 << 32) | NO_EDGE  # ordered_f32(0.0) | no-edge
+_INT64_MIN = -(1 << 63)
 
 
 @dataclass
@@ -46,8 +49,9 @@ class PathHit:
     entry: int
 
 
-def _unpack_score(packed_u64: np.ndarray) -> np.ndarray:
-    hi = (packed_u64 >> np.uint64(32)) & np.uint64(0x7FFFFFFF)
+def _unpack_score(packed_u64) -> np.ndarray:
+    arr = np.asarray(packed_u64, dtype=np.uint64)
+    hi = (arr >> np.uint64(32)) & np.uint64(0x7FFFFFFF)
     return hi.astype(np.uint32).view(np.float32)
 
 
@@ -55,9 +59,67 @@ def _np(a, dtype=None):
     """numpy view of a numpy array or (possibly device) torch tensor."""
     if a is None:
         return None
-    if hasattr(a, "cpu"):
+    if hasattr(a, "cpu") and not isinstance(a, np.ndarray):
         a = a.cpu().numpy()
     return np.asarray(a, dtype=dtype) if dtype else np.asarray(a)
+
+
+class _NumpyFetch:
+    """Label/edge/boost accessors over host arrays (CPU oracle path)."""
+
+    def __init__(self, levels, esrc, ecol, etyp, eweight, nboost):
+        self.levels = levels
+        self.esrc, self.etyp = esrc, etyp
+        self.eweight, self.nboost = eweight, nboost
+
+    def labels(self, d, nodes, g):
+        return self.levels[d][np.asarray(nodes, dtype=np.int64) * 2 + g]
+
+    def edge_src(self, e):
+        return self.esrc[np.asarray(e, dtype=np.int64)]
+
+    def edge_type(self, e):
+        return self.etyp[np.asarray(e, dtype=np.int64)]
+
+    def edge_w(self, e):
+        if self.eweight is None:
+            return None
+        return self.eweight[np.asarray(e, dtype=np.int64)]
+
+    def node_boost(self, v):
+        return self.nboost[np.asarray(v, dtype=np.int64)]
+
+
+class _DeviceFetch:
+    """Same accessors via small device gathers (labels stay in HBM)."""
+
+    def __init__(self, torch, levels_dev, src_t, et_t, ew_t, nb_t):
+        self.torch = torch
+        self.L = levels_dev  # list of int64 [N*2] device tensors
+        self.src_t, self.et_t = src_t, et_t
+        self.ew_t, self.nb_t = ew_t, nb_t
+        self.dev = src_t.device
+
+    def _idx(self, a):
+        return self.torch.from_numpy(np.asarray(a, dtype=np.int64)).to(self.dev)
+
+    def labels(self, d, nodes, g):
+        out = self.L[d][self._idx(np.asarray(nodes, dtype=np.int64) * 2 + g)]
+        return out.cpu().numpy().view(np.uint64)
+
+    def edge_src(self, e):
+        return self.src_t[self._idx(e)].cpu().numpy().astype(np.int64)
+
+    def edge_type(self, e):
+        return self.et_t[self._idx(e)].cpu().numpy()
+
+    def edge_w(self, e):
+        if self.ew_t is None:
+            return None
+        return self.ew_t[self._idx(e)].cpu().numpy()
+
+    def node_boost(self, v):
+        return self.nb_t[self._idx(v)].cpu().numpy()
 
 
 def run_path_dp(
@@ -70,11 +132,7 @@ def run_path_dp(
     k: int = 100,
     device=None,
 ) -> list[PathHit]:
-    """Full DP + reconstruction.  numpy in, PathHit list out.
-
-    On a CUDA ``device`` the relaxation runs the HIP kernel with all label
-    levels resident in HBM; otherwise the numpy oracle.
-    """
+    """Full DP + reconstruction.  Returns ranked PathHit list."""
     use_gpu = device is not None and getattr(device, "type", str(device)).startswith("cuda")
     N = num_nodes
 
@@ -82,55 +140,94 @@ def run_path_dp(
     entries = _np(entries, np.int64)
     seed[entries * 2] = np.uint64(_SEED_PACK)
 
+    eb_np = _np(etype_boost, np.float32)
+    eg_np = _np(etype_gate)
+    ng_np = _np(node_gate)
+
     if use_gpu:
-        levels = _run_gpu(edge_src, col, etype, edge_weight, seed, node_boost,
-                          etype_boost, etype_trav, etype_gate, node_gate,
-                          max_depth, device)
+        import torch
+
+        levels_dev, src_t, et_t, ew_t, nb_t = _run_gpu(
+            edge_src, col, etype, edge_weight, seed, node_boost,
+            etype_boost, etype_trav, etype_gate, node_gate, max_depth, device)
+        fetch = _DeviceFetch(torch, levels_dev, src_t, et_t, ew_t, nb_t)
+        cand_nodes, cand_depth, cand_packed = _select_device(
+            torch, levels_dev, target_mask, k, device)
     else:
         from agentbom_amd.ops import cpu_ref
 
-        edge_src, col, etype = _np(edge_src), _np(col), _np(etype)
-        edge_weight = _np(edge_weight)
-        node_boost, etype_boost = _np(node_boost), _np(etype_boost)
-        etype_trav, etype_gate = _np(etype_trav), _np(etype_gate)
-        node_gate = _np(node_gate)
+        esrc_np, col_np, et_np = _np(edge_src), _np(col), _np(etype)
+        ew_np = _np(edge_weight)
+        nb_np = _np(node_boost, np.float32)
+        tv_np = _np(etype_trav)
         levels = [seed]
         for _ in range(max_depth):
             levels.append(cpu_ref.path_relax(
-                edge_src, col, etype, edge_weight, levels[-1], node_boost,
-                etype_boost, etype_trav, etype_gate, node_gate))
+                esrc_np, col_np, et_np, ew_np, levels[-1], nb_np,
+                eb_np, tv_np, eg_np, ng_np))
+        fetch = _NumpyFetch(levels, esrc_np, col_np, et_np, ew_np, nb_np)
+        cand_nodes, cand_depth, cand_packed = _select_numpy(levels, _np(target_mask), k)
 
-    # best gated label per target node over all depths >= 1
-    tgt_nodes = np.nonzero(_np(target_mask))[0]
-    if not len(tgt_nodes):
-        return []
-    stacked = np.stack([lv[tgt_nodes * 2 + 1] for lv in levels[1:]])  # [D, T]
-    best_d = np.argmax(stacked, axis=0)
-    best = stacked[best_d, np.arange(len(tgt_nodes))]
-    hit = best != 0
-    if not hit.any():
-        return []
-    cand_nodes = tgt_nodes[hit]
-    cand_depth = best_d[hit] + 1
-    cand_packed = best[hit]
-    cand_score = _unpack_score(cand_packed)
-    # rank: score desc, node id asc (deterministic); take extra candidates
-    # because cyclic reconstructions may be rejected
-    order = np.lexsort((cand_nodes, -cand_score))
-    order = order[: max(k * 2, k + 16)]
-
+    # reconstruct in rank order; first accepted candidate per target wins
     out: list[PathHit] = []
-    esrc, ecol, etyp = _np(edge_src), _np(col), _np(etype)
-    ew_np, nb_np, eb_np = _np(edge_weight), _np(node_boost), _np(etype_boost)
-    eg_np, ng_np = _np(etype_gate), _np(node_gate)
-    for i in order:
-        if len(out) >= k:
+    done_targets: set[int] = set()
+    attempts = 0
+    max_attempts = max(64, 16 * k)
+    for node, depth, packed in zip(cand_nodes, cand_depth, cand_packed):
+        if len(out) >= k or attempts >= max_attempts:
             break
-        hitp = _reconstruct(levels, int(cand_nodes[i]), int(cand_depth[i]),
-                            esrc, ecol, etyp, ew_np, nb_np, eb_np, eg_np, ng_np)
-        if hitp is not None:
-            out.append(hitp)
+        t = int(node)
+        if t in done_targets:
+            continue
+        attempts += 1
+        hit = _reconstruct(fetch, t, int(depth), eb_np, eg_np, ng_np)
+        if hit is not None:
+            out.append(hit)
+            done_targets.add(t)
     return out
+
+
+def _select_numpy(levels, target_mask, k):
+    """All (target, depth) gated candidates ranked by score (numpy)."""
+    tgt = np.nonzero(target_mask)[0]
+    if not len(tgt):
+        return [], [], []
+    stacked = np.stack([lv[tgt * 2 + 1] for lv in levels[1:]])  # [D, T]
+    d_idx, t_idx = np.nonzero(stacked)
+    if not len(d_idx):
+        return [], [], []
+    packed = stacked[d_idx, t_idx]
+    nodes = tgt[t_idx]
+    # exact u64 ordering: ~packed ascending == packed (score) descending
+    order = np.lexsort((nodes, ~packed))
+    return nodes[order], d_idx[order] + 1, packed[order]
+
+
+def _select_device(torch, levels_dev, target_mask, k, device):
+    """Device-side candidate selection: no full-level host copies."""
+    tm = target_mask
+    if not torch.is_tensor(tm):
+        tm = torch.from_numpy(np.ascontiguousarray(tm)).to(device)
+    tgt = torch.nonzero(tm.to(torch.bool)).flatten()
+    if not tgt.numel():
+        return [], [], []
+    gidx = tgt * 2 + 1
+    rows = [lv[gidx] for lv in levels_dev[1:]]
+    stacked = torch.stack(rows)  # [D, T] int64 (valid labels are negative)
+    flat = stacked.flatten()
+    valid = flat != 0
+    ranked = torch.where(valid, flat, torch.full_like(flat, _INT64_MIN))
+    n_take = min(int(valid.sum().item()), max(64, 16 * k))
+    if n_take == 0:
+        return [], [], []
+    topv, topi = torch.topk(ranked, n_take)
+    keep = topv != _INT64_MIN
+    topv, topi = topv[keep], topi[keep]
+    T = tgt.numel()
+    d_idx = (topi // T).cpu().numpy()
+    nodes = tgt[(topi % T)].cpu().numpy()
+    packed = topv.cpu().numpy().view(np.uint64)
+    return nodes, d_idx + 1, packed
 
 
 def _run_gpu(edge_src, col, etype, edge_weight, seed, node_boost, etype_boost,
@@ -163,54 +260,65 @@ def _run_gpu(edge_src, col, etype, edge_weight, seed, node_boost, etype_boost,
         native.path_relax(src_t, col_t, et_t, ew_t, levels_dev[-1], nxt,
                           nb_t, eb_t, tv_t, eg_t, ng_t)
         levels_dev.append(nxt)
-    return [lv.cpu().numpy().view(np.uint64) for lv in levels_dev]
+    return levels_dev, src_t, et_t, ew_t, nb_t
 
 
-def _reconstruct(levels, node: int, depth: int, esrc, ecol, etyp, edge_weight,
-                 node_boost, etype_boost, etype_gate, node_gate) -> Optional[PathHit]:
+def _reconstruct(fetch, node: int, depth: int, eb_np, eg_np, ng_np) -> Optional[PathHit]:
     """Walk winner edges back to the entry; None for non-simple walks.
 
-    Gate-source disambiguation mirrors the kernel's atomicMax: the stored
-    score equals max over feasible predecessor labels + step; ties prefer
-    the gated predecessor (fixed rule, documented in the module docstring).
-    """
+    Gate-source disambiguation mirrors the kernel: the stored score equals
+    the feasible predecessor label + step; a predecessor label whose own
+    predecessor is this hop's target was blocked by the 2-cycle guard and
+    is not feasible.  Ties prefer the gated predecessor (fixed rule)."""
     f32 = np.float32
     g = 1
     v, d = node, depth
     nodes = [node]
     edges: list[int] = []
     etypes: list[int] = []
-    score = float(_unpack_score(np.array([levels[d][v * 2 + g]], dtype=np.uint64))[0])
+    packed0 = int(fetch.labels(d, [v], g)[0])
+    if packed0 == 0:
+        return None
+    score = float(_unpack_score([packed0])[0])
     while d > 0:
-        packed = int(levels[d][v * 2 + g])
+        packed = int(fetch.labels(d, [v], g)[0])
         if packed == 0:
             return None
         e = packed & NO_EDGE
         if e == NO_EDGE:
-            break  # reached a seed label
-        u = int(esrc[e])
-        et = int(etyp[e])
-        step = f32(etype_boost[et]) + f32(node_boost[v])
-        if edge_weight is not None:
-            step = f32(step + f32(edge_weight[e]) * f32(0.3))
-        my_score = _unpack_score(np.array([packed], dtype=np.uint64))[0]
-        gate_edge = bool(etype_gate[et]) or (node_gate is not None and bool(node_gate[v]))
-        prev = levels[d - 1]
+            break  # seed label
+        u = int(fetch.edge_src([e])[0])
+        et = int(fetch.edge_type([e])[0])
+        step = f32(eb_np[et]) + f32(fetch.node_boost([v])[0])
+        w = fetch.edge_w([e])
+        if w is not None:
+            step = f32(step + f32(w[0]) * f32(0.3))
+        my_score = _unpack_score([packed])[0]
+        gate_edge = bool(eg_np[et]) or (ng_np is not None and bool(ng_np[v]))
+        prev = fetch.labels(d - 1, [u, u], [0, 1])
+        lu0, lu1 = int(prev[0]), int(prev[1])
+
+        def feasible(lab):
+            if not lab:
+                return False
+            if f32(_unpack_score([lab])[0] + step) != my_score:
+                return False
+            pe = lab & NO_EDGE
+            if pe != NO_EDGE and int(fetch.edge_src([pe])[0]) == v:
+                return False  # blocked by the 2-cycle guard in relax
+            return True
+
         g_next = None
         if g == 1:
-            lu1 = int(prev[u * 2 + 1])
-            if lu1 and f32(_unpack_score(np.array([lu1], dtype=np.uint64))[0] + step) == my_score:
+            if feasible(lu1):
                 g_next = 1
-            if g_next is None and gate_edge:
-                lu0 = int(prev[u * 2])
-                if lu0 and f32(_unpack_score(np.array([lu0], dtype=np.uint64))[0] + step) == my_score:
-                    g_next = 0
+            elif gate_edge and feasible(lu0):
+                g_next = 0
         else:
-            lu0 = int(prev[u * 2])
-            if lu0 and f32(_unpack_score(np.array([lu0], dtype=np.uint64))[0] + step) == my_score:
+            if feasible(lu0):
                 g_next = 0
         if g_next is None:
-            return None  # inconsistent label chain (shouldn't happen)
+            return None  # inconsistent label chain
         edges.append(int(e))
         etypes.append(et)
         nodes.append(u)

@@ -218,14 +218,24 @@ def path_relax(edge_src, col, etype, edge_weight, cur, node_boost,
     if node_gate is not None:
         gate = gate | node_gate[v].astype(bool)
 
+    esrc_all = np.asarray(edge_src, dtype=np.int64)
+
+    def not_backtrack(labels):
+        # predecessor-avoidance: block relaxing straight back to the node
+        # the label came from (matches the kernel's 2-cycle guard)
+        pe = (labels & np.uint64(0xFFFFFFFF)).astype(np.int64)
+        no_edge = pe == 0xFFFFFFFF
+        pred = np.where(no_edge, -1, esrc_all[np.where(no_edge, 0, pe)])
+        return no_edge | (pred != v)
+
     lu0 = cur[u * 2]
-    m = lu0 != 0
+    m = (lu0 != 0) & not_backtrack(lu0)
     if m.any():
         s = path_unordered_f32(lu0[m] >> np.uint64(32)) + step[m]
         cand = path_pack(s, e_idx[m])
         np.maximum.at(nxt, v[m] * 2 + gate[m], cand)
     lu1 = cur[u * 2 + 1]
-    m = lu1 != 0
+    m = (lu1 != 0) & not_backtrack(lu1)
     if m.any():
         s = path_unordered_f32(lu1[m] >> np.uint64(32)) + step[m]
         cand = path_pack(s, e_idx[m])

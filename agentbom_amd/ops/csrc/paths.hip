@@ -69,13 +69,22 @@ __global__ void path_relax_kernel(
         if (edge_weight) step += edge_weight[e] * 0.3f;
         bool gate = etype_gate[et] || (node_gate && node_gate[v]);
         if (lu0) {
-            float s = unordered_f32((uint32_t)(lu0 >> 32)) + step;
-            unsigned long long cand = pack_label(s, (uint32_t)e);
-            atomicMax(&nxt[(long long)v * 2 + (gate ? 1 : 0)], cand);
+            // predecessor-avoidance: never relax straight back to the node
+            // this label came from (kills 2-cycles, the dominant non-simple
+            // walk class on lateral agent<->server edges)
+            uint32_t pe = (uint32_t)lu0;
+            if (pe == 0xFFFFFFFFu || edge_src[pe] != v) {
+                float s = unordered_f32((uint32_t)(lu0 >> 32)) + step;
+                unsigned long long cand = pack_label(s, (uint32_t)e);
+                atomicMax(&nxt[(long long)v * 2 + (gate ? 1 : 0)], cand);
+            }
         }
         if (lu1) {
-            float s = unordered_f32((uint32_t)(lu1 >> 32)) + step;
-            atomicMax(&nxt[(long long)v * 2 + 1], pack_label(s, (uint32_t)e));
+            uint32_t pe = (uint32_t)lu1;
+            if (pe == 0xFFFFFFFFu || edge_src[pe] != v) {
+                float s = unordered_f32((uint32_t)(lu1 >> 32)) + step;
+                atomicMax(&nxt[(long long)v * 2 + 1], pack_label(s, (uint32_t)e));
+            }
         }
     }
 }

@@ -404,12 +404,13 @@ def test_path_relax_parity(estate, dev):
 
     seed[entries * 2] = np.uint64(_SEED_PACK)
 
-    gpu_levels = _run_gpu(src, dst, et, None, seed, nb, eb, tv, eg, ng,
-                          4, dev)
+    levels_dev, *_ = _run_gpu(src, dst, et, None, seed, nb, eb, tv, eg, ng,
+                              4, dev)
     cur = seed
     for d in range(1, 5):
         cur = cpu_ref.path_relax(src, dst, et, None, cur, nb, eb, tv, eg, ng)
-        assert np.array_equal(gpu_levels[d], cur), f"hop {d} labels diverge"
+        got = levels_dev[d].cpu().numpy().view(np.uint64)
+        assert np.array_equal(got, cur), f"hop {d} labels diverge"
 
 
 def test_engine_attack_paths_gpu_matches_cpu(estate, dev):
