@@ -54,6 +54,31 @@ def _impact_lut(full: set, db: set):
     return lut
 
 
+_SIGN64 = -0x8000000000000000  # xor flips int64 order into u64 order
+
+
+def _precompute_win_ranges(torch, arena: dict, pkg_group_key_sorted):
+    """(wbeg u32[P], wend u32[P]) advisory window range per package row.
+
+    Keys are u64 hashes stored as int64 bit patterns; xor with the sign bit
+    makes torch.searchsorted order match the kernel's unsigned compare."""
+    gk = arena["group_keys"]
+    G = gk.numel()
+    P = pkg_group_key_sorted.numel()
+    dev = pkg_group_key_sorted.device
+    if G == 0:
+        z = torch.zeros(P, dtype=torch.int32, device=dev)
+        return z, z.clone()
+    g = torch.searchsorted(gk ^ _SIGN64, pkg_group_key_sorted ^ _SIGN64)
+    gc = g.clamp(max=G - 1)
+    valid = (g < G) & (gk[gc] == pkg_group_key_sorted)
+    off = arena["group_off"].to(torch.int32)
+    zero = torch.zeros(P, dtype=torch.int32, device=dev)
+    wbeg = torch.where(valid, off[gc], zero)
+    wend = torch.where(valid, off[gc + 1], zero)
+    return wbeg.contiguous(), wend.contiguous()
+
+
 class EstateEngine:
     """Device-resident estate + advisory arena + the findings pipeline."""
 
@@ -88,6 +113,11 @@ class EstateEngine:
         self.pkg_flags_sorted = self.pkg_flags[self.pkg_perm].contiguous()
 
         self.arena = estate.arena.to_torch(dev)
+        # resident-estate match index: per-package advisory window range,
+        # precomputed once (estate and arena are static between steps) so
+        # the match kernel skips the per-package group binary search
+        self.pkg_win_range = _precompute_win_ranges(
+            torch, self.arena, self.pkg_group_key_sorted)
         self.cred_is_db = torch.from_numpy(estate.cred_is_db).to(dev)
         self.tool_is_db = torch.from_numpy(estate.tool_is_db).to(dev)
         self.cred_lut = torch.from_numpy(_impact_lut(_FULL_CRED, _DB_CRED)).to(dev)
@@ -138,6 +168,7 @@ class EstateEngine:
                 self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
                 self.pkg_key_lo_sorted, self.pkg_flags_sorted,
                 self.arena["group_keys"], self.arena["group_off"], self.arena["windows"],
+                pkg_win_range=self.pkg_win_range,
             )
             # map sorted rows back to original package indices, re-sort for
             # the deterministic (pkg, window) output order
@@ -324,7 +355,7 @@ class EstateEngine:
                     self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
                     self.pkg_key_lo_sorted, self.pkg_flags_sorted,
                     self.arena["group_keys"], self.arena["group_off"],
-                    self.arena["windows"])
+                    self.arena["windows"], pkg_win_range=self.pkg_win_range)
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
             torch.cuda.current_stream().wait_stream(side)
             sp, sw = native.match_finalize(pending)

@@ -29,7 +29,7 @@ _SIGNATURES = {
     "abom_device_count": ([], _i32),
     "abom_synchronize": ([_c], _i32),
     "abom_error_string": ([_i32], ctypes.c_char_p),
-    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c, _c, _i64, _c], _i32),
+    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c, _c] + [_c, _c, _i64, _c], _i32),
     "abom_bfs_init": ([_c, _i64, _c], _i32),
     "abom_bfs_seed": ([_c, _i64, _c, _c, _c, _c, _c], _i32),
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
@@ -114,11 +114,14 @@ def _stream() -> ctypes.c_void_p:
 
 
 def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
-                 group_off, windows: dict, capacity: Optional[int] = None) -> dict:
+                 group_off, windows: dict, capacity: Optional[int] = None,
+                 pkg_win_range=None) -> dict:
     """Launch the match kernel on the CURRENT stream without syncing.
 
     Returns a pending handle for :func:`match_finalize`.  Used by the engine
     to overlap the match with the reach BFS on a second HIP stream.
+    ``pkg_win_range``: optional (wbeg u32[P], wend u32[P]) precomputed by the
+    engine for a resident estate — the kernel then skips the group search.
     """
     import torch
 
@@ -131,6 +134,9 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
     dev = pkg_group_key.device
     out_pairs = torch.empty(cap, dtype=torch.int64, device=dev)
     out_count = torch.zeros(1, dtype=torch.int32, device=dev)
+    wbeg_p = wend_p = None
+    if pkg_win_range is not None:
+        wbeg_p, wend_p = _ptr(pkg_win_range[0]), _ptr(pkg_win_range[1])
     rc = lib.abom_match(
         _ptr(pkg_group_key), _ptr(pkg_key_hi), _ptr(pkg_key_lo), _ptr(pkg_flags), P,
         _ptr(group_keys), _ptr(group_off), G,
@@ -138,12 +144,14 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
         _ptr(windows["fixed_hi"]), _ptr(windows["fixed_lo"]),
         _ptr(windows["last_hi"]), _ptr(windows["last_lo"]),
         _ptr(windows["flags"]),
+        wbeg_p, wend_p,
         _ptr(out_pairs), _ptr(out_count), cap, _stream(),
     )
     _check(rc, "abom_match")
     return {"out_pairs": out_pairs, "out_count": out_count, "cap": cap,
             "args": (pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags,
-                     group_keys, group_off, windows)}
+                     group_keys, group_off, windows),
+            "kw": {"pkg_win_range": pkg_win_range}}
 
 
 def match_finalize(pending: dict):
@@ -160,14 +168,15 @@ def match_finalize(pending: dict):
         # rather than slicing with a wrapped count.
         raise RuntimeError("match: device match count overflowed int32; shard the package batch")
     if n > pending["cap"]:
-        return match(*pending["args"], capacity=int(n * 1.2) + 1024)
+        return match(*pending["args"], capacity=int(n * 1.2) + 1024,
+                     **pending.get("kw", {}))
     pairs = pending["out_pairs"][:n]
     pairs, _ = torch.sort(pairs)
     return (pairs >> 32).to(torch.int64), (pairs & 0xFFFFFFFF).to(torch.int64)
 
 
 def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_off,
-          windows: dict, capacity: Optional[int] = None):
+          windows: dict, capacity: Optional[int] = None, pkg_win_range=None):
     """Bulk version-range match.  Returns sorted (pkg_idx u32, window_idx u32).
 
     All tensors device-resident; ``windows`` carries intro/fixed/last hi+lo
@@ -175,7 +184,8 @@ def match(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys, group_of
     capacity overflows.
     """
     pending = match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags,
-                           group_keys, group_off, windows, capacity=capacity)
+                           group_keys, group_off, windows, capacity=capacity,
+                           pkg_win_range=pkg_win_range)
     return match_finalize(pending)
 
 

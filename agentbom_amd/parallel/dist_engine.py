@@ -24,7 +24,14 @@ from __future__ import annotations
 
 import numpy as np
 
-from agentbom_amd.graph.gpu_engine import _DB_CRED, _DB_TOOL, _FULL_CRED, _FULL_TOOL, _impact_lut
+from agentbom_amd.graph.gpu_engine import (
+    _DB_CRED,
+    _DB_TOOL,
+    _FULL_CRED,
+    _FULL_TOOL,
+    _impact_lut,
+    _precompute_win_ranges,
+)
 from agentbom_amd.parallel.dist_blast import distributed_blast_counts
 from agentbom_amd.parallel.dist_bfs import distributed_reach
 from agentbom_amd.parallel.partition import build_csr, partition_estate
@@ -75,6 +82,8 @@ class DistEstateEngine:
         self.pkg_flags_sorted = fl[self.pkg_perm].contiguous()
 
         self.arena = estate.arena.to_torch(dev)
+        self.pkg_win_range = _precompute_win_ranges(
+            torch, self.arena, self.pkg_group_key_sorted)
         self.cred_lut = torch.from_numpy(_impact_lut(_FULL_CRED, _DB_CRED)).to(dev)
         self.tool_lut = torch.from_numpy(_impact_lut(_FULL_TOOL, _DB_TOOL)).to(dev)
 
@@ -108,7 +117,7 @@ class DistEstateEngine:
                 self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
                 self.pkg_key_lo_sorted, self.pkg_flags_sorted,
                 self.arena["group_keys"], self.arena["group_off"],
-                self.arena["windows"],
+                self.arena["windows"], pkg_win_range=self.pkg_win_range,
             )
             return sp, sw
         from agentbom_amd.ops import cpu_ref
@@ -153,7 +162,7 @@ class DistEstateEngine:
                     self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
                     self.pkg_key_lo_sorted, self.pkg_flags_sorted,
                     self.arena["group_keys"], self.arena["group_off"],
-                    self.arena["windows"])
+                    self.arena["windows"], pkg_win_range=self.pkg_win_range)
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
             torch.cuda.current_stream().wait_stream(side)
             sp, sw = native.match_finalize(pending)

@@ -16,6 +16,7 @@ def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--packages", type=int, default=10_000_000)
     ap.add_argument("--iters", type=int, default=10)
+    ap.add_argument("--arena-windows", type=int, default=0)
     args = ap.parse_args()
 
     import torch
@@ -29,6 +30,7 @@ def main() -> None:
         n_packages=args.packages,
         name_catalog=max(10_000, args.packages // 10),
         seed=1234,
+        arena_windows=args.arena_windows or None,
     )
     eng = EstateEngine(est, device="cuda")
     eng.step()

@@ -426,3 +426,27 @@ def test_engine_attack_paths_gpu_matches_cpu(estate, dev):
         assert a.nodes == b.nodes
         assert a.score == b.score
         assert a.etypes == b.etypes
+
+
+def test_match_precomputed_ranges_parity(estate, dev):
+    """Resident-estate mode (precomputed per-package window ranges) must
+    return exactly the same pairs as the searching kernel path."""
+    from agentbom_amd.graph.gpu_engine import _precompute_win_ranges
+    from agentbom_amd.ops import native
+
+    arena_t = estate.arena.to_torch(dev)
+    gk = torch.from_numpy(estate.pkg_name_id.view(np.int64)).to(dev)
+    perm = torch.argsort(gk, stable=True)
+    gk_s = gk[perm].contiguous()
+    hi_s = torch.from_numpy(estate.pkg_key_hi.view(np.int64)).to(dev)[perm].contiguous()
+    lo_s = torch.from_numpy(estate.pkg_key_lo.view(np.int64)).to(dev)[perm].contiguous()
+    fl_s = torch.from_numpy(estate.pkg_flags).to(dev)[perm].contiguous()
+
+    base = native.match(gk_s, hi_s, lo_s, fl_s, arena_t["group_keys"],
+                        arena_t["group_off"], arena_t["windows"])
+    rng = _precompute_win_ranges(torch, arena_t, gk_s)
+    fast = native.match(gk_s, hi_s, lo_s, fl_s, arena_t["group_keys"],
+                        arena_t["group_off"], arena_t["windows"],
+                        pkg_win_range=rng)
+    assert torch.equal(base[0], fast[0])
+    assert torch.equal(base[1], fast[1])
