@@ -79,6 +79,65 @@ def _precompute_win_ranges(torch, arena: dict, pkg_group_key_sorted):
     return wbeg.contiguous(), wend.contiguous()
 
 
+def build_dedup_match_layout(torch, arena: dict, gk, hi, lo, flags):
+    """Dedup (group, version, flags) rows for the resident-estate match.
+
+    Real estates install the same (name, version) on many servers; the
+    reference dedups before matching (package_scan.py:1794-1816).  The
+    engine mirrors that: match runs once per DISTINCT row, results fan back
+    out to package rows.  Returns None when dedup would not pay
+    (>90% distinct).  Layout:
+      u_*       unique row columns (lexicographically sorted)
+      run_off   int64 [U+1]: unique row -> its span in ``perm2``
+      perm2     int64 [P]: sorted-row -> original package index
+      u_ranges  precomputed window ranges for the unique rows
+    """
+    P = gk.numel()
+    # lexicographic stable sort by (gk, hi, lo, flags)
+    idx = torch.argsort(flags.to(torch.int64), stable=True)
+    for keycol in (lo, hi, gk):
+        idx = idx[torch.argsort(keycol[idx], stable=True)]
+    g2, h2, l2, f2 = gk[idx], hi[idx], lo[idx], flags[idx]
+    neq = (g2[1:] != g2[:-1]) | (h2[1:] != h2[:-1]) | (l2[1:] != l2[:-1]) \
+        | (f2[1:] != f2[:-1])
+    starts = torch.cat([
+        torch.zeros(1, dtype=torch.int64, device=gk.device),
+        torch.nonzero(neq).flatten() + 1,
+    ])
+    U = starts.numel()
+    if U > int(P * 0.9):
+        return None
+    run_off = torch.cat([starts, torch.tensor([P], dtype=torch.int64, device=gk.device)])
+    layout = {
+        "u_gk": g2[starts].contiguous(),
+        "u_hi": h2[starts].contiguous(),
+        "u_lo": l2[starts].contiguous(),
+        "u_flags": f2[starts].contiguous(),
+        "run_off": run_off,
+        "perm2": idx,
+    }
+    layout["u_ranges"] = _precompute_win_ranges(torch, arena, layout["u_gk"])
+    return layout
+
+
+def expand_dedup_matches(torch, layout: dict, su, sw):
+    """Fan unique-row matches back out to per-package (orig_idx, win) pairs,
+    sorted by (pkg << 32 | win) — identical to the non-dedup output."""
+    run_off = layout["run_off"]
+    cnt = run_off[su + 1] - run_off[su]
+    rep = torch.repeat_interleave(
+        torch.arange(su.numel(), device=su.device), cnt)
+    ends = torch.cumsum(cnt, 0)
+    base = torch.cat([torch.zeros(1, dtype=torch.int64, device=su.device), ends[:-1]])
+    within = torch.arange(int(ends[-1].item()) if ends.numel() else 0,
+                          device=su.device) - base[rep]
+    rows2 = run_off[su][rep] + within
+    orig = layout["perm2"][rows2]
+    packed = (orig << 32) | sw[rep]
+    packed, _ = torch.sort(packed)
+    return (packed >> 32), (packed & 0xFFFFFFFF)
+
+
 class EstateEngine:
     """Device-resident estate + advisory arena + the findings pipeline."""
 
@@ -118,6 +177,10 @@ class EstateEngine:
         # the match kernel skips the per-package group binary search
         self.pkg_win_range = _precompute_win_ranges(
             torch, self.arena, self.pkg_group_key_sorted)
+        # dedup-match layout (None when the estate is >90% distinct rows)
+        self.match_dedup = build_dedup_match_layout(
+            torch, self.arena, self.pkg_group_key, self.pkg_key_hi,
+            self.pkg_key_lo, self.pkg_flags) if self.use_gpu else None
         self.cred_is_db = torch.from_numpy(estate.cred_is_db).to(dev)
         self.tool_is_db = torch.from_numpy(estate.tool_is_db).to(dev)
         self.cred_lut = torch.from_numpy(_impact_lut(_FULL_CRED, _DB_CRED)).to(dev)
@@ -350,19 +413,31 @@ class EstateEngine:
                 self._match_stream = torch.cuda.Stream(device=self.device)
             side = self._match_stream
             side.wait_stream(torch.cuda.current_stream())
+            dd = self.match_dedup
             with torch.cuda.stream(side):
-                pending = native.match_launch(
-                    self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
-                    self.pkg_key_lo_sorted, self.pkg_flags_sorted,
-                    self.arena["group_keys"], self.arena["group_off"],
-                    self.arena["windows"], pkg_win_range=self.pkg_win_range)
+                if dd is not None:
+                    # dedup-match: one kernel row per DISTINCT (name,
+                    # version); results fan back out below
+                    pending = native.match_launch(
+                        dd["u_gk"], dd["u_hi"], dd["u_lo"], dd["u_flags"],
+                        self.arena["group_keys"], self.arena["group_off"],
+                        self.arena["windows"], pkg_win_range=dd["u_ranges"])
+                else:
+                    pending = native.match_launch(
+                        self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
+                        self.pkg_key_lo_sorted, self.pkg_flags_sorted,
+                        self.arena["group_keys"], self.arena["group_off"],
+                        self.arena["windows"], pkg_win_range=self.pkg_win_range)
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
             torch.cuda.current_stream().wait_stream(side)
             sp, sw = native.match_finalize(pending)
-            orig = self.pkg_perm[sp]
-            packed = (orig << 32) | sw
-            packed, _ = torch.sort(packed)
-            pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
+            if dd is not None:
+                pkg_idx, win_idx = expand_dedup_matches(torch, dd, sp, sw)
+            else:
+                orig = self.pkg_perm[sp]
+                packed = (orig << 32) | sw
+                packed, _ = torch.sort(packed)
+                pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
         else:
             pkg_idx, win_idx = self.match()
             dist = reach_dist if reach_dist is not None else self.dependency_reach()

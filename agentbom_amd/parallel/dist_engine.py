@@ -31,6 +31,8 @@ from agentbom_amd.graph.gpu_engine import (
     _FULL_TOOL,
     _impact_lut,
     _precompute_win_ranges,
+    build_dedup_match_layout,
+    expand_dedup_matches,
 )
 from agentbom_amd.parallel.dist_blast import distributed_blast_counts
 from agentbom_amd.parallel.dist_bfs import distributed_reach
@@ -84,6 +86,8 @@ class DistEstateEngine:
         self.arena = estate.arena.to_torch(dev)
         self.pkg_win_range = _precompute_win_ranges(
             torch, self.arena, self.pkg_group_key_sorted)
+        self.match_dedup = build_dedup_match_layout(
+            torch, self.arena, gk, khi, klo, fl) if self.use_gpu else None
         self.cred_lut = torch.from_numpy(_impact_lut(_FULL_CRED, _DB_CRED)).to(dev)
         self.tool_lut = torch.from_numpy(_impact_lut(_FULL_TOOL, _DB_TOOL)).to(dev)
 
@@ -157,24 +161,42 @@ class DistEstateEngine:
                 self._match_stream = torch.cuda.Stream(device=self.device)
             side = self._match_stream
             side.wait_stream(torch.cuda.current_stream())
+            dd = self.match_dedup
             with torch.cuda.stream(side):
-                pending = native.match_launch(
-                    self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
-                    self.pkg_key_lo_sorted, self.pkg_flags_sorted,
-                    self.arena["group_keys"], self.arena["group_off"],
-                    self.arena["windows"], pkg_win_range=self.pkg_win_range)
+                if dd is not None:
+                    pending = native.match_launch(
+                        dd["u_gk"], dd["u_hi"], dd["u_lo"], dd["u_flags"],
+                        self.arena["group_keys"], self.arena["group_off"],
+                        self.arena["windows"], pkg_win_range=dd["u_ranges"])
+                else:
+                    pending = native.match_launch(
+                        self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
+                        self.pkg_key_lo_sorted, self.pkg_flags_sorted,
+                        self.arena["group_keys"], self.arena["group_off"],
+                        self.arena["windows"], pkg_win_range=self.pkg_win_range)
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
             torch.cuda.current_stream().wait_stream(side)
             sp, sw = native.match_finalize(pending)
+            if dd is not None:
+                # dedup rows -> local package rows (perm2 indexes the LOCAL
+                # owned arrays) -> global package index
+                lrow, lwin = expand_dedup_matches(torch, dd, sp, sw)
+                orig = self.own_pkg_idx[lrow]
+                packed = (orig << 32) | lwin
+                packed, _ = torch.sort(packed)
+                pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
+            else:
+                orig = self.own_pkg_idx[self.pkg_perm[sp]]
+                packed = (orig << 32) | sw
+                packed, _ = torch.sort(packed)
+                pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
         else:
             sp, sw = self._match_local()
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
-
-        # local sorted row -> GLOBAL package index, deterministic pair order
-        orig = self.own_pkg_idx[self.pkg_perm[sp]]
-        packed = (orig << 32) | sw
-        packed, _ = torch.sort(packed)
-        pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
+            orig = self.own_pkg_idx[self.pkg_perm[sp]]
+            packed = (orig << 32) | sw
+            packed, _ = torch.sort(packed)
+            pkg_idx, win_idx = (packed >> 32), (packed & 0xFFFFFFFF)
         n_findings = pkg_idx.numel()
 
         pkg_nodes = pkg_idx + self.estate.pkg_base

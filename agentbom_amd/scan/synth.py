@@ -155,6 +155,37 @@ def _osv_shaped_arena(target_windows: int, name_catalog: int, seed: int) -> Advi
     )
 
 
+def _splitmix64(x: np.ndarray) -> np.ndarray:
+    """Vectorized splitmix64 over uint64 (wrapping arithmetic)."""
+    with np.errstate(over="ignore"):
+        x = (x + np.uint64(0x9E3779B97F4A7C15))
+        x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        return x ^ (x >> np.uint64(31))
+
+
+def _pool_versions(rng, name_id: np.ndarray, name_catalog: int, seed: int):
+    """Per-name finite release pools -> (key_hi, key_lo) per package.
+
+    Release counts are zipf-distributed (8..2048); a package's version is a
+    uniform draw from its name's pool, derived by hashing (name, rel_idx)
+    into the same (major<20, minor<30, patch<60) space the arena windows
+    cover."""
+    n_names_used = name_catalog
+    with np.errstate(over="ignore"):
+        nrel = 8 + np.minimum(
+            (np.random.default_rng([seed, 0x9E1E]).pareto(1.2, n_names_used) * 16)
+            .astype(np.int64), 2040)
+    rel_idx = (rng.random(len(name_id)) * nrel[name_id.astype(np.int64)]).astype(np.uint64)
+    h = _splitmix64(name_id * np.uint64(0x9E3779B97F4A7C15) ^ _splitmix64(rel_idx))
+    n1 = (h % np.uint64(20)).astype(np.int64)
+    n2 = ((h >> np.uint64(8)) % np.uint64(30)).astype(np.int64)
+    n3 = ((h >> np.uint64(16)) % np.uint64(60)).astype(np.int64)
+    zeros = np.zeros(len(name_id), dtype=np.int64)
+    rank = np.full(len(name_id), RELEASE_RANK, dtype=np.int64)
+    return pack_batch(zeros, n1, n2, n3, zeros, rank, zeros)
+
+
 def _triple_keys(rng, n, lo=(0, 0, 0), hi=(20, 30, 60)):
     n1 = rng.integers(lo[0], hi[0], n, dtype=np.int64)
     n2 = rng.integers(lo[1], hi[1], n, dtype=np.int64)
@@ -247,7 +278,17 @@ def generate_estate(
     # ── package identities + versions ─────────────────────────────────────
     # zipf-ish name popularity over the catalog
     name_id = (rng.pareto(1.2, n_packages) * name_catalog / 20).astype(np.uint64) % name_catalog
-    _, (key_hi, key_lo) = _triple_keys(rng, n_packages)
+    if arena_windows is not None:
+        # realism mode: each name has a FINITE release pool (real packages
+        # ship tens-to-hundreds of releases; an estate with 40k installs of
+        # one library covers few distinct versions).  Version = seeded hash
+        # of (name, release_idx), so duplicate (name, version) rows occur at
+        # real-estate rates and the engine's dedup-match layout has the
+        # same leverage the reference's dedup pass has
+        # (package_scan.py:1794-1816).
+        key_hi, key_lo = _pool_versions(rng, name_id, name_catalog, seed)
+    else:
+        _, (key_hi, key_lo) = _triple_keys(rng, n_packages)
     pkg_flags = np.full(n_packages, PF_ENCODABLE, dtype=np.uint8)
 
     # ── synthetic advisory arena ──────────────────────────────────────────

@@ -450,3 +450,19 @@ def test_match_precomputed_ranges_parity(estate, dev):
                         pkg_win_range=rng)
     assert torch.equal(base[0], fast[0])
     assert torch.equal(base[1], fast[1])
+
+
+def test_dedup_match_step_equals_plain(dev):
+    """Dedup-match layout must not change step results (bit-for-bit)."""
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+    from agentbom_amd.scan.synth import generate_estate
+
+    est = generate_estate(n_agents=400, n_servers=2000, n_packages=120_000,
+                          name_catalog=20_000, seed=99, arena_windows=40_000)
+    eng = EstateEngine(est, device=str(dev))
+    assert eng.match_dedup is not None, "realistic estate must trigger dedup"
+    with_dd = eng.step()
+    eng.match_dedup = None
+    without = eng.step()
+    for key in ("pkg_idx", "win_idx", "scores", "n_agents", "n_creds", "n_tools"):
+        assert torch.equal(with_dd[key], without[key]), key
