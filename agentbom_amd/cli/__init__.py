@@ -982,10 +982,19 @@ def db_group() -> None:
 
 
 @db_group.command(name="sync")
-@click.option("--source", type=click.Choice(["osv", "demo"]), default="demo")
+@click.option("--source", type=click.Choice(["osv", "epss", "kev", "nvd", "demo"]),
+              default="demo")
 @click.option("--path", type=click.Path(), default=None)
-def db_sync_cmd(source: str, path: Optional[str]) -> None:
-    """Populate the local advisory DB (offline image: demo source only)."""
+@click.option("--ecosystem", "ecosystems", multiple=True,
+              help="OSV bulk ecosystems (repeatable), e.g. --ecosystem npm")
+@click.option("--from-file", "from_file", type=click.Path(exists=True), default=None,
+              help="Offline file-drop ingest (OSV dir/zip, EPSS csv, KEV json)")
+def db_sync_cmd(source: str, path: Optional[str], ecosystems: tuple[str, ...],
+                from_file: Optional[str]) -> None:
+    """Sync the local advisory DB from live sources or file drops.
+
+    Live sources honor offline mode (AGENT_BOM_OFFLINE / --offline posture):
+    in offline mode only --from-file and demo work."""
     from agentbom_amd.db.store import AdvisoryStore, default_db_path
 
     store = AdvisoryStore(path or default_db_path())
@@ -994,9 +1003,35 @@ def db_sync_cmd(source: str, path: Optional[str]) -> None:
 
         n = store.ingest_windows(demo_advisory_windows())
         click.echo(f"ingested {n} demo advisory windows into {store.path}")
-    else:
-        click.echo("OSV sync requires network access (not available in this build).", err=True)
+        return
+    if from_file:
+        from agentbom_amd.db import osv_ingest
+
+        fn = {"osv": osv_ingest.sync_osv, "epss": osv_ingest.sync_epss,
+              "kev": osv_ingest.sync_kev}.get(source)
+        if fn is None:
+            click.echo(f"--from-file not supported for {source}", err=True)
+            sys.exit(1)
+        n = fn(store, from_file)
+        click.echo(f"ingested {n} {source} records from {from_file}")
+        return
+    from agentbom_amd.db import live
+    from agentbom_amd.utils.http_client import OfflineError
+
+    try:
+        if source == "osv":
+            ecos = list(ecosystems) or ["npm", "pypi"]
+            n = live.sync_osv_bulk(store, ecos)
+        elif source == "epss":
+            n = live.sync_epss_live(store)
+        elif source == "kev":
+            n = live.sync_kev_live(store)
+        else:
+            n = live.sync_nvd_live(store)
+    except OfflineError as exc:
+        click.echo(f"{exc} — use --from-file for air-gapped ingest", err=True)
         sys.exit(1)
+    click.echo(f"synced {n} {source} records into {store.path}")
 
 
 @db_group.command(name="enrich")

@@ -108,6 +108,7 @@ class ScanOptions:
     ignore_ids: frozenset = frozenset()
     vendor_advisories: bool = True  # AMD PSIRT / NVIDIA CSAF / Intel feeds
     threat_intel: bool = True       # local IOC store enrichment
+    live_osv: bool = False          # query api.osv.dev for the package batch
 
 
 def _match_packages(
@@ -207,7 +208,19 @@ def scan_agents(
     _mark("malicious_screen")
 
     # ── matching ───────────────────────────────────────────────────────────
-    arena = build_arena(list(advisory_windows), include_unfixed=options.include_unfixed)
+    advisory_windows = list(advisory_windows)
+    if options.live_osv and not options.offline:
+        # live OSV batch supplement (reference package_scan.py:573): merges
+        # into the same arena; failures warn and fall back to local windows
+        # (fail-open with a recorded coverage gap, never silently clean)
+        try:
+            from agentbom_amd.db.live import osv_windows_for_packages
+
+            advisory_windows.extend(osv_windows_for_packages(unique))
+        except Exception as exc:  # OfflineError, transport failure, ...
+            warnings.append(f"live OSV query unavailable: {exc}; "
+                            "matched against the local advisory DB only")
+    arena = build_arena(advisory_windows, include_unfixed=options.include_unfixed)
     use_gpu = options.use_gpu
     if use_gpu is None:
         try:

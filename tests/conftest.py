@@ -16,6 +16,7 @@ import pytest
 def pytest_configure(config: pytest.Config) -> None:
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
     config.addinivalue_line("markers", "slow: long-running scale tests")
+    config.addinivalue_line("markers", "network: requires real network egress")
 
 
 def _has_gpu() -> bool:
@@ -31,12 +32,16 @@ HAS_GPU = _has_gpu()
 
 
 def pytest_collection_modifyitems(config: pytest.Config, items: list[pytest.Item]) -> None:
-    if HAS_GPU:
-        return
+    import os
+
     skip_gpu = pytest.mark.skip(reason="no GPU in this environment")
+    skip_net = pytest.mark.skip(reason="network tests disabled (AGENT_BOM_TEST_NETWORK=1)")
+    net_on = os.environ.get("AGENT_BOM_TEST_NETWORK") == "1"
     for item in items:
-        if "gpu" in item.keywords:
+        if not HAS_GPU and "gpu" in item.keywords:
             item.add_marker(skip_gpu)
+        if not net_on and "network" in item.keywords:
+            item.add_marker(skip_net)
 
 
 @pytest.fixture(autouse=True)
