@@ -109,6 +109,8 @@ class ScanOptions:
     vendor_advisories: bool = True  # AMD PSIRT / NVIDIA CSAF / Intel feeds
     threat_intel: bool = True       # local IOC store enrichment
     live_osv: bool = False          # query api.osv.dev for the package batch
+    transitive: bool = False        # expand the dependency tree via registries
+    max_depth: int = 3              # transitive expansion depth bound
 
 
 def _match_packages(
@@ -185,6 +187,36 @@ def scan_agents(
         now = _time.perf_counter()
         perf[f"{stage}_ms"] = round((now - _t0) * 1000, 3)
         _t0 = now
+
+    # ── transitive expansion + registry version resolution ─────────────────
+    if options.transitive and not options.offline:
+        try:
+            from agentbom_amd.scan.transitive import (
+                _MetaCache,
+                expand_transitive,
+                resolve_package_versions,
+            )
+            from agentbom_amd.utils.http_client import create_client
+
+            client = create_client()
+            cache = _MetaCache()
+            n_resolved = n_added = 0
+            for agent in agents:
+                for server in agent.mcp_servers:
+                    n_resolved += resolve_package_versions(server.packages, client, cache)
+                    children = expand_transitive(server.packages,
+                                                 max_depth=options.max_depth,
+                                                 client=client, cache=cache)
+                    server.packages.extend(children)
+                    n_added += len(children)
+            if n_resolved or n_added:
+                warnings.append(
+                    f"transitive expansion: +{n_added} packages, "
+                    f"{n_resolved} versions resolved from registries")
+        except Exception as exc:
+            warnings.append(f"transitive expansion unavailable: {exc}; "
+                            "scanned declared dependencies only")
+    _mark("transitive")
 
     # ── collect + dedup packages across the estate ─────────────────────────
     pkg_refs: dict[tuple[str, str, str], list[Package]] = {}
