@@ -305,8 +305,13 @@ def discover_all(
     agent_types: Optional[list[AgentType]] = None,
     project_root: Optional[str] = None,
     include_installed: bool = True,
+    include_runtime: bool = True,
 ) -> list[Agent]:
-    """Union of global + project config surfaces, deduped by canonical id."""
+    """Union of global + project config surfaces, deduped by canonical id.
+
+    ``include_runtime`` adds the live surfaces: Docker MCP Toolkit,
+    compose services, running processes, Kubernetes (each best-effort —
+    absent substrate yields nothing, never an error)."""
     agents = discover_global_configs(agent_types)
     agents += discover_project_configs(project_root)
     if include_installed:
@@ -315,6 +320,24 @@ def discover_all(
             a for a in discover_installed_not_configured(agent_types)
             if a.agent_type not in configured
         ]
+    if include_runtime:
+        from agentbom_amd.scan.discovery_runtime import (
+            discover_compose_mcp_servers,
+            discover_docker_mcp,
+            discover_k8s_mcp_servers,
+            discover_running_processes,
+        )
+
+        for collect in (discover_docker_mcp,
+                        lambda: discover_compose_mcp_servers(project_root),
+                        discover_running_processes,
+                        discover_k8s_mcp_servers):
+            try:
+                found = collect()
+            except Exception:
+                found = None  # runtime surfaces are strictly best-effort
+            if found is not None:
+                agents.append(found)
     seen: set[str] = set()
     unique = []
     for a in agents:
