@@ -50,16 +50,17 @@ def verify_oidc_bearer(token: str,
                        issuer: Optional[str] = None,
                        audience: Optional[str] = None,
                        now: Optional[float] = None) -> dict[str, Any]:
-    """Validate an HS256 JWT; returns its claims or raises AuthError.
+    """Validate an OIDC bearer JWT; returns its claims or raises AuthError.
 
-    Checks: structure, alg allow-list (HS256 only — 'none' and RS* are
-    rejected), signature, exp/nbf with 60 s leeway, and iss/aud when the
-    deployment pins them (env or args)."""
-    secret = secret if secret is not None else os.environ.get("AGENT_BOM_OIDC_SECRET")
+    Dispatch by header alg (strict allow-list, 'none' always rejected):
+    - RS256 -> api/oidc.py JWKS verification (pure-stdlib RSA) when
+      AGENT_BOM_OIDC_JWKS / AGENT_BOM_OIDC_JWKS_URL is configured;
+    - HS256 -> operator-shared-secret HMAC (AGENT_BOM_OIDC_SECRET), the
+      air-gapped deployment shape.
+    Both paths require a numeric exp (60 s leeway), check nbf, and pin
+    iss/aud when configured."""
     issuer = issuer if issuer is not None else os.environ.get("AGENT_BOM_OIDC_ISSUER")
     audience = audience if audience is not None else os.environ.get("AGENT_BOM_OIDC_AUDIENCE")
-    if not secret:
-        raise AuthError("OIDC not configured (AGENT_BOM_OIDC_SECRET unset)")
     parts = token.split(".")
     if len(parts) != 3:
         raise AuthError("malformed JWT")
@@ -70,8 +71,15 @@ def verify_oidc_bearer(token: str,
     except Exception as exc:
         raise AuthError(f"undecodable JWT: {exc}") from None
     alg = header.get("alg")
+    if alg == "RS256":
+        from agentbom_amd.api.oidc import verify_rs256_bearer
+
+        return verify_rs256_bearer(token, issuer=issuer, audience=audience, now=now)
+    secret = secret if secret is not None else os.environ.get("AGENT_BOM_OIDC_SECRET")
+    if not secret:
+        raise AuthError("OIDC not configured (AGENT_BOM_OIDC_SECRET unset)")
     if alg != "HS256":
-        raise AuthError(f"alg {alg!r} not accepted (HS256 only in this build)")
+        raise AuthError(f"alg {alg!r} not accepted (RS256/HS256 only)")
     expected = hmac.new(secret.encode(), f"{parts[0]}.{parts[1]}".encode(),
                         hashlib.sha256).digest()
     if not hmac.compare_digest(sig, expected):

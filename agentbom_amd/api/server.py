@@ -45,6 +45,9 @@ class _State:
         self.reports: dict[str, Any] = {}
         self.graphs: dict[str, Any] = {}
         self.latest_scan: Optional[str] = None
+        # tenant scoping: scan ownership + per-tenant latest pointers
+        self.scan_tenant: dict[str, str] = {}
+        self.latest_by_tenant: dict[str, str] = {}
         self.lock = threading.Lock()
         self.metrics: dict[str, float] = {
             "scans_total": 0, "scan_failures_total": 0, "graph_queries_total": 0,
@@ -113,7 +116,9 @@ def create_app() -> FastAPI:
     state.delegation = DelegationTokens()
     state.quotas = QuotaTracker()
     state.scim_users = ScimUserStore()
-    oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET"))
+    oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET")
+                        or os.environ.get("AGENT_BOM_OIDC_JWKS")
+                        or os.environ.get("AGENT_BOM_OIDC_JWKS_URL"))
 
     def _resolve_role(request: Request, x_api_key: Optional[str]) -> Optional[str]:
         """API key (static table or SCIM-bound) > OIDC bearer > delegation
@@ -130,6 +135,8 @@ def create_app() -> FastAPI:
             except AuthError:
                 return None
             request.state.principal = f"oidc:{claims.get('sub', '?')}"
+            request.state.tenant_id = str(
+                claims.get("tenant_id") or claims.get("tid") or "default")
             return role_from_claims(claims)
         dtok = request.headers.get("X-Delegation-Token")
         if dtok and state.delegation.enabled:
@@ -238,13 +245,16 @@ def create_app() -> FastAPI:
                 return
             job["steps"].append({"step": "graph_persist", "at": _now()})
             snapshot_id = None
+            tenant = state.scan_tenant.get(job_id, "default")
             if state.graph_store is not None:
-                snapshot_id = state.graph_store.save_snapshot(graph, scan_id=job_id)
+                snapshot_id = state.graph_store.save_snapshot(
+                    graph, scan_id=job_id, tenant_id=tenant)
             job["snapshot_id"] = snapshot_id
             with state.lock:
                 state.reports[job_id] = report
                 state.graphs[job_id] = graph
                 state.latest_scan = job_id
+                state.latest_by_tenant[tenant] = job_id
             job["result"] = {"summary": to_json(report)["summary"]}
             job["status"] = "done"
             job["steps"].append({"step": "done", "at": _now()})
@@ -273,13 +283,15 @@ def create_app() -> FastAPI:
 
             raise HTTPException(status_code=429, detail="scan concurrency limit reached",
                                 headers={"Retry-After": str(retry_after)})
+        tenant = getattr(request.state, "tenant_id", "default") or "default"
         try:
-            return _submit_scan_inner(req)
+            return _submit_scan_inner(req, tenant)
         finally:
             state.backpressure.release()
 
-    def _submit_scan_inner(req: ScanRequest) -> dict:
+    def _submit_scan_inner(req: ScanRequest, tenant: str = "default") -> dict:
         job_id = str(uuid.uuid4())
+        state.scan_tenant[job_id] = tenant
         state.jobs[job_id] = {
             "id": job_id, "status": "pending", "submitted_at": _now(),
             "steps": [], "result": None, "error": None,
@@ -345,9 +357,9 @@ def create_app() -> FastAPI:
         return json.loads(json.dumps(to_json(report), default=str))
 
     @app.get("/v1/findings", dependencies=[Depends(auth)])
-    def findings(severity: Optional[str] = None, limit: int = 100,
+    def findings(request: Request, severity: Optional[str] = None, limit: int = 100,
                  view: Optional[str] = None) -> dict:
-        report = _latest_report()
+        report = _latest_report(request)
         if view:
             from agentbom_amd.output import finding_views as fv
 
@@ -367,20 +379,33 @@ def create_app() -> FastAPI:
 
     # ── graph reads ────────────────────────────────────────────────────────
 
-    def _latest_report():
-        if state.latest_scan is None:
-            raise HTTPException(status_code=404, detail="no scan yet — POST /v1/scan first")
-        return state.reports[state.latest_scan]
+    def _tenant_of(request: Optional[Request]) -> str:
+        if request is None:
+            return "default"
+        return getattr(request.state, "tenant_id", "default") or "default"
 
-    def _latest_graph():
-        if state.latest_scan is None:
-            raise HTTPException(status_code=404, detail="no scan yet — POST /v1/scan first")
+    def _latest_scan_for(request: Optional[Request]) -> str:
+        """Tenant-scoped latest scan id (cross-tenant reads 404)."""
+        tenant = _tenant_of(request)
+        sid = state.latest_by_tenant.get(tenant)
+        if sid is None and tenant == "default":
+            sid = state.latest_scan  # untenanted/legacy scans
+        if sid is None:
+            raise HTTPException(status_code=404,
+                                detail="no scan yet for this tenant — POST /v1/scan first")
+        return sid
+
+    def _latest_report(request: Optional[Request] = None):
+        return state.reports[_latest_scan_for(request)]
+
+    def _latest_graph(request: Optional[Request] = None):
+        sid = _latest_scan_for(request)
         state.metrics["graph_queries_total"] += 1
-        return state.graphs[state.latest_scan]
+        return state.graphs[sid]
 
     @app.get("/v1/graph", dependencies=[Depends(auth)])
-    def graph_summary(limit: int = 100) -> dict:
-        g = _latest_graph()
+    def graph_summary(request: Request, limit: int = 100) -> dict:
+        g = _latest_graph(request)
         node_ids = sorted(g.nodes)[:limit]
         return {
             "node_count": g.node_count,
@@ -390,17 +415,17 @@ def create_app() -> FastAPI:
         }
 
     @app.get("/v1/graph/search", dependencies=[Depends(auth)])
-    def graph_search(q: str = "", entity_type: Optional[str] = None, limit: int = 100) -> dict:
+    def graph_search(request: Request, q: str = "", entity_type: Optional[str] = None, limit: int = 100) -> dict:
         from agentbom_amd.graph.types import EntityType
 
-        g = _latest_graph()
+        g = _latest_graph(request)
         et = [EntityType(entity_type)] if entity_type else None
         nodes = g.search(query=q, entity_types=et, limit=limit)
         return {"total": len(nodes), "nodes": [n.to_dict() for n in nodes]}
 
     @app.get("/v1/graph/node/{node_id:path}/neighbors", dependencies=[Depends(auth)])
-    def graph_neighbors(node_id: str, direction: str = "both") -> dict:
-        g = _latest_graph()
+    def graph_neighbors(request: Request, node_id: str, direction: str = "both") -> dict:
+        g = _latest_graph(request)
         if node_id not in g.nodes:
             raise HTTPException(status_code=404, detail="node not found")
         return {
@@ -409,10 +434,10 @@ def create_app() -> FastAPI:
         }
 
     @app.get("/v1/graph/paths", dependencies=[Depends(auth)])
-    def graph_paths(source: Optional[str] = None, target: Optional[str] = None,
+    def graph_paths(request: Request, source: Optional[str] = None, target: Optional[str] = None,
                     limit: int = 25) -> dict:
         """Attack-path / blast-radius drilldown (the BASELINE p50 metric)."""
-        g = _latest_graph()
+        g = _latest_graph(request)
         if source and target:
             path = g.shortest_path(source, target)
             return {"paths": [path] if path else []}
@@ -422,15 +447,15 @@ def create_app() -> FastAPI:
         return {"path_count": len(paths), "paths": [p.to_dict() for p in paths]}
 
     @app.get("/v1/graph/attack-paths", dependencies=[Depends(auth)])
-    def graph_attack_paths(limit: int = 25) -> dict:
-        return graph_paths(limit=limit)
+    def graph_attack_paths(request: Request, limit: int = 25) -> dict:
+        return graph_paths(request, limit=limit)
 
     @app.get("/v1/graph/exposure-paths", dependencies=[Depends(auth)])
-    def graph_exposure_paths(limit: int = 50) -> dict:
+    def graph_exposure_paths(request: Request, limit: int = 50) -> dict:
         from agentbom_amd.models import blast_radius_to_finding
         from agentbom_amd.output.exposure_path import exposure_path_for_finding
 
-        report = _latest_report()
+        report = _latest_report(request)
         paths = [
             exposure_path_for_finding(blast_radius_to_finding(br), rank=i + 1)
             for i, br in enumerate(report.blast_radii[:limit])
@@ -439,36 +464,36 @@ def create_app() -> FastAPI:
                 "path_count": len(paths), "paths": paths}
 
     @app.post("/v1/graph/query", dependencies=[Depends(auth)])
-    def graph_query(req: GraphQueryRequest) -> dict:
-        g = _latest_graph()
+    def graph_query(request: Request, req: GraphQueryRequest) -> dict:
+        g = _latest_graph(request)
         if req.start not in g.nodes:
             raise HTTPException(status_code=404, detail="start node not found")
         return g.traverse_subgraph(req.start, max_depth=req.max_depth, max_nodes=req.max_nodes)
 
     @app.get("/v1/graph/impact/{node_id:path}", dependencies=[Depends(auth)])
-    def graph_impact(node_id: str, max_hops: int = 4) -> dict:
-        g = _latest_graph()
+    def graph_impact(request: Request, node_id: str, max_hops: int = 4) -> dict:
+        g = _latest_graph(request)
         if node_id not in g.nodes:
             raise HTTPException(status_code=404, detail="node not found")
         return g.impact_of(node_id, max_hops=min(max_hops, 4))
 
     @app.get("/v1/graph/rollup", dependencies=[Depends(auth)])
-    def graph_rollup() -> dict:
+    def graph_rollup(request: Request) -> dict:
         from agentbom_amd.graph.rollup import rollup_view
 
-        return rollup_view(_latest_graph())
+        return rollup_view(_latest_graph(request))
 
     @app.get("/v1/graph/rollup/{container_id:path}", dependencies=[Depends(auth)])
-    def graph_drilldown(container_id: str) -> dict:
+    def graph_drilldown(request: Request, container_id: str) -> dict:
         from agentbom_amd.graph.rollup import drill_down
 
-        return drill_down(_latest_graph(), container_id)
+        return drill_down(_latest_graph(request), container_id)
 
     @app.get("/v1/graph/should-i-deploy", dependencies=[Depends(auth)])
-    def should_i_deploy() -> dict:
+    def should_i_deploy(request: Request) -> dict:
         from agentbom_amd.utils import config as cfg
 
-        report = _latest_report()
+        report = _latest_report(request)
         max_risk = max((br.risk_score for br in report.blast_radii), default=0.0) * 10
         has_malicious = any(br.package.is_malicious for br in report.blast_radii)
         has_kev = any(br.vulnerability.is_kev for br in report.blast_radii)
@@ -505,17 +530,17 @@ def create_app() -> FastAPI:
         return state.graph_store.diff_snapshots(old, new)
 
     @app.get("/v1/findings/toxic-combinations", dependencies=[Depends(auth)])
-    def toxic_combinations() -> dict:
-        report = _latest_report()
+    def toxic_combinations(request: Request) -> dict:
+        report = _latest_report(request)
         return {"combinations": report.toxic_combination_findings_data or []}
 
     @app.get("/v1/graph/evidence-manifest", dependencies=[Depends(auth)])
-    def graph_evidence_manifest() -> dict:
+    def graph_evidence_manifest(request: Request) -> dict:
         import hashlib
 
         if state.graph_store is not None:
             return state.graph_store.evidence_manifest(scan_id=state.latest_scan)
-        g = _latest_graph()
+        g = _latest_graph(request)
         digest = hashlib.sha256(
             json.dumps(g.to_dict(), sort_keys=True, default=str).encode()
         ).hexdigest()
@@ -582,10 +607,10 @@ def create_app() -> FastAPI:
                 "entries": state.audit_entries[-limit:]}
 
     @app.get("/v1/compliance/{framework}/report", dependencies=[Depends(auth)])
-    def compliance_report(framework: str) -> dict:
+    def compliance_report(request: Request, framework: str) -> dict:
         from agentbom_amd.models import FRAMEWORK_TAG_FIELDS
 
-        report = _latest_report()
+        report = _latest_report(request)
         field_name = next((f for f, slug in FRAMEWORK_TAG_FIELDS if slug == framework), None)
         if field_name is None:
             raise HTTPException(status_code=404, detail=f"unknown framework {framework!r}")
@@ -604,28 +629,28 @@ def create_app() -> FastAPI:
                 "controls": dict(sorted(counts.items())), "findings": rows}
 
     @app.get("/v1/graph/attack-flow", dependencies=[Depends(auth)])
-    def attack_flow(cve: Optional[str] = None, min_severity: Optional[str] = None,
+    def attack_flow(request: Request, cve: Optional[str] = None, min_severity: Optional[str] = None,
                     agent: Optional[str] = None) -> dict:
         from agentbom_amd.output.flow_fmt import build_attack_flow
 
         state.metrics["graph_queries_total"] += 1
-        return build_attack_flow(_latest_report(), cve=cve,
+        return build_attack_flow(_latest_report(request), cve=cve,
                                  min_severity=min_severity, agent=agent)
 
     @app.get("/v1/mesh", dependencies=[Depends(auth)])
-    def agent_mesh() -> dict:
+    def agent_mesh(request: Request) -> dict:
         from agentbom_amd.output.flow_fmt import build_agent_mesh
 
-        return build_agent_mesh(_latest_report())
+        return build_agent_mesh(_latest_report(request))
 
     @app.get("/v1/findings/delta", dependencies=[Depends(auth)])
-    def findings_delta(format: str = "ndjson") -> dict:
+    def findings_delta(request: Request, format: str = "ndjson") -> dict:
         """Delta events (new/resolved/changed) since the previous call."""
         from agentbom_amd.output.delta_stream import DeltaStreamer
 
         if getattr(state, "delta_streamer", None) is None:
             state.delta_streamer = DeltaStreamer(fmt="ndjson")
-        events = state.delta_streamer.emit(_latest_report())
+        events = state.delta_streamer.emit(_latest_report(request))
         if format == "ocsf":
             from agentbom_amd.output.delta_stream import _to_ocsf_event
 
@@ -634,13 +659,13 @@ def create_app() -> FastAPI:
         return {"watermark": state.delta_streamer.watermark, "events": events}
 
     @app.get("/v1/remediation", dependencies=[Depends(auth)])
-    def remediation(script: bool = False) -> Any:
+    def remediation(request: Request, script: bool = False) -> Any:
         from agentbom_amd.scan.remediation import (
             remediation_commands,
             remediation_script,
         )
 
-        report = _latest_report()
+        report = _latest_report(request)
         if script:
             from fastapi.responses import PlainTextResponse
 
@@ -649,11 +674,11 @@ def create_app() -> FastAPI:
         return {"commands": remediation_commands(report)}
 
     @app.get("/v1/posture", dependencies=[Depends(auth)])
-    def posture() -> dict:
+    def posture(request: Request) -> dict:
         from agentbom_amd.scan.auth_posture import assess_a2a, assess_estate
         from agentbom_amd.scan.self_posture import evaluate_self_posture
 
-        report = _latest_report()
+        report = _latest_report(request)
         return {
             "mcp_auth_posture": assess_estate(report.agents),
             "a2a_auth_posture": assess_a2a(report.agents,
