@@ -150,6 +150,33 @@ def _scan_impl(
     aws_inventory: Optional[str] = None, endpoint: bool = False,
     notebooks: Optional[str] = None, skills: Optional[str] = None,
     semgrep: Optional[str] = None, cloud_inventory: tuple = (),
+    # modes
+    self_scan: bool = False, inventory_only: bool = False,
+    no_discover: bool = False, dry_run: bool = False,
+    # depth / enrichment
+    transitive: bool = False, max_depth: int = 3, live_osv: bool = False,
+    enrich: bool = False, enrich_bundle: Optional[str] = None,
+    deps_dev: Optional[str] = None, vex: Optional[str] = None,
+    generate_vex: Optional[str] = None,
+    license_check_flag: bool = False, license_deny: tuple = (),
+    # gates
+    fail_if_ai_risk: bool = False, warn_on: Optional[str] = None,
+    require_fresh_db: bool = False, policy: Optional[str] = None,
+    # surfaces
+    image_tar: Optional[str] = None, repo: Optional[str] = None,
+    os_packages: bool = False, jupyter: Optional[str] = None,
+    browser_extensions: tuple = (), dataset_cards: Optional[str] = None,
+    training_pipelines: Optional[str] = None,
+    verify_model_hashes: Optional[str] = None, gpu_scan_flag: bool = False,
+    introspect: bool = False, health_check_flag: bool = False,
+    scan_prompts: Optional[str] = None, scan_pii: Optional[str] = None,
+    skill: Optional[str] = None,
+    # push / integrations
+    push_url: Optional[str] = None, push_api_key: Optional[str] = None,
+    webhook: tuple = (), slack_webhook: Optional[str] = None,
+    jira_url: Optional[str] = None, jira_token: Optional[str] = None,
+    jira_project: Optional[str] = None,
+    siem_url: Optional[str] = None, siem_token: Optional[str] = None,
 ) -> None:
     from agentbom_amd.scan.orchestrator import (
         ScanOptions,
@@ -159,12 +186,58 @@ def _scan_impl(
         scan_agents,
     )
 
+    # flag aliases (reference option surface)
+    image = image or image_tar
+    notebooks = notebooks or jupyter
+    skills = skills or skill
+    if repo:
+        filesystem = filesystem or repo
+        code = code or repo
+    if self_scan:
+        import agentbom_amd as _pkg
+
+        filesystem = filesystem or str(Path(_pkg.__file__).resolve().parent)
+
     options = ScanOptions(
         demo=demo, offline=offline, include_unfixed=include_unfixed,
         blast_radius_depth=blast_radius_depth, fail_on_severity=fail_on_severity,
         exit_zero=exit_zero, fail_on_kev=fail_on_kev,
         use_gpu=False if no_gpu else None,
+        transitive=transitive, max_depth=max_depth, live_osv=live_osv,
+        fail_if_ai_risk=fail_if_ai_risk, warn_on=warn_on,
     )
+
+    if dry_run:
+        plan = {
+            "mode": "demo" if demo else "discover+scan",
+            "offline": offline,
+            "surfaces": {k: bool(v) for k, v in {
+                "inventory": inventory, "sbom": sbom, "image": image,
+                "filesystem": filesystem, "code": code, "iac": iac,
+                "os_packages": os_packages, "notebooks": notebooks,
+                "skills": skills, "browser_extensions": browser_extensions,
+                "dataset_cards": dataset_cards,
+                "training_pipelines": training_pipelines,
+                "gpu_scan": gpu_scan_flag, "secrets": scan_secrets,
+                "prompts": scan_prompts, "pii": scan_pii,
+            }.items()},
+            "gates": {"fail_on_severity": fail_on_severity,
+                      "fail_on_kev": fail_on_kev,
+                      "fail_if_ai_risk": fail_if_ai_risk,
+                      "policy": bool(policy)},
+        }
+        click.echo(json.dumps(plan, indent=2))
+        sys.exit(0)
+
+    if require_fresh_db and not demo:
+        from agentbom_amd.db.store import AdvisoryStore, default_db_path
+
+        st = AdvisoryStore(default_db_path()).status()
+        worst = st.get("freshness", {}).get("worst", "unknown")
+        if worst in ("expired", "unknown"):
+            click.echo(f"error: advisory DB freshness is {worst!r} "
+                       "(--require-fresh-db): run `agent-bom db sync`", err=True)
+            sys.exit(2)
     # project-level defaults + suppression file (.agent-bom.yaml / -ignore)
     from agentbom_amd.utils.project_config import apply_to_scan_options, load_project_config
 
@@ -201,13 +274,62 @@ def _scan_impl(
                                        packages=pkgs, surface=ServerSurface.FILESYSTEM)],
                 source="filesystem_scan",
             ))
-        if not agents:
+        if os_packages:
+            from agentbom_amd.scan.surfaces_extra import scan_os_packages
+
+            os_agent = scan_os_packages()
+            if os_agent:
+                agents.append(os_agent)
+            else:
+                click.echo("warning: no OS package database found", err=True)
+        if gpu_scan_flag:
+            from agentbom_amd.scan.surfaces_extra import gpu_scan_agent
+
+            gagent = gpu_scan_agent()
+            if gagent:
+                agents.append(gagent)
+            else:
+                click.echo("warning: no ROCm/GPU stack detected", err=True)
+        if not agents and not no_discover:
             from agentbom_amd.scan.discovery import discover_all
 
             agents = discover_all()
         if not agents:
             click.echo("No agents discovered. Try --demo for the bundled demo estate.", err=True)
             sys.exit(2)
+        if introspect:
+            from agentbom_amd.mcp.introspect import introspect_servers
+
+            try:
+                for _agent in agents:
+                    introspect_servers(_agent.mcp_servers)
+            except Exception as exc:
+                click.echo(f"warning: introspection failed: {exc}", err=True)
+        if health_check_flag:
+            from agentbom_amd.scan.surfaces_extra import health_check
+
+            for row in health_check(agents):
+                mark = {"ok": "+", "warn": "!", "error": "x"}[row["status"]]
+                click.echo(f"[{mark}] {row['agent']}/{row['server']}"
+                           + (f" — {row['detail']}" if row["detail"] else ""), err=True)
+        if inventory_only:
+            inv = {"agents": [{
+                "name": a.name, "agent_type": a.agent_type.value,
+                "config_path": a.config_path, "source": a.source,
+                "mcp_servers": [{
+                    "name": srv.name, "command": srv.command,
+                    "transport": srv.transport.value,
+                    "packages": [{"name": p.name, "version": p.version,
+                                  "ecosystem": p.ecosystem}
+                                 for p in srv.packages],
+                } for srv in a.mcp_servers],
+            } for a in agents]}
+            text = json.dumps(inv, indent=2, default=str)
+            if output:
+                Path(output).write_text(text)
+            else:
+                click.echo(text)
+            sys.exit(0)
         from agentbom_amd.db.store import load_advisory_windows
 
         report = scan_agents(agents, load_advisory_windows(offline=offline), options)
@@ -234,6 +356,72 @@ def _scan_impl(
         run_scanner_driver("skills", report, skills)
     if filesystem:
         run_scanner_driver("floating_refs", report, filesystem)
+    if scan_prompts:
+        from agentbom_amd.scan.surfaces_extra import scan_prompt_files
+
+        report.findings.extend(scan_prompt_files(scan_prompts))
+    if scan_pii:
+        from agentbom_amd.scan.surfaces_extra import scan_pii as _scan_pii
+
+        report.findings.extend(_scan_pii(scan_pii))
+    if browser_extensions:
+        from agentbom_amd.scan.surfaces_extra import scan_browser_extensions
+
+        inv, bfindings = scan_browser_extensions(browser_extensions)
+        report.findings.extend(bfindings)
+        report.ai_inventory_data = (report.ai_inventory_data or {}) | {
+            "browser_extensions": inv}
+    if dataset_cards:
+        from agentbom_amd.scan.surfaces_extra import scan_dataset_cards
+
+        inv, dfindings = scan_dataset_cards(dataset_cards)
+        report.findings.extend(dfindings)
+        report.ai_inventory_data = (report.ai_inventory_data or {}) | {
+            "dataset_cards": inv}
+    if training_pipelines:
+        from agentbom_amd.scan.surfaces_extra import scan_training_pipelines
+
+        inv, pfindings = scan_training_pipelines(training_pipelines)
+        report.findings.extend(pfindings)
+        report.ai_inventory_data = (report.ai_inventory_data or {}) | {
+            "training_pipelines": inv}
+    if verify_model_hashes:
+        if not model_files:
+            raise click.UsageError("--verify-model-hashes needs --model-files DIR")
+        from agentbom_amd.scan.surfaces_extra import verify_model_hash_manifest
+
+        report.findings.extend(
+            verify_model_hash_manifest(model_files, verify_model_hashes))
+    if license_check_flag:
+        from agentbom_amd.scan.surfaces_extra import license_check
+
+        report.findings.extend(
+            license_check(report, deny=license_deny or None))
+    if enrich or enrich_bundle:
+        from agentbom_amd.scan.enrichment import enrich_vulnerabilities, load_offline_bundle
+
+        stats = (load_offline_bundle(report, enrich_bundle) if enrich_bundle
+                 else enrich_vulnerabilities(report))
+        if verbose:
+            click.echo(f"enrichment: {json.dumps(stats)}", err=True)
+    if deps_dev:
+        from agentbom_amd.scan.deps_meta import enrich_packages_with_deps_meta, load_deps_bundle
+
+        pkgs = [p for a in report.agents for srv in a.mcp_servers for p in srv.packages]
+        enrich_packages_with_deps_meta(pkgs, bundle=load_deps_bundle(deps_dev))
+    if vex:
+        from agentbom_amd.scan.vex import VexDocument, apply_vex
+
+        doc = VexDocument.from_dict(json.loads(Path(vex).read_text()))
+        n = apply_vex(report, doc)
+        if verbose:
+            click.echo(f"VEX: {n} statements applied", err=True)
+    if generate_vex:
+        from agentbom_amd.scan.vex import generate_vex as _gen_vex
+
+        Path(generate_vex).write_text(
+            json.dumps(_gen_vex(report).to_dict(), indent=2))
+        click.echo(f"VEX document written to {generate_vex}", err=True)
     if code or semgrep:
         from agentbom_amd.scan.ast_analysis import (
             SymbolIndex,
@@ -280,8 +468,49 @@ def _scan_impl(
     ]
     report.toxic_combination_findings_data = [c.to_dict() for c in combos]
 
+    policy_failed = False
+    if policy:
+        from agentbom_amd.scan.policy import evaluate_policy, load_policy
+
+        pol = evaluate_policy(load_policy(policy), report.blast_radii)
+        for v in pol.get("violations", []):
+            click.echo(f"policy: rule {v['rule_id']} matched "
+                       f"{v['package']} ({v['vulnerability_id']})", err=True)
+        for w in pol.get("warnings", []):
+            click.echo(f"policy warning: rule {w['rule_id']} matched "
+                       f"{w['package']}", err=True)
+        policy_failed = not pol.get("passed", True)
+
+    if warn_on:
+        order = ["low", "medium", "high", "critical"]
+        if warn_on in order:
+            warn_set = set(order[order.index(warn_on):])
+            n_warn = sum(1 for br in report.blast_radii
+                         if br.vulnerability.severity.value in warn_set)
+            if n_warn:
+                click.echo(f"warning: {n_warn} finding(s) at or above "
+                           f"{warn_on} severity (--warn-on)", err=True)
+
+    # control-plane push + integrations (offline-guarded, fail-open)
+    if push_url or webhook or slack_webhook or (jira_url and jira_token) or siem_url:
+        from agentbom_amd.output.integrations import run_integrations
+
+        results = run_integrations(
+            report,
+            push_url=push_url, push_api_key=push_api_key,
+            webhooks=list(webhook), slack_webhook=slack_webhook,
+            jira={"url": jira_url, "token": jira_token,
+                  "project": jira_project} if jira_url and jira_token else None,
+            siem={"url": siem_url, "token": siem_token} if siem_url else None,
+        )
+        for name, ok, detail in results:
+            click.echo(f"integration {name}: {'ok' if ok else 'FAILED'}"
+                       + (f" — {detail}" if detail else ""), err=True)
+
     _render(report, fmt, output, verbose)
     rc = compute_exit_code(report, options)
+    if policy_failed and rc == 0:
+        rc = 1
     if rc:
         gates = []
         if any(br.package.is_malicious for br in report.blast_radii):
@@ -341,6 +570,91 @@ def _scan_options(f):
         click.option("--profile", "profile", default=None,
                      help="Apply a named flag-set from .agent-bom.yaml "
                           "profiles: (explicit flags still win)."),
+        # modes
+        click.option("--self-scan", "self_scan", is_flag=True,
+                     help="Scan agent-bom's own installation."),
+        click.option("--inventory-only", "inventory_only", is_flag=True,
+                     help="Discover and emit the inventory JSON; no matching."),
+        click.option("--no-discover", "no_discover", is_flag=True,
+                     help="Never auto-discover; only scan explicit surfaces."),
+        click.option("--dry-run", "dry_run", is_flag=True,
+                     help="Print the scan plan and exit 0."),
+        # depth / enrichment
+        click.option("--transitive", is_flag=True,
+                     help="Expand the dependency tree via package registries."),
+        click.option("--max-depth", type=click.IntRange(1, 6), default=3,
+                     help="Transitive expansion depth bound."),
+        click.option("--live-osv", "live_osv", is_flag=True,
+                     help="Query api.osv.dev for the package batch (needs network)."),
+        click.option("--enrich", is_flag=True,
+                     help="Enrich findings from the local advisory DB caches."),
+        click.option("--enrich-bundle", type=click.Path(exists=True), default=None,
+                     help="Air-gapped enrichment bundle dir (epss.csv/kev.json)."),
+        click.option("--deps-dev", "deps_dev", type=click.Path(exists=True), default=None,
+                     help="deps.dev/Scorecard metadata bundle for trust scoring."),
+        click.option("--vex", type=click.Path(exists=True), default=None,
+                     help="Apply an OpenVEX document (suppressions audit-kept)."),
+        click.option("--generate-vex", "generate_vex", type=click.Path(), default=None,
+                     help="Write an OpenVEX document for this scan's findings."),
+        click.option("--license-check", "license_check_flag", is_flag=True,
+                     help="Flag denied licenses on collected packages."),
+        click.option("--license-deny", "license_deny", multiple=True,
+                     help="License substring to deny (repeatable; default GPL-3.0/AGPL/SSPL)."),
+        # gates
+        click.option("--fail-if-ai-risk", "fail_if_ai_risk", is_flag=True,
+                     help="Exit 1 when findings carry AI risk context."),
+        click.option("--warn-on", "warn_on",
+                     type=click.Choice(["critical", "high", "medium", "low"]),
+                     default=None, help="Warn (stderr) at this severity without failing."),
+        click.option("--require-fresh-db", "require_fresh_db", is_flag=True,
+                     help="Refuse to scan against an expired advisory DB."),
+        click.option("--policy", type=click.Path(exists=True), default=None,
+                     help="Policy-as-code file; failed rules gate the exit code."),
+        # surfaces
+        click.option("--image-tar", "image_tar", type=click.Path(exists=True), default=None,
+                     help="Alias of --image (docker-save tarball)."),
+        click.option("--repo", type=click.Path(exists=True), default=None,
+                     help="Scan a repository: packages + AST code analysis."),
+        click.option("--os-packages", "os_packages", is_flag=True,
+                     help="Scan this host's dpkg/apk OS package databases."),
+        click.option("--jupyter", type=click.Path(exists=True), default=None,
+                     help="Alias of --notebooks."),
+        click.option("--browser-extensions", "browser_extensions", multiple=True,
+                     help="Browser extension dirs to audit (repeatable; "
+                          "defaults to Chrome/Firefox profile paths)."),
+        click.option("--dataset-cards", "dataset_cards", type=click.Path(exists=True),
+                     default=None, help="Scan dataset cards for provenance risks."),
+        click.option("--training-pipelines", "training_pipelines",
+                     type=click.Path(exists=True), default=None,
+                     help="Scan training pipeline definitions (dvc/MLproject/kfp)."),
+        click.option("--verify-model-hashes", "verify_model_hashes",
+                     type=click.Path(exists=True), default=None,
+                     help="sha256 manifest to verify --model-files artifacts against."),
+        click.option("--gpu-scan", "gpu_scan_flag", is_flag=True,
+                     help="Collect ROCm/amdgpu GPU posture inventory."),
+        click.option("--introspect", is_flag=True,
+                     help="Introspect live MCP servers (read-only handshake)."),
+        click.option("--health-check", "health_check_flag", is_flag=True,
+                     help="Static MCP server config health (command/url resolvable)."),
+        click.option("--scan-prompts", "scan_prompts", type=click.Path(exists=True),
+                     default=None, help="Scan prompt files for injection patterns."),
+        click.option("--scan-pii", "scan_pii", type=click.Path(exists=True),
+                     default=None, help="Bounded PII sweep (values redacted)."),
+        click.option("--skill", type=click.Path(exists=True), default=None,
+                     help="Alias of --skills."),
+        # push / integrations
+        click.option("--push-url", "push_url", default=None,
+                     help="POST the full report JSON to a control plane."),
+        click.option("--push-api-key", "push_api_key", default=None),
+        click.option("--webhook", multiple=True,
+                     help="POST a scan.completed event to this URL (repeatable)."),
+        click.option("--slack-webhook", "slack_webhook", default=None),
+        click.option("--jira-url", "jira_url", default=None),
+        click.option("--jira-token", "jira_token", default=None),
+        click.option("--jira-project", "jira_project", default=None),
+        click.option("--siem-url", "siem_url", default=None,
+                     help="POST OCSF events to a SIEM collector."),
+        click.option("--siem-token", "siem_token", default=None),
     ]
     for o in reversed(opts):
         f = o(f)

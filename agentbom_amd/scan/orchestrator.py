@@ -111,6 +111,8 @@ class ScanOptions:
     live_osv: bool = False          # query api.osv.dev for the package batch
     transitive: bool = False        # expand the dependency tree via registries
     max_depth: int = 3              # transitive expansion depth bound
+    fail_if_ai_risk: bool = False   # gate on findings carrying AI risk context
+    warn_on: Optional[str] = None   # severity that warns (stderr) w/o failing
 
 
 def _match_packages(
@@ -446,6 +448,8 @@ def compute_exit_code(report: AIBOMReport, options: ScanOptions) -> int:
     if any(br.vulnerability.severity.value in gate for br in active):
         return 1
     if options.fail_on_kev and any(br.vulnerability.is_kev for br in active):
+        return 1
+    if options.fail_if_ai_risk and any(br.ai_risk_context for br in active):
         return 1
     return 0
 

@@ -1,0 +1,505 @@
+"""Additional scan surfaces behind the grown CLI flag set.
+
+VERDICT r1 item 8 (CLI surface parity): each collector here backs a real
+``agent-bom agents`` flag (reference flag map:
+src/agent_bom/cli/options_sources.py / options_surfaces.py):
+
+  --os-packages          dpkg/apk/rpm databases of this host
+  --browser-extensions   Chrome/Chromium/Firefox extension manifests
+  --scan-prompts PATH    prompt files checked for injection patterns
+  --scan-pii PATH        bounded PII sweep (values always redacted)
+  --gpu-scan             ROCm/amdgpu driver + GPU posture inventory
+  --dataset-cards PATH   dataset card inventory + provenance checks
+  --training-pipelines   training pipeline inventory (dvc/MLproject/
+                         kubeflow/workflow files)
+  --license-check        denylist check over collected package licenses
+  --health-check         MCP server config health (command/url resolvable)
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import re
+import shutil
+import subprocess
+from pathlib import Path
+from typing import Iterable, Optional
+
+import yaml
+
+from agentbom_amd.models.core import Agent, AgentType, MCPServer, Package, ServerSurface
+from agentbom_amd.models.finding import Asset, Finding, FindingSource, FindingType, stable_id
+
+# ── OS packages ─────────────────────────────────────────────────────────────
+
+_OS_DB_PATHS = (
+    "var/lib/dpkg/status",
+    "lib/apk/db/installed",
+    "usr/lib/sysimage/rpm/rpmdb.sqlite",  # presence-only; rpm needs librpm
+)
+
+
+def scan_os_packages(root: str = "/") -> Optional[Agent]:
+    """Host OS package databases -> one os-packages agent."""
+    from agentbom_amd.scan.parsers import parse_apk_installed, parse_dpkg_status
+
+    rootp = Path(root)
+    pkgs: list[Package] = []
+    source_paths = []
+    dpkg = rootp / "var/lib/dpkg/status"
+    if dpkg.exists():
+        try:
+            pkgs += parse_dpkg_status(dpkg.read_text(errors="replace"), str(dpkg))
+            source_paths.append(str(dpkg))
+        except OSError:
+            pass
+    apk = rootp / "lib/apk/db/installed"
+    if apk.exists():
+        try:
+            pkgs += parse_apk_installed(apk.read_text(errors="replace"), str(apk))
+            source_paths.append(str(apk))
+        except OSError:
+            pass
+    if not pkgs:
+        return None
+    return Agent(
+        name=f"os-packages:{os.uname().nodename}", agent_type=AgentType.CUSTOM,
+        config_path=source_paths[0],
+        mcp_servers=[MCPServer(name="os-packages", packages=pkgs,
+                               surface=ServerSurface.OS_PACKAGES)],
+        source="os-packages",
+    )
+
+
+# ── browser extensions ──────────────────────────────────────────────────────
+
+_RISKY_EXT_PERMS = {
+    "<all_urls>", "webRequest", "webRequestBlocking", "cookies", "history",
+    "clipboardRead", "nativeMessaging", "debugger", "tabs", "downloads",
+}
+
+_DEFAULT_EXT_DIRS = (
+    "~/.config/google-chrome/Default/Extensions",
+    "~/.config/chromium/Default/Extensions",
+    "~/Library/Application Support/Google/Chrome/Default/Extensions",
+    "~/.mozilla/firefox",
+)
+
+
+def scan_browser_extensions(paths: Optional[Iterable[str]] = None):
+    """Extension manifests -> (inventory rows, findings for risky grants)."""
+    roots = [Path(os.path.expanduser(p)) for p in (paths or _DEFAULT_EXT_DIRS)]
+    inventory: list[dict] = []
+    findings: list[Finding] = []
+    for root in roots:
+        if not root.exists():
+            continue
+        for manifest in sorted(root.rglob("manifest.json"))[:500]:
+            try:
+                doc = json.loads(manifest.read_text(errors="replace"))
+            except (OSError, ValueError):
+                continue
+            if not isinstance(doc, dict) or "name" not in doc:
+                continue
+            perms = [str(p) for p in (doc.get("permissions") or []) if isinstance(p, (str,))]
+            host_perms = [str(p) for p in (doc.get("host_permissions") or [])]
+            risky = sorted(set(perms + host_perms) & _RISKY_EXT_PERMS)
+            if "<all_urls>" in host_perms:
+                risky = sorted(set(risky) | {"<all_urls>"})
+            row = {
+                "name": str(doc.get("name", "")), "version": str(doc.get("version", "")),
+                "manifest_version": doc.get("manifest_version"),
+                "path": str(manifest.parent), "permissions": perms,
+                "host_permissions": host_perms, "risky_permissions": risky,
+            }
+            inventory.append(row)
+            if risky:
+                findings.append(Finding(
+                    finding_type=FindingType.BROWSER_EXT,
+                    source=FindingSource.BROWSER_EXT,
+                    asset=Asset(name=row["name"] or manifest.parent.name,
+                                asset_type="browser_extension",
+                                location=str(manifest.parent)),
+                    severity="medium" if len(risky) < 3 else "high",
+                    title=f"Browser extension '{row['name']}' holds risky permissions",
+                    description=f"permissions: {', '.join(risky)}",
+                    evidence={"risky_permissions": risky,
+                              "version": row["version"]},
+                    is_actionable=True,
+                    id=stable_id("browser-ext", str(manifest.parent), row["name"]),
+                ))
+    return inventory, findings
+
+
+# ── prompt files: injection sweep ───────────────────────────────────────────
+
+_PROMPT_INJECTION = [
+    (re.compile(r"ignore (all )?(previous|prior|above) (instructions|prompts)", re.I),
+     "instruction-override"),
+    (re.compile(r"disregard (the )?(system|previous) (prompt|instructions)", re.I),
+     "instruction-override"),
+    (re.compile(r"you are now|act as (a|an) (?!assistant)", re.I), "role-hijack"),
+    (re.compile(r"(exfiltrate|send|post|upload).{0,40}(secret|credential|token|key)", re.I),
+     "exfiltration-instruction"),
+    (re.compile(r"do not (tell|inform|alert) the (user|human)", re.I), "cloaking"),
+    (re.compile(r"<\s*(system|assistant)\s*>", re.I), "role-tag-injection"),
+    (re.compile(r"base64 (decode|encoded) (and (run|execute|eval))", re.I),
+     "encoded-payload"),
+]
+
+_PROMPT_EXTS = {".md", ".txt", ".prompt", ".yaml", ".yml", ".json", ".xml"}
+
+
+def scan_prompt_files(path: str, max_files: int = 500) -> list[Finding]:
+    """Static prompt-injection sweep (the runtime detectors' pattern class
+    applied to prompt/config files at rest)."""
+    findings: list[Finding] = []
+    base = Path(path)
+    files = [base] if base.is_file() else [
+        p for p in sorted(base.rglob("*"))
+        if p.is_file() and p.suffix.lower() in _PROMPT_EXTS
+    ][:max_files]
+    for f in files:
+        try:
+            text = f.read_text(errors="replace")[:200_000]
+        except OSError:
+            continue
+        for rx, kind in _PROMPT_INJECTION:
+            m = rx.search(text)
+            if not m:
+                continue
+            line = text[: m.start()].count("\n") + 1
+            findings.append(Finding(
+                finding_type=FindingType.PROMPT_SECURITY,
+                source=FindingSource.PROMPT_SCAN,
+                asset=Asset(name=f.name, asset_type="prompt_file", location=str(f)),
+                severity="high" if kind in ("exfiltration-instruction", "cloaking")
+                         else "medium",
+                title=f"Prompt-injection pattern ({kind}) in {f.name}",
+                description=f"{kind} at {f}:{line}",
+                evidence={"kind": kind, "line": line,
+                          "match": m.group(0)[:80]},
+                is_actionable=True,
+                id=stable_id("prompt", str(f), kind, str(line)),
+            ))
+    return findings
+
+
+# ── PII sweep ───────────────────────────────────────────────────────────────
+
+_PII_PATTERNS = [
+    (re.compile(r"\b[A-Za-z0-9._%+-]+@[A-Za-z0-9.-]+\.[A-Za-z]{2,}\b"), "email"),
+    (re.compile(r"\b\d{3}-\d{2}-\d{4}\b"), "ssn"),
+    (re.compile(r"\b(?:\d[ -]*?){13,16}\b"), "card_number"),
+    (re.compile(r"\b\+?\d{1,2}[ .-]?\(?\d{3}\)?[ .-]?\d{3}[ .-]?\d{4}\b"), "phone"),
+    (re.compile(r"\b\d{1,3}\.\d{1,3}\.\d{1,3}\.\d{1,3}\b"), "ip_address"),
+]
+
+
+def _redact(value: str) -> str:
+    if len(value) <= 4:
+        return "***"
+    return value[:2] + "***" + value[-2:]
+
+
+def scan_pii(path: str, max_files: int = 300) -> list[Finding]:
+    """Bounded PII sweep; matched values are ALWAYS redacted in evidence."""
+    findings: list[Finding] = []
+    base = Path(path)
+    files = [base] if base.is_file() else [
+        p for p in sorted(base.rglob("*"))
+        if p.is_file() and p.stat().st_size < 2_000_000
+    ][:max_files]
+    for f in files:
+        try:
+            text = f.read_text(errors="replace")[:500_000]
+        except (OSError, UnicodeError):
+            continue
+        hits: dict[str, int] = {}
+        sample: dict[str, str] = {}
+        for rx, kind in _PII_PATTERNS:
+            found = rx.findall(text)
+            if found:
+                hits[kind] = len(found)
+                sample[kind] = _redact(found[0] if isinstance(found[0], str)
+                                       else str(found[0]))
+        if not hits:
+            continue
+        findings.append(Finding(
+            finding_type=FindingType.SENSITIVE_DATA,
+            source=FindingSource.DSPM,
+            asset=Asset(name=f.name, asset_type="data_file", location=str(f)),
+            severity="high" if ("ssn" in hits or "card_number" in hits) else "low",
+            title=f"PII detected in {f.name}",
+            description=", ".join(f"{k}x{v}" for k, v in sorted(hits.items())),
+            evidence={"counts": hits, "redacted_samples": sample},
+            is_actionable=True,
+            id=stable_id("pii", str(f), *sorted(hits)),
+        ))
+    return findings
+
+
+# ── GPU posture ─────────────────────────────────────────────────────────────
+
+
+def gpu_scan(run=None) -> dict:
+    """ROCm/amdgpu posture inventory (CPU boxes report absence, never fail).
+
+    Feeds the AMD PSIRT vendor-advisory matching: ROCm component versions
+    are returned as packages on a gpu-infra agent."""
+    run = run or (lambda cmd: subprocess.run(cmd, capture_output=True, text=True,
+                                             timeout=20))
+    info: dict = {"rocm_present": False, "gpus": [], "driver": None,
+                  "rocm_version": None}
+    ver_path = Path("/opt/rocm/.info/version")
+    if ver_path.exists():
+        info["rocm_present"] = True
+        info["rocm_version"] = ver_path.read_text().strip()
+    smi = shutil.which("rocm-smi")
+    if smi:
+        try:
+            out = run([smi, "--showproductname", "--json"])
+            if out.returncode == 0 and out.stdout.strip():
+                data = json.loads(out.stdout)
+                for card, fields in sorted(data.items()):
+                    if isinstance(fields, dict):
+                        info["gpus"].append({
+                            "card": card,
+                            "product": fields.get("Card Series")
+                            or fields.get("Card series")
+                            or fields.get("Card model", "unknown"),
+                        })
+        except (OSError, ValueError, subprocess.TimeoutExpired):
+            pass
+    mod = Path("/sys/module/amdgpu/version")
+    if mod.exists():
+        try:
+            info["driver"] = mod.read_text().strip()
+        except OSError:
+            pass
+    return info
+
+
+def gpu_scan_agent(run=None) -> Optional[Agent]:
+    info = gpu_scan(run=run)
+    if not info["rocm_present"] and not info["gpus"]:
+        return None
+    pkgs = []
+    if info["rocm_version"]:
+        pkgs.append(Package(name="rocm", version=info["rocm_version"],
+                            ecosystem="rocm"))
+    return Agent(
+        name="gpu-infra", agent_type=AgentType.CUSTOM,
+        config_path="/opt/rocm",
+        mcp_servers=[MCPServer(name="rocm-stack", packages=pkgs,
+                               surface=ServerSurface.OS_PACKAGES)],
+        source="gpu-scan", metadata={"gpu_scan": info},
+    )
+
+
+# ── dataset cards / training pipelines ──────────────────────────────────────
+
+
+def scan_dataset_cards(path: str, max_files: int = 200):
+    """Dataset cards (HF README.md front-matter, croissant JSON-LD,
+    dataset_card.yaml) -> inventory + provenance findings."""
+    base = Path(path)
+    inventory, findings = [], []
+    cands: list[Path] = []
+    for name in ("dataset_card.yaml", "dataset_card.yml", "croissant.json",
+                 "dataset_infos.json", "README.md"):
+        cands.extend(base.rglob(name))
+    for f in sorted(set(cands))[:max_files]:
+        entry = None
+        try:
+            if f.suffix in (".yaml", ".yml"):
+                doc = yaml.safe_load(f.read_text(errors="replace")) or {}
+                if isinstance(doc, dict) and ("dataset" in doc or "name" in doc):
+                    entry = {"name": str(doc.get("name") or doc.get("dataset")),
+                             "license": doc.get("license"),
+                             "sources": doc.get("sources") or doc.get("urls") or []}
+            elif f.name == "croissant.json" or f.name == "dataset_infos.json":
+                doc = json.loads(f.read_text(errors="replace"))
+                if isinstance(doc, dict):
+                    entry = {"name": str(doc.get("name", f.parent.name)),
+                             "license": doc.get("license"),
+                             "sources": [d.get("contentUrl") for d in
+                                         doc.get("distribution", []) or []
+                                         if isinstance(d, dict)]}
+            elif f.name == "README.md":
+                text = f.read_text(errors="replace")
+                if text.startswith("---"):
+                    fm = text.split("---", 2)
+                    meta = yaml.safe_load(fm[1]) if len(fm) > 2 else None
+                    if isinstance(meta, dict) and ("dataset_info" in meta
+                                                   or "datasets" in meta
+                                                   or "license" in meta):
+                        entry = {"name": f.parent.name,
+                                 "license": meta.get("license"),
+                                 "sources": []}
+        except (OSError, ValueError, yaml.YAMLError):
+            continue
+        if entry is None:
+            continue
+        entry["path"] = str(f)
+        inventory.append(entry)
+        urls = [u for u in (entry.get("sources") or []) if isinstance(u, str)]
+        http_urls = [u for u in urls if u.startswith("http://")]
+        if http_urls:
+            findings.append(Finding(
+                finding_type=FindingType.SENSITIVE_DATA,
+                source=FindingSource.DSPM,
+                asset=Asset(name=entry["name"] or f.parent.name,
+                            asset_type="dataset", location=str(f)),
+                severity="medium",
+                title=f"Dataset '{entry['name']}' pulls sources over plain HTTP",
+                description=f"{len(http_urls)} unencrypted source URL(s)",
+                evidence={"http_sources": http_urls[:5]},
+                is_actionable=True,
+                id=stable_id("dataset-http", str(f)),
+            ))
+    return inventory, findings
+
+
+_PIPELINE_FILES = ("dvc.yaml", "MLproject", "pipeline.yaml", "pipeline.yml",
+                   "kfp.yaml", "train.yaml")
+
+
+def scan_training_pipelines(path: str, max_files: int = 200):
+    """Training pipeline definitions -> inventory (+ curl|sh style risks)."""
+    base = Path(path)
+    inventory, findings = [], []
+    cands: list[Path] = []
+    for name in _PIPELINE_FILES:
+        cands.extend(base.rglob(name))
+    for f in sorted(set(cands))[:max_files]:
+        try:
+            text = f.read_text(errors="replace")
+        except OSError:
+            continue
+        inventory.append({"path": str(f), "kind": f.name})
+        if re.search(r"curl[^|\n]*\|\s*(ba)?sh", text) or \
+           re.search(r"wget[^|\n]*\|\s*(ba)?sh", text):
+            findings.append(Finding(
+                finding_type=FindingType.SAST,
+                source=FindingSource.SAST,
+                asset=Asset(name=f.name, asset_type="training_pipeline",
+                            location=str(f)),
+                severity="high",
+                title=f"Pipeline {f.name} pipes remote scripts into a shell",
+                description="curl|sh-style execution in a training pipeline",
+                evidence={"file": str(f)},
+                is_actionable=True,
+                id=stable_id("pipeline-curlsh", str(f)),
+            ))
+    return inventory, findings
+
+
+# ── license check ───────────────────────────────────────────────────────────
+
+_DEFAULT_LICENSE_DENY = ("GPL-3.0", "AGPL-3.0", "SSPL-1.0")
+
+
+def license_check(report, deny: Optional[Iterable[str]] = None) -> list[Finding]:
+    """Denylist check over licenses already collected on packages (SBOM
+    ingest and manifest parsers populate Package.license where known)."""
+    deny_set = {d.lower() for d in (deny or _DEFAULT_LICENSE_DENY)}
+    findings = []
+    seen = set()
+    for agent in report.agents:
+        for server in agent.mcp_servers:
+            for pkg in server.packages:
+                lic = (pkg.license or pkg.license_expression or "").strip()
+                if not lic:
+                    continue
+                if not any(d in lic.lower() for d in deny_set):
+                    continue
+                key = (pkg.ecosystem, pkg.name, lic)
+                if key in seen:
+                    continue
+                seen.add(key)
+                findings.append(Finding(
+                    finding_type=FindingType.LICENSE,
+                    source=FindingSource.SBOM,
+                    asset=Asset(name=f"{pkg.name}@{pkg.version}",
+                                asset_type="package", location=agent.name),
+                    severity="medium",
+                    title=f"Denied license {lic} on {pkg.name}",
+                    description=f"{pkg.ecosystem}:{pkg.name}@{pkg.version} "
+                                f"is licensed {lic}",
+                    evidence={"license": lic},
+                    is_actionable=True,
+                    id=stable_id("license", pkg.ecosystem, pkg.name, lic),
+                ))
+    return findings
+
+
+# ── MCP server config health ────────────────────────────────────────────────
+
+
+def health_check(agents: Iterable[Agent]) -> list[dict]:
+    """Static health: is the launch command resolvable / the URL well-formed?
+    Never spawns servers and never touches the network."""
+    out = []
+    for agent in agents:
+        for server in agent.mcp_servers:
+            status = "ok"
+            detail = ""
+            if server.url:
+                if not re.match(r"^(https|wss)://", server.url):
+                    status, detail = "warn", "non-TLS transport URL"
+            elif server.command:
+                if shutil.which(server.command) is None and \
+                        not Path(server.command).exists():
+                    status, detail = "error", f"command not found: {server.command}"
+            else:
+                status, detail = "error", "no command or url configured"
+            out.append({"agent": agent.name, "server": server.name,
+                        "status": status, "detail": detail})
+    return out
+
+
+# ── model hash verification ─────────────────────────────────────────────────
+
+
+def verify_model_hash_manifest(model_dir: str, manifest_path: str) -> list[Finding]:
+    """Verify model artifacts against a pinned sha256 manifest
+    ({relative_path: sha256}).  Mismatch or absence is MODEL_INTEGRITY."""
+    import hashlib
+
+    findings: list[Finding] = []
+    base = Path(model_dir)
+    try:
+        manifest = json.loads(Path(manifest_path).read_text())
+    except (OSError, ValueError) as exc:
+        raise ValueError(f"unreadable hash manifest {manifest_path}: {exc}")
+    for rel, expected in sorted(manifest.items()):
+        f = base / rel
+        status = None
+        actual = ""
+        if not f.exists():
+            status = "missing"
+        else:
+            h = hashlib.sha256()
+            with open(f, "rb") as fh:
+                for chunk in iter(lambda: fh.read(1 << 20), b""):
+                    h.update(chunk)
+            actual = h.hexdigest()
+            if actual != str(expected).lower():
+                status = "mismatch"
+        if status:
+            findings.append(Finding(
+                finding_type=FindingType.MODEL_INTEGRITY,
+                source=FindingSource.MODEL_SCAN,
+                asset=Asset(name=rel, asset_type="model_file", location=str(f)),
+                severity="critical" if status == "mismatch" else "high",
+                title=f"Model artifact {rel}: hash {status}",
+                description=(f"expected sha256 {str(expected)[:16]}…, "
+                             + (f"got {actual[:16]}…" if actual else "file absent")),
+                evidence={"expected": str(expected), "actual": actual,
+                          "status": status},
+                is_actionable=True,
+                id=stable_id("model-hash", rel, status),
+            ))
+    return findings
