@@ -103,5 +103,129 @@ def render_report(report: AIBOMReport, console: Optional[Console] = None,
                 d_node.add(chain)
         console.print(root)
 
+    _render_remediation(report, console)
+    _render_framework_summary(report, console)
+    _render_exposure_paths(report, console, max_paths=5 if not verbose else 25)
+    _render_other_findings(report, console, verbose)
+    _render_scan_performance(report, console, verbose)
+
     if report.warnings:
         console.print(Panel("\n".join(report.warnings[:10]), title="warnings", style="yellow"))
+    gaps = [i for i in report.scan_run.issues
+            if getattr(i, "kind", "") == "coverage_gap"] if report.scan_run else []
+    if gaps:
+        console.print(Panel(
+            "\n".join(f"{g.scanner}: {g.message}" for g in gaps[:10]),
+            title="coverage gaps — PARTIAL evidence never renders as clean",
+            style="red"))
+
+
+def _render_remediation(report: AIBOMReport, console: Console) -> None:
+    """Prioritized remediation plan: one row per (package, fix), findings
+    grouped — the operator's 'what do I upgrade first' view."""
+    from agentbom_amd.output.json_fmt import _build_remediation_json
+
+    plan = _build_remediation_json(report)
+    if not plan:
+        return
+    table = Table(title="Remediation plan (highest risk first)")
+    table.add_column("#", justify="right")
+    table.add_column("package")
+    table.add_column("upgrade to")
+    table.add_column("vulns", justify="right")
+    table.add_column("action")
+    table.add_column("risk", justify="right")
+    for i, item in enumerate(plan[:15], 1):
+        risk = item.get("max_risk_score", 0)
+        style = ("bold red" if risk >= 8 else "red" if risk >= 6
+                 else "yellow" if risk >= 4 else "")
+        table.add_row(
+            str(i), item.get("package", "?"),
+            item.get("fix_version") or "[dim]no fix yet[/dim]",
+            str(len(item.get("vulns", []))),
+            f"[{style}]{item.get('action', '')[:40]}[/{style}]" if style
+            else item.get("action", "")[:40],
+            f"{risk:.1f}",
+        )
+    console.print(table)
+
+
+def _render_framework_summary(report: AIBOMReport, console: Console) -> None:
+    from agentbom_amd.output.json_fmt import _build_framework_summary
+
+    summary = _build_framework_summary(report.blast_radii)
+    rows = [(fw, data) for fw, data in sorted(summary.items())
+            if isinstance(data, dict) and data.get("tagged_findings")]
+    if not rows:
+        return
+    table = Table(title="Compliance posture (16 frameworks)")
+    table.add_column("framework")
+    table.add_column("tagged findings", justify="right")
+    table.add_column("top controls")
+    for fw, data in rows[:16]:
+        controls = data.get("controls") or {}
+        top = sorted(controls.items(), key=lambda kv: -kv[1])[:4] \
+            if isinstance(controls, dict) else [(c, "") for c in list(controls)[:4]]
+        table.add_row(fw, str(data.get("tagged_findings", 0)),
+                      ", ".join(f"{c} ({n})" if n != "" else str(c)
+                                for c, n in top))
+    console.print(table)
+
+
+def _render_exposure_paths(report: AIBOMReport, console: Console,
+                           max_paths: int = 5) -> None:
+    """Ranked exposure paths: entry -> hops -> crown jewel, with fix."""
+    from agentbom_amd.models import blast_radius_to_finding
+    from agentbom_amd.output.exposure_path import exposure_path_for_finding
+
+    active = active_blast_radii(report.blast_radii)
+    if not active:
+        return
+    tree = Tree("[bold]Top exposure paths[/bold]")
+    for i, br in enumerate(active[:max_paths], 1):
+        path = exposure_path_for_finding(blast_radius_to_finding(br), rank=i)
+        hops = path.get("hops") or []
+        label = " [dim]->[/dim] ".join(str(h.get("label", h)) if isinstance(h, dict)
+                                       else str(h) for h in hops) or path.get("label", "")
+        node = tree.add(f"#{i} [bold]{path.get('riskScore', br.risk_score):.1f}[/bold] {label}")
+        fix = path.get("fix")
+        if fix:
+            node.add(f"[green]fix:[/green] {fix}")
+    console.print(tree)
+
+
+def _render_other_findings(report: AIBOMReport, console: Console,
+                           verbose: bool) -> None:
+    """Non-CVE findings stream: secrets, IaC, prompts, toxic combos, ..."""
+    others = [f for f in (report.findings or [])
+              if f.finding_type.value not in ("CVE",)]
+    if not others:
+        return
+    by_type: dict[str, list] = {}
+    for f in others:
+        by_type.setdefault(f.finding_type.value, []).append(f)
+    table = Table(title="Other findings")
+    table.add_column("type")
+    table.add_column("count", justify="right")
+    table.add_column("worst", justify="center")
+    table.add_column("example")
+    order = {"critical": 0, "high": 1, "medium": 2, "low": 3}
+    for ftype, items in sorted(by_type.items()):
+        worst = min((f.severity for f in items), key=lambda s: order.get(s, 9))
+        style = _SEV_STYLE.get(worst, "")
+        example = items[0].title if items else ""
+        table.add_row(ftype, str(len(items)), f"[{style}]{worst}[/{style}]",
+                      example[:60])
+    console.print(table)
+
+
+def _render_scan_performance(report: AIBOMReport, console: Console,
+                             verbose: bool) -> None:
+    perf = report.scan_performance_data or {}
+    if not perf or not verbose:
+        return
+    parts = [f"{k.removesuffix('_ms')} {v:.1f}ms" for k, v in perf.items()
+             if isinstance(v, (int, float))]
+    if parts:
+        console.print(Panel(" · ".join(parts), title="scan performance",
+                            style="dim"))

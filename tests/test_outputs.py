@@ -302,3 +302,51 @@ class TestNarratives:
         assert "1/3 checks pass" in text
         assert "CIS-1.1" in text and "no MFA" in text
         assert "UNVERIFIED" in text
+
+
+class TestConsoleRenderSections:
+    """Console render parity sections: remediation
+    plan, compliance posture, exposure-path tree, coverage-gap panel."""
+
+    def _render(self, report, verbose=True, width=120):
+        import io
+
+        from rich.console import Console
+
+        from agentbom_amd.output.console_render import render_report
+
+        buf = io.StringIO()
+        render_report(report, console=Console(file=buf, width=width),
+                      verbose=verbose)
+        return buf.getvalue()
+
+    def test_sections_present(self):
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        out = self._render(run_demo_scan())
+        assert "Remediation plan" in out
+        assert "Compliance posture" in out
+        assert "Top exposure paths" in out
+        assert "Findings (by risk)" in out
+        # remediation carries a concrete action for the demo PyYAML RCE
+        assert "pyyaml" in out
+
+    def test_other_findings_table(self):
+        from agentbom_amd.models.finding import (
+            Asset,
+            Finding,
+            FindingSource,
+            FindingType,
+        )
+        from agentbom_amd.scan.orchestrator import run_demo_scan
+
+        report = run_demo_scan()
+        report.findings.append(Finding(
+            finding_type=FindingType.CREDENTIAL_EXPOSURE,
+            source=FindingSource.SECRET_SCAN,
+            asset=Asset(name="env", asset_type="file", location="/app/.env"),
+            severity="high", title="AWS key in .env", description="x",
+            id="sec-1"))
+        out = self._render(report)
+        assert "Other findings" in out
+        assert "CREDENTIAL_EXPOSURE" in out
