@@ -1375,19 +1375,55 @@ def db_status_cmd(path: Optional[str]) -> None:
 
 
 @main.command(name="proxy", context_settings={"ignore_unknown_options": True})
-@click.argument("command", nargs=-1, required=True, type=click.UNPROCESSED)
+@click.argument("command", nargs=-1, required=False, type=click.UNPROCESSED)
+@click.option("--url", default=None,
+              help="Wrap a remote HTTP/SSE MCP server instead of a command.")
 @click.option("--policy", "policy_path", type=click.Path(exists=True), default=None)
 @click.option("--audit-log", type=click.Path(), default=None)
 @click.option("--block-on-warn", is_flag=True)
-def proxy_cmd(command: tuple[str, ...], policy_path: Optional[str],
-              audit_log: Optional[str], block_on_warn: bool) -> None:
-    """Wrap a target MCP server with inline runtime detectors."""
-    from agentbom_amd.runtime.proxy import AuditLog, McpProxy, ProxyPolicy
+@click.option("--sandbox", is_flag=True,
+              help="Run the wrapped command inside a docker/podman sandbox "
+                   "(read-only, no network, caps dropped).")
+@click.option("--sandbox-runtime", type=click.Choice(["docker", "podman"]),
+              default="docker")
+@click.option("--sandbox-image", default="node:20-slim")
+@click.option("--sandbox-network", is_flag=True,
+              help="Allow network inside the sandbox (default: none).")
+def proxy_cmd(command: tuple[str, ...], url: Optional[str],
+              policy_path: Optional[str], audit_log: Optional[str],
+              block_on_warn: bool, sandbox: bool, sandbox_runtime: str,
+              sandbox_image: str, sandbox_network: bool) -> None:
+    """Wrap a target MCP server (stdio command or HTTP/SSE URL) with
+    inline runtime detectors, policy and a hash-chained audit log."""
+    from agentbom_amd.runtime.proxy import (
+        AuditLog,
+        HttpMcpProxy,
+        McpProxy,
+        ProxyPolicy,
+        sandboxed_proxy,
+    )
 
     policy = ProxyPolicy.load(policy_path)
     if block_on_warn:
         policy.block_on_warn = True
-    proxy = McpProxy(list(command), policy=policy, audit=AuditLog(audit_log))
+    audit = AuditLog(audit_log)
+    if url:
+        if command:
+            raise click.UsageError("pass either a COMMAND or --url, not both")
+        proxy = HttpMcpProxy(url, policy=policy, audit=audit)
+    elif not command:
+        raise click.UsageError("pass the MCP server COMMAND or --url")
+    elif sandbox:
+        try:
+            proxy = sandboxed_proxy(list(command), policy=policy, audit=audit,
+                                    runtime=sandbox_runtime,
+                                    image=sandbox_image,
+                                    network=sandbox_network)
+        except RuntimeError as exc:
+            click.echo(f"error: {exc}", err=True)
+            sys.exit(2)
+    else:
+        proxy = McpProxy(list(command), policy=policy, audit=audit)
     sys.exit(proxy.run())
 
 
@@ -1401,7 +1437,11 @@ def gateway_group() -> None:
 @click.option("--port", type=int, default=8787)
 @click.option("--upstream", "upstreams", multiple=True,
               help="name=http://host:port upstream registrations")
-def gateway_serve_cmd(host: str, port: int, upstreams: tuple[str, ...]) -> None:
+@click.option("--registry", "registry_path", type=click.Path(exists=True),
+              default=None,
+              help="Upstream registry file (yaml/json: upstreams: [{name,url}])")
+def gateway_serve_cmd(host: str, port: int, upstreams: tuple[str, ...],
+                      registry_path: Optional[str] = None) -> None:
     """Serve the gateway relay over HTTP (/mcp/{upstream})."""
     import uvicorn
     from fastapi import FastAPI
@@ -1409,7 +1449,13 @@ def gateway_serve_cmd(host: str, port: int, upstreams: tuple[str, ...]) -> None:
     from agentbom_amd.runtime.gateway import Gateway, Upstream
 
     gw = Gateway()
-    for spec in upstreams:
+    registry_specs = []
+    if registry_path:
+        from agentbom_amd.runtime.gateway import load_upstream_registry
+
+        registry_specs = [f"{e['name']}={e['url']}"
+                          for e in load_upstream_registry(registry_path)]
+    for spec in list(upstreams) + registry_specs:
         name, _, url = spec.partition("=")
 
         def make_handler(u):

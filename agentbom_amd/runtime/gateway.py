@@ -76,6 +76,9 @@ class Gateway:
     def __init__(self) -> None:
         self.upstreams: dict[str, Upstream] = {}
         self.pipeline = DetectorPipeline()
+        self.identity_gate = IdentityGate()
+        self.drift_gate = DriftGate()
+        self.cost_gate = CostAnomalyGate()
         self.metrics = {"relays_total": 0, "blocked_total": 0, "redactions_total": 0,
                         "breaker_rejections_total": 0}
 
@@ -86,7 +89,9 @@ class Gateway:
         if name in self.upstreams:
             self.upstreams[name].quarantined = on
 
-    def relay(self, upstream_name: str, frame: dict[str, Any]) -> dict[str, Any]:
+    def relay(self, upstream_name: str, frame: dict[str, Any],
+              principal: Optional[str] = None,
+              cost: float = 1.0) -> dict[str, Any]:
         up = self.upstreams.get(upstream_name)
         msg_id = frame.get("id")
 
@@ -98,6 +103,18 @@ class Gateway:
         if up.quarantined:
             self.metrics["blocked_total"] += 1
             return err(-32002, f"upstream {upstream_name!r} is quarantined")
+        # conditional-access gates: revoked identity, catalog drift, cost
+        if principal is not None and not self.identity_gate.allow(principal):
+            self.metrics["blocked_total"] += 1
+            return err(-32005, f"identity {principal!r} is revoked")
+        if not self.drift_gate.allow(upstream_name):
+            self.metrics["blocked_total"] += 1
+            return err(-32006, f"upstream {upstream_name!r} tool catalog "
+                       "drifted; operator re-approval required")
+        if principal is not None and not self.cost_gate.record_and_check(
+                principal, cost):
+            self.metrics["blocked_total"] += 1
+            return err(-32007, f"cost budget exhausted for {principal!r}")
         action, alerts = self.pipeline.inspect(frame)
         if action == "block":
             self.metrics["blocked_total"] += 1
@@ -112,6 +129,14 @@ class Gateway:
         except Exception as exc:  # noqa: BLE001 — upstream boundary
             up.breaker.record_failure()
             return err(-32004, f"upstream error: {exc}")
+
+        # drift observation: pin/compare tools/list catalogs (rug-pull gate)
+        if frame.get("method") == "tools/list" and isinstance(response, dict):
+            tools = (response.get("result") or {}).get("tools")
+            if tools is not None and not self.drift_gate.observe(upstream_name, tools):
+                self.metrics["blocked_total"] += 1
+                return err(-32006, f"upstream {upstream_name!r} tool catalog "
+                           "drifted; operator re-approval required")
 
         # response-side inspection + DLP
         r_action, r_alerts = self.pipeline.inspect(response)
@@ -131,3 +156,99 @@ class Gateway:
                 response["result"] = _json.loads(redacted)
         self.metrics["relays_total"] += 1
         return response
+
+
+# ── conditional-access gates (reference gateway_server.py:321-703) ─────────
+
+
+@dataclass
+class IdentityGate:
+    """Revoked-identity gate: requests from revoked principals never relay."""
+
+    revoked: set = field(default_factory=set)
+
+    def revoke(self, principal: str) -> None:
+        self.revoked.add(principal)
+
+    def restore(self, principal: str) -> None:
+        self.revoked.discard(principal)
+
+    def allow(self, principal: Optional[str]) -> bool:
+        return principal not in self.revoked
+
+
+@dataclass
+class DriftGate:
+    """Tool-catalog drift gate: an upstream whose tools/list answer changes
+    from its pinned hash is blocked until an operator re-approves
+    (rug-pull protection — the reference's drift gate)."""
+
+    pinned: dict = field(default_factory=dict)   # upstream -> catalog hash
+    drifted: set = field(default_factory=set)
+
+    @staticmethod
+    def catalog_hash(tools: Any) -> str:
+        import hashlib
+        import json as _json
+
+        canon = _json.dumps(tools, sort_keys=True, default=str)
+        return hashlib.sha256(canon.encode()).hexdigest()
+
+    def observe(self, upstream: str, tools: Any) -> bool:
+        """Record a tools/list catalog; returns False when drift blocks."""
+        h = self.catalog_hash(tools)
+        pinned = self.pinned.get(upstream)
+        if pinned is None:
+            self.pinned[upstream] = h
+            return True
+        if h != pinned:
+            self.drifted.add(upstream)
+            return False
+        return True
+
+    def approve(self, upstream: str, tools: Any) -> None:
+        """Operator re-approval: repin the current catalog."""
+        self.pinned[upstream] = self.catalog_hash(tools)
+        self.drifted.discard(upstream)
+
+    def allow(self, upstream: str) -> bool:
+        return upstream not in self.drifted
+
+
+@dataclass
+class CostAnomalyGate:
+    """Per-principal sliding-window cost budget: calls beyond the budget in
+    the window are blocked (cost-anomaly gate, gateway_server.py:321)."""
+
+    budget_per_window: float = 1000.0
+    window_s: float = 3600.0
+    spend: dict = field(default_factory=dict)  # principal -> [(t, cost)]
+
+    def record_and_check(self, principal: str, cost: float = 1.0,
+                         now: Optional[float] = None) -> bool:
+        now = time.monotonic() if now is None else now
+        hist = [(t, c) for t, c in self.spend.get(principal, [])
+                if now - t < self.window_s]
+        total = sum(c for _, c in hist) + cost
+        hist.append((now, cost))
+        self.spend[principal] = hist
+        return total <= self.budget_per_window
+
+
+def load_upstream_registry(path) -> list[dict]:
+    """Upstream registry file: [{name, url, quarantined?, budget?}...]
+    (yaml or json).  The gateway CLI registers each entry."""
+    import json as _json
+    from pathlib import Path as _Path
+
+    import yaml as _yaml
+
+    text = _Path(path).read_text()
+    data = _yaml.safe_load(text) if str(path).endswith((".yaml", ".yml")) \
+        else _json.loads(text)
+    ups = data.get("upstreams", data) if isinstance(data, dict) else data
+    out = []
+    for entry in ups or []:
+        if isinstance(entry, dict) and entry.get("name") and entry.get("url"):
+            out.append(entry)
+    return out

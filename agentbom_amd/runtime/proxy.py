@@ -190,3 +190,165 @@ class McpProxy:
         except KeyboardInterrupt:
             proc.terminate()
             return 130
+
+
+# ── HTTP / SSE transport wrap (reference proxy.py HTTP+SSE frames) ─────────
+
+
+def parse_sse_stream(text: str) -> list[dict[str, Any]]:
+    """Parse a text/event-stream body into JSON-RPC frames.
+
+    Handles multi-line ``data:`` continuation and ignores comments/other
+    fields (the subset MCP's SSE transport uses)."""
+    frames: list[dict[str, Any]] = []
+    data_lines: list[str] = []
+    for raw in text.splitlines() + [""]:
+        line = raw.rstrip("\r")
+        if line.startswith("data:"):
+            data_lines.append(line[5:].lstrip())
+            continue
+        if line == "" and data_lines:
+            payload = "\n".join(data_lines)
+            data_lines = []
+            try:
+                frames.append(json.loads(payload))
+            except json.JSONDecodeError:
+                continue
+    return frames
+
+
+class HttpMcpProxy:
+    """stdio-facing proxy for an HTTP(S)/SSE MCP server.
+
+    The client speaks newline JSON-RPC on stdio; each frame passes the
+    SAME decide()/audit path as the stdio proxy, then POSTs to the
+    upstream URL.  ``application/json`` responses relay directly; a
+    ``text/event-stream`` response is parsed into frames and each relayed
+    after response-side inspection.  Offline mode refuses construction —
+    wrapping a remote server is inherently a network operation."""
+
+    def __init__(self, url: str, policy: Optional[ProxyPolicy] = None,
+                 audit: Optional[AuditLog] = None,
+                 pipeline: Optional[DetectorPipeline] = None,
+                 client=None, headers: Optional[dict[str, str]] = None):
+        from agentbom_amd.utils.http_client import check_offline, create_client
+
+        check_offline(url)
+        self.url = url
+        self.headers = dict(headers or {})
+        self._inner = McpProxy([], policy=policy, audit=audit, pipeline=pipeline)
+        self.client = client or create_client(timeout=60.0)
+
+    @property
+    def blocked(self) -> int:
+        return self._inner.blocked
+
+    @property
+    def relayed(self) -> int:
+        return self._inner.relayed
+
+    def forward(self, frame: dict[str, Any]) -> list[dict[str, Any]]:
+        """One request frame -> response frame(s) (SSE may yield several)."""
+        action, alerts = self._inner.decide(frame, "client->server")
+        self._inner.audit.record({
+            "direction": "client->server", "transport": "http",
+            "method": frame.get("method"), "id": frame.get("id"),
+            "action": action, "alerts": alerts,
+        })
+        if action == "block":
+            self._inner.blocked += 1
+            return [{"jsonrpc": "2.0", "id": frame.get("id"),
+                     "error": {"code": -32000,
+                               "message": "blocked by agent-bom proxy: "
+                               + "; ".join(a["message"] for a in alerts[:3])}}]
+        resp = self.client.post(self.url, json=frame, headers={
+            "Accept": "application/json, text/event-stream", **self.headers})
+        ctype = resp.headers.get("content-type", "")
+        if ctype.startswith("text/event-stream"):
+            replies = parse_sse_stream(resp.text)
+        else:
+            try:
+                replies = [resp.json()]
+            except ValueError:
+                replies = []
+        out: list[dict[str, Any]] = []
+        for reply in replies:
+            r_action, r_alerts = self._inner.decide(reply, "server->client")
+            self._inner.audit.record({
+                "direction": "server->client", "transport": "http",
+                "id": reply.get("id"), "action": r_action, "alerts": r_alerts,
+            })
+            if r_action == "block":
+                self._inner.blocked += 1
+                out.append({"jsonrpc": "2.0", "id": reply.get("id"),
+                            "error": {"code": -32000,
+                                      "message": "upstream response blocked"}})
+                continue
+            self._inner.relayed += 1
+            out.append(reply)
+        return out
+
+    def run(self) -> int:
+        """stdio loop: read frames from stdin, emit replies on stdout."""
+        for line in sys.stdin:
+            line = line.strip()
+            if not line:
+                continue
+            try:
+                frame = json.loads(line)
+            except json.JSONDecodeError:
+                continue
+            for reply in self.forward(frame):
+                sys.stdout.write(json.dumps(reply) + "\n")
+                sys.stdout.flush()
+        return 0
+
+
+# ── opt-in sandbox (reference proxy_sandbox.py) ─────────────────────────────
+
+
+def sandbox_command(command: list[str], runtime: str = "docker",
+                    image: str = "node:20-slim", network: bool = False,
+                    mounts: Optional[list[tuple[str, str]]] = None,
+                    memory: str = "512m", pids: int = 256) -> list[str]:
+    """Wrap a server launch in a container sandbox (docker/podman).
+
+    Deny-by-default posture: no network (unless opted in), read-only
+    root, memory/pids limits, all capabilities dropped, no privilege
+    escalation.  Returns the argv; the caller execs it when the runtime
+    is present (sandbox_available)."""
+    if runtime not in ("docker", "podman"):
+        raise ValueError(f"unsupported sandbox runtime {runtime!r}")
+    argv = [
+        runtime, "run", "-i", "--rm",
+        "--read-only",
+        "--cap-drop=ALL",
+        "--security-opt", "no-new-privileges",
+        f"--memory={memory}",
+        f"--pids-limit={pids}",
+    ]
+    if not network:
+        argv.append("--network=none")
+    for host, cont in mounts or []:
+        argv += ["-v", f"{host}:{cont}:ro"]
+    argv.append(image)
+    argv += command
+    return argv
+
+
+def sandbox_available(runtime: str = "docker") -> bool:
+    import shutil
+
+    return shutil.which(runtime) is not None
+
+
+def sandboxed_proxy(command: list[str], policy: Optional[ProxyPolicy] = None,
+                    audit: Optional[AuditLog] = None, runtime: str = "docker",
+                    **sandbox_kw) -> McpProxy:
+    """McpProxy whose child runs inside the container sandbox."""
+    if not sandbox_available(runtime):
+        raise RuntimeError(
+            f"sandbox runtime {runtime!r} not installed; run without "
+            "--sandbox or install docker/podman")
+    return McpProxy(sandbox_command(command, runtime=runtime, **sandbox_kw),
+                    policy=policy, audit=audit)
