@@ -466,3 +466,76 @@ def test_dedup_match_step_equals_plain(dev):
     without = eng.step()
     for key in ("pkg_idx", "win_idx", "scores", "n_agents", "n_creds", "n_tools"):
         assert torch.equal(with_dd[key], without[key]), key
+
+
+def test_impact_query_truncation_invariants(dev):
+    """Oracle-check the TRUNCATED impact-query paths:
+    slab overflow (max_nodes < neighborhood) and the LDS hash-full path
+    (> 8192 distinct nodes).  Under truncation the kernel still guarantees:
+    a duplicate-free subset of the true <=max_hops neighborhood, reported
+    hop >= oracle min-hop (never better than BFS), entry 0 = the query at
+    hop 0, count <= max_nodes, truncated flag set."""
+    import numpy as np_
+
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+    from agentbom_amd.ops import cpu_ref, native
+    from agentbom_amd.scan.synth import generate_estate
+
+    # dense small estate: every server shares packages widely -> the reverse
+    # neighborhood of a popular package blows past both caps
+    est = generate_estate(n_agents=500, n_servers=600, n_packages=150_000,
+                          name_catalog=200, extra_pkg_share=2.0, seed=13)
+    eng = EstateEngine(est, device=str(dev))
+
+    # pick the highest reverse-degree package node (a zipf-head hub)
+    row_off = eng.rev["row_off"].cpu().numpy()
+    deg = np_.diff(row_off)
+    hub = int(np_.argmax(deg[est.pkg_base:])) + est.pkg_base
+    q = torch.tensor([hub], dtype=torch.int64, device=dev)
+
+    def check_invariants(nodes, hops, counts, trunc, source, oracle_hops,
+                         max_nodes, expect_trunc):
+        n = int(counts[0].item())
+        assert int(trunc[0].item()) == (1 if expect_trunc else 0)
+        assert 0 < n <= max_nodes
+        got_nodes = nodes[0, :n].cpu().numpy()
+        got_hops = hops[0, :n].cpu().numpy()
+        assert got_nodes[0] == source and got_hops[0] == 0
+        assert len(set(got_nodes.tolist())) == n, "duplicate nodes reported"
+        for node, hop in zip(got_nodes.tolist(), got_hops.tolist()):
+            assert node in oracle_hops, f"node {node} outside the true neighborhood"
+            assert oracle_hops[node] <= hop <= 4, (
+                f"hop {hop} for node {node} beats oracle {oracle_hops[node]}")
+
+    rev_oracle = cpu_ref.impact_query(
+        row_off, eng.rev["col"].cpu().numpy(),
+        np_.array([hub], dtype=np_.int64), max_hops=4, max_nodes=1 << 30)[0][0]
+    for max_nodes in (64, 1024):
+        out = eng.blast_radius_query(q, max_hops=4, max_nodes=max_nodes)
+        check_invariants(*out, source=hub, oracle_hops=rev_oracle,
+                         max_nodes=max_nodes,
+                         expect_trunc=len(rev_oracle) > max_nodes)
+
+    # LDS hash-full path (> 8192 distinct nodes): forward query from the
+    # heaviest agent — agents fan out to servers -> packages/creds/tools
+    fwd_off = eng.fwd["row_off"].cpu().numpy()
+    fwd_col = eng.fwd["col"].cpu().numpy()
+    fdeg = np_.diff(fwd_off)
+    heavy_agent = int(np_.argmax(fdeg[:est.n_agents]))
+    fwd_oracle = cpu_ref.impact_query(
+        fwd_off, fwd_col, np_.array([heavy_agent], dtype=np_.int64),
+        max_hops=4, max_nodes=1 << 30)[0][0]
+    assert len(fwd_oracle) > 8192, "estate too small to fill the LDS hash"
+    qn = torch.tensor([heavy_agent], dtype=torch.int32, device=dev)
+    out = native.impact_query(eng.fwd["row_off"], eng.fwd["col"], qn,
+                              etype=None, allowed_mask=0xFFFFFFFF,
+                              max_hops=4, max_nodes=16384)
+    check_invariants(*out, source=heavy_agent, oracle_hops=fwd_oracle,
+                     max_nodes=16384, expect_trunc=True)
+
+    # sanity: with generous caps on a small-neighborhood query, no truncation
+    tail = int(np_.argmin(deg[est.pkg_base:])) + est.pkg_base
+    _, _, counts2, trunc2 = eng.blast_radius_query(
+        torch.tensor([tail], dtype=torch.int64, device=dev),
+        max_hops=2, max_nodes=4096)
+    assert int(trunc2[0].item()) == 0
