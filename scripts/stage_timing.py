@@ -45,6 +45,23 @@ def main() -> None:
         return (time.perf_counter() - t0) / args.iters * 1000, out
 
     t_match, (pkg_idx, win_idx) = timed(eng.match)
+    # dedup-match path (what step() actually runs)
+    t_match_dd = None
+    if eng.match_dedup is not None:
+        from agentbom_amd.ops import native
+
+        dd = eng.match_dedup
+
+        def dd_match():
+            sp, sw = native.match(
+                dd["u_gk"], dd["u_hi"], dd["u_lo"], dd["u_flags"],
+                eng.arena["group_keys"], eng.arena["group_off"],
+                eng.arena["windows"], pkg_win_range=dd["u_ranges"])
+            from agentbom_amd.graph.gpu_engine import expand_dedup_matches
+
+            return expand_dedup_matches(eng.torch, dd, sp, sw)
+
+        t_match_dd, _ = timed(dd_match)
     t_reach, dist = timed(eng.dependency_reach)
     pkg_nodes = pkg_idx + est.pkg_base
     t_counts, counts = timed(lambda: eng.blast_counts(pkg_nodes))
@@ -54,6 +71,8 @@ def main() -> None:
         "packages": args.packages,
         "findings": int(pkg_idx.numel()),
         "match_ms": round(t_match, 3),
+        "match_dedup_ms": round(t_match_dd, 3) if t_match_dd is not None else None,
+        "unique_rows": int(eng.match_dedup["u_gk"].numel()) if eng.match_dedup else None,
         "reach_bfs_ms": round(t_reach, 3),
         "blast_counts_ms": round(t_counts, 3),
         "full_step_ms": round(t_step, 3),
