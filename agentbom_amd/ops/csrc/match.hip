@@ -100,6 +100,45 @@ __global__ void match_kernel(
         const uint64_t khi = pkg_key_hi[p];
         const uint64_t klo = pkg_key_lo[p];
 
+        // Cooperative wave walk: when the whole wave shares one window
+        // range (zipf-head groups under the sorted/dedup layouts, where
+        // the walk cost concentrates), each WINDOW is loaded once by one
+        // lane (coalesced across lanes) and broadcast by shuffle — 64x
+        // less L1 traffic than every lane re-walking the same run.
+        const uint32_t wbeg0 = (uint32_t)__shfl((int)wbeg, 0, 64);
+        const uint32_t wend0 = (uint32_t)__shfl((int)wend, 0, 64);
+        const unsigned long long same =
+            __ballot(wbeg == wbeg0 && wend == wend0);
+        if (same == ~0ull && wend - wbeg >= 16) {
+            for (uint32_t base = wbeg; base < wend; base += 64) {
+                const uint32_t w = base + lane;
+                uint8_t f = 0;
+                uint64_t ihi = 0, ilo = 0, fhi = 0, flo = 0, lhi = 0, llo = 0;
+                if (w < wend) {
+                    f = w_flags[w];
+                    ihi = w_intro_hi[w]; ilo = w_intro_lo[w];
+                    fhi = w_fixed_hi[w]; flo = w_fixed_lo[w];
+                    lhi = w_last_hi[w];  llo = w_last_lo[w];
+                }
+                const int nwin = (int)min((uint32_t)64, wend - base);
+                for (int j = 0; j < nwin; ++j) {
+                    const uint8_t fj = (uint8_t)__shfl((int)f, j, 64);
+                    if (fj & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
+                    const uint64_t jihi = __shfl(ihi, j, 64), jilo = __shfl(ilo, j, 64);
+                    const uint64_t jfhi = __shfl(fhi, j, 64), jflo = __shfl(flo, j, 64);
+                    const uint64_t jlhi = __shfl(lhi, j, 64), jllo = __shfl(llo, j, 64);
+                    if ((fj & WF_HAS_INTRO) && key_lt(khi, klo, jihi, jilo)) continue;
+                    if ((fj & WF_HAS_FIXED) && key_ge(khi, klo, jfhi, jflo)) continue;
+                    if ((fj & WF_HAS_LAST) && key_gt(khi, klo, jlhi, jllo)) continue;
+                    const unsigned idx = atomicAdd(out_count, 1u);
+                    if ((long long)idx < capacity) {
+                        out_pairs[idx] = ((uint64_t)p << 32) | (uint64_t)(base + j);
+                    }
+                }
+            }
+            continue;
+        }
+
         for (uint32_t w = wbeg; w < wend; ++w) {
             const uint8_t f = w_flags[w];
             if (f & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
