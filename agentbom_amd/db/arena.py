@@ -118,6 +118,19 @@ class AdvisoryArena:
         def t(a, dtype):
             return torch.from_numpy(np.ascontiguousarray(a)).to(device=device, dtype=dtype)
 
+        # AoS mirror: one 64-byte line per window [ihi ilo fhi flo lhi llo
+        # flags pad] — the match walk reads ONE cache line per window
+        # instead of touching 7 column streams (tail groups are short runs,
+        # so SoA pays ~7 lines for 56 bytes of data)
+        W = self.num_windows
+        packed = np.zeros((W, 8), dtype=np.uint64)
+        packed[:, 0] = self.intro_hi
+        packed[:, 1] = self.intro_lo
+        packed[:, 2] = self.fixed_hi
+        packed[:, 3] = self.fixed_lo
+        packed[:, 4] = self.last_hi
+        packed[:, 5] = self.last_lo
+        packed[:, 6] = self.flags.astype(np.uint64)
         return {
             "group_keys": t(self.group_keys.view(np.int64), torch.int64),
             "group_off": t(self.group_off.view(np.int32), torch.int32),
@@ -129,6 +142,7 @@ class AdvisoryArena:
                 "last_hi": t(self.last_hi.view(np.int64), torch.int64),
                 "last_lo": t(self.last_lo.view(np.int64), torch.int64),
                 "flags": t(self.flags, torch.uint8),
+                "packed": t(packed.view(np.int64), torch.int64),
             },
             "severity": t(self.severity, torch.uint8),
             "cvss": t(self.cvss, torch.float32),

@@ -45,6 +45,7 @@ __global__ void match_kernel(
     const uint64_t* __restrict__ w_last_hi,
     const uint64_t* __restrict__ w_last_lo,
     const uint8_t* __restrict__ w_flags,
+    const uint64_t* __restrict__ w_packed,        // [W*8] AoS 64B/window (nullable)
     const uint32_t* __restrict__ pkg_wbeg,        // [P] precomputed ranges
     const uint32_t* __restrict__ pkg_wend,        //     (nullable)
     uint64_t* __restrict__ out_pairs,             // [capacity]
@@ -139,6 +140,22 @@ __global__ void match_kernel(
             continue;
         }
 
+        if (w_packed) {
+            // AoS walk: one 64-byte line per window
+            for (uint32_t w = wbeg; w < wend; ++w) {
+                const uint64_t* rec = w_packed + (uint64_t)w * 8;
+                const uint8_t f = (uint8_t)rec[6];
+                if (f & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
+                if ((f & WF_HAS_INTRO) && key_lt(khi, klo, rec[0], rec[1])) continue;
+                if ((f & WF_HAS_FIXED) && key_ge(khi, klo, rec[2], rec[3])) continue;
+                if ((f & WF_HAS_LAST) && key_gt(khi, klo, rec[4], rec[5])) continue;
+                const unsigned idx = atomicAdd(out_count, 1u);
+                if ((long long)idx < capacity) {
+                    out_pairs[idx] = ((uint64_t)p << 32) | (uint64_t)w;
+                }
+            }
+            continue;
+        }
         for (uint32_t w = wbeg; w < wend; ++w) {
             const uint8_t f = w_flags[w];
             if (f & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
@@ -176,6 +193,7 @@ extern "C" int abom_match(
     const void* w_fixed_hi, const void* w_fixed_lo,
     const void* w_last_hi, const void* w_last_lo,
     const void* w_flags,
+    const void* w_packed,                        // nullable AoS windows
     const void* pkg_wbeg, const void* pkg_wend,  // nullable precomputed ranges
     void* out_pairs, void* out_count, long long capacity, void* stream) {
     const int block = 256;
@@ -188,6 +206,7 @@ extern "C" int abom_match(
                        (const uint64_t*)w_fixed_hi, (const uint64_t*)w_fixed_lo,
                        (const uint64_t*)w_last_hi, (const uint64_t*)w_last_lo,
                        (const uint8_t*)w_flags,
+                       (const uint64_t*)w_packed,
                        (const uint32_t*)pkg_wbeg, (const uint32_t*)pkg_wend,
                        (uint64_t*)out_pairs,
                        (unsigned int*)out_count, capacity);

@@ -29,7 +29,7 @@ _SIGNATURES = {
     "abom_device_count": ([], _i32),
     "abom_synchronize": ([_c], _i32),
     "abom_error_string": ([_i32], ctypes.c_char_p),
-    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c, _c] + [_c, _c, _i64, _c], _i32),
+    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c] + [_c, _c] + [_c, _c, _i64, _c], _i32),
     "abom_bfs_init": ([_c, _i64, _c], _i32),
     "abom_bfs_seed": ([_c, _i64, _c, _c, _c, _c, _c], _i32),
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
@@ -144,6 +144,7 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
         _ptr(windows["fixed_hi"]), _ptr(windows["fixed_lo"]),
         _ptr(windows["last_hi"]), _ptr(windows["last_lo"]),
         _ptr(windows["flags"]),
+        _ptr(windows["packed"]) if "packed" in windows else None,
         wbeg_p, wend_p,
         _ptr(out_pairs), _ptr(out_count), cap, _stream(),
     )
