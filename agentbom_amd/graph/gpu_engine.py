@@ -521,6 +521,90 @@ class EstateEngine:
             "reach_dist": dist,
         }
 
+    def rollup(self, step_res=None):
+        """Estate CONTAINS-tree collapse on device.
+
+        The estate tree is two levels (agent -> server -> package), so the
+        collapse is two segmented reductions kept on the GPU: per-server
+        severity histograms over contained findings (severity_histogram
+        HIP kernel on the GPU path), then index_add-merged up the USES
+        edges into per-agent aggregates.  Returns device tensors:
+        histograms [n,6] (codes 5..0 mapped to columns 0..5), worst code,
+        finding counts — the estate-scale analog of graph/rollup.py's
+        Python CONTAINS walk (reference graph/rollup.py:313-580).
+        """
+        torch = self.torch
+        est = self.estate
+        if step_res is None:
+            step_res = self.step()
+        win_idx = step_res["win_idx"].to(torch.int64)
+        sev_code = self.arena["severity"].to(torch.int64)[win_idx]  # 5..1
+        sev_col = (5 - sev_code).clamp(0, 5)  # 0=critical .. 5=unknown
+        pkg_nodes = (step_res["pkg_idx"] + est.pkg_base).to(torch.int64)
+
+        # finding -> containing servers (reverse CONTAINS), severity carried
+        uniq = torch.unique(pkg_nodes)
+        pos_of_uniq = torch.arange(uniq.numel(), device=self.device)
+        srv, carry = self._expand(self.rev, uniq, pos_of_uniq, ET_CONTAINS)
+        # distinct (package, server): duplicate CONTAINS edges count once
+        pair = torch.unique((carry << 32) | srv)
+        carry = pair >> 32
+        srv = pair & 0xFFFFFFFF
+        # map each (finding) to each server containing its package
+        fpos = torch.searchsorted(uniq, pkg_nodes)
+        # (uniq_pos -> servers) join back onto findings: build per-uniq run
+        order = torch.argsort(carry, stable=True)
+        srv_sorted = srv[order]
+        carry_sorted = carry[order]
+        run_counts = torch.bincount(carry_sorted, minlength=uniq.numel())
+        run_off = torch.zeros(uniq.numel() + 1, dtype=torch.int64, device=self.device)
+        torch.cumsum(run_counts, 0, out=run_off[1:])
+        cnt_f = run_counts[fpos]
+        rep = torch.repeat_interleave(
+            torch.arange(pkg_nodes.numel(), device=self.device), cnt_f)
+        ends = torch.cumsum(cnt_f, 0)
+        base = torch.cat([torch.zeros(1, dtype=torch.int64, device=self.device),
+                          ends[:-1]])
+        within = torch.arange(int(ends[-1].item()) if ends.numel() else 0,
+                              device=self.device) - base[rep]
+        srv_per = srv_sorted[run_off[fpos[rep]] + within]
+        sev_per = sev_col[rep]
+
+        srv_local = (srv_per - est.server_base).to(torch.int64)
+        if self.use_gpu:
+            from agentbom_amd.ops import native
+
+            srv_hist = native.severity_histogram(
+                srv_local.to(torch.int64).contiguous(),
+                sev_per.to(torch.uint8).contiguous(), est.n_servers)
+            srv_hist = srv_hist.to(torch.int64)
+        else:
+            flat = torch.bincount(srv_local * 6 + sev_per,
+                                  minlength=est.n_servers * 6)
+            srv_hist = flat.view(est.n_servers, 6)
+
+        # server -> agent merge (reverse USES rows of servers)
+        srv_ids = torch.arange(est.n_servers, device=self.device) + est.server_base
+        ag, srv_carry = self._expand(self.rev, srv_ids,
+                                     torch.arange(est.n_servers, device=self.device),
+                                     ET_USES)
+        agent_hist = torch.zeros(est.n_agents, 6, dtype=torch.int64,
+                                 device=self.device)
+        agent_hist.index_add_(0, ag, srv_hist[srv_carry])
+
+        def worst_of(hist):
+            has = hist > 0
+            first = torch.argmax(has.to(torch.int8), dim=1)
+            any_f = has.any(dim=1)
+            return torch.where(any_f, first, torch.full_like(first, 6))
+
+        return {
+            "server_hist": srv_hist, "agent_hist": agent_hist,
+            "server_worst": worst_of(srv_hist), "agent_worst": worst_of(agent_hist),
+            "server_findings": srv_hist.sum(dim=1),
+            "agent_findings": agent_hist.sum(dim=1),
+        }
+
     # ── attack / exposure paths (GPU path DP) ──────────────────────────────
 
     ET_LATERAL = 4  # reversed USES edge (server -> agent): lateral movement

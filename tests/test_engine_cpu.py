@@ -180,3 +180,41 @@ class TestOsvShapedArena:
         res = EstateEngine(osv_estate, device="cpu").step()
         rate = res["n_findings"] / osv_estate.n_packages
         assert 0.001 < rate < 0.25, f"finding rate {rate} out of realistic band"
+
+
+def test_rollup_tree_collapse_bruteforce(estate, engine):
+    """Engine rollup (device CONTAINS collapse) vs a dict-based recount."""
+    res = engine.step()
+    roll = engine.rollup(res)
+    from agentbom_amd.scan.synth import ET_CONTAINS, ET_USES
+
+    srv_of_pkg: dict[int, set] = {}
+    ag_of_srv: dict[int, set] = {}
+    for s, d, t in zip(estate.edge_src, estate.edge_dst, estate.edge_type):
+        if t == ET_CONTAINS:
+            srv_of_pkg.setdefault(int(d), set()).add(int(s))
+        elif t == ET_USES:
+            ag_of_srv.setdefault(int(d), set()).add(int(s))
+
+    import numpy as np_
+
+    exp_srv = np_.zeros((estate.n_servers, 6), dtype=np_.int64)
+    sev = estate.arena.severity[res["win_idx"].numpy()]
+    for pkg_i, sev_code in zip(res["pkg_idx"].numpy(), sev):
+        node = int(pkg_i) + estate.pkg_base
+        col = max(0, min(5, 5 - int(sev_code)))
+        for s in srv_of_pkg.get(node, ()):  # every server containing it
+            exp_srv[s - estate.server_base, col] += 1
+    assert np_.array_equal(roll["server_hist"].numpy(), exp_srv)
+
+    exp_ag = np_.zeros((estate.n_agents, 6), dtype=np_.int64)
+    for s in range(estate.n_servers):
+        for a in ag_of_srv.get(s + estate.server_base, ()):  # agents using it
+            exp_ag[a] += exp_srv[s]
+    assert np_.array_equal(roll["agent_hist"].numpy(), exp_ag)
+
+    # worst codes: 0=critical..5, 6=no findings
+    w = roll["server_worst"].numpy()
+    for s in range(estate.n_servers):
+        nz = np_.nonzero(exp_srv[s])[0]
+        assert w[s] == (nz[0] if len(nz) else 6)

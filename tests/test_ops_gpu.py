@@ -539,3 +539,19 @@ def test_impact_query_truncation_invariants(dev):
         torch.tensor([tail], dtype=torch.int64, device=dev),
         max_hops=2, max_nodes=4096)
     assert int(trunc2[0].item()) == 0
+
+
+def test_rollup_gpu_matches_cpu(estate, dev):
+    """Device rollup (severity_histogram kernel + index_add merge) == CPU."""
+    eng_g = EstateEngineForRollup = None
+    from agentbom_amd.graph.gpu_engine import EstateEngine
+
+    g = EstateEngine(estate, device=str(dev))
+    res_g = g.step()
+    roll_g = g.rollup(res_g)
+    c = EstateEngine(estate, device="cpu")
+    roll_c = c.rollup(c.step())
+    for key in ("server_hist", "agent_hist", "server_worst", "agent_worst",
+                "server_findings", "agent_findings"):
+        assert torch.equal(roll_g[key].cpu().to(torch.int64),
+                           roll_c[key].to(torch.int64)), key
