@@ -543,7 +543,6 @@ def test_impact_query_truncation_invariants(dev):
 
 def test_rollup_gpu_matches_cpu(estate, dev):
     """Device rollup (severity_histogram kernel + index_add merge) == CPU."""
-    eng_g = EstateEngineForRollup = None
     from agentbom_amd.graph.gpu_engine import EstateEngine
 
     g = EstateEngine(estate, device=str(dev))
