@@ -18,7 +18,7 @@ import click
 from agentbom_amd import __version__
 
 FORMATS = [
-    "console", "json", "html", "sarif", "cyclonedx", "spdx", "spdx2", "ocsf",
+    "console", "json", "html", "sarif", "cyclonedx", "spdx", "spdx2", "spdx3", "ocsf",
     "csv", "markdown", "plain", "junit", "prometheus", "parquet", "pdf",
     "svg", "badge", "graph", "mermaid", "dot", "graphml", "cypher",
 ]
@@ -56,6 +56,10 @@ def _render(report, fmt: str, output: Optional[str], verbose: bool) -> None:
         from agentbom_amd.output.spdx_fmt import to_spdx
 
         text = json.dumps(to_spdx(report), indent=2)
+    elif fmt == "spdx3":
+        from agentbom_amd.output.spdx_fmt import to_spdx3
+
+        text = json.dumps(to_spdx3(report), indent=2)
     elif fmt == "html":
         from agentbom_amd.output.html_fmt import to_html
 

@@ -39,8 +39,9 @@ _CTRL_RE = re.compile(r"[\x00-\x08\x0b-\x1f\x7f]")
 
 
 def _safe_text(value: Any, max_len: int = 1000) -> str:
-    text = _CTRL_RE.sub("", str(value or ""))
-    return text[:max_len]
+    from agentbom_amd.utils.security import fence_untrusted
+
+    return fence_untrusted(value, max_len=max_len)
 
 
 class IntrospectionError(Exception):

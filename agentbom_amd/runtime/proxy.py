@@ -74,8 +74,10 @@ class AuditLog:
         self._lock = threading.Lock()
 
     def record(self, entry: dict[str, Any]) -> None:
+        from agentbom_amd.utils.security import redact_structure
+
         with self._lock:
-            entry = dict(entry)
+            entry = redact_structure(dict(entry))
             entry["ts"] = time.time()
             entry["prev_hash"] = self._prev_hash
             body = json.dumps(entry, sort_keys=True, default=str)

@@ -350,3 +350,24 @@ class TestConsoleRenderSections:
         out = self._render(report)
         assert "Other findings" in out
         assert "CREDENTIAL_EXPOSURE" in out
+
+
+def test_spdx3_jsonld_shape():
+    from agentbom_amd.output.spdx_fmt import to_spdx3
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    doc = to_spdx3(run_demo_scan())
+    assert doc["@context"].endswith("spdx-context.jsonld")
+    graph = doc["@graph"]
+    types = [e.get("type") for e in graph]
+    assert "CreationInfo" in types and "SpdxDocument" in types
+    pkgs = [e for e in graph if e.get("type") == "software_Package"]
+    assert pkgs, "no packages exported"
+    for p in pkgs:
+        assert p["spdxId"].startswith("urn:agent-bom:pkg:")
+        purl = p["externalIdentifier"][0]["identifier"]
+        assert purl.startswith("pkg:")
+    sdoc = next(e for e in graph if e.get("type") == "SpdxDocument")
+    rels = [e for e in graph if e.get("type") == "Relationship"]
+    assert len(rels) == len(sdoc["rootElement"]) == len(pkgs)
+    assert {r["to"][0] for r in rels} == {p["spdxId"] for p in pkgs}
