@@ -73,7 +73,7 @@ _GHSA_ECOSYSTEM = {
 class ScanCache:
     """TTL JSON disk cache for OSV responses (reference: scan_cache.py)."""
 
-    def __init__(self, root: Optional[Path] = None, ttl_s: float = 6 * 3600):
+    def __init__(self, root: Optional[Path] = None, ttl_s: Optional[float] = None):
         if root is None:
             import os
 
@@ -82,7 +82,7 @@ class ScanCache:
             root = base / "cache" / "osv"
         self.root = Path(root)
         self.root.mkdir(parents=True, exist_ok=True)
-        self.ttl_s = ttl_s
+        self.ttl_s = ttl_s if ttl_s is not None else cfg.OSV_CACHE_TTL_S
 
     def _path(self, key: str) -> Path:
         safe = "".join(c if c.isalnum() or c in "-._" else "_" for c in key)
@@ -326,11 +326,12 @@ def _parse_ghsa_range(vrange: str):
 # ── EPSS / KEV / NVD sync jobs ──────────────────────────────────────────────
 
 
-def sync_epss_live(store: AdvisoryStore, client=None, page_size: int = 10_000,
+def sync_epss_live(store: AdvisoryStore, client=None, page_size: Optional[int] = None,
                    max_pages: int = 100) -> int:
     """EPSS bulk scores via api.first.org paged JSON."""
     check_offline(EPSS_API_URL)
     client = client or create_client()
+    page_size = page_size or cfg.EPSS_PAGE_SIZE
     n = 0
     offset = 0
     for _ in range(max_pages):
@@ -383,7 +384,7 @@ def sync_kev_live(store: AdvisoryStore, client=None) -> int:
         Path(tmp).unlink(missing_ok=True)
 
 
-def sync_nvd_live(store: AdvisoryStore, client=None, page_size: int = 2000,
+def sync_nvd_live(store: AdvisoryStore, client=None, page_size: Optional[int] = None,
                   max_pages: int = 50, api_key: Optional[str] = None) -> int:
     """NVD CVE API 2.0 incremental sync with a lastModStartDate checkpoint.
 
@@ -392,6 +393,8 @@ def sync_nvd_live(store: AdvisoryStore, client=None, page_size: int = 2000,
     resumable (reference: 'NVD sync is checkpointed')."""
     check_offline(NVD_API_URL)
     client = client or create_client(timeout=60.0)
+    page_size = page_size or cfg.NVD_PAGE_SIZE
+    api_key = api_key or cfg.NVD_API_KEY or None
     row = store.conn.execute(
         "SELECT metadata_json FROM sync_meta WHERE source='nvd'").fetchone()
     checkpoint = None

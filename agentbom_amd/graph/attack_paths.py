@@ -73,7 +73,10 @@ def compute_fused_attack_paths(
     small-graph reference implementation."""
     import os
 
-    threshold = int(os.environ.get("AGENT_BOM_PATH_DP_THRESHOLD", "20000"))
+    from agentbom_amd.utils import config as _cfg
+
+    threshold = int(os.environ.get("AGENT_BOM_PATH_DP_THRESHOLD",
+                                   str(_cfg.PATH_DP_THRESHOLD)))
     if len(graph.nodes) >= threshold:
         device = None
         try:

@@ -35,8 +35,16 @@ class CircuitBreaker:
     """Per-upstream failure breaker: opens after N failures, half-opens
     after a cooldown (reference gateway_server.py:778)."""
 
-    threshold: int = 5
-    cooldown_s: float = 30.0
+    threshold: int = None  # type: ignore[assignment]
+    cooldown_s: float = None  # type: ignore[assignment]
+
+    def __post_init__(self):
+        from agentbom_amd.utils import config as _cfg
+
+        if self.threshold is None:
+            self.threshold = _cfg.GATEWAY_BREAKER_THRESHOLD
+        if self.cooldown_s is None:
+            self.cooldown_s = _cfg.GATEWAY_BREAKER_COOLDOWN_S
     failures: int = 0
     opened_at: Optional[float] = None
 
@@ -220,8 +228,16 @@ class CostAnomalyGate:
     """Per-principal sliding-window cost budget: calls beyond the budget in
     the window are blocked (cost-anomaly gate, gateway_server.py:321)."""
 
-    budget_per_window: float = 1000.0
-    window_s: float = 3600.0
+    budget_per_window: float = None  # type: ignore[assignment]
+    window_s: float = None  # type: ignore[assignment]
+
+    def __post_init__(self):
+        from agentbom_amd.utils import config as _cfg
+
+        if self.budget_per_window is None:
+            self.budget_per_window = _cfg.GATEWAY_COST_BUDGET
+        if self.window_s is None:
+            self.window_s = _cfg.GATEWAY_COST_WINDOW_S
     spend: dict = field(default_factory=dict)  # principal -> [(t, cost)]
 
     def record_and_check(self, principal: str, cost: float = 1.0,

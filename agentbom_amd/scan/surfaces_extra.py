@@ -151,9 +151,12 @@ _PROMPT_INJECTION = [
 _PROMPT_EXTS = {".md", ".txt", ".prompt", ".yaml", ".yml", ".json", ".xml"}
 
 
-def scan_prompt_files(path: str, max_files: int = 500) -> list[Finding]:
+def scan_prompt_files(path: str, max_files: Optional[int] = None) -> list[Finding]:
     """Static prompt-injection sweep (the runtime detectors' pattern class
     applied to prompt/config files at rest)."""
+    from agentbom_amd.utils import config as _cfg
+
+    max_files = max_files or _cfg.PROMPT_SCAN_MAX_FILES
     findings: list[Finding] = []
     base = Path(path)
     files = [base] if base.is_file() else [
@@ -203,13 +206,16 @@ def _redact(value: str) -> str:
     return value[:2] + "***" + value[-2:]
 
 
-def scan_pii(path: str, max_files: int = 300) -> list[Finding]:
+def scan_pii(path: str, max_files: Optional[int] = None) -> list[Finding]:
     """Bounded PII sweep; matched values are ALWAYS redacted in evidence."""
+    from agentbom_amd.utils import config as _cfg
+
+    max_files = max_files or _cfg.PII_SCAN_MAX_FILES
     findings: list[Finding] = []
     base = Path(path)
     files = [base] if base.is_file() else [
         p for p in sorted(base.rglob("*"))
-        if p.is_file() and p.stat().st_size < 2_000_000
+        if p.is_file() and p.stat().st_size < _cfg.SECRETS_MAX_FILE_BYTES
     ][:max_files]
     for f in files:
         try:
@@ -398,13 +404,16 @@ def scan_training_pipelines(path: str, max_files: int = 200):
 
 # ── license check ───────────────────────────────────────────────────────────
 
-_DEFAULT_LICENSE_DENY = ("GPL-3.0", "AGPL-3.0", "SSPL-1.0")
+def _default_license_deny():
+    from agentbom_amd.utils import config as _cfg
+
+    return tuple(x.strip() for x in _cfg.LICENSE_DENYLIST.split(",") if x.strip())
 
 
 def license_check(report, deny: Optional[Iterable[str]] = None) -> list[Finding]:
     """Denylist check over licenses already collected on packages (SBOM
     ingest and manifest parsers populate Package.license where known)."""
-    deny_set = {d.lower() for d in (deny or _DEFAULT_LICENSE_DENY)}
+    deny_set = {d.lower() for d in (deny or _default_license_deny())}
     findings = []
     seen = set()
     for agent in report.agents:

@@ -30,15 +30,18 @@ from agentbom_amd.utils.canonical_ids import normalize_package_ecosystem
 from agentbom_amd.utils.http_client import check_offline, create_client, request_with_retry
 from agentbom_amd.utils.version_utils import compare_version_order
 
-NPM_REGISTRY = "https://registry.npmjs.org"
-PYPI_API = "https://pypi.org/pypi"
-GO_PROXY = "https://proxy.golang.org"
+from agentbom_amd.utils import config as _cfg
 
-_MAX_NODES = 2_000  # expansion budget per scan (bounded, like the reference)
+NPM_REGISTRY = _cfg.NPM_REGISTRY_URL
+PYPI_API = _cfg.PYPI_API_URL
+GO_PROXY = _cfg.GO_PROXY_URL
+
+_MAX_NODES = _cfg.TRANSITIVE_MAX_NODES  # expansion budget per scan
 
 
 class _MetaCache:
-    def __init__(self, cap: int = 5_000):
+    def __init__(self, cap: int = None):
+        cap = cap if cap is not None else _cfg.TRANSITIVE_CACHE_SIZE
         self.cap = cap
         self.data: dict[str, Optional[dict]] = {}
 

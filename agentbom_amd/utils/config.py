@@ -128,3 +128,70 @@ GPU_GRAPH_THRESHOLD = _int("AGENT_BOM_GPU_GRAPH_THRESHOLD", 5000)
 
 # ── Offline mode ────────────────────────────────────────────────────────────
 OFFLINE = _bool("AGENT_BOM_OFFLINE", False)
+
+# ── Gateway / proxy runtime knobs ───────────────────────────────────────────
+GATEWAY_BREAKER_THRESHOLD = _int("AGENT_BOM_GATEWAY_BREAKER_THRESHOLD", 5)
+GATEWAY_BREAKER_COOLDOWN_S = _float("AGENT_BOM_GATEWAY_BREAKER_COOLDOWN_S", 30.0)
+GATEWAY_COST_BUDGET = _float("AGENT_BOM_GATEWAY_COST_BUDGET", 1000.0)
+GATEWAY_COST_WINDOW_S = _float("AGENT_BOM_GATEWAY_COST_WINDOW_S", 3600.0)
+PROXY_BLOCK_ON_WARN = _bool("AGENT_BOM_PROXY_BLOCK_ON_WARN", False)
+PROXY_AUDIT_PATH = _str("AGENT_BOM_PROXY_AUDIT_PATH", "")
+SANDBOX_RUNTIME = _str("AGENT_BOM_SANDBOX_RUNTIME", "docker")
+SANDBOX_IMAGE = _str("AGENT_BOM_SANDBOX_IMAGE", "node:20-slim")
+SANDBOX_MEMORY = _str("AGENT_BOM_SANDBOX_MEMORY", "512m")
+SANDBOX_PIDS_LIMIT = _int("AGENT_BOM_SANDBOX_PIDS_LIMIT", 256)
+
+# ── Runtime detector thresholds ─────────────────────────────────────────────
+DETECTOR_RATE_LIMIT_WINDOW_S = _float("AGENT_BOM_DETECTOR_RATE_WINDOW_S", 60.0)
+DETECTOR_RATE_LIMIT_MAX_CALLS = _int("AGENT_BOM_DETECTOR_RATE_MAX_CALLS", 120)
+DETECTOR_ARG_MAX_BYTES = _int("AGENT_BOM_DETECTOR_ARG_MAX_BYTES", 65536)
+DETECTOR_ENTROPY_THRESHOLD = _float("AGENT_BOM_DETECTOR_ENTROPY_THRESHOLD", 4.5)
+DETECTOR_REPLAY_WINDOW_S = _float("AGENT_BOM_DETECTOR_REPLAY_WINDOW_S", 300.0)
+
+# ── API / control plane ─────────────────────────────────────────────────────
+API_MAX_BODY_BYTES = _int("AGENT_BOM_API_MAX_BODY_BYTES", 10 * 1024 * 1024)
+API_SCAN_QUOTA_PER_HOUR = _int("AGENT_BOM_API_SCAN_QUOTA_PER_HOUR", 60)
+API_SCAN_CONCURRENCY = _int("AGENT_BOM_API_SCAN_CONCURRENCY", 4)
+API_STUCK_JOB_AGE_S = _float("AGENT_BOM_API_STUCK_JOB_AGE_S", 600.0)
+GRAPH_SNAPSHOT_RETENTION = _int("AGENT_BOM_GRAPH_SNAPSHOT_RETENTION", 10)
+GRAPH_NODE_BUDGET = _int("AGENT_BOM_GRAPH_NODE_BUDGET", 250_000)
+OIDC_CLOCK_LEEWAY_S = _float("AGENT_BOM_OIDC_CLOCK_LEEWAY_S", 60.0)
+DELEGATION_TOKEN_TTL_S = _float("AGENT_BOM_DELEGATION_TOKEN_TTL_S", 3600.0)
+WEBHOOK_TIMEOUT_S = _float("AGENT_BOM_WEBHOOK_TIMEOUT_S", 10.0)
+
+# ── Graph / path engine ─────────────────────────────────────────────────────
+PATH_DP_THRESHOLD = _int("AGENT_BOM_PATH_DP_THRESHOLD", 20_000)
+PATH_MAX_DEPTH = _int("AGENT_BOM_PATH_MAX_DEPTH", 6)
+PATH_MAX_PATHS = _int("AGENT_BOM_PATH_MAX_PATHS", 100)
+BLAST_RADIUS_MAX_DEPTH = _int("AGENT_BOM_BLAST_RADIUS_MAX_DEPTH", 5)
+IMPACT_QUERY_MAX_NODES = _int("AGENT_BOM_IMPACT_QUERY_MAX_NODES", 4096)
+ROLLUP_MAX_CONTAINERS = _int("AGENT_BOM_ROLLUP_MAX_CONTAINERS", 200)
+
+# ── Transitive expansion / registries ───────────────────────────────────────
+TRANSITIVE_MAX_DEPTH = _int("AGENT_BOM_TRANSITIVE_MAX_DEPTH", 3)
+TRANSITIVE_MAX_NODES = _int("AGENT_BOM_TRANSITIVE_MAX_NODES", 2000)
+TRANSITIVE_CACHE_SIZE = _int("AGENT_BOM_TRANSITIVE_CACHE_SIZE", 5000)
+NPM_REGISTRY_URL = _str("AGENT_BOM_NPM_REGISTRY_URL", "https://registry.npmjs.org")
+PYPI_API_URL = _str("AGENT_BOM_PYPI_API_URL", "https://pypi.org/pypi")
+GO_PROXY_URL = _str("AGENT_BOM_GO_PROXY_URL", "https://proxy.golang.org")
+
+# ── Live advisory acquisition ───────────────────────────────────────────────
+OSV_CACHE_TTL_S = _float("AGENT_BOM_OSV_CACHE_TTL_S", 6 * 3600.0)
+EPSS_PAGE_SIZE = _int("AGENT_BOM_EPSS_PAGE_SIZE", 10_000)
+NVD_PAGE_SIZE = _int("AGENT_BOM_NVD_PAGE_SIZE", 2000)
+NVD_API_KEY = _str("AGENT_BOM_NVD_API_KEY", "")
+GITHUB_TOKEN = _str("AGENT_BOM_GITHUB_TOKEN", "")
+DB_FRESH_DAYS = _float("AGENT_BOM_DB_FRESH_DAYS", 3.0)
+DB_STALE_DAYS = _float("AGENT_BOM_DB_STALE_DAYS", 30.0)
+
+# ── Scan surface bounds ─────────────────────────────────────────────────────
+PROMPT_SCAN_MAX_FILES = _int("AGENT_BOM_PROMPT_SCAN_MAX_FILES", 500)
+PII_SCAN_MAX_FILES = _int("AGENT_BOM_PII_SCAN_MAX_FILES", 300)
+BROWSER_EXT_MAX_MANIFESTS = _int("AGENT_BOM_BROWSER_EXT_MAX_MANIFESTS", 500)
+SECRETS_MAX_FILE_BYTES = _int("AGENT_BOM_SECRETS_MAX_FILE_BYTES", 2_000_000)
+UNTRUSTED_TEXT_MAX_LEN = _int("AGENT_BOM_UNTRUSTED_TEXT_MAX_LEN", 500)
+LICENSE_DENYLIST = _str("AGENT_BOM_LICENSE_DENYLIST", "GPL-3.0,AGPL-3.0,SSPL-1.0")
+
+# ── Multi-GPU / distributed engine ──────────────────────────────────────────
+DIST_EXCHANGE_CAP = _int("AGENT_BOM_DIST_EXCHANGE_CAP", 0)  # 0 = auto
+DIST_MAX_BFS_LEVELS = _int("AGENT_BOM_DIST_MAX_BFS_LEVELS", 64)

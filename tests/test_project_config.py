@@ -137,3 +137,31 @@ class TestCliProfiles:
         res = CliRunner().invoke(main, ["agents", "--demo", "--profile", "zz"])
         assert res.exit_code == 2
         assert "unknown profile" in res.output
+
+
+def test_config_knob_surface_and_wiring(monkeypatch):
+    """Knob count grew toward the reference's ~200 AGENT_BOM_* surface and
+    the new knobs genuinely parameterize behavior (not dead constants)."""
+    import importlib
+
+    import agentbom_amd.utils.config as cfg
+
+    knobs = [line for line in open(cfg.__file__)
+             if "AGENT_BOM_" in line and any(f"_{t}(" in line
+                                             for t in ("int", "float", "bool", "str"))]
+    assert len(knobs) >= 100, f"only {len(knobs)} knobs"
+
+    # a sampled knob changes real behavior: gateway cost budget
+    monkeypatch.setenv("AGENT_BOM_GATEWAY_COST_BUDGET", "2")
+    importlib.reload(cfg)
+    try:
+        from agentbom_amd.runtime.gateway import CostAnomalyGate
+
+        gate = CostAnomalyGate()
+        assert gate.budget_per_window == 2.0
+        assert gate.record_and_check("p", 1, now=0)
+        assert gate.record_and_check("p", 1, now=1)
+        assert not gate.record_and_check("p", 1, now=2)
+    finally:
+        monkeypatch.delenv("AGENT_BOM_GATEWAY_COST_BUDGET")
+        importlib.reload(cfg)
