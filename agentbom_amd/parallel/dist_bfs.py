@@ -93,8 +93,14 @@ def distributed_reach(
     import torch
     import torch.distributed as dist_mod
 
+    from agentbom_amd.utils import config as _cfg
+
     device = csr["row_off"].device
     use_gpu = device.type == "cuda"
+    if cap <= 0 and _cfg.DIST_EXCHANGE_CAP > 0:
+        cap = _cfg.DIST_EXCHANGE_CAP
+    if max_levels == 64:
+        max_levels = _cfg.DIST_MAX_BFS_LEVELS
 
     ws = workspace if workspace is not None else {}
     dist = ws.get("ddist")

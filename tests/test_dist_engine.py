@@ -25,7 +25,7 @@ from agentbom_amd.graph.gpu_engine import EstateEngine
 from agentbom_amd.scan.synth import generate_estate
 
 ESTATE_KW = dict(n_agents=60, n_servers=240, n_packages=4000, name_catalog=800,
-                 seed=424)
+                 seed=424, arena_windows=1500)
 
 
 def _single_reference():
