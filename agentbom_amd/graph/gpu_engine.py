@@ -574,8 +574,9 @@ class EstateEngine:
         if self.use_gpu:
             from agentbom_amd.ops import native
 
+            # the histogram kernel reads u32 owner indices
             srv_hist = native.severity_histogram(
-                srv_local.to(torch.int64).contiguous(),
+                srv_local.to(torch.int32).contiguous(),
                 sev_per.to(torch.uint8).contiguous(), est.n_servers)
             srv_hist = srv_hist.to(torch.int64)
         else:
