@@ -138,6 +138,28 @@ def expand_dedup_matches(torch, layout: dict, su, sw):
     return (packed >> 32), (packed & 0xFFFFFFFF)
 
 
+
+def rank_order(torch, scores):
+    """Deterministic finding rank: score desc (0.01 quantization), index asc.
+
+    Scores are clipped to [0, 10] and serialized rounded, so 0.01 buckets
+    preserve every visible distinction; packing (bucket, index) into ONE
+    integer key turns the rank into a single int sort instead of a
+    float+index argsort (round-1 roadmap lever #2).  Same formula on CPU
+    and GPU -> bit-equal order.
+    """
+    n = scores.numel()
+    if n == 0:
+        return torch.empty(0, dtype=torch.int64, device=scores.device)
+    q = (scores * 100).round().clamp(0, 1000).to(torch.int64)
+    bits = max(1, (n - 1).bit_length())
+    key = ((1000 - q) << bits) | torch.arange(n, device=scores.device)
+    if (1010 << bits) < (1 << 31):
+        key = key.to(torch.int32)
+    skey, _ = torch.sort(key)
+    return (skey.to(torch.int64)) & ((1 << bits) - 1)
+
+
 class EstateEngine:
     """Device-resident estate + advisory arena + the findings pipeline."""
 
@@ -462,7 +484,7 @@ class EstateEngine:
             return {
                 "n_findings": int(n_findings),
                 "scores": scores,
-                "order": torch.argsort(scores, descending=True, stable=True),
+                "order": rank_order(torch, scores),
                 "pkg_idx": pkg_idx,
                 "win_idx": win_idx,
                 "n_agents": n_agents,
@@ -507,8 +529,8 @@ class EstateEngine:
             )
         )
 
-        # deterministic rank: score desc, finding index asc
-        order = torch.argsort(scores, descending=True, stable=True)
+        # deterministic rank: quantized score desc, finding index asc
+        order = rank_order(torch, scores)
         return {
             "n_findings": int(n_findings),
             "scores": scores,

@@ -249,7 +249,9 @@ class DistEstateEngine:
                     reach.numpy(),
                 ))
 
-        order = torch.argsort(scores, descending=True, stable=True)
+        from agentbom_amd.graph.gpu_engine import rank_order
+
+        order = rank_order(torch, scores)
         return {
             "n_findings": int(n_findings),
             "scores": scores,
