@@ -96,4 +96,89 @@ def to_html(report: AIBOMReport) -> str:
 </table>
 <h2>Blast radius</h2>
 {''.join(details)}
+{_remediation_html(report)}
+{_frameworks_html(report)}
+{_exposure_paths_html(report)}
+{_other_findings_html(report)}
+{_warnings_html(report)}
 </body></html>"""
+
+
+def _remediation_html(report: AIBOMReport) -> str:
+    from agentbom_amd.output.json_fmt import _build_remediation_json
+
+    plan = _build_remediation_json(report)
+    if not plan:
+        return ""
+    rows = "".join(
+        f"<tr><td>{i}</td><td><code>{_e(p.get('package'))}</code></td>"
+        f"<td>{_e(p.get('fix_version') or 'no fix yet')}</td>"
+        f"<td>{len(p.get('vulns', []))}</td>"
+        f"<td>{_e(p.get('action', ''))}</td>"
+        f"<td>{p.get('max_risk_score', 0):.1f}</td></tr>"
+        for i, p in enumerate(plan[:20], 1))
+    return ("<h2>Remediation plan</h2><table><tr><th>#</th><th>package</th>"
+            "<th>upgrade to</th><th>vulns</th><th>action</th><th>risk</th></tr>"
+            f"{rows}</table>")
+
+
+def _frameworks_html(report: AIBOMReport) -> str:
+    from agentbom_amd.output.json_fmt import _build_framework_summary
+
+    summary = _build_framework_summary(report.blast_radii)
+    rows = []
+    for fw, data in sorted(summary.items()):
+        if not isinstance(data, dict) or not data.get("tagged_findings"):
+            continue
+        controls = data.get("controls") or {}
+        top = sorted(controls.items(), key=lambda kv: -kv[1])[:4] \
+            if isinstance(controls, dict) else []
+        rows.append(f"<tr><td>{_e(fw)}</td><td>{data['tagged_findings']}</td>"
+                    f"<td>{_e(', '.join(f'{c} ({n})' for c, n in top))}</td></tr>")
+    if not rows:
+        return ""
+    return ("<h2>Compliance posture</h2><table><tr><th>framework</th>"
+            "<th>tagged findings</th><th>top controls</th></tr>"
+            + "".join(rows) + "</table>")
+
+
+def _exposure_paths_html(report: AIBOMReport) -> str:
+    from agentbom_amd.models import active_blast_radii, blast_radius_to_finding
+    from agentbom_amd.output.exposure_path import exposure_path_for_finding
+
+    active = active_blast_radii(report.blast_radii)[:10]
+    if not active:
+        return ""
+    items = []
+    for i, br in enumerate(active, 1):
+        path = exposure_path_for_finding(blast_radius_to_finding(br), rank=i)
+        hops = path.get("hops") or []
+        label = " → ".join(_e(h.get("label", h)) if isinstance(h, dict) else _e(h)
+                           for h in hops) or _e(path.get("label", ""))
+        fix = path.get("fix")
+        items.append(f"<li><b>{path.get('riskScore', br.risk_score):.1f}</b> {label}"
+                     + (f" — <i>fix: {_e(fix)}</i>" if fix else "") + "</li>")
+    return "<h2>Top exposure paths</h2><ol>" + "".join(items) + "</ol>"
+
+
+def _other_findings_html(report: AIBOMReport) -> str:
+    others = [f for f in (report.findings or [])
+              if f.finding_type.value != "CVE"]
+    if not others:
+        return ""
+    by_type: dict[str, list] = {}
+    for f in others:
+        by_type.setdefault(f.finding_type.value, []).append(f)
+    rows = "".join(
+        f"<tr><td>{_e(t)}</td><td>{len(items)}</td>"
+        f"<td>{_e(items[0].title[:80])}</td></tr>"
+        for t, items in sorted(by_type.items()))
+    return ("<h2>Other findings</h2><table><tr><th>type</th><th>count</th>"
+            f"<th>example</th></tr>{rows}</table>")
+
+
+def _warnings_html(report: AIBOMReport) -> str:
+    if not report.warnings:
+        return ""
+    items = "".join(f"<li>{_e(w)}</li>" for w in report.warnings[:15])
+    return f"<h2>Warnings</h2><ul>{items}</ul>"

@@ -371,3 +371,15 @@ def test_spdx3_jsonld_shape():
     rels = [e for e in graph if e.get("type") == "Relationship"]
     assert len(rels) == len(sdoc["rootElement"]) == len(pkgs)
     assert {r["to"][0] for r in rels} == {p["spdxId"] for p in pkgs}
+
+
+def test_html_report_sections():
+    from agentbom_amd.output.html_fmt import to_html
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    h = to_html(run_demo_scan())
+    for section in ("Remediation plan", "Compliance posture",
+                    "Top exposure paths", "Blast radius"):
+        assert section in h, section
+    assert "<script" not in h  # self-contained, no external/active content
+    assert "pyyaml" in h
