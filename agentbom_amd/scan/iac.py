@@ -63,6 +63,24 @@ _TF_RULES = [
      "wildcard IAM action grant", "high", "T1078"),
     ("TF007", re.compile(r"enable_logging\s*=\s*false|logging\s*{\s*}"),
      "audit logging disabled", "medium", "T1562"),
+    ("TF008", re.compile(r'block_public_acls\s*=\s*false|block_public_policy\s*=\s*false'),
+     "S3 public access block disabled", "high", "T1530"),
+    ("TF009", re.compile(r'skip_final_snapshot\s*=\s*true'),
+     "database deleted without a final snapshot", "low", "T1485"),
+    ("TF010", re.compile(r'(?m)^\s*versioning\s*{\s*enabled\s*=\s*false'),
+     "bucket versioning disabled (ransom/delete resilience)", "medium", "T1485"),
+    ("TF011", re.compile(r'force_ssl\s*=\s*false|require_secure_transport\s*=\s*"?OFF'),
+     "TLS not enforced on data store", "high", "T1557"),
+    ("TF012", re.compile(r'associate_public_ip_address\s*=\s*true'),
+     "compute instance with public IP", "medium", "T1190"),
+    ("TF013", re.compile(r'(?i)kms_key_id\s*=\s*(""|null)'),
+     "customer-managed KMS key absent", "low", "T1530"),
+    ("TF014", re.compile(r'min_tls_version\s*=\s*"?(TLS1_0|TLS1_1|1\.0|1\.1)'),
+     "legacy TLS version allowed", "medium", "T1557"),
+    ("TF015", re.compile(r'(?m)^\s*ingress\s*{[^}]*from_port\s*=\s*(22|3389)\b[\s\S]{0,200}?0\.0\.0\.0/0'),
+     "SSH/RDP admin port open to the internet", "critical", "T1021"),
+    ("TF016", re.compile(r'(?i)authorized_networks\s*{\s*[^}]*0\.0\.0\.0/0'),
+     "cloud SQL authorized network open to the internet", "critical", "T1190"),
 ]
 
 _K8S_RULES = [
@@ -82,6 +100,20 @@ _K8S_RULES = [
      "cluster-admin role binding", "critical", "T1078"),
     ("K8S008", re.compile(r"imagePullPolicy:\s*Never"),
      "imagePullPolicy Never (stale/unscanned image)", "low", "T1525"),
+    ("K8S009", re.compile(r"hostPID:\s*true|hostIPC:\s*true"),
+     "pod shares host PID/IPC namespace", "high", "T1611"),
+    ("K8S010", re.compile(r"(?m)^\s*capabilities:\s*\n\s*add:[\s\S]{0,80}?(SYS_ADMIN|NET_ADMIN|ALL)\b"),
+     "dangerous Linux capability added", "high", "T1611"),
+    ("K8S011", re.compile(r"automountServiceAccountToken:\s*true"),
+     "service-account token automounted", "medium", "T1528"),
+    ("K8S012", re.compile(r"(?i)(?:env|value):[^\n]*(?:PASSWORD|SECRET|TOKEN|API_KEY)[^\n]*value:\s*\S+"),
+     "secret literal in pod env", "high", "T1552"),
+    ("K8S013", re.compile(r"(?m)^\s*securityContext:\s*{\s*}\s*$"),
+     "empty securityContext (no hardening)", "low", "T1611"),
+    ("K8S014", re.compile(r"type:\s*NodePort"),
+     "NodePort service exposes every node", "medium", "T1190"),
+    ("K8S015", re.compile(r"(?i)seccompProfile:\s*\n\s*type:\s*Unconfined"),
+     "seccomp unconfined", "high", "T1611"),
 ]
 
 _DOCKER_RULES = [
@@ -95,6 +127,17 @@ _DOCKER_RULES = [
      "unpinned base image (no tag or digest)", "medium", "T1525"),
     ("DKR005", re.compile(r"--no-check-certificate|--insecure\b|-k\s"),
      "TLS verification disabled in build", "medium", "T1557"),
+    ("DKR006", re.compile(r"(?m)^\s*ADD\s+https?://"),
+     "remote ADD without checksum verification", "medium", "T1195"),
+    ("DKR007", re.compile(r"(?m)^\s*EXPOSE\s+22\b"),
+     "SSH port exposed from a container", "medium", "T1021"),
+    ("DKR008", re.compile(r"(?m)apt-get[^\n]*install(?![^\n]*--no-install-recommends)"),
+     "apt install without --no-install-recommends (bloat/attack surface)",
+     "low", "T1105"),
+    ("DKR009", re.compile(r"(?m)^\s*COPY\s+(\.|\./|/)\s"),
+     "COPY of the whole build context (secrets risk)", "low", "T1552"),
+    ("DKR010", re.compile(r"(?m)chmod\s+(-R\s+)?0?777"),
+     "world-writable permissions set in image", "medium", "T1222"),
 ]
 
 _COMPOSE_RULES = [
@@ -104,6 +147,14 @@ _COMPOSE_RULES = [
      "docker socket mounted into container", "critical", "T1611"),
     ("CMP003", re.compile(r"network_mode:\s*[\"']?host"),
      "compose service on host network", "high", "T1610"),
+    ("CMP004", re.compile(r"(?m)^\s*-?\s*[\"']?\d+:\d+[\"']?\s*$[\s\S]{0,4}"),
+     "service port published to all interfaces", "low", "T1190"),
+    ("CMP005", re.compile(r"(?i)^\s*(?:-\s*)?\w*(PASSWORD|SECRET|TOKEN|API_KEY)\w*[=:]\s*\S+", re.M),
+     "secret literal in compose environment", "high", "T1552"),
+    ("CMP006", re.compile(r"cap_add:[\s\S]{0,60}?(SYS_ADMIN|NET_ADMIN|ALL)\b"),
+     "dangerous capability added to compose service", "high", "T1611"),
+    ("CMP007", re.compile(r"pid:\s*[\"']?host|ipc:\s*[\"']?host"),
+     "compose service shares host PID/IPC", "high", "T1611"),
 ]
 
 # GitHub Actions workflow supply-chain rules (reference: github_actions.py)

@@ -153,3 +153,80 @@ class TestGoSdk:
         assert self.GO_SRC.count("{") == self.GO_SRC.count("}")
         assert "TODO" not in self.GO_SRC
         assert self.GO_SRC.startswith("// Package agentbom")
+
+
+class TestExtendedIacRules:
+    """Round-2 rule-pack depth: sampled checks per family."""
+
+    def _scan(self, text, path):
+        from agentbom_amd.scan.iac import scan_iac_text
+
+        return {f.rule_id for f in scan_iac_text(text, path)}
+
+    def test_terraform_new_rules(self):
+        tf = '''
+resource "aws_s3_bucket_public_access_block" "b" {
+  block_public_acls = false
+}
+resource "aws_db_instance" "d" {
+  skip_final_snapshot = true
+}
+resource "aws_instance" "i" {
+  associate_public_ip_address = true
+}
+resource "aws_security_group" "sg" {
+  ingress {
+    from_port = 22
+    to_port = 22
+    cidr_blocks = ["0.0.0.0/0"]
+  }
+}
+'''
+        ids = self._scan(tf, "main.tf")
+        assert {"TF008", "TF009", "TF012", "TF015"} <= ids
+
+    def test_k8s_new_rules(self):
+        manifest = '''
+apiVersion: v1
+kind: Pod
+spec:
+  hostPID: true
+  automountServiceAccountToken: true
+  containers:
+    - name: c
+      securityContext:
+        capabilities:
+          add: ["SYS_ADMIN"]
+---
+apiVersion: v1
+kind: Service
+spec:
+  type: NodePort
+'''
+        ids = self._scan(manifest, "pod.yaml")
+        assert {"K8S009", "K8S010", "K8S011", "K8S014"} <= ids
+
+    def test_dockerfile_new_rules(self):
+        df = '''
+FROM ubuntu:22.04
+ADD https://example.com/tool.tar.gz /opt/
+RUN apt-get update && apt-get install -y curl
+RUN chmod -R 777 /app
+EXPOSE 22
+'''
+        ids = self._scan(df, "Dockerfile")
+        assert {"DKR006", "DKR007", "DKR008", "DKR010"} <= ids
+
+    def test_compose_new_rules(self):
+        compose = '''
+services:
+  app:
+    image: corp/app:1
+    pid: host
+    cap_add:
+      - NET_ADMIN
+    environment:
+      - DB_PASSWORD=hunter2
+'''
+        ids = self._scan(compose, "docker-compose.yml")
+        assert {"CMP005", "CMP006", "CMP007"} <= ids
