@@ -94,6 +94,56 @@ def evaluate_aws_inventory(inv: dict[str, Any]) -> list[CisCheckResult]:
     if trail is not None:
         _check(results, "CIS-3.1", "CloudTrail enabled in all regions", "high",
                "cloudtrail", bool(trail.get("MultiRegion", False)))
+        _check(results, "CIS-3.2", "CloudTrail log file validation enabled", "medium",
+               "cloudtrail", bool(trail.get("LogFileValidation", True)))
+        _check(results, "CIS-3.7", "CloudTrail logs encrypted with KMS", "medium",
+               "cloudtrail", bool(trail.get("KmsKeyId")))
+
+    root = inv.get("root_account")
+    if root is not None:
+        _check(results, "CIS-1.5", "Root account MFA enabled", "critical",
+               "iam:root", bool(root.get("MFAEnabled", False)))
+        _check(results, "CIS-1.7", "Root access keys do not exist", "critical",
+               "iam:root", not root.get("AccessKeysPresent", False))
+        _check(results, "CIS-1.1", "Root account not used in 90 days", "high",
+               "iam:root", not root.get("RecentlyUsed", False))
+
+    pw = inv.get("password_policy")
+    if pw is not None:
+        _check(results, "CIS-1.8", "Password policy minimum length >= 14", "medium",
+               "iam:password-policy", int(pw.get("MinimumLength", 0)) >= 14)
+        _check(results, "CIS-1.9", "Password reuse prevented (>= 24)", "low",
+               "iam:password-policy", int(pw.get("ReusePrevention", 0)) >= 24)
+
+    for vol in inv.get("ebs_volumes", []) or []:
+        vid = vol.get("VolumeId", "?")
+        _check(results, "CIS-2.2.1", "EBS volume encrypted", "medium",
+               f"ebs:{vid}", bool(vol.get("Encrypted", True)))
+
+    for fn in inv.get("lambda_functions", []) or []:
+        name = fn.get("FunctionName", "?")
+        env = (fn.get("Environment") or {}).get("Variables") or {}
+        leaked = [k for k in env
+                  if any(m in k.upper() for m in ("SECRET", "PASSWORD", "TOKEN",
+                                                  "API_KEY", "ACCESS_KEY"))]
+        _check(results, "LMB-1", "Lambda env free of secret-named variables",
+               "high", f"lambda:{name}", not leaked,
+               f"secret-named env: {', '.join(leaked[:3])}" if leaked else "")
+        _check(results, "LMB-2", "Lambda not exposed via public URL without auth",
+               "high", f"lambda:{name}",
+               not (fn.get("FunctionUrlAuthType") == "NONE"))
+
+    for kms in inv.get("kms_keys", []) or []:
+        kid = kms.get("KeyId", "?")
+        _check(results, "CIS-3.8", "KMS key rotation enabled", "medium",
+               f"kms:{kid}", bool(kms.get("RotationEnabled", True)))
+
+    for eks in inv.get("eks_clusters", []) or []:
+        name = eks.get("Name", "?")
+        _check(results, "EKS-1", "EKS API endpoint not public", "high",
+               f"eks:{name}", not eks.get("EndpointPublicAccess", False))
+        _check(results, "EKS-2", "EKS control-plane logging enabled", "medium",
+               f"eks:{name}", bool(eks.get("LoggingEnabled", True)))
 
     # GPU/AI infrastructure posture (cloud/gpu_infra.py analog)
     for ep in inv.get("inference_endpoints", []) or []:

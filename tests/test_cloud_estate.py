@@ -219,3 +219,39 @@ class TestDispatchAndCli:
         assert any(s.status.value in ("partial", "unavailable")
                    for s in report.scan_run.scopes
                    if s.name == "cloud_estate") or report.scan_run.issues
+
+
+class TestExtendedAwsChecks:
+    """Round-2 CIS pack depth: root/password-policy/EBS/Lambda/KMS/EKS."""
+
+    def test_new_check_families(self):
+        from agentbom_amd.scan.cloud import evaluate_aws_inventory
+
+        inv = {
+            "root_account": {"MFAEnabled": False, "AccessKeysPresent": True,
+                             "RecentlyUsed": True},
+            "password_policy": {"MinimumLength": 8, "ReusePrevention": 3},
+            "ebs_volumes": [{"VolumeId": "vol-1", "Encrypted": False}],
+            "lambda_functions": [{"FunctionName": "f1",
+                                  "Environment": {"Variables":
+                                                  {"DB_PASSWORD": "x"}},
+                                  "FunctionUrlAuthType": "NONE"}],
+            "kms_keys": [{"KeyId": "k1", "RotationEnabled": False}],
+            "eks_clusters": [{"Name": "c1", "EndpointPublicAccess": True,
+                              "LoggingEnabled": False}],
+            "cloudtrail": {"MultiRegion": True, "LogFileValidation": False},
+        }
+        results = evaluate_aws_inventory(inv)
+        fails = {r.check_id for r in results if r.status == "fail"}
+        assert {"CIS-1.5", "CIS-1.7", "CIS-1.1", "CIS-1.8", "CIS-1.9",
+                "CIS-2.2.1", "LMB-1", "LMB-2", "CIS-3.8", "EKS-1", "EKS-2",
+                "CIS-3.2"} <= fails
+
+    def test_clean_inventory_passes(self):
+        from agentbom_amd.scan.cloud import evaluate_aws_inventory
+
+        inv = {"root_account": {"MFAEnabled": True, "AccessKeysPresent": False},
+               "password_policy": {"MinimumLength": 16, "ReusePrevention": 24},
+               "eks_clusters": [{"Name": "c", "EndpointPublicAccess": False}]}
+        results = evaluate_aws_inventory(inv)
+        assert all(r.status == "pass" for r in results)
