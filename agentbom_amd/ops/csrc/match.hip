@@ -170,87 +170,6 @@ __global__ void match_kernel(
     }
 }
 
-// Dual-row walk (resident-estate ranges + packed-AoS path): each thread
-// owns rows 2t and 2t+1.  Adjacent rows in the sorted/dedup layouts almost
-// always share their group, so the window run is loaded ONCE per thread
-// and tested against TWO independent keys — the walk is dependent-load
-// latency-bound (PMC: MemUnitStalled ~0, VALUBusy 4%), so the second
-// in-flight compare chain is near-free ILP.
-__global__ void match_pair_kernel(
-    const uint64_t* __restrict__ pkg_key_hi,
-    const uint64_t* __restrict__ pkg_key_lo,
-    const uint8_t* __restrict__ pkg_flags,
-    long long num_packages,
-    const uint64_t* __restrict__ w_packed,
-    const uint32_t* __restrict__ pkg_wbeg,
-    const uint32_t* __restrict__ pkg_wend,
-    uint64_t* __restrict__ out_pairs,
-    unsigned int* __restrict__ out_count,
-    long long capacity) {
-    const long long stride = (long long)gridDim.x * blockDim.x;
-    const long long half = (num_packages + 1) >> 1;
-    for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < half;
-         t += stride) {
-        const long long p0 = t * 2;
-        const long long p1 = p0 + 1;
-        bool v0 = (pkg_flags[p0] & PF_ENCODABLE) != 0;
-        bool v1 = p1 < num_packages && (pkg_flags[p1] & PF_ENCODABLE) != 0;
-        uint32_t b0 = 0, e0 = 0, b1 = 0, e1 = 0;
-        if (v0) { b0 = pkg_wbeg[p0]; e0 = pkg_wend[p0]; v0 = b0 < e0; }
-        if (v1) { b1 = pkg_wbeg[p1]; e1 = pkg_wend[p1]; v1 = b1 < e1; }
-        if (!v0 && !v1) continue;
-        const uint64_t k0hi = v0 ? pkg_key_hi[p0] : 0, k0lo = v0 ? pkg_key_lo[p0] : 0;
-        const uint64_t k1hi = v1 ? pkg_key_hi[p1] : 0, k1lo = v1 ? pkg_key_lo[p1] : 0;
-        if (v0 && v1 && b0 == b1 && e0 == e1) {
-            for (uint32_t w = b0; w < e0; ++w) {
-                const uint64_t* rec = w_packed + (uint64_t)w * 8;
-                const uint8_t f = (uint8_t)rec[6];
-                if (f & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
-                const uint64_t ihi = rec[0], ilo = rec[1];
-                const uint64_t fhi = rec[2], flo = rec[3];
-                const uint64_t lhi = rec[4], llo = rec[5];
-                bool m0 = !((f & WF_HAS_INTRO) && key_lt(k0hi, k0lo, ihi, ilo))
-                       && !((f & WF_HAS_FIXED) && key_ge(k0hi, k0lo, fhi, flo))
-                       && !((f & WF_HAS_LAST) && key_gt(k0hi, k0lo, lhi, llo));
-                bool m1 = !((f & WF_HAS_INTRO) && key_lt(k1hi, k1lo, ihi, ilo))
-                       && !((f & WF_HAS_FIXED) && key_ge(k1hi, k1lo, fhi, flo))
-                       && !((f & WF_HAS_LAST) && key_gt(k1hi, k1lo, lhi, llo));
-                if (m0) {
-                    const unsigned idx = atomicAdd(out_count, 1u);
-                    if ((long long)idx < capacity)
-                        out_pairs[idx] = ((uint64_t)p0 << 32) | (uint64_t)w;
-                }
-                if (m1) {
-                    const unsigned idx = atomicAdd(out_count, 1u);
-                    if ((long long)idx < capacity)
-                        out_pairs[idx] = ((uint64_t)p1 << 32) | (uint64_t)w;
-                }
-            }
-            continue;
-        }
-        for (int which = 0; which < 2; ++which) {
-            const bool v = which == 0 ? v0 : v1;
-            if (!v) continue;
-            const long long p = which == 0 ? p0 : p1;
-            const uint32_t wb = which == 0 ? b0 : b1;
-            const uint32_t we = which == 0 ? e0 : e1;
-            const uint64_t khi = which == 0 ? k0hi : k1hi;
-            const uint64_t klo = which == 0 ? k0lo : k1lo;
-            for (uint32_t w = wb; w < we; ++w) {
-                const uint64_t* rec = w_packed + (uint64_t)w * 8;
-                const uint8_t f = (uint8_t)rec[6];
-                if (f & (WF_CPU_FALLBACK | WF_UNFIXED_SUPPRESSED)) continue;
-                if ((f & WF_HAS_INTRO) && key_lt(khi, klo, rec[0], rec[1])) continue;
-                if ((f & WF_HAS_FIXED) && key_ge(khi, klo, rec[2], rec[3])) continue;
-                if ((f & WF_HAS_LAST) && key_gt(khi, klo, rec[4], rec[5])) continue;
-                const unsigned idx = atomicAdd(out_count, 1u);
-                if ((long long)idx < capacity)
-                    out_pairs[idx] = ((uint64_t)p << 32) | (uint64_t)w;
-            }
-        }
-    }
-}
-
 }  // namespace abom
 
 namespace abom {
@@ -278,18 +197,6 @@ extern "C" int abom_match(
     const void* pkg_wbeg, const void* pkg_wend,  // nullable precomputed ranges
     void* out_pairs, void* out_count, long long capacity, void* stream) {
     const int block = 256;
-    if (pkg_wbeg && w_packed) {
-        // serving mode: dual-row ILP walk over the packed AoS windows
-        const int grid2 = abom::match_grid_for((num_packages + 1) / 2, block);
-        hipLaunchKernelGGL(abom::match_pair_kernel, dim3(grid2), dim3(block), 0,
-                           (hipStream_t)stream,
-                           (const uint64_t*)pkg_key_hi, (const uint64_t*)pkg_key_lo,
-                           (const uint8_t*)pkg_flags, num_packages,
-                           (const uint64_t*)w_packed,
-                           (const uint32_t*)pkg_wbeg, (const uint32_t*)pkg_wend,
-                           (uint64_t*)out_pairs, (unsigned int*)out_count, capacity);
-        return (int)hipGetLastError();
-    }
     const int grid = abom::match_grid_for(num_packages, block);
     hipLaunchKernelGGL(abom::match_kernel, dim3(grid), dim3(block), 0, (hipStream_t)stream,
                        (const uint64_t*)pkg_group_key, (const uint64_t*)pkg_key_hi,
