@@ -86,7 +86,7 @@ def _run_world(world: int, tmp_path):
 
 
 @pytest.mark.timeout(360)
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 3, 4])
 def test_dist_pipeline_equals_single_engine(world, tmp_path):
     _run_world(world, tmp_path)
     est, ref = _single_reference()
