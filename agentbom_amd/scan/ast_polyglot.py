@@ -146,6 +146,72 @@ _COMMENT_LINE = {
 }
 
 
+
+
+# languages whose line comments start with '#' (others use '//')
+_HASH_COMMENT_LANGS = {"ruby", "php"}  # php supports both # and //
+
+
+def strip_comments_and_strings(text: str, lang: str) -> str:
+    """Blank out string literals and comments, preserving line structure.
+
+    A small shared lexer (state machine) rather than per-line regex: sinks
+    inside strings ("use eval() carefully"), line comments and /* block
+    comments */ no longer false-positive, and block comments spanning
+    lines are handled.  Stripped characters become spaces so line/column
+    numbers survive."""
+    out = list(text)
+    i = 0
+    n = len(text)
+    hash_comments = lang in _HASH_COMMENT_LANGS
+
+    def blank(a: int, b: int) -> None:
+        for j in range(a, min(b, n)):
+            if out[j] != "\n":
+                out[j] = " "
+
+    while i < n:
+        c = text[i]
+        nxt = text[i + 1] if i + 1 < n else ""
+        if c == "/" and nxt == "/" and lang != "ruby":
+            j = text.find("\n", i)
+            j = n if j == -1 else j
+            blank(i, j)
+            i = j
+        elif c == "#" and hash_comments:
+            j = text.find("\n", i)
+            j = n if j == -1 else j
+            blank(i, j)
+            i = j
+        elif c == "/" and nxt == "*":
+            j = text.find("*/", i + 2)
+            j = n if j == -1 else j + 2
+            blank(i, j)
+            i = j
+        elif c in ("\"", "'", "`"):
+            q = c
+            j = i + 1
+            while j < n:
+                if text[j] == "\\":
+                    j += 2
+                    continue
+                if text[j] == q or (q != "`" and text[j] == "\n"):
+                    break
+                j += 1
+            blank(i + 1, j)  # keep the quotes, blank the contents
+            i = min(j + 1, n)
+        else:
+            i += 1
+    return "".join(out)
+
+
+# user-input source identifiers per language family (taint heuristic)
+_TAINT_SOURCES = re.compile(
+    r"\b(req|request|params|query|body|argv|args|stdin|input|r\.URL|"
+    r"r\.Form|getenv|environ|os\.Args|\$_GET|\$_POST|\$_REQUEST|"
+    r"event|payload|untrusted)\b", re.IGNORECASE)
+
+
 def language_for(path: str | Path) -> str | None:
     return _LANG_BY_SUFFIX.get(Path(path).suffix.lower())
 
@@ -162,15 +228,17 @@ def _enclosing_function(lines_before: list[str], lang: str) -> str:
 
 
 def analyze_source(text: str, path: str, lang: str) -> tuple[list[AstFinding], set[str]]:
-    """(findings, called symbols) for one non-Python source file."""
+    """(findings, called symbols) for one non-Python source file.
+
+    Analysis runs over a comment/string-stripped view (lexer pass) so
+    sinks quoted in strings or commented out never fire; snippets come
+    from the ORIGINAL text for readable evidence."""
     findings: list[AstFinding] = []
     calls: set[str] = set()
-    comment = _COMMENT_LINE.get(lang, "//")
-    lines = text.splitlines()
+    code = strip_comments_and_strings(text, lang)
+    lines = code.splitlines()
+    orig_lines = text.splitlines()
     for ln, line in enumerate(lines, start=1):
-        stripped = line.strip()
-        if stripped.startswith(comment):
-            continue
         for m in _CALL_RE.finditer(line):
             name = m.group(1)
             if name not in _CALL_STOPWORDS:
@@ -178,15 +246,21 @@ def analyze_source(text: str, path: str, lang: str) -> tuple[list[AstFinding], s
         for pattern, call, category, severity, cwe in _SINKS.get(lang, []):
             sm = pattern.search(line)
             if sm:
+                tail = line[sm.start():]
+                # taint: a user-input source identifier in the call, or a
+                # non-literal argument (anything but blanked quotes/consts)
+                arg_region = tail[: tail.find(")") + 1 or len(tail)]
+                tainted = bool(_TAINT_SOURCES.search(tail)) or bool(
+                    re.search(r"\(\s*[A-Za-z_][\w.\[\]]*", arg_region))
                 findings.append(AstFinding(
                     file=path, line=ln, call=call, category=category,
                     severity=severity, cwe=cwe,
                     entrypoint=_enclosing_function(lines[:ln], lang),
-                    snippet=stripped[:160],
-                    tainted="(" in line and not re.search(
-                        r"\(\s*['\"][^'\"]*['\"]\s*\)", line[sm.start():]),
+                    snippet=(orig_lines[ln - 1].strip()[:160]
+                             if ln <= len(orig_lines) else ""),
+                    tainted=tainted,
                 ))
-    for dm in _DEFS.get(lang, re.compile(r"$^")).finditer(text):
+    for dm in _DEFS.get(lang, re.compile(r"$^")).finditer(code):
         name = next((g for g in dm.groups() if g), None)
         if name:
             calls.add(name)
