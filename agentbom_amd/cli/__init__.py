@@ -262,7 +262,16 @@ def _scan_impl(
         if image:
             from agentbom_amd.scan.oci import oci_result_to_agent, scan_image
 
-            result = scan_image(image)
+            if Path(image).exists():
+                result = scan_image(image)
+            else:
+                from agentbom_amd.scan.oci_registry import scan_image_registry
+
+                if offline:
+                    click.echo("error: --image with a registry reference "
+                               "needs network (offline mode)", err=True)
+                    sys.exit(2)
+                result = scan_image_registry(image)
             for w in result.warnings:
                 click.echo(f"warning: {w}", err=True)
             agents.append(oci_result_to_agent(result))
@@ -544,8 +553,10 @@ def _scan_options(f):
         click.option("--no-gpu", is_flag=True, help="Force the CPU match path."),
         click.option("--sbom", type=click.Path(exists=True), default=None,
                      help="Scan an existing CycloneDX/SPDX SBOM."),
-        click.option("--image", type=click.Path(exists=True), default=None,
-                     help="Scan a container image (docker-save tar or OCI layout)."),
+        click.option("--image", default=None,
+                     help="Scan a container image: docker-save tar, OCI "
+                          "layout dir, or a registry reference "
+                          "(e.g. ghcr.io/org/app:tag — needs network)."),
         click.option("--filesystem", type=click.Path(exists=True), default=None,
                      help="Extract + scan packages from a directory tree."),
         click.option("--scan-secrets", is_flag=True,
