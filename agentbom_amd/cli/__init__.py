@@ -175,6 +175,10 @@ def _scan_impl(
     introspect: bool = False, health_check_flag: bool = False,
     scan_prompts: Optional[str] = None, scan_pii: Optional[str] = None,
     skill: Optional[str] = None,
+    # live cloud collectors
+    aws_live: bool = False, azure_live: Optional[str] = None,
+    azure_token: Optional[str] = None, gcp_live: Optional[str] = None,
+    gcp_token: Optional[str] = None,
     # push / integrations
     push_url: Optional[str] = None, push_api_key: Optional[str] = None,
     webhook: tuple = (), slack_webhook: Optional[str] = None,
@@ -359,6 +363,49 @@ def _scan_impl(
         run_scanner_driver("cloud_cis", report, aws_inventory)
     for spec in cloud_inventory:
         run_scanner_driver("cloud_estate", report, spec)
+    if aws_live or azure_live or gcp_live:
+        from agentbom_amd.scan.cloud import cis_result_to_finding
+
+        def _live(name, fn):
+            try:
+                results = fn()
+            except Exception as exc:
+                click.echo(f"warning: {name} live collection failed: {exc}; "
+                           "use the exported-inventory path instead", err=True)
+                return
+            report.findings.extend(
+                f for f in (cis_result_to_finding(r, name) for r in results)
+                if f is not None)
+
+        if aws_live:
+            def _aws():
+                from agentbom_amd.scan.cloud import evaluate_aws_inventory
+                from agentbom_amd.scan.cloud_live import AwsCollector
+
+                return evaluate_aws_inventory(AwsCollector().collect_inventory())
+            _live("aws", _aws)
+        if azure_live:
+            if not azure_token:
+                raise click.UsageError("--azure-live needs --azure-token")
+
+            def _azure():
+                from agentbom_amd.scan.cloud_estate import evaluate_azure_inventory
+                from agentbom_amd.scan.cloud_live import collect_azure_inventory
+
+                return evaluate_azure_inventory(
+                    collect_azure_inventory(azure_live, azure_token))
+            _live("azure", _azure)
+        if gcp_live:
+            if not gcp_token:
+                raise click.UsageError("--gcp-live needs --gcp-token")
+
+            def _gcp():
+                from agentbom_amd.scan.cloud_estate import evaluate_gcp_inventory
+                from agentbom_amd.scan.cloud_live import collect_gcp_inventory
+
+                return evaluate_gcp_inventory(
+                    collect_gcp_inventory(gcp_live, gcp_token))
+            _live("gcp", _gcp)
     if endpoint:
         run_scanner_driver("endpoint", report)
     if iac:
@@ -657,6 +704,16 @@ def _scan_options(f):
                      default=None, help="Bounded PII sweep (values redacted)."),
         click.option("--skill", type=click.Path(exists=True), default=None,
                      help="Alias of --skills."),
+        # live cloud collectors
+        click.option("--aws-live", "aws_live", is_flag=True,
+                     help="Collect the AWS inventory live (SigV4, env creds) "
+                          "and run the CIS pack on it."),
+        click.option("--azure-live", "azure_live", default=None, metavar="SUB_ID",
+                     help="Collect the Azure inventory live (ARM REST)."),
+        click.option("--azure-token", "azure_token", default=None),
+        click.option("--gcp-live", "gcp_live", default=None, metavar="PROJECT",
+                     help="Collect the GCP inventory live (REST)."),
+        click.option("--gcp-token", "gcp_token", default=None),
         # push / integrations
         click.option("--push-url", "push_url", default=None,
                      help="POST the full report JSON to a control plane."),
