@@ -175,6 +175,7 @@ def _scan_impl(
     introspect: bool = False, health_check_flag: bool = False,
     scan_prompts: Optional[str] = None, scan_pii: Optional[str] = None,
     skill: Optional[str] = None,
+    github_actions: Optional[str] = None, repo_inventory: Optional[str] = None,
     # live cloud collectors
     aws_live: bool = False, azure_live: Optional[str] = None,
     azure_token: Optional[str] = None, gcp_live: Optional[str] = None,
@@ -416,6 +417,10 @@ def _scan_impl(
         run_scanner_driver("skills", report, skills)
     if filesystem:
         run_scanner_driver("floating_refs", report, filesystem)
+    if github_actions or repo:
+        run_scanner_driver("ci_workflows", report, github_actions or repo)
+    if repo_inventory or repo:
+        run_scanner_driver("repo_inventory", report, repo_inventory or repo)
     if scan_prompts:
         from agentbom_amd.scan.surfaces_extra import scan_prompt_files
 
@@ -622,6 +627,15 @@ def _scan_options(f):
                      help="Scan Jupyter notebooks (pip installs, secrets in outputs, sinks)."),
         click.option("--skills", type=click.Path(exists=True), default=None,
                      help="Scan agent skill bundles (SKILL.md injection/grants/scripts)."),
+        click.option("--github-actions", "github_actions",
+                     type=click.Path(exists=True), default=None,
+                     help="Scan .github/workflows for agentic-CI usage + "
+                          "pipeline hardening (implied by --repo)."),
+        click.option("--repo-inventory", "repo_inventory",
+                     type=click.Path(exists=True), default=None,
+                     help="Collect the project directory/file/import "
+                          "inventory for the code-graph overlays (implied "
+                          "by --repo)."),
         click.option("--semgrep", type=click.Path(exists=True), default=None,
                      help="Ingest a semgrep --json result file (SAST findings "
                           "+ symbol-reachability joins)."),

@@ -176,4 +176,29 @@ def build_unified_graph_from_report_json(data: dict[str, Any],
                     g.add_edge(UnifiedEdge(p_id, v_id, RelationshipType.VULNERABLE_TO,
                                            weight=_sev_weight(v.get("severity", ""))))
                     g.add_edge(UnifiedEdge(v_id, p_id, RelationshipType.AFFECTS))
+    apply_report_overlays(g, data)
     return g
+
+
+def apply_report_overlays(graph, data: dict[str, Any]) -> dict[str, dict]:
+    """Phase-B overlays over a built graph, driven by the report JSON blocks
+    (reference: builder.py Phase-B overlay appliers, :1188-1253).
+
+    Order matters: repo structure places the file/dir nodes the code-graph
+    and CI overlays stitch onto; CNAPP sets exposure flags the ASPM
+    reachability verdicts read.  Every overlay no-ops on absent input."""
+    from agentbom_amd.graph.overlays_code import (
+        apply_aspm_overlay,
+        apply_ci_graph_overlay,
+        apply_cnapp_overlay,
+        apply_code_graph_overlay,
+        apply_repo_structure_overlay,
+    )
+
+    return {
+        "repo_structure": apply_repo_structure_overlay(graph, data),
+        "code_graph": apply_code_graph_overlay(graph, data),
+        "ci_graph": apply_ci_graph_overlay(graph, data),
+        "cnapp": apply_cnapp_overlay(graph),
+        "aspm": apply_aspm_overlay(graph, data),
+    }

@@ -670,4 +670,18 @@ def to_json(report: AIBOMReport) -> dict[str, Any]:
         "scan_performance": report.scan_performance_data,
         "intel_matches": report.intel_matches or [],
         "estate_score": _estate_score(report),
+        # optional side blocks travel under their reference key names; absent
+        # blocks are omitted entirely (contract: no null-noise keys)
+        **{key: value for key, value in (
+            ("project_inventory", report.project_inventory_data),
+            ("iac_findings", (report.iac_findings_data or {}).get("findings")
+             if report.iac_findings_data else None),
+            ("ai_inventory", report.ai_inventory_data),
+            ("license_report", report.license_report),
+            ("vex", report.vex_data),
+            ("context_graph", report.context_graph_data),
+            ("enforcement", report.enforcement_data),
+            ("delta", report.delta_data),
+        ) if value is not None},
+        **dict(sorted(report.extra_data.items())),
     }

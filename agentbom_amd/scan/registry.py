@@ -222,6 +222,27 @@ def _run_floating_refs(report: AIBOMReport, target: str) -> int:
     return len(refs)
 
 
+def _run_ci_workflows(report: AIBOMReport, target: str) -> int:
+    from agentbom_amd.scan.ci_workflows import scan_github_actions
+
+    agents, warnings = scan_github_actions(target)
+    report.agents.extend(agents)
+    report.warnings.extend(warnings)
+    if "github-actions" not in report.scan_sources:
+        report.scan_sources.append("github-actions")
+    return len(agents)
+
+
+def _run_repo_inventory(report: AIBOMReport, target: str) -> int:
+    from agentbom_amd.scan.repo_inventory import collect_project_inventory
+
+    inv = collect_project_inventory(target)
+    if inv is None:
+        return 0
+    report.project_inventory_data = inv
+    return len(inv["files"])
+
+
 def _ensure_builtins() -> None:
     if "secrets" in _REGISTRY:
         return
@@ -244,4 +265,10 @@ def _ensure_builtins() -> None:
         "skills", "skills", _run_skills, failure_mode=WARN_CONTINUE))
     register_scanner(ScannerRegistration(
         "floating_refs", "pinning", _run_floating_refs,
+        failure_mode=WARN_CONTINUE))
+    register_scanner(ScannerRegistration(
+        "ci_workflows", "code", _run_ci_workflows,
+        failure_mode=WARN_CONTINUE))
+    register_scanner(ScannerRegistration(
+        "repo_inventory", "code", _run_repo_inventory,
         failure_mode=WARN_CONTINUE))
