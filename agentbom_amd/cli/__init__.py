@@ -538,12 +538,19 @@ def _scan_impl(
         from agentbom_amd.scan.policy import evaluate_policy, load_policy
 
         pol = evaluate_policy(load_policy(policy), report.blast_radii)
-        for v in pol.get("violations", []):
-            click.echo(f"policy: rule {v['rule_id']} matched "
-                       f"{v['package']} ({v['vulnerability_id']})", err=True)
-        for w in pol.get("warnings", []):
-            click.echo(f"policy warning: rule {w['rule_id']} matched "
-                       f"{w['package']}", err=True)
+        if fmt == "console":
+            from rich.console import Console
+
+            from agentbom_amd.output.console_render import print_policy_results
+
+            print_policy_results(pol, Console(stderr=True))
+        else:
+            for v in pol.get("violations", []):
+                click.echo(f"policy: rule {v['rule_id']} matched "
+                           f"{v['package']} ({v['vulnerability_id']})", err=True)
+            for w in pol.get("warnings", []):
+                click.echo(f"policy warning: rule {w['rule_id']} matched "
+                           f"{w['package']}", err=True)
         policy_failed = not pol.get("passed", True)
 
     if warn_on:
@@ -1586,13 +1593,21 @@ def policy_check_cmd(policy_file: str, demo: bool, dry_run: bool) -> None:
 @main.command(name="diff")
 @click.argument("old_report", type=click.Path(exists=True))
 @click.argument("new_report", type=click.Path(exists=True))
-def diff_cmd(old_report: str, new_report: str) -> None:
+@click.option("--pretty", is_flag=True,
+              help="Render a rich console diff instead of JSON.")
+def diff_cmd(old_report: str, new_report: str, pretty: bool) -> None:
     """Diff two scan report JSON files (new/resolved findings)."""
     from agentbom_amd.scan.history import diff_reports
 
     old = json.loads(Path(old_report).read_text())
     new = json.loads(Path(new_report).read_text())
-    click.echo(json.dumps(diff_reports(old, new), indent=2, default=str))
+    result = diff_reports(old, new)
+    if pretty:
+        from agentbom_amd.output.console_render import print_diff
+
+        print_diff(result)
+    else:
+        click.echo(json.dumps(result, indent=2, default=str))
 
 
 @main.command(name="bench")

@@ -108,6 +108,11 @@ def render_report(report: AIBOMReport, console: Optional[Console] = None,
     _render_exposure_paths(report, console, max_paths=5 if not verbose else 25)
     _render_other_findings(report, console, verbose)
     _render_scan_performance(report, console, verbose)
+    if verbose:
+        print_severity_chart(report, console)
+        print_agent_tree(report, console)
+        print_cis_findings(report, console)
+        print_posture_summary(report, console)
 
     if report.warnings:
         console.print(Panel("\n".join(report.warnings[:10]), title="warnings", style="yellow"))
@@ -229,3 +234,189 @@ def _render_scan_performance(report: AIBOMReport, console: Console,
     if parts:
         console.print(Panel(" · ".join(parts), title="scan performance",
                             style="dim"))
+
+
+# ── standalone printers (reference console_render.py public surface) ───────
+
+
+def print_agent_tree(report: AIBOMReport,
+                     console: Optional[Console] = None) -> None:
+    """Full inventory tree: agent → server → (credentials, tools, packages),
+    vulnerable packages highlighted (reference print_agent_tree, :540)."""
+    console = console or Console()
+    root = Tree(f"[bold]Estate inventory[/bold] — {report.total_agents} "
+                f"agents, {report.total_servers} servers")
+    for agent in report.agents:
+        a_node = root.add(
+            f"[magenta]{agent.name}[/magenta] ({agent.agent_type.value})"
+            + (f" [dim]{agent.config_path}[/dim]" if agent.config_path else ""))
+        for srv in agent.mcp_servers:
+            vuln_ct = srv.total_vulnerabilities
+            badge = f" [red]{vuln_ct} vulns[/red]" if vuln_ct else ""
+            s_node = a_node.add(f"[cyan]{srv.name}[/cyan] "
+                                f"({srv.transport.value}){badge}")
+            if srv.credential_names:
+                c = s_node.add(f"[red]credentials "
+                               f"({len(srv.credential_names)})[/red]")
+                for name in srv.credential_names[:6]:
+                    c.add(name)
+            if srv.tools:
+                t = s_node.add(f"tools ({len(srv.tools)})")
+                for tool in srv.tools[:6]:
+                    t.add(tool.name)
+            vuln_pkgs = srv.vulnerable_packages
+            shown = vuln_pkgs[:8] if vuln_pkgs else srv.packages[:5]
+            if srv.packages:
+                p = s_node.add(f"packages ({len(srv.packages)})")
+                for pkg in shown:
+                    sevs = {v.severity.value for v in pkg.vulnerabilities}
+                    worst = next((s for s in ("critical", "high", "medium",
+                                              "low") if s in sevs), None)
+                    style = _SEV_STYLE.get(worst or "", "")
+                    mark = f" [{style}]{worst}[/{style}]" if worst else ""
+                    p.add(f"{pkg.name}@{pkg.version}{mark}")
+    console.print(root)
+
+
+def print_severity_chart(report: AIBOMReport,
+                         console: Optional[Console] = None,
+                         width: int = 40) -> None:
+    """Horizontal severity bar chart (reference print_severity_chart, :2168)."""
+    console = console or Console()
+    counts = report.severity_counts()
+    total = max(sum(counts.get(s, 0) for s in
+                    ("critical", "high", "medium", "low")), 1)
+    table = Table(title="Severity distribution", show_header=False,
+                  show_lines=False, box=None, padding=(0, 1))
+    table.add_column("sev", width=9)
+    table.add_column("bar")
+    table.add_column("n", justify="right")
+    for sev in ("critical", "high", "medium", "low"):
+        n = counts.get(sev, 0)
+        bar = "█" * max(round(width * n / total), 1 if n else 0)
+        style = _SEV_STYLE.get(sev, "")
+        table.add_row(f"[{style}]{sev}[/{style}]",
+                      f"[{style}]{bar}[/{style}]", str(n))
+    console.print(table)
+
+
+def print_cis_findings(report: AIBOMReport,
+                       console: Optional[Console] = None,
+                       show_passed: bool = False) -> None:
+    """Cloud CIS benchmark results table (reference print_cis_findings,
+    :1691): failures first, per-check evidence + remediation line."""
+    console = console or Console()
+    rows = report.extra_data.get("cis_benchmark_data") or []
+    if not rows:
+        return
+    order = {"fail": 0, "warn": 1, "manual": 2, "pass": 3}
+    rows = sorted((r for r in rows if isinstance(r, dict)),
+                  key=lambda r: (order.get(str(r.get("status")), 9),
+                                 str(r.get("check_id"))))
+    failed = [r for r in rows if r.get("status") == "fail"]
+    table = Table(title=f"Cloud CIS benchmark — {len(failed)} failing "
+                        f"of {len(rows)} checks")
+    table.add_column("check")
+    table.add_column("status")
+    table.add_column("title")
+    table.add_column("evidence")
+    for r in rows:
+        status = str(r.get("status", "?"))
+        if status == "pass" and not show_passed:
+            continue
+        style = {"fail": "bold red", "warn": "yellow",
+                 "manual": "dim", "pass": "green"}.get(status, "")
+        table.add_row(str(r.get("check_id", "?")),
+                      f"[{style}]{status}[/{style}]",
+                      str(r.get("title", ""))[:60],
+                      str(r.get("evidence", r.get("detail", "")))[:50])
+    console.print(table)
+
+
+def print_diff(diff: dict, console: Optional[Console] = None,
+               quiet: bool = False) -> None:
+    """Render a scan-over-scan diff (scan/history.diff_reports output):
+    new findings red, resolved green, package inventory delta."""
+    console = console or Console()
+    new = diff.get("new_findings") or []
+    resolved = diff.get("resolved_findings") or []
+    if new:
+        table = Table(title=f"[red]New findings ({len(new)})[/red]")
+        table.add_column("vulnerability")
+        table.add_column("package")
+        table.add_column("severity")
+        table.add_column("risk", justify="right")
+        for row in new:
+            sev = str(row.get("severity", "?"))
+            style = _SEV_STYLE.get(sev, "")
+            table.add_row(str(row.get("vulnerability_id", "?")),
+                          str(row.get("package_name",
+                                      row.get("package", "?"))),
+                          f"[{style}]{sev}[/{style}]",
+                          f"{float(row.get('risk_score', 0) or 0):.1f}")
+        console.print(table)
+    if resolved and not quiet:
+        table = Table(title=f"[green]Resolved ({len(resolved)})[/green]")
+        table.add_column("vulnerability")
+        table.add_column("package")
+        for row in resolved:
+            table.add_row(str(row.get("vulnerability_id", "?")),
+                          str(row.get("package", "?")))
+        console.print(table)
+    added = diff.get("packages_added") or []
+    removed = diff.get("packages_removed") or []
+    console.print(
+        f"unchanged: {diff.get('unchanged_count', 0)} · "
+        f"packages [green]+{len(added)}[/green]/[red]-{len(removed)}[/red]"
+        + (f" · drift: {', '.join(added[:5])}" if added else ""))
+
+
+def print_policy_results(policy_result: dict,
+                         console: Optional[Console] = None) -> None:
+    """Policy gate outcome (reference print_policy_results, :2131)."""
+    console = console or Console()
+    violations = policy_result.get("violations") or []
+    warnings = policy_result.get("warnings") or []
+    if not violations and not warnings:
+        console.print(f"[green]policy: PASS[/green] "
+                      f"({policy_result.get('rules_evaluated', 0)} rules)")
+        return
+    table = Table(title="Policy gate")
+    table.add_column("rule")
+    table.add_column("action")
+    table.add_column("vulnerability")
+    table.add_column("package")
+    table.add_column("risk", justify="right")
+    for hit in (*violations, *warnings):
+        action = str(hit.get("action", "fail"))
+        style = "bold red" if action == "fail" else "yellow"
+        table.add_row(str(hit.get("rule_id", "?")),
+                      f"[{style}]{action}[/{style}]",
+                      str(hit.get("vulnerability_id", "")),
+                      str(hit.get("package", "")),
+                      f"{float(hit.get('risk_score', 0) or 0):.1f}")
+    console.print(table)
+    verdict = "[bold red]FAIL[/bold red]" if violations else "[yellow]WARN[/yellow]"
+    console.print(f"policy: {verdict} — {len(violations)} violations, "
+                  f"{len(warnings)} warnings")
+
+
+def print_posture_summary(report: AIBOMReport,
+                          console: Optional[Console] = None) -> None:
+    """One-panel estate posture: grade + the dominant risk drivers
+    (reference print_posture_summary, :323)."""
+    from agentbom_amd.output.json_fmt import _estate_score
+
+    console = console or Console()
+    score = _estate_score(report)
+    grade = str(score.get("grade", "?"))
+    style = {"A": "green", "B": "green", "C": "yellow",
+             "D": "red", "F": "bold red"}.get(grade[:1], "")
+    drivers = score.get("drivers") or score.get("top_drivers") or []
+    lines = [f"estate grade: [{style}]{grade}[/{style}] "
+             f"(score {score.get('score', '?')})"]
+    lines += [f"• {d}" for d in list(drivers)[:5]]
+    counts = report.severity_counts()
+    lines.append(f"critical {counts['critical']} · high {counts['high']} · "
+                 f"KEV {sum(1 for br in report.blast_radii if br.vulnerability.is_kev)}")
+    console.print(Panel("\n".join(lines), title="security posture"))
