@@ -122,15 +122,27 @@ def build_dedup_match_layout(torch, arena: dict, gk, hi, lo, flags):
 
 def expand_dedup_matches(torch, layout: dict, su, sw):
     """Fan unique-row matches back out to per-package (orig_idx, win) pairs,
-    sorted by (pkg << 32 | win) — identical to the non-dedup output."""
+    sorted by (pkg << 32 | win) — identical to the non-dedup output.
+
+    Accepts UNSORTED (su, sw) — the one F-sized value sort at the end
+    canonicalizes, so match_finalize can skip its pair sort (sort=False).
+    (A cross-product construction replacing the F-int64 sort with a stable
+    int32 argsort was measured SLOWER — 2.50 vs 2.38 ms/step at 10M pkgs:
+    argsort sorts (key, index) pairs internally, so nothing was saved and
+    the extra expansion kernels were pure overhead.  See
+    profiles/r02_progress.md.)"""
+    dev = su.device
+    if su.numel() == 0:
+        e = torch.empty(0, dtype=torch.int64, device=dev)
+        return e, e.clone()
     run_off = layout["run_off"]
     cnt = run_off[su + 1] - run_off[su]
     rep = torch.repeat_interleave(
-        torch.arange(su.numel(), device=su.device), cnt)
+        torch.arange(su.numel(), device=dev), cnt)
     ends = torch.cumsum(cnt, 0)
-    base = torch.cat([torch.zeros(1, dtype=torch.int64, device=su.device), ends[:-1]])
+    base = torch.cat([torch.zeros(1, dtype=torch.int64, device=dev), ends[:-1]])
     within = torch.arange(int(ends[-1].item()) if ends.numel() else 0,
-                          device=su.device) - base[rep]
+                          device=dev) - base[rep]
     rows2 = run_off[su][rep] + within
     orig = layout["perm2"][rows2]
     packed = (orig << 32) | sw[rep]
@@ -452,7 +464,8 @@ class EstateEngine:
                         self.arena["windows"], pkg_win_range=self.pkg_win_range)
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
             torch.cuda.current_stream().wait_stream(side)
-            sp, sw = native.match_finalize(pending)
+            # downstream always re-canonicalizes; skip the pair sort
+            sp, sw = native.match_finalize(pending, sort=False)
             if dd is not None:
                 pkg_idx, win_idx = expand_dedup_matches(torch, dd, sp, sw)
             else:

@@ -155,11 +155,16 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
             "kw": {"pkg_win_range": pkg_win_range}}
 
 
-def match_finalize(pending: dict):
-    """Read the pending match count and produce sorted (pkg_idx, win_idx).
+def match_finalize(pending: dict, sort: bool = True):
+    """Read the pending match count and produce (pkg_idx, win_idx).
 
     On capacity overflow (count > cap) the match is relaunched synchronously
     with a grown buffer — rare, the launch heuristic is P//4.
+
+    ``sort=False`` skips the canonical (pkg<<32|win) sort — for callers that
+    impose their own order downstream (the dedup-expand path re-sorts after
+    fanning out to package rows, so sorting the distinct-row pairs here is
+    pure waste).
     """
     import torch
 
@@ -169,10 +174,12 @@ def match_finalize(pending: dict):
         # rather than slicing with a wrapped count.
         raise RuntimeError("match: device match count overflowed int32; shard the package batch")
     if n > pending["cap"]:
-        return match(*pending["args"], capacity=int(n * 1.2) + 1024,
-                     **pending.get("kw", {}))
+        relaunched = match_launch(*pending["args"], capacity=int(n * 1.2) + 1024,
+                                  **pending.get("kw", {}))
+        return match_finalize(relaunched, sort=sort)
     pairs = pending["out_pairs"][:n]
-    pairs, _ = torch.sort(pairs)
+    if sort:
+        pairs, _ = torch.sort(pairs)
     return (pairs >> 32).to(torch.int64), (pairs & 0xFFFFFFFF).to(torch.int64)
 
 

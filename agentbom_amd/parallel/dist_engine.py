@@ -176,7 +176,8 @@ class DistEstateEngine:
                         self.arena["windows"], pkg_win_range=self.pkg_win_range)
             dist = reach_dist if reach_dist is not None else self.dependency_reach()
             torch.cuda.current_stream().wait_stream(side)
-            sp, sw = native.match_finalize(pending)
+            # downstream always re-canonicalizes; skip the pair sort
+            sp, sw = native.match_finalize(pending, sort=False)
             if dd is not None:
                 # dedup rows -> local package rows (perm2 indexes the LOCAL
                 # owned arrays) -> global package index
