@@ -148,32 +148,131 @@ def ip_in_any_cidr(ip: str, cidrs: list[str]) -> bool:
 
 
 @dataclass
+class AccessContext:
+    """Request-time context a conditional-access policy evaluates against.
+
+    Unsupplied attributes ("" / [] / None) FAIL CLOSED against any policy
+    condition that constrains them — an unprovable "in prod", "from this
+    CIDR" or "device compliant" claim is treated as not met (reference
+    agent_identity_store.py AccessContext semantics)."""
+
+    scope: str = ""
+    tool_name: str = ""
+    environment: str = ""
+    source_ip: str = ""
+    device_id: str = ""
+    groups: list[str] = field(default_factory=list)
+    client_id: str = ""
+    device_managed: Optional[bool] = None
+    device_compliant: Optional[bool] = None
+    device_disk_encrypted: Optional[bool] = None
+    at: Optional[datetime] = None
+
+
+@dataclass
 class ConditionalAccessPolicy:
-    """Deny-by-condition policy layered over live identities."""
+    """Context-aware access rule (ABAC) layered over live identities.
+
+    ``effect="require"`` permits only when every configured condition holds;
+    ``effect="deny"`` blocks when every condition holds (deny wins, then
+    priority/policy_id order).  Empty scope/condition lists mean "any" /
+    "unconstrained"."""
 
     policy_id: str
     name: str
     scopes: list[str] = field(default_factory=list)  # empty = all scopes
     allowed_cidrs: list[str] = field(default_factory=list)
     deny_outside_hours: Optional[tuple[int, int]] = None  # (start_h, end_h) UTC
+    effect: str = "require"  # require | deny
+    status: str = "active"   # active | disabled
+    priority: int = 100
+    tools: list[str] = field(default_factory=list)
+    allowed_environments: list[str] = field(default_factory=list)
+    allowed_hours_utc: list[int] = field(default_factory=list)
+    allowed_weekdays: list[int] = field(default_factory=list)  # 0=Mon..6=Sun
+    allowed_devices: list[str] = field(default_factory=list)
+    allowed_groups: list[str] = field(default_factory=list)
+    allowed_clients: list[str] = field(default_factory=list)
+    require_device_managed: bool = False
+    require_device_compliant: bool = False
+    require_device_disk_encrypted: bool = False
 
-    def applies_to(self, scope: str) -> bool:
-        return not self.scopes or any(_scope_match(p, scope) for p in self.scopes)
+    def applies_to(self, scope: str, tool: str = "") -> bool:
+        scope_ok = not self.scopes or any(_scope_match(p, scope)
+                                          for p in self.scopes)
+        tool_ok = not self.tools or "*" in self.tools or tool in self.tools
+        return scope_ok and tool_ok
 
     def conditions_met(self, source_ip: Optional[str] = None,
-                       at: Optional[datetime] = None) -> tuple[bool, str]:
+                       at: Optional[datetime] = None,
+                       ctx: Optional[AccessContext] = None) -> tuple[bool, str]:
+        """(met, reason_if_not).  Every configured condition must hold."""
+        c = ctx or AccessContext(source_ip=source_ip or "", at=at)
+        if source_ip and not c.source_ip:
+            c.source_ip = source_ip
+        now = c.at or at or _now()
         if self.allowed_cidrs:
-            if not source_ip:
+            if not c.source_ip:
                 return False, f"policy {self.name}: source ip required"
-            if not ip_in_any_cidr(source_ip, self.allowed_cidrs):
-                return False, f"policy {self.name}: ip {source_ip} outside allowed ranges"
+            if not ip_in_any_cidr(c.source_ip, self.allowed_cidrs):
+                return False, (f"policy {self.name}: ip {c.source_ip} "
+                               "outside allowed ranges")
         if self.deny_outside_hours:
-            h = (at or _now()).hour
+            h = now.hour
             lo, hi = self.deny_outside_hours
             inside = lo <= h < hi if lo <= hi else (h >= lo or h < hi)
             if not inside:
-                return False, f"policy {self.name}: outside allowed hours {lo}-{hi} UTC"
+                return False, (f"policy {self.name}: outside allowed hours "
+                               f"{lo}-{hi} UTC")
+        if self.allowed_hours_utc and now.hour not in set(self.allowed_hours_utc):
+            return False, f"policy {self.name}: hour {now.hour} UTC not allowed"
+        if self.allowed_weekdays and now.weekday() not in set(self.allowed_weekdays):
+            return False, f"policy {self.name}: weekday not allowed"
+        if self.allowed_environments and c.environment.strip().lower() not in {
+                e.strip().lower() for e in self.allowed_environments}:
+            return False, (f"policy {self.name}: environment "
+                           f"{c.environment or '(unset)'} not allowed")
+        if self.allowed_devices and (not c.device_id
+                                     or c.device_id not in set(self.allowed_devices)):
+            return False, f"policy {self.name}: device not on allow-list"
+        if self.allowed_groups and not (set(c.groups) & set(self.allowed_groups)):
+            return False, f"policy {self.name}: caller not in an allowed group"
+        if self.allowed_clients and (not c.client_id
+                                     or c.client_id not in set(self.allowed_clients)):
+            return False, f"policy {self.name}: client app not allowed"
+        if self.require_device_managed and c.device_managed is not True:
+            return False, f"policy {self.name}: unmanaged device"
+        if self.require_device_compliant and c.device_compliant is not True:
+            return False, f"policy {self.name}: non-compliant device"
+        if self.require_device_disk_encrypted and c.device_disk_encrypted is not True:
+            return False, f"policy {self.name}: disk encryption not proven"
         return True, ""
+
+
+def evaluate_conditional_access(policies: list[ConditionalAccessPolicy],
+                                scope: str,
+                                ctx: AccessContext) -> tuple[bool, str, str]:
+    """(allowed, reason, policy_id) with deny precedence.
+
+    Any applying active ``deny`` policy whose conditions hold blocks; then
+    any applying ``require`` policy whose conditions do NOT hold blocks.
+    Ordering: (priority, policy_id).  Empty/none-applying ⇒ allow."""
+    applicable = sorted(
+        (p for p in policies
+         if p.status == "active" and p.applies_to(scope, ctx.tool_name)),
+        key=lambda p: (p.priority, p.policy_id))
+    for pol in applicable:
+        if pol.effect == "deny":
+            met, _ = pol.conditions_met(ctx=ctx)
+            if met:
+                return False, f"blocked by conditional-access policy '{pol.name}'", \
+                    pol.policy_id
+    for pol in applicable:
+        if pol.effect == "require":
+            met, why = pol.conditions_met(ctx=ctx)
+            if not met:
+                return False, why, pol.policy_id
+    return True, "", ""
 
 
 _SCHEMA = """
@@ -327,8 +426,14 @@ class AgentIdentityStore:
             return True
 
     def verify(self, raw_token: str, tool: Optional[str] = None,
-               source_ip: Optional[str] = None) -> dict[str, Any]:
-        """Verify a presented token: live, tool-scoped, conditional access."""
+               source_ip: Optional[str] = None,
+               ctx: Optional[AccessContext] = None) -> dict[str, Any]:
+        """Verify a presented token: live, tool-scoped, conditional access.
+
+        ``ctx`` carries the full ABAC request context (environment, device,
+        groups, client, posture); absent attributes fail closed against any
+        policy that constrains them.  Deny-effect policies take precedence
+        (evaluate_conditional_access)."""
         ident = self.get_by_token_hash(hash_token(raw_token))
         if ident is None:
             return {"valid": False, "reason": "unknown token"}
@@ -338,13 +443,17 @@ class AgentIdentityStore:
         if tool and not ident.tool_allowed(tool):
             return {"valid": False, "reason": f"tool {tool!r} not in allowed_tools",
                     "identity_id": ident.identity_id}
+        c = ctx or AccessContext()
+        if source_ip and not c.source_ip:
+            c.source_ip = source_ip
+        if tool and not c.tool_name:
+            c.tool_name = tool
+        policies = list(self._policies.values())
         for scope in ident.scopes or ["*"]:
-            for pol in self._policies.values():
-                if pol.applies_to(scope):
-                    ok, why = pol.conditions_met(source_ip=source_ip)
-                    if not ok:
-                        return {"valid": False, "reason": why,
-                                "identity_id": ident.identity_id}
+            allowed, why, pol_id = evaluate_conditional_access(policies, scope, c)
+            if not allowed:
+                return {"valid": False, "reason": why, "policy_id": pol_id,
+                        "identity_id": ident.identity_id}
         return {"valid": True, "identity_id": ident.identity_id,
                 "agent_name": ident.agent_name, "scopes": ident.scopes}
 

@@ -81,12 +81,15 @@ class Gateway:
     """Secure-by-default relay: quarantine gate -> detectors -> breaker ->
     upstream -> response inspection -> DLP redaction."""
 
-    def __init__(self) -> None:
+    def __init__(self, identity_store=None) -> None:
         self.upstreams: dict[str, Upstream] = {}
         self.pipeline = DetectorPipeline()
         self.identity_gate = IdentityGate()
         self.drift_gate = DriftGate()
         self.cost_gate = CostAnomalyGate()
+        # optional identity.lifecycle.AgentIdentityStore: supplies the ABAC
+        # conditional-access policies relay() evaluates per call
+        self.identity_store = identity_store
         self.metrics = {"relays_total": 0, "blocked_total": 0, "redactions_total": 0,
                         "breaker_rejections_total": 0}
 
@@ -99,7 +102,10 @@ class Gateway:
 
     def relay(self, upstream_name: str, frame: dict[str, Any],
               principal: Optional[str] = None,
-              cost: float = 1.0) -> dict[str, Any]:
+              cost: float = 1.0, access_ctx=None) -> dict[str, Any]:
+        """``access_ctx``: an identity.lifecycle.AccessContext — when present
+        together with ``self.identity_store`` policies, full ABAC
+        conditional-access (deny-wins, fail-closed unknowns) gates the call."""
         up = self.upstreams.get(upstream_name)
         msg_id = frame.get("id")
 
@@ -111,10 +117,23 @@ class Gateway:
         if up.quarantined:
             self.metrics["blocked_total"] += 1
             return err(-32002, f"upstream {upstream_name!r} is quarantined")
-        # conditional-access gates: revoked identity, catalog drift, cost
+        # conditional-access gates: revoked identity, ABAC policy, catalog
+        # drift, cost
         if principal is not None and not self.identity_gate.allow(principal):
             self.metrics["blocked_total"] += 1
             return err(-32005, f"identity {principal!r} is revoked")
+        if access_ctx is not None and self.identity_store is not None:
+            from agentbom_amd.identity.lifecycle import evaluate_conditional_access
+
+            if not access_ctx.tool_name and frame.get("method") == "tools/call":
+                access_ctx.tool_name = str(
+                    (frame.get("params") or {}).get("name") or "")
+            allowed, why, _pid = evaluate_conditional_access(
+                self.identity_store.list_conditional_policies(),
+                principal or "*", access_ctx)
+            if not allowed:
+                self.metrics["blocked_total"] += 1
+                return err(-32008, f"conditional access: {why}")
         if not self.drift_gate.allow(upstream_name):
             self.metrics["blocked_total"] += 1
             return err(-32006, f"upstream {upstream_name!r} tool catalog "

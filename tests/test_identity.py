@@ -245,3 +245,94 @@ class TestNhiOverlay:
         assert posture["total_identities"] == 3
         assert posture["by_provider"]["okta"] == 2
         assert posture["by_provider"]["agent-bom"] == 1
+
+
+class TestConditionalAccessABAC:
+    """Reference agent_identity_store.py ABAC semantics: deny-wins,
+    fail-closed unknowns, device/group/client/posture conditions."""
+
+    def _store_with(self, policy):
+        from agentbom_amd.identity.lifecycle import AgentIdentityStore
+
+        store = AgentIdentityStore()
+        ident, raw = store.issue("bot", scopes=["scan:read"])
+        store.put_conditional_policy(policy)
+        return store, raw
+
+    def test_deny_effect_wins(self):
+        from agentbom_amd.identity.lifecycle import AccessContext
+
+        store, raw = self._store_with(ConditionalAccessPolicy(
+            policy_id="d1", name="block-prod-writes", effect="deny",
+            allowed_environments=["prod"]))
+        # in prod: deny conditions hold -> blocked
+        denied = store.verify(raw, ctx=AccessContext(environment="prod"))
+        assert denied["valid"] is False and "block-prod-writes" in denied["reason"]
+        # in staging: deny conditions do not hold -> allowed
+        assert store.verify(raw, ctx=AccessContext(environment="staging"))["valid"]
+
+    def test_group_membership_fails_closed(self):
+        from agentbom_amd.identity.lifecycle import AccessContext
+
+        store, raw = self._store_with(ConditionalAccessPolicy(
+            policy_id="g1", name="sre-only", allowed_groups=["sre"]))
+        # no groups supplied -> fail closed
+        assert store.verify(raw)["valid"] is False
+        assert store.verify(raw, ctx=AccessContext(groups=["dev"]))["valid"] is False
+        assert store.verify(raw, ctx=AccessContext(groups=["dev", "sre"]))["valid"]
+
+    def test_device_posture_fails_closed_on_unknown(self):
+        from agentbom_amd.identity.lifecycle import AccessContext
+
+        store, raw = self._store_with(ConditionalAccessPolicy(
+            policy_id="p1", name="managed-only", require_device_managed=True,
+            require_device_disk_encrypted=True))
+        assert store.verify(raw)["valid"] is False  # unknown posture
+        partial = AccessContext(device_managed=True)  # encryption unknown
+        assert store.verify(raw, ctx=partial)["valid"] is False
+        good = AccessContext(device_managed=True, device_disk_encrypted=True)
+        assert store.verify(raw, ctx=good)["valid"]
+
+    def test_client_and_device_allowlists(self):
+        from agentbom_amd.identity.lifecycle import AccessContext
+
+        store, raw = self._store_with(ConditionalAccessPolicy(
+            policy_id="c1", name="claude-only", allowed_clients=["claude-code"],
+            allowed_devices=["wks-7"]))
+        ok = AccessContext(client_id="claude-code", device_id="wks-7")
+        assert store.verify(raw, ctx=ok)["valid"]
+        wrong_client = AccessContext(client_id="other", device_id="wks-7")
+        assert store.verify(raw, ctx=wrong_client)["valid"] is False
+
+    def test_weekday_and_hour_windows(self):
+        import datetime as dt
+
+        from agentbom_amd.identity.lifecycle import AccessContext
+
+        store, raw = self._store_with(ConditionalAccessPolicy(
+            policy_id="w1", name="business-hours",
+            allowed_weekdays=[0, 1, 2, 3, 4], allowed_hours_utc=list(range(8, 18))))
+        monday_noon = dt.datetime(2026, 9, 14, 12, 0,
+                                  tzinfo=dt.timezone.utc)  # Monday
+        sunday = dt.datetime(2026, 9, 13, 12, 0, tzinfo=dt.timezone.utc)
+        assert store.verify(raw, ctx=AccessContext(at=monday_noon))["valid"]
+        assert store.verify(raw, ctx=AccessContext(at=sunday))["valid"] is False
+
+    def test_disabled_and_priority(self):
+        from agentbom_amd.identity.lifecycle import (
+            AccessContext,
+            evaluate_conditional_access,
+        )
+
+        off = ConditionalAccessPolicy(policy_id="z", name="off", effect="deny",
+                                      status="disabled")
+        deny = ConditionalAccessPolicy(policy_id="a", name="deny-all",
+                                       effect="deny", priority=1)
+        req = ConditionalAccessPolicy(policy_id="b", name="needs-prod",
+                                      allowed_environments=["prod"], priority=2)
+        allowed, reason, pid = evaluate_conditional_access([off], "s",
+                                                           AccessContext())
+        assert allowed
+        allowed, reason, pid = evaluate_conditional_access(
+            [req, deny], "s", AccessContext(environment="prod"))
+        assert not allowed and pid == "a"  # deny precedence over satisfied require
