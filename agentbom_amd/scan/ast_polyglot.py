@@ -264,6 +264,30 @@ def analyze_source(text: str, path: str, lang: str) -> tuple[list[AstFinding], s
         name = next((g for g in dm.groups() if g), None)
         if name:
             calls.add(name)
+
+    # taint-lite structural pass (Go/JS): argument-level untrusted-identifier
+    # escalation + guard de-escalation over the SAME stripped view.  Upgrades
+    # severity of regex hits on the same line; adds sink calls the line
+    # patterns missed (method-style spellings).
+    if lang in ("go", "js"):
+        from agentbom_amd.scan.ast_taint import analyze_taint
+
+        seen = {(f.line, f.category) for f in findings}
+        by_line_cat = {(f.line, f.category): f for f in findings}
+        for tf in analyze_taint(code, path, lang):
+            key = (tf.line, tf.category)
+            if key in seen:
+                hit = by_line_cat[key]
+                if tf.untrusted_args and not tf.guarded:
+                    hit.severity = "critical"
+                    hit.tainted = True
+                continue
+            findings.append(AstFinding(
+                file=path, line=tf.line, call=tf.call, category=tf.category,
+                severity=tf.severity, cwe=tf.cwe, entrypoint=tf.caller,
+                snippet=(orig_lines[tf.line - 1].strip()[:160]
+                         if tf.line <= len(orig_lines) else ""),
+                tainted=bool(tf.untrusted_args)))
     return findings, calls
 
 

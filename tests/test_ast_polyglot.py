@@ -13,9 +13,11 @@ CASES = [
     ("js", "srv.ts", 'const { execSync } = require("child_process");\n'
                      'execSync(cmd)', "child_process.exec", "critical"),
     ("js", "dom.jsx", 'el.innerHTML = data', "innerHTML", "high"),
+    # taint-lite escalates these: args.../userInput are untrusted-shaped
     ("go", "main.go", 'out, _ := exec.Command(name, args...).Output()',
-     "exec.Command", "high"),
-    ("go", "tpl.go", 'return template.HTML(userInput)', "template.HTML", "high"),
+     "exec.Command", "critical"),
+    ("go", "tpl.go", 'return template.HTML(userInput)', "template.HTML",
+     "critical"),
     ("java", "App.java",
      'Process p = Runtime.getRuntime().exec(cmd);', "Runtime.exec", "critical"),
     ("java", "Ser.java", 'ObjectInputStream in = new ObjectInputStream(s);'
