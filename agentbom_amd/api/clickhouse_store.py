@@ -13,6 +13,7 @@ import json
 import time
 from typing import Any, Optional
 
+from agentbom_amd.utils import config as cfg
 from agentbom_amd.utils.http_client import check_offline, create_client, request_with_retry
 
 _SCHEMA = """
@@ -42,7 +43,7 @@ class ClickHouseAnalyticsStore:
         check_offline(url)
         self.url = url.rstrip("/")
         self.database = database
-        self.client = client or create_client(timeout=30.0)
+        self.client = client or create_client(timeout=cfg.CLICKHOUSE_TIMEOUT_S)
         self.headers: dict[str, str] = {}
         if username:
             self.headers["X-ClickHouse-User"] = username

@@ -17,6 +17,7 @@ import time
 from typing import Any, Optional
 from urllib.parse import urlencode
 
+from agentbom_amd.utils import config as cfg
 from agentbom_amd.utils.http_client import check_offline, create_client, request_with_retry
 
 _DDL = """
@@ -47,7 +48,7 @@ class SnowflakeStore:
         self.database = database
         self.schema = schema
         self.warehouse = warehouse
-        self.client = client or create_client(timeout=60.0)
+        self.client = client or create_client(timeout=cfg.SNOWFLAKE_TIMEOUT_S)
 
     def _exec(self, statement: str, bindings: Optional[dict] = None) -> dict:
         payload: dict[str, Any] = {"statement": statement, "timeout": 60}
@@ -155,11 +156,13 @@ def public_key_fingerprint(n: int, e: int) -> str:
 
 
 def keypair_jwt(account: str, user: str, n: int, e: int, d: int,
-                lifetime_s: int = 3600, now: Optional[int] = None) -> str:
+                lifetime_s: Optional[int] = None, now: Optional[int] = None) -> str:
     """Mint the key-pair bearer JWT for the SQL API
     (iss = ACCOUNT.USER.<fingerprint>, sub = ACCOUNT.USER)."""
     from agentbom_amd.api.oidc import rs256_sign
 
+    if lifetime_s is None:
+        lifetime_s = cfg.SNOWFLAKE_JWT_LIFETIME_S
     qualified = f"{account.upper()}.{user.upper()}"
     iat = int(now if now is not None else time.time())
     claims = {"iss": f"{qualified}.{public_key_fingerprint(n, e)}",

@@ -44,16 +44,17 @@ from agentbom_amd.utils import config as cfg
 from agentbom_amd.utils.canonical_ids import normalize_package_ecosystem
 from agentbom_amd.utils.http_client import check_offline, create_client, request_with_retry
 
-OSV_API_URL = "https://api.osv.dev/v1"
+# endpoints are config knobs so mirrors / air-gapped relays can be pointed
+# at without code changes (AGENT_BOM_OSV_API_URL etc.)
+OSV_API_URL = cfg.OSV_API_URL
 OSV_BATCH_URL = f"{OSV_API_URL}/querybatch"
-OSV_BULK_URL = "https://osv-vulnerabilities.storage.googleapis.com"
-GHSA_API_URL = "https://api.github.com/advisories"
-EPSS_API_URL = "https://api.first.org/data/v1/epss"
-KEV_URL = ("https://www.cisa.gov/sites/default/files/feeds/"
-           "known_exploited_vulnerabilities.json")
-NVD_API_URL = "https://services.nvd.nist.gov/rest/json/cves/2.0"
+OSV_BULK_URL = cfg.OSV_BULK_URL
+GHSA_API_URL = cfg.GHSA_API_URL
+EPSS_API_URL = cfg.EPSS_API_URL
+KEV_URL = cfg.KEV_URL
+NVD_API_URL = cfg.NVD_API_URL
 
-_BATCH_SIZE = 1000
+_BATCH_SIZE = cfg.OSV_BATCH_CHUNK
 
 # OSV ecosystem spellings for the bulk export / query API
 _OSV_ECOSYSTEM = {

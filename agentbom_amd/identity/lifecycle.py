@@ -26,6 +26,8 @@ from dataclasses import dataclass, field
 from datetime import datetime, timedelta, timezone
 from typing import Any, Optional
 
+from agentbom_amd.utils import config as cfg
+
 TOKEN_PREFIX = "abi"
 
 
@@ -366,9 +368,11 @@ class AgentIdentityStore:
 
     def issue(self, agent_name: str, scopes: Optional[list[str]] = None,
               allowed_tools: Optional[list[str]] = None,
-              ttl_hours: float = 24.0, actor: str = "api",
+              ttl_hours: Optional[float] = None, actor: str = "api",
               reason: str = "initial issue") -> tuple[AgentIdentity, str]:
         """Issue a new identity; returns (identity, raw_token) — raw shown once."""
+        if ttl_hours is None:
+            ttl_hours = cfg.IDENTITY_TOKEN_TTL_HOURS
         with self._lock:
             raw, prefix, thash = generate_token()
             now = _now()
@@ -386,10 +390,15 @@ class AgentIdentityStore:
             self._audit("identity.issue", actor, reason, ident.identity_id)
             return ident, raw
 
-    def rotate(self, identity_id: str, overlap_minutes: float = 15.0,
-               ttl_hours: float = 24.0, actor: str = "api",
+    def rotate(self, identity_id: str,
+               overlap_minutes: Optional[float] = None,
+               ttl_hours: Optional[float] = None, actor: str = "api",
                reason: str = "rotation") -> tuple[Optional[AgentIdentity], Optional[str]]:
         """Rotate: new token, old identity stays live through the overlap window."""
+        if overlap_minutes is None:
+            overlap_minutes = cfg.IDENTITY_ROTATE_OVERLAP_MIN
+        if ttl_hours is None:
+            ttl_hours = cfg.IDENTITY_TOKEN_TTL_HOURS
         with self._lock:
             old = self.get(identity_id)
             if old is None or not old.is_live():
@@ -496,7 +505,9 @@ class AgentIdentityStore:
     # ── JIT grants ────────────────────────────────────────────────────────
 
     def grant_jit(self, identity_id: str, scopes: list[str], reason: str,
-                  granted_by: str, ttl_minutes: float = 60.0) -> Optional[AgentJITGrant]:
+                  granted_by: str, ttl_minutes: Optional[float] = None) -> Optional[AgentJITGrant]:
+        if ttl_minutes is None:
+            ttl_minutes = cfg.JIT_GRANT_TTL_MINUTES
         with self._lock:
             if self.get(identity_id) is None:
                 return None

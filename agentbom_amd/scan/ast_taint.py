@@ -290,8 +290,10 @@ _ASSIGN = re.compile(
 def _tainted_locals(body: str, seeds: set[str]) -> set[str]:
     """Local names assigned from untrusted expressions (2 propagation
     rounds — enough for the assign-then-use chains a lexer pass can see)."""
+    from agentbom_amd.utils import config as cfg
+
     tainted = set(seeds)
-    for _ in range(2):
+    for _ in range(max(1, cfg.TAINT_PROPAGATION_ROUNDS)):
         grew = False
         for line in body.splitlines():
             m = _ASSIGN.match(line)

@@ -46,9 +46,9 @@ def _py_imports(path: Path, limit: int = 64) -> list[str]:
     return sorted(mods)[:limit]
 
 
-def collect_project_inventory(root: str, *, max_depth: int = 6,
-                              max_files: int = 4000,
-                              max_entries_per_dir: int = 400,
+def collect_project_inventory(root: str, *, max_depth: Optional[int] = None,
+                              max_files: Optional[int] = None,
+                              max_entries_per_dir: Optional[int] = None,
                               parse_imports: bool = True) -> Optional[dict[str, Any]]:
     """Walk ``root`` → {"directories": [...], "files": [...], "truncated"}.
 
@@ -56,6 +56,12 @@ def collect_project_inventory(root: str, *, max_depth: int = 6,
     manifest names.  File records: path, kind (source|config|manifest),
     language for sources, and ``imports`` for Python files.
     """
+    from agentbom_amd.utils import config as cfg
+
+    max_depth = cfg.REPO_INVENTORY_MAX_DEPTH if max_depth is None else max_depth
+    max_files = cfg.REPO_INVENTORY_MAX_FILES if max_files is None else max_files
+    max_entries_per_dir = (cfg.REPO_INVENTORY_MAX_DIR_ENTRIES
+                           if max_entries_per_dir is None else max_entries_per_dir)
     base = Path(root).expanduser()
     if not base.is_dir():
         return None

@@ -195,3 +195,33 @@ LICENSE_DENYLIST = _str("AGENT_BOM_LICENSE_DENYLIST", "GPL-3.0,AGPL-3.0,SSPL-1.0
 # ── Multi-GPU / distributed engine ──────────────────────────────────────────
 DIST_EXCHANGE_CAP = _int("AGENT_BOM_DIST_EXCHANGE_CAP", 0)  # 0 = auto
 DIST_MAX_BFS_LEVELS = _int("AGENT_BOM_DIST_MAX_BFS_LEVELS", 64)
+
+# ── Live feed endpoints (override for mirrors / air-gapped relays) ─────────
+OSV_API_URL = _str("AGENT_BOM_OSV_API_URL", "https://api.osv.dev/v1")
+OSV_BULK_URL = _str("AGENT_BOM_OSV_BULK_URL",
+                    "https://osv-vulnerabilities.storage.googleapis.com")
+GHSA_API_URL = _str("AGENT_BOM_GHSA_API_URL", "https://api.github.com/advisories")
+EPSS_API_URL = _str("AGENT_BOM_EPSS_API_URL", "https://api.first.org/data/v1/epss")
+KEV_URL = _str("AGENT_BOM_KEV_URL",
+               "https://www.cisa.gov/sites/default/files/feeds/"
+               "known_exploited_vulnerabilities.json")
+NVD_API_URL = _str("AGENT_BOM_NVD_API_URL",
+                   "https://services.nvd.nist.gov/rest/json/cves/2.0")
+OSV_BATCH_CHUNK = _int("AGENT_BOM_OSV_BATCH_CHUNK", 1000)
+
+# ── Identity lifecycle ──────────────────────────────────────────────────────
+IDENTITY_TOKEN_TTL_HOURS = _float("AGENT_BOM_IDENTITY_TOKEN_TTL_HOURS", 24.0)
+IDENTITY_ROTATE_OVERLAP_MIN = _float("AGENT_BOM_IDENTITY_ROTATE_OVERLAP_MIN", 15.0)
+JIT_GRANT_TTL_MINUTES = _float("AGENT_BOM_JIT_GRANT_TTL_MINUTES", 60.0)
+
+# ── Repo inventory / CI scanning bounds ────────────────────────────────────
+REPO_INVENTORY_MAX_FILES = _int("AGENT_BOM_REPO_INVENTORY_MAX_FILES", 4000)
+REPO_INVENTORY_MAX_DEPTH = _int("AGENT_BOM_REPO_INVENTORY_MAX_DEPTH", 6)
+REPO_INVENTORY_MAX_DIR_ENTRIES = _int(
+    "AGENT_BOM_REPO_INVENTORY_MAX_DIR_ENTRIES", 400)
+TAINT_PROPAGATION_ROUNDS = _int("AGENT_BOM_TAINT_PROPAGATION_ROUNDS", 2)
+
+# ── Optional analytics stores ───────────────────────────────────────────────
+CLICKHOUSE_TIMEOUT_S = _float("AGENT_BOM_CLICKHOUSE_TIMEOUT_S", 30.0)
+SNOWFLAKE_TIMEOUT_S = _float("AGENT_BOM_SNOWFLAKE_TIMEOUT_S", 60.0)
+SNOWFLAKE_JWT_LIFETIME_S = _int("AGENT_BOM_SNOWFLAKE_JWT_LIFETIME_S", 3600)
