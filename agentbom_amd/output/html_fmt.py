@@ -35,6 +35,11 @@ details { margin: 0.5rem 0; background: #161b22; border: 1px solid #30363d;
           border-radius: 8px; padding: 0.5rem 1rem; }
 summary { cursor: pointer; }
 code { background: #21262d; padding: 0.1rem 0.3rem; border-radius: 4px; }
+.posture { background: #161b22; border-radius: 8px; padding: 0.8rem 1.2rem;
+           margin: 1rem 0; display: flex; align-items: center; gap: 1rem; }
+.posture .grade { font-size: 2.2rem; font-weight: 800; }
+.posture ul { margin: 0; font-size: 0.85rem; color: #8b949e; }
+svg text { fill: #e6edf3; }
 """
 
 
@@ -89,6 +94,8 @@ def to_html(report: AIBOMReport) -> str:
 <div class="card med"><div class="num">{counts['medium']}</div>medium</div>
 <div class="card ok"><div class="num">{counts['low']}</div>low</div>
 </div>
+{_posture_html(report)}
+{_severity_chart_svg(counts)}
 <h2>Findings</h2>
 <table><tr><th>risk</th><th>severity</th><th>vulnerability</th><th>package</th>
 <th>reach</th><th>agents</th><th>creds</th><th>tools</th><th>fix</th></tr>
@@ -99,9 +106,77 @@ def to_html(report: AIBOMReport) -> str:
 {_remediation_html(report)}
 {_frameworks_html(report)}
 {_exposure_paths_html(report)}
+{_inventory_tree_html(report)}
 {_other_findings_html(report)}
 {_warnings_html(report)}
 </body></html>"""
+
+
+def _posture_html(report: AIBOMReport) -> str:
+    """Estate grade banner (scan/risk.estate_exec_score)."""
+    from agentbom_amd.scan.risk import estate_exec_score
+
+    score = estate_exec_score(report)
+    grade = str(score.get("grade", "?"))
+    color = {"A": "#2e7d32", "B": "#558b2f", "C": "#f9a825",
+             "D": "#e65100", "F": "#c62828"}.get(grade[:1], "#666")
+    drivers = "".join(f"<li>{_e(d)}</li>" for d in (score.get("drivers") or [])[:5])
+    return (f'<div class="posture" style="border-left:6px solid {color}">'
+            f'<span class="grade" style="color:{color}">{_e(grade)}</span>'
+            f'<span>estate score {score.get("score", "?")}/100</span>'
+            f"<ul>{drivers}</ul></div>")
+
+
+def _severity_chart_svg(counts: dict) -> str:
+    """Inline SVG horizontal severity bars — no external assets."""
+    sev_colors = [("critical", "#c62828"), ("high", "#e65100"),
+                  ("medium", "#f9a825"), ("low", "#2e7d32")]
+    total = max(sum(counts.get(s, 0) for s, _ in sev_colors), 1)
+    bars = []
+    for i, (sev, color) in enumerate(sev_colors):
+        n = counts.get(sev, 0)
+        w = max(round(360 * n / total), 2 if n else 0)
+        y = 8 + i * 26
+        bars.append(
+            f'<text x="0" y="{y + 13}" font-size="12">{sev}</text>'
+            f'<rect x="70" y="{y}" width="{w}" height="18" fill="{color}"/>'
+            f'<text x="{76 + w}" y="{y + 13}" font-size="12">{n}</text>')
+    return (f'<svg width="480" height="118" role="img" '
+            f'aria-label="severity distribution">{"".join(bars)}</svg>')
+
+
+def _inventory_tree_html(report: AIBOMReport) -> str:
+    """Nested inventory: agent → server → creds/tools/vulnerable pkgs."""
+    if not report.agents:
+        return ""
+    parts = ["<h2>Estate inventory</h2>"]
+    for agent in report.agents[:200]:
+        servers = []
+        for srv in agent.mcp_servers:
+            vulns = srv.total_vulnerabilities
+            badge = (f' <span class="badge kev">{vulns} vulns</span>'
+                     if vulns else "")
+            inner = []
+            if srv.credential_names:
+                inner.append("<p><b>credentials:</b> "
+                             + _e(", ".join(srv.credential_names[:8])) + "</p>")
+            if srv.tools:
+                inner.append("<p><b>tools:</b> "
+                             + _e(", ".join(t.name for t in srv.tools[:8]))
+                             + "</p>")
+            vp = srv.vulnerable_packages
+            if vp:
+                inner.append("<p><b>vulnerable:</b> " + _e(", ".join(
+                    f"{p.name}@{p.version}" for p in vp[:10])) + "</p>")
+            servers.append(
+                f"<details><summary><code>{_e(srv.name)}</code> "
+                f"({_e(srv.transport.value)}, {len(srv.packages)} pkgs)"
+                f"{badge}</summary>{''.join(inner)}</details>")
+        parts.append(
+            f"<details><summary><b>{_e(agent.name)}</b> "
+            f"({_e(agent.agent_type.value)}, {len(agent.mcp_servers)} "
+            f"servers)</summary>{''.join(servers)}</details>")
+    return "".join(parts)
 
 
 def _remediation_html(report: AIBOMReport) -> str:

@@ -94,3 +94,14 @@ def test_render_report_verbose_includes_tree(report):
     assert "Estate inventory" in text       # agent tree
     assert "Severity distribution" in text  # chart
     assert "security posture" in text       # posture panel
+
+
+def test_html_report_sections(report):
+    from agentbom_amd.output.html_fmt import to_html
+
+    h = to_html(report)
+    assert "<svg" in h and 'aria-label="severity distribution"' in h
+    assert "estate score" in h and 'class="grade"' in h
+    assert "Estate inventory" in h
+    for agent in report.agents[:2]:
+        assert agent.name in h
