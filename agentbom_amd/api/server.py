@@ -433,6 +433,48 @@ def create_app() -> FastAPI:
             "neighbors": [g.nodes[n].to_dict() for n in g.neighbors(node_id, direction)],
         }
 
+    @app.get("/v1/graph/node/{node_id:path}/impact", dependencies=[Depends(auth)])
+    def graph_node_impact(request: Request, node_id: str, max_hops: int = 4) -> dict:
+        """Bounded downstream impact of one node (container.impact_of)."""
+        g = _latest_graph(request)
+        if node_id not in g.nodes:
+            raise HTTPException(status_code=404, detail="node not found")
+        return g.impact_of(node_id, max_hops=min(max_hops, 8))
+
+    @app.get("/v1/graph/node/{node_id:path}", dependencies=[Depends(auth)])
+    def graph_node(request: Request, node_id: str) -> dict:
+        g = _latest_graph(request)
+        if node_id not in g.nodes:
+            raise HTTPException(status_code=404, detail="node not found")
+        edges = [e.to_dict() for e in g.edges
+                 if e.source == node_id or e.target == node_id][:200]
+        return {"node": g.nodes[node_id].to_dict(), "edges": edges}
+
+    @app.get("/v1/graph/centrality", dependencies=[Depends(auth)])
+    def graph_centrality(request: Request, top_n: int = 20) -> dict:
+        g = _latest_graph(request)
+        return {"centrality": [{"id": nid, "degree": deg}
+                               for nid, deg in g.degree_centrality(top_n)]}
+
+    @app.get("/v1/graph/bottlenecks", dependencies=[Depends(auth)])
+    def graph_bottlenecks(request: Request, top_n: int = 10,
+                          sample: int = 64) -> dict:
+        """Approx-betweenness choke points (lateral-movement bottlenecks)."""
+        g = _latest_graph(request)
+        rows = g.bottlenecks(top_n=top_n, sample=min(sample, 256))
+        return {"bottlenecks": [{"id": nid, "score": round(score, 4)}
+                                for nid, score in rows]}
+
+    @app.get("/v1/graph/view/{view_name}", dependencies=[Depends(auth)])
+    def graph_view(request: Request, view_name: str) -> dict:
+        """Typed subgraph views: inventory / attack-path / lateral /
+        compliance / runtime."""
+        g = _latest_graph(request)
+        try:
+            return g.view(view_name)
+        except (KeyError, ValueError) as exc:
+            raise HTTPException(status_code=404, detail=str(exc))
+
     @app.get("/v1/graph/paths", dependencies=[Depends(auth)])
     def graph_paths(request: Request, source: Optional[str] = None, target: Optional[str] = None,
                     limit: int = 25) -> dict:

@@ -58,6 +58,22 @@ class TestApi:
         assert q["nodes"]
         imp = client.get("/v1/graph/impact/pkg:pypi:pyyaml@5.3").json()
         assert imp["total_impacted"] > 0
+
+    def test_graph_analytics_endpoints(self, client):
+        node = client.get("/v1/graph/node/agent:cursor").json()
+        assert node["node"]["id"] == "agent:cursor" and node["edges"]
+        imp = client.get("/v1/graph/node/agent:cursor/impact?max_hops=3").json()
+        assert imp["total_impacted"] > 0
+        cent = client.get("/v1/graph/centrality?top_n=5").json()
+        assert len(cent["centrality"]) == 5
+        assert cent["centrality"][0]["degree"] >= cent["centrality"][-1]["degree"]
+        bn = client.get("/v1/graph/bottlenecks?top_n=3").json()
+        assert bn["bottlenecks"] and all("score" in b for b in bn["bottlenecks"])
+        for view in ("inventory", "attack-path", "lateral"):
+            v = client.get(f"/v1/graph/view/{view}").json()
+            assert v["view"] == view
+        assert client.get("/v1/graph/view/bogus").status_code == 404
+        assert client.get("/v1/graph/node/missing:x").status_code == 404
         r = client.get("/v1/graph/rollup").json()
         assert r["containers"]
         d = client.get("/v1/graph/should-i-deploy").json()
