@@ -151,18 +151,26 @@ def expand_dedup_matches(torch, layout: dict, su, sw):
 
 
 
+_RANK_WS: dict = {}
+
+
 def rank_order(torch, scores):
     """Deterministic finding rank: score desc (0.01 quantization), index asc.
 
     Scores are clipped to [0, 10] and serialized rounded, so 0.01 buckets
-    preserve every visible distinction; packing (bucket, index) into ONE
-    integer key turns the rank into a single int sort instead of a
-    float+index argsort (round-1 roadmap lever #2).  Same formula on CPU
-    and GPU -> bit-equal order.
+    preserve every visible distinction.  On GPU the rank is the stable
+    1024-bin counting-sort kernel (ops/csrc/rank.hip — four small launches,
+    no rocprim); the CPU path packs (bucket,
+    index) into one integer key and sorts.  Both produce the identical
+    (bucket asc, index asc) order -> bit-equal CPU/GPU.
     """
     n = scores.numel()
     if n == 0:
         return torch.empty(0, dtype=torch.int64, device=scores.device)
+    if scores.is_cuda:
+        from agentbom_amd.ops import native
+
+        return native.rank_bucket_order(scores.contiguous(), _RANK_WS)
     q = (scores * 100).round().clamp(0, 1000).to(torch.int64)
     bits = max(1, (n - 1).bit_length())
     key = ((1000 - q) << bits) | torch.arange(n, device=scores.device)

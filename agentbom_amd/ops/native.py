@@ -42,6 +42,7 @@ _SIGNATURES = {
     "abom_severity_histogram": ([_c, _c, _c, _i64, _c], _i32),
     "abom_score_gather": ([_c] * 11 + [_c] * 4 + [_i64, ctypes.POINTER(ctypes.c_float), _c], _i32),
     "abom_path_relax": ([_c] * 11 + [_i64, _c], _i32),
+    "abom_rank_order": ([_c, _i64, _c, _c, _i32, _c, _c], _i32),
 }
 
 
@@ -489,3 +490,36 @@ def path_relax(edge_src, col, etype, edge_weight, cur, nxt, node_boost,
         etype.numel(), _stream(),
     )
     _check(rc, "abom_path_relax")
+
+
+_RANK_BINS = 1024
+_RANK_CHUNK = 256
+
+
+def rank_bucket_order(scores, workspace: Optional[dict] = None):
+    """Deterministic finding rank via the stable counting-sort kernel.
+
+    Bit-equal to gpu_engine.rank_order's quantized key sort (bucket asc,
+    index asc) — four small launches, no rocprim (ops/csrc/rank.hip)."""
+    import torch
+
+    lib = load()
+    n = scores.numel()
+    dev = scores.device
+    if n == 0:
+        return torch.empty(0, dtype=torch.int64, device=dev)
+    nblocks = (n + _RANK_CHUNK - 1) // _RANK_CHUNK
+    ws = workspace if workspace is not None else {}
+    hist = ws.get("rank_hist")
+    if hist is None or hist.numel() < nblocks * _RANK_BINS:
+        hist = torch.empty(nblocks * _RANK_BINS, dtype=torch.int32, device=dev)
+        ws["rank_hist"] = hist
+    scratch = ws.get("rank_scratch")
+    if scratch is None:
+        scratch = torch.empty(2 * _RANK_BINS, dtype=torch.int32, device=dev)
+        ws["rank_scratch"] = scratch
+    order = torch.empty(n, dtype=torch.int64, device=dev)
+    rc = lib.abom_rank_order(_ptr(scores), n, _ptr(hist), _ptr(scratch),
+                             nblocks, _ptr(order), _stream())
+    _check(rc, "abom_rank_order")
+    return order

@@ -554,3 +554,32 @@ def test_rollup_gpu_matches_cpu(estate, dev):
                 "server_findings", "agent_findings"):
         assert torch.equal(roll_g[key].cpu().to(torch.int64),
                            roll_c[key].to(torch.int64)), key
+
+
+def test_rank_kernel_matches_torch_key_sort(dev):
+    """Counting-sort rank kernel == the quantized key sort, incl. ties,
+    clamping, odd sizes, and all-equal scores."""
+    from agentbom_amd.graph.gpu_engine import rank_order
+    from agentbom_amd.ops import native
+
+    def oracle(scores_cpu):
+        n = scores_cpu.numel()
+        q = (scores_cpu * 100).round().clamp(0, 1000).to(torch.int64)
+        bits = max(1, (n - 1).bit_length())
+        key = ((1000 - q) << bits) | torch.arange(n)
+        skey, _ = torch.sort(key)
+        return skey & ((1 << bits) - 1)
+
+    gen = torch.Generator().manual_seed(7)
+    for n in (1, 63, 64, 257, 4096, 100_001, 500_000):
+        scores = torch.rand(n, generator=gen) * 12.0 - 1.0  # exercises clamp
+        scores[::7] = 9.87  # heavy tie bucket
+        got = native.rank_bucket_order(scores.to(dev).contiguous())
+        assert torch.equal(got.cpu(), oracle(scores)), f"n={n}"
+    # all-equal: rank must be identity (index asc within the single bucket)
+    eq = torch.full((1000,), 5.0, device=dev)
+    assert torch.equal(native.rank_bucket_order(eq).cpu(),
+                       torch.arange(1000))
+    # engine-level: rank_order on device routes through the kernel
+    s = torch.rand(2048, generator=gen).to(dev)
+    assert torch.equal(rank_order(torch, s).cpu(), oracle(s.cpu()))
