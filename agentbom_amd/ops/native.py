@@ -29,7 +29,7 @@ _SIGNATURES = {
     "abom_device_count": ([], _i32),
     "abom_synchronize": ([_c], _i32),
     "abom_error_string": ([_i32], ctypes.c_char_p),
-    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c] + [_c, _c] + [_c, _c, _i64, _c], _i32),
+    "abom_match": ([_c] * 4 + [_i64] + [_c, _c, _i64] + [_c] * 7 + [_c] + [_c, _c] + [_c] + [_c, _c, _i64, _c], _i32),
     "abom_bfs_init": ([_c, _i64, _c], _i32),
     "abom_bfs_seed": ([_c, _i64, _c, _c, _c, _c, _c], _i32),
     "abom_bfs_expand": ([_c, _c, _c, _u32, _c, _i64, _c, _u32, _c, _c, _c, _c, _c, _i64, _c], _i32),
@@ -116,7 +116,7 @@ def _stream() -> ctypes.c_void_p:
 
 def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
                  group_off, windows: dict, capacity: Optional[int] = None,
-                 pkg_win_range=None) -> dict:
+                 pkg_win_range=None, order=None) -> dict:
     """Launch the match kernel on the CURRENT stream without syncing.
 
     Returns a pending handle for :func:`match_finalize`.  Used by the engine
@@ -138,6 +138,7 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
     wbeg_p = wend_p = None
     if pkg_win_range is not None:
         wbeg_p, wend_p = _ptr(pkg_win_range[0]), _ptr(pkg_win_range[1])
+    order_p = _ptr(order) if order is not None else None
     rc = lib.abom_match(
         _ptr(pkg_group_key), _ptr(pkg_key_hi), _ptr(pkg_key_lo), _ptr(pkg_flags), P,
         _ptr(group_keys), _ptr(group_off), G,
@@ -146,14 +147,14 @@ def match_launch(pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags, group_keys,
         _ptr(windows["last_hi"]), _ptr(windows["last_lo"]),
         _ptr(windows["flags"]),
         _ptr(windows["packed"]) if "packed" in windows else None,
-        wbeg_p, wend_p,
+        wbeg_p, wend_p, order_p,
         _ptr(out_pairs), _ptr(out_count), cap, _stream(),
     )
     _check(rc, "abom_match")
     return {"out_pairs": out_pairs, "out_count": out_count, "cap": cap,
             "args": (pkg_group_key, pkg_key_hi, pkg_key_lo, pkg_flags,
                      group_keys, group_off, windows),
-            "kw": {"pkg_win_range": pkg_win_range}}
+            "kw": {"pkg_win_range": pkg_win_range, "order": order}}
 
 
 def match_finalize(pending: dict, sort: bool = True):

@@ -167,7 +167,8 @@ class DistEstateEngine:
                     pending = native.match_launch(
                         dd["u_gk"], dd["u_hi"], dd["u_lo"], dd["u_flags"],
                         self.arena["group_keys"], self.arena["group_off"],
-                        self.arena["windows"], pkg_win_range=dd["u_ranges"])
+                        self.arena["windows"], pkg_win_range=dd["u_ranges"],
+                        order=dd["heavy_order"])
                 else:
                     pending = native.match_launch(
                         self.pkg_group_key_sorted, self.pkg_key_hi_sorted,
