@@ -91,7 +91,7 @@ def create_app() -> FastAPI:
         key_roles[api_key] = "admin"  # legacy single-key = admin
 
     _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
-                       "/scim", "/v1/delegation-tokens")
+                       "/scim", "/v1/delegation-tokens", "/v1/costs/budget")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -116,6 +116,9 @@ def create_app() -> FastAPI:
     state.delegation = DelegationTokens()
     state.quotas = QuotaTracker()
     state.scim_users = ScimUserStore()
+    from agentbom_amd.api.cost_store import SQLiteCostStore
+
+    state.costs = SQLiteCostStore()
     oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS_URL"))
@@ -634,6 +637,69 @@ def create_app() -> FastAPI:
     def delete_schedule(schedule_id: str) -> None:
         if state.scheduler is None or not state.scheduler.remove(schedule_id):
             raise HTTPException(status_code=404, detail="schedule not found")
+
+    @app.post("/v1/costs/records", status_code=201, dependencies=[Depends(auth)])
+    def ingest_costs(request: Request, payload: dict) -> dict:
+        """Batch LLM cost ingest (OTel GenAI span shape)."""
+        from agentbom_amd.api.cost_store import LLMCostRecord
+
+        rows = payload.get("records", [])
+        if not isinstance(rows, list) or not rows:
+            raise HTTPException(status_code=400, detail="records must be a non-empty list")
+        tenant = _tenant_of(request)
+        recs = []
+        for row in rows[:10_000]:
+            if not isinstance(row, dict) or "agent" not in row:
+                raise HTTPException(status_code=400,
+                                    detail="each record needs at least agent+cost_usd")
+            recs.append(LLMCostRecord(
+                tenant_id=tenant, agent=str(row["agent"]),
+                cost_usd=float(row.get("cost_usd", 0.0)),
+                model=str(row.get("model", "")),
+                tokens_in=int(row.get("tokens_in", 0)),
+                tokens_out=int(row.get("tokens_out", 0)),
+                observed_at=str(row.get("observed_at", "")),
+                cost_center=str(row.get("cost_center", "")),
+                tags={str(k): str(v) for k, v in
+                      (row.get("tags") or {}).items()}))
+        return {"ingested": state.costs.add_records(recs)}
+
+    @app.get("/v1/costs/summary", dependencies=[Depends(auth)])
+    def cost_summary(request: Request, agent: Optional[str] = None,
+                     since: Optional[str] = None) -> dict:
+        from agentbom_amd.api.cost_store import budget_status, summarize
+
+        tenant = _tenant_of(request)
+        records = state.costs.list_records(tenant, agent=agent, since=since)
+        out = summarize(records)
+        budget = state.costs.get_budget(tenant, agent=agent or "")
+        out["budget"] = budget_status(
+            sum(r.cost_usd for r in records), budget)
+        return out
+
+    @app.get("/v1/costs/forecast", dependencies=[Depends(auth)])
+    def cost_forecast(request: Request, agent: Optional[str] = None) -> dict:
+        from agentbom_amd.api.cost_store import forecast_spend
+
+        tenant = _tenant_of(request)
+        records = state.costs.list_records(tenant, agent=agent)
+        budget = state.costs.get_budget(tenant, agent=agent or "") \
+            or state.costs.get_budget(tenant)
+        return forecast_spend(records, budget)
+
+    @app.put("/v1/costs/budget", dependencies=[Depends(auth)])
+    def set_cost_budget(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.cost_store import CostBudget
+
+        if "limit_usd" not in payload:
+            raise HTTPException(status_code=400, detail="limit_usd required")
+        budget = CostBudget(
+            tenant_id=_tenant_of(request),
+            limit_usd=float(payload["limit_usd"]),
+            agent=str(payload.get("agent", "")),
+            cost_center=str(payload.get("cost_center", "")))
+        state.costs.set_budget(budget)
+        return budget.to_dict()
 
     @app.post("/v1/proxy/audit", dependencies=[Depends(auth)])
     def ingest_audit(payload: dict) -> dict:
