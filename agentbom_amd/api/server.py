@@ -91,7 +91,8 @@ def create_app() -> FastAPI:
         key_roles[api_key] = "admin"  # legacy single-key = admin
 
     _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
-                       "/scim", "/v1/delegation-tokens", "/v1/costs/budget")
+                       "/scim", "/v1/delegation-tokens", "/v1/costs/budget",
+                       "/v1/exceptions")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -117,8 +118,10 @@ def create_app() -> FastAPI:
     state.quotas = QuotaTracker()
     state.scim_users = ScimUserStore()
     from agentbom_amd.api.cost_store import SQLiteCostStore
+    from agentbom_amd.api.exceptions_store import ExceptionStore
 
     state.costs = SQLiteCostStore()
+    state.exceptions = ExceptionStore()
     oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS_URL"))
@@ -637,6 +640,70 @@ def create_app() -> FastAPI:
     def delete_schedule(schedule_id: str) -> None:
         if state.scheduler is None or not state.scheduler.remove(schedule_id):
             raise HTTPException(status_code=404, detail="schedule not found")
+
+    @app.post("/v1/exceptions", status_code=201, dependencies=[Depends(auth)])
+    def request_exception(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.exceptions_store import VulnException
+
+        for key in ("vuln_id", "package_name", "reason"):
+            if not payload.get(key):
+                raise HTTPException(status_code=400, detail=f"{key} required")
+        exc = VulnException(
+            vuln_id=str(payload["vuln_id"]),
+            package_name=str(payload["package_name"]),
+            server_name=str(payload.get("server_name", "")),
+            reason=str(payload["reason"]),
+            requested_by=getattr(request.state, "principal", "anonymous"),
+            expires_at=str(payload.get("expires_at", "")),
+            tenant_id=_tenant_of(request))
+        return state.exceptions.request(exc).to_dict()
+
+    @app.get("/v1/exceptions", dependencies=[Depends(auth)])
+    def list_exceptions(request: Request, status: Optional[str] = None) -> dict:
+        rows = state.exceptions.list(_tenant_of(request), status=status)
+        return {"total": len(rows), "exceptions": [e.to_dict() for e in rows]}
+
+    @app.post("/v1/exceptions/{exception_id}/approve", dependencies=[Depends(auth)])
+    def approve_exception(request: Request, exception_id: str,
+                          payload: Optional[dict] = None) -> dict:
+        payload = payload or {}
+        exc = state.exceptions.approve(
+            exception_id, actor=getattr(request.state, "principal", "?"),
+            reason=str(payload.get("reason", "")),
+            tenant_id=_tenant_of(request),
+            ttl_days=float(payload.get("ttl_days", 90.0)))
+        if exc is None:
+            raise HTTPException(status_code=409,
+                                detail="unknown exception or invalid transition")
+        return exc.to_dict()
+
+    @app.post("/v1/exceptions/{exception_id}/reject", dependencies=[Depends(auth)])
+    def reject_exception(request: Request, exception_id: str,
+                         payload: Optional[dict] = None) -> dict:
+        exc = state.exceptions.reject(
+            exception_id, actor=getattr(request.state, "principal", "?"),
+            reason=str((payload or {}).get("reason", "")),
+            tenant_id=_tenant_of(request))
+        if exc is None:
+            raise HTTPException(status_code=409,
+                                detail="unknown exception or invalid transition")
+        return exc.to_dict()
+
+    @app.post("/v1/exceptions/{exception_id}/revoke", dependencies=[Depends(auth)])
+    def revoke_exception(request: Request, exception_id: str,
+                         payload: Optional[dict] = None) -> dict:
+        exc = state.exceptions.revoke(
+            exception_id, actor=getattr(request.state, "principal", "?"),
+            reason=str((payload or {}).get("reason", "")),
+            tenant_id=_tenant_of(request))
+        if exc is None:
+            raise HTTPException(status_code=409,
+                                detail="unknown exception or invalid transition")
+        return exc.to_dict()
+
+    @app.get("/v1/exceptions/audit", dependencies=[Depends(auth)])
+    def exceptions_audit() -> dict:
+        return {"chain_valid": state.exceptions.audit_chain_valid()}
 
     @app.post("/v1/costs/records", status_code=201, dependencies=[Depends(auth)])
     def ingest_costs(request: Request, payload: dict) -> dict:
