@@ -92,7 +92,7 @@ def create_app() -> FastAPI:
 
     _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
                        "/scim", "/v1/delegation-tokens", "/v1/costs/budget",
-                       "/v1/exceptions")
+                       "/v1/exceptions", "/v1/blueprints", "/v1/drift-incidents")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -120,8 +120,11 @@ def create_app() -> FastAPI:
     from agentbom_amd.api.cost_store import SQLiteCostStore
     from agentbom_amd.api.exceptions_store import ExceptionStore
 
+    from agentbom_amd.api.blueprints import BlueprintStore
+
     state.costs = SQLiteCostStore()
     state.exceptions = ExceptionStore()
+    state.blueprints = BlueprintStore()
     oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS_URL"))
@@ -640,6 +643,116 @@ def create_app() -> FastAPI:
     def delete_schedule(schedule_id: str) -> None:
         if state.scheduler is None or not state.scheduler.remove(schedule_id):
             raise HTTPException(status_code=404, detail="schedule not found")
+
+    @app.post("/v1/blueprints", status_code=201, dependencies=[Depends(auth)])
+    def create_blueprint(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.blueprints import BlueprintComposition
+
+        if not payload.get("name"):
+            raise HTTPException(status_code=400, detail="name required")
+        bp = state.blueprints.create(
+            _tenant_of(request), str(payload["name"]),
+            BlueprintComposition.from_dict(payload.get("composition") or {}),
+            author=getattr(request.state, "principal", "?"),
+            owner=str(payload.get("owner", "")),
+            seeded_from=str(payload.get("seeded_from", "")))
+        return bp.to_dict()
+
+    @app.get("/v1/blueprints", dependencies=[Depends(auth)])
+    def list_blueprints(request: Request) -> dict:
+        rows = state.blueprints.list(_tenant_of(request))
+        return {"total": len(rows), "blueprints": [b.to_dict() for b in rows]}
+
+    @app.get("/v1/blueprints/{blueprint_id}/versions", dependencies=[Depends(auth)])
+    def blueprint_versions(request: Request, blueprint_id: str) -> dict:
+        if state.blueprints.get(_tenant_of(request), blueprint_id) is None:
+            raise HTTPException(status_code=404, detail="blueprint not found")
+        return {"versions": [v.to_dict() for v in
+                             state.blueprints.list_versions(blueprint_id)]}
+
+    @app.post("/v1/blueprints/{blueprint_id}/versions", status_code=201,
+              dependencies=[Depends(auth)])
+    def blueprint_new_draft(request: Request, blueprint_id: str,
+                            payload: dict) -> dict:
+        from agentbom_amd.api.blueprints import BlueprintComposition
+
+        v = state.blueprints.create_draft(
+            _tenant_of(request), blueprint_id,
+            BlueprintComposition.from_dict(payload.get("composition") or {}),
+            author=getattr(request.state, "principal", "?"))
+        if v is None:
+            raise HTTPException(status_code=404, detail="blueprint not found")
+        return v.to_dict()
+
+    @app.post("/v1/blueprints/{blueprint_id}/versions/{version}/submit",
+              dependencies=[Depends(auth)])
+    def blueprint_submit(blueprint_id: str, version: int) -> dict:
+        v = state.blueprints.submit(blueprint_id, version)
+        if v is None:
+            raise HTTPException(status_code=409, detail="not a draft version")
+        return v.to_dict()
+
+    @app.post("/v1/blueprints/{blueprint_id}/versions/{version}/approve",
+              dependencies=[Depends(auth)])
+    def blueprint_approve(request: Request, blueprint_id: str, version: int,
+                          payload: Optional[dict] = None) -> dict:
+        from agentbom_amd.api.blueprints import BlueprintApprovalError
+
+        try:
+            v = state.blueprints.approve(
+                _tenant_of(request), blueprint_id, version,
+                approver=getattr(request.state, "principal", ""),
+                note=str((payload or {}).get("note", "")))
+        except BlueprintApprovalError as exc:
+            raise HTTPException(status_code=409, detail=str(exc))
+        return v.to_dict()
+
+    @app.get("/v1/blueprints/{blueprint_id}/diff", dependencies=[Depends(auth)])
+    def blueprint_diff(blueprint_id: str, from_version: int,
+                       to_version: int) -> dict:
+        from agentbom_amd.api.blueprints import diff_versions
+
+        d = diff_versions(state.blueprints, blueprint_id, from_version,
+                          to_version)
+        if d is None:
+            raise HTTPException(status_code=404, detail="version not found")
+        return d
+
+    @app.post("/v1/blueprints/{blueprint_id}/drift", dependencies=[Depends(auth)])
+    def blueprint_drift(request: Request, blueprint_id: str) -> dict:
+        """Evaluate the approved composition against the latest scan;
+        opens incidents for every deviation."""
+        from agentbom_amd.api.blueprints import evaluate_drift
+
+        tenant = _tenant_of(request)
+        bp = state.blueprints.get(tenant, blueprint_id)
+        if bp is None:
+            raise HTTPException(status_code=404, detail="blueprint not found")
+        if bp.current_approved_version == 0:
+            raise HTTPException(status_code=409, detail="no approved version")
+        version = state.blueprints.get_version(blueprint_id,
+                                               bp.current_approved_version)
+        report = _latest_report(request)
+        incidents = [state.blueprints.record_incident(i)
+                     for i in evaluate_drift(bp, version, report)]
+        return {"incidents_opened": len(incidents), "incidents": incidents}
+
+    @app.get("/v1/drift-incidents", dependencies=[Depends(auth)])
+    def list_drift_incidents(request: Request,
+                             status: Optional[str] = None) -> dict:
+        rows = state.blueprints.list_incidents(_tenant_of(request), status)
+        return {"total": len(rows), "incidents": rows}
+
+    @app.post("/v1/drift-incidents/{incident_id}/resolve",
+              dependencies=[Depends(auth)])
+    def resolve_drift_incident(request: Request, incident_id: str,
+                               payload: Optional[dict] = None) -> dict:
+        doc = state.blueprints.resolve_incident(
+            incident_id, actor=getattr(request.state, "principal", "?"),
+            note=str((payload or {}).get("note", "")))
+        if doc is None:
+            raise HTTPException(status_code=404, detail="incident not found")
+        return doc
 
     @app.post("/v1/exceptions", status_code=201, dependencies=[Depends(auth)])
     def request_exception(request: Request, payload: dict) -> dict:
