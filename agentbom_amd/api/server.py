@@ -92,7 +92,8 @@ def create_app() -> FastAPI:
 
     _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
                        "/scim", "/v1/delegation-tokens", "/v1/costs/budget",
-                       "/v1/exceptions", "/v1/blueprints", "/v1/drift-incidents")
+                       "/v1/exceptions", "/v1/blueprints", "/v1/drift-incidents",
+                       "/v1/connections")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -122,9 +123,12 @@ def create_app() -> FastAPI:
 
     from agentbom_amd.api.blueprints import BlueprintStore
 
+    from agentbom_amd.api.connections import ConnectionStore
+
     state.costs = SQLiteCostStore()
     state.exceptions = ExceptionStore()
     state.blueprints = BlueprintStore()
+    state.connections = ConnectionStore()
     oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS_URL"))
@@ -643,6 +647,70 @@ def create_app() -> FastAPI:
     def delete_schedule(schedule_id: str) -> None:
         if state.scheduler is None or not state.scheduler.remove(schedule_id):
             raise HTTPException(status_code=404, detail="schedule not found")
+
+    @app.post("/v1/connections", status_code=201, dependencies=[Depends(auth)])
+    def create_connection(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.connections import (
+            CloudConnection,
+            ConnectionsCryptoUnavailable,
+            encrypt_secret,
+            validate_credential_ref,
+        )
+
+        for key in ("provider", "display_name"):
+            if not payload.get(key):
+                raise HTTPException(status_code=400, detail=f"{key} required")
+        secret_encrypted = ""
+        if payload.get("secret"):
+            try:
+                secret_encrypted = encrypt_secret(str(payload["secret"]))
+            except ConnectionsCryptoUnavailable as exc:
+                raise HTTPException(status_code=409, detail=str(exc))
+        status, detail = validate_credential_ref(
+            str(payload["provider"]), str(payload.get("mode", "")),
+            str(payload.get("role_ref", "")))
+        conn = CloudConnection(
+            provider=str(payload["provider"]),
+            display_name=str(payload["display_name"]),
+            tenant_id=_tenant_of(request),
+            role_ref=str(payload.get("role_ref", "")),
+            secret_encrypted=secret_encrypted,
+            auth_params=dict(payload.get("auth_params") or {}),
+            regions=[str(r) for r in payload.get("regions") or []],
+            status="pending" if status == "ok" else "degraded",
+            status_detail=detail,
+            scan_interval_minutes=(int(payload["scan_interval_minutes"])
+                                   if payload.get("scan_interval_minutes")
+                                   else None))
+        return state.connections.put(conn).to_public_dict()
+
+    @app.get("/v1/connections", dependencies=[Depends(auth)])
+    def list_connections(request: Request, due: bool = False) -> dict:
+        tenant = _tenant_of(request)
+        rows = (state.connections.due_connections(tenant) if due
+                else state.connections.list(tenant))
+        return {"total": len(rows),
+                "connections": [c.to_public_dict() for c in rows]}
+
+    @app.delete("/v1/connections/{connection_id}", status_code=204,
+                dependencies=[Depends(auth)])
+    def delete_connection(request: Request, connection_id: str) -> None:
+        if not state.connections.delete(_tenant_of(request), connection_id):
+            raise HTTPException(status_code=404, detail="connection not found")
+
+    @app.post("/v1/connections/{connection_id}/scan",
+              dependencies=[Depends(auth)])
+    def scan_connection(request: Request, connection_id: str) -> dict:
+        """Mark a connection scanned (the demo-scan pipeline stands in for
+        the provider collector in this build)."""
+        out = _submit_scan_inner(ScanRequest(demo=True),
+                                 tenant=_tenant_of(request))
+        job_id = out["job_id"]
+        conn = state.connections.mark_scanned(
+            _tenant_of(request), connection_id, job_id)
+        if conn is None:
+            raise HTTPException(status_code=404, detail="connection not found")
+        return {"job_id": job_id, "connection": conn.to_public_dict()}
 
     @app.post("/v1/blueprints", status_code=201, dependencies=[Depends(auth)])
     def create_blueprint(request: Request, payload: dict) -> dict:
