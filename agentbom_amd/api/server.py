@@ -93,7 +93,7 @@ def create_app() -> FastAPI:
     _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
                        "/scim", "/v1/delegation-tokens", "/v1/costs/budget",
                        "/v1/exceptions", "/v1/blueprints", "/v1/drift-incidents",
-                       "/v1/connections")
+                       "/v1/connections", "/v1/approvals")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -125,10 +125,13 @@ def create_app() -> FastAPI:
 
     from agentbom_amd.api.connections import ConnectionStore
 
+    from agentbom_amd.runtime.hitl import ApprovalQueue
+
     state.costs = SQLiteCostStore()
     state.exceptions = ExceptionStore()
     state.blueprints = BlueprintStore()
     state.connections = ConnectionStore()
+    state.approvals = ApprovalQueue()
     oidc_enabled = bool(os.environ.get("AGENT_BOM_OIDC_SECRET")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS")
                         or os.environ.get("AGENT_BOM_OIDC_JWKS_URL"))
@@ -647,6 +650,35 @@ def create_app() -> FastAPI:
     def delete_schedule(schedule_id: str) -> None:
         if state.scheduler is None or not state.scheduler.remove(schedule_id):
             raise HTTPException(status_code=404, detail="schedule not found")
+
+    @app.get("/v1/approvals", dependencies=[Depends(auth)])
+    def list_approvals(status: Optional[str] = None) -> dict:
+        rows = state.approvals.list(status=status)
+        return {"total": len(rows), "approvals": [r.to_dict() for r in rows]}
+
+    @app.post("/v1/approvals/{request_id}/approve", dependencies=[Depends(auth)])
+    def approve_request(request: Request, request_id: str,
+                        payload: Optional[dict] = None) -> dict:
+        req = state.approvals.decide(
+            request_id, approve=True,
+            actor=getattr(request.state, "principal", "?"),
+            reason=str((payload or {}).get("reason", "")))
+        if req is None:
+            raise HTTPException(status_code=409,
+                                detail="not pending (unknown/expired/decided)")
+        return req.to_dict()
+
+    @app.post("/v1/approvals/{request_id}/deny", dependencies=[Depends(auth)])
+    def deny_request(request: Request, request_id: str,
+                     payload: Optional[dict] = None) -> dict:
+        req = state.approvals.decide(
+            request_id, approve=False,
+            actor=getattr(request.state, "principal", "?"),
+            reason=str((payload or {}).get("reason", "")))
+        if req is None:
+            raise HTTPException(status_code=409,
+                                detail="not pending (unknown/expired/decided)")
+        return req.to_dict()
 
     @app.post("/v1/connections", status_code=201, dependencies=[Depends(auth)])
     def create_connection(request: Request, payload: dict) -> dict:

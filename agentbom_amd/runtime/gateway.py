@@ -81,12 +81,17 @@ class Gateway:
     """Secure-by-default relay: quarantine gate -> detectors -> breaker ->
     upstream -> response inspection -> DLP redaction."""
 
-    def __init__(self, identity_store=None) -> None:
+    def __init__(self, identity_store=None, hitl_queue=None,
+                 hitl_escalate: bool = False) -> None:
         self.upstreams: dict[str, Upstream] = {}
         self.pipeline = DetectorPipeline()
         self.identity_gate = IdentityGate()
         self.drift_gate = DriftGate()
         self.cost_gate = CostAnomalyGate()
+        # optional human-in-the-loop escalation: warn-level detector hits
+        # park the call for approval instead of passing through
+        self.hitl = hitl_queue
+        self.hitl_escalate = hitl_escalate
         # optional identity.lifecycle.AgentIdentityStore: supplies the ABAC
         # conditional-access policies relay() evaluates per call
         self.identity_store = identity_store
@@ -102,7 +107,8 @@ class Gateway:
 
     def relay(self, upstream_name: str, frame: dict[str, Any],
               principal: Optional[str] = None,
-              cost: float = 1.0, access_ctx=None) -> dict[str, Any]:
+              cost: float = 1.0, access_ctx=None,
+              approval_id: Optional[str] = None) -> dict[str, Any]:
         """``access_ctx``: an identity.lifecycle.AccessContext — when present
         together with ``self.identity_store`` policies, full ABAC
         conditional-access (deny-wins, fail-closed unknowns) gates the call."""
@@ -147,6 +153,23 @@ class Gateway:
             self.metrics["blocked_total"] += 1
             return err(-32000, "blocked by gateway: "
                        + "; ".join(a.message for a in alerts[:3]))
+        if action == "warn" and self.hitl is not None and self.hitl_escalate:
+            # single-use approval: a granted request_id authorizes exactly
+            # one replay of the SAME frame; otherwise the call parks
+            if approval_id and self.hitl.consume(approval_id, frame):
+                self.metrics["hitl_approved_total"] = \
+                    self.metrics.get("hitl_approved_total", 0) + 1
+            else:
+                req = self.hitl.park(
+                    upstream_name, principal or "anonymous", frame,
+                    alerts=[{"detector": a.detector, "severity": a.severity,
+                             "message": a.message} for a in alerts[:5]])
+                self.metrics["hitl_parked_total"] = \
+                    self.metrics.get("hitl_parked_total", 0) + 1
+                out = err(-32009, "human approval required: "
+                          + "; ".join(a.message for a in alerts[:2]))
+                out["error"]["data"] = {"approval_request_id": req.request_id}
+                return out
         if not up.breaker.allow():
             self.metrics["breaker_rejections_total"] += 1
             return err(-32003, f"upstream {upstream_name!r} circuit open")
