@@ -70,6 +70,8 @@ class _State:
         from agentbom_amd.api.webhooks import WebhookRegistry
 
         self.webhooks = WebhookRegistry()
+        # Idempotency-Key replay cache: (tenant, key) -> first response
+        self.idempotency: dict[tuple, dict] = {}
 
 
 def create_app() -> FastAPI:
@@ -286,8 +288,21 @@ def create_app() -> FastAPI:
             state.metrics["scan_failures_total"] += 1
 
     @app.post("/v1/scan", status_code=201, dependencies=[Depends(auth)])
-    def submit_scan(req: ScanRequest, request: Request) -> dict:
+    def submit_scan(req: ScanRequest, request: Request,
+                    idempotency_key: Optional[str] = Header(
+                        default=None, alias="Idempotency-Key")) -> dict:
         principal = getattr(request.state, "principal", "anonymous")
+        # Idempotency-Key: a retried POST with the same key returns the FIRST
+        # submission's response instead of launching a duplicate scan
+        # (reference api/idempotency_store.py semantics, per-tenant scoping)
+        idem_scope = None
+        if idempotency_key:
+            idem_scope = (getattr(request.state, "tenant_id", "default"),
+                          idempotency_key)
+            with state.lock:
+                cached = state.idempotency.get(idem_scope)
+            if cached is not None:
+                return cached
         ok, retry = state.quotas.check_and_record(principal)
         if not ok:
             raise HTTPException(status_code=429,
@@ -301,7 +316,13 @@ def create_app() -> FastAPI:
                                 headers={"Retry-After": str(retry_after)})
         tenant = getattr(request.state, "tenant_id", "default") or "default"
         try:
-            return _submit_scan_inner(req, tenant)
+            out = _submit_scan_inner(req, tenant)
+            if idem_scope is not None:
+                with state.lock:
+                    if len(state.idempotency) > 10_000:  # bounded replay cache
+                        state.idempotency.clear()
+                    state.idempotency[idem_scope] = out
+            return out
         finally:
             state.backpressure.release()
 

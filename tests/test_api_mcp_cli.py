@@ -349,3 +349,14 @@ class TestIdentitySkillsHistoryCli:
         assert out.exit_code == 0
         doc = _json.loads(out.output)
         assert doc["resolved_findings"][0]["vulnerability_id"] == "CVE-1"
+
+
+def test_scan_idempotency_key(client):
+    r1 = client.post("/v1/scan", json={"demo": True},
+                     headers={"Idempotency-Key": "retry-abc"})
+    r2 = client.post("/v1/scan", json={"demo": True},
+                     headers={"Idempotency-Key": "retry-abc"})
+    assert r1.json()["job_id"] == r2.json()["job_id"]  # replayed, not re-run
+    r3 = client.post("/v1/scan", json={"demo": True},
+                     headers={"Idempotency-Key": "other-key"})
+    assert r3.json()["job_id"] != r1.json()["job_id"]
