@@ -117,18 +117,45 @@ def build_dedup_match_layout(torch, arena: dict, gk, hi, lo, flags):
         "perm2": idx,
     }
     layout["u_ranges"] = _precompute_win_ranges(torch, arena, layout["u_gk"])
-    # Optional heavy-first match schedule (AGENT_BOM_MATCH_HEAVY_FIRST=1):
-    # visit rows in window-count-descending order so zipf-head walks launch
-    # first.  Measured SLOWER at 10M pkgs (2.44 vs 2.27 ms/step — the
-    # permuted per-row gathers cost more than the kernel tail they remove;
-    # profiles/r02_rank_kernel.md), so OFF by default; kept behind the knob
-    # for bigger-arena experiments where the tail dominates.
+    # Match schedule experiments (AGENT_BOM_MATCH_SCHEDULE):
+    #   row   — rows in window-count-descending order.  Measured SLOWER
+    #           (2.44 vs 2.27 ms/step): equal-wcnt rows from DIFFERENT
+    #           groups interleave, so wave lanes stop sharing a window
+    #           range and the cooperative walk (64x traffic saving on
+    #           zipf heads) stops firing.
+    #   group — GROUPS ordered by descending total work (rows x windows),
+    #           rows within a group stay adjacent and in original order:
+    #           cooperative walk intact, heavy groups launch first.
+    #   off   — natural lexicographic order (default).
     import os
 
-    if os.environ.get("AGENT_BOM_MATCH_HEAVY_FIRST", "0") == "1":
+    mode = os.environ.get("AGENT_BOM_MATCH_SCHEDULE", "off")
+    if mode in ("1", "row"):
         wcnt = (layout["u_ranges"][1] - layout["u_ranges"][0]).to(torch.int64)
         layout["heavy_order"] = torch.argsort(
             -wcnt, stable=True).to(torch.int32).contiguous()
+    elif mode == "group":
+        wbeg, wend = layout["u_ranges"]
+        wcnt = (wend - wbeg).to(torch.int64)
+        # group id = first row index of the run of equal wbeg (rows are
+        # lexicographically sorted, so equal-group rows are adjacent)
+        wb = wbeg.to(torch.int64)
+        new_grp = torch.cat([torch.ones(1, dtype=torch.bool, device=wb.device),
+                             wb[1:] != wb[:-1]])
+        grp = torch.cumsum(new_grp.to(torch.int64), 0) - 1
+        n_grp = int(grp[-1].item()) + 1 if grp.numel() else 0
+        rows_per = torch.bincount(grp, minlength=n_grp)
+        gw = torch.zeros(n_grp, dtype=torch.int64, device=wb.device)
+        gw.scatter_add_(0, grp, wcnt)  # total windows walked per group
+        # stable argsort by descending group work -> group visit order
+        grp_rank = torch.empty(n_grp, dtype=torch.int64, device=wb.device)
+        grp_rank[torch.argsort(-gw, stable=True)] = torch.arange(
+            n_grp, device=wb.device)
+        # row key = (group rank, original row index) — one stable sort
+        U = wb.numel()
+        bits = max(1, (U - 1).bit_length())
+        key = (grp_rank[grp] << bits) | torch.arange(U, device=wb.device)
+        layout["heavy_order"] = torch.argsort(key).to(torch.int32).contiguous()
     else:
         layout["heavy_order"] = None
     return layout
