@@ -576,6 +576,10 @@ def to_json(report: AIBOMReport) -> dict[str, Any]:
         if not f.first_seen:
             f.first_seen = scan_observed_at
     unified = [f.to_dict() for f in export_findings]
+    from agentbom_amd.models.maestro import classify_finding, maestro_summary
+
+    for row in unified:
+        row["maestro_layer"] = classify_finding(row).value
     finding_summary = _build_finding_summary(unified)
     asset_inventory = _build_asset_inventory(unified)
 
@@ -659,6 +663,7 @@ def to_json(report: AIBOMReport) -> dict[str, Any]:
         },
         "findings": unified,
         "threat_framework_summary": _build_framework_summary(report.blast_radii),
+        "maestro_summary": maestro_summary(unified),
         "scorecard_summary": {
             "total_packages": report.total_packages,
             "with_scorecard": sum(
