@@ -608,6 +608,16 @@ def create_app() -> FastAPI:
         old = old or snaps[1]["snapshot_id"]
         return state.graph_store.diff_snapshots(old, new)
 
+    @app.get("/v1/findings/reach", dependencies=[Depends(auth)])
+    def findings_reach(request: Request, band: Optional[str] = None) -> dict:
+        """Effective-reach triage view: composite + band per finding."""
+        from agentbom_amd.graph.effective_reach import effective_reach_summary
+
+        out = effective_reach_summary(_latest_report(request))
+        if band:
+            out["findings"] = [r for r in out["findings"] if r["band"] == band]
+        return out
+
     @app.get("/v1/findings/toxic-combinations", dependencies=[Depends(auth)])
     def toxic_combinations(request: Request) -> dict:
         report = _latest_report(request)
