@@ -176,6 +176,7 @@ def _scan_impl(
     scan_prompts: Optional[str] = None, scan_pii: Optional[str] = None,
     skill: Optional[str] = None,
     github_actions: Optional[str] = None, repo_inventory: Optional[str] = None,
+    k8s_posture: Optional[str] = None,
     # live cloud collectors
     aws_live: bool = False, azure_live: Optional[str] = None,
     azure_token: Optional[str] = None, gcp_live: Optional[str] = None,
@@ -421,6 +422,8 @@ def _scan_impl(
         run_scanner_driver("ci_workflows", report, github_actions or repo)
     if repo_inventory or repo:
         run_scanner_driver("repo_inventory", report, repo_inventory or repo)
+    if k8s_posture:
+        run_scanner_driver("kspm", report, k8s_posture)
     if scan_prompts:
         from agentbom_amd.scan.surfaces_extra import scan_prompt_files
 
@@ -638,6 +641,10 @@ def _scan_options(f):
                      type=click.Path(exists=True), default=None,
                      help="Scan .github/workflows for agentic-CI usage + "
                           "pipeline hardening (implied by --repo)."),
+        click.option("--k8s-posture", "k8s_posture",
+                     type=click.Path(exists=True), default=None,
+                     help="Evaluate an exported Kubernetes cluster inventory "
+                          "(pods/RBAC) against the KSPM posture checks."),
         click.option("--repo-inventory", "repo_inventory",
                      type=click.Path(exists=True), default=None,
                      help="Collect the project directory/file/import "

@@ -222,6 +222,19 @@ def _run_floating_refs(report: AIBOMReport, target: str) -> int:
     return len(refs)
 
 
+def _run_kspm(report: AIBOMReport, target: str) -> int:
+    from agentbom_amd.scan.kspm import kspm_finding_to_finding, scan_cluster_posture
+
+    res = scan_cluster_posture(target)
+    report.extra_data["kspm_posture"] = res.to_evidence_dict()
+    report.findings.extend(kspm_finding_to_finding(f) for f in res.findings)
+    if res.status == "partial":
+        report.warnings.append(
+            "KSPM posture is PARTIAL: one or more collectors were denied or "
+            "failed — absent evidence is never a clean pass")
+    return len(res.findings)
+
+
 def _run_ci_workflows(report: AIBOMReport, target: str) -> int:
     from agentbom_amd.scan.ci_workflows import scan_github_actions
 
@@ -269,6 +282,8 @@ def _ensure_builtins() -> None:
     register_scanner(ScannerRegistration(
         "ci_workflows", "code", _run_ci_workflows,
         failure_mode=WARN_CONTINUE))
+    register_scanner(ScannerRegistration(
+        "kspm", "cloud_cis", _run_kspm, failure_mode=FAIL_CLOSED))
     register_scanner(ScannerRegistration(
         "repo_inventory", "code", _run_repo_inventory,
         failure_mode=WARN_CONTINUE))
