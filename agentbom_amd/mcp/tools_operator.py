@@ -279,16 +279,24 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
             by_status[r.status] += 1
         return {"summary": dict(by_status), "checks": [r.to_dict() for r in results]}
 
-    @tool("kspm_cluster_posture", "Kubernetes posture: scan manifests for "
-                                  "privileged pods, host mounts, missing limits.",
+    @tool("kspm_cluster_posture", "Kubernetes posture: exported cluster "
+                                  "inventory (runtime pods/RBAC, evidence "
+                                  "envelope) or a manifest tree.",
           {"type": "object", "properties": {"path": {"type": "string"}},
            "required": ["path"]})
     def kspm_cluster_posture(path: str) -> dict:
+        import os as _os
+
+        if _os.path.isfile(path):
+            from agentbom_amd.scan.kspm import scan_cluster_posture
+
+            return scan_cluster_posture(path).to_evidence_dict()
         from agentbom_amd.scan.iac import scan_iac_tree
 
         findings = [f.to_dict() for f in scan_iac_tree(path)
                     if f.rule_id.startswith("K8S")]
-        return {"finding_count": len(findings), "findings": findings}
+        return {"schema_version": "kspm.manifests.v1",
+                "finding_count": len(findings), "findings": findings}
 
     @tool("marketplace_check", "Pre-install trust check for an MCP server "
                                "package: typosquat, malicious list, advisories.",
@@ -927,6 +935,54 @@ def register_operator_tools(server) -> None:  # noqa: C901 — one registrar
         return trust_score(
             Package(name=name, version=version or "0.0.0", ecosystem=ecosystem),
             advisory_windows=load_advisory_windows(offline=True))
+
+    @tool("effective_reach", "Effective-reach triage bands per finding: "
+                             "composite of CVSS/EPSS/KEV/tool-capability/"
+                             "credential-tier/agent-breadth.",
+          {"type": "object", "properties": {
+              "band": {"type": "string",
+                       "enum": ["green", "amber", "red", "pulsing-red"]}}})
+    def effective_reach(band: str = "") -> dict:
+        from agentbom_amd.graph.effective_reach import effective_reach_summary
+
+        report, _g = server._ensure_scan()
+        out = effective_reach_summary(report)
+        if band:
+            out["findings"] = [r for r in out["findings"]
+                               if r["band"] == band]
+        return out
+
+    @tool("maestro_posture", "MAESTRO KC-layer posture: findings classified "
+                             "by agentic-AI layer (KC1 models .. KC6 infra).")
+    def maestro_posture() -> dict:
+        from agentbom_amd.models.maestro import maestro_summary
+
+        report, _g = server._ensure_scan()
+        return maestro_summary([f.to_dict() for f in report.to_findings()])
+
+    @tool("cost_runway", "Budget runway from OTel GenAI span records: "
+                         "burn-rate basis, projected period spend, days "
+                         "remaining, exhaustion timestamp.",
+          {"type": "object", "properties": {
+              "spans_path": {"type": "string"},
+              "budget_usd": {"type": "number"}},
+           "required": ["spans_path"]})
+    def cost_runway(spans_path: str, budget_usd: float = 0.0) -> dict:
+        from agentbom_amd.api.cost_store import (
+            CostBudget,
+            LLMCostRecord,
+            forecast_spend,
+        )
+
+        records = []
+        for row in _load_jsonl(spans_path):
+            records.append(LLMCostRecord(
+                tenant_id="local", agent=str(row.get("agent", "?")),
+                cost_usd=float(row.get("cost_usd", 0.0)),
+                model=str(row.get("model", "")),
+                observed_at=str(row.get("observed_at", ""))))
+        budget = CostBudget("local", budget_usd) if budget_usd else None
+        return forecast_spend(records, budget)
 
     @tool("tool_metrics", "Per-tool call/latency/error counters for this "
                           "MCP session.")

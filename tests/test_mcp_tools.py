@@ -489,3 +489,43 @@ class TestYoucomSearch:
         s.youcom_client = boom
         out = self._call(s, query="q")
         assert "failed" in out["error"] and out["results"] == []
+
+
+class TestPostureTools:
+    def test_effective_reach_tool(self, server):
+        out = call(server, "effective_reach")
+        assert out["findings"] and sum(out["bands"].values()) >= 1
+        band = out["findings"][0]["band"]
+        filtered = call(server, "effective_reach", band=band)
+        assert all(r["band"] == band for r in filtered["findings"])
+
+    def test_maestro_posture_tool(self, server):
+        out = call(server, "maestro_posture")
+        assert out["layers"] and all(r["layer"].startswith("KC")
+                                     for r in out["layers"])
+
+    def test_kspm_runtime_inventory_file(self, server, tmp_path):
+        import json as _json
+
+        doc = {"pods": {"items": [{"metadata": {"name": "p", "namespace": "d"},
+                                   "spec": {"hostNetwork": True,
+                                            "containers": []}}]}}
+        f = tmp_path / "cluster.json"
+        f.write_text(_json.dumps(doc))
+        out = call(server, "kspm_cluster_posture", path=str(f))
+        assert out["schema_version"] == "kspm.cluster.posture.v1"
+        assert out["finding_count"] >= 1
+        assert out["status"] == "partial"  # rbac collectors absent
+
+    def test_cost_runway_tool(self, server, tmp_path):
+        import datetime as dt
+        import json as _json
+
+        now = dt.datetime.now(dt.timezone.utc)
+        rows = [{"agent": "bot", "cost_usd": 1.0, "model": "opus",
+                 "observed_at": (now - dt.timedelta(hours=h)).isoformat()}
+                for h in (1, 5)]
+        f = tmp_path / "spans.jsonl"
+        f.write_text("\n".join(_json.dumps(r) for r in rows))
+        out = call(server, "cost_runway", spans_path=str(f), budget_usd=100.0)
+        assert out["status"] == "ok" and out["days_remaining"] > 0
