@@ -72,6 +72,7 @@ class _State:
         self.webhooks = WebhookRegistry()
         # Idempotency-Key replay cache: (tenant, key) -> first response
         self.idempotency: dict[tuple, dict] = {}
+        self.lifecycle = None  # AssetTracker, created on first observe
 
 
 def create_app() -> FastAPI:
@@ -607,6 +608,31 @@ def create_app() -> FastAPI:
         new = new or snaps[0]["snapshot_id"]
         old = old or snaps[1]["snapshot_id"]
         return state.graph_store.diff_snapshots(old, new)
+
+    @app.post("/v1/findings/lifecycle/observe", dependencies=[Depends(auth)])
+    def lifecycle_observe(request: Request) -> dict:
+        """Fold the latest scan into the finding-lifecycle tracker
+        (first_seen / resolved / reopened / MTTR)."""
+        from agentbom_amd.output.json_fmt import to_json
+
+        if state.lifecycle is None:
+            from agentbom_amd.scan.history import AssetTracker
+
+            state.lifecycle = AssetTracker(":memory:")
+        delta = state.lifecycle.observe_scan(to_json(_latest_report(request)))
+        return delta
+
+    @app.get("/v1/findings/lifecycle", dependencies=[Depends(auth)])
+    def lifecycle_summary() -> dict:
+        if state.lifecycle is None:
+            return {"observed_scans": 0, "mttr_seconds": None}
+        mttr = state.lifecycle.mttr_seconds()
+        rows = state.lifecycle.conn.execute(
+            "SELECT COUNT(*), SUM(resolved_at IS NOT NULL) FROM"
+            " finding_lifecycle").fetchone()
+        return {"tracked_findings": int(rows[0] or 0),
+                "resolved": int(rows[1] or 0),
+                "mttr_seconds": round(mttr, 1) if mttr is not None else None}
 
     @app.get("/v1/findings/reach", dependencies=[Depends(auth)])
     def findings_reach(request: Request, band: Optional[str] = None) -> dict:

@@ -99,3 +99,19 @@ def test_reach_endpoint():
     one_band = out["findings"][0]["band"]
     filtered = client.get(f"/v1/findings/reach?band={one_band}").json()
     assert all(r["band"] == one_band for r in filtered["findings"])
+
+
+def test_finding_lifecycle_endpoints():
+    from starlette.testclient import TestClient
+
+    from agentbom_amd.api.server import create_app
+
+    client = TestClient(create_app())
+    client.post("/v1/scan", json={"demo": True})
+    assert client.get("/v1/findings/lifecycle").json()["observed_scans"] == 0
+    d1 = client.post("/v1/findings/lifecycle/observe").json()
+    assert d1["new"] > 0 and d1["active"] == d1["new"]
+    d2 = client.post("/v1/findings/lifecycle/observe").json()
+    assert d2["new"] == 0 and d2["resolved"] == 0  # same scan, steady state
+    s = client.get("/v1/findings/lifecycle").json()
+    assert s["tracked_findings"] == d1["active"] and s["resolved"] == 0
