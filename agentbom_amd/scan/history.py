@@ -81,8 +81,11 @@ class AssetTracker:
 
     def __init__(self, path: Optional[str | Path] = None):
         db = Path(path) if path else _history_dir().parent / "assets.db"
-        db.parent.mkdir(parents=True, exist_ok=True)
-        self.conn = sqlite3.connect(str(db))
+        if str(db) != ":memory:":
+            db.parent.mkdir(parents=True, exist_ok=True)
+        # API threads share the tracker; sqlite guards are relaxed and the
+        # caller (server/state) serializes writes per request
+        self.conn = sqlite3.connect(str(db), check_same_thread=False)
         self.conn.execute(
             """CREATE TABLE IF NOT EXISTS finding_lifecycle (
                 finding_key TEXT PRIMARY KEY,
