@@ -73,6 +73,7 @@ class _State:
         # Idempotency-Key replay cache: (tenant, key) -> first response
         self.idempotency: dict[tuple, dict] = {}
         self.lifecycle = None  # AssetTracker, created on first observe
+        self.exports = None  # ExportManager, created on first use
 
 
 def create_app() -> FastAPI:
@@ -707,6 +708,62 @@ def create_app() -> FastAPI:
     def delete_schedule(schedule_id: str) -> None:
         if state.scheduler is None or not state.scheduler.remove(schedule_id):
             raise HTTPException(status_code=404, detail="schedule not found")
+
+    def _exports(request: Request):
+        if state.exports is None:
+            from agentbom_amd.api.exports import ExportManager
+
+            state.exports = ExportManager(
+                get_report=lambda: _latest_report(request),
+                file_root=os.environ.get("AGENT_BOM_EXPORT_ROOT", "."))
+        return state.exports
+
+    @app.post("/v1/exports/destinations", status_code=201,
+              dependencies=[Depends(auth)])
+    def add_export_destination(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.exports import ExportDestination
+
+        if not payload.get("target"):
+            raise HTTPException(status_code=400, detail="target required")
+        try:
+            dest = ExportDestination(
+                target=str(payload["target"]),
+                format=str(payload.get("format", "json")),
+                name=str(payload.get("name", "")),
+                tenant_id=_tenant_of(request))
+        except ValueError as exc:
+            raise HTTPException(status_code=400, detail=str(exc))
+        return _exports(request).add_destination(dest).to_dict()
+
+    @app.get("/v1/exports/destinations", dependencies=[Depends(auth)])
+    def list_export_destinations(request: Request) -> dict:
+        mgr = _exports(request)
+        return {"destinations": [d.to_dict()
+                                 for d in mgr.destinations.values()]}
+
+    @app.post("/v1/exports/schedules", status_code=201,
+              dependencies=[Depends(auth)])
+    def add_export_schedule(request: Request, payload: dict) -> dict:
+        mgr = _exports(request)
+        sched = mgr.add_schedule(str(payload.get("destination_id", "")),
+                                 float(payload.get("interval_s", 3600)))
+        if sched is None:
+            raise HTTPException(status_code=404, detail="unknown destination")
+        mgr.start()
+        return sched.to_dict()
+
+    @app.get("/v1/exports/schedules", dependencies=[Depends(auth)])
+    def list_export_schedules(request: Request) -> dict:
+        mgr = _exports(request)
+        return {"schedules": [s.to_dict() for s in mgr.schedules.values()],
+                "recent_deliveries": mgr.deliveries[-20:]}
+
+    @app.post("/v1/exports/run/{destination_id}", dependencies=[Depends(auth)])
+    def run_export_now(request: Request, destination_id: str) -> dict:
+        out = _exports(request).run_export(destination_id)
+        if not out.get("ok") and out.get("error") == "unknown destination":
+            raise HTTPException(status_code=404, detail="unknown destination")
+        return out
 
     @app.get("/v1/approvals", dependencies=[Depends(auth)])
     def list_approvals(status: Optional[str] = None) -> dict:
