@@ -74,6 +74,7 @@ class _State:
         self.idempotency: dict[tuple, dict] = {}
         self.lifecycle = None  # AssetTracker, created on first observe
         self.exports = None  # ExportManager, created on first use
+        self.oauth_as = None  # OAuthAuthorizationServer (AGENT_BOM_OAUTH_AS=1)
 
 
 def create_app() -> FastAPI:
@@ -192,6 +193,92 @@ def create_app() -> FastAPI:
     @app.get("/healthz")
     def healthz() -> dict:
         return {"status": "ok", "version": __version__}
+
+    # ── OAuth 2.1 AS for the MCP auth spec (opt-in: AGENT_BOM_OAUTH_AS=1;
+    #    key generation costs seconds, so the broker is lazy + gated) ──────
+
+    def _oauth_as():
+        if state.oauth_as is None:
+            if os.environ.get("AGENT_BOM_OAUTH_AS", "0") != "1":
+                raise HTTPException(status_code=404,
+                                    detail="OAuth broker not enabled "
+                                           "(AGENT_BOM_OAUTH_AS=1)")
+            from agentbom_amd.api.oauth_as import OAuthAuthorizationServer
+
+            issuer = os.environ.get("AGENT_BOM_OAUTH_ISSUER",
+                                    "https://agent-bom.local")
+            state.oauth_as = OAuthAuthorizationServer(issuer=issuer)
+        return state.oauth_as
+
+    @app.get("/.well-known/oauth-authorization-server")
+    def oauth_metadata() -> dict:
+        return _oauth_as().metadata()
+
+    @app.get("/oauth/jwks.json")
+    def oauth_jwks() -> dict:
+        return _oauth_as().jwks()
+
+    @app.post("/oauth/register", status_code=201)
+    def oauth_register(payload: dict) -> dict:
+        from agentbom_amd.api.oauth_as import OAuthError
+
+        try:
+            return _oauth_as().register_client(
+                [str(u) for u in payload.get("redirect_uris") or []],
+                client_name=str(payload.get("client_name", "")),
+                confidential=(payload.get("token_endpoint_auth_method")
+                              == "client_secret_basic"))
+        except OAuthError as exc:
+            raise HTTPException(status_code=exc.status, detail=exc.to_dict())
+
+    @app.post("/oauth/authorize")
+    def oauth_authorize(payload: dict) -> dict:
+        """Programmatic consent (this build has no browser UI); returns
+        the code + state for the client's redirect handling."""
+        from agentbom_amd.api.oauth_as import OAuthError
+
+        try:
+            return _oauth_as().authorize(
+                str(payload.get("client_id", "")),
+                str(payload.get("redirect_uri", "")),
+                str(payload.get("code_challenge", "")),
+                str(payload.get("code_challenge_method", "")),
+                scope=str(payload.get("scope", "mcp")),
+                state=str(payload.get("state", "")))
+        except OAuthError as exc:
+            raise HTTPException(status_code=exc.status, detail=exc.to_dict())
+
+    @app.post("/oauth/token")
+    def oauth_token(payload: dict, request: Request) -> dict:
+        import base64 as _b64
+
+        from agentbom_amd.api.oauth_as import OAuthError
+
+        srv = _oauth_as()
+        grant = str(payload.get("grant_type", ""))
+        try:
+            if grant == "authorization_code":
+                return srv.token_authorization_code(
+                    str(payload.get("code", "")),
+                    str(payload.get("client_id", "")),
+                    str(payload.get("redirect_uri", "")),
+                    str(payload.get("code_verifier", "")))
+            if grant == "client_credentials":
+                cid = str(payload.get("client_id", ""))
+                secret = str(payload.get("client_secret", ""))
+                authz = request.headers.get("Authorization", "")
+                if authz.startswith("Basic "):
+                    try:
+                        raw = _b64.b64decode(authz[6:]).decode()
+                        cid, _, secret = raw.partition(":")
+                    except Exception:
+                        pass
+                return srv.token_client_credentials(cid, secret,
+                                                    scope=str(payload.get(
+                                                        "scope", "mcp")))
+            raise OAuthError("unsupported_grant_type", grant)
+        except OAuthError as exc:
+            raise HTTPException(status_code=exc.status, detail=exc.to_dict())
 
     @app.get("/metrics")
     def metrics() -> Any:
