@@ -800,8 +800,11 @@ def create_app() -> FastAPI:
         if state.exports is None:
             from agentbom_amd.api.exports import ExportManager
 
+            # resolve the DEFAULT tenant's latest report at delivery time
+            # (binding the creating request would freeze its tenant forever
+            # and leak the request object into the scheduler thread)
             state.exports = ExportManager(
-                get_report=lambda: _latest_report(request),
+                get_report=lambda: _latest_report(None),
                 file_root=os.environ.get("AGENT_BOM_EXPORT_ROOT", "."))
         return state.exports
 
