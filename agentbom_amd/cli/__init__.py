@@ -176,7 +176,7 @@ def _scan_impl(
     scan_prompts: Optional[str] = None, scan_pii: Optional[str] = None,
     skill: Optional[str] = None,
     github_actions: Optional[str] = None, repo_inventory: Optional[str] = None,
-    k8s_posture: Optional[str] = None,
+    k8s_posture: Optional[str] = None, exceptions_db: Optional[str] = None,
     # live cloud collectors
     aws_live: bool = False, azure_live: Optional[str] = None,
     azure_token: Optional[str] = None, gcp_live: Optional[str] = None,
@@ -424,6 +424,16 @@ def _scan_impl(
         run_scanner_driver("repo_inventory", report, repo_inventory or repo)
     if k8s_posture:
         run_scanner_driver("kspm", report, k8s_posture)
+    if exceptions_db:
+        from agentbom_amd.api.exceptions_store import (
+            ExceptionStore,
+            apply_exceptions_to_report,
+        )
+
+        waived = apply_exceptions_to_report(report, ExceptionStore(exceptions_db))
+        if waived:
+            click.echo(f"exceptions: {waived} finding(s) suppressed by "
+                       "approved waivers", err=True)
     if scan_prompts:
         from agentbom_amd.scan.surfaces_extra import scan_prompt_files
 
@@ -641,6 +651,11 @@ def _scan_options(f):
                      type=click.Path(exists=True), default=None,
                      help="Scan .github/workflows for agentic-CI usage + "
                           "pipeline hardening (implied by --repo)."),
+        click.option("--exceptions-db", "exceptions_db",
+                     type=click.Path(exists=True), default=None,
+                     help="Apply APPROVED, unexpired waivers from an "
+                          "exception store (suppression contract; expired "
+                          "waivers never suppress)."),
         click.option("--k8s-posture", "k8s_posture",
                      type=click.Path(exists=True), default=None,
                      help="Evaluate an exported Kubernetes cluster inventory "

@@ -117,3 +117,33 @@ class TestApi:
         assert v["status"] == "revoked"
         assert client.get("/v1/exceptions/audit").json()["chain_valid"]
         assert client.post("/v1/exceptions", json={}).status_code == 400
+
+
+def test_cli_exceptions_db(tmp_path):
+    """--exceptions-db applies approved waivers in a CLI scan."""
+    import json
+    import subprocess
+    import sys
+
+    db = tmp_path / "waivers.db"
+    store = ExceptionStore(str(db))
+    # find a demo finding to waive
+    from agentbom_amd.scan.orchestrator import run_demo_scan
+
+    target = run_demo_scan().blast_radii[0]
+    exc = store.request(_exc(vuln_id=target.vulnerability.id,
+                             package_name=target.package.name))
+    store.approve(exc.exception_id, "secops")
+    out = subprocess.run(
+        [sys.executable, "-m", "agentbom_amd.cli", "scan", "--demo",
+         "--exceptions-db", str(db), "--format", "json", "--exit-zero"],
+        capture_output=True, text=True, timeout=240)
+    # demo contains a malicious package: that gate fails closed even with
+    # --exit-zero, so only the waiver mechanics are asserted here
+    assert out.returncode in (0, 1), out.stderr[-500:]
+    assert "suppressed by" in out.stderr
+    doc = json.loads(out.stdout)
+    waived = [r for r in doc["blast_radius"]
+              if r["suppression_id"] == exc.exception_id]
+    assert waived and all(r["suppression_state"] == "exception"
+                          for r in waived)
