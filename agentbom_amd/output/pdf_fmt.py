@@ -39,6 +39,9 @@ def to_pdf_bytes(report: AIBOMReport) -> bytes:
     counts = report.severity_counts()
     pages: list[list[tuple[int, int, str, int]]] = []
 
+    from agentbom_amd.scan.risk import estate_exec_score
+
+    score = estate_exec_score(report)
     header = [
         (_MARGIN, _PAGE_H - _MARGIN, "agent-bom AI-BOM Scan Report", 18),
         (_MARGIN, _PAGE_H - _MARGIN - 24,
@@ -48,10 +51,13 @@ def to_pdf_bytes(report: AIBOMReport) -> bytes:
          f"findings: {report.total_vulnerabilities}   critical: "
          f"{counts['critical']}   high: {counts['high']}   medium: "
          f"{counts['medium']}   low: {counts['low']}", 11),
-        (_MARGIN, _PAGE_H - _MARGIN - 64, "risk  severity  vulnerability"
+        (_MARGIN, _PAGE_H - _MARGIN - 56,
+         f"estate posture: grade {score.get('grade', '?')} "
+         f"(score {score.get('score', '?')}/100)", 11),
+        (_MARGIN, _PAGE_H - _MARGIN - 80, "risk  severity  vulnerability"
          "          package                    reachability", 10),
     ]
-    y = _PAGE_H - _MARGIN - 64 - _LINE_H
+    y = _PAGE_H - _MARGIN - 80 - _LINE_H
     current = list(header)
     for br in report.blast_radii:
         if y < _MARGIN:
@@ -68,6 +74,30 @@ def to_pdf_bytes(report: AIBOMReport) -> bytes:
             f"{br.package.name}@{br.package.version:<12s}"
             f"  {br.reachability}{kev}{mal}", 9))
         y -= _LINE_H
+
+    # remediation section (same prioritized plan as console/HTML)
+    from agentbom_amd.output.json_fmt import _build_remediation_json
+
+    plan = _build_remediation_json(report)
+    if plan:
+        needed = (len(plan[:15]) + 3) * _LINE_H
+        if y - needed < _MARGIN:
+            pages.append(current)
+            current = [(_MARGIN, _PAGE_H - _MARGIN,
+                        "agent-bom scan report (continued)", 12)]
+            y = _PAGE_H - _MARGIN - 2 * _LINE_H
+        y -= _LINE_H
+        current.append((_MARGIN, y, "Remediation plan (highest risk first)",
+                        12))
+        y -= _LINE_H
+        for i, item in enumerate(plan[:15], 1):
+            current.append((
+                _MARGIN, y,
+                f"{i:>2d}. {item.get('package', '?'):<34s} -> "
+                f"{item.get('fix_version') or 'no fix yet':<14s} "
+                f"({len(item.get('vulns', []))} vulns, max risk "
+                f"{item.get('max_risk_score', 0):.1f})", 9))
+            y -= _LINE_H
     pages.append(current)
 
     # assemble objects: 1 catalog, 2 pages-tree, 3 font, then per page
