@@ -2,7 +2,7 @@
 
 Reference: src/agent_bom/graph/types.py (42 EntityTypes, 14 semantic
 layers, 49 RelationshipTypes, 4 NodeStatus, 6 layouts) — cited whole by
-docs/graph/CONTRACT.md; these values are the persisted wire contract.
+docs/graph_contract.json; these values are the persisted wire contract.
 
 ``REL_CODE``/``ENTITY_CODE`` add the compact u8 encodings the GPU CSR
 engine carries per edge/node (traversal classes become 32-bit masks).
