@@ -75,6 +75,7 @@ class _State:
         self.lifecycle = None  # AssetTracker, created on first observe
         self.exports = None  # ExportManager, created on first use
         self.oauth_as = None  # OAuthAuthorizationServer (AGENT_BOM_OAUTH_AS=1)
+        self.model_keys = None  # ModelKeyBroker, created on first use
 
 
 def create_app() -> FastAPI:
@@ -98,7 +99,7 @@ def create_app() -> FastAPI:
     _WRITE_PREFIXES = ("/v1/identities", "/v1/schedules", "/v1/fleet",
                        "/scim", "/v1/delegation-tokens", "/v1/costs/budget",
                        "/v1/exceptions", "/v1/blueprints", "/v1/drift-incidents",
-                       "/v1/connections", "/v1/approvals")
+                       "/v1/connections", "/v1/approvals", "/v1/model-keys")
 
     def _role_allows(role: str, method: str, path: str) -> bool:
         if role == "admin":
@@ -854,6 +855,61 @@ def create_app() -> FastAPI:
         if not out.get("ok") and out.get("error") == "unknown destination":
             raise HTTPException(status_code=404, detail="unknown destination")
         return out
+
+    def _model_keys(request: Request):
+        if state.model_keys is None:
+            from agentbom_amd.api.model_keys import ModelKeyBroker
+
+            state.model_keys = ModelKeyBroker()
+        return state.model_keys
+
+    @app.post("/v1/model-keys/providers", status_code=201,
+              dependencies=[Depends(auth)])
+    def register_model_provider_key(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.connections import ConnectionsCryptoUnavailable
+
+        for k in ("provider", "key"):
+            if not payload.get(k):
+                raise HTTPException(status_code=400, detail=f"{k} required")
+        try:
+            rec = _model_keys(request).register_provider_key(
+                str(payload["provider"]), str(payload["key"]),
+                tenant_id=_tenant_of(request),
+                label=str(payload.get("label", "")))
+        except ConnectionsCryptoUnavailable as exc:
+            raise HTTPException(status_code=409, detail=str(exc))
+        return rec.to_public_dict()
+
+    @app.post("/v1/model-keys/virtual", status_code=201,
+              dependencies=[Depends(auth)])
+    def mint_virtual_model_key(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.model_keys import ModelKeyBrokerError
+
+        try:
+            rec, raw = _model_keys(request).mint_virtual_key(
+                str(payload.get("provider_key_id", "")),
+                tenant_id=_tenant_of(request),
+                holder=str(payload.get("holder", "")),
+                model_allowlist=[str(m) for m in
+                                 payload.get("model_allowlist") or []],
+                ttl_hours=float(payload.get("ttl_hours", 24.0)))
+        except ModelKeyBrokerError as exc:
+            raise HTTPException(status_code=404, detail=str(exc))
+        return {**rec.to_public_dict(), "virtual_key": raw}  # raw shown ONCE
+
+    @app.get("/v1/model-keys/virtual", dependencies=[Depends(auth)])
+    def list_virtual_model_keys(request: Request) -> dict:
+        return {"virtual_keys":
+                _model_keys(request).list_virtual_keys(_tenant_of(request))}
+
+    @app.post("/v1/model-keys/virtual/{virtual_key_id}/revoke",
+              dependencies=[Depends(auth)])
+    def revoke_virtual_model_key(request: Request,
+                                 virtual_key_id: str) -> dict:
+        if not _model_keys(request).revoke_virtual_key(
+                virtual_key_id, tenant_id=_tenant_of(request)):
+            raise HTTPException(status_code=404, detail="unknown virtual key")
+        return {"revoked": virtual_key_id}
 
     @app.get("/v1/approvals", dependencies=[Depends(auth)])
     def list_approvals(status: Optional[str] = None) -> dict:

@@ -90,3 +90,29 @@ class TestFailClosed:
 
         with pytest.raises(ConnectionsCryptoUnavailable):
             broker.register_provider_key("openai", "sk-x")
+
+
+class TestApi:
+    def test_endpoints(self, monkeypatch):
+        monkeypatch.setenv("AGENT_BOM_CONNECTIONS_KEY",
+                           _ENV["AGENT_BOM_CONNECTIONS_KEY"])
+        from starlette.testclient import TestClient
+
+        from agentbom_amd.api.server import create_app
+
+        client = TestClient(create_app())
+        r = client.post("/v1/model-keys/providers", json={
+            "provider": "anthropic", "key": "sk-ant-REAL-999",
+            "label": "prod"})
+        assert r.status_code == 201 and "key" not in r.json()
+        pkid = r.json()["provider_key_id"]
+        v = client.post("/v1/model-keys/virtual", json={
+            "provider_key_id": pkid, "holder": "agent-a"}).json()
+        assert v["virtual_key"].startswith("abvk_")
+        listed = client.get("/v1/model-keys/virtual").json()["virtual_keys"]
+        assert listed and "sk-ant-REAL-999" not in str(listed)
+        assert client.post(
+            f"/v1/model-keys/virtual/{v['virtual_key_id']}/revoke"
+        ).json()["revoked"] == v["virtual_key_id"]
+        assert client.post("/v1/model-keys/virtual",
+                           json={"provider_key_id": "mpk-x"}).status_code == 404
