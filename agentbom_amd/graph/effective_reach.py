@@ -23,7 +23,7 @@ instead of re-walking a context graph.
 from __future__ import annotations
 
 import re
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Optional
 
 from agentbom_amd.scan.risk import ToolCapability, classify_mcp_tool

@@ -17,7 +17,7 @@ tests.  Discovery itself stays behind the nhi.py enable flags.
 from __future__ import annotations
 
 import re
-from typing import Any, Iterable, Optional
+from typing import Iterable, Optional
 
 from agentbom_amd.utils.http_client import check_offline, create_client, request_with_retry
 

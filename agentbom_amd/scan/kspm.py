@@ -18,7 +18,7 @@ from __future__ import annotations
 import json
 from dataclasses import dataclass, field
 from pathlib import Path
-from typing import Any, Callable, Optional
+from typing import Any, Optional
 
 _BUILTIN_ROLE_PREFIXES = ("system:", "cluster-admin", "admin", "edit", "view")
 
