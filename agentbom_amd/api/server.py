@@ -76,6 +76,7 @@ class _State:
         self.exports = None  # ExportManager, created on first use
         self.oauth_as = None  # OAuthAuthorizationServer (AGENT_BOM_OAUTH_AS=1)
         self.model_keys = None  # ModelKeyBroker, created on first use
+        self.hub = None  # ComplianceHub, created on first use
 
 
 def create_app() -> FastAPI:
@@ -1253,6 +1254,42 @@ def create_app() -> FastAPI:
     def read_audit(limit: int = 100) -> dict:
         return {"total": len(state.audit_entries),
                 "entries": state.audit_entries[-limit:]}
+
+    def _hub(request: Request):
+        if state.hub is None:
+            from agentbom_amd.api.hub import ComplianceHub
+
+            state.hub = ComplianceHub()
+        return state.hub
+
+    @app.post("/v1/compliance/ingest", dependencies=[Depends(auth)])
+    def hub_ingest(request: Request, payload: dict) -> dict:
+        """External-scan / connector finding ingest into the hub."""
+        if not payload.get("source"):
+            raise HTTPException(status_code=400, detail="source required")
+        findings = payload.get("findings")
+        if not isinstance(findings, list):
+            raise HTTPException(status_code=400,
+                                detail="findings must be a list")
+        return _hub(request).ingest(
+            _tenant_of(request), str(payload["source"]), findings,
+            reconcile_absent=bool(payload.get("reconcile_absent", True)))
+
+    @app.get("/v1/compliance/hub/overview", dependencies=[Depends(auth)])
+    def hub_overview(request: Request) -> dict:
+        return _hub(request).overview(_tenant_of(request))
+
+    @app.get("/v1/compliance/hub/findings", dependencies=[Depends(auth)])
+    def hub_findings(request: Request, source: Optional[str] = None,
+                     status: str = "open", limit: int = 1000) -> dict:
+        rows = _hub(request).findings(_tenant_of(request), source=source,
+                                      status=status, limit=min(limit, 10_000))
+        return {"total": len(rows), "findings": rows}
+
+    @app.get("/v1/compliance/hub/ledger", dependencies=[Depends(auth)])
+    def hub_ledger(request: Request, limit: int = 100) -> dict:
+        return {"events": _hub(request).ledger(_tenant_of(request),
+                                               limit=min(limit, 1000))}
 
     @app.get("/v1/compliance/{framework}/report", dependencies=[Depends(auth)])
     def compliance_report(request: Request, framework: str) -> dict:
