@@ -1228,6 +1228,23 @@ def create_app() -> FastAPI:
             or state.costs.get_budget(tenant)
         return forecast_spend(records, budget)
 
+    @app.get("/v1/costs/anomalies", dependencies=[Depends(auth)])
+    def cost_anomalies(request: Request,
+                       z_threshold: float = 3.5) -> dict:
+        """Robust (median+MAD) per-agent spend outliers + hourly EWMA
+        spikes over the stored cost records."""
+        from agentbom_amd.api.anomaly import (
+            detect_cost_anomalies,
+            detect_temporal_cost_anomalies,
+        )
+
+        records = state.costs.list_records(_tenant_of(request))
+        spend: dict[str, float] = {}
+        for rec in records:
+            spend[rec.agent] = spend.get(rec.agent, 0.0) + rec.cost_usd
+        return {"agents": detect_cost_anomalies(spend, z_threshold),
+                "temporal": detect_temporal_cost_anomalies(records)}
+
     @app.put("/v1/costs/budget", dependencies=[Depends(auth)])
     def set_cost_budget(request: Request, payload: dict) -> dict:
         from agentbom_amd.api.cost_store import CostBudget
