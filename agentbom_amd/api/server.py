@@ -1179,6 +1179,18 @@ def create_app() -> FastAPI:
     def exceptions_audit() -> dict:
         return {"chain_valid": state.exceptions.audit_chain_valid()}
 
+    @app.get("/v1/audit/verify", dependencies=[Depends(auth)])
+    def audit_verify() -> dict:
+        """Consolidated audit-integrity verdict: every hash chain the
+        control plane maintains, verified in one read (a broken chain
+        anywhere fails the whole verdict — tamper evidence is only as
+        strong as its weakest ledger)."""
+        chains = {"exceptions": state.exceptions.audit_chain_valid()}
+        if state.identity_store is not None:
+            chains["identities"] = state.identity_store.audit_chain_valid()
+        return {"chains": chains, "all_valid": all(chains.values()),
+                "ingested_proxy_entries": len(state.audit_entries)}
+
     @app.post("/v1/costs/records", status_code=201, dependencies=[Depends(auth)])
     def ingest_costs(request: Request, payload: dict) -> dict:
         """Batch LLM cost ingest (OTel GenAI span shape)."""

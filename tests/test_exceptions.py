@@ -147,3 +147,19 @@ def test_cli_exceptions_db(tmp_path):
               if r["suppression_id"] == exc.exception_id]
     assert waived and all(r["suppression_state"] == "exception"
                           for r in waived)
+
+
+def test_consolidated_audit_verify():
+    from starlette.testclient import TestClient
+
+    from agentbom_amd.api.server import create_app
+
+    client = TestClient(create_app())
+    # exercise both chains
+    r = client.post("/v1/exceptions", json={
+        "vuln_id": "CVE-1", "package_name": "p", "reason": "r"})
+    client.post(f"/v1/exceptions/{r.json()['exception_id']}/approve")
+    client.post("/v1/identities", json={"agent_name": "bot"})
+    out = client.get("/v1/audit/verify").json()
+    assert out["all_valid"] and out["chains"]["exceptions"]
+    assert out["chains"].get("identities", True)
