@@ -77,6 +77,7 @@ class _State:
         self.oauth_as = None  # OAuthAuthorizationServer (AGENT_BOM_OAUTH_AS=1)
         self.model_keys = None  # ModelKeyBroker, created on first use
         self.hub = None  # ComplianceHub, created on first use
+        self.datasets = None  # DatasetVersionStore, created on first use
 
 
 def create_app() -> FastAPI:
@@ -1178,6 +1179,52 @@ def create_app() -> FastAPI:
     @app.get("/v1/exceptions/audit", dependencies=[Depends(auth)])
     def exceptions_audit() -> dict:
         return {"chain_valid": state.exceptions.audit_chain_valid()}
+
+    def _datasets(request: Request):
+        if state.datasets is None:
+            from agentbom_amd.api.dataset_versions import DatasetVersionStore
+
+            state.datasets = DatasetVersionStore()
+        return state.datasets
+
+    @app.post("/v1/datasets/{dataset_id}/versions", status_code=201,
+              dependencies=[Depends(auth)])
+    def register_dataset_version(request: Request, dataset_id: str,
+                                 payload: dict) -> dict:
+        from agentbom_amd.api.dataset_versions import (
+            DatasetVersionConflict,
+            DatasetVersionRecord,
+        )
+
+        if not payload.get("version_id"):
+            raise HTTPException(status_code=400, detail="version_id required")
+        rec = DatasetVersionRecord(
+            tenant_id=_tenant_of(request), dataset_id=dataset_id,
+            version_id=str(payload["version_id"]),
+            source=str(payload.get("source", "")),
+            artifact_uri=payload.get("artifact_uri"),
+            digest=payload.get("digest"),
+            digest_algorithm=str(payload.get("digest_algorithm", "sha256")),
+            metadata=dict(payload.get("metadata") or {}))
+        try:
+            return _datasets(request).put(rec).to_dict()
+        except DatasetVersionConflict as exc:
+            raise HTTPException(status_code=409, detail=str(exc))
+
+    @app.get("/v1/datasets/{dataset_id}/versions", dependencies=[Depends(auth)])
+    def list_dataset_versions(request: Request, dataset_id: str) -> dict:
+        rows = _datasets(request).list(_tenant_of(request), dataset_id)
+        return {"total": len(rows), "versions": [r.to_dict() for r in rows]}
+
+    @app.post("/v1/datasets/{dataset_id}/versions/{version_id}/verify",
+              dependencies=[Depends(auth)])
+    def verify_dataset_version(request: Request, dataset_id: str,
+                               version_id: str, payload: dict) -> dict:
+        if not payload.get("digest"):
+            raise HTTPException(status_code=400, detail="digest required")
+        return _datasets(request).verify(
+            _tenant_of(request), dataset_id, version_id,
+            str(payload["digest"]))
 
     @app.get("/v1/audit/verify", dependencies=[Depends(auth)])
     def audit_verify() -> dict:
