@@ -97,6 +97,11 @@ class Gateway:
         self.identity_store = identity_store
         self.metrics = {"relays_total": 0, "blocked_total": 0, "redactions_total": 0,
                         "breaker_rejections_total": 0}
+        # bounded relay-activity history (reference gateway_activity_store):
+        # every decision with outcome + latency, newest last
+        from collections import deque
+
+        self.activity = deque(maxlen=1000)
 
     def register(self, upstream: Upstream) -> None:
         self.upstreams[upstream.name] = upstream
@@ -109,6 +114,38 @@ class Gateway:
               principal: Optional[str] = None,
               cost: float = 1.0, access_ctx=None,
               approval_id: Optional[str] = None) -> dict[str, Any]:
+        """Relay with activity recording (outcome + error code + latency)."""
+        t0 = time.monotonic()
+        out = self._relay_inner(upstream_name, frame, principal=principal,
+                                cost=cost, access_ctx=access_ctx,
+                                approval_id=approval_id)
+        err = out.get("error") if isinstance(out, dict) else None
+        self.activity.append({
+            "ts": time.time(), "upstream": upstream_name,
+            "principal": principal or "anonymous",
+            "method": str(frame.get("method", "?")),
+            "tool": str((frame.get("params") or {}).get("name", "")),
+            "outcome": "error" if err else "ok",
+            "code": err.get("code") if err else None,
+            "latency_ms": round((time.monotonic() - t0) * 1000, 3)})
+        return out
+
+    def activity_summary(self) -> dict[str, Any]:
+        """Aggregate view over the bounded history (per-outcome/per-code)."""
+        from collections import Counter
+
+        codes = Counter(a["code"] for a in self.activity
+                        if a["code"] is not None)
+        return {"window": len(self.activity),
+                "ok": sum(1 for a in self.activity if a["outcome"] == "ok"),
+                "errors_by_code": {str(k): v for k, v in sorted(codes.items())},
+                "principals": len({a["principal"] for a in self.activity}),
+                "recent": list(self.activity)[-20:]}
+
+    def _relay_inner(self, upstream_name: str, frame: dict[str, Any],
+                     principal: Optional[str] = None,
+                     cost: float = 1.0, access_ctx=None,
+                     approval_id: Optional[str] = None) -> dict[str, Any]:
         """``access_ctx``: an identity.lifecycle.AccessContext — when present
         together with ``self.identity_store`` policies, full ABAC
         conditional-access (deny-wins, fail-closed unknowns) gates the call."""

@@ -46,3 +46,35 @@ class TestGatewayConditionalAccess:
         frame = {"jsonrpc": "2.0", "id": 3, "method": "tools/call",
                  "params": {"name": "deploy"}}
         assert gw.relay("up", frame, principal="bot").get("result") == {"ok": True}
+
+
+class TestActivityHistory:
+    def test_relay_activity_recorded_and_summarized(self):
+        from agentbom_amd.runtime.gateway import Gateway, Upstream
+
+        gw = Gateway()
+        gw.register(Upstream(name="up", handler=lambda f: {
+            "jsonrpc": "2.0", "id": f.get("id"), "result": {}}))
+        gw.relay("up", {"jsonrpc": "2.0", "id": 1, "method": "tools/list"},
+                 principal="bot")
+        gw.relay("missing", {"jsonrpc": "2.0", "id": 2,
+                             "method": "tools/call",
+                             "params": {"name": "x"}}, principal="bot")
+        assert len(gw.activity) == 2
+        first, second = list(gw.activity)
+        assert first["outcome"] == "ok" and first["latency_ms"] >= 0
+        assert second["outcome"] == "error" and second["code"] == -32001
+        s = gw.activity_summary()
+        assert s["ok"] == 1 and s["errors_by_code"] == {"-32001": 1}
+        assert s["principals"] == 1 and len(s["recent"]) == 2
+
+    def test_bounded_history(self):
+        from agentbom_amd.runtime.gateway import Gateway, Upstream
+
+        gw = Gateway()
+        gw.register(Upstream(name="up", handler=lambda f: {
+            "jsonrpc": "2.0", "id": f.get("id"), "result": {}}))
+        for i in range(1100):
+            gw.relay("up", {"jsonrpc": "2.0", "id": i,
+                            "method": "tools/list"})
+        assert len(gw.activity) == 1000  # ring buffer
