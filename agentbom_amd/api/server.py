@@ -78,6 +78,7 @@ class _State:
         self.model_keys = None  # ModelKeyBroker, created on first use
         self.hub = None  # ComplianceHub, created on first use
         self.datasets = None  # DatasetVersionStore, created on first use
+        self.evaluations = None  # EvaluationStore, created on first use
 
 
 def create_app() -> FastAPI:
@@ -1225,6 +1226,45 @@ def create_app() -> FastAPI:
         return _datasets(request).verify(
             _tenant_of(request), dataset_id, version_id,
             str(payload["digest"]))
+
+    def _evals(request: Request):
+        if state.evaluations is None:
+            from agentbom_amd.api.evaluations import EvaluationStore
+
+            state.evaluations = EvaluationStore(
+                dataset_store=_datasets(request))
+        return state.evaluations
+
+    @app.post("/v1/evaluations", status_code=201, dependencies=[Depends(auth)])
+    def record_evaluation(request: Request, payload: dict) -> dict:
+        from agentbom_amd.api.evaluations import EvaluationRun
+
+        if not payload.get("name"):
+            raise HTTPException(status_code=400, detail="name required")
+        run = EvaluationRun(
+            tenant_id=_tenant_of(request), name=str(payload["name"]),
+            status=str(payload.get("status", "completed")),
+            dataset_id=payload.get("dataset_id"),
+            dataset_version_id=payload.get("dataset_version_id"),
+            model=payload.get("model"),
+            prompt_hash=payload.get("prompt_hash"),
+            scores={str(k): float(v) for k, v in
+                    (payload.get("scores") or {}).items()},
+            cases=list(payload.get("cases") or [])[:1000],
+            metadata=dict(payload.get("metadata") or {}))
+        try:
+            return _evals(request).put(run).to_dict()
+        except ValueError as exc:
+            raise HTTPException(status_code=409, detail=str(exc))
+
+    @app.get("/v1/evaluations", dependencies=[Depends(auth)])
+    def list_evaluations(request: Request, name: Optional[str] = None) -> dict:
+        rows = _evals(request).list(_tenant_of(request), name=name)
+        return {"total": len(rows), "runs": [r.to_dict() for r in rows]}
+
+    @app.get("/v1/evaluations/compare/{name}", dependencies=[Depends(auth)])
+    def compare_evaluations(request: Request, name: str) -> dict:
+        return _evals(request).compare(_tenant_of(request), name)
 
     @app.get("/v1/audit/verify", dependencies=[Depends(auth)])
     def audit_verify() -> dict:
