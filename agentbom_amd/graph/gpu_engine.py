@@ -167,11 +167,12 @@ def expand_dedup_matches(torch, layout: dict, su, sw):
 
     Accepts UNSORTED (su, sw) — the one F-sized value sort at the end
     canonicalizes, so match_finalize can skip its pair sort (sort=False).
-    (A cross-product construction replacing the F-int64 sort with a stable
-    int32 argsort was measured SLOWER — 2.50 vs 2.38 ms/step at 10M pkgs:
-    argsort sorts (key, index) pairs internally, so nothing was saved and
-    the extra expansion kernels were pure overhead.  See
-    profiles/r02_progress.md.)"""
+    A cross-product construction replacing the int64 value sort with a
+    stable int32 argsort was A/B-measured at BOTH scales and rejected
+    (2.50 vs 2.38 ms at 10M pkgs; 18.96 vs 18.69 ms at 100M): argsort
+    internally sorts (key, index) pairs — the same 8 bytes/element the
+    packed sort moves — so nothing is saved at any size; see
+    profiles/r02_rank_kernel.md."""
     dev = su.device
     if su.numel() == 0:
         e = torch.empty(0, dtype=torch.int64, device=dev)
